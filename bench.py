@@ -1,0 +1,183 @@
+#!/usr/bin/env python3
+"""Driver bench contract — flagship serving step (BASELINE.json config 4).
+
+Workload: sklearn-RandomForest inference with RCCL batch fan-out on MI355X
+(the reference's headline metric names this config: "packaged layer MB +
+cold-start->first-inference ms, sklearn RF 1/2/4/8 GPU"). One step =
+one served batch: scatter input shards root->ranks over xGMI, gfx950 RF
+traversal kernel per rank, all_gather of the predicted labels (SURVEY.md
+§2.2 N15). Weak scaling: per-GPU shard is fixed, global batch = B * N.
+
+The JSON line's `value` is whole-job throughput (samples/s over all N GPUs);
+the metric's artifact-size and cold-start components are measured once (rank
+0, single-GPU mode) and reported alongside as `layer_mb` and
+`coldstart_first_inference_ms` (the reference publishes no numbers —
+BASELINE.md — so vs_baseline is null).
+
+Launch (driver): python -m torch.distributed.run --nnodes=1 --nproc-per-node N
+  --master-addr 127.0.0.1 --master-port P bench.py --gpus N --steps K --warmup W
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+if REPO not in sys.path:
+    sys.path.insert(0, REPO)
+
+import torch  # noqa: E402
+
+
+def measure_coldstart_and_layer(handler: str, per_gpu_batch: int) -> dict:
+    """Build (or reuse) the sklearn layer + measure one cold start on 1 GPU."""
+    from lambdipy_amd.builder import build_recipe_artifacts
+    from lambdipy_amd.packager import assemble_layer, make_zip, zip_report
+    from lambdipy_amd.resolver import parse_requirements, resolve
+    from lambdipy_amd.cli import _install_handlers
+    from lambdipy_amd.runtime import cold_start
+
+    layer_dir = os.path.join(REPO, ".lpx_bench_layer")
+    zip_path = os.path.join(REPO, ".lpx_bench_layer.zip")
+    if not os.path.exists(zip_path):
+        res = resolve(parse_requirements("scikit-learn\n"))
+        assemble_layer(res, layer_dir, wheel_builder=build_recipe_artifacts)
+        _install_handlers(layer_dir, ["rf"])
+        make_zip(layer_dir, zip_path)
+    sizes = zip_report(layer_dir, zip_path)
+    out = {
+        "layer_mb": round(sizes.unzipped_mb, 2),
+        "layer_zipped_mb": round(sizes.zipped_mb, 2),
+    }
+    if torch.cuda.is_available():
+        cs = cold_start(zip_path, handler=handler, gpus=1,
+                        env={"LPX_RF_BATCH": str(per_gpu_batch)}, timeout_s=600)
+        out["coldstart_first_inference_ms"] = round(cs.total_ms, 1)
+        out["coldstart_phases"] = cs.handler_report.get("phases", [])
+    return out
+
+
+def main() -> int:
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--batch", type=int, default=1_048_576, help="samples per GPU")
+    p.add_argument("--trees", type=int, default=256)
+    p.add_argument("--depth", type=int, default=10)
+    p.add_argument("--features", type=int, default=32)
+    p.add_argument("--classes", type=int, default=8)
+    p.add_argument("--no-coldstart", action="store_true")
+    args = p.parse_args()
+
+    from lambdipy_amd.parallel import fanout
+    from lambdipy_amd.shims.lpx_sklearn import ForestSoA
+
+    ctx = fanout.init_fanout()
+    n = ctx.world_size
+    if n == 1 and args.gpus > 1:
+        print("WARNING: --gpus > 1 but WORLD_SIZE == 1 (launch via torchrun)",
+              file=sys.stderr)
+    use_gpu = torch.cuda.is_available()
+    dev = ctx.device
+
+    # model: built on root, broadcast over RCCL (same path as the handler)
+    if ctx.is_root:
+        forest = ForestSoA.random(args.trees, args.depth, args.features,
+                                  args.classes, seed=17, device=dev)
+    else:
+        nn = 2 ** (args.depth + 1) - 1
+        nl = 2 ** args.depth
+        forest = ForestSoA(
+            torch.empty(args.trees * nn, 4, dtype=torch.int32, device=dev),
+            torch.empty(args.trees * nl, args.classes, dtype=torch.float32, device=dev),
+            torch.empty(args.trees, dtype=torch.int32, device=dev),
+            args.trees, args.classes)
+    fanout.broadcast_tensors(ctx, [forest.nodes, forest.leaf_values, forest.roots])
+
+    # synthetic global batch lives on root (weak scaling: B per GPU)
+    counts = [args.batch] * n
+    total = args.batch * n
+    X_root = None
+    if ctx.is_root:
+        g = torch.Generator(device="cpu").manual_seed(23)
+        X_root = torch.randn(total, args.features, generator=g,
+                             dtype=torch.float32).to(dev)
+
+    def step():
+        shard = fanout.scatter_rows(ctx, X_root, counts, (args.features,),
+                                    torch.float32)
+        proba = forest.predict_proba_t(shard)
+        labels = proba.argmax(dim=1).to(torch.int32)
+        all_labels = fanout.all_gather_rows(ctx, labels, counts)
+        return all_labels
+
+    def barrier_sync():
+        if n > 1:
+            import torch.distributed as dist
+            dist.barrier()
+        if use_gpu:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        step()
+    barrier_sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        out = step()
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks (slowest rank defines the job)
+    if n > 1:
+        import torch.distributed as dist
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=dev if ctx.backend == "nccl" else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    ms_per_step = elapsed / args.steps * 1e3
+    samples_per_s = total * args.steps / elapsed
+
+    if ctx.is_root:
+        extras = {}
+        if not args.no_coldstart and n == 1 and "RANK" not in os.environ:
+            try:
+                extras = measure_coldstart_and_layer("rf", args.batch)
+            except Exception as e:  # noqa: BLE001
+                extras = {"coldstart_error": str(e)[-500:]}
+        line = {
+            "metric": "sklearn_rf_fanout_samples_per_s",
+            "value": round(samples_per_s, 1),
+            "unit": "samples/s",
+            "n_gpus": n,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "f32",
+            "data": "synthetic",
+            "config": {
+                "model": f"sklearn-rf-{args.trees}x{args.depth}",
+                "global_batch": total,
+                "per_gpu_batch": args.batch,
+                "features": args.features,
+                "classes": args.classes,
+                "parallelism": f"dp{n}",
+                "baseline_metric": "packaged layer MB + cold-start->first-inference ms, sklearn RF 1/2/4/8 GPU",
+            },
+        }
+        line.update(extras)
+        print(json.dumps(line), flush=True)
+    fanout.destroy()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -1,0 +1,173 @@
+"""lpx — the lambdipy-shaped CLI.
+
+Commands mirror the reference's surface (build/publish — SURVEY.md §2.1 R1)
+plus the MI355X runtime half (run/bench). Config surface, like the
+reference's, is three tiers: CLI flags, env (LPX_*), recipe JSON (N3).
+"""
+
+from __future__ import annotations
+
+import json
+import os
+import shutil
+import sys
+
+import click
+
+from .utils.log import get_logger
+
+log = get_logger("cli")
+
+DEFAULT_RELEASE_DIR = os.environ.get(
+    "LPX_RELEASES", os.path.expanduser("~/.lpx/releases")
+)
+
+
+def _install_handlers(build_dir: str, handlers):
+    """Copy handler modules + fan-out support into the layer."""
+    pkg = os.path.dirname(os.path.abspath(__file__))
+    hsrc = os.path.join(pkg, "runtime", "handlers")
+    dest = os.path.join(build_dir, "lpx_handlers")
+    os.makedirs(dest, exist_ok=True)
+    with open(os.path.join(dest, "__init__.py"), "w") as f:
+        f.write("# lambdipy_amd layer handlers\n")
+    from .runtime.coldstart import HANDLERS
+
+    for h in handlers:
+        mod = HANDLERS.get(h, h)
+        src = os.path.join(hsrc, mod + ".py")
+        if not os.path.exists(src):
+            raise click.ClickException(f"unknown handler {h!r}")
+        shutil.copy2(src, os.path.join(dest, mod + ".py"))
+    # fan-out support ships as a top-level module
+    shutil.copy2(os.path.join(pkg, "parallel", "fanout.py"),
+                 os.path.join(build_dir, "lpx_fanout.py"))
+
+
+@click.group()
+def main():
+    """MI355X-native scientific-Python packager + runtime."""
+
+
+@main.command()
+@click.option("-r", "--requirements", default="requirements.txt",
+              type=click.Path(exists=True), help="requirements file to build")
+@click.option("-o", "--output", default="build", help="layer output directory")
+@click.option("-z", "--zip", "zip_path", default=None,
+              help="also produce a zip artifact at this path")
+@click.option("--handler", "handlers", multiple=True,
+              help="handler(s) to ship in the layer (rf, matmul, fftspmv, requests, resnet50)")
+@click.option("--no-transitive", is_flag=True, help="skip transitive resolution")
+@click.option("--json", "as_json", is_flag=True, help="print the report as JSON")
+def build(requirements, output, zip_path, handlers, no_transitive, as_json):
+    """Resolve requirements, build recipe wheels, assemble the layer."""
+    from .builder import build_recipe_artifacts
+    from .packager import assemble_layer
+    from .resolver import parse_requirements_file, resolve
+
+    reqs = parse_requirements_file(requirements)
+    res = resolve(reqs, follow_transitive=not no_transitive)
+    log.info("resolved %d deps (%d recipe hits)", len(res.deps), len(res.recipe_hits))
+    report = assemble_layer(
+        res, output, zip_path=zip_path, wheel_builder=build_recipe_artifacts
+    )
+    if handlers:
+        _install_handlers(output, handlers)
+        if zip_path:  # re-zip with handlers included
+            from .packager import make_zip, zip_report
+            make_zip(output, zip_path)
+            report.sizes = zip_report(output, zip_path)
+    d = report.to_dict()
+    if as_json:
+        click.echo(json.dumps(d, indent=1))
+    else:
+        click.echo(f"layer: {output}  ({d.get('unzipped_mb')} MB unzipped"
+                   + (f", {d.get('zipped_mb')} MB zipped" if zip_path else "") + ")")
+        for dep in d["deps"]:
+            click.echo(f"  - {dep['name']}=={dep.get('version','?')} [{dep.get('action')}]")
+
+
+@main.command()
+@click.argument("zip_path", type=click.Path(exists=True))
+@click.option("--name", default=None, help="release name (default: zip filename)")
+@click.option("--release-dir", default=DEFAULT_RELEASE_DIR, show_default=True)
+def publish(zip_path, name, release_dir):
+    """Store a built layer in the local artifact store.
+
+    Offline analog of the reference's GitHub-Releases upload:
+    artifacts land in a content-addressed local store that later builds reuse."""
+    import hashlib
+
+    os.makedirs(release_dir, exist_ok=True)
+    h = hashlib.sha256()
+    with open(zip_path, "rb") as f:
+        for chunk in iter(lambda: f.read(1 << 20), b""):
+            h.update(chunk)
+    digest = h.hexdigest()[:16]
+    base = name or os.path.basename(zip_path)
+    dest = os.path.join(release_dir, f"{base}.{digest}.zip")
+    shutil.copy2(zip_path, dest)
+    meta = {"name": base, "sha256_16": digest, "bytes": os.path.getsize(zip_path)}
+    with open(dest + ".json", "w") as f:
+        json.dump(meta, f)
+    click.echo(f"published {dest}")
+
+
+@main.command()
+@click.argument("layer", type=click.Path(exists=True))
+@click.option("--handler", default="rf", show_default=True)
+@click.option("--gpus", default=1, show_default=True)
+@click.option("--timeout", default=600.0, show_default=True)
+@click.option("--json", "as_json", is_flag=True)
+def run(layer, handler, gpus, timeout, as_json):
+    """Cold-start a handler from a layer (zip or dir) and time to first
+    inference."""
+    from .runtime import cold_start
+
+    res = cold_start(layer, handler=handler, gpus=gpus, timeout_s=timeout)
+    d = res.to_dict()
+    if as_json:
+        click.echo(json.dumps(d, indent=1))
+    else:
+        click.echo(f"cold-start->first-inference: {d['coldstart_ms']} ms "
+                   f"({handler}, {gpus} GPU)")
+
+
+@main.command()
+@click.argument("layer", type=click.Path(exists=True))
+@click.option("--handler", default="rf", show_default=True)
+@click.option("--gpus", default="1", help="comma list, e.g. 1,2,4,8")
+@click.option("--repeats", default=1, show_default=True)
+@click.option("--out", default=None, help="append JSON lines to this file")
+def bench(layer, handler, gpus, repeats, out):
+    """Cold-start scaling sweep: layer MB + cold-start ms per GPU count."""
+    from .runtime import cold_start
+
+    rows = []
+    for g in [int(x) for x in gpus.split(",")]:
+        best = None
+        for _ in range(repeats):
+            r = cold_start(layer, handler=handler, gpus=g)
+            if best is None or r.total_ms < best.total_ms:
+                best = r
+        rows.append(best.to_dict())
+        click.echo(json.dumps(rows[-1]))
+    if out:
+        with open(out, "a") as f:
+            for row in rows:
+                f.write(json.dumps(row) + "\n")
+
+
+@main.command()
+def recipes():
+    """List the recipe registry."""
+    from .recipes import get_registry
+
+    reg = get_registry()
+    for name in reg.names():
+        r = reg.get(name)
+        click.echo(f"{name:15s} {r.kind:7s} versions={r.versions} kernels={','.join(r.kernels) or '-'}")
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,124 @@
+"""Python API over the gfx950 extension.
+
+Loading contract (fail-loud): the compiled ``_lpx_ops``
+extension lives NEXT TO this file (in-tree / in-wheel, AOT gfx950). If torch
+sees a GPU and the extension is missing or broken, every op raises — there is
+deliberately no eager/CPU fallback on GPU machines, so a silently-slow path
+cannot masquerade as the native one.
+"""
+
+from __future__ import annotations
+
+import importlib
+import math
+from typing import Optional
+
+import torch
+
+_ext = None
+_ext_error: Optional[BaseException] = None
+
+try:
+    _ext = importlib.import_module("._lpx_ops", __package__)
+except Exception as e:  # noqa: BLE001 — record and re-raise at use time
+    _ext = None
+    _ext_error = e
+
+
+def extension_available() -> bool:
+    return _ext is not None
+
+
+def extension_error() -> Optional[BaseException]:
+    return _ext_error
+
+
+def require_extension():
+    if _ext is None:
+        raise ImportError(
+            "lambdipy_amd ops extension (_lpx_ops, gfx950) is not built/loadable. "
+            "Build it in-tree with `PYTORCH_ROCM_ARCH=gfx950 python setup.py "
+            f"build_ext --inplace`. Original error: {_ext_error!r}"
+        )
+    return _ext
+
+
+def ext():
+    """The raw extension module (fail-loud)."""
+    return require_extension()
+
+
+# ---- dense ---------------------------------------------------------------
+
+def gemm_bf16(A: torch.Tensor, Bt: torch.Tensor, bf16_out: bool = False) -> torch.Tensor:
+    """C = A[M,K] @ Bt[N,K]^T — NT layout, bf16 MFMA kernel."""
+    return require_extension().gemm_bf16(A.contiguous(), Bt.contiguous(), bf16_out)
+
+
+def matmul_bf16(A: torch.Tensor, B: torch.Tensor, bf16_out: bool = False) -> torch.Tensor:
+    """C = A @ B for row-major B (transposes B on device into NT layout)."""
+    return gemm_bf16(A, B.t().contiguous(), bf16_out)
+
+
+def gemv(A: torch.Tensor, x: torch.Tensor) -> torch.Tensor:
+    return require_extension().gemv(A.contiguous(), x.contiguous())
+
+
+# ---- sparse --------------------------------------------------------------
+
+def spmv_csr(indptr: torch.Tensor, indices: torch.Tensor, data: torch.Tensor,
+             x: torch.Tensor) -> torch.Tensor:
+    return require_extension().spmv_csr(
+        indptr.contiguous(), indices.contiguous(), data.contiguous(), x.contiguous()
+    )
+
+
+# ---- FFT -----------------------------------------------------------------
+
+_twiddle_cache = {}
+
+
+def _twiddles(n: int, device: torch.device) -> torch.Tensor:
+    key = (n, str(device))
+    t = _twiddle_cache.get(key)
+    if t is None:
+        k = torch.arange(n // 2, dtype=torch.float64)
+        ang = -2.0 * math.pi * k / n
+        t = torch.complex(torch.cos(ang), torch.sin(ang)).to(torch.complex64).to(device)
+        _twiddle_cache[key] = t
+    return t
+
+
+def fft(x: torch.Tensor) -> torch.Tensor:
+    """Batched 1-D FFT over the last axis (complex64, N = power of two)."""
+    x = x.contiguous()
+    n = x.shape[-1]
+    return require_extension().fft_c64(x, _twiddles(n, x.device), False)
+
+
+def ifft(x: torch.Tensor) -> torch.Tensor:
+    x = x.contiguous()
+    n = x.shape[-1]
+    return require_extension().fft_c64(x, _twiddles(n, x.device), True)
+
+
+# ---- pairwise ------------------------------------------------------------
+
+def pairwise_sqeuclidean(X: torch.Tensor, Y: torch.Tensor) -> torch.Tensor:
+    return require_extension().pairwise_sqeuclidean(X.contiguous(), Y.contiguous())
+
+
+# ---- RandomForest --------------------------------------------------------
+
+def rf_predict_proba_sum(nodes: torch.Tensor, leaf_values: torch.Tensor,
+                         roots: torch.Tensor, X: torch.Tensor) -> torch.Tensor:
+    """Per-class sums over trees: out [S, C]. Chunks classes by 8 (kernel CMAX)."""
+    e = require_extension()
+    C = leaf_values.shape[1]
+    if C <= 8:
+        return e.rf_predict(nodes, leaf_values, roots, X.contiguous())
+    outs = []
+    for c0 in range(0, C, 8):
+        lv = leaf_values[:, c0 : c0 + 8].contiguous()
+        outs.append(e.rf_predict(nodes, lv, roots, X.contiguous()))
+    return torch.cat(outs, dim=1)

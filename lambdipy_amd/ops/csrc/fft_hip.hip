@@ -1,0 +1,168 @@
+#include "hip/hip_runtime.h"
+// Batched 1-D complex64 FFT for gfx950 — Stockham radix-2 in LDS.
+//
+// Design: one workgroup per batch element; the whole
+// sequence (N <= 4096 complex64) lives in two LDS ping-pong buffers
+// (2 x 32 KiB at N=4096, within the 160 KiB/CU budget). Stockham autosort
+// needs no bit-reversal pass; log2(N) stages, each stage N/2 butterflies
+// spread over 256 threads. Twiddles are a host-precomputed [N/2] complex
+// table (exp(-2*pi*i*k/N)) read through L2/L3 (broadcast across blocks) —
+// on-device sincos would turn this memory-bound kernel VALU-bound
+// (guide Appendix B, trig-heavy ops).
+//
+// Larger N falls back to a global-memory Stockham (same recurrence, tensors
+// ping-pong in HBM) — still bandwidth-bound, log2(N) passes.
+
+#include "lpx_common.h"
+
+namespace lpx {
+
+namespace {
+
+constexpr int THREADS = 256;
+
+__device__ __forceinline__ f32x2 cmul(f32x2 a, f32x2 b) {
+  return f32x2{a.x * b.x - a.y * b.y, a.x * b.y + a.y * b.x};
+}
+
+template <bool INV>
+__global__ __launch_bounds__(THREADS) void fft_lds_kernel(
+    const float* __restrict__ xg, float* __restrict__ outg,
+    const float* __restrict__ tw, int batch, int n, int log2n) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  f32x2* buf0 = reinterpret_cast<f32x2*>(smem);
+  f32x2* buf1 = buf0 + n;
+
+  const int t = threadIdx.x;
+  const int half = n >> 1;
+
+  for (int b = blockIdx.x; b < batch; b += gridDim.x) {
+    const f32x2* src = reinterpret_cast<const f32x2*>(xg) + (long)b * n;
+    // load (coalesced 8 B/lane)
+    for (int i = t; i < n; i += THREADS) buf0[i] = src[i];
+    __syncthreads();
+
+    f32x2* x = buf0;
+    f32x2* y = buf1;
+    int s = 1;
+    for (int stage = 0; stage < log2n; ++stage) {
+      const int m = n >> (stage + 1);  // half-length at this stage
+      for (int u = t; u < half; u += THREADS) {
+        const int p = u / s;
+        const int q = u - p * s;  // u % s
+        const f32x2 a = x[q + s * p];
+        const f32x2 bv = x[q + s * (p + m)];
+        f32x2 w = reinterpret_cast<const f32x2*>(tw)[(long)p * s];
+        if (INV) w.y = -w.y;
+        y[q + s * (2 * p)] = f32x2{a.x + bv.x, a.y + bv.y};
+        y[q + s * (2 * p + 1)] = cmul(f32x2{a.x - bv.x, a.y - bv.y}, w);
+      }
+      __syncthreads();
+      f32x2* tmp = x; x = y; y = tmp;
+      s <<= 1;
+    }
+
+    f32x2* dst_lds = x;  // result lives in x after the final swap
+    f32x2* dst = reinterpret_cast<f32x2*>(outg) + (long)b * n;
+    if (INV) {
+      const float inv_n = 1.0f / (float)n;
+      for (int i = t; i < n; i += THREADS)
+        dst[i] = f32x2{dst_lds[i].x * inv_n, dst_lds[i].y * inv_n};
+    } else {
+      for (int i = t; i < n; i += THREADS) dst[i] = dst_lds[i];
+    }
+    __syncthreads();
+  }
+}
+
+// global-memory fallback for n > 4096: one stage per kernel launch
+template <bool INV>
+__global__ __launch_bounds__(THREADS) void fft_global_stage(
+    const float* __restrict__ xg, float* __restrict__ yg,
+    const float* __restrict__ tw, int batch, int n, int s, int m) {
+  const int half = n >> 1;
+  const long total = (long)batch * half;
+  for (long idx = (long)blockIdx.x * THREADS + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * THREADS) {
+    const int b = (int)(idx / half);
+    const int u = (int)(idx - (long)b * half);
+    const int p = u / s;
+    const int q = u - p * s;
+    const f32x2* x = reinterpret_cast<const f32x2*>(xg) + (long)b * n;
+    f32x2* y = reinterpret_cast<f32x2*>(yg) + (long)b * n;
+    const f32x2 a = x[q + s * p];
+    const f32x2 bv = x[q + s * (p + m)];
+    f32x2 w = reinterpret_cast<const f32x2*>(tw)[(long)p * s];
+    if (INV) w.y = -w.y;
+    y[q + s * (2 * p)] = f32x2{a.x + bv.x, a.y + bv.y};
+    y[q + s * (2 * p + 1)] = cmul(f32x2{a.x - bv.x, a.y - bv.y}, w);
+  }
+}
+
+__global__ __launch_bounds__(THREADS) void scale_kernel(float* __restrict__ d,
+                                                        long total,
+                                                        float scale) {
+  for (long i = (long)blockIdx.x * THREADS + threadIdx.x; i < total;
+       i += (long)gridDim.x * THREADS)
+    d[i] *= scale;
+}
+
+}  // namespace
+
+hipError_t fft_c64(const float* x, float* out, const float* twiddles,
+                   int batch, int n, bool inverse, hipStream_t stream) {
+  if (n <= 0 || (n & (n - 1)) != 0) return hipErrorInvalidValue;
+  int log2n = 0;
+  while ((1 << log2n) < n) ++log2n;
+  if (n == 1) {
+    return hipMemcpyAsync(out, x, (size_t)batch * 8, hipMemcpyDeviceToDevice,
+                          stream);
+  }
+  if (n <= 4096) {
+    const size_t lds = (size_t)2 * n * sizeof(f32x2);
+    unsigned blocks = batch < 8192 ? (unsigned)batch : 8192u;
+    if (inverse)
+     hipLaunchKernelGGL(( fft_lds_kernel<true>), dim3(dim3(blocks)), dim3(dim3(THREADS)), lds, stream, 
+          x, out, twiddles, batch, n, log2n);
+    else
+     hipLaunchKernelGGL(( fft_lds_kernel<false>), dim3(dim3(blocks)), dim3(dim3(THREADS)), lds, stream, 
+          x, out, twiddles, batch, n, log2n);
+    return hipGetLastError();
+  }
+  // global fallback: ping-pong between out and a scratch half... we use out
+  // and x alternately; log2n stages. Requires log2n passes; final must land
+  // in out. We bounce out <-> (float*)x is not allowed (x const). Caller
+  // provides out only, so allocate scratch here (acceptable: n>4096 path is
+  // rare and not cold-start critical).
+  float* scratch = nullptr;
+  hipError_t e = hipMallocAsync(&scratch, (size_t)batch * n * 8, stream);
+  if (e != hipSuccess) return e;
+  const float* src = x;
+  float* dsts[2] = {(log2n % 2 == 1) ? out : scratch,
+                    (log2n % 2 == 1) ? scratch : out};
+  long halfw = (long)batch * (n >> 1);
+  unsigned blocks = (unsigned)((halfw + THREADS - 1) / THREADS);
+  if (blocks > 4096u) blocks = 4096u;
+  int s = 1;
+  for (int stage = 0; stage < log2n; ++stage) {
+    const int m = n >> (stage + 1);
+    float* dst = dsts[stage % 2];
+    if (inverse)
+     hipLaunchKernelGGL(( fft_global_stage<true>), dim3(dim3(blocks)), dim3(dim3(THREADS)), 0, stream, 
+          src, dst, twiddles, batch, n, s, m);
+    else
+     hipLaunchKernelGGL(( fft_global_stage<false>), dim3(dim3(blocks)), dim3(dim3(THREADS)), 0, stream, 
+          src, dst, twiddles, batch, n, s, m);
+    src = dst;
+    s <<= 1;
+  }
+  if (inverse) {
+    long total = (long)batch * n * 2;
+   hipLaunchKernelGGL(( scale_kernel), dim3(dim3(blocks)), dim3(dim3(THREADS)), 0, stream, out, total,
+                                                             1.0f / n);
+  }
+  hipFreeAsync(scratch, stream);
+  return hipGetLastError();
+}
+
+}  // namespace lpx

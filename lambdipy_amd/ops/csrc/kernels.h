@@ -1,0 +1,54 @@
+// lambdipy_amd ops — plain C++ launcher API (no torch headers here).
+// Each launcher enqueues on the given hipStream_t and returns hipError_t.
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+
+namespace lpx {
+
+// C[M,N] (f32) = A[M,K] (bf16, row-major) @ Bt[N,K]^T (bf16, row-major).
+// NT layout: both operands K-contiguous (torch Linear weight layout).
+hipError_t gemm_bf16_nt(const uint16_t* A, const uint16_t* Bt, float* C,
+                        int M, int N, int K, hipStream_t stream);
+// Same but C stored as bf16.
+hipError_t gemm_bf16_nt_bf16out(const uint16_t* A, const uint16_t* Bt, uint16_t* C,
+                                int M, int N, int K, hipStream_t stream);
+
+// Batched GEMV: y[b,m] = sum_k A[b,m,k] * x[b,k].  bf16 in, f32 out.
+hipError_t gemv_bf16(const uint16_t* A, const uint16_t* x, float* y,
+                     int B, int M, int K, hipStream_t stream);
+// f32 variant.
+hipError_t gemv_f32(const float* A, const float* x, float* y,
+                    int B, int M, int K, hipStream_t stream);
+
+// CSR SpMV: y = A @ x (f32, int32 indices). mean_nnz picks the variant.
+hipError_t spmv_csr_f32(const int* indptr, const int* indices, const float* data,
+                        const float* x, float* y, int n_rows, long nnz,
+                        hipStream_t stream);
+
+// Batched 1-D complex64 FFT, N = power of two (<= 4096), Stockham in LDS.
+// x/out: [batch][N] interleaved (re,im) f32. twiddles: [N/2] complex
+// (host-precomputed, exp(-2*pi*i*k/N)). inverse: conjugate + 1/N scale.
+hipError_t fft_c64(const float* x, float* out, const float* twiddles,
+                   int batch, int n, bool inverse, hipStream_t stream);
+
+// Pairwise squared-euclidean: D[i,j] = ||X[i]-Y[j]||^2, f32 (exact-f32 MFMA).
+// xsq/ysq: precomputed row norms (f32). X [M][K], Y [N][K], both row-major.
+hipError_t pairwise_sqeuclidean_f32(const float* X, const float* Y,
+                                    const float* xsq, const float* ysq,
+                                    float* D, int M, int N, int K,
+                                    hipStream_t stream);
+// Row-wise squared norms: out[i] = sum_k X[i,k]^2.
+hipError_t row_sqnorm_f32(const float* X, float* out, int M, int K,
+                          hipStream_t stream);
+
+// RandomForest inference over a SoA forest (16-B nodes packed in `nodes4`:
+// {int feat; float thresh; int left; int right}, feat<0 => leaf and `left`
+// indexes leaf_values[C]). X [S][F] f32 row-major. out [S][C] f32: per-class
+// sums over trees (caller argmaxes / divides by n_trees).
+hipError_t rf_predict_f32(const int* nodes4, const float* leaf_values,
+                          const int* roots, int n_trees, const float* X,
+                          float* out, int S, int F, int C, hipStream_t stream);
+
+}  // namespace lpx

@@ -1,0 +1,208 @@
+// lambdipy_amd ops — PyTorch-ROCm bindings for the gfx950 kernel library.
+//
+// ROCm-native: streams come from c10::hip (no CUDA-named API, nothing for
+// hipify to rewrite — BASELINE.json forbids CUDA-compat shims/dual paths).
+// Tensors live on the ROCm device (torch reports it as "cuda").
+
+#include <torch/extension.h>
+
+#include <c10/hip/HIPGuard.h>
+#include <c10/hip/HIPStream.h>
+
+#include "kernels.h"
+
+namespace {
+
+#define CHECK_DEV(x)                                                        \
+  TORCH_CHECK((x).is_cuda(), #x " must be on the ROCm device (got CPU)");   \
+  TORCH_CHECK((x).is_contiguous(), #x " must be contiguous")
+
+hipStream_t current_stream(const at::Tensor& t) {
+  return c10::hip::getCurrentHIPStream(t.device().index()).stream();
+}
+
+void check_hip(hipError_t e, const char* what) {
+  TORCH_CHECK(e == hipSuccess, what, " failed: ", hipGetErrorString(e));
+}
+
+// ---- GEMM ----------------------------------------------------------------
+
+at::Tensor gemm_bf16(const at::Tensor& A, const at::Tensor& Bt, bool bf16_out) {
+  CHECK_DEV(A);
+  CHECK_DEV(Bt);
+  TORCH_CHECK(A.scalar_type() == at::kBFloat16 && Bt.scalar_type() == at::kBFloat16,
+              "gemm_bf16 expects bf16 operands");
+  TORCH_CHECK(A.dim() == 2 && Bt.dim() == 2, "gemm_bf16 expects 2-D operands");
+  const int64_t M = A.size(0), K = A.size(1), N = Bt.size(0);
+  TORCH_CHECK(Bt.size(1) == K, "gemm_bf16: A is [M,K], Bt must be [N,K] (NT layout)");
+  c10::hip::HIPGuard guard(A.device());
+  auto C = at::empty({M, N}, A.options().dtype(bf16_out ? at::kBFloat16 : at::kFloat));
+  const auto* a = reinterpret_cast<const uint16_t*>(A.data_ptr());
+  const auto* b = reinterpret_cast<const uint16_t*>(Bt.data_ptr());
+  if (bf16_out)
+    check_hip(lpx::gemm_bf16_nt_bf16out(a, b, reinterpret_cast<uint16_t*>(C.data_ptr()),
+                                        (int)M, (int)N, (int)K, current_stream(A)),
+              "gemm_bf16_nt");
+  else
+    check_hip(lpx::gemm_bf16_nt(a, b, C.data_ptr<float>(), (int)M, (int)N, (int)K,
+                                current_stream(A)),
+              "gemm_bf16_nt");
+  return C;
+}
+
+// ---- GEMV ----------------------------------------------------------------
+
+at::Tensor gemv(const at::Tensor& A_in, const at::Tensor& x_in) {
+  auto A = A_in, x = x_in;
+  CHECK_DEV(A);
+  CHECK_DEV(x);
+  bool squeeze = false;
+  if (A.dim() == 2) {  // [M,K] @ [K]
+    A = A.unsqueeze(0);
+    x = x.unsqueeze(0);
+    squeeze = true;
+  }
+  TORCH_CHECK(A.dim() == 3 && x.dim() == 2, "gemv expects A [B,M,K], x [B,K]");
+  const int64_t B = A.size(0), M = A.size(1), K = A.size(2);
+  TORCH_CHECK(x.size(0) == B && x.size(1) == K, "gemv shape mismatch");
+  c10::hip::HIPGuard guard(A.device());
+  auto y = at::empty({B, M}, A.options().dtype(at::kFloat));
+  if (A.scalar_type() == at::kBFloat16) {
+    TORCH_CHECK(x.scalar_type() == at::kBFloat16, "gemv: dtype mismatch");
+    check_hip(lpx::gemv_bf16(reinterpret_cast<const uint16_t*>(A.data_ptr()),
+                             reinterpret_cast<const uint16_t*>(x.data_ptr()),
+                             y.data_ptr<float>(), (int)B, (int)M, (int)K,
+                             current_stream(A)),
+              "gemv_bf16");
+  } else if (A.scalar_type() == at::kFloat) {
+    TORCH_CHECK(x.scalar_type() == at::kFloat, "gemv: dtype mismatch");
+    check_hip(lpx::gemv_f32(A.data_ptr<float>(), x.data_ptr<float>(),
+                            y.data_ptr<float>(), (int)B, (int)M, (int)K,
+                            current_stream(A)),
+              "gemv_f32");
+  } else {
+    TORCH_CHECK(false, "gemv supports bf16 and f32");
+  }
+  return squeeze ? y.squeeze(0) : y;
+}
+
+// ---- CSR SpMV ------------------------------------------------------------
+
+at::Tensor spmv_csr(const at::Tensor& indptr, const at::Tensor& indices,
+                    const at::Tensor& data, const at::Tensor& x) {
+  CHECK_DEV(indptr);
+  CHECK_DEV(indices);
+  CHECK_DEV(data);
+  CHECK_DEV(x);
+  TORCH_CHECK(indptr.scalar_type() == at::kInt && indices.scalar_type() == at::kInt,
+              "spmv_csr expects int32 indptr/indices");
+  TORCH_CHECK(data.scalar_type() == at::kFloat && x.scalar_type() == at::kFloat,
+              "spmv_csr expects f32 data/x");
+  const int64_t n_rows = indptr.numel() - 1;
+  c10::hip::HIPGuard guard(x.device());
+  auto y = at::empty({n_rows}, x.options());
+  check_hip(lpx::spmv_csr_f32(indptr.data_ptr<int>(), indices.data_ptr<int>(),
+                              data.data_ptr<float>(), x.data_ptr<float>(),
+                              y.data_ptr<float>(), (int)n_rows,
+                              (long)data.numel(), current_stream(x)),
+            "spmv_csr_f32");
+  return y;
+}
+
+// ---- FFT -----------------------------------------------------------------
+
+at::Tensor fft_c64(const at::Tensor& x, const at::Tensor& twiddles, bool inverse) {
+  CHECK_DEV(x);
+  CHECK_DEV(twiddles);
+  TORCH_CHECK(x.scalar_type() == at::kComplexFloat, "fft_c64 expects complex64");
+  TORCH_CHECK(twiddles.scalar_type() == at::kComplexFloat, "twiddles must be complex64");
+  const int64_t n = x.size(-1);
+  TORCH_CHECK(n >= 1 && (n & (n - 1)) == 0, "fft_c64: N must be a power of two");
+  TORCH_CHECK(twiddles.numel() >= n / 2, "twiddle table too small");
+  const int64_t batch = x.numel() / n;
+  c10::hip::HIPGuard guard(x.device());
+  auto out = at::empty_like(x);
+  check_hip(lpx::fft_c64(reinterpret_cast<const float*>(x.data_ptr()),
+                         reinterpret_cast<float*>(out.data_ptr()),
+                         reinterpret_cast<const float*>(twiddles.data_ptr()),
+                         (int)batch, (int)n, inverse, current_stream(x)),
+            "fft_c64");
+  return out;
+}
+
+// ---- pairwise ------------------------------------------------------------
+
+at::Tensor pairwise_sqeuclidean(const at::Tensor& X, const at::Tensor& Y) {
+  CHECK_DEV(X);
+  CHECK_DEV(Y);
+  TORCH_CHECK(X.scalar_type() == at::kFloat && Y.scalar_type() == at::kFloat,
+              "pairwise_sqeuclidean expects f32");
+  TORCH_CHECK(X.dim() == 2 && Y.dim() == 2 && X.size(1) == Y.size(1),
+              "pairwise_sqeuclidean expects [M,K], [N,K]");
+  const int64_t M = X.size(0), N = Y.size(0), K = X.size(1);
+  c10::hip::HIPGuard guard(X.device());
+  auto xsq = at::empty({M}, X.options());
+  auto ysq = at::empty({N}, Y.options());
+  auto D = at::empty({M, N}, X.options());
+  auto s = current_stream(X);
+  check_hip(lpx::row_sqnorm_f32(X.data_ptr<float>(), xsq.data_ptr<float>(), (int)M,
+                                (int)K, s),
+            "row_sqnorm_f32");
+  check_hip(lpx::row_sqnorm_f32(Y.data_ptr<float>(), ysq.data_ptr<float>(), (int)N,
+                                (int)K, s),
+            "row_sqnorm_f32");
+  check_hip(lpx::pairwise_sqeuclidean_f32(X.data_ptr<float>(), Y.data_ptr<float>(),
+                                          xsq.data_ptr<float>(), ysq.data_ptr<float>(),
+                                          D.data_ptr<float>(), (int)M, (int)N, (int)K, s),
+            "pairwise_sqeuclidean_f32");
+  return D;
+}
+
+// ---- RandomForest --------------------------------------------------------
+
+at::Tensor rf_predict(const at::Tensor& nodes, const at::Tensor& leaf_values,
+                      const at::Tensor& roots, const at::Tensor& X) {
+  CHECK_DEV(nodes);
+  CHECK_DEV(leaf_values);
+  CHECK_DEV(roots);
+  CHECK_DEV(X);
+  TORCH_CHECK(nodes.scalar_type() == at::kInt && nodes.dim() == 2 && nodes.size(1) == 4,
+              "nodes must be int32 [n_nodes, 4]");
+  TORCH_CHECK(leaf_values.scalar_type() == at::kFloat && leaf_values.dim() == 2,
+              "leaf_values must be f32 [n_leaf_rows, C]");
+  TORCH_CHECK(roots.scalar_type() == at::kInt, "roots must be int32");
+  TORCH_CHECK(X.scalar_type() == at::kFloat && X.dim() == 2, "X must be f32 [S,F]");
+  const int64_t S = X.size(0), F = X.size(1), C = leaf_values.size(1);
+  TORCH_CHECK(C <= 8, "rf_predict kernel handles C<=8 per call (wrapper chunks)");
+  c10::hip::HIPGuard guard(X.device());
+  auto out = at::empty({S, C}, X.options());
+  check_hip(lpx::rf_predict_f32(nodes.data_ptr<int>(), leaf_values.data_ptr<float>(),
+                                roots.data_ptr<int>(), (int)roots.numel(),
+                                X.data_ptr<float>(), out.data_ptr<float>(), (int)S,
+                                (int)F, (int)C, current_stream(X)),
+            "rf_predict_f32");
+  return out;
+}
+
+std::string build_info() {
+  return std::string("lambdipy_amd ops: gfx950 (CDNA4), HIP ") +
+         std::to_string(HIP_VERSION_MAJOR) + "." + std::to_string(HIP_VERSION_MINOR);
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "lambdipy_amd gfx950 kernel library (MFMA/LDS-tiled HIP)";
+  m.def("gemm_bf16", &gemm_bf16, "C = A[M,K] @ Bt[N,K]^T (bf16 MFMA)",
+        py::arg("A"), py::arg("Bt"), py::arg("bf16_out") = false);
+  m.def("gemv", &gemv, "y[b,m] = A[b,m,:] . x[b,:]", py::arg("A"), py::arg("x"));
+  m.def("spmv_csr", &spmv_csr, "y = CSR(A) @ x", py::arg("indptr"),
+        py::arg("indices"), py::arg("data"), py::arg("x"));
+  m.def("fft_c64", &fft_c64, "batched complex64 FFT (Stockham)", py::arg("x"),
+        py::arg("twiddles"), py::arg("inverse") = false);
+  m.def("pairwise_sqeuclidean", &pairwise_sqeuclidean,
+        "D[i,j] = ||X_i - Y_j||^2 (f32 MFMA)", py::arg("X"), py::arg("Y"));
+  m.def("rf_predict", &rf_predict, "RandomForest per-class sums over trees",
+        py::arg("nodes"), py::arg("leaf_values"), py::arg("roots"), py::arg("X"));
+  m.def("build_info", &build_info);
+}

@@ -1,0 +1,161 @@
+#include "hip/hip_runtime.h"
+// Pairwise squared-euclidean distance for gfx950 — exact-f32 MFMA.
+//
+// D[i,j] = ||X[i]||^2 - 2*X[i].Y[j] + ||Y[j]||^2
+//
+// Design: GEMM-shaped. The -2ab term runs on the f32-input
+// MFMA (v_mfma_f32_16x16x4_f32 — exact f32 at the 157 TF vector-f32 rate,
+// guide §3 'FP32-input MFMA'), with row norms fused in the epilogue. 64x64
+// tile per 256-thread block, BK=16, +1-float padded LDS rows (conflict-free
+// ds_read_b32 — guide G4). sklearn.metrics.pairwise parity: exact f32
+// numerics (bitwise an fmaf chain), no bf16 downcast.
+
+#include "lpx_common.h"
+
+namespace lpx {
+
+namespace {
+
+constexpr int BM = 64, BN = 64, BK = 16;
+constexpr int THREADS = 256;
+constexpr int ROWF = BK + 1;  // padded row, floats
+
+__device__ __forceinline__ void stage_f32(const float* __restrict__ src,
+                                          int rows_total, int K, int row0,
+                                          int k0, float* __restrict__ lds) {
+  // 64 rows x 16 f32 per tile = 4 KiB; 256 threads x 1 float4.
+  const int t = threadIdx.x;          // chunk = t: row = t/4, block = t%4
+  const int r = t >> 2, kb = (t & 3) * 4;
+  const int grow = row0 + r, gk = k0 + kb;
+  f32x4 v = {0.f, 0.f, 0.f, 0.f};
+  if (grow < rows_total) {
+    if (gk + 4 <= K) {
+      v = *reinterpret_cast<const f32x4*>(src + (long)grow * K + gk);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        if (gk + j < K) v[j] = src[(long)grow * K + gk + j];
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < 4; ++j) lds[r * ROWF + kb + j] = v[j];
+}
+
+__global__ __launch_bounds__(THREADS) void pairwise_kernel(
+    const float* __restrict__ X, const float* __restrict__ Y,
+    const float* __restrict__ xsq, const float* __restrict__ ysq,
+    float* __restrict__ D, int M, int N, int K, unsigned tiles_m,
+    unsigned tiles_n) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* const fbase = reinterpret_cast<float*>(smem);
+  // buffer layout: [X0 | X1 | Y0 | Y1], each BM*ROWF floats
+  const int BUF = BM * ROWF;
+
+  const unsigned nwg = tiles_m * tiles_n;
+  const unsigned bid = lpx_xcd_swizzle(blockIdx.x, nwg);
+  const int bm = (int)(bid / tiles_n) * BM;
+  const int bn = (int)(bid % tiles_n) * BN;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wr = wid >> 1, wc = wid & 1;  // wave covers 32x32
+
+  f32x4 acc[2][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int fr = lane & 15, fk = lane >> 4;  // fragment row, k-lane (0..3)
+
+  int cur = 0;
+  stage_f32(X, M, K, bm, 0, fbase);
+  stage_f32(Y, N, K, bn, 0, fbase + 2 * BUF);
+  __syncthreads();
+
+  const int ksteps = (K + BK - 1) / BK;
+  for (int ks = 0; ks < ksteps; ++ks) {
+    float* const xCur = fbase + cur * BUF;
+    float* const yCur = fbase + 2 * BUF + cur * BUF;
+    if (ks + 1 < ksteps) {
+      stage_f32(X, M, K, bm, (ks + 1) * BK, fbase + (cur ^ 1) * BUF);
+      stage_f32(Y, N, K, bn, (ks + 1) * BK, fbase + 2 * BUF + (cur ^ 1) * BUF);
+    }
+#pragma unroll
+    for (int kk = 0; kk < 4; ++kk) {
+      float a0 = xCur[(wr * 32 + fr) * ROWF + kk * 4 + fk];
+      float a1 = xCur[(wr * 32 + 16 + fr) * ROWF + kk * 4 + fk];
+      float b0 = yCur[(wc * 32 + fr) * ROWF + kk * 4 + fk];
+      float b1 = yCur[(wc * 32 + 16 + fr) * ROWF + kk * 4 + fk];
+      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1, acc[1][1], 0, 0, 0);
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  const int orow0 = bm + wr * 32 + (lane >> 4) * 4;
+  const int ocol0 = bn + wc * 32 + (lane & 15);
+#pragma unroll
+  for (int mf = 0; mf < 2; ++mf) {
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = orow0 + mf * 16 + r;
+        const int col = ocol0 + nf * 16;
+        if (row < M && col < N) {
+          float d = xsq[row] - 2.0f * acc[mf][nf][r] + ysq[col];
+          D[(long)row * N + col] = d > 0.f ? d : 0.f;
+        }
+      }
+    }
+  }
+}
+
+__global__ __launch_bounds__(THREADS) void row_sqnorm_kernel(
+    const float* __restrict__ X, float* __restrict__ out, int M, long K) {
+  const int lane = threadIdx.x & 63;
+  const int wave = blockIdx.x * (THREADS / 64) + (threadIdx.x >> 6);
+  const int nwaves = gridDim.x * (THREADS / 64);
+  for (int row = wave; row < M; row += nwaves) {
+    const float* r = X + (long)row * K;
+    float s = 0.f;
+    long k = (long)lane * 4;
+    const long kend = K & ~3L;
+    for (; k + 4 <= kend; k += 64 * 4) {
+      const f32x4 v = *reinterpret_cast<const f32x4*>(r + k);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) s += v[j] * v[j];
+    }
+    for (long kt = kend + lane; kt < K; kt += 64) s += r[kt] * r[kt];
+    s = lpx_wave_allreduce_sum(s);
+    if (lane == 0) out[row] = s;
+  }
+}
+
+}  // namespace
+
+hipError_t pairwise_sqeuclidean_f32(const float* X, const float* Y,
+                                    const float* xsq, const float* ysq,
+                                    float* D, int M, int N, int K,
+                                    hipStream_t stream) {
+  const unsigned tiles_m = lpx_ceil_div(M, BM), tiles_n = lpx_ceil_div(N, BN);
+  const size_t lds = sizeof(float) * 4u * BM * ROWF;
+ hipLaunchKernelGGL(( pairwise_kernel), dim3(dim3(tiles_m * tiles_n)), dim3(dim3(THREADS)), lds, stream, 
+      X, Y, xsq, ysq, D, M, N, K, tiles_m, tiles_n);
+  return hipGetLastError();
+}
+
+hipError_t row_sqnorm_f32(const float* X, float* out, int M, int K,
+                          hipStream_t stream) {
+  unsigned blocks = lpx_ceil_div(M, THREADS / 64);
+  if (blocks > 2048u) blocks = 2048u;
+ hipLaunchKernelGGL(( row_sqnorm_kernel), dim3(dim3(blocks ? blocks : 1)), dim3(dim3(THREADS)), 0, stream, 
+      X, out, M, (long)K);
+  return hipGetLastError();
+}
+
+}  // namespace lpx

@@ -1,0 +1,122 @@
+// RandomForest inference for gfx950 — SoA forest traversal.
+//
+// THE metric workload (BASELINE.json config 4). Design (SURVEY.md §2.2 N9 +
+// §7 hard-part 4 'RF traversal on SIMD-64'):
+//   - forest in SoA: one int4 per node {feat, thresh-bits, left, right};
+//     feat < 0 marks a leaf and `left` indexes the leaf-value table row.
+//     16-B nodes -> one global_load_dwordx4 gather per step; the whole forest
+//     (a few MB) is L2/L3-resident, so gathers are cache-served.
+//   - one lane = one sample; a 256-thread block owns a 256-sample tile whose
+//     feature rows are staged in LDS ([256][F+1] f32, +1 pad to spread banks)
+//     when F <= 63, so the per-step feature gather never re-touches HBM.
+//   - two trees walked concurrently per lane (independent pointer-chase
+//     chains -> 2 node loads in flight, hides L2 latency).
+//   - per-class accumulators live in registers (compile-time CMAX=8 unroll;
+//     the Python wrapper chunks wider classifiers over class groups).
+//   - divergence: lanes exit at different depths; finished chains idle until
+//     the wave's while-condition clears — depth-synchronous enough for
+//     similar-depth CART trees.
+
+#include "lpx_common.h"
+
+namespace lpx {
+
+namespace {
+
+constexpr int THREADS = 256;
+constexpr int CMAX = 8;
+constexpr int F_LDS_MAX = 63;  // 256*(63+1)*4 = 64 KiB LDS
+
+template <bool STAGE_X>
+__global__ __launch_bounds__(THREADS) void rf_kernel(
+    const i32x4* __restrict__ nodes, const float* __restrict__ leaf_values,
+    const int* __restrict__ roots, int n_trees, const float* __restrict__ X,
+    float* __restrict__ out, int S, int F, int C) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* xl = reinterpret_cast<float*>(smem);  // [256][F+1] when STAGE_X
+  const int t = threadIdx.x;
+  const int rowf = F + 1;
+
+  for (int tile = blockIdx.x * THREADS; tile < S; tile += gridDim.x * THREADS) {
+    const int s = tile + t;
+    const int tile_n = min(THREADS, S - tile);
+    if (STAGE_X) {
+      // cooperative coalesced load of the tile's rows
+      const long total = (long)tile_n * F;
+      for (long i = t; i < total; i += THREADS) {
+        const int r = (int)(i / F), c = (int)(i - (long)r * F);
+        xl[r * rowf + c] = X[((long)tile + r) * F + c];
+      }
+      __syncthreads();
+    }
+
+    if (s < S) {
+      const float* xrow = STAGE_X ? (xl + t * rowf) : (X + (long)s * F);
+      float acc[CMAX];
+#pragma unroll
+      for (int c = 0; c < CMAX; ++c) acc[c] = 0.f;
+
+      for (int tr = 0; tr < n_trees; tr += 2) {
+        int nd0 = roots[tr];
+        int nd1 = (tr + 1 < n_trees) ? roots[tr + 1] : -1;
+        while (nd0 >= 0 || nd1 >= 0) {
+          if (nd0 >= 0) {
+            const i32x4 n0 = nodes[nd0];
+            if (n0.x >= 0) {
+              const float xv = xrow[n0.x];
+              nd0 = (xv <= __int_as_float(n0.y)) ? n0.z : n0.w;
+            } else {
+              const float* lv = leaf_values + (long)n0.z * C;
+#pragma unroll
+              for (int c = 0; c < CMAX; ++c)
+                if (c < C) acc[c] += lv[c];
+              nd0 = -1;
+            }
+          }
+          if (nd1 >= 0) {
+            const i32x4 n1 = nodes[nd1];
+            if (n1.x >= 0) {
+              const float xv = xrow[n1.x];
+              nd1 = (xv <= __int_as_float(n1.y)) ? n1.z : n1.w;
+            } else {
+              const float* lv = leaf_values + (long)n1.z * C;
+#pragma unroll
+              for (int c = 0; c < CMAX; ++c)
+                if (c < C) acc[c] += lv[c];
+              nd1 = -1;
+            }
+          }
+        }
+      }
+#pragma unroll
+      for (int c = 0; c < CMAX; ++c)
+        if (c < C) out[(long)s * C + c] = acc[c];
+    }
+    if (STAGE_X) __syncthreads();
+  }
+}
+
+}  // namespace
+
+hipError_t rf_predict_f32(const int* nodes4, const float* leaf_values,
+                          const int* roots, int n_trees, const float* X,
+                          float* out, int S, int F, int C,
+                          hipStream_t stream) {
+  if (C > CMAX) return hipErrorInvalidValue;  // Python wrapper chunks classes
+  unsigned blocks = lpx_ceil_div(S, THREADS);
+  if (blocks > 4096u) blocks = 4096u;
+  if (blocks == 0) blocks = 1;
+  const bool stage = F <= F_LDS_MAX;
+  const size_t lds = stage ? (size_t)THREADS * (F + 1) * sizeof(float) : 0;
+  if (stage)
+    rf_kernel<true><<<dim3(blocks), dim3(THREADS), lds, stream>>>(
+        reinterpret_cast<const i32x4*>(nodes4), leaf_values, roots, n_trees, X,
+        out, S, F, C);
+  else
+    rf_kernel<false><<<dim3(blocks), dim3(THREADS), 0, stream>>>(
+        reinterpret_cast<const i32x4*>(nodes4), leaf_values, roots, n_trees, X,
+        out, S, F, C);
+  return hipGetLastError();
+}
+
+}  // namespace lpx

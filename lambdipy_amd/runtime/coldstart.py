@@ -1,0 +1,160 @@
+"""Cold-start harness — the metric path.
+
+Times unzip -> fresh-interpreter spawn -> import -> hipInit -> model ->
+first inference, per GPU count. t0 is taken BEFORE the unzip; t1 when rank 0
+prints its LPX_RESULT line. For g > 1 the harness spawns g processes itself
+(one per GPU, RANK/WORLD_SIZE/MASTER_* env) — no torchrun wrapper overhead
+in the timed region.
+"""
+
+from __future__ import annotations
+
+import json
+import os
+import shutil
+import socket
+import subprocess
+import sys
+import tempfile
+import time
+import zipfile
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional
+
+from ..utils.log import get_logger
+
+log = get_logger("runtime")
+
+HANDLERS = {
+    "requests": "requests_handler",
+    "matmul": "matmul_handler",
+    "fftspmv": "fftspmv_handler",
+    "rf": "rf_handler",
+    "resnet50": "resnet50_handler",
+}
+
+
+@dataclass
+class ColdStartResult:
+    handler: str
+    gpus: int
+    total_ms: float  # t0 (pre-unzip) -> rank-0 LPX_RESULT
+    unzip_ms: float
+    handler_report: dict
+    layer_mb: Optional[float] = None
+    returncodes: List[int] = field(default_factory=list)
+
+    def to_dict(self) -> dict:
+        return {
+            "handler": self.handler,
+            "gpus": self.gpus,
+            "coldstart_ms": round(self.total_ms, 1),
+            "unzip_ms": round(self.unzip_ms, 1),
+            "layer_mb": self.layer_mb,
+            "handler_report": self.handler_report,
+        }
+
+
+def _free_port() -> int:
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    p = s.getsockname()[1]
+    s.close()
+    return p
+
+
+def cold_start(
+    layer: str,
+    handler: str = "rf",
+    gpus: int = 1,
+    *,
+    env: Optional[Dict[str, str]] = None,
+    timeout_s: float = 600.0,
+    keep_dir: bool = False,
+) -> ColdStartResult:
+    """Run one cold start of `handler` from `layer` (zip path or dir)."""
+    mod = HANDLERS.get(handler, handler)
+    layer_mb = None
+    t0 = time.perf_counter()
+
+    tmpdir = None
+    if os.path.isfile(layer) and layer.endswith(".zip"):
+        layer_mb = os.path.getsize(layer) / 2**20
+        tmpdir = tempfile.mkdtemp(prefix="lpx-layer-")
+        with zipfile.ZipFile(layer) as zf:
+            zf.extractall(tmpdir)
+        layer_dir = tmpdir
+    else:
+        layer_dir = layer
+    unzip_ms = (time.perf_counter() - t0) * 1e3
+
+    base_env = dict(os.environ)
+    if env:
+        base_env.update(env)
+    # the layer provides its packages; the image provides python + torch
+    base_env["PYTHONPATH"] = layer_dir + (
+        os.pathsep + base_env["PYTHONPATH"] if base_env.get("PYTHONPATH") else ""
+    )
+    base_env.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+
+    code = f"import lpx_handlers.{mod} as h; h.main()"
+    procs = []
+    if gpus <= 1:
+        procs.append(
+            subprocess.Popen([sys.executable, "-c", code], env=base_env,
+                             stdout=subprocess.PIPE, stderr=subprocess.PIPE,
+                             text=True)
+        )
+    else:
+        port = _free_port()
+        for r in range(gpus):
+            e = dict(base_env)
+            e.update(
+                RANK=str(r), LOCAL_RANK=str(r), WORLD_SIZE=str(gpus),
+                MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
+            )
+            procs.append(
+                subprocess.Popen([sys.executable, "-c", code], env=e,
+                                 stdout=subprocess.PIPE, stderr=subprocess.PIPE,
+                                 text=True)
+            )
+
+    report = {}
+    total_ms = None
+    try:
+        out0, err0 = procs[0].communicate(timeout=timeout_s)
+        total_ms = (time.perf_counter() - t0) * 1e3
+        for line in out0.splitlines():
+            if line.startswith("LPX_RESULT "):
+                report = json.loads(line[len("LPX_RESULT "):])
+                break
+        if not report:
+            raise RuntimeError(
+                f"handler produced no LPX_RESULT (rc={procs[0].returncode})\n"
+                f"stdout:\n{out0[-4000:]}\nstderr:\n{err0[-4000:]}"
+            )
+        rcs = [procs[0].returncode]
+        for p in procs[1:]:
+            try:
+                p.communicate(timeout=60)
+            except subprocess.TimeoutExpired:
+                p.kill()
+            rcs.append(p.returncode)
+    finally:
+        for p in procs:
+            if p.poll() is None:
+                p.kill()
+        if tmpdir and not keep_dir:
+            shutil.rmtree(tmpdir, ignore_errors=True)
+
+    if procs[0].returncode != 0:
+        raise RuntimeError(f"handler rank 0 exited {procs[0].returncode}")
+    return ColdStartResult(
+        handler=handler,
+        gpus=gpus,
+        total_ms=total_ms,
+        unzip_ms=unzip_ms,
+        handler_report=report,
+        layer_mb=round(layer_mb, 2) if layer_mb is not None else None,
+        returncodes=rcs,
+    )

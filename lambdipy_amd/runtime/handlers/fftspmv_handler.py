@@ -1,0 +1,68 @@
+"""Config-3 handler: batched FFT + CSR SpMV on gfx950 (numpy+scipy layer)."""
+
+from __future__ import annotations
+
+import json
+import os
+import time
+
+
+def main() -> dict:
+    t0 = time.perf_counter()
+    phases = []
+
+    def mark(name, start):
+        phases.append({"phase": name, "dur_ms": (time.perf_counter() - start) * 1e3})
+
+    t = time.perf_counter()
+    import torch
+    mark("import_torch", t)
+
+    t = time.perf_counter()
+    try:
+        import lpx_scipy
+    except ImportError:
+        from lambdipy_amd.shims import lpx_scipy
+    mark("import_shim", t)
+
+    batch = int(os.environ.get("LPX_FFT_BATCH", "1024"))
+    n = int(os.environ.get("LPX_FFT_N", "1024"))
+    rows = int(os.environ.get("LPX_SPMV_ROWS", "262144"))
+    nnz_per_row = int(os.environ.get("LPX_SPMV_NNZ", "32"))
+
+    t = time.perf_counter()
+    dev = torch.device("cuda", 0)
+    torch.manual_seed(0)
+    x = torch.randn(batch, n, dtype=torch.float32, device=dev)
+    xc = torch.complex(x, torch.zeros_like(x))
+    # synthetic CSR: fixed nnz/row, random columns
+    idx = torch.randint(0, rows, (rows, nnz_per_row), device=dev, dtype=torch.int32)
+    idx, _ = torch.sort(idx, dim=1)
+    indices = idx.flatten().contiguous()
+    indptr = torch.arange(0, (rows + 1) * nnz_per_row, nnz_per_row,
+                          device=dev, dtype=torch.int32)
+    data = torch.randn(rows * nnz_per_row, dtype=torch.float32, device=dev)
+    vec = torch.randn(rows, dtype=torch.float32, device=dev)
+    torch.cuda.synchronize()
+    mark("hip_init_and_data", t)
+
+    t = time.perf_counter()
+    f = lpx_scipy.fft_t(xc)
+    y = lpx_scipy.csr_matvec_t(indptr, indices, data, vec)
+    torch.cuda.synchronize()
+    mark("first_inference", t)
+
+    out = {
+        "handler": "fftspmv",
+        "fft": {"batch": batch, "n": n},
+        "spmv": {"rows": rows, "nnz_per_row": nnz_per_row},
+        "checksum": float(f.real.sum() + y.sum()),
+        "total_ms": (time.perf_counter() - t0) * 1e3,
+        "phases": phases,
+    }
+    print("LPX_RESULT " + json.dumps(out), flush=True)
+    return out
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,55 @@
+import json
+import os
+
+import pytest
+from click.testing import CliRunner
+
+from lambdipy_amd.cli import main as cli_main
+
+
+@pytest.fixture()
+def runner():
+    return CliRunner()
+
+
+def test_recipes_listing(runner):
+    r = runner.invoke(cli_main, ["recipes"])
+    assert r.exit_code == 0, r.output
+    assert "numpy" in r.output and "scikit-learn" in r.output
+
+
+def test_build_and_run_requests_layer(runner, tmp_path):
+    """Config 1 end-to-end: build -> zip -> cold-start (CPU only)."""
+    req = tmp_path / "requirements.txt"
+    req.write_text("requests\n")
+    out = str(tmp_path / "build")
+    zp = str(tmp_path / "layer.zip")
+    r = runner.invoke(
+        cli_main,
+        ["build", "-r", str(req), "-o", out, "-z", zp,
+         "--handler", "requests", "--json"],
+    )
+    assert r.exit_code == 0, r.output
+    rep = json.loads(r.output[r.output.index("{"):])
+    assert rep["within_250mb_unzipped"]
+    assert os.path.exists(os.path.join(out, "lpx_handlers", "requests_handler.py"))
+    assert os.path.exists(os.path.join(out, "lpx_fanout.py"))
+
+    from lambdipy_amd.runtime import cold_start
+
+    res = cold_start(zp, handler="requests", gpus=1, timeout_s=300)
+    assert res.handler_report["handler"] == "requests"
+    assert res.total_ms > 0
+    assert res.layer_mb is not None and res.layer_mb < 10
+
+
+def test_publish(runner, tmp_path):
+    zp = tmp_path / "x.zip"
+    import zipfile
+
+    with zipfile.ZipFile(zp, "w") as zf:
+        zf.writestr("a.txt", "hi")
+    rel = tmp_path / "releases"
+    r = runner.invoke(cli_main, ["publish", str(zp), "--release-dir", str(rel)])
+    assert r.exit_code == 0, r.output
+    assert any(f.endswith(".zip") for f in os.listdir(rel))

@@ -1,0 +1,93 @@
+"""ForestSoA conversion correctness — validated on CPU against sklearn
+(the GPU kernel is validated against this same SoA via a numpy traversal
+reference; see test_ops_gpu.py for the on-device check)."""
+
+import numpy as np
+import pytest
+import torch
+
+from lambdipy_amd.shims.lpx_sklearn import ForestSoA
+
+
+def soa_predict_proba_cpu(f: ForestSoA, X: np.ndarray) -> np.ndarray:
+    """Reference traversal of the SoA arrays (mirrors the HIP kernel)."""
+    nodes = f.nodes.cpu().numpy()
+    thr_f = np.ascontiguousarray(nodes[:, 1]).view(np.float32)
+    leaves = f.leaf_values.cpu().numpy()
+    roots = f.roots.cpu().numpy()
+    S = X.shape[0]
+    out = np.zeros((S, f.n_classes), dtype=np.float64)
+    for s in range(S):
+        for r in roots:
+            nd = int(r)
+            while nodes[nd, 0] >= 0:
+                feat = nodes[nd, 0]
+                nd = int(nodes[nd, 2]) if X[s, feat] <= thr_f[nd] else int(nodes[nd, 3])
+            out[s] += leaves[nodes[nd, 2]]
+    return (out / f.n_trees).astype(np.float32)
+
+
+@pytest.fixture(scope="module")
+def fitted_forest():
+    sklearn = pytest.importorskip("sklearn")
+    from sklearn.ensemble import RandomForestClassifier
+
+    rng = np.random.default_rng(0)
+    X = rng.standard_normal((400, 12)).astype(np.float32)
+    y = (X[:, 0] + X[:, 3] * X[:, 5] > 0).astype(int) + (X[:, 2] > 1).astype(int)
+    clf = RandomForestClassifier(n_estimators=12, max_depth=6, random_state=0)
+    clf.fit(X, y)
+    return clf, X
+
+
+def test_from_sklearn_matches_predict_proba(fitted_forest):
+    clf, X = fitted_forest
+    f = ForestSoA.from_sklearn(clf, device=None)
+    got = soa_predict_proba_cpu(f, X[:50])
+    want = clf.predict_proba(X[:50]).astype(np.float32)
+    np.testing.assert_allclose(got, want, rtol=1e-5, atol=1e-6)
+
+
+def test_from_sklearn_predict_classes(fitted_forest):
+    clf, X = fitted_forest
+    f = ForestSoA.from_sklearn(clf, device=None)
+    proba = soa_predict_proba_cpu(f, X[:100])
+    got = np.asarray(f.classes)[proba.argmax(axis=1)]
+    want = clf.predict(X[:100])
+    assert (got == want).mean() > 0.99  # ties may break differently
+
+
+def test_random_forest_shapes():
+    f = ForestSoA.random(n_trees=4, depth=3, n_features=8, n_classes=3,
+                         seed=1, device=None)
+    n_nodes_per_tree = 2**4 - 1
+    assert f.nodes.shape == (4 * n_nodes_per_tree, 4)
+    assert f.leaf_values.shape == (4 * 8, 3)
+    assert f.roots.tolist() == [0, 15, 30, 45]
+    # leaf distributions normalized
+    s = f.leaf_values.sum(dim=1)
+    assert torch.allclose(s, torch.ones_like(s), atol=1e-5)
+
+
+def test_random_forest_traversal_terminates():
+    f = ForestSoA.random(n_trees=2, depth=4, n_features=5, n_classes=2,
+                         seed=2, device=None)
+    X = np.random.default_rng(3).standard_normal((16, 5)).astype(np.float32)
+    proba = soa_predict_proba_cpu(f, X)
+    np.testing.assert_allclose(proba.sum(axis=1), 1.0, rtol=1e-5)
+
+
+def test_regressor_conversion():
+    sklearn = pytest.importorskip("sklearn")
+    from sklearn.ensemble import RandomForestRegressor
+
+    rng = np.random.default_rng(0)
+    X = rng.standard_normal((300, 6)).astype(np.float32)
+    y = X[:, 0] * 2 + np.sin(X[:, 1])
+    reg = RandomForestRegressor(n_estimators=8, max_depth=5, random_state=0)
+    reg.fit(X, y)
+    f = ForestSoA.from_sklearn(reg, device=None)
+    assert f.n_classes == 1
+    got = soa_predict_proba_cpu(f, X[:40])[:, 0]
+    want = reg.predict(X[:40]).astype(np.float32)
+    np.testing.assert_allclose(got, want, rtol=1e-4, atol=1e-5)

@@ -1,0 +1,214 @@
+"""GPU numerics tests: every gfx950 kernel vs a plain PyTorch fp32 reference
+(asymmetric operands — transpose-detecting, guide §5.4 rule 16) plus
+determinism double-runs. Run on the MI355X box: pytest tests -m gpu."""
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda"
+
+
+@pytest.fixture(scope="module")
+def ops():
+    from lambdipy_amd import ops as _ops
+
+    # fail-loud contract: on a GPU box the extension must be present
+    assert _ops.extension_available(), f"extension missing: {_ops.extension_error()}"
+    return _ops
+
+
+def _rand(shape, dtype=torch.float32, seed=0):
+    g = torch.Generator(device="cpu").manual_seed(seed)
+    return torch.randn(*shape, generator=g, dtype=torch.float32).to(DEV).to(dtype)
+
+
+# ---- GEMM ----------------------------------------------------------------
+
+@pytest.mark.parametrize("m,n,k", [(256, 256, 256), (128, 128, 64),
+                                   (512, 384, 320), (100, 130, 70),
+                                   (4096, 4096, 4096)])
+def test_gemm_bf16_vs_torch(ops, m, n, k):
+    A = _rand((m, k), torch.bfloat16, seed=m + n)
+    Bt = _rand((n, k), torch.bfloat16, seed=m * 3 + k)
+    C = ops.gemm_bf16(A, Bt)
+    ref = (A.float() @ Bt.float().t())
+    err = (C - ref).abs().max().item()
+    scale = ref.abs().max().item() + 1e-6
+    assert err / scale < 2e-2, f"rel err {err/scale}"
+
+
+def test_gemm_bf16_out_dtype(ops):
+    A = _rand((128, 128), torch.bfloat16)
+    Bt = _rand((128, 128), torch.bfloat16, seed=9)
+    C = ops.gemm_bf16(A, Bt, bf16_out=True)
+    assert C.dtype == torch.bfloat16
+    ref = (A.float() @ Bt.float().t())
+    assert (C.float() - ref).abs().max().item() / (ref.abs().max().item() + 1e-6) < 3e-2
+
+
+def test_gemm_deterministic(ops):
+    A = _rand((512, 512), torch.bfloat16, seed=4)
+    Bt = _rand((512, 512), torch.bfloat16, seed=5)
+    c1 = ops.gemm_bf16(A, Bt)
+    c2 = ops.gemm_bf16(A, Bt)
+    assert torch.equal(c1, c2)
+
+
+# ---- GEMV ----------------------------------------------------------------
+
+@pytest.mark.parametrize("b,m,k", [(1, 256, 1024), (8, 512, 4096), (3, 100, 77)])
+def test_gemv_bf16(ops, b, m, k):
+    A = _rand((b, m, k), torch.bfloat16, seed=b)
+    x = _rand((b, k), torch.bfloat16, seed=b + 1)
+    y = ops.gemv(A, x)
+    ref = torch.einsum("bmk,bk->bm", A.float(), x.float())
+    err = (y - ref).abs().max().item() / (ref.abs().max().item() + 1e-6)
+    assert err < 2e-2, err
+
+
+def test_gemv_f32_2d(ops):
+    A = _rand((300, 500), torch.float32, seed=7)
+    x = _rand((500,), torch.float32, seed=8)
+    y = ops.gemv(A, x)
+    ref = A @ x
+    assert torch.allclose(y, ref, rtol=1e-4, atol=1e-4)
+
+
+# ---- SpMV ----------------------------------------------------------------
+
+def _random_csr(rows, cols, nnz_per_row, seed=0):
+    rng = np.random.default_rng(seed)
+    indices = np.sort(
+        rng.integers(0, cols, (rows, nnz_per_row)).astype(np.int32), axis=1
+    ).reshape(-1)
+    indptr = np.arange(0, (rows + 1) * nnz_per_row, nnz_per_row, dtype=np.int32)
+    data = rng.standard_normal(rows * nnz_per_row).astype(np.float32)
+    return indptr, indices, data
+
+
+@pytest.mark.parametrize("rows,cols,nnz", [(1000, 800, 7), (512, 512, 100),
+                                           (3000, 3000, 33)])
+def test_spmv_vs_dense(ops, rows, cols, nnz):
+    indptr, indices, data = _random_csr(rows, cols, nnz, seed=rows)
+    x = np.random.default_rng(1).standard_normal(cols).astype(np.float32)
+    ti = torch.from_numpy(indptr).to(DEV)
+    tj = torch.from_numpy(indices).to(DEV)
+    td = torch.from_numpy(data).to(DEV)
+    tx = torch.from_numpy(x).to(DEV)
+    y = ops.spmv_csr(ti, tj, td, tx).cpu().numpy()
+    # dense reference
+    dense = np.zeros((rows, cols), dtype=np.float64)
+    for r in range(rows):
+        for j in range(indptr[r], indptr[r + 1]):
+            dense[r, indices[j]] += data[j]
+    ref = dense @ x.astype(np.float64)
+    np.testing.assert_allclose(y, ref, rtol=1e-4, atol=1e-4)
+
+
+# ---- FFT -----------------------------------------------------------------
+
+@pytest.mark.parametrize("batch,n", [(4, 256), (16, 1024), (2, 4096),
+                                     (3, 8192), (128, 64), (1, 2)])
+def test_fft_vs_torch(ops, batch, n):
+    g = torch.Generator(device="cpu").manual_seed(n)
+    x = (torch.randn(batch, n, generator=g) +
+         1j * torch.randn(batch, n, generator=g)).to(torch.complex64).to(DEV)
+    got = ops.fft(x)
+    ref = torch.fft.fft(x.to(torch.complex128))
+    err = (got.to(torch.complex128) - ref).abs().max().item()
+    scale = ref.abs().max().item()
+    assert err / scale < 1e-4, err / scale
+
+
+def test_ifft_roundtrip(ops):
+    g = torch.Generator(device="cpu").manual_seed(0)
+    x = (torch.randn(8, 512, generator=g) +
+         1j * torch.randn(8, 512, generator=g)).to(torch.complex64).to(DEV)
+    rt = ops.ifft(ops.fft(x))
+    assert (rt - x).abs().max().item() < 1e-3
+
+
+# ---- pairwise ------------------------------------------------------------
+
+@pytest.mark.parametrize("m,n,k", [(128, 128, 32), (500, 300, 64), (64, 64, 7),
+                                   (1000, 1000, 128)])
+def test_pairwise_vs_cdist(ops, m, n, k):
+    X = _rand((m, k), seed=m)
+    Y = _rand((n, k), seed=n + 1)
+    D = ops.pairwise_sqeuclidean(X, Y)
+    ref = torch.cdist(X.double(), Y.double()) ** 2
+    err = (D.double() - ref).abs().max().item()
+    assert err < 1e-2 * (ref.max().item() + 1), err
+
+
+# ---- RandomForest --------------------------------------------------------
+
+def test_rf_vs_cpu_reference(ops):
+    from lambdipy_amd.shims.lpx_sklearn import ForestSoA
+    from tests.test_forest import soa_predict_proba_cpu
+
+    f = ForestSoA.random(n_trees=31, depth=8, n_features=20, n_classes=3,
+                         seed=11, device=None)
+    rng = np.random.default_rng(5)
+    X = rng.standard_normal((500, 20)).astype(np.float32)
+    ref = soa_predict_proba_cpu(f, X)
+    fg = f.to(DEV)
+    got = fg.predict_proba_t(torch.from_numpy(X).to(DEV)).cpu().numpy()
+    np.testing.assert_allclose(got, ref, rtol=1e-4, atol=1e-5)
+
+
+def test_rf_vs_sklearn(ops):
+    sklearn = pytest.importorskip("sklearn")
+    from sklearn.ensemble import RandomForestClassifier
+    from lambdipy_amd.shims.lpx_sklearn import ForestSoA
+
+    rng = np.random.default_rng(0)
+    X = rng.standard_normal((600, 16)).astype(np.float32)
+    y = ((X[:, 0] > 0) & (X[:, 2] < 1)).astype(int) + (X[:, 5] > 0.5)
+    clf = RandomForestClassifier(n_estimators=20, max_depth=8, random_state=0)
+    clf.fit(X, y)
+    f = ForestSoA.from_sklearn(clf, device=DEV)
+    got = f.predict_proba(X)
+    want = clf.predict_proba(X).astype(np.float32)
+    np.testing.assert_allclose(got, want, rtol=1e-4, atol=1e-5)
+
+
+def test_rf_many_classes_chunking(ops):
+    from lambdipy_amd.shims.lpx_sklearn import ForestSoA
+    from tests.test_forest import soa_predict_proba_cpu
+
+    f = ForestSoA.random(n_trees=6, depth=5, n_features=10, n_classes=11,
+                         seed=3, device=None)
+    rng = np.random.default_rng(6)
+    X = rng.standard_normal((100, 10)).astype(np.float32)
+    ref = soa_predict_proba_cpu(f, X)
+    got = f.to(DEV).predict_proba_t(torch.from_numpy(X).to(DEV)).cpu().numpy()
+    np.testing.assert_allclose(got, ref, rtol=1e-4, atol=1e-5)
+
+
+def test_rf_wide_features_no_lds_path(ops):
+    """F > 63 exercises the non-staged (global X) kernel variant."""
+    from lambdipy_amd.shims.lpx_sklearn import ForestSoA
+    from tests.test_forest import soa_predict_proba_cpu
+
+    f = ForestSoA.random(n_trees=5, depth=6, n_features=100, n_classes=2,
+                         seed=9, device=None)
+    rng = np.random.default_rng(10)
+    X = rng.standard_normal((200, 100)).astype(np.float32)
+    ref = soa_predict_proba_cpu(f, X)
+    got = f.to(DEV).predict_proba_t(torch.from_numpy(X).to(DEV)).cpu().numpy()
+    np.testing.assert_allclose(got, ref, rtol=1e-4, atol=1e-5)
+
+
+def test_rf_deterministic(ops):
+    from lambdipy_amd.shims.lpx_sklearn import ForestSoA
+
+    f = ForestSoA.random(n_trees=64, depth=10, n_features=32, n_classes=8,
+                         seed=1, device=DEV)
+    X = _rand((4096, 32))
+    p1 = f.predict_proba_t(X)
+    p2 = f.predict_proba_t(X)
+    assert torch.equal(p1, p2)

@@ -1,0 +1,50 @@
+import pytest
+
+from lambdipy_amd.recipes import BuildRecipe, RecipeError, get_registry
+
+
+def test_registry_loads_core_set():
+    reg = get_registry()
+    assert {"requests", "numpy", "scipy", "scikit-learn", "torch"} <= set(reg.names())
+
+
+def test_version_matching():
+    reg = get_registry()
+    assert reg.match("numpy", "1.26.4") is not None
+    assert reg.match("numpy", "0.9") is None
+    assert reg.match("nonexistent", "1.0") is None
+
+
+def test_native_recipes_have_kernels_and_shims():
+    reg = get_registry()
+    for name in ("numpy", "scipy", "scikit-learn", "torch"):
+        r = reg.get(name)
+        assert r.kind == "native"
+        assert r.kernels, name
+        assert r.shim, name
+
+
+def test_recipe_schema_validation():
+    with pytest.raises(RecipeError):
+        BuildRecipe.from_dict({"recipe_format": 1, "name": "x", "versions": ">=1"})
+    with pytest.raises(RecipeError):
+        BuildRecipe.from_dict(
+            {"recipe_format": 99, "name": "x", "versions": ">=1", "kind": "pure"}
+        )
+    with pytest.raises(RecipeError):
+        BuildRecipe.from_dict(
+            {"recipe_format": 1, "name": "x", "versions": ">=1", "kind": "weird"}
+        )
+    # native without kernels
+    with pytest.raises(RecipeError):
+        BuildRecipe.from_dict(
+            {"recipe_format": 1, "name": "x", "versions": ">=1", "kind": "native"}
+        )
+
+
+def test_provides_defaults_to_name():
+    r = BuildRecipe.from_dict(
+        {"recipe_format": 1, "name": "Foo_Bar", "versions": ">=1", "kind": "pure"}
+    )
+    assert r.name == "foo-bar"
+    assert r.provides == ["foo-bar"]
