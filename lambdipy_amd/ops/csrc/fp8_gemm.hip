@@ -1,0 +1,255 @@
+// fp8 (OCP e4m3) GEMM + quantize + NHWC im2col for gfx950 — config 5.
+//
+// ResNet-50 inference path (BASELINE.json config 5 'fp8 MFMA'):
+//   - quantize_fp8: f32/bf16 -> e4m3 with a per-tensor scale (elementwise,
+//     vectorized 4 B/lane out). gfx950 fp8 is OCP e4m3fn — NOT MI300X fnuz
+//     (guide §4).
+//   - im2col_fp8: NHWC activations -> [M = N*HO*WO, K = R*S*C] e4m3 patch
+//     matrix, quantization fused (coalesced along C).
+//   - gemm_fp8_nt: C[M,N] f32 = A[M,K] @ Bt[N,K]^T on
+//     v_mfma_f32_16x16x32_fp8_fp8, same 128x128 LDS-tiled structure as the
+//     bf16 GEMM (gemm_bf16.hip), BK=32, padded 48-B LDS rows. The f32
+//     result is descaled by (scale_a * scale_b) in the epilogue, with
+//     optional bias + relu fused (conv epilogue).
+//
+// Fragment map (v_mfma_f32_16x16x32_fp8_fp8): lane l holds 8 e4m3 bytes of
+// A at [row = l&15][k = (l>>4)*8 + 0..7]; same map for Bt rows; D as bf16
+// variant (col = l&15, row = (l>>4)*4 + reg).
+
+#include "lpx_common.h"
+
+namespace lpx {
+
+namespace {
+
+constexpr int THREADS = 256;
+
+// ---- f32 -> e4m3 (OCP, saturating) --------------------------------------
+__device__ __forceinline__ unsigned char f32_to_e4m3(float x) {
+  // OCP e4m3fn: 1s4e3m, bias 7, max finite 448, no inf, nan = 0x7f/0xff.
+  union { float f; uint32_t u; } c;
+  c.f = x;
+  const uint32_t sign = (c.u >> 24) & 0x80;
+  float ax = fabsf(x);
+  if (!(ax == ax)) return (unsigned char)(sign | 0x7f);  // nan
+  if (ax > 448.f) ax = 448.f;                            // satfinite
+  c.f = ax;
+  int32_t e = (int32_t)((c.u >> 23) & 0xff) - 127;
+  uint32_t m = c.u & 0x7fffff;
+  uint32_t out;
+  if (ax == 0.f) {
+    out = 0;
+  } else if (e < -6) {
+    // subnormal: value = q * 2^-9, q in [0,8), round-to-nearest-even
+    float scaled = ax * 512.f;  // 2^9
+    uint32_t q = (uint32_t)scaled;
+    float frac = scaled - (float)q;
+    if (frac > 0.5f || (frac == 0.5f && (q & 1))) ++q;
+    out = (q >= 8) ? (1u << 3) : q;  // q==8 rounds up into normal range
+  } else {
+    // normal: exponent field e+7 in [1,15], 3-bit mantissa, RNE
+    uint32_t mant = m >> 20;           // top 3 bits
+    uint32_t rest = m & 0xfffff;
+    uint32_t half = 0x80000;
+    if (rest > half || (rest == half && (mant & 1))) {
+      ++mant;
+      if (mant == 8) { mant = 0; ++e; }
+    }
+    if (e > 8) { e = 8; mant = 7; }    // saturate at 448 (e=8 -> field 15)
+    out = ((uint32_t)(e + 7) << 3) | mant;
+  }
+  return (unsigned char)(sign | out);
+}
+
+__global__ __launch_bounds__(THREADS) void quantize_fp8_kernel(
+    const float* __restrict__ x, unsigned char* __restrict__ out, long n,
+    float inv_scale) {
+  for (long i = (long)blockIdx.x * THREADS + threadIdx.x; i < n;
+       i += (long)gridDim.x * THREADS)
+    out[i] = f32_to_e4m3(x[i] * inv_scale);
+}
+
+// ---- NHWC im2col (fused quantize) ---------------------------------------
+// x: [N, H, W, C] f32. out: [N*HO*WO, R*S*C] e4m3 rows (patch matrix).
+__global__ __launch_bounds__(THREADS) void im2col_fp8_kernel(
+    const float* __restrict__ x, unsigned char* __restrict__ out, int N,
+    int H, int W, int C, int R, int S, int HO, int WO, int stride, int pad,
+    float inv_scale) {
+  const long M = (long)N * HO * WO;
+  const int K = R * S * C;
+  for (long idx = (long)blockIdx.x * THREADS + threadIdx.x; idx < M * K;
+       idx += (long)gridDim.x * THREADS) {
+    const long m = idx / K;
+    const int k = (int)(idx - m * K);
+    const int c = k % C;
+    const int s = (k / C) % S;
+    const int r = k / (C * S);
+    const int wo = (int)(m % WO);
+    const int ho = (int)((m / WO) % HO);
+    const int n = (int)(m / ((long)WO * HO));
+    const int hi = ho * stride + r - pad;
+    const int wi = wo * stride + s - pad;
+    float v = 0.f;
+    if (hi >= 0 && hi < H && wi >= 0 && wi < W)
+      v = x[(((long)n * H + hi) * W + wi) * C + c];
+    out[idx] = f32_to_e4m3(v * inv_scale);
+  }
+}
+
+// ---- fp8 GEMM ------------------------------------------------------------
+
+constexpr int BM = 128, BN = 128, BK = 32;
+constexpr int ROW_BYTES = BK + 16;  // 48-B padded LDS row (32 fp8 + pad)
+
+typedef __attribute__((ext_vector_type(2))) int fp8frag;  // 8 e4m3 bytes
+
+__device__ __forceinline__ void stage_fp8_nt(const unsigned char* __restrict__ src,
+                                             int rows_total, int K, int row0,
+                                             int k0, char* __restrict__ lds) {
+  // 128 rows x 32 B per tile = 4 KiB; 256 threads x 1 chunk of 16 B.
+  const int t = threadIdx.x;
+  const int r = t >> 1;
+  const int kb = (t & 1) * 16;
+  const int grow = row0 + r;
+  const int gk = k0 + kb;
+  i32x4 v = {0, 0, 0, 0};
+  if (grow < rows_total) {
+    if (gk + 16 <= K) {
+      v = *reinterpret_cast<const i32x4*>(src + (long)grow * K + gk);
+    } else {
+      unsigned char z[16] = {0};
+      for (int j = 0; j < 16; ++j)
+        if (gk + j < K) z[j] = src[(long)grow * K + gk + j];
+      v = *reinterpret_cast<const i32x4*>(z);
+    }
+  }
+  *reinterpret_cast<i32x4*>(lds + r * ROW_BYTES + kb) = v;
+}
+
+template <bool RELU, bool BIAS>
+__global__ __launch_bounds__(THREADS) void gemm_fp8_nt_kernel(
+    const unsigned char* __restrict__ A, const unsigned char* __restrict__ Bt,
+    float* __restrict__ C, const float* __restrict__ bias, int M, int N,
+    int K, float descale, unsigned tiles_m, unsigned tiles_n) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* const sbase = smem;
+  const int BUF = BM * ROW_BYTES;
+
+  const unsigned nwg = tiles_m * tiles_n;
+  const unsigned bid = lpx_xcd_swizzle(blockIdx.x, nwg);
+  const int bm = (int)(bid / tiles_n) * BM;
+  const int bn = (int)(bid % tiles_n) * BN;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wr = wid >> 1, wc = wid & 1;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int frag_row = lane & 15;
+  const int frag_kb = lane >> 4;  // 8-byte k block (0..3)
+
+  int cur = 0;
+  stage_fp8_nt(A, M, K, bm, 0, sbase);
+  stage_fp8_nt(Bt, N, K, bn, 0, sbase + 2 * BUF);
+  __syncthreads();
+
+  const int ksteps = (K + BK - 1) / BK;
+  for (int ks = 0; ks < ksteps; ++ks) {
+    char* const aCur = sbase + cur * BUF;
+    char* const bCur = sbase + 2 * BUF + cur * BUF;
+    if (ks + 1 < ksteps) {
+      stage_fp8_nt(A, M, K, bm, (ks + 1) * BK, sbase + (cur ^ 1) * BUF);
+      stage_fp8_nt(Bt, N, K, bn, (ks + 1) * BK, sbase + 2 * BUF + (cur ^ 1) * BUF);
+    }
+    fp8frag aF[4], bF[4];
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf) {
+      const int r = wr * 64 + mf * 16 + frag_row;
+      aF[mf] = *reinterpret_cast<const fp8frag*>(aCur + r * ROW_BYTES + frag_kb * 8);
+    }
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      const int r = wc * 64 + nf * 16 + frag_row;
+      bF[nf] = *reinterpret_cast<const fp8frag*>(bCur + r * ROW_BYTES + frag_kb * 8);
+    }
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf)
+        acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+            __builtin_bit_cast(long, aF[mf]), __builtin_bit_cast(long, bF[nf]),
+            acc[mf][nf], 0, 0, 0);
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  const int orow0 = bm + wr * 64 + (lane >> 4) * 4;
+  const int ocol0 = bn + wc * 64 + (lane & 15);
+#pragma unroll
+  for (int mf = 0; mf < 4; ++mf) {
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = orow0 + mf * 16 + r;
+        const int col = ocol0 + nf * 16;
+        if (row < M && col < N) {
+          float v = acc[mf][nf][r] * descale;
+          if constexpr (BIAS) v += bias[col];
+          if constexpr (RELU) v = v > 0.f ? v : 0.f;
+          C[(long)row * N + col] = v;
+        }
+      }
+    }
+  }
+}
+
+}  // namespace
+
+hipError_t quantize_fp8(const float* x, unsigned char* out, long n,
+                        float inv_scale, hipStream_t stream) {
+  unsigned blocks = (unsigned)((n + THREADS - 1) / THREADS);
+  if (blocks > 2048u) blocks = 2048u;
+  quantize_fp8_kernel<<<dim3(blocks ? blocks : 1), dim3(THREADS), 0, stream>>>(
+      x, out, n, inv_scale);
+  return hipGetLastError();
+}
+
+hipError_t im2col_fp8(const float* x, unsigned char* out, int N, int H, int W,
+                      int C, int R, int S, int HO, int WO, int stride, int pad,
+                      float inv_scale, hipStream_t stream) {
+  const long total = (long)N * HO * WO * R * S * C;
+  unsigned blocks = (unsigned)((total + THREADS - 1) / THREADS);
+  if (blocks > 8192u) blocks = 8192u;
+  im2col_fp8_kernel<<<dim3(blocks ? blocks : 1), dim3(THREADS), 0, stream>>>(
+      x, out, N, H, W, C, R, S, HO, WO, stride, pad, inv_scale);
+  return hipGetLastError();
+}
+
+hipError_t gemm_fp8_nt(const unsigned char* A, const unsigned char* Bt,
+                       float* C, const float* bias, int M, int N, int K,
+                       float descale, bool relu, hipStream_t stream) {
+  const unsigned tiles_m = lpx_ceil_div(M, BM), tiles_n = lpx_ceil_div(N, BN);
+  const size_t lds = 4u * BM * ROW_BYTES;
+  dim3 grid(tiles_m * tiles_n), block(THREADS);
+  if (bias && relu)
+    gemm_fp8_nt_kernel<true, true><<<grid, block, lds, stream>>>(
+        A, Bt, C, bias, M, N, K, descale, tiles_m, tiles_n);
+  else if (bias)
+    gemm_fp8_nt_kernel<false, true><<<grid, block, lds, stream>>>(
+        A, Bt, C, bias, M, N, K, descale, tiles_m, tiles_n);
+  else if (relu)
+    gemm_fp8_nt_kernel<true, false><<<grid, block, lds, stream>>>(
+        A, Bt, C, bias, M, N, K, descale, tiles_m, tiles_n);
+  else
+    gemm_fp8_nt_kernel<false, false><<<grid, block, lds, stream>>>(
+        A, Bt, C, bias, M, N, K, descale, tiles_m, tiles_n);
+  return hipGetLastError();
+}
+
+}  // namespace lpx

@@ -51,4 +51,17 @@ hipError_t rf_predict_f32(const int* nodes4, const float* leaf_values,
                           const int* roots, int n_trees, const float* X,
                           float* out, int S, int F, int C, hipStream_t stream);
 
+// ---- fp8 (OCP e4m3) path — config 5 -------------------------------------
+// f32 -> e4m3 with inv_scale premultiply (per-tensor scaling).
+hipError_t quantize_fp8(const float* x, unsigned char* out, long n,
+                        float inv_scale, hipStream_t stream);
+// NHWC [N,H,W,C] f32 -> [N*HO*WO, R*S*C] e4m3 patch matrix (im2col+quantize).
+hipError_t im2col_fp8(const float* x, unsigned char* out, int N, int H, int W,
+                      int C, int R, int S, int HO, int WO, int stride, int pad,
+                      float inv_scale, hipStream_t stream);
+// C[M,N] f32 = descale * (A[M,K] @ Bt[N,K]^T) (+bias) (relu) — fp8 MFMA.
+hipError_t gemm_fp8_nt(const unsigned char* A, const unsigned char* Bt,
+                       float* C, const float* bias, int M, int N, int K,
+                       float descale, bool relu, hipStream_t stream);
+
 }  // namespace lpx

@@ -6,8 +6,8 @@
 
 #include <torch/extension.h>
 
-#include <c10/hip/HIPGuard.h>
-#include <c10/hip/HIPStream.h>
+#include <ATen/hip/impl/HIPGuardImplMasqueradingAsCUDA.h>
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
 
 #include "kernels.h"
 
@@ -18,7 +18,7 @@ namespace {
   TORCH_CHECK((x).is_contiguous(), #x " must be contiguous")
 
 hipStream_t current_stream(const at::Tensor& t) {
-  return c10::hip::getCurrentHIPStream(t.device().index()).stream();
+  return c10::hip::getCurrentHIPStreamMasqueradingAsCUDA(t.device().index()).stream();
 }
 
 void check_hip(hipError_t e, const char* what) {
@@ -35,7 +35,7 @@ at::Tensor gemm_bf16(const at::Tensor& A, const at::Tensor& Bt, bool bf16_out) {
   TORCH_CHECK(A.dim() == 2 && Bt.dim() == 2, "gemm_bf16 expects 2-D operands");
   const int64_t M = A.size(0), K = A.size(1), N = Bt.size(0);
   TORCH_CHECK(Bt.size(1) == K, "gemm_bf16: A is [M,K], Bt must be [N,K] (NT layout)");
-  c10::hip::HIPGuard guard(A.device());
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(A.device());
   auto C = at::empty({M, N}, A.options().dtype(bf16_out ? at::kBFloat16 : at::kFloat));
   const auto* a = reinterpret_cast<const uint16_t*>(A.data_ptr());
   const auto* b = reinterpret_cast<const uint16_t*>(Bt.data_ptr());
@@ -65,7 +65,7 @@ at::Tensor gemv(const at::Tensor& A_in, const at::Tensor& x_in) {
   TORCH_CHECK(A.dim() == 3 && x.dim() == 2, "gemv expects A [B,M,K], x [B,K]");
   const int64_t B = A.size(0), M = A.size(1), K = A.size(2);
   TORCH_CHECK(x.size(0) == B && x.size(1) == K, "gemv shape mismatch");
-  c10::hip::HIPGuard guard(A.device());
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(A.device());
   auto y = at::empty({B, M}, A.options().dtype(at::kFloat));
   if (A.scalar_type() == at::kBFloat16) {
     TORCH_CHECK(x.scalar_type() == at::kBFloat16, "gemv: dtype mismatch");
@@ -99,7 +99,7 @@ at::Tensor spmv_csr(const at::Tensor& indptr, const at::Tensor& indices,
   TORCH_CHECK(data.scalar_type() == at::kFloat && x.scalar_type() == at::kFloat,
               "spmv_csr expects f32 data/x");
   const int64_t n_rows = indptr.numel() - 1;
-  c10::hip::HIPGuard guard(x.device());
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
   auto y = at::empty({n_rows}, x.options());
   check_hip(lpx::spmv_csr_f32(indptr.data_ptr<int>(), indices.data_ptr<int>(),
                               data.data_ptr<float>(), x.data_ptr<float>(),
@@ -120,7 +120,7 @@ at::Tensor fft_c64(const at::Tensor& x, const at::Tensor& twiddles, bool inverse
   TORCH_CHECK(n >= 1 && (n & (n - 1)) == 0, "fft_c64: N must be a power of two");
   TORCH_CHECK(twiddles.numel() >= n / 2, "twiddle table too small");
   const int64_t batch = x.numel() / n;
-  c10::hip::HIPGuard guard(x.device());
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
   auto out = at::empty_like(x);
   check_hip(lpx::fft_c64(reinterpret_cast<const float*>(x.data_ptr()),
                          reinterpret_cast<float*>(out.data_ptr()),
@@ -140,7 +140,7 @@ at::Tensor pairwise_sqeuclidean(const at::Tensor& X, const at::Tensor& Y) {
   TORCH_CHECK(X.dim() == 2 && Y.dim() == 2 && X.size(1) == Y.size(1),
               "pairwise_sqeuclidean expects [M,K], [N,K]");
   const int64_t M = X.size(0), N = Y.size(0), K = X.size(1);
-  c10::hip::HIPGuard guard(X.device());
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(X.device());
   auto xsq = at::empty({M}, X.options());
   auto ysq = at::empty({N}, Y.options());
   auto D = at::empty({M, N}, X.options());
@@ -174,7 +174,7 @@ at::Tensor rf_predict(const at::Tensor& nodes, const at::Tensor& leaf_values,
   TORCH_CHECK(X.scalar_type() == at::kFloat && X.dim() == 2, "X must be f32 [S,F]");
   const int64_t S = X.size(0), F = X.size(1), C = leaf_values.size(1);
   TORCH_CHECK(C <= 8, "rf_predict kernel handles C<=8 per call (wrapper chunks)");
-  c10::hip::HIPGuard guard(X.device());
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(X.device());
   auto out = at::empty({S, C}, X.options());
   check_hip(lpx::rf_predict_f32(nodes.data_ptr<int>(), leaf_values.data_ptr<float>(),
                                 roots.data_ptr<int>(), (int)roots.numel(),
@@ -182,6 +182,61 @@ at::Tensor rf_predict(const at::Tensor& nodes, const at::Tensor& leaf_values,
                                 (int)F, (int)C, current_stream(X)),
             "rf_predict_f32");
   return out;
+}
+
+// ---- fp8 path (config 5) -------------------------------------------------
+
+at::Tensor quantize_fp8(const at::Tensor& x, double scale) {
+  CHECK_DEV(x);
+  TORCH_CHECK(x.scalar_type() == at::kFloat, "quantize_fp8 expects f32");
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
+  auto out = at::empty(x.sizes(), x.options().dtype(at::kByte));
+  check_hip(lpx::quantize_fp8(x.data_ptr<float>(), out.data_ptr<uint8_t>(),
+                              x.numel(), (float)(1.0 / scale), current_stream(x)),
+            "quantize_fp8");
+  return out;
+}
+
+at::Tensor im2col_fp8(const at::Tensor& x, int64_t R, int64_t S, int64_t stride,
+                      int64_t pad, double scale) {
+  CHECK_DEV(x);
+  TORCH_CHECK(x.scalar_type() == at::kFloat && x.dim() == 4,
+              "im2col_fp8 expects NHWC f32 [N,H,W,C]");
+  const int64_t N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+  const int64_t HO = (H + 2 * pad - R) / stride + 1;
+  const int64_t WO = (W + 2 * pad - S) / stride + 1;
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
+  auto out = at::empty({N * HO * WO, R * S * C}, x.options().dtype(at::kByte));
+  check_hip(lpx::im2col_fp8(x.data_ptr<float>(), out.data_ptr<uint8_t>(), (int)N,
+                            (int)H, (int)W, (int)C, (int)R, (int)S, (int)HO,
+                            (int)WO, (int)stride, (int)pad, (float)(1.0 / scale),
+                            current_stream(x)),
+            "im2col_fp8");
+  return out;
+}
+
+at::Tensor gemm_fp8(const at::Tensor& A, const at::Tensor& Bt, double descale,
+                    const c10::optional<at::Tensor>& bias, bool relu) {
+  CHECK_DEV(A);
+  CHECK_DEV(Bt);
+  TORCH_CHECK(A.scalar_type() == at::kByte && Bt.scalar_type() == at::kByte,
+              "gemm_fp8 expects e4m3 bytes");
+  TORCH_CHECK(A.dim() == 2 && Bt.dim() == 2 && A.size(1) == Bt.size(1),
+              "gemm_fp8 expects A [M,K], Bt [N,K]");
+  const int64_t M = A.size(0), K = A.size(1), N = Bt.size(0);
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(A.device());
+  auto C = at::empty({M, N}, A.options().dtype(at::kFloat));
+  const float* bptr = nullptr;
+  if (bias.has_value()) {
+    CHECK_DEV(bias.value());
+    TORCH_CHECK(bias->numel() == N, "bias must be [N]");
+    bptr = bias->data_ptr<float>();
+  }
+  check_hip(lpx::gemm_fp8_nt(A.data_ptr<uint8_t>(), Bt.data_ptr<uint8_t>(),
+                             C.data_ptr<float>(), bptr, (int)M, (int)N, (int)K,
+                             (float)descale, relu, current_stream(A)),
+            "gemm_fp8_nt");
+  return C;
 }
 
 std::string build_info() {
@@ -204,5 +259,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "D[i,j] = ||X_i - Y_j||^2 (f32 MFMA)", py::arg("X"), py::arg("Y"));
   m.def("rf_predict", &rf_predict, "RandomForest per-class sums over trees",
         py::arg("nodes"), py::arg("leaf_values"), py::arg("roots"), py::arg("X"));
+  m.def("quantize_fp8", &quantize_fp8, "f32 -> OCP e4m3 bytes (x / scale)",
+        py::arg("x"), py::arg("scale") = 1.0);
+  m.def("im2col_fp8", &im2col_fp8, "NHWC f32 -> [M, R*S*C] e4m3 patches",
+        py::arg("x"), py::arg("R"), py::arg("S"), py::arg("stride"),
+        py::arg("pad"), py::arg("scale") = 1.0);
+  m.def("gemm_fp8", &gemm_fp8, "fp8 e4m3 MFMA GEMM with fused descale/bias/relu",
+        py::arg("A"), py::arg("Bt"), py::arg("descale") = 1.0,
+        py::arg("bias") = py::none(), py::arg("relu") = false);
   m.def("build_info", &build_info);
 }

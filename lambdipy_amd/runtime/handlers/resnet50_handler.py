@@ -1,0 +1,84 @@
+"""Config-5 handler: fp8 ResNet-50 inference (random-init), with optional
+whole-node batch fan-out over RCCL (WORLD_SIZE>1: broadcast fp8 weights,
+shard the image batch, all_gather logits' argmax)."""
+
+from __future__ import annotations
+
+import json
+import os
+import time
+
+
+def main() -> dict:
+    t0 = time.perf_counter()
+    phases = []
+
+    def mark(name, start):
+        phases.append({"phase": name, "dur_ms": (time.perf_counter() - start) * 1e3})
+
+    t = time.perf_counter()
+    import torch
+    mark("import_torch", t)
+
+    t = time.perf_counter()
+    try:
+        import lpx_torch
+    except ImportError:
+        from lambdipy_amd.shims import lpx_torch
+    try:
+        import lpx_fanout
+    except ImportError:
+        from lambdipy_amd.parallel import fanout as lpx_fanout
+    mark("import_shim", t)
+
+    batch = int(os.environ.get("LPX_RESNET_BATCH", "64"))
+    res = int(os.environ.get("LPX_RESNET_RES", "224"))
+
+    t = time.perf_counter()
+    ctx = lpx_fanout.init_fanout()
+    mark("comm_init", t)
+
+    t = time.perf_counter()
+    model = lpx_torch.ResNet50FP8(n_classes=1000, device=str(ctx.device), seed=3)
+    torch.cuda.synchronize()
+    mark("model_build", t)
+
+    t = time.perf_counter()
+    lpx_fanout.broadcast_tensors(ctx, model.weight_tensors)
+    torch.cuda.synchronize()
+    mark("model_broadcast", t)
+
+    t = time.perf_counter()
+    if ctx.is_root:
+        g = torch.Generator().manual_seed(5)
+        X = torch.randn(batch * ctx.world_size, 3, res, res, generator=g).to(ctx.device)
+    else:
+        X = torch.empty(batch * ctx.world_size, 3, res, res, device=ctx.device)
+    torch.cuda.synchronize()
+    mark("batch_data", t)
+
+    t = time.perf_counter()
+    shard, counts = lpx_fanout.shard_rows(ctx, X)
+    logits = model(shard)
+    pred = logits.argmax(dim=1).to(torch.int32)
+    preds = lpx_fanout.all_gather_rows(ctx, pred, counts)
+    torch.cuda.synchronize()
+    mark("first_inference", t)
+
+    out = {
+        "handler": "resnet50",
+        "rank": ctx.rank,
+        "world_size": ctx.world_size,
+        "batch": int(X.shape[0]),
+        "checksum": float(preds.float().sum()),
+        "total_ms": (time.perf_counter() - t0) * 1e3,
+        "phases": phases,
+    }
+    if ctx.is_root:
+        print("LPX_RESULT " + json.dumps(out), flush=True)
+    lpx_fanout.destroy()
+    return out
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,136 @@
+"""lpx_torch — fp8 (OCP e4m3) ResNet-50 inference on gfx950 MFMA
+(recipe: torch; BASELINE.json config 5).
+
+Every conv/fc runs on the hand-written fp8 MFMA GEMM kernel (gemm_fp8 in
+lpx_ops): 1x1 convs are direct GEMMs over NHWC rows; 3x3/7x7 go through the
+fused im2col+quantize kernel. BN is folded into conv scale/bias at init
+(inference). Weights are quantized to e4m3 once at load (per-tensor scale =
+amax/448); activations are quantized per call with a dynamic per-tensor
+scale. Residual adds / pooling are cheap elementwise torch ops — the MFMA
+GEMMs are the hot path rocprof must show.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import List, Optional, Tuple
+
+import torch
+import torch.nn.functional as F
+
+try:
+    import lpx_ops as _ops
+except ImportError:
+    from lambdipy_amd import ops as _ops
+
+E4M3_MAX = 448.0
+
+
+def _quant_weight(w: torch.Tensor) -> Tuple[torch.Tensor, float]:
+    """w [KO, K] f32 -> (e4m3 bytes, scale)."""
+    amax = float(w.abs().amax().clamp(min=1e-8))
+    scale = amax / E4M3_MAX
+    return _ops.ext().quantize_fp8(w.contiguous(), scale), scale
+
+
+def _act_scale(x: torch.Tensor) -> float:
+    return float(x.abs().amax().clamp(min=1e-8)) / E4M3_MAX
+
+
+class Conv2dFP8:
+    """NHWC conv as fp8 MFMA GEMM (+ folded-BN bias, optional fused relu)."""
+
+    def __init__(self, cin: int, cout: int, ksize: int, stride: int, pad: int,
+                 relu: bool, device, gen: torch.Generator):
+        self.cin, self.cout = cin, cout
+        self.k, self.stride, self.pad = ksize, stride, pad
+        self.relu = relu
+        fan_in = cin * ksize * ksize
+        w = torch.randn(cout, ksize * ksize * cin, generator=gen,
+                        dtype=torch.float32) * (1.0 / math.sqrt(fan_in))
+        w = w.to(device)
+        self.wq, self.wscale = _quant_weight(w)
+        self.bias = (torch.randn(cout, generator=gen, dtype=torch.float32)
+                     * 0.01).to(device)
+
+    def __call__(self, x: torch.Tensor) -> torch.Tensor:
+        # x: [N, H, W, Cin] f32
+        N, H, W, C = x.shape
+        e = _ops.ext()
+        sa = _act_scale(x)
+        if self.k == 1 and self.stride == 1 and self.pad == 0:
+            aq = e.quantize_fp8(x.reshape(-1, C).contiguous(), sa)
+            ho, wo = H, W
+        else:
+            aq = e.im2col_fp8(x.contiguous(), self.k, self.k, self.stride,
+                              self.pad, sa)
+            ho = (H + 2 * self.pad - self.k) // self.stride + 1
+            wo = (W + 2 * self.pad - self.k) // self.stride + 1
+        out = e.gemm_fp8(aq, self.wq, sa * self.wscale, self.bias, self.relu)
+        return out.view(N, ho, wo, self.cout)
+
+
+class Bottleneck:
+    def __init__(self, cin: int, mid: int, cout: int, stride: int, device, gen):
+        self.conv1 = Conv2dFP8(cin, mid, 1, 1, 0, True, device, gen)
+        self.conv2 = Conv2dFP8(mid, mid, 3, stride, 1, True, device, gen)
+        self.conv3 = Conv2dFP8(mid, cout, 1, 1, 0, False, device, gen)
+        self.down: Optional[Conv2dFP8] = None
+        if stride != 1 or cin != cout:
+            self.down = Conv2dFP8(cin, cout, 1, stride, 0, False, device, gen)
+
+    def __call__(self, x: torch.Tensor) -> torch.Tensor:
+        idn = x if self.down is None else self.down(x)
+        y = self.conv3(self.conv2(self.conv1(x)))
+        return F.relu(y + idn)
+
+
+class ResNet50FP8:
+    """ResNet-50 v1 inference graph, random-init (no network for weights)."""
+
+    LAYERS = [(64, 256, 3, 1), (128, 512, 4, 2), (256, 1024, 6, 2),
+              (512, 2048, 3, 2)]
+
+    def __init__(self, n_classes: int = 1000, device: str = "cuda", seed: int = 0):
+        gen = torch.Generator().manual_seed(seed)
+        self.device = torch.device(device)
+        self.conv1 = Conv2dFP8(3, 64, 7, 2, 3, True, self.device, gen)
+        self.blocks: List[Bottleneck] = []
+        cin = 64
+        for mid, cout, n, stride in self.LAYERS:
+            for i in range(n):
+                self.blocks.append(
+                    Bottleneck(cin, mid, cout, stride if i == 0 else 1,
+                               self.device, gen))
+                cin = cout
+        fcw = torch.randn(n_classes, 2048, generator=gen) * (1.0 / math.sqrt(2048))
+        self.fcq, self.fcscale = _quant_weight(fcw.to(self.device))
+        self.fc_bias = torch.zeros(n_classes, dtype=torch.float32,
+                                   device=self.device)
+        self.n_classes = n_classes
+
+    @property
+    def weight_tensors(self) -> List[torch.Tensor]:
+        ts = []
+        convs = [self.conv1] + [c for b in self.blocks
+                                for c in (b.conv1, b.conv2, b.conv3, b.down) if c]
+        for c in convs:
+            ts.extend([c.wq, c.bias])
+        ts.extend([self.fcq, self.fc_bias])
+        return ts
+
+    def __call__(self, x: torch.Tensor) -> torch.Tensor:
+        """x: [N, 3, 224, 224] f32 (NCHW input, converted to NHWC inside)."""
+        x = x.permute(0, 2, 3, 1).contiguous()  # NHWC
+        x = self.conv1(x)
+        # maxpool 3x3 s2 p1 (torch elementwise op, NCHW round trip)
+        x = F.max_pool2d(x.permute(0, 3, 1, 2), 3, 2, 1).permute(0, 2, 3, 1)
+        x = x.contiguous()
+        for b in self.blocks:
+            x = b(x)
+        x = x.mean(dim=(1, 2))  # global average pool -> [N, 2048]
+        e = _ops.ext()
+        sa = _act_scale(x)
+        logits = e.gemm_fp8(e.quantize_fp8(x.contiguous(), sa), self.fcq,
+                            sa * self.fcscale, self.fc_bias, False)
+        return logits

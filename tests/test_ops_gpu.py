@@ -212,3 +212,81 @@ def test_rf_deterministic(ops):
     p1 = f.predict_proba_t(X)
     p2 = f.predict_proba_t(X)
     assert torch.equal(p1, p2)
+
+
+# ---- fp8 (config 5 path) -------------------------------------------------
+
+def test_quantize_fp8_matches_torch_e4m3(ops):
+    g = torch.Generator(device="cpu").manual_seed(0)
+    x = (torch.randn(4096, generator=g) * 100).to(DEV)
+    got = ops.ext().quantize_fp8(x, 1.0)
+    ref = x.to(torch.float8_e4m3fn).view(torch.uint8)
+    # allow the rare off-by-one at exact rounding ties
+    mism = (got != ref).float().mean().item()
+    assert mism < 1e-3, f"{mism*100:.3f}% mismatch vs torch e4m3fn"
+
+
+def test_gemm_fp8_vs_dequant_ref(ops):
+    e = ops.ext()
+    g = torch.Generator(device="cpu").manual_seed(1)
+    M, N, K = 256, 192, 128
+    A = torch.randn(M, K, generator=g).to(DEV)
+    Bt = torch.randn(N, K, generator=g).to(DEV)
+    sa, sb = 0.25, 0.5
+    Aq = e.quantize_fp8(A, sa)
+    Btq = e.quantize_fp8(Bt, sb)
+    C = e.gemm_fp8(Aq, Btq, sa * sb)
+    # reference: dequantized operands, f32 matmul
+    Ad = Aq.view(torch.float8_e4m3fn).float() * sa
+    Btd = Btq.view(torch.float8_e4m3fn).float() * sb
+    ref = Ad @ Btd.t()
+    err = (C - ref).abs().max().item() / (ref.abs().max().item() + 1e-6)
+    assert err < 1e-3, err
+
+
+def test_gemm_fp8_bias_relu(ops):
+    e = ops.ext()
+    g = torch.Generator(device="cpu").manual_seed(2)
+    M, N, K = 128, 128, 64
+    A = torch.randn(M, K, generator=g).to(DEV)
+    Bt = torch.randn(N, K, generator=g).to(DEV)
+    bias = torch.randn(N, generator=g).to(DEV)
+    Aq = e.quantize_fp8(A, 1.0)
+    Btq = e.quantize_fp8(Bt, 1.0)
+    C = e.gemm_fp8(Aq, Btq, 1.0, bias, True)
+    Ad = Aq.view(torch.float8_e4m3fn).float()
+    Btd = Btq.view(torch.float8_e4m3fn).float()
+    ref = torch.relu(Ad @ Btd.t() + bias)
+    assert (C - ref).abs().max().item() < 1e-2
+
+
+def test_im2col_fp8_matches_unfold(ops):
+    e = ops.ext()
+    g = torch.Generator(device="cpu").manual_seed(3)
+    N, H, W, C = 2, 14, 14, 8
+    R = S = 3
+    stride, pad = 2, 1
+    x = torch.randn(N, H, W, C, generator=g).to(DEV)
+    got = e.im2col_fp8(x, R, S, stride, pad, 1.0).view(torch.float8_e4m3fn).float()
+    # reference via torch.nn.functional.unfold on NCHW
+    import torch.nn.functional as TF
+    xn = x.permute(0, 3, 1, 2).contiguous()  # NCHW
+    ref = TF.unfold(xn, (R, S), padding=pad, stride=stride)  # [N, C*R*S, L]
+    HO = (H + 2 * pad - R) // stride + 1
+    WO = (W + 2 * pad - S) // stride + 1
+    # unfold col order is (c, r, s); ours is (r, s, c)
+    ref = ref.view(N, C, R, S, HO * WO).permute(0, 4, 2, 3, 1).reshape(N * HO * WO, R * S * C)
+    refq = ref.to(torch.float8_e4m3fn).float()
+    assert (got - refq).abs().max().item() < 1e-6
+
+
+def test_resnet50_fp8_forward(ops):
+    from lambdipy_amd.shims.lpx_torch import ResNet50FP8
+
+    model = ResNet50FP8(n_classes=100, device=DEV, seed=1)
+    x = torch.randn(2, 3, 64, 64, generator=torch.Generator().manual_seed(0)).to(DEV)
+    y1 = model(x)
+    assert y1.shape == (2, 100)
+    assert torch.isfinite(y1).all()
+    y2 = model(x)
+    assert torch.equal(y1, y2)  # deterministic inference
