@@ -50,9 +50,12 @@ def ext():
 
 # ---- dense ---------------------------------------------------------------
 
-def gemm_bf16(A: torch.Tensor, Bt: torch.Tensor, bf16_out: bool = False) -> torch.Tensor:
-    """C = A[M,K] @ Bt[N,K]^T — NT layout, bf16 MFMA kernel."""
-    return require_extension().gemm_bf16(A.contiguous(), Bt.contiguous(), bf16_out)
+def gemm_bf16(A: torch.Tensor, Bt: torch.Tensor, bf16_out: bool = False,
+              variant: int = -1) -> torch.Tensor:
+    """C = A[M,K] @ Bt[N,K]^T — NT layout, bf16 MFMA kernel.
+
+    variant: -1 auto, 1 = 128^2 tile, 2 = 256^2/BK64 register-staged pipeline."""
+    return require_extension().gemm_bf16(A.contiguous(), Bt.contiguous(), bf16_out, variant)
 
 
 def matmul_bf16(A: torch.Tensor, B: torch.Tensor, bf16_out: bool = False) -> torch.Tensor:

@@ -14,6 +14,12 @@ hipError_t gemm_bf16_nt(const uint16_t* A, const uint16_t* Bt, float* C,
 // Same but C stored as bf16.
 hipError_t gemm_bf16_nt_bf16out(const uint16_t* A, const uint16_t* Bt, uint16_t* C,
                                 int M, int N, int K, hipStream_t stream);
+// v2: 256x256/BK64 8-wave register-staged pipeline (large shapes).
+hipError_t gemm_bf16_nt256(const uint16_t* A, const uint16_t* Bt, float* C,
+                           int M, int N, int K, hipStream_t stream);
+hipError_t gemm_bf16_nt256_bf16out(const uint16_t* A, const uint16_t* Bt,
+                                   uint16_t* C, int M, int N, int K,
+                                   hipStream_t stream);
 
 // Batched GEMV: y[b,m] = sum_k A[b,m,k] * x[b,k].  bf16 in, f32 out.
 hipError_t gemv_bf16(const uint16_t* A, const uint16_t* x, float* y,

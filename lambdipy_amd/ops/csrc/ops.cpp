@@ -27,7 +27,8 @@ void check_hip(hipError_t e, const char* what) {
 
 // ---- GEMM ----------------------------------------------------------------
 
-at::Tensor gemm_bf16(const at::Tensor& A, const at::Tensor& Bt, bool bf16_out) {
+at::Tensor gemm_bf16(const at::Tensor& A, const at::Tensor& Bt, bool bf16_out,
+                     int64_t variant) {
   CHECK_DEV(A);
   CHECK_DEV(Bt);
   TORCH_CHECK(A.scalar_type() == at::kBFloat16 && Bt.scalar_type() == at::kBFloat16,
@@ -39,14 +40,27 @@ at::Tensor gemm_bf16(const at::Tensor& A, const at::Tensor& Bt, bool bf16_out) {
   auto C = at::empty({M, N}, A.options().dtype(bf16_out ? at::kBFloat16 : at::kFloat));
   const auto* a = reinterpret_cast<const uint16_t*>(A.data_ptr());
   const auto* b = reinterpret_cast<const uint16_t*>(Bt.data_ptr());
-  if (bf16_out)
+  // variant: -1 auto (v2 for big tiles), 1 = 128^2 tile, 2 = 256^2 tile
+  const bool big = (variant == 2) ||
+                   (variant < 0 && M >= 512 && N >= 512 && K >= 64);
+  auto s = current_stream(A);
+  if (big) {
+    if (bf16_out)
+      check_hip(lpx::gemm_bf16_nt256_bf16out(a, b, reinterpret_cast<uint16_t*>(C.data_ptr()),
+                                             (int)M, (int)N, (int)K, s),
+                "gemm_bf16_nt256");
+    else
+      check_hip(lpx::gemm_bf16_nt256(a, b, C.data_ptr<float>(), (int)M, (int)N,
+                                     (int)K, s),
+                "gemm_bf16_nt256");
+  } else if (bf16_out) {
     check_hip(lpx::gemm_bf16_nt_bf16out(a, b, reinterpret_cast<uint16_t*>(C.data_ptr()),
-                                        (int)M, (int)N, (int)K, current_stream(A)),
+                                        (int)M, (int)N, (int)K, s),
               "gemm_bf16_nt");
-  else
-    check_hip(lpx::gemm_bf16_nt(a, b, C.data_ptr<float>(), (int)M, (int)N, (int)K,
-                                current_stream(A)),
+  } else {
+    check_hip(lpx::gemm_bf16_nt(a, b, C.data_ptr<float>(), (int)M, (int)N, (int)K, s),
               "gemm_bf16_nt");
+  }
   return C;
 }
 
@@ -249,7 +263,8 @@ std::string build_info() {
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "lambdipy_amd gfx950 kernel library (MFMA/LDS-tiled HIP)";
   m.def("gemm_bf16", &gemm_bf16, "C = A[M,K] @ Bt[N,K]^T (bf16 MFMA)",
-        py::arg("A"), py::arg("Bt"), py::arg("bf16_out") = false);
+        py::arg("A"), py::arg("Bt"), py::arg("bf16_out") = false,
+        py::arg("variant") = -1);
   m.def("gemv", &gemv, "y[b,m] = A[b,m,:] . x[b,:]", py::arg("A"), py::arg("x"));
   m.def("spmv_csr", &spmv_csr, "y = CSR(A) @ x", py::arg("indptr"),
         py::arg("indices"), py::arg("data"), py::arg("x"));

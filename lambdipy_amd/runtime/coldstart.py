@@ -55,6 +55,36 @@ class ColdStartResult:
         }
 
 
+def _parallel_extract(zip_path: str, dest: str, workers: int = 8) -> None:
+    """Multi-threaded zip extraction (zlib inflate releases the GIL; ~2-4x
+    faster than ZipFile.extractall on a >100 MB layer — cold-start critical)."""
+    from concurrent.futures import ThreadPoolExecutor
+
+    with zipfile.ZipFile(zip_path) as zf:
+        infos = [i for i in zf.infolist() if not i.filename.endswith("/")]
+    dirs = {os.path.dirname(i.filename) for i in infos}
+    for d in sorted(dirs):
+        if d:
+            os.makedirs(os.path.join(dest, d), exist_ok=True)
+    # largest-first round-robin shards; each worker opens its OWN ZipFile
+    # (a shared handle serializes on its lock and is slower than extractall)
+    infos.sort(key=lambda i: i.file_size, reverse=True)
+    shards = [infos[w::workers] for w in range(workers)]
+
+    def _shard(shard):
+        with zipfile.ZipFile(zip_path) as z:
+            for info in shard:
+                out = os.path.join(dest, info.filename)
+                with z.open(info) as src, open(out, "wb") as dst:
+                    shutil.copyfileobj(src, dst, 1 << 20)
+                mode = (info.external_attr >> 16) & 0o777
+                if mode:
+                    os.chmod(out, mode)
+
+    with ThreadPoolExecutor(max_workers=workers) as ex:
+        list(ex.map(_shard, shards))
+
+
 def _free_port() -> int:
     s = socket.socket()
     s.bind(("127.0.0.1", 0))
@@ -75,14 +105,31 @@ def cold_start(
     """Run one cold start of `handler` from `layer` (zip path or dir)."""
     mod = HANDLERS.get(handler, handler)
     layer_mb = None
+
+    # failure handling (SURVEY.md §5): degrade to the largest available GPU
+    # count (8 -> 4 -> 2 -> 1) instead of hanging in comm init
+    if gpus > 1:
+        try:
+            import torch
+
+            avail = torch.cuda.device_count() if torch.cuda.is_available() else 0
+        except Exception:
+            avail = 0
+        if avail and gpus > avail:
+            g = avail
+            while g & (g - 1):
+                g &= g - 1  # largest power of two <= avail
+            log.warning("requested %d GPUs but only %d available; degrading to %d",
+                        gpus, avail, g)
+            gpus = max(g, 1)
+
     t0 = time.perf_counter()
 
     tmpdir = None
     if os.path.isfile(layer) and layer.endswith(".zip"):
         layer_mb = os.path.getsize(layer) / 2**20
         tmpdir = tempfile.mkdtemp(prefix="lpx-layer-")
-        with zipfile.ZipFile(layer) as zf:
-            zf.extractall(tmpdir)
+        _parallel_extract(layer, tmpdir)
         layer_dir = tmpdir
     else:
         layer_dir = layer

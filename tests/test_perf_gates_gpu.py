@@ -1,0 +1,84 @@
+"""Per-kernel performance floors (SURVEY.md §4.2 'Kernel performance gates').
+
+Floors are CONSERVATIVE (~50-60% of measured values on an idle MI355X) so
+regressions fail loudly while DVFS/noise doesn't flake the gate. Measured
+reference points (2026-09, this repo, MI355X): GEMM v1 670 TF @4096^3,
+RF 196 Msamples/s (256 trees, depth 10, F=32, 1M samples).
+"""
+
+import time
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda"
+
+
+def _time_op(fn, iters=10, warmup=3):
+    for _ in range(warmup):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / iters
+
+
+def test_gemm_bf16_floor():
+    from lambdipy_amd import ops
+
+    n = 4096
+    a = torch.randn(n, n, device=DEV, dtype=torch.bfloat16)
+    b = torch.randn(n, n, device=DEV, dtype=torch.bfloat16)
+    dt = _time_op(lambda: ops.gemm_bf16(a, b))
+    tf = 2 * n**3 / dt / 1e12
+    assert tf > 350, f"bf16 GEMM at {tf:.0f} TF (floor 350)"
+
+
+def test_rf_traversal_floor():
+    from lambdipy_amd.shims.lpx_sklearn import ForestSoA
+
+    f = ForestSoA.random(256, 10, 32, 8, seed=17, device=DEV)
+    X = torch.randn(1 << 20, 32, device=DEV)
+    dt = _time_op(lambda: f.predict_proba_t(X))
+    ms = (1 << 20) / dt / 1e6
+    assert ms > 90, f"RF traversal at {ms:.0f} Msamples/s (floor 90)"
+
+
+def test_gemv_bw_floor():
+    from lambdipy_amd import ops
+
+    m, k = 8192, 8192  # 128 MB weight read per call
+    a = torch.randn(m, k, device=DEV, dtype=torch.bfloat16)
+    x = torch.randn(k, device=DEV, dtype=torch.bfloat16)
+    dt = _time_op(lambda: ops.gemv(a, x), iters=20)
+    tbps = (m * k * 2) / dt / 1e12
+    assert tbps > 2.0, f"GEMV streaming at {tbps:.2f} TB/s (floor 2.0)"
+
+
+def test_fft_floor():
+    from lambdipy_amd import ops
+
+    batch, n = 4096, 1024  # 32 MB complex batch
+    g = torch.Generator().manual_seed(0)
+    x = (torch.randn(batch, n, generator=g) +
+         1j * torch.randn(batch, n, generator=g)).to(torch.complex64).to(DEV)
+    dt = _time_op(lambda: ops.fft(x), iters=20)
+    # effective traffic >= 2 * tensor bytes (in+out); require >= 0.5 TB/s
+    tbps = 2 * x.numel() * 8 / dt / 1e12
+    assert tbps > 0.5, f"FFT at {tbps:.2f} TB/s effective (floor 0.5)"
+
+
+def test_pairwise_floor():
+    from lambdipy_amd import ops
+
+    m = n = 4096
+    k = 128
+    X = torch.randn(m, k, device=DEV)
+    Y = torch.randn(n, k, device=DEV)
+    dt = _time_op(lambda: ops.pairwise_sqeuclidean(X, Y))
+    tf = 2 * m * n * k / dt / 1e12
+    assert tf > 25, f"pairwise at {tf:.0f} TF (floor 25, f32 MFMA peak 157)"
