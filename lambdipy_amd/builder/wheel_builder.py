@@ -63,28 +63,66 @@ def ensure_ops_extension(allow_build: bool = True) -> str:
     return p
 
 
-def _install_lpx_ops(build_dir: str) -> dict:
-    """Bundle the ops package into the layer as ``lpx_ops``."""
+def _read(path: str) -> bytes:
+    with open(path, "rb") as f:
+        return f.read()
+
+
+def emit_ops_wheel(dist_dir: str) -> str:
+    """Emit (or reuse) the lpx_ops ROCm wheel: loader + AOT gfx950 .so."""
+    from .wheelgen import write_wheel
+
     so = ensure_ops_extension()
-    dest = os.path.join(build_dir, "lpx_ops")
-    os.makedirs(dest, exist_ok=True)
-    for fn in ("__init__.py", "api.py"):
-        shutil.copy2(os.path.join(OPS_DIR, fn), os.path.join(dest, fn))
-    shutil.copy2(so, os.path.join(dest, os.path.basename(so)))
-    return {
-        "so": os.path.basename(so),
-        "so_mb": round(os.path.getsize(so) / 2**20, 2),
+    files = {
+        "lpx_ops/__init__.py": _read(os.path.join(OPS_DIR, "__init__.py")),
+        "lpx_ops/api.py": _read(os.path.join(OPS_DIR, "api.py")),
+        f"lpx_ops/{os.path.basename(so)}": _read(so),
     }
+    return write_wheel(
+        dist_dir, "lpx_ops", files, platform_specific=True,
+        summary="lambdipy_amd gfx950 kernel library (MFMA/LDS HIP, AOT)",
+    )
 
 
-def build_recipe_artifacts(recipe: BuildRecipe, dep: ResolvedDep, build_dir: str) -> dict:
-    """The wheel_builder hook for assemble_layer (native recipes only)."""
-    info = _install_lpx_ops(build_dir)
+def emit_shim_wheel(recipe: BuildRecipe, dist_dir: str) -> str:
+    """Emit (or reuse) the per-recipe shim wheel (depends on lpx_ops)."""
+    from .wheelgen import write_wheel
+
+    shim_src = os.path.join(SHIMS_DIR, recipe.shim + ".py")
+    if not os.path.exists(shim_src):
+        raise FileNotFoundError(f"recipe {recipe.name}: shim {recipe.shim} not found")
+    return write_wheel(
+        dist_dir, recipe.shim, {recipe.shim + ".py": _read(shim_src)},
+        requires=["lpx_ops"],
+        summary=f"{recipe.name} GPU shim (kernels: {', '.join(recipe.kernels)})",
+    )
+
+
+def _unpack_wheel(wheel_path: str, build_dir: str) -> None:
+    from ..packager.assemble import _vendor_wheel
+
+    _vendor_wheel(wheel_path, build_dir)
+
+
+def build_recipe_artifacts(recipe: BuildRecipe, dep: ResolvedDep, build_dir: str,
+                           dist_dir: Optional[str] = None) -> dict:
+    """The wheel_builder hook for assemble_layer (native recipes only).
+
+    Emits real ROCm wheels into the wheel cache, then unpacks them into the
+    layer (the reference's prebuilt-else-build flow)."""
+    from .wheelgen import WHEEL_CACHE
+
+    dist = dist_dir or WHEEL_CACHE
+    ops_whl = emit_ops_wheel(dist)
+    _unpack_wheel(ops_whl, build_dir)
+    info = {
+        "wheels": [os.path.basename(ops_whl)],
+        "so_mb": round(os.path.getsize(ensure_ops_extension()) / 2**20, 2),
+    }
     if recipe.shim:
-        shim_src = os.path.join(SHIMS_DIR, recipe.shim + ".py")
-        if not os.path.exists(shim_src):
-            raise FileNotFoundError(f"recipe {recipe.name}: shim {recipe.shim} not found")
-        shutil.copy2(shim_src, os.path.join(build_dir, recipe.shim + ".py"))
+        shim_whl = emit_shim_wheel(recipe, dist)
+        _unpack_wheel(shim_whl, build_dir)
+        info["wheels"].append(os.path.basename(shim_whl))
         info["shim"] = recipe.shim
     info["kernels"] = recipe.kernels
     return info

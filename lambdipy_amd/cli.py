@@ -139,15 +139,24 @@ def run(layer, handler, gpus, timeout, as_json):
 @click.option("--gpus", default="1", help="comma list, e.g. 1,2,4,8")
 @click.option("--repeats", default=1, show_default=True)
 @click.option("--out", default=None, help="append JSON lines to this file")
-def bench(layer, handler, gpus, repeats, out):
+@click.option("--rocprof", "rocprof_dir", default=None,
+              help="wrap the g=1 run in rocprofv3 --kernel-trace --stats, "
+                   "writing to this directory (MFMA/LDS evidence)")
+def bench(layer, handler, gpus, repeats, out, rocprof_dir):
     """Cold-start scaling sweep: layer MB + cold-start ms per GPU count."""
+    import shutil as _sh
+
     from .runtime import cold_start
 
     rows = []
     for g in [int(x) for x in gpus.split(",")]:
+        wrap = None
+        if rocprof_dir and g == 1 and _sh.which("rocprofv3"):
+            wrap = ["rocprofv3", "--kernel-trace", "--stats", "-d", rocprof_dir,
+                    "-o", f"{handler}_g1", "--"]
         best = None
         for _ in range(repeats):
-            r = cold_start(layer, handler=handler, gpus=g)
+            r = cold_start(layer, handler=handler, gpus=g, wrap_cmd=wrap)
             if best is None or r.total_ms < best.total_ms:
                 best = r
         rows.append(best.to_dict())

@@ -101,6 +101,7 @@ def cold_start(
     env: Optional[Dict[str, str]] = None,
     timeout_s: float = 600.0,
     keep_dir: bool = False,
+    wrap_cmd: Optional[List[str]] = None,
 ) -> ColdStartResult:
     """Run one cold start of `handler` from `layer` (zip path or dir)."""
     mod = HANDLERS.get(handler, handler)
@@ -145,10 +146,13 @@ def cold_start(
     base_env.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
 
     code = f"import lpx_handlers.{mod} as h; h.main()"
+    argv = [sys.executable, "-c", code]
+    if wrap_cmd:  # e.g. rocprofv3 --kernel-trace --stats -d DIR --
+        argv = list(wrap_cmd) + argv
     procs = []
     if gpus <= 1:
         procs.append(
-            subprocess.Popen([sys.executable, "-c", code], env=base_env,
+            subprocess.Popen(argv, env=base_env,
                              stdout=subprocess.PIPE, stderr=subprocess.PIPE,
                              text=True)
         )
@@ -161,7 +165,8 @@ def cold_start(
                 MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port),
             )
             procs.append(
-                subprocess.Popen([sys.executable, "-c", code], env=e,
+                subprocess.Popen(argv if r == 0 else [sys.executable, "-c", code],
+                                 env=e,
                                  stdout=subprocess.PIPE, stderr=subprocess.PIPE,
                                  text=True)
             )
