@@ -6,11 +6,16 @@ Imports the vendored `requests` from the layer and builds a Request object
 from __future__ import annotations
 
 import json
+import os
 import time
 
 
 def main() -> dict:
     t0 = time.perf_counter()
+    ready = os.environ.get("LPX_READY_FILE")
+    if ready:  # overlapped cold start: layer may still be extracting
+        while not os.path.exists(ready):
+            time.sleep(0.002)
     import requests
 
     req = requests.Request("GET", "https://example.invalid/lpx").prepare()

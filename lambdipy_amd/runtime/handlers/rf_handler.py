@@ -30,6 +30,14 @@ def main() -> dict:
     import torch
     mark("import_torch", t)
 
+    # layer may still be extracting (overlapped cold start) — wait for it
+    ready = os.environ.get("LPX_READY_FILE")
+    if ready:
+        t = time.perf_counter()
+        while not os.path.exists(ready):
+            time.sleep(0.002)
+        mark("wait_layer_extract", t)
+
     t = time.perf_counter()
     try:
         import lpx_sklearn
