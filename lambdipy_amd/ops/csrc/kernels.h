@@ -56,6 +56,11 @@ hipError_t row_sqnorm_f32(const float* X, float* out, int M, int K,
 hipError_t rf_predict_f32(const int* nodes4, const float* leaf_values,
                           const int* roots, int n_trees, const float* X,
                           float* out, int S, int F, int C, hipStream_t stream);
+// ILP-sweep entry (ilp in {1,2,4,8}) for perf A/B.
+hipError_t rf_predict_f32_ilp(const int* nodes4, const float* leaf_values,
+                              const int* roots, int n_trees, const float* X,
+                              float* out, int S, int F, int C, int ilp,
+                              hipStream_t stream);
 
 // ---- fp8 (OCP e4m3) path — config 5 -------------------------------------
 // f32 -> e4m3 with inv_scale premultiply (per-tensor scaling).

@@ -175,7 +175,8 @@ at::Tensor pairwise_sqeuclidean(const at::Tensor& X, const at::Tensor& Y) {
 // ---- RandomForest --------------------------------------------------------
 
 at::Tensor rf_predict(const at::Tensor& nodes, const at::Tensor& leaf_values,
-                      const at::Tensor& roots, const at::Tensor& X) {
+                      const at::Tensor& roots, const at::Tensor& X,
+                      int64_t ilp) {
   CHECK_DEV(nodes);
   CHECK_DEV(leaf_values);
   CHECK_DEV(roots);
@@ -190,11 +191,19 @@ at::Tensor rf_predict(const at::Tensor& nodes, const at::Tensor& leaf_values,
   TORCH_CHECK(C <= 8, "rf_predict kernel handles C<=8 per call (wrapper chunks)");
   c10::hip::HIPGuardMasqueradingAsCUDA guard(X.device());
   auto out = at::empty({S, C}, X.options());
-  check_hip(lpx::rf_predict_f32(nodes.data_ptr<int>(), leaf_values.data_ptr<float>(),
-                                roots.data_ptr<int>(), (int)roots.numel(),
-                                X.data_ptr<float>(), out.data_ptr<float>(), (int)S,
-                                (int)F, (int)C, current_stream(X)),
-            "rf_predict_f32");
+  if (ilp <= 0) {
+    check_hip(lpx::rf_predict_f32(nodes.data_ptr<int>(), leaf_values.data_ptr<float>(),
+                                  roots.data_ptr<int>(), (int)roots.numel(),
+                                  X.data_ptr<float>(), out.data_ptr<float>(), (int)S,
+                                  (int)F, (int)C, current_stream(X)),
+              "rf_predict_f32");
+  } else {
+    check_hip(lpx::rf_predict_f32_ilp(nodes.data_ptr<int>(), leaf_values.data_ptr<float>(),
+                                      roots.data_ptr<int>(), (int)roots.numel(),
+                                      X.data_ptr<float>(), out.data_ptr<float>(), (int)S,
+                                      (int)F, (int)C, (int)ilp, current_stream(X)),
+              "rf_predict_f32_ilp");
+  }
   return out;
 }
 
@@ -273,7 +282,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pairwise_sqeuclidean", &pairwise_sqeuclidean,
         "D[i,j] = ||X_i - Y_j||^2 (f32 MFMA)", py::arg("X"), py::arg("Y"));
   m.def("rf_predict", &rf_predict, "RandomForest per-class sums over trees",
-        py::arg("nodes"), py::arg("leaf_values"), py::arg("roots"), py::arg("X"));
+        py::arg("nodes"), py::arg("leaf_values"), py::arg("roots"), py::arg("X"),
+        py::arg("ilp") = 0);
   m.def("quantize_fp8", &quantize_fp8, "f32 -> OCP e4m3 bytes (x / scale)",
         py::arg("x"), py::arg("scale") = 1.0);
   m.def("im2col_fp8", &im2col_fp8, "NHWC f32 -> [M, R*S*C] e4m3 patches",
