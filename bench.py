@@ -90,13 +90,8 @@ def main() -> int:
         forest = ForestSoA.random(args.trees, args.depth, args.features,
                                   args.classes, seed=17, device=dev)
     else:
-        nn = 2 ** (args.depth + 1) - 1
-        nl = 2 ** args.depth
-        forest = ForestSoA(
-            torch.empty(args.trees * nn, 4, dtype=torch.int32, device=dev),
-            torch.empty(args.trees * nl, args.classes, dtype=torch.float32, device=dev),
-            torch.empty(args.trees, dtype=torch.int32, device=dev),
-            args.trees, args.classes)
+        forest = ForestSoA.alloc(args.trees, args.depth, args.features,
+                                 args.classes, device=dev)
     fanout.broadcast_tensors(ctx, [forest.nodes, forest.leaf_values, forest.roots])
 
     # synthetic global batch lives on root (weak scaling: B per GPU)

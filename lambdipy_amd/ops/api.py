@@ -114,14 +114,18 @@ def pairwise_sqeuclidean(X: torch.Tensor, Y: torch.Tensor) -> torch.Tensor:
 # ---- RandomForest --------------------------------------------------------
 
 def rf_predict_proba_sum(nodes: torch.Tensor, leaf_values: torch.Tensor,
-                         roots: torch.Tensor, X: torch.Tensor) -> torch.Tensor:
-    """Per-class sums over trees: out [S, C]. Chunks classes by 8 (kernel CMAX)."""
+                         roots: torch.Tensor, X: torch.Tensor,
+                         packed: bool = False) -> torch.Tensor:
+    """Per-class sums over trees: out [S, C]. Chunks classes by 8 (kernel CMAX).
+
+    packed: nodes are the depth-first 8-B layout ([n,2] int32)."""
     e = require_extension()
+    fn = e.rf_predict8 if packed else e.rf_predict
     C = leaf_values.shape[1]
     if C <= 8:
-        return e.rf_predict(nodes, leaf_values, roots, X.contiguous())
+        return fn(nodes, leaf_values, roots, X.contiguous())
     outs = []
     for c0 in range(0, C, 8):
         lv = leaf_values[:, c0 : c0 + 8].contiguous()
-        outs.append(e.rf_predict(nodes, lv, roots, X.contiguous()))
+        outs.append(fn(nodes, lv, roots, X.contiguous()))
     return torch.cat(outs, dim=1)

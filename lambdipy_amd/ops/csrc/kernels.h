@@ -56,6 +56,12 @@ hipError_t row_sqnorm_f32(const float* X, float* out, int M, int K,
 hipError_t rf_predict_f32(const int* nodes4, const float* leaf_values,
                           const int* roots, int n_trees, const float* X,
                           float* out, int S, int F, int C, hipStream_t stream);
+// Packed 8-B depth-first nodes (v2): meta = feat<<21 | right_delta; left
+// child implicit (nd+1); leaf feat=0x7FF, val = leaf row. Halves L2 traffic.
+hipError_t rf_predict8_f32(const int* nodes8, const float* leaf_values,
+                           const int* roots, int n_trees, const float* X,
+                           float* out, int S, int F, int C, int ilp,
+                           hipStream_t stream);
 // ILP-sweep entry (ilp in {1,2,4,8}) for perf A/B.
 hipError_t rf_predict_f32_ilp(const int* nodes4, const float* leaf_values,
                               const int* roots, int n_trees, const float* X,

@@ -262,6 +262,33 @@ at::Tensor gemm_fp8(const at::Tensor& A, const at::Tensor& Bt, double descale,
   return C;
 }
 
+at::Tensor rf_predict8(const at::Tensor& nodes8, const at::Tensor& leaf_values,
+                       const at::Tensor& roots, const at::Tensor& X,
+                       int64_t ilp) {
+  CHECK_DEV(nodes8);
+  CHECK_DEV(leaf_values);
+  CHECK_DEV(roots);
+  CHECK_DEV(X);
+  TORCH_CHECK(nodes8.scalar_type() == at::kInt && nodes8.dim() == 2 &&
+              nodes8.size(1) == 2, "nodes8 must be int32 [n_nodes, 2]");
+  TORCH_CHECK(leaf_values.scalar_type() == at::kFloat && leaf_values.dim() == 2,
+              "leaf_values must be f32 [n_leaf_rows, C]");
+  TORCH_CHECK(roots.scalar_type() == at::kInt, "roots must be int32");
+  TORCH_CHECK(X.scalar_type() == at::kFloat && X.dim() == 2, "X must be f32 [S,F]");
+  const int64_t S = X.size(0), F = X.size(1), C = leaf_values.size(1);
+  TORCH_CHECK(C <= 8, "rf_predict8 kernel handles C<=8 per call");
+  TORCH_CHECK(F <= 2046, "rf_predict8 packs feat in 11 bits (F<=2046)");
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(X.device());
+  auto out = at::empty({S, C}, X.options());
+  check_hip(lpx::rf_predict8_f32(nodes8.data_ptr<int>(), leaf_values.data_ptr<float>(),
+                                 roots.data_ptr<int>(), (int)roots.numel(),
+                                 X.data_ptr<float>(), out.data_ptr<float>(), (int)S,
+                                 (int)F, (int)C, (int)(ilp > 0 ? ilp : 2),
+                                 current_stream(X)),
+            "rf_predict8_f32");
+  return out;
+}
+
 std::string build_info() {
   return std::string("lambdipy_amd ops: gfx950 (CDNA4), HIP ") +
          std::to_string(HIP_VERSION_MAJOR) + "." + std::to_string(HIP_VERSION_MINOR);
@@ -284,6 +311,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rf_predict", &rf_predict, "RandomForest per-class sums over trees",
         py::arg("nodes"), py::arg("leaf_values"), py::arg("roots"), py::arg("X"),
         py::arg("ilp") = 0);
+  m.def("rf_predict8", &rf_predict8,
+        "RandomForest per-class sums (packed 8-B depth-first nodes)",
+        py::arg("nodes8"), py::arg("leaf_values"), py::arg("roots"),
+        py::arg("X"), py::arg("ilp") = 0);
   m.def("quantize_fp8", &quantize_fp8, "f32 -> OCP e4m3 bytes (x / scale)",
         py::arg("x"), py::arg("scale") = 1.0);
   m.def("im2col_fp8", &im2col_fp8, "NHWC f32 -> [M, R*S*C] e4m3 patches",

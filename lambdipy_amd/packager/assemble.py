@@ -151,6 +151,21 @@ def assemble_layer(
                 {"name": dep.name, "version": dep.version, "action": "deduped"}
             )
             continue
+        if recipe is not None and not recipe.vendor_upstream:
+            # runtime-provided upstream (e.g. torch): ship only the recipe's
+            # ROCm artifacts; record the runtime requirement in the manifest
+            log.info("runtime-provided %s==%s (recipe %s ships shim+kernels only)",
+                     dep.name, dep.version, recipe.name)
+            report.deps.append(
+                {"name": dep.name, "version": dep.version, "source": "runtime",
+                 "action": "runtime-provided",
+                 "runtime_requires": recipe.runtime_requires}
+            )
+            if wheel_builder is not None and recipe.kind == "native":
+                info = wheel_builder(recipe, dep, build_dir)
+                if info:
+                    report.deps[-1]["rocm_wheel"] = info
+            continue
         log.info(
             "vendor %s==%s [%s%s]",
             dep.name,

@@ -74,14 +74,8 @@ def main() -> dict:
             forest = lpx_sklearn.ForestSoA.random(
                 n_trees, depth, n_features, n_classes, seed=17, device=device)
         else:
-            n_nodes_t = 2 ** (depth + 1) - 1
-            n_leaves_t = 2 ** depth
-            forest = lpx_sklearn.ForestSoA(
-                torch.empty(n_trees * n_nodes_t, 4, dtype=torch.int32, device=device),
-                torch.empty(n_trees * n_leaves_t, n_classes, dtype=torch.float32,
-                            device=device),
-                torch.empty(n_trees, dtype=torch.int32, device=device),
-                n_trees, n_classes)
+            forest = lpx_sklearn.ForestSoA.alloc(
+                n_trees, depth, n_features, n_classes, device=device)
     if use_gpu:
         torch.cuda.synchronize()
     mark("model_build", t)

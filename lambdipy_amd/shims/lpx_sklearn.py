@@ -22,24 +22,37 @@ except ImportError:
 _DEV = "cuda"
 
 
+MAX_PACKED_FEAT = 2046   # feat field is 11 bits (0x7FF = leaf sentinel)
+MAX_PACKED_DELTA = (1 << 21) - 1
+
+
 class ForestSoA:
-    """Device-resident SoA forest: nodes int32 [n,4] {feat, thresh-bits,
-    left, right}, leaf feat=-1 and left = row into leaf_values [L, C]."""
+    """Device-resident SoA forest.
+
+    Two node layouts (the HIP kernels accept both):
+      - classic: int32 [n,4] {feat, thresh-bits, left, right}; leaf feat=-1,
+        left = leaf_values row.
+      - packed (default when it fits): int32 [n,2] {meta, val} in DEPTH-FIRST
+        order — meta = feat<<21 | right_delta, left child implicit at nd+1,
+        leaf meta feat==0x7FF and val = leaf row. Half the node bytes and the
+        '<=' chain is cacheline-local (the traversal kernel is L2-BW-bound).
+    """
 
     def __init__(self, nodes: torch.Tensor, leaf_values: torch.Tensor,
                  roots: torch.Tensor, n_trees: int, n_classes: int,
-                 classes: Optional[np.ndarray] = None):
+                 classes: Optional[np.ndarray] = None, packed: bool = False):
         self.nodes = nodes
         self.leaf_values = leaf_values
         self.roots = roots
         self.n_trees = n_trees
         self.n_classes = n_classes
+        self.packed = packed
         self.classes = classes if classes is not None else np.arange(n_classes)
 
     def to(self, device) -> "ForestSoA":
         return ForestSoA(self.nodes.to(device), self.leaf_values.to(device),
                          self.roots.to(device), self.n_trees, self.n_classes,
-                         self.classes)
+                         self.classes, self.packed)
 
     @property
     def nbytes(self) -> int:
@@ -90,7 +103,73 @@ class ForestSoA:
         rt = torch.tensor(roots, dtype=torch.int32)
         f = cls(nodes, leaves, rt, len(estimators), n_classes,
                 clf.classes_ if is_clf else None)
+        f = f.pack()  # depth-first 8-B layout when it fits
         return f.to(device) if device else f
+
+    def pack(self) -> "ForestSoA":
+        """Convert classic [n,4] nodes to packed depth-first [n,2] nodes.
+
+        Returns self unchanged when the forest does not fit the packed
+        limits (F > 2046, subtree > 2^21 nodes) or is already packed."""
+        if self.packed:
+            return self
+        nodes = self.nodes.cpu().numpy()
+        n = nodes.shape[0]
+        feat = nodes[:, 0]
+        if feat.max(initial=-1) > MAX_PACKED_FEAT:
+            return self
+        left, right = nodes[:, 2], nodes[:, 3]
+        # subtree sizes: sklearn/our layouts put children after parents per
+        # tree, so a reverse sweep sees children first
+        size = np.ones(n, dtype=np.int64)
+        internal = feat >= 0
+        order = np.arange(n - 1, -1, -1)
+        for i in order:
+            if internal[i]:
+                size[i] = 1 + size[left[i]] + size[right[i]]
+        new_nodes = np.empty((n, 2), dtype=np.uint32)
+        new_roots = np.empty(self.n_trees, dtype=np.int32)
+        pos = 0
+        roots = self.roots.cpu().numpy()
+        thr_bits = nodes[:, 1].astype(np.int64).view(np.uint64).astype(np.uint32)             if False else nodes[:, 1].view(np.uint32) if nodes[:, 1].flags['C_CONTIGUOUS']             else np.ascontiguousarray(nodes[:, 1]).view(np.uint32)
+        for t, root in enumerate(roots):
+            new_roots[t] = pos
+            stack = [int(root)]
+            while stack:
+                i = stack.pop()
+                if internal[i]:
+                    delta = int(size[left[i]]) + 1
+                    if delta > MAX_PACKED_DELTA:
+                        return self
+                    new_nodes[pos, 0] = (int(feat[i]) << 21) | delta
+                    new_nodes[pos, 1] = thr_bits[i]  # f32 threshold bits
+                    stack.append(int(right[i]))
+                    stack.append(int(left[i]))
+                else:
+                    new_nodes[pos, 0] = 0x7FF << 21
+                    new_nodes[pos, 1] = nodes[i, 2]  # leaf row
+                pos += 1
+        assert pos == n
+        new_nodes = new_nodes.view(np.int32)
+        return ForestSoA(torch.from_numpy(new_nodes).contiguous(),
+                         self.leaf_values, torch.from_numpy(new_roots),
+                         self.n_trees, self.n_classes, self.classes,
+                         packed=True)
+
+    @classmethod
+    def alloc(cls, n_trees: int, depth: int, n_features: int, n_classes: int,
+              device=_DEV) -> "ForestSoA":
+        """Empty forest with the shapes random(...) produces (for non-root
+        ranks receiving the model via broadcast)."""
+        n_nodes = n_trees * (2 ** (depth + 1) - 1)
+        n_leaves = n_trees * (2 ** depth)
+        packed = n_features <= MAX_PACKED_FEAT and depth <= 20
+        shape = (n_nodes, 2) if packed else (n_nodes, 4)
+        return cls(torch.empty(*shape, dtype=torch.int32, device=device),
+                   torch.empty(n_leaves, n_classes, dtype=torch.float32,
+                               device=device),
+                   torch.empty(n_trees, dtype=torch.int32, device=device),
+                   n_trees, n_classes, packed=packed)
 
     @classmethod
     def random(cls, n_trees: int, depth: int, n_features: int, n_classes: int,
@@ -125,13 +204,15 @@ class ForestSoA:
         f = cls(torch.from_numpy(all_nodes).contiguous(),
                 torch.from_numpy(all_leaves).contiguous(),
                 torch.from_numpy(roots).contiguous(), n_trees, n_classes)
+        if n_features <= MAX_PACKED_FEAT and depth <= 20:
+            f = f.pack()
         return f.to(device) if device else f
 
     # ---- inference -------------------------------------------------------
 
     def predict_proba_t(self, X: torch.Tensor) -> torch.Tensor:
         sums = _ops.rf_predict_proba_sum(self.nodes, self.leaf_values,
-                                         self.roots, X)
+                                         self.roots, X, packed=self.packed)
         return sums / float(self.n_trees)
 
     def predict_t(self, X: torch.Tensor) -> torch.Tensor:

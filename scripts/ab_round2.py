@@ -65,3 +65,17 @@ def main():
 
 if __name__ == "__main__":
     main()
+
+
+def rf_packed_ab():
+    """Packed vs classic RF traversal A/B (appended after the v2 kernel)."""
+    from lambdipy_amd.shims.lpx_sklearn import ForestSoA
+    import numpy as np
+
+    for trees, depth in [(256, 10), (512, 12)]:
+        fp = ForestSoA.random(trees, depth, 32, 8, seed=17, device="cuda")
+        assert fp.packed
+        X = torch.randn(1 << 20, 32, device="cuda")
+        dt = timeit(lambda: fp.predict_proba_t(X), iters=15)
+        print(f"PERF rf-packed {trees}x{depth}: {dt*1e3:.3f} ms "
+              f"{(1<<20)/dt/1e6:.0f} Msamples/s")

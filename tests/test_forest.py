@@ -10,20 +10,31 @@ from lambdipy_amd.shims.lpx_sklearn import ForestSoA
 
 
 def soa_predict_proba_cpu(f: ForestSoA, X: np.ndarray) -> np.ndarray:
-    """Reference traversal of the SoA arrays (mirrors the HIP kernel)."""
+    """Reference traversal of the SoA arrays (mirrors the HIP kernels;
+    handles both the classic [n,4] and packed depth-first [n,2] layouts)."""
     nodes = f.nodes.cpu().numpy()
     thr_f = np.ascontiguousarray(nodes[:, 1]).view(np.float32)
     leaves = f.leaf_values.cpu().numpy()
     roots = f.roots.cpu().numpy()
     S = X.shape[0]
     out = np.zeros((S, f.n_classes), dtype=np.float64)
-    for s in range(S):
-        for r in roots:
-            nd = int(r)
-            while nodes[nd, 0] >= 0:
-                feat = nodes[nd, 0]
-                nd = int(nodes[nd, 2]) if X[s, feat] <= thr_f[nd] else int(nodes[nd, 3])
-            out[s] += leaves[nodes[nd, 2]]
+    if f.packed:
+        meta = np.ascontiguousarray(nodes[:, 0]).view(np.uint32)
+        for s in range(S):
+            for r in roots:
+                nd = int(r)
+                while (meta[nd] >> 21) != 0x7FF:
+                    feat = int(meta[nd] >> 21)
+                    nd = nd + 1 if X[s, feat] <= thr_f[nd] else nd + int(meta[nd] & 0x1FFFFF)
+                out[s] += leaves[nodes[nd, 1]]
+    else:
+        for s in range(S):
+            for r in roots:
+                nd = int(r)
+                while nodes[nd, 0] >= 0:
+                    feat = nodes[nd, 0]
+                    nd = int(nodes[nd, 2]) if X[s, feat] <= thr_f[nd] else int(nodes[nd, 3])
+                out[s] += leaves[nodes[nd, 2]]
     return (out / f.n_trees).astype(np.float32)
 
 
@@ -61,7 +72,8 @@ def test_random_forest_shapes():
     f = ForestSoA.random(n_trees=4, depth=3, n_features=8, n_classes=3,
                          seed=1, device=None)
     n_nodes_per_tree = 2**4 - 1
-    assert f.nodes.shape == (4 * n_nodes_per_tree, 4)
+    assert f.packed  # small forest fits the packed 8-B layout
+    assert f.nodes.shape == (4 * n_nodes_per_tree, 2)
     assert f.leaf_values.shape == (4 * 8, 3)
     assert f.roots.tolist() == [0, 15, 30, 45]
     # leaf distributions normalized
@@ -91,3 +103,20 @@ def test_regressor_conversion():
     got = soa_predict_proba_cpu(f, X[:40])[:, 0]
     want = reg.predict(X[:40]).astype(np.float32)
     np.testing.assert_allclose(got, want, rtol=1e-4, atol=1e-5)
+
+
+def test_pack_roundtrip_equivalence():
+    """Packed traversal must agree exactly with the classic layout."""
+    sklearn = pytest.importorskip("sklearn")
+    from sklearn.ensemble import RandomForestClassifier
+
+    rng = np.random.default_rng(2)
+    X = rng.standard_normal((300, 9)).astype(np.float32)
+    y = (X[:, 0] > 0).astype(int) + (X[:, 4] > 0.3)
+    clf = RandomForestClassifier(n_estimators=9, max_depth=7, random_state=1)
+    clf.fit(X, y)
+    f = ForestSoA.from_sklearn(clf, device=None)
+    assert f.packed
+    got = soa_predict_proba_cpu(f, X[:80])
+    want = clf.predict_proba(X[:80]).astype(np.float32)
+    np.testing.assert_allclose(got, want, rtol=1e-5, atol=1e-6)
