@@ -63,10 +63,6 @@ def main():
         print(f"PERF rf512d12 ilp={ilp}: {dt*1e3:.3f} ms {(1<<20)/dt/1e6:.0f} Msamples/s")
 
 
-if __name__ == "__main__":
-    main()
-
-
 def rf_packed_ab():
     """Packed vs classic RF traversal A/B (appended after the v2 kernel)."""
     from lambdipy_amd.shims.lpx_sklearn import ForestSoA
@@ -79,3 +75,8 @@ def rf_packed_ab():
         dt = timeit(lambda: fp.predict_proba_t(X), iters=15)
         print(f"PERF rf-packed {trees}x{depth}: {dt*1e3:.3f} ms "
               f"{(1<<20)/dt/1e6:.0f} Msamples/s")
+
+
+if __name__ == "__main__":
+    main()
+    rf_packed_ab()
