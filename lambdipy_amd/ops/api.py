@@ -115,12 +115,22 @@ def pairwise_sqeuclidean(X: torch.Tensor, Y: torch.Tensor) -> torch.Tensor:
 
 def rf_predict_proba_sum(nodes: torch.Tensor, leaf_values: torch.Tensor,
                          roots: torch.Tensor, X: torch.Tensor,
-                         packed: bool = False) -> torch.Tensor:
+                         packed: bool = False, chunks=None) -> torch.Tensor:
     """Per-class sums over trees: out [S, C]. Chunks classes by 8 (kernel CMAX).
 
-    packed: nodes are the depth-first 8-B layout ([n,2] int32)."""
+    packed: nodes are the depth-first 8-B layout ([n,2] int32).
+    chunks: optional (chunk_tree_begin, max_chunk_nodes) — routes to the
+    forest-in-LDS kernel (fastest; needs packed nodes)."""
     e = require_extension()
-    fn = e.rf_predict8 if packed else e.rf_predict
+    if packed and chunks is not None:
+        ctb, max_nodes = chunks
+
+        def fn(n, lv, r, x):
+            return e.rf_predict8_lds(n, lv, r, ctb, max_nodes, x)
+    elif packed:
+        fn = e.rf_predict8
+    else:
+        fn = e.rf_predict
     C = leaf_values.shape[1]
     if C <= 8:
         return fn(nodes, leaf_values, roots, X.contiguous())

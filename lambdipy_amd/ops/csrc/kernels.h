@@ -62,6 +62,14 @@ hipError_t rf_predict8_f32(const int* nodes8, const float* leaf_values,
                            const int* roots, int n_trees, const float* X,
                            float* out, int S, int F, int C, int ilp,
                            hipStream_t stream);
+// Forest-in-LDS chunked traversal (v3): whole-tree chunks staged in LDS
+// (host-computed chunk_tree_begin[n_chunks+1]); out accumulates across
+// chunks. One block per CU.
+hipError_t rf_predict8_lds_f32(const int* nodes8, const float* leaf_values,
+                               const int* roots, const int* chunk_tree_begin,
+                               int n_chunks, int max_chunk_nodes, int n_trees,
+                               int total_nodes, const float* X, float* out,
+                               int S, int F, int C, hipStream_t stream);
 // ILP-sweep entry (ilp in {1,2,4,8}) for perf A/B.
 hipError_t rf_predict_f32_ilp(const int* nodes4, const float* leaf_values,
                               const int* roots, int n_trees, const float* X,

@@ -289,6 +289,33 @@ at::Tensor rf_predict8(const at::Tensor& nodes8, const at::Tensor& leaf_values,
   return out;
 }
 
+at::Tensor rf_predict8_lds(const at::Tensor& nodes8, const at::Tensor& leaf_values,
+                           const at::Tensor& roots, const at::Tensor& chunk_tree_begin,
+                           int64_t max_chunk_nodes, const at::Tensor& X) {
+  CHECK_DEV(nodes8);
+  CHECK_DEV(leaf_values);
+  CHECK_DEV(roots);
+  CHECK_DEV(chunk_tree_begin);
+  CHECK_DEV(X);
+  TORCH_CHECK(nodes8.scalar_type() == at::kInt && nodes8.dim() == 2 &&
+              nodes8.size(1) == 2, "nodes8 must be int32 [n_nodes, 2]");
+  TORCH_CHECK(chunk_tree_begin.scalar_type() == at::kInt,
+              "chunk_tree_begin must be int32 [n_chunks+1]");
+  const int64_t S = X.size(0), F = X.size(1), C = leaf_values.size(1);
+  TORCH_CHECK(C <= 8 && F <= 2046, "rf_predict8_lds: C<=8, F<=2046");
+  const int64_t n_chunks = chunk_tree_begin.numel() - 1;
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(X.device());
+  auto out = at::empty({S, C}, X.options());
+  check_hip(lpx::rf_predict8_lds_f32(
+                nodes8.data_ptr<int>(), leaf_values.data_ptr<float>(),
+                roots.data_ptr<int>(), chunk_tree_begin.data_ptr<int>(),
+                (int)n_chunks, (int)max_chunk_nodes, (int)roots.numel(),
+                (int)nodes8.size(0), X.data_ptr<float>(), out.data_ptr<float>(),
+                (int)S, (int)F, (int)C, current_stream(X)),
+            "rf_predict8_lds_f32");
+  return out;
+}
+
 std::string build_info() {
   return std::string("lambdipy_amd ops: gfx950 (CDNA4), HIP ") +
          std::to_string(HIP_VERSION_MAJOR) + "." + std::to_string(HIP_VERSION_MINOR);
@@ -311,6 +338,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rf_predict", &rf_predict, "RandomForest per-class sums over trees",
         py::arg("nodes"), py::arg("leaf_values"), py::arg("roots"), py::arg("X"),
         py::arg("ilp") = 0);
+  m.def("rf_predict8_lds", &rf_predict8_lds,
+        "RandomForest per-class sums (forest chunks staged in LDS)",
+        py::arg("nodes8"), py::arg("leaf_values"), py::arg("roots"),
+        py::arg("chunk_tree_begin"), py::arg("max_chunk_nodes"), py::arg("X"));
   m.def("rf_predict8", &rf_predict8,
         "RandomForest per-class sums (packed 8-B depth-first nodes)",
         py::arg("nodes8"), py::arg("leaf_values"), py::arg("roots"),

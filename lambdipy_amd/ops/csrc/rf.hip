@@ -168,7 +168,130 @@ __global__ __launch_bounds__(THREADS) void rf8_kernel(
   }
 }
 
+// ---- forest-in-LDS chunked kernel (v3) ----------------------------------
+// The v1/v2 kernels are L2 *line-throughput* bound: every node step pulls a
+// 64-B line for 8-16 B of node, and 4 blocks/CU thrash the 32-KiB L1 so the
+// depth-first locality is lost. v3 stages a CHUNK of whole trees (packed 8-B
+// nodes) into LDS and traverses from there: node fetches become ds_read_b64
+// (no line over-fetch, ~75 TB/s aggregate), forest bytes stream on-chip once
+// per block per chunk. One block per CU; per-class partial sums accumulate
+// into out[] across chunks (block owns its sample tiles exclusively).
+//
+// Host splits trees into chunks fitting: LDS = chunk nodes * 8 + X tile
+// (256 x (F+1) x 4 when F <= F_LDS_MAX) <= 160 KiB.
+
+template <bool STAGE_X, int ILP>
+__global__ __launch_bounds__(THREADS) void rf_lds_kernel(
+    const i32x2* __restrict__ nodes, const float* __restrict__ leaf_values,
+    const int* __restrict__ roots, const int* __restrict__ chunk_tree_begin,
+    int n_chunks, int n_trees, int total_nodes, const float* __restrict__ X,
+    float* __restrict__ out, int S, int F, int C, int max_chunk_nodes) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  i32x2* lnodes = reinterpret_cast<i32x2*>(smem);                // chunk nodes
+  float* xl = reinterpret_cast<float*>(smem + (size_t)max_chunk_nodes * 8);
+  const int t = threadIdx.x;
+  const int rowf = F + 1;
+
+  for (int ch = 0; ch < n_chunks; ++ch) {
+    const int tree0 = chunk_tree_begin[ch];
+    const int tree1 = chunk_tree_begin[ch + 1];
+    const int node0 = roots[tree0];
+    const int node1 = (tree1 < n_trees) ? roots[tree1] : total_nodes;
+    const int count = node1 - node0;  // host guarantees <= max_chunk_nodes
+    __syncthreads();  // previous iteration done with LDS
+    for (int i = t; i < count; i += THREADS) lnodes[i] = nodes[node0 + i];
+    __syncthreads();
+
+    for (int tile = blockIdx.x * THREADS; tile < S;
+         tile += gridDim.x * THREADS) {
+      const int s = tile + t;
+      const int tile_n = min(THREADS, S - tile);
+      if (STAGE_X) {
+        const long total = (long)tile_n * F;
+        for (long i = t; i < total; i += THREADS) {
+          const int r = (int)(i / F), c = (int)(i - (long)r * F);
+          xl[r * rowf + c] = X[((long)tile + r) * F + c];
+        }
+        __syncthreads();
+      }
+      if (s < S) {
+        const float* xrow = STAGE_X ? (xl + t * rowf) : (X + (long)s * F);
+        float acc[CMAX];
+#pragma unroll
+        for (int c = 0; c < CMAX; ++c) acc[c] = 0.f;
+        for (int tr = tree0; tr < tree1; tr += ILP) {
+          int nd[ILP];
+#pragma unroll
+          for (int i = 0; i < ILP; ++i)
+            nd[i] = (tr + i < tree1) ? roots[tr + i] - node0 : -1;
+          bool any = true;
+          while (any) {
+            any = false;
+#pragma unroll
+            for (int i = 0; i < ILP; ++i) {
+              if (nd[i] >= 0) {
+                const i32x2 n = lnodes[nd[i]];
+                const unsigned meta = (unsigned)n.x;
+                const int feat = (int)(meta >> 21);
+                if (feat != LEAF_FEAT) {
+                  const float xv = xrow[feat];
+                  nd[i] = (xv <= __int_as_float(n.y))
+                              ? nd[i] + 1
+                              : nd[i] + (int)(meta & 0x1FFFFF);
+                  any = true;
+                } else {
+                  const float* lv = leaf_values + (long)n.y * C;
+#pragma unroll
+                  for (int c = 0; c < CMAX; ++c)
+                    if (c < C) acc[c] += lv[c];
+                  nd[i] = -1;
+                }
+              }
+            }
+          }
+        }
+        float* orow = out + (long)s * C;
+        if (ch == 0) {
+#pragma unroll
+          for (int c = 0; c < CMAX; ++c)
+            if (c < C) orow[c] = acc[c];
+        } else {
+#pragma unroll
+          for (int c = 0; c < CMAX; ++c)
+            if (c < C) orow[c] += acc[c];
+        }
+      }
+      if (STAGE_X) __syncthreads();
+    }
+  }
+}
+
 }  // namespace
+
+hipError_t rf_predict8_lds_f32(const int* nodes8, const float* leaf_values,
+                               const int* roots, const int* chunk_tree_begin,
+                               int n_chunks, int max_chunk_nodes, int n_trees,
+                               int total_nodes, const float* X, float* out,
+                               int S, int F, int C, hipStream_t stream) {
+  if (C > CMAX) return hipErrorInvalidValue;
+  const bool stage = F <= F_LDS_MAX;
+  const size_t xbytes = stage ? (size_t)THREADS * (F + 1) * sizeof(float) : 0;
+  const size_t lds = (size_t)max_chunk_nodes * 8 + xbytes;
+  if (lds > 160 * 1024) return hipErrorInvalidValue;
+  unsigned blocks = lpx_ceil_div(S, THREADS);
+  if (blocks > 256u) blocks = 256u;  // one block per CU (full-LDS kernel)
+  if (blocks == 0) blocks = 1;
+  const auto* n8 = reinterpret_cast<const i32x2*>(nodes8);
+  if (stage)
+    rf_lds_kernel<true, 2><<<dim3(blocks), dim3(THREADS), lds, stream>>>(
+        n8, leaf_values, roots, chunk_tree_begin, n_chunks, n_trees,
+        total_nodes, X, out, S, F, C, max_chunk_nodes);
+  else
+    rf_lds_kernel<false, 2><<<dim3(blocks), dim3(THREADS), lds, stream>>>(
+        n8, leaf_values, roots, chunk_tree_begin, n_chunks, n_trees,
+        total_nodes, X, out, S, F, C, max_chunk_nodes);
+  return hipGetLastError();
+}
 
 hipError_t rf_predict8_f32(const int* nodes8, const float* leaf_values,
                            const int* roots, int n_trees, const float* X,

@@ -210,9 +210,50 @@ class ForestSoA:
 
     # ---- inference -------------------------------------------------------
 
+    _chunk_cache = None
+
+    def _plan_chunks(self, F: int):
+        """Greedy whole-tree chunking for the forest-in-LDS kernel.
+
+        Returns (chunk_tree_begin int32 device tensor [n_chunks+1],
+        max_chunk_nodes) or None when a single tree exceeds the LDS budget
+        (fall back to the global-memory kernel)."""
+        if not self.packed:
+            return None
+        if self._chunk_cache is not None and self._chunk_cache[0] == F:
+            return self._chunk_cache[1]
+        xbytes = 256 * (F + 1) * 4 if F <= 63 else 0
+        cap = (160 * 1024 - xbytes - 2048) // 8  # nodes per chunk
+        roots = self.roots.cpu().numpy()
+        total = self.nodes.shape[0]
+        counts = np.diff(np.append(roots, total))
+        plan = None
+        if counts.max(initial=0) <= cap and cap > 0:
+            begins = [0]
+            cur = 0
+            for i, c in enumerate(counts):
+                if cur + c > cap:
+                    begins.append(i)
+                    cur = 0
+                cur += int(c)
+            begins.append(len(counts))
+            ctb = torch.tensor(begins, dtype=torch.int32,
+                               device=self.nodes.device)
+            starts = np.array(begins[:-1])
+            ends = np.array(begins[1:])
+            node_b = roots[starts]
+            node_e = np.where(ends < len(roots), 
+                              np.append(roots, total)[ends], total)
+            max_nodes = int((node_e - node_b).max())
+            plan = (ctb, max_nodes)
+        self._chunk_cache = (F, plan)
+        return plan
+
     def predict_proba_t(self, X: torch.Tensor) -> torch.Tensor:
+        chunks = self._plan_chunks(X.shape[1]) if self.packed else None
         sums = _ops.rf_predict_proba_sum(self.nodes, self.leaf_values,
-                                         self.roots, X, packed=self.packed)
+                                         self.roots, X, packed=self.packed,
+                                         chunks=chunks)
         return sums / float(self.n_trees)
 
     def predict_t(self, X: torch.Tensor) -> torch.Tensor:

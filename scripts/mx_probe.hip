@@ -1,33 +1,43 @@
-// MX-scaled fp8 MFMA layout probe (gfx950): one v_mfma_scale_f32_32x32x64
-// with assumed fragment map A[row=l&31][k=(l>>5)*32+0..31] (32 B/lane),
-// per-lane E8M0 scales. Verifies against host reference via unit scales
-// (127 -> 2^0) and a non-trivial scale pattern.
+// MX-scaled fp8 MFMA scale-mapping probe (gfx950).
+// Fragment map verified (unit scales pass): A[row=l&31][k=(l>>5)*32+0..31].
+// This probe sweeps per-lane scale sourcing hypotheses:
+//   mode 0: sa = S[l&31][l>>5]                  (scale of lane's own block)
+//   mode 1: sa = S[l>>1][l&1]                   (row-major pairs over lanes)
+//   mode 2: sa = S[l&31][0] | S[l&31][1] << 8   (both blocks in one reg)
+//   mode 3: mode 2 with opsel = 1
 #include <hip/hip_runtime.h>
 #include <stdint.h>
 
 typedef __attribute__((ext_vector_type(8))) int i32x8;
 typedef __attribute__((ext_vector_type(16))) float f32x16;
 
+__device__ int pick_scale(const uint8_t* S, int l, int mode) {
+  switch (mode & 3) {
+    case 0: return S[(l & 31) * 2 + (l >> 5)];
+    case 1: return S[(l >> 1) * 2 + (l & 1)];
+    default: return (int)S[(l & 31) * 2] | ((int)S[(l & 31) * 2 + 1] << 8);
+  }
+}
+
 extern "C" __global__ void mx_probe_kernel(const uint8_t* __restrict__ A,
                                            const uint8_t* __restrict__ B,
                                            const uint8_t* __restrict__ scaleA,
                                            const uint8_t* __restrict__ scaleB,
-                                           float* __restrict__ D) {
-  // A: [32][64] fp8 row-major; B: [64][32] -> we pass Bt[32][64] (col j holds
-  // B[k][j] as Bt[j][k]); scaleA: [32][2] per (row, kblock); scaleB: [32][2]
-  // per (col, kblock). One wave.
+                                           float* __restrict__ D, int mode) {
   const int l = threadIdx.x;
-  const int rc = l & 31;          // row of A / col of B
-  const int kb = l >> 5;          // k-block (0..1), 32 elems each
+  const int rc = l & 31;
+  const int kb = l >> 5;
   i32x8 a = *reinterpret_cast<const i32x8*>(A + rc * 64 + kb * 32);
   i32x8 b = *reinterpret_cast<const i32x8*>(B + rc * 64 + kb * 32);
   f32x16 c;
 #pragma unroll
   for (int i = 0; i < 16; ++i) c[i] = 0.f;
-  const int sa = scaleA[rc * 2 + kb];
-  const int sb = scaleB[rc * 2 + kb];
-  c = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(a, b, c, 0, 0, 0, sa, 0, sb);
-  // D map (32x32): col = l&31, row = (reg&3) + 8*(reg>>2) + 4*(l>>5)
+  const int sa = pick_scale(scaleA, l, mode);
+  const int sb = pick_scale(scaleB, l, mode);
+  if (mode == 3)
+    c = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(a, b, c, 0, 0, 1, sa, 1, sb);
+  else
+    c = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(a, b, c, 0, 0, 0, sa, 0, sb);
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
     const int row = (r & 3) + 8 * (r >> 2) + 4 * (l >> 5);
@@ -38,7 +48,7 @@ extern "C" __global__ void mx_probe_kernel(const uint8_t* __restrict__ A,
 
 extern "C" int run_mx_probe(const uint8_t* hA, const uint8_t* hB,
                             const uint8_t* hSA, const uint8_t* hSB,
-                            float* hD) {
+                            float* hD, int mode) {
   uint8_t *dA, *dB, *dSA, *dSB;
   float* dD;
   (void)hipMalloc(&dA, 32 * 64);
@@ -50,7 +60,8 @@ extern "C" int run_mx_probe(const uint8_t* hA, const uint8_t* hB,
   (void)hipMemcpy(dB, hB, 32 * 64, hipMemcpyHostToDevice);
   (void)hipMemcpy(dSA, hSA, 64, hipMemcpyHostToDevice);
   (void)hipMemcpy(dSB, hSB, 64, hipMemcpyHostToDevice);
-  hipLaunchKernelGGL(mx_probe_kernel, dim3(1), dim3(64), 0, 0, dA, dB, dSA, dSB, dD);
+  hipLaunchKernelGGL(mx_probe_kernel, dim3(1), dim3(64), 0, 0, dA, dB, dSA,
+                     dSB, dD, mode);
   hipError_t e = hipDeviceSynchronize();
   (void)hipMemcpy(hD, dD, 32 * 32 * 4, hipMemcpyDeviceToHost);
   (void)hipFree(dA); (void)hipFree(dB); (void)hipFree(dSA); (void)hipFree(dSB); (void)hipFree(dD);
