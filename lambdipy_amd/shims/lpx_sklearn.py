@@ -249,8 +249,14 @@ class ForestSoA:
         self._chunk_cache = (F, plan)
         return plan
 
+    # forest-in-LDS routing measured SLOWER on MI355X (latency-bound at one
+    # 4-wave block/CU + per-chunk X restaging) — global packed kernel is the
+    # default; set use_lds_chunks=True to A/B the LDS variant.
+    use_lds_chunks = False
+
     def predict_proba_t(self, X: torch.Tensor) -> torch.Tensor:
-        chunks = self._plan_chunks(X.shape[1]) if self.packed else None
+        chunks = (self._plan_chunks(X.shape[1])
+                  if (self.packed and self.use_lds_chunks) else None)
         sums = _ops.rf_predict_proba_sum(self.nodes, self.leaf_values,
                                          self.roots, X, packed=self.packed,
                                          chunks=chunks)

@@ -19,6 +19,21 @@ __device__ int pick_scale(const uint8_t* S, int l, int mode) {
   }
 }
 
+// lane-id-encoding modes for mapping derivation (scripts/mx_probe.py):
+//  4: sa = 127 + (l&7),        sb = 127
+//  5: sa = 127 + ((l>>3)&7),   sb = 127
+//  6: sa = 127,                sb = 127 + (l&7)
+//  7: sa = 127,                sb = 127 + ((l>>3)&7)
+__device__ void probe_scales(int l, int mode, int* sa, int* sb) {
+  const int lo = 127 + (l & 7);
+  const int hi = 127 + ((l >> 3) & 7);
+  *sa = 127; *sb = 127;
+  if (mode == 4) *sa = lo;
+  else if (mode == 5) *sa = hi;
+  else if (mode == 6) *sb = lo;
+  else if (mode == 7) *sb = hi;
+}
+
 extern "C" __global__ void mx_probe_kernel(const uint8_t* __restrict__ A,
                                            const uint8_t* __restrict__ B,
                                            const uint8_t* __restrict__ scaleA,
@@ -32,8 +47,13 @@ extern "C" __global__ void mx_probe_kernel(const uint8_t* __restrict__ A,
   f32x16 c;
 #pragma unroll
   for (int i = 0; i < 16; ++i) c[i] = 0.f;
-  const int sa = pick_scale(scaleA, l, mode);
-  const int sb = pick_scale(scaleB, l, mode);
+  int sa, sb;
+  if (mode >= 4) {
+    probe_scales(l, mode, &sa, &sb);
+  } else {
+    sa = pick_scale(scaleA, l, mode);
+    sb = pick_scale(scaleB, l, mode);
+  }
   if (mode == 3)
     c = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(a, b, c, 0, 0, 1, sa, 1, sb);
   else
