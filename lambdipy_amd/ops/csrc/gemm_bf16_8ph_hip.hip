@@ -1,0 +1,219 @@
+#include "hip/hip_runtime.h"
+// bf16 GEMM v3 for gfx950 — 256x256/BK64, glds-staged 8-phase pipeline.
+//
+// The guide's fast-template class (cdna_hip_programming.md §5 'The 256^2
+// 8-phase template'): 512 threads (8 waves), 2 K-tiles per 8 phases, each
+// phase computes ONE 128x128 tile-quadrant (rh, ch) = ((p>>1)&1, p&1) —
+// every wave contributes a 64x32 sub-tile, 16 MFMAs — while ONE half-tile
+// (128 rows x 64 k of A or B) is prefetched by global_load_lds (2
+// instructions per thread). Counted s_waitcnt vmcnt(6) per phase (never 0 in
+// the loop) keeps 3 half-tiles in flight across raw s_barriers; staging lands
+// in the slot the previous-parity tile freed one phase earlier:
+//   phase 4T+0 reads {A-lo,B-lo}, +1 {A-lo,B-hi}, +2 {A-hi,B-lo}, +3 {A-hi,B-hi}
+//   phase 4T+0 stages (T+1, B-hi), +1 (T+1, A-hi), +2 (T+2, A-lo), +3 (T+2, B-lo)
+// LDS 128 KiB: 2 parities x (A,B) x 2 halves x 16 KiB, lane-linear glds
+// image with the st_16x32 XOR swizzle applied to the SOURCE address and the
+// ds_read address (rule 21; 16-B cell q <-> q ^ (((q>>5)&1)<<1)).
+// Requires M,N % 256 == 0 and K % 64 == 0 (launcher falls back to v2).
+
+#include "lpx_common.h"
+
+namespace lpx {
+
+namespace {
+
+constexpr int BM8 = 256, BK8 = 64;
+constexpr int THREADS8 = 512;
+constexpr int HALF_BYTES = 128 * 128;  // 128 rows x 128 B
+constexpr int PARITY_BYTES = 4 * HALF_BYTES;  // A-lo A-hi B-lo B-hi
+
+using bf16frag = bf16x8;
+
+__device__ __forceinline__ int swz_cell(int q) {
+  return q ^ (((q >> 5) & 1) << 1);
+}
+
+// stage one half-tile: hh 0=A-lo 1=A-hi 2=B-lo 3=B-hi of K-tile `tile`
+__device__ __forceinline__ void stage_half8(const uint16_t* __restrict__ A,
+                                            const uint16_t* __restrict__ Bt,
+                                            long K, int bm, int bn, int tile,
+                                            int hh, int ksteps,
+                                            char* __restrict__ smem, int w,
+                                            int lane) {
+  const int tc = tile < ksteps ? tile : ksteps - 1;  // clamp (never read)
+  const long k0 = (long)tc * BK8;
+  const uint16_t* src = (hh < 2) ? A : Bt;
+  const int row0 = ((hh < 2) ? bm : bn) + (hh & 1) * 128;
+  char* const base = smem + (size_t)(tile & 1) * PARITY_BYTES +
+                     (size_t)hh * HALF_BYTES;
+#pragma unroll
+  for (int j = 0; j < 2; ++j) {
+    const int qbase = (j * 8 + w) * 64;  // 16-B cell stripe of this wave
+    const int q = qbase + lane;
+    const int qs = swz_cell(q);
+    const int r = qs >> 3, cell = qs & 7;
+    const uint16_t* g = src + (long)(row0 + r) * K + (k0 + cell * 8);
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)g,
+        (__attribute__((address_space(3))) void*)(base + qbase * 16), 16, 0, 0);
+  }
+}
+
+// which (tile, half) the cyclic schedule stages at phase (t, p)
+__device__ __forceinline__ void stage_slot(int t, int p, int* tile, int* hh) {
+  switch (p) {
+    case 0: *tile = t + 1; *hh = 3; break;  // (T+1, B-hi)
+    case 1: *tile = t + 1; *hh = 1; break;  // (T+1, A-hi)
+    case 2: *tile = t + 2; *hh = 0; break;  // (T+2, A-lo)
+    default: *tile = t + 2; *hh = 2; break; // (T+2, B-lo)
+  }
+}
+
+__device__ __forceinline__ const bf16frag* frag_ptr(const char* __restrict__ smem,
+                                                    int parity, int hh,
+                                                    int row_in_half, int cell) {
+  const int q = row_in_half * 8 + cell;
+  const int qs = swz_cell(q);
+  return reinterpret_cast<const bf16frag*>(
+      smem + (size_t)parity * PARITY_BYTES + (size_t)hh * HALF_BYTES + qs * 16);
+}
+
+template <bool BF16_OUT>
+__global__ __launch_bounds__(THREADS8, 2) void gemm_bf16_8ph_kernel(
+    const uint16_t* __restrict__ A, const uint16_t* __restrict__ Bt,
+    void* __restrict__ Cout, int M, int N, int K, unsigned tiles_m,
+    unsigned tiles_n) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+
+  const unsigned nwg = tiles_m * tiles_n;
+  const unsigned bid = lpx_xcd_swizzle(blockIdx.x, nwg);
+  const int bm = (int)(bid / tiles_n) * BM8;
+  const int bn = (int)(bid % tiles_n) * BM8;
+
+  const int lane = threadIdx.x & 63;
+  const int w = threadIdx.x >> 6;       // wave id 0..7
+  const int wsr = w >> 2;               // wave sub-row within quadrant (0..1)
+  const int wsc = w & 3;                // wave sub-col within quadrant (0..3)
+
+  const int fr = lane & 15;             // fragment row
+  const int fkb = lane >> 4;            // 8-wide k block within 32-deep step
+
+  // acc[quad][mf 0..3][nf 0..1]
+  f32x4 acc[4][4][2];
+#pragma unroll
+  for (int q = 0; q < 4; ++q)
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 2; ++j) acc[q][i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int ksteps = (K + BK8 - 1) / BK8;
+
+  // prologue: tile0 {A-lo, B-lo, B-hi, A-hi} + tile1 {A-lo, B-lo}
+  stage_half8(A, Bt, K, bm, bn, 0, 0, ksteps, smem, w, lane);
+  stage_half8(A, Bt, K, bm, bn, 0, 2, ksteps, smem, w, lane);
+  stage_half8(A, Bt, K, bm, bn, 0, 3, ksteps, smem, w, lane);
+  stage_half8(A, Bt, K, bm, bn, 0, 1, ksteps, smem, w, lane);
+  stage_half8(A, Bt, K, bm, bn, 1, 0, ksteps, smem, w, lane);
+  stage_half8(A, Bt, K, bm, bn, 1, 2, ksteps, smem, w, lane);
+  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");  // tile0 A-lo/B-lo landed
+  __builtin_amdgcn_s_barrier();
+
+  for (int t = 0; t < ksteps; ++t) {
+    const int parity = t & 1;
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      const int rh = (p >> 1) & 1;  // quadrant row half
+      const int ch = p & 1;        // quadrant col half
+      // 12 fragment reads for this quadrant
+      bf16frag aF[4][2], bF[2][2];
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const int cell = ks * 4 + fkb;
+#pragma unroll
+        for (int mf = 0; mf < 4; ++mf)
+          aF[mf][ks] = *frag_ptr(smem, parity, rh,
+                                 wsr * 64 + mf * 16 + fr, cell);
+#pragma unroll
+        for (int nf = 0; nf < 2; ++nf)
+          bF[nf][ks] = *frag_ptr(smem, parity, 2 + ch,
+                                 wsc * 32 + nf * 16 + fr, cell);
+      }
+      // stage the slot this phase frees
+      int st, sh;
+      stage_slot(t, p, &st, &sh);
+      stage_half8(A, Bt, K, bm, bn, st, sh, ksteps, smem, w, lane);
+
+      __builtin_amdgcn_s_barrier();
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 2; ++nf)
+#pragma unroll
+          for (int ks = 0; ks < 2; ++ks)
+            acc[p][mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                aF[mf][ks], bF[nf][ks], acc[p][mf][nf], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+
+  // epilogue: quad (rh, ch); rows rh*128 + wsr*64 + mf*16 + (l>>4)*4 + r
+  const int orow_l = (lane >> 4) * 4;
+  const int ocol_l = lane & 15;
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    const int rh = (p >> 1) & 1, ch = p & 1;
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf) {
+#pragma unroll
+      for (int nf = 0; nf < 2; ++nf) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int row = bm + rh * 128 + wsr * 64 + mf * 16 + orow_l + r;
+          const int col = bn + ch * 128 + wsc * 32 + nf * 16 + ocol_l;
+          if constexpr (BF16_OUT)
+            reinterpret_cast<uint16_t*>(Cout)[(long)row * N + col] =
+                (uint16_t)lpx_f32_to_bf16(acc[p][mf][nf][r]);
+          else
+            reinterpret_cast<float*>(Cout)[(long)row * N + col] =
+                acc[p][mf][nf][r];
+        }
+      }
+    }
+  }
+}
+
+}  // namespace
+
+template <bool BF16_OUT>
+static hipError_t launch8ph(const uint16_t* A, const uint16_t* Bt, void* C,
+                            int M, int N, int K, hipStream_t stream) {
+  const unsigned tiles_m = (unsigned)(M / BM8), tiles_n = (unsigned)(N / BM8);
+  const size_t lds = 2 * PARITY_BYTES;  // 128 KiB
+ hipLaunchKernelGGL(( gemm_bf16_8ph_kernel<BF16_OUT>), dim3(dim3(tiles_m * tiles_n)), dim3(dim3(THREADS8)),
+                                   lds, stream, A, Bt, C, M, N, K, tiles_m,
+                                                  tiles_n);
+  return hipGetLastError();
+}
+
+bool gemm_bf16_8ph_supported(int M, int N, int K) {
+  return M % BM8 == 0 && N % BM8 == 0 && K % BK8 == 0 && K >= BK8;
+}
+
+hipError_t gemm_bf16_nt8ph(const uint16_t* A, const uint16_t* Bt, float* C,
+                           int M, int N, int K, hipStream_t stream) {
+  return launch8ph<false>(A, Bt, C, M, N, K, stream);
+}
+
+hipError_t gemm_bf16_nt8ph_bf16out(const uint16_t* A, const uint16_t* Bt,
+                                   uint16_t* C, int M, int N, int K,
+                                   hipStream_t stream) {
+  return launch8ph<true>(A, Bt, C, M, N, K, stream);
+}
+
+}  // namespace lpx

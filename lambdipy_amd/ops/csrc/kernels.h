@@ -14,6 +14,13 @@ hipError_t gemm_bf16_nt(const uint16_t* A, const uint16_t* Bt, float* C,
 // Same but C stored as bf16.
 hipError_t gemm_bf16_nt_bf16out(const uint16_t* A, const uint16_t* Bt, uint16_t* C,
                                 int M, int N, int K, hipStream_t stream);
+// v3: glds-staged 8-phase pipeline (M,N%256==0, K%64==0).
+bool gemm_bf16_8ph_supported(int M, int N, int K);
+hipError_t gemm_bf16_nt8ph(const uint16_t* A, const uint16_t* Bt, float* C,
+                           int M, int N, int K, hipStream_t stream);
+hipError_t gemm_bf16_nt8ph_bf16out(const uint16_t* A, const uint16_t* Bt,
+                                   uint16_t* C, int M, int N, int K,
+                                   hipStream_t stream);
 // v2: 256x256/BK64 8-wave register-staged pipeline (large shapes).
 hipError_t gemm_bf16_nt256(const uint16_t* A, const uint16_t* Bt, float* C,
                            int M, int N, int K, hipStream_t stream);

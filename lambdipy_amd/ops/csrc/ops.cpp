@@ -40,10 +40,24 @@ at::Tensor gemm_bf16(const at::Tensor& A, const at::Tensor& Bt, bool bf16_out,
   auto C = at::empty({M, N}, A.options().dtype(bf16_out ? at::kBFloat16 : at::kFloat));
   const auto* a = reinterpret_cast<const uint16_t*>(A.data_ptr());
   const auto* b = reinterpret_cast<const uint16_t*>(Bt.data_ptr());
-  // variant: -1 auto (v2 for big tiles), 1 = 128^2 tile, 2 = 256^2 tile
+  // variant: -1 auto, 1 = 128^2, 2 = 256^2 reg-staged, 3 = 256^2 8-phase glds
+  auto s = current_stream(A);
+  const bool want8 = (variant == 3) ||
+                     (variant < 0 && M >= 1024 && N >= 1024 &&
+                      lpx::gemm_bf16_8ph_supported((int)M, (int)N, (int)K));
+  if (want8 && lpx::gemm_bf16_8ph_supported((int)M, (int)N, (int)K)) {
+    if (bf16_out)
+      check_hip(lpx::gemm_bf16_nt8ph_bf16out(a, b, reinterpret_cast<uint16_t*>(C.data_ptr()),
+                                             (int)M, (int)N, (int)K, s),
+                "gemm_bf16_nt8ph");
+    else
+      check_hip(lpx::gemm_bf16_nt8ph(a, b, C.data_ptr<float>(), (int)M, (int)N,
+                                     (int)K, s),
+                "gemm_bf16_nt8ph");
+    return C;
+  }
   const bool big = (variant == 2) ||
                    (variant < 0 && M >= 512 && N >= 512 && K >= 64);
-  auto s = current_stream(A);
   if (big) {
     if (bf16_out)
       check_hip(lpx::gemm_bf16_nt256_bf16out(a, b, reinterpret_cast<uint16_t*>(C.data_ptr()),

@@ -34,9 +34,14 @@ def perf(m, n, k, variant, iters=20):
 
 
 if __name__ == "__main__":
-    for shape in [(256, 256, 256), (512, 512, 64), (300, 700, 100),
-                  (1024, 1024, 1024), (4096, 4096, 4096)]:
-        check(*shape, 2)
-    for v in (1, 2):
+    import sys as _sys
+    variants = [int(x) for x in (_sys.argv[1:] or ["2", "3"])]
+    for v in variants:
+        for shape in [(256, 256, 256), (512, 512, 64), (512, 768, 128),
+                      (1024, 1024, 1024), (4096, 4096, 4096)]:
+            if v == 3 and any(s % 256 for s in shape[:2]) or (v == 3 and shape[2] % 64):
+                continue
+            check(*shape, v)
+    for v in variants:
         perf(4096, 4096, 4096, v)
         perf(8192, 8192, 8192, v, iters=8)
