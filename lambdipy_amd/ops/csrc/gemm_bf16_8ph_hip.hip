@@ -12,8 +12,12 @@
 //   phase 4T+0 reads {A-lo,B-lo}, +1 {A-lo,B-hi}, +2 {A-hi,B-lo}, +3 {A-hi,B-hi}
 //   phase 4T+0 stages (T+1, B-hi), +1 (T+1, A-hi), +2 (T+2, A-lo), +3 (T+2, B-lo)
 // LDS 128 KiB: 2 parities x (A,B) x 2 halves x 16 KiB, lane-linear glds
-// image with the st_16x32 XOR swizzle applied to the SOURCE address and the
-// ds_read address (rule 21; 16-B cell q <-> q ^ (((q>>5)&1)<<1)).
+// image with a per-row ROTATION permutation applied to the SOURCE address
+// and the ds_read address (rule 21): phys 16-B cell = row*8 + (cell+row)%8.
+// Verified conflict-free for the b128 fragment-read lane groups (each
+// group's 16 lanes land on 16 distinct 16-B slots of the 256-B bank row),
+// and the glds write stays lane-linear (no write conflicts); the source
+// still touches whole 128-B row segments, so coalescing is preserved.
 // Requires M,N % 256 == 0 and K % 64 == 0 (launcher falls back to v2).
 
 #include "lpx_common.h"
@@ -29,8 +33,14 @@ constexpr int PARITY_BYTES = 4 * HALF_BYTES;  // A-lo A-hi B-lo B-hi
 
 using bf16frag = bf16x8;
 
-__device__ __forceinline__ int swz_cell(int q) {
-  return q ^ (((q >> 5) & 1) << 1);
+// logical (row, cell) -> physical 16-B cell index (rotation per row)
+__device__ __forceinline__ int phys_cell(int row, int cell) {
+  return row * 8 + ((cell + row) & 7);
+}
+// physical cell q -> logical cell within its row (staging side)
+__device__ __forceinline__ int src_cell(int q) {
+  const int row = q >> 3;
+  return ((q & 7) - row) & 7;
 }
 
 // stage one half-tile: hh 0=A-lo 1=A-hi 2=B-lo 3=B-hi of K-tile `tile`
@@ -50,8 +60,7 @@ __device__ __forceinline__ void stage_half8(const uint16_t* __restrict__ A,
   for (int j = 0; j < 2; ++j) {
     const int qbase = (j * 8 + w) * 64;  // 16-B cell stripe of this wave
     const int q = qbase + lane;
-    const int qs = swz_cell(q);
-    const int r = qs >> 3, cell = qs & 7;
+    const int r = q >> 3, cell = src_cell(q);
     const uint16_t* g = src + (long)(row0 + r) * K + (k0 + cell * 8);
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) void*)g,
@@ -72,10 +81,9 @@ __device__ __forceinline__ void stage_slot(int t, int p, int* tile, int* hh) {
 __device__ __forceinline__ const bf16frag* frag_ptr(const char* __restrict__ smem,
                                                     int parity, int hh,
                                                     int row_in_half, int cell) {
-  const int q = row_in_half * 8 + cell;
-  const int qs = swz_cell(q);
+  const int q = phys_cell(row_in_half, cell);
   return reinterpret_cast<const bf16frag*>(
-      smem + (size_t)parity * PARITY_BYTES + (size_t)hh * HALF_BYTES + qs * 16);
+      smem + (size_t)parity * PARITY_BYTES + (size_t)hh * HALF_BYTES + q * 16);
 }
 
 template <bool BF16_OUT>

@@ -84,3 +84,47 @@ def test_single_process_fanout():
     assert shard.shape == (5, 2) and counts == [5]
     out = fanout.all_gather_rows(ctx, shard, counts)
     assert torch.equal(out, X)
+
+
+SCATTER_WORKER = r"""
+import os, sys
+import torch
+sys.path.insert(0, os.environ["LPX_REPO"])
+from lambdipy_amd.parallel import fanout
+
+ctx = fanout.init_fanout(backend="gloo")
+counts = [3] * ctx.world_size  # weak-scaling shape from bench.py
+F = 4
+X_root = None
+if ctx.is_root:
+    X_root = torch.arange(sum(counts) * F, dtype=torch.float32).reshape(-1, F)
+shard = fanout.scatter_rows(ctx, X_root, counts, (F,), torch.float32)
+assert shard.shape == (3, F), shard.shape
+off = 3 * ctx.rank
+expect = torch.arange(sum(counts) * F, dtype=torch.float32).reshape(-1, F)[off:off+3]
+assert torch.equal(shard, expect), f"rank {ctx.rank} scatter mismatch"
+# 1-D labels all_gather (bench.py's prediction gather)
+labels = (shard[:, 0] * 10).to(torch.int32)
+full = fanout.all_gather_rows(ctx, labels, counts)
+assert full.shape[0] == sum(counts)
+if ctx.is_root:
+    print("SCATTER_OK")
+fanout.destroy()
+"""
+
+
+def test_scatter_rows_gloo():
+    world = 2
+    port = 29710
+    procs = []
+    for r in range(world):
+        env = dict(os.environ, RANK=str(r), LOCAL_RANK=str(r),
+                   WORLD_SIZE=str(world), MASTER_ADDR="127.0.0.1",
+                   MASTER_PORT=str(port), LPX_REPO=REPO)
+        procs.append(subprocess.Popen([sys.executable, "-c", SCATTER_WORKER],
+                                      env=env, stdout=subprocess.PIPE,
+                                      stderr=subprocess.PIPE, text=True))
+    outs = [p.communicate(timeout=180) for p in procs]
+    for p, (o, e) in zip(procs, outs):
+        assert p.returncode == 0, f"{o}\n{e}"
+    assert any("SCATTER_OK" in o for o, _ in outs)
