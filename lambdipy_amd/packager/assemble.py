@@ -214,6 +214,19 @@ def assemble_layer(
     with open(os.path.join(build_dir, "lpx-manifest.json"), "w") as f:
         json.dump(manifest, f, indent=1, sort_keys=True)
 
+    # lockfile: the fully pinned set this layer was built from (reference
+    # analog: the pinned dep list the resolver produced — SURVEY.md §2.1 R2)
+    lock = {
+        "lock_format": 1,
+        "deps": [
+            {"name": d.name, "version": d.version, "source": d.source,
+             "recipe": d.recipe}
+            for d in resolution.deps
+        ],
+    }
+    with open(os.path.join(build_dir, "lpx-lock.json"), "w") as f:
+        json.dump(lock, f, indent=1, sort_keys=True)
+
     if zip_path:
         make_zip(build_dir, zip_path)
     report.sizes = zip_report(build_dir, zip_path)
