@@ -86,7 +86,7 @@ __device__ __forceinline__ const bf16frag* frag_ptr(const char* __restrict__ sme
       smem + (size_t)parity * PARITY_BYTES + (size_t)hh * HALF_BYTES + q * 16);
 }
 
-template <bool BF16_OUT>
+template <bool BF16_OUT, int VM, bool LG0>
 __global__ __launch_bounds__(THREADS8, 2) void gemm_bf16_8ph_kernel(
     const uint16_t* __restrict__ A, const uint16_t* __restrict__ Bt,
     void* __restrict__ Cout, int M, int N, int K, unsigned tiles_m,
@@ -153,7 +153,7 @@ __global__ __launch_bounds__(THREADS8, 2) void gemm_bf16_8ph_kernel(
       stage_half8(A, Bt, K, bm, bn, st, sh, ksteps, smem, w, lane);
 
       __builtin_amdgcn_s_barrier();
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      if constexpr (LG0) asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int mf = 0; mf < 4; ++mf)
@@ -164,7 +164,10 @@ __global__ __launch_bounds__(THREADS8, 2) void gemm_bf16_8ph_kernel(
             acc[p][mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                 aF[mf][ks], bF[nf][ks], acc[p][mf][nf], 0, 0, 0);
       __builtin_amdgcn_s_setprio(0);
-      asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+      if constexpr (VM == 8)
+        asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
       __builtin_amdgcn_s_barrier();
     }
   }
@@ -198,14 +201,14 @@ __global__ __launch_bounds__(THREADS8, 2) void gemm_bf16_8ph_kernel(
 
 }  // namespace
 
-template <bool BF16_OUT>
+template <bool BF16_OUT, int VM = 6, bool LG0 = true>
 static hipError_t launch8ph(const uint16_t* A, const uint16_t* Bt, void* C,
                             int M, int N, int K, hipStream_t stream) {
   const unsigned tiles_m = (unsigned)(M / BM8), tiles_n = (unsigned)(N / BM8);
   const size_t lds = 2 * PARITY_BYTES;  // 128 KiB
- hipLaunchKernelGGL(( gemm_bf16_8ph_kernel<BF16_OUT>), dim3(dim3(tiles_m * tiles_n)), dim3(dim3(THREADS8)),
-                                   lds, stream, A, Bt, C, M, N, K, tiles_m,
-                                                  tiles_n);
+ hipLaunchKernelGGL(( gemm_bf16_8ph_kernel<BF16_OUT, VM, LG0>), dim3(dim3(tiles_m * tiles_n)),
+                                            dim3(dim3(THREADS8)), lds, stream, 
+      A, Bt, C, M, N, K, tiles_m, tiles_n);
   return hipGetLastError();
 }
 
@@ -216,6 +219,18 @@ bool gemm_bf16_8ph_supported(int M, int N, int K) {
 hipError_t gemm_bf16_nt8ph(const uint16_t* A, const uint16_t* Bt, float* C,
                            int M, int N, int K, hipStream_t stream) {
   return launch8ph<false>(A, Bt, C, M, N, K, stream);
+}
+
+// tuning variants for within-probe A/B (scripts/ab_gemm8.py)
+hipError_t gemm_bf16_nt8ph_tune(const uint16_t* A, const uint16_t* Bt,
+                                float* C, int M, int N, int K, int tune,
+                                hipStream_t stream) {
+  switch (tune) {
+    case 1: return launch8ph<false, 8, true>(A, Bt, C, M, N, K, stream);
+    case 2: return launch8ph<false, 6, false>(A, Bt, C, M, N, K, stream);
+    case 3: return launch8ph<false, 8, false>(A, Bt, C, M, N, K, stream);
+    default: return launch8ph<false, 6, true>(A, Bt, C, M, N, K, stream);
+  }
 }
 
 hipError_t gemm_bf16_nt8ph_bf16out(const uint16_t* A, const uint16_t* Bt,

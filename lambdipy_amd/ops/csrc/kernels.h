@@ -21,6 +21,9 @@ hipError_t gemm_bf16_nt8ph(const uint16_t* A, const uint16_t* Bt, float* C,
 hipError_t gemm_bf16_nt8ph_bf16out(const uint16_t* A, const uint16_t* Bt,
                                    uint16_t* C, int M, int N, int K,
                                    hipStream_t stream);
+hipError_t gemm_bf16_nt8ph_tune(const uint16_t* A, const uint16_t* Bt,
+                                float* C, int M, int N, int K, int tune,
+                                hipStream_t stream);
 // v2: 256x256/BK64 8-wave register-staged pipeline (large shapes).
 hipError_t gemm_bf16_nt256(const uint16_t* A, const uint16_t* Bt, float* C,
                            int M, int N, int K, hipStream_t stream);

@@ -330,6 +330,21 @@ at::Tensor rf_predict8_lds(const at::Tensor& nodes8, const at::Tensor& leaf_valu
   return out;
 }
 
+at::Tensor gemm_bf16_tune8(const at::Tensor& A, const at::Tensor& Bt, int64_t tune) {
+  CHECK_DEV(A);
+  CHECK_DEV(Bt);
+  const int64_t M = A.size(0), K = A.size(1), N = Bt.size(0);
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(A.device());
+  auto C = at::empty({M, N}, A.options().dtype(at::kFloat));
+  check_hip(lpx::gemm_bf16_nt8ph_tune(
+                reinterpret_cast<const uint16_t*>(A.data_ptr()),
+                reinterpret_cast<const uint16_t*>(Bt.data_ptr()),
+                C.data_ptr<float>(), (int)M, (int)N, (int)K, (int)tune,
+                current_stream(A)),
+            "gemm_bf16_nt8ph_tune");
+  return C;
+}
+
 std::string build_info() {
   return std::string("lambdipy_amd ops: gfx950 (CDNA4), HIP ") +
          std::to_string(HIP_VERSION_MAJOR) + "." + std::to_string(HIP_VERSION_MINOR);
@@ -368,5 +383,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_fp8", &gemm_fp8, "fp8 e4m3 MFMA GEMM with fused descale/bias/relu",
         py::arg("A"), py::arg("Bt"), py::arg("descale") = 1.0,
         py::arg("bias") = py::none(), py::arg("relu") = false);
+  m.def("gemm_bf16_tune8", &gemm_bf16_tune8, "8-phase GEMM tuning variants",
+        py::arg("A"), py::arg("Bt"), py::arg("tune") = 0);
   m.def("build_info", &build_info);
 }
