@@ -370,6 +370,8 @@ hipError_t im2col_fp8(const float* x, unsigned char* out, int N, int H, int W,
 hipError_t gemm_fp8_nt(const unsigned char* A, const unsigned char* Bt,
                        float* C, const float* bias, int M, int N, int K,
                        float descale, bool relu, hipStream_t stream) {
+  if (M >= 1024 && N >= 1024 && gemm_fp8_8ph_supported(M, N, K))
+    return gemm_fp8_nt8ph(A, Bt, C, bias, M, N, K, descale, relu, stream);
   const bool big = (M >= 512 && N >= 512 && K >= 64);
   if (big) {
     const unsigned tiles_m = lpx_ceil_div(M, BM2), tiles_n = lpx_ceil_div(N, BN2);

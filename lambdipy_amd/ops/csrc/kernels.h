@@ -98,5 +98,10 @@ hipError_t im2col_fp8(const float* x, unsigned char* out, int N, int H, int W,
 hipError_t gemm_fp8_nt(const unsigned char* A, const unsigned char* Bt,
                        float* C, const float* bias, int M, int N, int K,
                        float descale, bool relu, hipStream_t stream);
+// fp8 v3: glds 8-phase pipeline (M,N%256==0, K%128==0).
+bool gemm_fp8_8ph_supported(int M, int N, int K);
+hipError_t gemm_fp8_nt8ph(const unsigned char* A, const unsigned char* Bt,
+                          float* C, const float* bias, int M, int N, int K,
+                          float descale, bool relu, hipStream_t stream);
 
 }  // namespace lpx
