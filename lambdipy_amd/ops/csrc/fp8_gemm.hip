@@ -17,6 +17,7 @@
 // variant (col = l&15, row = (l>>4)*4 + reg).
 
 #include "lpx_common.h"
+#include "kernels.h"
 
 namespace lpx {
 
