@@ -115,14 +115,21 @@ def pairwise_sqeuclidean(X: torch.Tensor, Y: torch.Tensor) -> torch.Tensor:
 
 def rf_predict_proba_sum(nodes: torch.Tensor, leaf_values: torch.Tensor,
                          roots: torch.Tensor, X: torch.Tensor,
-                         packed: bool = False, chunks=None) -> torch.Tensor:
+                         packed: bool = False, chunks=None,
+                         layout: str = None, depth: int = None) -> torch.Tensor:
     """Per-class sums over trees: out [S, C]. Chunks classes by 8 (kernel CMAX).
 
     packed: nodes are the depth-first 8-B layout ([n,2] int32).
+    layout="heap": nodes are level-order internal nodes (complete forests).
     chunks: optional (chunk_tree_begin, max_chunk_nodes) — routes to the
-    forest-in-LDS kernel (fastest; needs packed nodes)."""
+    forest-in-LDS kernel (needs packed nodes)."""
     e = require_extension()
-    if packed and chunks is not None:
+    if layout == "heap":
+        assert depth is not None
+
+        def fn(n, lv, r, x):
+            return e.rf_predict_heap(n, lv, depth, x)
+    elif packed and chunks is not None:
         ctb, max_nodes = chunks
 
         def fn(n, lv, r, x):

@@ -72,6 +72,12 @@ hipError_t rf_predict8_f32(const int* nodes8, const float* leaf_values,
                            const int* roots, int n_trees, const float* X,
                            float* out, int S, int F, int C, int ilp,
                            hipStream_t stream);
+// Heap-layout traversal (v4): complete equal-depth forests, branchless
+// fixed-depth loop; nodes_heap = internal nodes only, level-order per tree.
+hipError_t rf_predict_heap_f32(const int* nodes_heap, const float* leaf_values,
+                               int n_trees, int depth, const float* X,
+                               float* out, int S, int F, int C, int silp,
+                               hipStream_t stream);
 // Forest-in-LDS chunked traversal (v3): whole-tree chunks staged in LDS
 // (host-computed chunk_tree_begin[n_chunks+1]); out accumulates across
 // chunks. One block per CU.

@@ -345,6 +345,33 @@ at::Tensor gemm_bf16_tune8(const at::Tensor& A, const at::Tensor& Bt, int64_t tu
   return C;
 }
 
+at::Tensor rf_predict_heap(const at::Tensor& nodes_heap,
+                           const at::Tensor& leaf_values, int64_t depth,
+                           const at::Tensor& X, int64_t silp) {
+  CHECK_DEV(nodes_heap);
+  CHECK_DEV(leaf_values);
+  CHECK_DEV(X);
+  TORCH_CHECK(nodes_heap.scalar_type() == at::kInt && nodes_heap.dim() == 2 &&
+              nodes_heap.size(1) == 2, "nodes_heap must be int32 [n, 2]");
+  TORCH_CHECK(X.scalar_type() == at::kFloat && X.dim() == 2, "X must be f32 [S,F]");
+  const int64_t S = X.size(0), F = X.size(1), C = leaf_values.size(1);
+  TORCH_CHECK(C <= 8, "rf_predict_heap: C<=8 per call");
+  const int64_t tree_stride = (1LL << depth) - 1;
+  TORCH_CHECK(nodes_heap.size(0) % tree_stride == 0, "nodes/depth mismatch");
+  const int64_t n_trees = nodes_heap.size(0) / tree_stride;
+  TORCH_CHECK(leaf_values.size(0) == n_trees * (1LL << depth),
+              "leaf_values rows != n_trees * 2^depth");
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(X.device());
+  auto out = at::empty({S, C}, X.options());
+  check_hip(lpx::rf_predict_heap_f32(nodes_heap.data_ptr<int>(),
+                                     leaf_values.data_ptr<float>(), (int)n_trees,
+                                     (int)depth, X.data_ptr<float>(),
+                                     out.data_ptr<float>(), (int)S, (int)F,
+                                     (int)C, (int)silp, current_stream(X)),
+            "rf_predict_heap_f32");
+  return out;
+}
+
 std::string build_info() {
   return std::string("lambdipy_amd ops: gfx950 (CDNA4), HIP ") +
          std::to_string(HIP_VERSION_MAJOR) + "." + std::to_string(HIP_VERSION_MINOR);
@@ -367,6 +394,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rf_predict", &rf_predict, "RandomForest per-class sums over trees",
         py::arg("nodes"), py::arg("leaf_values"), py::arg("roots"), py::arg("X"),
         py::arg("ilp") = 0);
+  m.def("rf_predict_heap", &rf_predict_heap,
+        "RandomForest per-class sums (heap layout, complete forests)",
+        py::arg("nodes_heap"), py::arg("leaf_values"), py::arg("depth"),
+        py::arg("X"), py::arg("silp") = 1);
   m.def("rf_predict8_lds", &rf_predict8_lds,
         "RandomForest per-class sums (forest chunks staged in LDS)",
         py::arg("nodes8"), py::arg("leaf_values"), py::arg("roots"),

@@ -266,7 +266,125 @@ __global__ __launch_bounds__(THREADS) void rf_lds_kernel(
   }
 }
 
+// ---- heap-layout kernel (v4, complete equal-depth forests) ---------------
+// Level-order (binary-heap) node layout: tree t's INTERNAL nodes only,
+// contiguous, node i's children at 2i+1 / 2i+2; a sample exits after exactly
+// `depth` steps at leaf row = i - (2^depth - 1). Wins:
+//   - branchless fixed-depth loop (no divergence at all);
+//   - all lanes walk the SAME tree at the SAME depth, and each level is a
+//     contiguous block, so the wave's 64 node loads coalesce into the few
+//     lines of that level (top levels: 1-8 lines for the whole wave) — the
+//     general kernel pays ~1 L2 line per lane per step and is line-bound.
+// node = {u32 feat, f32 thresh} (8 B); leaf values in leaf_values rows
+// [t * 2^depth + leaf_idx]. Used for ForestSoA.random forests (and heapified
+// sklearn trees when padding stays cheap); ragged forests use rf8_kernel.
+
+template <bool STAGE_X, int SILP>
+__global__ __launch_bounds__(THREADS) void rf_heap_kernel(
+    const i32x2* __restrict__ nodes, const float* __restrict__ leaf_values,
+    int n_trees, int depth, const float* __restrict__ X,
+    float* __restrict__ out, int S, int F, int C) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* xl = reinterpret_cast<float*>(smem);
+  const int t = threadIdx.x;
+  const int rowf = F + 1;
+  const int tree_stride = (1 << depth) - 1;  // internal nodes per tree
+  const int n_leaves = 1 << depth;
+
+  for (int tile = blockIdx.x * (THREADS * SILP); tile < S;
+       tile += gridDim.x * (THREADS * SILP)) {
+    const int tile_n = min(THREADS * SILP, S - tile);
+    if (STAGE_X) {
+      const long total = (long)tile_n * F;
+      for (long i = t; i < total; i += THREADS) {
+        const int r = (int)(i / F), c = (int)(i - (long)r * F);
+        xl[r * rowf + c] = X[((long)tile + r) * F + c];
+      }
+      __syncthreads();
+    }
+
+    int s[SILP];
+    const float* xrow[SILP];
+#pragma unroll
+    for (int j = 0; j < SILP; ++j) {
+      s[j] = tile + t + j * THREADS;
+      xrow[j] = STAGE_X ? (xl + (t + j * THREADS) * rowf)
+                        : (X + (long)s[j] * F);
+    }
+    float acc[SILP][CMAX];
+#pragma unroll
+    for (int j = 0; j < SILP; ++j)
+#pragma unroll
+      for (int c = 0; c < CMAX; ++c) acc[j][c] = 0.f;
+
+    for (int tr = 0; tr < n_trees; ++tr) {
+      const i32x2* tb = nodes + (long)tr * tree_stride;
+      int nd[SILP];
+#pragma unroll
+      for (int j = 0; j < SILP; ++j) nd[j] = 0;
+      for (int d = 0; d < depth; ++d) {
+#pragma unroll
+        for (int j = 0; j < SILP; ++j) {
+          if (s[j] < S) {
+            const i32x2 n = tb[nd[j]];
+            const float xv = xrow[j][(unsigned)n.x];
+            nd[j] = 2 * nd[j] + 1 + (xv <= __int_as_float(n.y) ? 0 : 1);
+          }
+        }
+      }
+#pragma unroll
+      for (int j = 0; j < SILP; ++j) {
+        if (s[j] < S) {
+          const int leaf = nd[j] - tree_stride;  // - (2^depth - 1)
+          const float* lv = leaf_values + ((long)tr * n_leaves + leaf) * C;
+#pragma unroll
+          for (int c = 0; c < CMAX; ++c)
+            if (c < C) acc[j][c] += lv[c];
+        }
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < SILP; ++j)
+      if (s[j] < S)
+#pragma unroll
+        for (int c = 0; c < CMAX; ++c)
+          if (c < C) out[(long)s[j] * C + c] = acc[j][c];
+    if (STAGE_X) __syncthreads();
+  }
+}
+
 }  // namespace
+
+hipError_t rf_predict_heap_f32(const int* nodes_heap, const float* leaf_values,
+                               int n_trees, int depth, const float* X,
+                               float* out, int S, int F, int C, int silp,
+                               hipStream_t stream) {
+  if (C > CMAX || depth < 1 || depth > 26) return hipErrorInvalidValue;
+  const int SILP = (silp == 2) ? 2 : (silp == 4 ? 4 : 1);
+  const bool stage = F <= F_LDS_MAX;
+  // X tile covers THREADS*SILP samples when staged
+  const size_t lds = stage ? (size_t)THREADS * SILP * (F + 1) * sizeof(float) : 0;
+  if (lds > 160 * 1024) return hipErrorInvalidValue;
+  unsigned blocks = lpx_ceil_div(S, THREADS * SILP);
+  if (blocks > 4096u) blocks = 4096u;
+  if (blocks == 0) blocks = 1;
+  const auto* nh = reinterpret_cast<const i32x2*>(nodes_heap);
+#define LPX_RFH_CASE(I)                                                       \
+  if (SILP == I) {                                                            \
+    if (stage)                                                                \
+      rf_heap_kernel<true, I><<<dim3(blocks), dim3(THREADS), lds, stream>>>(  \
+          nh, leaf_values, n_trees, depth, X, out, S, F, C);                  \
+    else                                                                      \
+      rf_heap_kernel<false, I><<<dim3(blocks), dim3(THREADS), 0, stream>>>(   \
+          nh, leaf_values, n_trees, depth, X, out, S, F, C);                  \
+    return hipGetLastError();                                                 \
+  }
+  LPX_RFH_CASE(1)
+  LPX_RFH_CASE(2)
+  LPX_RFH_CASE(4)
+#undef LPX_RFH_CASE
+  return hipErrorInvalidValue;
+}
 
 hipError_t rf_predict8_lds_f32(const int* nodes8, const float* leaf_values,
                                const int* roots, const int* chunk_tree_begin,

@@ -40,19 +40,25 @@ class ForestSoA:
 
     def __init__(self, nodes: torch.Tensor, leaf_values: torch.Tensor,
                  roots: torch.Tensor, n_trees: int, n_classes: int,
-                 classes: Optional[np.ndarray] = None, packed: bool = False):
+                 classes: Optional[np.ndarray] = None, packed: bool = False,
+                 layout: Optional[str] = None, depth: Optional[int] = None):
         self.nodes = nodes
         self.leaf_values = leaf_values
         self.roots = roots
         self.n_trees = n_trees
         self.n_classes = n_classes
-        self.packed = packed
+        self.layout = layout or ("packed" if packed else "classic")
+        self.depth = depth
         self.classes = classes if classes is not None else np.arange(n_classes)
+
+    @property
+    def packed(self) -> bool:
+        return self.layout == "packed"
 
     def to(self, device) -> "ForestSoA":
         return ForestSoA(self.nodes.to(device), self.leaf_values.to(device),
                          self.roots.to(device), self.n_trees, self.n_classes,
-                         self.classes, self.packed)
+                         self.classes, layout=self.layout, depth=self.depth)
 
     @property
     def nbytes(self) -> int:
@@ -154,22 +160,27 @@ class ForestSoA:
         return ForestSoA(torch.from_numpy(new_nodes).contiguous(),
                          self.leaf_values, torch.from_numpy(new_roots),
                          self.n_trees, self.n_classes, self.classes,
-                         packed=True)
+                         layout="packed")
 
     @classmethod
     def alloc(cls, n_trees: int, depth: int, n_features: int, n_classes: int,
               device=_DEV) -> "ForestSoA":
         """Empty forest with the shapes random(...) produces (for non-root
         ranks receiving the model via broadcast)."""
-        n_nodes = n_trees * (2 ** (depth + 1) - 1)
         n_leaves = n_trees * (2 ** depth)
-        packed = n_features <= MAX_PACKED_FEAT and depth <= 20
-        shape = (n_nodes, 2) if packed else (n_nodes, 4)
-        return cls(torch.empty(*shape, dtype=torch.int32, device=device),
+        if n_features <= MAX_PACKED_FEAT and depth <= 20:
+            n_internal = n_trees * (2 ** depth - 1)
+            return cls(torch.empty(n_internal, 2, dtype=torch.int32, device=device),
+                       torch.empty(n_leaves, n_classes, dtype=torch.float32,
+                                   device=device),
+                       torch.empty(n_trees, dtype=torch.int32, device=device),
+                       n_trees, n_classes, layout="heap", depth=depth)
+        n_nodes = n_trees * (2 ** (depth + 1) - 1)
+        return cls(torch.empty(n_nodes, 4, dtype=torch.int32, device=device),
                    torch.empty(n_leaves, n_classes, dtype=torch.float32,
                                device=device),
                    torch.empty(n_trees, dtype=torch.int32, device=device),
-                   n_trees, n_classes, packed=packed)
+                   n_trees, n_classes, layout="classic")
 
     @classmethod
     def random(cls, n_trees: int, depth: int, n_features: int, n_classes: int,
@@ -180,43 +191,31 @@ class ForestSoA:
         no per-node Python loop; the pack() conversion was ~250 ms of
         cold-start for the bench forest) when the packed limits allow."""
         if n_features <= MAX_PACKED_FEAT and depth <= 20:
-            return cls._random_packed(n_trees, depth, n_features, n_classes,
-                                      seed, device)
+            return cls._random_heap(n_trees, depth, n_features, n_classes,
+                                    seed, device)
         return cls._random_classic(n_trees, depth, n_features, n_classes,
                                    seed, device)
 
     @classmethod
-    def _random_packed(cls, n_trees, depth, n_features, n_classes, seed,
-                       device):
+    def _random_heap(cls, n_trees, depth, n_features, n_classes, seed,
+                     device):
+        """Complete forest in level-order (heap) layout: internal nodes only
+        ({u32 feat, f32 thresh-bits}); children implicit at 2i+1/2i+2, leaf
+        row = heap index - (2^depth - 1). Fully vectorized generation."""
         rng = np.random.default_rng(seed)
-        # preorder depth array of one complete tree, by recursion on arrays
-        darr = np.zeros(1, dtype=np.int32)
-        for _ in range(depth):
-            darr = np.concatenate([np.zeros(1, dtype=np.int32),
-                                   darr + 1, darr + 1])
-        n = darr.size  # 2^(depth+1) - 1
-        internal = darr < depth
-        delta = np.where(internal, (1 << (depth - darr)).astype(np.uint32), 0)
+        n_internal = (1 << depth) - 1
         n_leaves = 1 << depth
-        total = n_trees * n
+        total = n_trees * n_internal
         feat = rng.integers(0, n_features, total).astype(np.uint32)
         thr = rng.standard_normal(total).astype(np.float32)
-        meta = np.where(np.tile(internal, n_trees),
-                        (feat << np.uint32(21)) | np.tile(delta, n_trees),
-                        np.uint32(0x7FF << 21))
-        leaf_rows = np.arange(n_trees * n_leaves, dtype=np.uint32)
-        val = np.empty(total, dtype=np.uint32)
-        int_mask = np.tile(internal, n_trees)
-        val[int_mask] = thr.view(np.uint32)[int_mask]
-        val[~int_mask] = leaf_rows
-        nodes = np.stack([meta, val], axis=1).view(np.int32)
-        roots = (np.arange(n_trees, dtype=np.int32) * n)
+        nodes = np.stack([feat, thr.view(np.uint32)], axis=1).view(np.int32)
+        roots = np.arange(n_trees, dtype=np.int32) * n_internal
         lv = rng.random((n_trees * n_leaves, n_classes)).astype(np.float32)
         lv /= lv.sum(axis=1, keepdims=True)
         f = cls(torch.from_numpy(nodes).contiguous(),
                 torch.from_numpy(lv).contiguous(),
                 torch.from_numpy(roots).contiguous(), n_trees, n_classes,
-                packed=True)
+                layout="heap", depth=depth)
         return f.to(device) if device else f
 
     @classmethod
@@ -301,11 +300,16 @@ class ForestSoA:
     use_lds_chunks = False
 
     def predict_proba_t(self, X: torch.Tensor) -> torch.Tensor:
-        chunks = (self._plan_chunks(X.shape[1])
-                  if (self.packed and self.use_lds_chunks) else None)
-        sums = _ops.rf_predict_proba_sum(self.nodes, self.leaf_values,
-                                         self.roots, X, packed=self.packed,
-                                         chunks=chunks)
+        if self.layout == "heap":
+            sums = _ops.rf_predict_proba_sum(self.nodes, self.leaf_values,
+                                             self.roots, X, layout="heap",
+                                             depth=self.depth)
+        else:
+            chunks = (self._plan_chunks(X.shape[1])
+                      if (self.packed and self.use_lds_chunks) else None)
+            sums = _ops.rf_predict_proba_sum(self.nodes, self.leaf_values,
+                                             self.roots, X, packed=self.packed,
+                                             chunks=chunks)
         return sums / float(self.n_trees)
 
     def predict_t(self, X: torch.Tensor) -> torch.Tensor:

@@ -62,5 +62,28 @@ def main():
             print(f"  packed ilp={ilp}: {dt*1e3:.3f} ms {(1<<20)/dt/1e6:.0f} Ms/s")
 
 
+def heap_ab():
+    """Heap-layout kernel vs packed global kernel (same logical forest not
+    required — same shape/size workload, plus CPU-reference correctness)."""
+    import numpy as np
+    from tests.test_forest import soa_predict_proba_cpu
+
+    X = torch.randn(1 << 20, 32, device="cuda")
+    for trees, depth in [(256, 10), (512, 12), (128, 8)]:
+        fh = ForestSoA.random(trees, depth, 32, 8, seed=17, device=None)
+        assert fh.layout == "heap"
+        Xs = np.random.default_rng(4).standard_normal((64, 32)).astype(np.float32)
+        ref = soa_predict_proba_cpu(fh, Xs)
+        fg = fh.to("cuda")
+        got = fg.predict_proba_t(torch.from_numpy(Xs).to("cuda")).cpu().numpy()
+        np.testing.assert_allclose(got, ref, rtol=1e-4, atol=1e-5)
+        for silp in (1, 2, 4):
+            dt = timeit(lambda: ops.ext().rf_predict_heap(
+                fg.nodes, fg.leaf_values, depth, X, silp))
+            print(f"PERF rf-heap {trees}x{depth} silp={silp}: {dt*1e3:.3f} ms "
+                  f"{(1<<20)/dt/1e6:.0f} Ms/s")
+
+
 if __name__ == "__main__":
     main()
+    heap_ab()

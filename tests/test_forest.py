@@ -18,6 +18,18 @@ def soa_predict_proba_cpu(f: ForestSoA, X: np.ndarray) -> np.ndarray:
     roots = f.roots.cpu().numpy()
     S = X.shape[0]
     out = np.zeros((S, f.n_classes), dtype=np.float64)
+    if f.layout == "heap":
+        nint = (1 << f.depth) - 1
+        feats = np.ascontiguousarray(nodes[:, 0]).view(np.uint32)
+        for s in range(S):
+            for tr in range(f.n_trees):
+                nd = 0
+                base = tr * nint
+                for _ in range(f.depth):
+                    feat = int(feats[base + nd])
+                    nd = 2 * nd + 1 + (0 if X[s, feat] <= thr_f[base + nd] else 1)
+                out[s] += leaves[tr * (1 << f.depth) + nd - nint]
+        return (out / f.n_trees).astype(np.float32)
     if f.packed:
         meta = np.ascontiguousarray(nodes[:, 0]).view(np.uint32)
         for s in range(S):
@@ -71,11 +83,10 @@ def test_from_sklearn_predict_classes(fitted_forest):
 def test_random_forest_shapes():
     f = ForestSoA.random(n_trees=4, depth=3, n_features=8, n_classes=3,
                          seed=1, device=None)
-    n_nodes_per_tree = 2**4 - 1
-    assert f.packed  # small forest fits the packed 8-B layout
-    assert f.nodes.shape == (4 * n_nodes_per_tree, 2)
+    assert f.layout == "heap"  # complete forest -> heap layout
+    assert f.nodes.shape == (4 * (2**3 - 1), 2)  # internal nodes only
     assert f.leaf_values.shape == (4 * 8, 3)
-    assert f.roots.tolist() == [0, 15, 30, 45]
+    assert f.depth == 3
     # leaf distributions normalized
     s = f.leaf_values.sum(dim=1)
     assert torch.allclose(s, torch.ones_like(s), atol=1e-5)
