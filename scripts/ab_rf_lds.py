@@ -25,7 +25,7 @@ def timeit(fn, iters=15, warmup=3):
 def main():
     X = torch.randn(1 << 20, 32, device="cuda")
     for trees, depth in [(256, 10), (512, 12), (128, 8)]:
-        f = ForestSoA.random(trees, depth, 32, 8, seed=17, device="cuda")
+        f = ForestSoA._random_classic(trees, depth, 32, 8, seed=17, device=None).pack().to("cuda")
         plan = f._plan_chunks(32)
         assert plan is not None, "no chunk plan"
         ctb, mx = plan

@@ -45,7 +45,8 @@ def unpack_classic(fp: ForestSoA) -> ForestSoA:
 def main():
     X = torch.randn(1 << 20, 32, device="cuda")
     for trees, depth in [(256, 10), (512, 12)]:
-        fp = ForestSoA.random(trees, depth, 32, 8, seed=17, device="cuda")
+        fp = ForestSoA._random_classic(trees, depth, 32, 8, seed=17,
+                                       device=None).pack().to("cuda")
         assert fp.packed
         fc = unpack_classic(fp)
         pc = fc.predict_proba_t(X)
