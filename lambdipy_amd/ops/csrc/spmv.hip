@@ -29,13 +29,26 @@ __global__ __launch_bounds__(THREADS) void spmv_wave_row(
   for (int row = wave; row < n_rows; row += nwaves) {
     const int lo = indptr[row], hi = indptr[row + 1];
     float sum = 0.f;
-    for (int j = lo + lane; j < hi; j += 64) sum += data[j] * x[indices[j]];
+    if (((lo & 3) == 0) && (hi - lo) >= 1024) {
+      const int vend = lo + ((hi - lo) & ~3);
+      for (int j = lo + lane * 4; j < vend; j += 64 * 4) {
+        const i32x4 idx = *reinterpret_cast<const i32x4*>(indices + j);
+        const f32x4 d = *reinterpret_cast<const f32x4*>(data + j);
+        sum += d.x * x[idx.x] + d.y * x[idx.y] + d.z * x[idx.z] + d.w * x[idx.w];
+      }
+      for (int j = vend + lane; j < hi; j += 64) sum += data[j] * x[indices[j]];
+    } else {
+      for (int j = lo + lane; j < hi; j += 64) sum += data[j] * x[indices[j]];
+    }
     sum = lpx_wave_allreduce_sum(sum);
     if (lane == 0) y[row] = sum;
   }
 }
 
 // --- 8-lane segments, 8 rows per wave (short rows) ------------------------
+// Rows whose nnz range is 16-B aligned and a multiple of 4 take the
+// vectorized path: each slot loads int4 indices + float4 data (16 B/lane,
+// G13) — 4x fewer load instructions on the streaming data/indices arrays.
 __global__ __launch_bounds__(THREADS) void spmv_seg8(
     const int* __restrict__ indptr, const int* __restrict__ indices,
     const float* __restrict__ data, const float* __restrict__ x,
@@ -50,7 +63,15 @@ __global__ __launch_bounds__(THREADS) void spmv_seg8(
     float sum = 0.f;
     if (row < n_rows) {
       const int lo = indptr[row], hi = indptr[row + 1];
-      for (int j = lo + slot; j < hi; j += 8) sum += data[j] * x[indices[j]];
+      if (((lo & 3) == 0) && (((hi - lo) & 3) == 0)) {
+        for (int j = lo + slot * 4; j < hi; j += 32) {
+          const i32x4 idx = *reinterpret_cast<const i32x4*>(indices + j);
+          const f32x4 d = *reinterpret_cast<const f32x4*>(data + j);
+          sum += d.x * x[idx.x] + d.y * x[idx.y] + d.z * x[idx.z] + d.w * x[idx.w];
+        }
+      } else {
+        for (int j = lo + slot; j < hi; j += 8) sum += data[j] * x[indices[j]];
+      }
     }
     // reduce within the 8-lane segment
 #pragma unroll
