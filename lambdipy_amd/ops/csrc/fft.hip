@@ -24,6 +24,9 @@ __device__ __forceinline__ f32x2 cmul(f32x2 a, f32x2 b) {
   return f32x2{a.x * b.x - a.y * b.y, a.x * b.y + a.y * b.x};
 }
 
+// Stockham in LDS, radix-4 stages (+ one radix-2 stage when log2(n) is odd).
+// Half the stage count / barrier count of pure radix-2 (OTFFT-style
+// recurrence; twiddles w1 = tw[p*s], w2 = tw[2*p*s], w3 = w1*w2).
 template <bool INV>
 __global__ __launch_bounds__(THREADS) void fft_lds_kernel(
     const float* __restrict__ xg, float* __restrict__ outg,
@@ -34,31 +37,62 @@ __global__ __launch_bounds__(THREADS) void fft_lds_kernel(
 
   const int t = threadIdx.x;
   const int half = n >> 1;
+  const int quarter = n >> 2;
+  const f32x2* twc = reinterpret_cast<const f32x2*>(tw);
 
   for (int b = blockIdx.x; b < batch; b += gridDim.x) {
     const f32x2* src = reinterpret_cast<const f32x2*>(xg) + (long)b * n;
-    // load (coalesced 8 B/lane)
     for (int i = t; i < n; i += THREADS) buf0[i] = src[i];
     __syncthreads();
 
     f32x2* x = buf0;
     f32x2* y = buf1;
     int s = 1;
-    for (int stage = 0; stage < log2n; ++stage) {
-      const int m = n >> (stage + 1);  // half-length at this stage
-      for (int u = t; u < half; u += THREADS) {
+    int n_rem = n;
+    while (n_rem >= 4) {
+      const int m = n_rem >> 2;
+      for (int u = t; u < quarter; u += THREADS) {
         const int p = u / s;
-        const int q = u - p * s;  // u % s
+        const int q = u - p * s;
         const f32x2 a = x[q + s * p];
         const f32x2 bv = x[q + s * (p + m)];
-        f32x2 w = reinterpret_cast<const f32x2*>(tw)[(long)p * s];
+        const f32x2 c = x[q + s * (p + 2 * m)];
+        const f32x2 d = x[q + s * (p + 3 * m)];
+        f32x2 w1 = twc[(long)p * s];
+        f32x2 w2 = twc[(long)2 * p * s];
+        if (INV) { w1.y = -w1.y; w2.y = -w2.y; }
+        const f32x2 w3 = cmul(w1, w2);
+        const f32x2 apc = {a.x + c.x, a.y + c.y};
+        const f32x2 amc = {a.x - c.x, a.y - c.y};
+        const f32x2 bpd = {bv.x + d.x, bv.y + d.y};
+        // jbmd = j*(b-d) for forward, -j*(b-d) for inverse
+        f32x2 jbmd = {-(bv.y - d.y), bv.x - d.x};
+        if (INV) jbmd = f32x2{-jbmd.x, -jbmd.y};
+        y[q + s * (4 * p + 0)] = f32x2{apc.x + bpd.x, apc.y + bpd.y};
+        y[q + s * (4 * p + 1)] =
+            cmul(f32x2{amc.x - jbmd.x, amc.y - jbmd.y}, w1);
+        y[q + s * (4 * p + 2)] = cmul(f32x2{apc.x - bpd.x, apc.y - bpd.y}, w2);
+        y[q + s * (4 * p + 3)] =
+            cmul(f32x2{amc.x + jbmd.x, amc.y + jbmd.y}, w3);
+      }
+      __syncthreads();
+      f32x2* tmp = x; x = y; y = tmp;
+      s <<= 2;
+      n_rem >>= 2;
+    }
+    if (n_rem == 2) {  // final radix-2 stage (odd log2n); m = 1
+      for (int u = t; u < half; u += THREADS) {
+        const int p = u / s;
+        const int q = u - p * s;
+        const f32x2 a = x[q + s * p];
+        const f32x2 bv = x[q + s * (p + 1)];
+        f32x2 w = twc[(long)p * s];
         if (INV) w.y = -w.y;
         y[q + s * (2 * p)] = f32x2{a.x + bv.x, a.y + bv.y};
         y[q + s * (2 * p + 1)] = cmul(f32x2{a.x - bv.x, a.y - bv.y}, w);
       }
       __syncthreads();
       f32x2* tmp = x; x = y; y = tmp;
-      s <<= 1;
     }
 
     f32x2* dst_lds = x;  // result lives in x after the final swap
