@@ -290,3 +290,30 @@ def test_resnet50_fp8_forward(ops):
     assert torch.isfinite(y1).all()
     y2 = model(x)
     assert torch.equal(y1, y2)  # deterministic inference
+
+
+def test_rf_heap_vs_cpu_reference(ops):
+    from lambdipy_amd.shims.lpx_sklearn import ForestSoA
+    from tests.test_forest import soa_predict_proba_cpu
+
+    f = ForestSoA.random(n_trees=17, depth=7, n_features=12, n_classes=5,
+                         seed=21, device=None)
+    assert f.layout == "heap"
+    rng = np.random.default_rng(8)
+    X = rng.standard_normal((300, 12)).astype(np.float32)
+    ref = soa_predict_proba_cpu(f, X)
+    got = f.to(DEV).predict_proba_t(torch.from_numpy(X).to(DEV)).cpu().numpy()
+    np.testing.assert_allclose(got, ref, rtol=1e-4, atol=1e-5)
+
+
+def test_rf_heap_deterministic_and_silp(ops):
+    from lambdipy_amd.shims.lpx_sklearn import ForestSoA
+
+    f = ForestSoA.random(n_trees=64, depth=9, n_features=32, n_classes=8,
+                         seed=2, device=DEV)
+    X = _rand((65536, 32))
+    e = ops.ext()
+    p1 = e.rf_predict_heap(f.nodes, f.leaf_values, f.depth, X, 1)
+    p2 = e.rf_predict_heap(f.nodes, f.leaf_values, f.depth, X, 2)
+    p4 = e.rf_predict_heap(f.nodes, f.leaf_values, f.depth, X, 4)
+    assert torch.equal(p1, p2) and torch.equal(p1, p4)
