@@ -346,7 +346,167 @@ __global__ __launch_bounds__(THREADS2, 2) void gemm_fp8_nt256_kernel(
   }
 }
 
+// ---- implicit-GEMM fp8 conv (v2 structure) -------------------------------
+// out[m = (n,ho,wo)][ko] = sum_{r,s,c} xq[n][ho*st+r-p][wo*st+s-p][c] * w[ko][r,s,c]
+// The A staging gathers 16-B chunks of pre-quantized NHWC activations
+// directly (C % 16 == 0 so a chunk never crosses an (r,s) cell; spatial
+// padding and row tails are zero-filled) — no materialized patch matrix,
+// which saves writing+reading R*S x the activation bytes of im2col.
+
+struct ConvGeom {
+  int N, H, W, C, R, S, HO, WO, stride, pad;
+};
+
+__device__ __forceinline__ void stage_imp(const unsigned char* __restrict__ xq,
+                                          const ConvGeom g, int M, int K,
+                                          int row0, int k0,
+                                          i32x4 (&regs)[CHUNKS2]) {
+  const int t = threadIdx.x;
+#pragma unroll
+  for (int i = 0; i < CHUNKS2; ++i) {
+    const int c = t + i * THREADS2;
+    const int rt = c >> 2;               // tile row
+    const int cell = (c & 3) * 16;       // 16-B cell within BK
+    const int m = row0 + rt;
+    const int k = k0 + cell;
+    i32x4 v = {0, 0, 0, 0};
+    if (m < M && k + 16 <= K) {
+      const int c0 = k % g.C;
+      const int rs = k / g.C;
+      const int s = rs % g.S;
+      const int r = rs / g.S;
+      const int wo = m % g.WO;
+      const int ho = (m / g.WO) % g.HO;
+      const int n = m / (g.WO * g.HO);
+      const int hi = ho * g.stride + r - g.pad;
+      const int wi = wo * g.stride + s - g.pad;
+      if (hi >= 0 && hi < g.H && wi >= 0 && wi < g.W)
+        v = *reinterpret_cast<const i32x4*>(
+            xq + (((long)n * g.H + hi) * g.W + wi) * g.C + c0);
+    }
+    regs[i] = v;
+  }
+}
+
+template <bool RELU, bool BIAS>
+__global__ __launch_bounds__(THREADS2, 2) void conv_fp8_imp_kernel(
+    const unsigned char* __restrict__ xq, const unsigned char* __restrict__ Wq,
+    float* __restrict__ C, const float* __restrict__ bias, ConvGeom g, int M,
+    int N, int K, float descale, unsigned tiles_m, unsigned tiles_n) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* const lA = smem;
+  char* const lB = smem + BM2 * ROWB2;
+
+  const unsigned nwg = tiles_m * tiles_n;
+  const unsigned bid = lpx_xcd_swizzle(blockIdx.x, nwg);
+  const int bm = (int)(bid / tiles_n) * BM2;
+  const int bn = (int)(bid % tiles_n) * BN2;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wr = wid >> 2, wc = wid & 3;
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int frag_row = lane & 15;
+  const int frag_kb = lane >> 4;
+
+  i32x4 aRegs[CHUNKS2], bRegs[CHUNKS2];
+  stage_imp(xq, g, M, K, bm, 0, aRegs);
+  load_fp8_regs(Wq, N, K, bn, 0, bRegs);
+  write_fp8_lds(aRegs, lA);
+  write_fp8_lds(bRegs, lB);
+  __syncthreads();
+
+  const int ksteps = (K + BK2 - 1) / BK2;
+  for (int ks = 0; ks < ksteps; ++ks) {
+    if (ks + 1 < ksteps) {
+      stage_imp(xq, g, M, K, bm, (ks + 1) * BK2, aRegs);
+      load_fp8_regs(Wq, N, K, bn, (ks + 1) * BK2, bRegs);
+    }
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      const int off = kk * 32 + frag_kb * 8;
+      fp8frag aF[8], bF[4];
+#pragma unroll
+      for (int mf = 0; mf < 8; ++mf) {
+        const int r = wr * 128 + mf * 16 + frag_row;
+        aF[mf] = *reinterpret_cast<const fp8frag*>(lA + r * ROWB2 + off);
+      }
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf) {
+        const int r = wc * 64 + nf * 16 + frag_row;
+        bF[nf] = *reinterpret_cast<const fp8frag*>(lB + r * ROWB2 + off);
+      }
+#pragma unroll
+      for (int mf = 0; mf < 8; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 4; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+              __builtin_bit_cast(long, aF[mf]), __builtin_bit_cast(long, bF[nf]),
+              acc[mf][nf], 0, 0, 0);
+    }
+    __syncthreads();
+    if (ks + 1 < ksteps) {
+      write_fp8_lds(aRegs, lA);
+      write_fp8_lds(bRegs, lB);
+    }
+    __syncthreads();
+  }
+
+  const int orow0 = bm + wr * 128 + (lane >> 4) * 4;
+  const int ocol0 = bn + wc * 64 + (lane & 15);
+#pragma unroll
+  for (int mf = 0; mf < 8; ++mf) {
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = orow0 + mf * 16 + r;
+        const int col = ocol0 + nf * 16;
+        if (row < M && col < N) {
+          float v = acc[mf][nf][r] * descale;
+          if constexpr (BIAS) v += bias[col];
+          if constexpr (RELU) v = v > 0.f ? v : 0.f;
+          C[(long)row * N + col] = v;
+        }
+      }
+    }
+  }
+}
+
 }  // namespace
+
+hipError_t conv_fp8_implicit(const unsigned char* xq, const unsigned char* Wq,
+                             float* C, const float* bias, int Nb, int H, int W,
+                             int Cin, int R, int S, int HO, int WO, int stride,
+                             int pad, int KO, float descale, bool relu,
+                             hipStream_t stream) {
+  if (Cin % 16 != 0) return hipErrorInvalidValue;  // chunk alignment
+  ConvGeom g{Nb, H, W, Cin, R, S, HO, WO, stride, pad};
+  const int M = Nb * HO * WO;
+  const int K = R * S * Cin;
+  const unsigned tiles_m = lpx_ceil_div(M, BM2), tiles_n = lpx_ceil_div(KO, BN2);
+  const size_t lds = (size_t)2 * BM2 * ROWB2;
+  dim3 grid(tiles_m * tiles_n), block(THREADS2);
+  if (bias && relu)
+   hipLaunchKernelGGL(( conv_fp8_imp_kernel<true, true>), dim3(grid), dim3(block), lds, stream, 
+        xq, Wq, C, bias, g, M, KO, K, descale, tiles_m, tiles_n);
+  else if (bias)
+   hipLaunchKernelGGL(( conv_fp8_imp_kernel<false, true>), dim3(grid), dim3(block), lds, stream, 
+        xq, Wq, C, bias, g, M, KO, K, descale, tiles_m, tiles_n);
+  else if (relu)
+   hipLaunchKernelGGL(( conv_fp8_imp_kernel<true, false>), dim3(grid), dim3(block), lds, stream, 
+        xq, Wq, C, bias, g, M, KO, K, descale, tiles_m, tiles_n);
+  else
+   hipLaunchKernelGGL(( conv_fp8_imp_kernel<false, false>), dim3(grid), dim3(block), lds, stream, 
+        xq, Wq, C, bias, g, M, KO, K, descale, tiles_m, tiles_n);
+  return hipGetLastError();
+}
 
 hipError_t quantize_fp8(const float* x, unsigned char* out, long n,
                         float inv_scale, hipStream_t stream) {

@@ -104,6 +104,13 @@ hipError_t im2col_fp8(const float* x, unsigned char* out, int N, int H, int W,
 hipError_t gemm_fp8_nt(const unsigned char* A, const unsigned char* Bt,
                        float* C, const float* bias, int M, int N, int K,
                        float descale, bool relu, hipStream_t stream);
+// Implicit-GEMM fp8 conv: xq NHWC e4m3 (pre-quantized), Wq [KO, R*S*C];
+// out f32 [N*HO*WO, KO]; requires C % 16 == 0.
+hipError_t conv_fp8_implicit(const unsigned char* xq, const unsigned char* Wq,
+                             float* C, const float* bias, int Nb, int H, int W,
+                             int Cin, int R, int S, int HO, int WO, int stride,
+                             int pad, int KO, float descale, bool relu,
+                             hipStream_t stream);
 // fp8 v3: glds 8-phase pipeline (M,N%256==0, K%128==0).
 bool gemm_fp8_8ph_supported(int M, int N, int K);
 hipError_t gemm_fp8_nt8ph(const unsigned char* A, const unsigned char* Bt,

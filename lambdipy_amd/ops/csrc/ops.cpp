@@ -372,6 +372,36 @@ at::Tensor rf_predict_heap(const at::Tensor& nodes_heap,
   return out;
 }
 
+at::Tensor conv_fp8(const at::Tensor& xq, const at::Tensor& Wq,
+                    int64_t H, int64_t W, int64_t Cin, int64_t R, int64_t S,
+                    int64_t stride, int64_t pad, double descale,
+                    const c10::optional<at::Tensor>& bias, bool relu) {
+  CHECK_DEV(xq);
+  CHECK_DEV(Wq);
+  TORCH_CHECK(xq.scalar_type() == at::kByte && Wq.scalar_type() == at::kByte,
+              "conv_fp8 expects e4m3 bytes");
+  TORCH_CHECK(Cin % 16 == 0, "conv_fp8 requires C % 16 == 0");
+  const int64_t Nb = xq.numel() / (H * W * Cin);
+  const int64_t KO = Wq.size(0);
+  TORCH_CHECK(Wq.size(1) == R * S * Cin, "Wq must be [KO, R*S*C]");
+  const int64_t HO = (H + 2 * pad - R) / stride + 1;
+  const int64_t WO = (W + 2 * pad - S) / stride + 1;
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(xq.device());
+  auto out = at::empty({Nb * HO * WO, KO}, xq.options().dtype(at::kFloat));
+  const float* bptr = nullptr;
+  if (bias.has_value()) {
+    CHECK_DEV(bias.value());
+    bptr = bias->data_ptr<float>();
+  }
+  check_hip(lpx::conv_fp8_implicit(xq.data_ptr<uint8_t>(), Wq.data_ptr<uint8_t>(),
+                                   out.data_ptr<float>(), bptr, (int)Nb, (int)H,
+                                   (int)W, (int)Cin, (int)R, (int)S, (int)HO,
+                                   (int)WO, (int)stride, (int)pad, (int)KO,
+                                   (float)descale, relu, current_stream(xq)),
+            "conv_fp8_implicit");
+  return out;
+}
+
 std::string build_info() {
   return std::string("lambdipy_amd ops: gfx950 (CDNA4), HIP ") +
          std::to_string(HIP_VERSION_MAJOR) + "." + std::to_string(HIP_VERSION_MINOR);
@@ -411,6 +441,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("im2col_fp8", &im2col_fp8, "NHWC f32 -> [M, R*S*C] e4m3 patches",
         py::arg("x"), py::arg("R"), py::arg("S"), py::arg("stride"),
         py::arg("pad"), py::arg("scale") = 1.0);
+  m.def("conv_fp8", &conv_fp8, "implicit-GEMM fp8 conv (NHWC e4m3)",
+        py::arg("xq"), py::arg("Wq"), py::arg("H"), py::arg("W"), py::arg("Cin"),
+        py::arg("R"), py::arg("S"), py::arg("stride"), py::arg("pad"),
+        py::arg("descale") = 1.0, py::arg("bias") = py::none(),
+        py::arg("relu") = false);
   m.def("gemm_fp8", &gemm_fp8, "fp8 e4m3 MFMA GEMM with fused descale/bias/relu",
         py::arg("A"), py::arg("Bt"), py::arg("descale") = 1.0,
         py::arg("bias") = py::none(), py::arg("relu") = false);

@@ -58,15 +58,21 @@ class Conv2dFP8:
         N, H, W, C = x.shape
         e = _ops.ext()
         sa = _act_scale(x)
+        ho = (H + 2 * self.pad - self.k) // self.stride + 1
+        wo = (W + 2 * self.pad - self.k) // self.stride + 1
         if self.k == 1 and self.stride == 1 and self.pad == 0:
             aq = e.quantize_fp8(x.reshape(-1, C).contiguous(), sa)
-            ho, wo = H, W
+            out = e.gemm_fp8(aq, self.wq, sa * self.wscale, self.bias, self.relu)
+        elif C % 16 == 0:
+            # implicit GEMM: quantize NHWC once, gather patches in staging
+            xq = e.quantize_fp8(x.contiguous(), sa)
+            out = e.conv_fp8(xq, self.wq, H, W, C, self.k, self.k, self.stride,
+                             self.pad, sa * self.wscale, self.bias, self.relu)
         else:
+            # first layer (C=3): materialized im2col patch matrix
             aq = e.im2col_fp8(x.contiguous(), self.k, self.k, self.stride,
                               self.pad, sa)
-            ho = (H + 2 * self.pad - self.k) // self.stride + 1
-            wo = (W + 2 * self.pad - self.k) // self.stride + 1
-        out = e.gemm_fp8(aq, self.wq, sa * self.wscale, self.bias, self.relu)
+            out = e.gemm_fp8(aq, self.wq, sa * self.wscale, self.bias, self.relu)
         return out.view(N, ho, wo, self.cout)
 
 

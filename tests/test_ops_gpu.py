@@ -317,3 +317,34 @@ def test_rf_heap_deterministic_and_silp(ops):
     p2 = e.rf_predict_heap(f.nodes, f.leaf_values, f.depth, X, 2)
     p4 = e.rf_predict_heap(f.nodes, f.leaf_values, f.depth, X, 4)
     assert torch.equal(p1, p2) and torch.equal(p1, p4)
+
+
+def test_conv_fp8_implicit_vs_im2col(ops):
+    e = ops.ext()
+    g = torch.Generator().manual_seed(7)
+    N, H, W, C = 2, 14, 14, 32
+    KO, R, stride, pad = 64, 3, 2, 1
+    x = torch.randn(N, H, W, C, generator=g).to(DEV)
+    w = torch.randn(KO, R * R * C, generator=g).to(DEV)
+    xq = e.quantize_fp8(x, 1.0)
+    wq = e.quantize_fp8(w, 1.0)
+    got = e.conv_fp8(xq, wq, H, W, C, R, R, stride, pad, 1.0)
+    aq = e.im2col_fp8(x, R, R, stride, pad, 1.0)
+    # im2col quantizes from f32; quantize-then-gather must agree exactly here
+    ref = e.gemm_fp8(aq, wq, 1.0)
+    assert torch.allclose(got, ref, atol=1e-4), (got - ref).abs().max()
+
+
+def test_conv_fp8_bias_relu_stride1(ops):
+    e = ops.ext()
+    g = torch.Generator().manual_seed(8)
+    N, H, W, C = 1, 8, 8, 16
+    KO, R = 32, 3
+    x = torch.randn(N, H, W, C, generator=g).to(DEV)
+    w = torch.randn(KO, R * R * C, generator=g).to(DEV)
+    bias = torch.randn(KO, generator=g).to(DEV)
+    xq, wq = e.quantize_fp8(x, 1.0), e.quantize_fp8(w, 1.0)
+    got = e.conv_fp8(xq, wq, H, W, C, R, R, 1, 1, 1.0, bias, True)
+    aq = e.im2col_fp8(x, R, R, 1, 1, 1.0)
+    ref = e.gemm_fp8(aq, wq, 1.0, bias, True)
+    assert torch.allclose(got, ref, atol=1e-4)
