@@ -21,11 +21,19 @@ def main():
         ops.gemm_bf16(a, b, variant=2)
     torch.cuda.synchronize()
 
-    # RF traversal (config 4, the metric workload)
+    # RF traversal (config 4, the metric workload; heap-layout kernel)
     f = ForestSoA.random(256, 10, 32, 8, seed=17, device=dev)
     X = torch.randn(1 << 20, 32, device=dev)
     for _ in range(5):
         f.predict_proba_t(X)
+    torch.cuda.synchronize()
+
+    # implicit-GEMM fp8 conv (config 5 hot op)
+    e0 = ops.ext()
+    xq = e0.quantize_fp8(torch.randn(64, 28, 28, 256, device=dev), 0.05)
+    wq = e0.quantize_fp8(torch.randn(256, 3 * 3 * 256, device=dev), 0.05)
+    for _ in range(5):
+        e0.conv_fp8(xq, wq, 28, 28, 256, 3, 3, 1, 1, 0.0025)
     torch.cuda.synchronize()
 
     # fp8 GEMM (config 5)
