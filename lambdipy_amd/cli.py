@@ -168,6 +168,41 @@ def bench(layer, handler, gpus, repeats, out, rocprof_dir):
 
 
 @main.command()
+@click.argument("name")
+@click.option("-o", "--output", default=None, help="destination zip path")
+@click.option("--release-dir", default=DEFAULT_RELEASE_DIR, show_default=True)
+def fetch(name, output, release_dir):
+    """Fetch a published layer from the local artifact store.
+
+    Offline analog of the reference's prebuilt GitHub-Releases download
+    (SURVEY.md §2.1 R4): picks the newest artifact whose name matches,
+    verifies its content hash against the stored metadata, and copies it."""
+    import hashlib
+
+    candidates = sorted(
+        (f for f in os.listdir(release_dir) if f.endswith(".zip")
+         and f.startswith(name)),
+        key=lambda f: os.path.getmtime(os.path.join(release_dir, f)),
+    ) if os.path.isdir(release_dir) else []
+    if not candidates:
+        raise click.ClickException(f"no published artifact matches {name!r}")
+    src = os.path.join(release_dir, candidates[-1])
+    meta_path = src + ".json"
+    if os.path.exists(meta_path):
+        with open(meta_path) as f:
+            meta = json.load(f)
+        h = hashlib.sha256()
+        with open(src, "rb") as f:
+            for chunk in iter(lambda: f.read(1 << 20), b""):
+                h.update(chunk)
+        if h.hexdigest()[:16] != meta.get("sha256_16"):
+            raise click.ClickException(f"hash mismatch for {src} (corrupt store)")
+    dest = output or os.path.basename(src)
+    shutil.copy2(src, dest)
+    click.echo(f"fetched {dest}")
+
+
+@main.command()
 def recipes():
     """List the recipe registry."""
     from .recipes import get_registry

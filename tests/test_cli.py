@@ -53,3 +53,26 @@ def test_publish(runner, tmp_path):
     r = runner.invoke(cli_main, ["publish", str(zp), "--release-dir", str(rel)])
     assert r.exit_code == 0, r.output
     assert any(f.endswith(".zip") for f in os.listdir(rel))
+
+
+def test_publish_fetch_roundtrip(runner, tmp_path):
+    import zipfile
+
+    zp = tmp_path / "mylayer.zip"
+    with zipfile.ZipFile(zp, "w") as zf:
+        zf.writestr("m.txt", "content")
+    rel = str(tmp_path / "rel")
+    r = runner.invoke(cli_main, ["publish", str(zp), "--release-dir", rel])
+    assert r.exit_code == 0, r.output
+    out = str(tmp_path / "fetched.zip")
+    r = runner.invoke(cli_main, ["fetch", "mylayer", "-o", out,
+                                 "--release-dir", rel])
+    assert r.exit_code == 0, r.output
+    with zipfile.ZipFile(out) as zf:
+        assert zf.read("m.txt") == b"content"
+
+
+def test_fetch_missing(runner, tmp_path):
+    r = runner.invoke(cli_main, ["fetch", "nope", "--release-dir",
+                                 str(tmp_path / "empty")])
+    assert r.exit_code != 0
