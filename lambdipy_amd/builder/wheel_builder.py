@@ -42,6 +42,30 @@ def ops_extension_path() -> Optional[str]:
     return hits[0] if hits else None
 
 
+HSACO_PATH = os.path.join(OPS_DIR, "standalone_gfx950.hsaco")
+HSACO_SRC = os.path.join(OPS_DIR, "csrc", "standalone.hip")
+
+
+def ensure_standalone_hsaco(allow_build: bool = True) -> str:
+    """AOT code object for the torch-free runtime (hipcc --genco, gfx950).
+
+    Rebuilt when missing or older than its source; ships in-tree (the .hsaco
+    is git-ignored but travels with snapshots / into recipe wheels)."""
+    if os.path.exists(HSACO_PATH) and (
+        os.path.getmtime(HSACO_PATH) >= os.path.getmtime(HSACO_SRC)
+    ):
+        return HSACO_PATH
+    if not allow_build:
+        raise RuntimeError("standalone hsaco missing and allow_build=False")
+    log.info("building standalone gfx950 code object (hipcc --genco)...")
+    subprocess.run(
+        ["hipcc", "--genco", "--offload-arch=gfx950", "-O3", HSACO_SRC,
+         "-o", HSACO_PATH],
+        check=True,
+    )
+    return HSACO_PATH
+
+
 def ensure_ops_extension(allow_build: bool = True) -> str:
     """Return the built extension path, building it in-tree if needed."""
     p = ops_extension_path()
@@ -69,14 +93,17 @@ def _read(path: str) -> bytes:
 
 
 def emit_ops_wheel(dist_dir: str) -> str:
-    """Emit (or reuse) the lpx_ops ROCm wheel: loader + AOT gfx950 .so."""
+    """Emit (or reuse) the lpx_ops ROCm wheel: loader + AOT gfx950 .so +
+    the standalone .hsaco for the torch-free cold-start runtime."""
     from .wheelgen import write_wheel
 
     so = ensure_ops_extension()
+    hsaco = ensure_standalone_hsaco()
     files = {
         "lpx_ops/__init__.py": _read(os.path.join(OPS_DIR, "__init__.py")),
         "lpx_ops/api.py": _read(os.path.join(OPS_DIR, "api.py")),
         f"lpx_ops/{os.path.basename(so)}": _read(so),
+        f"lpx_ops/{os.path.basename(hsaco)}": _read(hsaco),
     }
     return write_wheel(
         dist_dir, "lpx_ops", files, platform_specific=True,

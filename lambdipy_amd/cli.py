@@ -39,9 +39,11 @@ def _install_handlers(build_dir: str, handlers):
         if not os.path.exists(src):
             raise click.ClickException(f"unknown handler {h!r}")
         shutil.copy2(src, os.path.join(dest, mod + ".py"))
-    # fan-out support ships as a top-level module
+    # fan-out + torch-free runtime support ship as top-level modules
     shutil.copy2(os.path.join(pkg, "parallel", "fanout.py"),
                  os.path.join(build_dir, "lpx_fanout.py"))
+    shutil.copy2(os.path.join(pkg, "shims", "lpx_rt.py"),
+                 os.path.join(build_dir, "lpx_rt.py"))
 
 
 @click.group()

@@ -1,0 +1,111 @@
+// Standalone code objects for the torch-free cold-start runtime (lpx_rt).
+//
+// Compiled with `hipcc --genco --offload-arch=gfx950` into
+// standalone_gfx950.hsaco and loaded at cold start via hipModuleLoad by the
+// ctypes shim (shims/lpx_rt.py) — no JIT, no libtorch in the critical path
+// ("loads precompiled .hsaco at import; cold-start ms is
+// the metric"). Kernels mirror the torch-extension versions (rf.hip):
+//   - lpx_rf_heap:   heap-layout RandomForest traversal (complete forests)
+//   - lpx_argmax_rows: per-row argmax of [S, C] f32 -> int32 labels
+//   - lpx_xgen:      synthetic ~N(0,1) feature matrix (xorshift + Box-Muller)
+
+#include <hip/hip_runtime.h>
+#include <stdint.h>
+
+#define THREADS 256
+#define CMAX 8
+#define F_LDS_MAX 63
+
+typedef __attribute__((ext_vector_type(2))) int i32x2;
+
+extern "C" __global__ __launch_bounds__(THREADS) void lpx_rf_heap(
+    const i32x2* __restrict__ nodes, const float* __restrict__ leaf_values,
+    int n_trees, int depth, const float* __restrict__ X,
+    float* __restrict__ out, int S, int F, int C) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* xl = reinterpret_cast<float*>(smem);
+  const int t = threadIdx.x;
+  const int rowf = F + 1;
+  const int tree_stride = (1 << depth) - 1;
+  const int n_leaves = 1 << depth;
+  const bool stage = F <= F_LDS_MAX;
+
+  for (int tile = blockIdx.x * THREADS; tile < S; tile += gridDim.x * THREADS) {
+    const int s = tile + t;
+    const int tile_n = min(THREADS, S - tile);
+    if (stage) {
+      const long total = (long)tile_n * F;
+      for (long i = t; i < total; i += THREADS) {
+        const int r = (int)(i / F), c = (int)(i - (long)r * F);
+        xl[r * rowf + c] = X[((long)tile + r) * F + c];
+      }
+      __syncthreads();
+    }
+    if (s < S) {
+      const float* xrow = stage ? (xl + t * rowf) : (X + (long)s * F);
+      float acc[CMAX];
+#pragma unroll
+      for (int c = 0; c < CMAX; ++c) acc[c] = 0.f;
+      for (int tr = 0; tr < n_trees; ++tr) {
+        const i32x2* tb = nodes + (long)tr * tree_stride;
+        int nd = 0;
+        for (int d = 0; d < depth; ++d) {
+          const i32x2 n = tb[nd];
+          const float xv = xrow[(unsigned)n.x];
+          nd = 2 * nd + 1 + (xv <= __int_as_float(n.y) ? 0 : 1);
+        }
+        const float* lv =
+            leaf_values + ((long)tr * n_leaves + nd - tree_stride) * C;
+#pragma unroll
+        for (int c = 0; c < CMAX; ++c)
+          if (c < C) acc[c] += lv[c];
+      }
+#pragma unroll
+      for (int c = 0; c < CMAX; ++c)
+        if (c < C) out[(long)s * C + c] = acc[c];
+    }
+    if (stage) __syncthreads();
+  }
+}
+
+extern "C" __global__ __launch_bounds__(THREADS) void lpx_argmax_rows(
+    const float* __restrict__ x, int* __restrict__ out, long rows, int cols) {
+  for (long r = (long)blockIdx.x * THREADS + threadIdx.x; r < rows;
+       r += (long)gridDim.x * THREADS) {
+    const float* p = x + r * cols;
+    int best = 0;
+    float bv = p[0];
+    for (int c = 1; c < cols; ++c) {
+      const float v = p[c];
+      if (v > bv) { bv = v; best = c; }
+    }
+    out[r] = best;
+  }
+}
+
+// xorshift128 + Box-Muller synthetic standard-normal features
+extern "C" __global__ __launch_bounds__(THREADS) void lpx_xgen(
+    float* __restrict__ out, long n, unsigned seed) {
+  const long tid = (long)blockIdx.x * THREADS + threadIdx.x;
+  unsigned s0 = 0x9E3779B9u ^ (unsigned)tid ^ seed;
+  unsigned s1 = 0x85EBCA6Bu + (unsigned)(tid * 2654435761u);
+  for (long i = tid * 2; i < n; i += (long)gridDim.x * THREADS * 2) {
+    // two xorshift steps -> two uniforms
+    unsigned x = s0;
+    const unsigned y = s1;
+    s0 = y;
+    x ^= x << 23;
+    s1 = x ^ y ^ (x >> 17) ^ (y >> 26);
+    const unsigned u1 = s1 + y;
+    x = s0; s0 = s1; unsigned xx = x ^ (x << 23);
+    s1 = xx ^ s0 ^ (xx >> 17) ^ (s0 >> 26);
+    const unsigned u2 = s1 + s0;
+    const float f1 = ((u1 >> 8) + 1.0f) * (1.0f / 16777217.0f);  // (0,1]
+    const float f2 = (u2 >> 8) * (1.0f / 16777216.0f);           // [0,1)
+    const float r = sqrtf(-2.0f * logf(f1));
+    float sn, cs;
+    __sincosf(6.2831853f * f2, &sn, &cs);
+    out[i] = r * cs;
+    if (i + 1 < n) out[i + 1] = r * sn;
+  }
+}

@@ -30,6 +30,7 @@ HANDLERS = {
     "matmul": "matmul_handler",
     "fftspmv": "fftspmv_handler",
     "rf": "rf_handler",
+    "rf_lite": "rf_lite_handler",
     "resnet50": "resnet50_handler",
 }
 

@@ -1,0 +1,236 @@
+"""lpx_rt — torch-free gfx950 runtime (ctypes over libamdhip64 + librccl).
+
+The cold-start-critical path of SURVEY.md §2.2 N11/N15: a handler imports
+this shim (~ms), hipModuleLoads the AOT standalone_gfx950.hsaco (no JIT) and
+launches kernels directly; multi-GPU fan-out uses RCCL's single-process
+ncclCommInitAll + broadcast/allGather over xGMI. No PyTorch import anywhere
+on this path — the ~0.8 s libtorch import leaves the metric.
+"""
+
+from __future__ import annotations
+
+import ctypes
+import os
+from typing import List, Optional, Sequence
+
+_hip = None
+_rccl = None
+
+
+def hip():
+    global _hip
+    if _hip is None:
+        _hip = ctypes.CDLL("libamdhip64.so", mode=ctypes.RTLD_GLOBAL)
+        _hip.hipGetErrorString.restype = ctypes.c_char_p
+    return _hip
+
+
+def rccl():
+    global _rccl
+    if _rccl is None:
+        _rccl = ctypes.CDLL("librccl.so", mode=ctypes.RTLD_GLOBAL)
+    return _rccl
+
+
+class HipError(RuntimeError):
+    pass
+
+
+def _ck(code: int, what: str = "hip call") -> None:
+    if code != 0:
+        msg = hip().hipGetErrorString(code)
+        raise HipError(f"{what} failed: {msg.decode() if msg else code}")
+
+
+# ---- device/memory -------------------------------------------------------
+
+def device_count() -> int:
+    n = ctypes.c_int(0)
+    _ck(hip().hipGetDeviceCount(ctypes.byref(n)), "hipGetDeviceCount")
+    return n.value
+
+
+def set_device(d: int) -> None:
+    _ck(hip().hipSetDevice(d), "hipSetDevice")
+
+
+def malloc(nbytes: int) -> int:
+    p = ctypes.c_void_p(0)
+    _ck(hip().hipMalloc(ctypes.byref(p), ctypes.c_size_t(nbytes)), "hipMalloc")
+    return p.value
+
+
+def _vp(p):
+    return p if isinstance(p, ctypes.c_void_p) else ctypes.c_void_p(p)
+
+
+def memcpy_h2d(dst: int, src, nbytes: int) -> None:
+    _ck(hip().hipMemcpy(_vp(dst), _vp(src), ctypes.c_size_t(nbytes), 1),
+        "hipMemcpy H2D")
+
+
+def memcpy_d2h(dst, src: int, nbytes: int) -> None:
+    _ck(hip().hipMemcpy(_vp(dst), _vp(src), ctypes.c_size_t(nbytes), 2),
+        "hipMemcpy D2H")
+
+
+def sync() -> None:
+    _ck(hip().hipDeviceSynchronize(), "hipDeviceSynchronize")
+
+
+# ---- module / kernels ----------------------------------------------------
+
+class Module:
+    """hipModule wrapper over the AOT .hsaco (found next to this file or in
+    lpx_ops/, falling back to the in-repo ops dir)."""
+
+    SEARCH = ("standalone_gfx950.hsaco",
+              os.path.join("lpx_ops", "standalone_gfx950.hsaco"))
+
+    def __init__(self, path: Optional[str] = None):
+        if path is None:
+            here = os.path.dirname(os.path.abspath(__file__))
+            candidates = [os.path.join(here, s) for s in self.SEARCH]
+            candidates += [os.path.join(os.path.dirname(here), "ops",
+                                        "standalone_gfx950.hsaco")]
+            for sp in __import__("sys").path:
+                candidates.append(os.path.join(sp, "lpx_ops",
+                                               "standalone_gfx950.hsaco"))
+            path = next((c for c in candidates if os.path.exists(c)), None)
+            if path is None:
+                raise HipError("standalone_gfx950.hsaco not found (AOT code "
+                               "object missing from the layer)")
+        self.path = path
+        self._mod = ctypes.c_void_p(0)
+        _ck(hip().hipModuleLoad(ctypes.byref(self._mod), path.encode()),
+            "hipModuleLoad")
+        self._fns = {}
+
+    def fn(self, name: str):
+        f = self._fns.get(name)
+        if f is None:
+            f = ctypes.c_void_p(0)
+            _ck(hip().hipModuleGetFunction(ctypes.byref(f), self._mod,
+                                           name.encode()),
+                f"hipModuleGetFunction({name})")
+            self._fns[name] = f
+        return f
+
+    def launch(self, name: str, grid: int, block: int, shared: int,
+               arg_struct: ctypes.Structure) -> None:
+        buf = ctypes.byref(arg_struct)
+        size = ctypes.c_size_t(ctypes.sizeof(arg_struct))
+        # HIP_LAUNCH_PARAM_BUFFER_POINTER=1, _BUFFER_SIZE=2, _END=3
+        extra = (ctypes.c_void_p * 5)(
+            ctypes.c_void_p(1), ctypes.cast(buf, ctypes.c_void_p),
+            ctypes.c_void_p(2), ctypes.cast(ctypes.byref(size), ctypes.c_void_p),
+            ctypes.c_void_p(3))
+        _ck(hip().hipModuleLaunchKernel(self.fn(name), grid, 1, 1, block, 1, 1,
+                                        shared, None, None, extra),
+            f"hipModuleLaunchKernel({name})")
+
+
+# ---- kernel arg structs (match standalone.hip signatures) ----------------
+
+class RfHeapArgs(ctypes.Structure):
+    _fields_ = [("nodes", ctypes.c_void_p), ("leaf_values", ctypes.c_void_p),
+                ("n_trees", ctypes.c_int), ("depth", ctypes.c_int),
+                ("X", ctypes.c_void_p), ("out", ctypes.c_void_p),
+                ("S", ctypes.c_int), ("F", ctypes.c_int), ("C", ctypes.c_int)]
+
+
+class ArgmaxArgs(ctypes.Structure):
+    _fields_ = [("x", ctypes.c_void_p), ("out", ctypes.c_void_p),
+                ("rows", ctypes.c_long), ("cols", ctypes.c_int)]
+
+
+class XgenArgs(ctypes.Structure):
+    _fields_ = [("out", ctypes.c_void_p), ("n", ctypes.c_long),
+                ("seed", ctypes.c_uint)]
+
+
+def rf_heap(mod: Module, nodes: int, leaves: int, n_trees: int, depth: int,
+            X: int, out: int, S: int, F: int, C: int) -> None:
+    blocks = min((S + 255) // 256, 4096) or 1
+    shared = 256 * (F + 1) * 4 if F <= 63 else 0
+    mod.launch("lpx_rf_heap", blocks, 256, shared,
+               RfHeapArgs(nodes, leaves, n_trees, depth, X, out, S, F, C))
+
+
+def argmax_rows(mod: Module, x: int, out: int, rows: int, cols: int) -> None:
+    blocks = min((rows + 255) // 256, 4096) or 1
+    mod.launch("lpx_argmax_rows", blocks, 256, 0, ArgmaxArgs(x, out, rows, cols))
+
+
+def xgen(mod: Module, out: int, n: int, seed: int) -> None:
+    blocks = min((n // 2 + 255) // 256, 4096) or 1
+    mod.launch("lpx_xgen", blocks, 256, 0, XgenArgs(out, n, seed))
+
+
+# ---- torch-free synthetic forest (heap layout) ---------------------------
+
+def random_heap_forest_np(n_trees: int, depth: int, n_features: int,
+                          n_classes: int, seed: int = 0):
+    """Complete forest in level-order (heap) layout, numpy only.
+
+    Returns (nodes int32 [n_trees*(2^d-1), 2] {feat, thresh-bits},
+    leaf_values f32 [n_trees*2^d, C] row-normalized). Identical arrays to
+    lpx_sklearn.ForestSoA.random(...) for the same arguments (that path
+    delegates here)."""
+    import numpy as np
+
+    rng = np.random.default_rng(seed)
+    n_internal = (1 << depth) - 1
+    n_leaves = 1 << depth
+    total = n_trees * n_internal
+    feat = rng.integers(0, n_features, total).astype(np.uint32)
+    thr = rng.standard_normal(total).astype(np.float32)
+    nodes = np.stack([feat, thr.view(np.uint32)], axis=1).view(np.int32)
+    lv = rng.random((n_trees * n_leaves, n_classes)).astype(np.float32)
+    lv /= lv.sum(axis=1, keepdims=True)
+    return np.ascontiguousarray(nodes), np.ascontiguousarray(lv)
+
+
+# ---- RCCL (single-process multi-GPU) -------------------------------------
+
+NCCL_FLOAT32 = 7
+NCCL_INT32 = 2
+
+
+class Comms:
+    """ncclCommInitAll over g devices in ONE process (xGMI fan-out without a
+    process-per-GPU rendezvous — the cold-start-cheap topology)."""
+
+    def __init__(self, ndev: int):
+        self.ndev = ndev
+        self.comms = (ctypes.c_void_p * ndev)()
+        devs = (ctypes.c_int * ndev)(*range(ndev))
+        r = rccl().ncclCommInitAll(self.comms, ndev, devs)
+        if r != 0:
+            raise HipError(f"ncclCommInitAll failed: {r}")
+
+    def broadcast(self, bufs: Sequence[int], count: int, dtype: int,
+                  root: int = 0) -> None:
+        rc = rccl()
+        rc.ncclGroupStart()
+        for d in range(self.ndev):
+            set_device(d)
+            # each rank passes device-local pointers; only root's send matters
+            rc.ncclBroadcast(ctypes.c_void_p(bufs[d]), ctypes.c_void_p(bufs[d]),
+                             ctypes.c_size_t(count), dtype, root,
+                             self.comms[d], None)
+        rc.ncclGroupEnd()
+
+    def all_gather(self, send: Sequence[int], recv: Sequence[int],
+                   count: int, dtype: int) -> None:
+        rc = rccl()
+        rc.ncclGroupStart()
+        for d in range(self.ndev):
+            set_device(d)
+            rc.ncclAllGather(ctypes.c_void_p(send[d]), ctypes.c_void_p(recv[d]),
+                             ctypes.c_size_t(count), dtype, self.comms[d], None)
+        rc.ncclGroupEnd()
+
+    def destroy(self) -> None:
+        for d in range(self.ndev):
+            rccl().ncclCommDestroy(self.comms[d])

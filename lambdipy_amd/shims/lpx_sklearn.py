@@ -201,17 +201,15 @@ class ForestSoA:
                      device):
         """Complete forest in level-order (heap) layout: internal nodes only
         ({u32 feat, f32 thresh-bits}); children implicit at 2i+1/2i+2, leaf
-        row = heap index - (2^depth - 1). Fully vectorized generation."""
-        rng = np.random.default_rng(seed)
-        n_internal = (1 << depth) - 1
-        n_leaves = 1 << depth
-        total = n_trees * n_internal
-        feat = rng.integers(0, n_features, total).astype(np.uint32)
-        thr = rng.standard_normal(total).astype(np.float32)
-        nodes = np.stack([feat, thr.view(np.uint32)], axis=1).view(np.int32)
-        roots = np.arange(n_trees, dtype=np.int32) * n_internal
-        lv = rng.random((n_trees * n_leaves, n_classes)).astype(np.float32)
-        lv /= lv.sum(axis=1, keepdims=True)
+        row = heap index - (2^depth - 1). Generation shared with the
+        torch-free runtime (lpx_rt.random_heap_forest_np)."""
+        try:
+            from lpx_rt import random_heap_forest_np
+        except ImportError:
+            from .lpx_rt import random_heap_forest_np
+        nodes, lv = random_heap_forest_np(n_trees, depth, n_features,
+                                          n_classes, seed)
+        roots = np.arange(n_trees, dtype=np.int32) * ((1 << depth) - 1)
         f = cls(torch.from_numpy(nodes).contiguous(),
                 torch.from_numpy(lv).contiguous(),
                 torch.from_numpy(roots).contiguous(), n_trees, n_classes,

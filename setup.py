@@ -22,6 +22,7 @@ sources = [
     os.path.join(CSRC, f)
     for f in sorted(os.listdir(CSRC))
     if f.endswith((".cpp", ".hip")) and not f.endswith(("_hip.cpp", "_hip.hip"))
+    and f != "standalone.hip"  # --genco code object, built by the builder
 ]
 
 setup(

@@ -348,3 +348,42 @@ def test_conv_fp8_bias_relu_stride1(ops):
     aq = e.im2col_fp8(x, R, R, 1, 1, 1.0)
     ref = e.gemm_fp8(aq, wq, 1.0, bias, True)
     assert torch.allclose(got, ref, atol=1e-4)
+
+
+# ---- torch-free runtime (lpx_rt + standalone .hsaco) ---------------------
+
+def test_lpx_rt_rf_matches_extension(ops):
+    """hipModule path (no-torch runtime) must agree with the torch extension."""
+    from lambdipy_amd.shims import lpx_rt
+
+    n_trees, depth, F, C = 32, 8, 16, 4
+    nodes_np, leaves_np = lpx_rt.random_heap_forest_np(n_trees, depth, F, C, 5)
+    nodes = torch.from_numpy(nodes_np).to(DEV)
+    leaves = torch.from_numpy(leaves_np).to(DEV)
+    X = _rand((20000, F), seed=12)
+    ref = ops.ext().rf_predict_heap(nodes, leaves, depth, X, 1)
+
+    mod = lpx_rt.Module()
+    out = torch.empty(20000, C, device=DEV)
+    lpx_rt.rf_heap(mod, nodes.data_ptr(), leaves.data_ptr(), n_trees, depth,
+                   X.data_ptr(), out.data_ptr(), 20000, F, C)
+    lpx_rt.sync()
+    assert torch.equal(out, ref)
+
+    labels = torch.empty(20000, dtype=torch.int32, device=DEV)
+    lpx_rt.argmax_rows(mod, out.data_ptr(), labels.data_ptr(), 20000, C)
+    lpx_rt.sync()
+    assert torch.equal(labels, ref.argmax(dim=1).to(torch.int32))
+
+
+def test_lpx_rt_xgen_distribution(ops):
+    from lambdipy_amd.shims import lpx_rt
+
+    mod = lpx_rt.Module()
+    n = 1 << 20
+    x = torch.empty(n, device=DEV)
+    lpx_rt.xgen(mod, x.data_ptr(), n, 7)
+    lpx_rt.sync()
+    assert torch.isfinite(x).all()
+    assert abs(x.mean().item()) < 0.01
+    assert abs(x.std().item() - 1.0) < 0.02
