@@ -18,7 +18,7 @@ def main():
     a = torch.randn(4096, 4096, device=dev, dtype=torch.bfloat16)
     b = torch.randn(4096, 4096, device=dev, dtype=torch.bfloat16)
     for _ in range(10):
-        ops.gemm_bf16(a, b, variant=2)
+        ops.gemm_bf16(a, b)
     torch.cuda.synchronize()
 
     # RF traversal (config 4, the metric workload; heap-layout kernel)
