@@ -41,8 +41,10 @@ __global__ __launch_bounds__(THREADS) void fft_lds_kernel(
   const f32x2* twc = reinterpret_cast<const f32x2*>(tw);
 
   for (int b = blockIdx.x; b < batch; b += gridDim.x) {
-    const f32x2* src = reinterpret_cast<const f32x2*>(xg) + (long)b * n;
-    for (int i = t; i < n; i += THREADS) buf0[i] = src[i];
+    // 16-B vectorized load (two complex64 per lane — G13)
+    const f32x4* src4 = reinterpret_cast<const f32x4*>(xg) + (long)b * (n >> 1);
+    f32x4* b04 = reinterpret_cast<f32x4*>(buf0);
+    for (int i = t; i < (n >> 1); i += THREADS) b04[i] = src4[i];
     __syncthreads();
 
     f32x2* x = buf0;
@@ -96,13 +98,16 @@ __global__ __launch_bounds__(THREADS) void fft_lds_kernel(
     }
 
     f32x2* dst_lds = x;  // result lives in x after the final swap
-    f32x2* dst = reinterpret_cast<f32x2*>(outg) + (long)b * n;
+    f32x4* dst4 = reinterpret_cast<f32x4*>(outg) + (long)b * (n >> 1);
+    const f32x4* src_lds4 = reinterpret_cast<const f32x4*>(dst_lds);
     if (INV) {
       const float inv_n = 1.0f / (float)n;
-      for (int i = t; i < n; i += THREADS)
-        dst[i] = f32x2{dst_lds[i].x * inv_n, dst_lds[i].y * inv_n};
+      for (int i = t; i < (n >> 1); i += THREADS) {
+        f32x4 v = src_lds4[i];
+        dst4[i] = f32x4{v.x * inv_n, v.y * inv_n, v.z * inv_n, v.w * inv_n};
+      }
     } else {
-      for (int i = t; i < n; i += THREADS) dst[i] = dst_lds[i];
+      for (int i = t; i < (n >> 1); i += THREADS) dst4[i] = src_lds4[i];
     }
     __syncthreads();
   }
