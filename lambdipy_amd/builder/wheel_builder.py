@@ -20,7 +20,6 @@ from __future__ import annotations
 
 import glob
 import os
-import shutil
 import subprocess
 import sys
 from typing import Optional

@@ -10,7 +10,6 @@ from __future__ import annotations
 import json
 import os
 import shutil
-import sys
 
 import click
 

@@ -10,7 +10,7 @@ from __future__ import annotations
 import os
 import re
 import shutil
-from typing import Iterable, List, Tuple
+from typing import Iterable, Tuple
 
 
 def _compile(pattern: str) -> "re.Pattern[str]":

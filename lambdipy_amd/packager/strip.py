@@ -10,7 +10,7 @@ from __future__ import annotations
 import os
 import shutil
 import subprocess
-from typing import List, Tuple
+from typing import Tuple
 
 _STRIP_CANDIDATES = [
     "/opt/rocm/lib/llvm/bin/llvm-strip",

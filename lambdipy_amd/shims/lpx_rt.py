@@ -11,7 +11,7 @@ from __future__ import annotations
 
 import ctypes
 import os
-from typing import List, Optional, Sequence
+from typing import Optional, Sequence
 
 _hip = None
 _rccl = None
