@@ -16,17 +16,19 @@ def main() -> dict:
     def mark(name, start):
         phases.append({"phase": name, "dur_ms": (time.perf_counter() - start) * 1e3})
 
-    t = time.perf_counter()
-    import torch
-    mark("import_torch", t)
-
-    # layer may still be extracting (overlapped cold start) — wait for it
+    # Wait for the layer BEFORE importing torch: torch itself imports numpy,
+    # and a half-extracted layer numpy on PYTHONPATH would poison the import
+    # (the torch-free rf_lite handler keeps the full unzip/import overlap).
     ready = os.environ.get("LPX_READY_FILE")
     if ready:
         t = time.perf_counter()
         while not os.path.exists(ready):
             time.sleep(0.002)
         mark("wait_layer_extract", t)
+
+    t = time.perf_counter()
+    import torch
+    mark("import_torch", t)
 
     t = time.perf_counter()
     try:
