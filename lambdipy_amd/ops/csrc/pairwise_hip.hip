@@ -16,33 +16,29 @@ namespace lpx {
 
 namespace {
 
-constexpr int BM = 128, BN = 128, BK = 16;
+constexpr int BM = 64, BN = 64, BK = 16;
 constexpr int THREADS = 256;
 constexpr int ROWF = BK + 1;  // padded row, floats
 
 __device__ __forceinline__ void stage_f32(const float* __restrict__ src,
                                           int rows_total, int K, int row0,
                                           int k0, float* __restrict__ lds) {
-  // 128 rows x 16 f32 per tile = 8 KiB; 256 threads x 2 float4 chunks.
-  const int t = threadIdx.x;
+  // 64 rows x 16 f32 per tile = 4 KiB; 256 threads x 1 float4.
+  const int t = threadIdx.x;          // chunk = t: row = t/4, block = t%4
+  const int r = t >> 2, kb = (t & 3) * 4;
+  const int grow = row0 + r, gk = k0 + kb;
+  f32x4 v = {0.f, 0.f, 0.f, 0.f};
+  if (grow < rows_total) {
+    if (gk + 4 <= K) {
+      v = *reinterpret_cast<const f32x4*>(src + (long)grow * K + gk);
+    } else {
 #pragma unroll
-  for (int i = 0; i < 2; ++i) {
-    const int c = t + i * THREADS;
-    const int r = c >> 2, kb = (c & 3) * 4;
-    const int grow = row0 + r, gk = k0 + kb;
-    f32x4 v = {0.f, 0.f, 0.f, 0.f};
-    if (grow < rows_total) {
-      if (gk + 4 <= K) {
-        v = *reinterpret_cast<const f32x4*>(src + (long)grow * K + gk);
-      } else {
-#pragma unroll
-        for (int j = 0; j < 4; ++j)
-          if (gk + j < K) v[j] = src[(long)grow * K + gk + j];
-      }
+      for (int j = 0; j < 4; ++j)
+        if (gk + j < K) v[j] = src[(long)grow * K + gk + j];
     }
-#pragma unroll
-    for (int j = 0; j < 4; ++j) lds[r * ROWF + kb + j] = v[j];
   }
+#pragma unroll
+  for (int j = 0; j < 4; ++j) lds[r * ROWF + kb + j] = v[j];
 }
 
 __global__ __launch_bounds__(THREADS) void pairwise_kernel(
@@ -62,13 +58,13 @@ __global__ __launch_bounds__(THREADS) void pairwise_kernel(
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
-  const int wr = wid >> 1, wc = wid & 1;  // wave covers 64x64
+  const int wr = wid >> 1, wc = wid & 1;  // wave covers 32x32
 
-  f32x4 acc[4][4];
+  f32x4 acc[2][2];
 #pragma unroll
-  for (int i = 0; i < 4; ++i)
+  for (int i = 0; i < 2; ++i)
 #pragma unroll
-    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+    for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
   const int fr = lane & 15, fk = lane >> 4;  // fragment row, k-lane (0..3)
 
@@ -87,30 +83,25 @@ __global__ __launch_bounds__(THREADS) void pairwise_kernel(
     }
 #pragma unroll
     for (int kk = 0; kk < 4; ++kk) {
-      float aV[4], bV[4];
-#pragma unroll
-      for (int mf = 0; mf < 4; ++mf)
-        aV[mf] = xCur[(wr * 64 + mf * 16 + fr) * ROWF + kk * 4 + fk];
-#pragma unroll
-      for (int nf = 0; nf < 4; ++nf)
-        bV[nf] = yCur[(wc * 64 + nf * 16 + fr) * ROWF + kk * 4 + fk];
-#pragma unroll
-      for (int mf = 0; mf < 4; ++mf)
-#pragma unroll
-        for (int nf = 0; nf < 4; ++nf)
-          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x4f32(
-              aV[mf], bV[nf], acc[mf][nf], 0, 0, 0);
+      float a0 = xCur[(wr * 32 + fr) * ROWF + kk * 4 + fk];
+      float a1 = xCur[(wr * 32 + 16 + fr) * ROWF + kk * 4 + fk];
+      float b0 = yCur[(wc * 32 + fr) * ROWF + kk * 4 + fk];
+      float b1 = yCur[(wc * 32 + 16 + fr) * ROWF + kk * 4 + fk];
+      acc[0][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1, b1, acc[1][1], 0, 0, 0);
     }
     __syncthreads();
     cur ^= 1;
   }
 
-  const int orow0 = bm + wr * 64 + (lane >> 4) * 4;
-  const int ocol0 = bn + wc * 64 + (lane & 15);
+  const int orow0 = bm + wr * 32 + (lane >> 4) * 4;
+  const int ocol0 = bn + wc * 32 + (lane & 15);
 #pragma unroll
-  for (int mf = 0; mf < 4; ++mf) {
+  for (int mf = 0; mf < 2; ++mf) {
 #pragma unroll
-    for (int nf = 0; nf < 4; ++nf) {
+    for (int nf = 0; nf < 2; ++nf) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int row = orow0 + mf * 16 + r;
