@@ -109,3 +109,67 @@ extern "C" __global__ __launch_bounds__(THREADS) void lpx_xgen(
     if (i + 1 < n) out[i + 1] = r * sn;
   }
 }
+
+// device-side synthetic forest generation (heap layout) — removes the numpy
+// import + model H2D upload from the cold-start critical path.
+extern "C" __global__ __launch_bounds__(THREADS) void lpx_forestgen_nodes(
+    i32x2* __restrict__ nodes, long n_internal_total, int n_features,
+    unsigned seed) {
+  const long tid = (long)blockIdx.x * THREADS + threadIdx.x;
+  for (long i = tid; i < n_internal_total; i += (long)gridDim.x * THREADS) {
+    unsigned s0 = 0x9E3779B9u ^ (unsigned)i ^ seed;
+    unsigned s1 = 0x85EBCA6Bu + (unsigned)(i * 2654435761u);
+    // advance a few steps to decorrelate
+    for (int k = 0; k < 3; ++k) {
+      unsigned x = s0; s0 = s1; x ^= x << 23;
+      s1 = x ^ s0 ^ (x >> 17) ^ (s0 >> 26);
+    }
+    const unsigned u1 = s1 + s0;
+    unsigned x = s0; s0 = s1; unsigned xx = x ^ (x << 23);
+    s1 = xx ^ s0 ^ (xx >> 17) ^ (s0 >> 26);
+    const unsigned u2 = s1 + s0;
+    const float f1 = ((u1 >> 8) + 1.0f) * (1.0f / 16777217.0f);
+    const float f2 = (u2 >> 8) * (1.0f / 16777216.0f);
+    float sn, cs;
+    __sincosf(6.2831853f * f2, &sn, &cs);
+    const float thresh = sqrtf(-2.0f * logf(f1)) * cs;
+    i32x2 nd;
+    nd.x = (int)(u2 % (unsigned)n_features);
+    nd.y = __float_as_int(thresh);
+    nodes[i] = nd;
+  }
+}
+
+extern "C" __global__ __launch_bounds__(THREADS) void lpx_forestgen_leaves(
+    float* __restrict__ leaves, long n_rows, int C, unsigned seed) {
+  const long tid = (long)blockIdx.x * THREADS + threadIdx.x;
+  for (long r = tid; r < n_rows; r += (long)gridDim.x * THREADS) {
+    unsigned s0 = 0xDEADBEEFu ^ (unsigned)r ^ seed;
+    unsigned s1 = 0xC2B2AE35u + (unsigned)(r * 2246822519u);
+    float v[CMAX];
+    float sum = 0.f;
+    for (int c = 0; c < C && c < CMAX; ++c) {
+      unsigned x = s0; s0 = s1; x ^= x << 23;
+      s1 = x ^ s0 ^ (x >> 17) ^ (s0 >> 26);
+      const float u = ((s1 + s0) >> 8) * (1.0f / 16777216.0f) + 1e-3f;
+      v[c] = u;
+      sum += u;
+    }
+    for (int c = 0; c < C && c < CMAX; ++c)
+      leaves[r * C + c] = v[c] / sum;
+  }
+}
+
+extern "C" __global__ __launch_bounds__(THREADS) void lpx_checksum_i32(
+    const int* __restrict__ x, unsigned long long* __restrict__ out, long n) {
+  long long local = 0;
+  for (long i = (long)blockIdx.x * THREADS + threadIdx.x; i < n;
+       i += (long)gridDim.x * THREADS)
+    local += x[i];
+  // wave reduce then one atomic per wave
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    local += __shfl_down(local, off, 64);
+  if ((threadIdx.x & 63) == 0)
+    atomicAdd(out, (unsigned long long)local);
+}

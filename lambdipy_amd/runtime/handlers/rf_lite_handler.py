@@ -30,7 +30,6 @@ def main() -> dict:
         mark("wait_layer_extract", t)
 
     t = time.perf_counter()
-    import numpy as np
     try:
         import lpx_rt
     except ImportError:
@@ -55,33 +54,46 @@ def main() -> dict:
         mods.append(lpx_rt.Module())
     mark("hip_init_module_load", t)
 
-    t = time.perf_counter()
-    nodes_np, leaves_np = lpx_rt.random_heap_forest_np(
-        n_trees, depth, n_features, n_classes, seed=17)
-    mark("model_build", t)
-
+    n_internal = n_trees * ((1 << depth) - 1)
+    n_leaf_rows = n_trees * (1 << depth)
+    nodes_bytes = n_internal * 8
+    leaves_bytes = n_leaf_rows * n_classes * 4
     total = batch * g
     xel = total * n_features
 
     t = time.perf_counter()
+    # one slab per device (a single hipMalloc; sub-buffers by offset)
+    sizes = [nodes_bytes, leaves_bytes, xel * 4, batch * n_classes * 4,
+             batch * 4, total * 4, 8]
+    offs = []
+    cur = 0
+    for sz in sizes:
+        offs.append(cur)
+        cur += (sz + 255) & ~255
     dev = []
     for d in range(g):
         lpx_rt.set_device(d)
-        b = {
-            "nodes": lpx_rt.malloc(nodes_np.nbytes),
-            "leaves": lpx_rt.malloc(leaves_np.nbytes),
-            "X": lpx_rt.malloc(xel * 4),
-            "out": lpx_rt.malloc(batch * n_classes * 4),
-            "labels": lpx_rt.malloc(batch * 4),
-            "labels_all": lpx_rt.malloc(total * 4),
-        }
+        slab = lpx_rt.malloc(cur)
+        b = {k: slab + o for k, o in zip(
+            ("nodes", "leaves", "X", "out", "labels", "labels_all", "ck"),
+            offs)}
         dev.append(b)
-    # model bytes land on GPU 0 only; RCCL broadcasts to the rest
-    lpx_rt.set_device(0)
-    lpx_rt.memcpy_h2d(dev[0]["nodes"], nodes_np.ctypes.data, nodes_np.nbytes)
-    lpx_rt.memcpy_h2d(dev[0]["leaves"], leaves_np.ctypes.data,
-                      leaves_np.nbytes)
-    mark("alloc_upload", t)
+    mark("alloc", t)
+
+    t = time.perf_counter()
+    hostgen = os.environ.get("LPX_RF_HOSTGEN") == "1"
+    if hostgen:
+        nodes_np, leaves_np = lpx_rt.random_heap_forest_np(
+            n_trees, depth, n_features, n_classes, seed=17)
+        lpx_rt.set_device(0)
+        lpx_rt.memcpy_h2d(dev[0]["nodes"], nodes_np.ctypes.data, nodes_bytes)
+        lpx_rt.memcpy_h2d(dev[0]["leaves"], leaves_np.ctypes.data, leaves_bytes)
+    else:
+        # model generated ON GPU 0 (random-init weights; no host bytes)
+        lpx_rt.set_device(0)
+        lpx_rt.forestgen(mods[0], dev[0]["nodes"], n_internal, n_features,
+                         dev[0]["leaves"], n_leaf_rows, n_classes, 17)
+    mark("model_build", t)
 
     comms = None
     t = time.perf_counter()
@@ -91,9 +103,9 @@ def main() -> dict:
 
     t = time.perf_counter()
     if comms:
-        comms.broadcast([b["nodes"] for b in dev], nodes_np.nbytes // 4,
+        comms.broadcast([b["nodes"] for b in dev], nodes_bytes // 4,
                         lpx_rt.NCCL_INT32, root=0)
-        comms.broadcast([b["leaves"] for b in dev], leaves_np.size,
+        comms.broadcast([b["leaves"] for b in dev], leaves_bytes // 4,
                         lpx_rt.NCCL_FLOAT32, root=0)
     mark("model_broadcast", t)
 
@@ -118,14 +130,18 @@ def main() -> dict:
         comms.all_gather([b["labels"] for b in dev],
                          [b["labels_all"] for b in dev], batch,
                          lpx_rt.NCCL_INT32)
+    import ctypes
+    lpx_rt.set_device(0)
+    src = dev[0]["labels_all"] if comms else dev[0]["labels"]
+    zero = ctypes.c_ulonglong(0)
+    lpx_rt.memcpy_h2d(dev[0]["ck"], ctypes.addressof(zero), 8)
+    lpx_rt.checksum_i32(mods[0], src, dev[0]["ck"], total)
     for d in range(g):
         lpx_rt.set_device(d)
         lpx_rt.sync()
-    labels = np.empty(total, dtype=np.int32)
+    ck = ctypes.c_ulonglong(0)
     lpx_rt.set_device(0)
-    lpx_rt.memcpy_d2h(labels.ctypes.data,
-                      dev[0]["labels_all"] if comms else dev[0]["labels"],
-                      total * 4)
+    lpx_rt.memcpy_d2h(ctypes.addressof(ck), dev[0]["ck"], 8)
     mark("first_inference", t)
 
     out = {
@@ -133,9 +149,9 @@ def main() -> dict:
         "world_size": g,
         "model": {"trees": n_trees, "depth": depth, "features": n_features,
                   "classes": n_classes,
-                  "model_mb": round((nodes_np.nbytes + leaves_np.nbytes) / 2**20, 2)},
+                  "model_mb": round((nodes_bytes + leaves_bytes) / 2**20, 2)},
         "batch": total,
-        "checksum": float(labels.sum()),
+        "checksum": float(ck.value),
         "total_ms": (time.perf_counter() - t0) * 1e3,
         "phases": phases,
     }

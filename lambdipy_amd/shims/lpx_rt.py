@@ -167,6 +167,37 @@ def xgen(mod: Module, out: int, n: int, seed: int) -> None:
     mod.launch("lpx_xgen", blocks, 256, 0, XgenArgs(out, n, seed))
 
 
+class ForestGenNodesArgs(ctypes.Structure):
+    _fields_ = [("nodes", ctypes.c_void_p), ("n", ctypes.c_long),
+                ("n_features", ctypes.c_int), ("seed", ctypes.c_uint)]
+
+
+class ForestGenLeavesArgs(ctypes.Structure):
+    _fields_ = [("leaves", ctypes.c_void_p), ("n_rows", ctypes.c_long),
+                ("C", ctypes.c_int), ("seed", ctypes.c_uint)]
+
+
+class ChecksumArgs(ctypes.Structure):
+    _fields_ = [("x", ctypes.c_void_p), ("out", ctypes.c_void_p),
+                ("n", ctypes.c_long)]
+
+
+def forestgen(mod: Module, nodes: int, n_internal_total: int, n_features: int,
+              leaves: int, n_leaf_rows: int, C: int, seed: int) -> None:
+    """Device-side heap-forest generation (no host model bytes at all)."""
+    b1 = min((n_internal_total + 255) // 256, 4096) or 1
+    mod.launch("lpx_forestgen_nodes", b1, 256, 0,
+               ForestGenNodesArgs(nodes, n_internal_total, n_features, seed))
+    b2 = min((n_leaf_rows + 255) // 256, 4096) or 1
+    mod.launch("lpx_forestgen_leaves", b2, 256, 0,
+               ForestGenLeavesArgs(leaves, n_leaf_rows, C, seed ^ 0x5bd1e995))
+
+
+def checksum_i32(mod: Module, x: int, out: int, n: int) -> None:
+    blocks = min((n + 255) // 256, 2048) or 1
+    mod.launch("lpx_checksum_i32", blocks, 256, 0, ChecksumArgs(x, out, n))
+
+
 # ---- torch-free synthetic forest (heap layout) ---------------------------
 
 def random_heap_forest_np(n_trees: int, depth: int, n_features: int,

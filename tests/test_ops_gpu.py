@@ -387,3 +387,39 @@ def test_lpx_rt_xgen_distribution(ops):
     assert torch.isfinite(x).all()
     assert abs(x.mean().item()) < 0.01
     assert abs(x.std().item() - 1.0) < 0.02
+
+
+def test_lpx_rt_forestgen(ops):
+    from lambdipy_amd.shims import lpx_rt
+
+    mod = lpx_rt.Module()
+    n_trees, depth, F, C = 8, 6, 12, 4
+    n_int = n_trees * (2**depth - 1)
+    n_rows = n_trees * 2**depth
+    nodes = torch.empty(n_int, 2, dtype=torch.int32, device=DEV)
+    leaves = torch.empty(n_rows, C, device=DEV)
+    lpx_rt.forestgen(mod, nodes.data_ptr(), n_int, F, leaves.data_ptr(),
+                     n_rows, C, 17)
+    lpx_rt.sync()
+    feats = nodes[:, 0]
+    assert (feats >= 0).all() and (feats < F).all()
+    thr = nodes[:, 1].view(torch.float32)
+    assert torch.isfinite(thr).all()
+    assert abs(thr.mean().item()) < 0.1
+    s = leaves.sum(dim=1)
+    assert torch.allclose(s, torch.ones_like(s), atol=1e-5)
+    # the generated forest must traverse cleanly through the rf kernel
+    X = _rand((4096, F), seed=3)
+    out = torch.empty(4096, C, device=DEV)
+    lpx_rt.rf_heap(mod, nodes.data_ptr(), leaves.data_ptr(), n_trees, depth,
+                   X.data_ptr(), out.data_ptr(), 4096, F, C)
+    lpx_rt.sync()
+    ps = out.sum(dim=1) / n_trees
+    assert torch.allclose(ps, torch.ones_like(ps), atol=1e-4)
+
+    # checksum kernel vs torch sum
+    labels = torch.randint(0, C, (100000,), dtype=torch.int32, device=DEV)
+    ck = torch.zeros(1, dtype=torch.int64, device=DEV)
+    lpx_rt.checksum_i32(mod, labels.data_ptr(), ck.data_ptr(), labels.numel())
+    lpx_rt.sync()
+    assert ck.item() == labels.sum().item()
