@@ -174,7 +174,14 @@ def cold_start(
     if wrap_cmd:  # e.g. rocprofv3 --kernel-trace --stats -d DIR --
         argv = list(wrap_cmd) + argv
     procs = []
-    if gpus <= 1:
+    if handler == "rf_lite" and gpus > 1:
+        # single-process multi-GPU (ncclCommInitAll) — no per-rank spawn
+        e = dict(base_env, LPX_RF_GPUS=str(gpus))
+        procs.append(
+            subprocess.Popen(argv, env=e, stdout=subprocess.PIPE,
+                             stderr=subprocess.PIPE, text=True)
+        )
+    elif gpus <= 1:
         procs.append(
             subprocess.Popen(argv, env=base_env,
                              stdout=subprocess.PIPE, stderr=subprocess.PIPE,
