@@ -85,7 +85,7 @@ __device__ __forceinline__ const bf16frag* frag_ptr(const char* __restrict__ sme
       smem + (size_t)parity * PARITY_BYTES + (size_t)hh * HALF_BYTES + q * 16);
 }
 
-template <bool BF16_OUT, int VM, bool LG0>
+template <bool BF16_OUT, int VM, bool LG0, int PRIO = 0>
 __global__ __launch_bounds__(THREADS8, 2) void gemm_bf16_8ph_kernel(
     const uint16_t* __restrict__ A, const uint16_t* __restrict__ Bt,
     void* __restrict__ Cout, int M, int N, int K, unsigned tiles_m,
@@ -126,6 +126,13 @@ __global__ __launch_bounds__(THREADS8, 2) void gemm_bf16_8ph_kernel(
   asm volatile("s_waitcnt vmcnt(4)" ::: "memory");  // tile0 A-lo/B-lo landed
   __builtin_amdgcn_s_barrier();
 
+  if constexpr (PRIO == 1) {
+    // T5 static form: the second-dispatched half loses VALU arbitration on
+    // every segment; one static priority raise, no per-phase flips.
+    if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+      __builtin_amdgcn_s_setprio(1);
+  }
+
   for (int t = 0; t < ksteps; ++t) {
     const int parity = t & 1;
 #pragma unroll
@@ -153,7 +160,7 @@ __global__ __launch_bounds__(THREADS8, 2) void gemm_bf16_8ph_kernel(
 
       __builtin_amdgcn_s_barrier();
       if constexpr (LG0) asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-      __builtin_amdgcn_s_setprio(1);
+      if constexpr (PRIO == 0) __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int mf = 0; mf < 4; ++mf)
 #pragma unroll
@@ -162,7 +169,7 @@ __global__ __launch_bounds__(THREADS8, 2) void gemm_bf16_8ph_kernel(
           for (int ks = 0; ks < 2; ++ks)
             acc[p][mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                 aF[mf][ks], bF[nf][ks], acc[p][mf][nf], 0, 0, 0);
-      __builtin_amdgcn_s_setprio(0);
+      if constexpr (PRIO == 0) __builtin_amdgcn_s_setprio(0);
       if constexpr (VM == 8)
         asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
       else
@@ -200,13 +207,13 @@ __global__ __launch_bounds__(THREADS8, 2) void gemm_bf16_8ph_kernel(
 
 }  // namespace
 
-template <bool BF16_OUT, int VM = 8, bool LG0 = false>
+template <bool BF16_OUT, int VM = 8, bool LG0 = false, int PRIO = 0>
 static hipError_t launch8ph(const uint16_t* A, const uint16_t* Bt, void* C,
                             int M, int N, int K, hipStream_t stream) {
   const unsigned tiles_m = (unsigned)(M / BM8), tiles_n = (unsigned)(N / BM8);
   const size_t lds = 2 * PARITY_BYTES;  // 128 KiB
-  gemm_bf16_8ph_kernel<BF16_OUT, VM, LG0><<<dim3(tiles_m * tiles_n),
-                                            dim3(THREADS8), lds, stream>>>(
+  gemm_bf16_8ph_kernel<BF16_OUT, VM, LG0, PRIO><<<dim3(tiles_m * tiles_n),
+                                                  dim3(THREADS8), lds, stream>>>(
       A, Bt, C, M, N, K, tiles_m, tiles_n);
   return hipGetLastError();
 }
@@ -228,6 +235,8 @@ hipError_t gemm_bf16_nt8ph_tune(const uint16_t* A, const uint16_t* Bt,
     case 1: return launch8ph<false, 8, true>(A, Bt, C, M, N, K, stream);
     case 2: return launch8ph<false, 6, false>(A, Bt, C, M, N, K, stream);
     case 3: return launch8ph<false, 8, false>(A, Bt, C, M, N, K, stream);
+    case 4: return launch8ph<false, 8, false, 1>(A, Bt, C, M, N, K, stream);
+    case 5: return launch8ph<false, 8, false, 2>(A, Bt, C, M, N, K, stream);
     default: return launch8ph<false, 6, true>(A, Bt, C, M, N, K, stream);
   }
 }
