@@ -423,3 +423,65 @@ def test_lpx_rt_forestgen(ops):
     lpx_rt.checksum_i32(mod, labels.data_ptr(), ck.data_ptr(), labels.numel())
     lpx_rt.sync()
     assert ck.item() == labels.sum().item()
+
+
+# ---- shim-level parity (the layer-facing API surfaces) -------------------
+
+def test_lpx_numpy_shim_matmul(ops):
+    from lambdipy_amd.shims import lpx_numpy
+
+    rng = np.random.default_rng(0)
+    a = rng.standard_normal((256, 300)).astype(np.float32)
+    b = rng.standard_normal((300, 192)).astype(np.float32)
+    got = lpx_numpy.matmul(a, b)
+    ref = a @ b
+    rel = np.abs(got - ref).max() / (np.abs(ref).max() + 1e-6)
+    assert rel < 2e-2, rel  # bf16 operand rounding
+
+
+def test_lpx_numpy_shim_matvec(ops):
+    from lambdipy_amd.shims import lpx_numpy
+
+    rng = np.random.default_rng(1)
+    a = rng.standard_normal((500, 700)).astype(np.float32)
+    x = rng.standard_normal(700).astype(np.float32)
+    got = lpx_numpy.matvec(a, x)
+    ref = a @ x
+    assert np.abs(got - ref).max() / (np.abs(ref).max() + 1e-6) < 2e-2
+
+
+def test_lpx_scipy_shim_fft(ops):
+    from lambdipy_amd.shims import lpx_scipy
+
+    rng = np.random.default_rng(2)
+    x = (rng.standard_normal((8, 512)) + 1j * rng.standard_normal((8, 512)))
+    got = lpx_scipy.fft(x)
+    ref = np.fft.fft(x.astype(np.complex64))
+    assert np.abs(got - ref).max() / np.abs(ref).max() < 1e-4
+
+
+def test_lpx_scipy_shim_csr(ops):
+    scipy = pytest.importorskip("scipy")
+    import scipy.sparse as sp
+    from lambdipy_amd.shims import lpx_scipy
+
+    rng = np.random.default_rng(3)
+    A = sp.random(800, 600, density=0.02, random_state=4, format="csr",
+                  dtype=np.float32)
+    x = rng.standard_normal(600).astype(np.float32)
+    got = lpx_scipy.csr_matvec(A.indptr, A.indices, A.data, x)
+    ref = A @ x
+    np.testing.assert_allclose(got, ref, rtol=1e-4, atol=1e-4)
+
+
+def test_lpx_sklearn_shim_pairwise(ops):
+    sklearn = pytest.importorskip("sklearn")
+    from sklearn.metrics.pairwise import euclidean_distances
+    from lambdipy_amd.shims import lpx_sklearn
+
+    rng = np.random.default_rng(4)
+    X = rng.standard_normal((300, 40)).astype(np.float32)
+    Y = rng.standard_normal((200, 40)).astype(np.float32)
+    got = lpx_sklearn.pairwise_sqeuclidean(X, Y)
+    ref = euclidean_distances(X, Y, squared=True)
+    assert np.abs(got - ref).max() < 1e-2
