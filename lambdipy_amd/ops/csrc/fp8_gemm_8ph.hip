@@ -112,6 +112,11 @@ __global__ __launch_bounds__(THREADS8, 2) void gemm_fp8_8ph_kernel(
   asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
   __builtin_amdgcn_s_barrier();
 
+  // T5 static form: one priority raise for the younger dispatch half, no
+  // per-phase flips (within-probe +23-24% on the bf16 8-phase kernel).
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);
+
   for (int t = 0; t < ksteps; ++t) {
     const int parity = t & 1;
 #pragma unroll
@@ -135,7 +140,6 @@ __global__ __launch_bounds__(THREADS8, 2) void gemm_fp8_8ph_kernel(
       stage_half8f(A, Bt, K, bm, bn, st, sh, ksteps, smem, w, lane);
 
       __builtin_amdgcn_s_barrier();
-      __builtin_amdgcn_s_setprio(1);
 #pragma unroll
       for (int mf = 0; mf < 4; ++mf)
 #pragma unroll
@@ -145,7 +149,6 @@ __global__ __launch_bounds__(THREADS8, 2) void gemm_fp8_8ph_kernel(
             acc[p][mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
                 __builtin_bit_cast(long, aF[mf][ks]),
                 __builtin_bit_cast(long, bF[nf][ks]), acc[p][mf][nf], 0, 0, 0);
-      __builtin_amdgcn_s_setprio(0);
       asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
       __builtin_amdgcn_s_barrier();
     }

@@ -86,7 +86,7 @@ __device__ __forceinline__ const bf16frag* frag_ptr(const char* __restrict__ sme
       smem + (size_t)parity * PARITY_BYTES + (size_t)hh * HALF_BYTES + q * 16);
 }
 
-template <bool BF16_OUT, int VM, bool LG0, int PRIO = 0>
+template <bool BF16_OUT, int VM, bool LG0, int PRIO>
 __global__ __launch_bounds__(THREADS8, 2) void gemm_bf16_8ph_kernel(
     const uint16_t* __restrict__ A, const uint16_t* __restrict__ Bt,
     void* __restrict__ Cout, int M, int N, int K, unsigned tiles_m,
@@ -208,7 +208,7 @@ __global__ __launch_bounds__(THREADS8, 2) void gemm_bf16_8ph_kernel(
 
 }  // namespace
 
-template <bool BF16_OUT, int VM = 8, bool LG0 = false, int PRIO = 0>
+template <bool BF16_OUT, int VM = 8, bool LG0 = false, int PRIO = 1>
 static hipError_t launch8ph(const uint16_t* A, const uint16_t* Bt, void* C,
                             int M, int N, int K, hipStream_t stream) {
   const unsigned tiles_m = (unsigned)(M / BM8), tiles_n = (unsigned)(N / BM8);
