@@ -61,6 +61,46 @@ extern "C" __global__ void mx_sweep_kernel(const uint8_t* __restrict__ A,
   }
 }
 
+// byte-1 variant of the sweep (perturbs scale byte1 instead of byte0)
+extern "C" __global__ void mx_sweep_kernel_b1(const uint8_t* __restrict__ A,
+                                              const uint8_t* __restrict__ B,
+                                              float* __restrict__ Dsweep) {
+  const int l = threadIdx.x;
+  const int rc = l & 31;
+  const int kb = l >> 5;
+  i32x8 a = *reinterpret_cast<const i32x8*>(A + rc * 64 + kb * 32);
+  i32x8 b = *reinterpret_cast<const i32x8*>(B + rc * 64 + kb * 32);
+  for (int L = 0; L < 64; ++L) {
+    f32x16 c;
+#pragma unroll
+    for (int i = 0; i < 16; ++i) c[i] = 0.f;
+    const int sa = 127 | (((l == L) ? 128 : 127) << 8);
+    c = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(a, b, c, 0, 0, 0, sa,
+                                                        0, 127);
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int row = (r & 3) + 8 * (r >> 2) + 4 * (l >> 5);
+      const int col = l & 31;
+      Dsweep[L * 1024 + row * 32 + col] = c[r];
+    }
+  }
+}
+
+extern "C" int run_mx_sweep_b1(const uint8_t* hA, const uint8_t* hB, float* hD) {
+  uint8_t *dA, *dB;
+  float* dD;
+  (void)hipMalloc(&dA, 32 * 64);
+  (void)hipMalloc(&dB, 32 * 64);
+  (void)hipMalloc(&dD, 64 * 1024 * 4);
+  (void)hipMemcpy(dA, hA, 32 * 64, hipMemcpyHostToDevice);
+  (void)hipMemcpy(dB, hB, 32 * 64, hipMemcpyHostToDevice);
+  hipLaunchKernelGGL(mx_sweep_kernel_b1, dim3(1), dim3(64), 0, 0, dA, dB, dD);
+  hipError_t e = hipDeviceSynchronize();
+  (void)hipMemcpy(hD, dD, 64 * 1024 * 4, hipMemcpyDeviceToHost);
+  (void)hipFree(dA); (void)hipFree(dB); (void)hipFree(dD);
+  return (int)e;
+}
+
 extern "C" int run_mx_sweep(const uint8_t* hA, const uint8_t* hB, float* hD) {
   uint8_t *dA, *dB;
   float* dD;
