@@ -16,9 +16,12 @@ from .api import (
     extension_error,
     fft,
     gemm_bf16,
+    gemm_mx,
     gemv,
     ifft,
     matmul_bf16,
+    matmul_mx,
+    quantize_mx,
     pairwise_sqeuclidean,
     require_extension,
     rf_predict_proba_sum,
@@ -32,6 +35,9 @@ __all__ = [
     "require_extension",
     "gemm_bf16",
     "matmul_bf16",
+    "quantize_mx",
+    "gemm_mx",
+    "matmul_mx",
     "gemv",
     "spmv_csr",
     "fft",

@@ -67,6 +67,29 @@ def gemv(A: torch.Tensor, x: torch.Tensor) -> torch.Tensor:
     return require_extension().gemv(A.contiguous(), x.contiguous())
 
 
+def quantize_mx(x: torch.Tensor):
+    """f32 [R, K%64==0] -> (e4m3 bytes in MFMA-shuffled order, E8M0 scales
+    [R, K/64, 2]).  One shared 2^e scale per 32-element MX block; block b of
+    group g covers k quarters {b, b+2} (the hardware lane coverage of
+    v_mfma_scale_f32_32x32x64_f8f6f4, decoded in scripts/mx_sweep.py)."""
+    q, sc = require_extension().quantize_mx(x.contiguous())
+    return q, sc
+
+
+def gemm_mx(A: torch.Tensor, sA: torch.Tensor, Bt: torch.Tensor,
+            sB: torch.Tensor) -> torch.Tensor:
+    """MX block-scaled fp8 GEMM: C = dequant(A) @ dequant(Bt)^T, f32 out.
+    Operands/scales come from quantize_mx; scales are applied in hardware."""
+    return require_extension().gemm_mx(A, sA, Bt, sB)
+
+
+def matmul_mx(A: torch.Tensor, B: torch.Tensor) -> torch.Tensor:
+    """Convenience: quantize both f32 operands to MX fp8 and multiply."""
+    qa, sa = quantize_mx(A)
+    qb, sb = quantize_mx(B.t().contiguous())
+    return gemm_mx(qa, sa, qb, sb)
+
+
 # ---- sparse --------------------------------------------------------------
 
 def spmv_csr(indptr: torch.Tensor, indices: torch.Tensor, data: torch.Tensor,

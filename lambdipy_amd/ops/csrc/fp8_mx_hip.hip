@@ -1,0 +1,271 @@
+#include "hip/hip_runtime.h"
+// MX block-scaled fp8 GEMM for gfx950 — v_mfma_scale_f32_32x32x64_f8f6f4.
+//
+// The ONLY path to the ~4.6 PF fp8 ceiling (guide §3: non-scaled fp8 runs at
+// the bf16 rate; MX K=64/128 doubles it). Empirically derived semantics
+// (scripts/mx_sweep.py, one-lane scale perturbation with quarter-marked
+// data): for the 32x32x64 shape, lane l sources A row (l&31); its 32 operand
+// bytes cover k-quarters {kb, kb+2} (kb = l>>5, quarters = 16 elements), and
+// its SINGLE scale byte (always byte0 of the scale VGPR; byte1 unused)
+// applies to exactly those 32 elements. So:
+//   - operands are packed SHUFFLED per 64-k group: [q0 | q2 | q1 | q3]
+//     (lane kb=0 reads bytes [0,32) = original k [0,16)+[32,48); kb=1 reads
+//     [32,64) = k [16,32)+[48,64)) — then each lane's bytes are contiguous;
+//   - MX blocks are (row, 64-k-group, kb): 32 elements with one shared E8M0
+//     scale, quantized by quantize_mx_fp8 below (amax -> 2^e, e4m3 payload);
+//   - the MFMA applies scales in hardware; NO epilogue descale is needed.
+// C/D layout (32x32): col = l&31, row = (r&3) + 8*(r>>2) + 4*(l>>5) — probed.
+
+#include "lpx_common.h"
+
+namespace lpx {
+
+namespace {
+
+constexpr int THREADS = 256;
+
+typedef __attribute__((ext_vector_type(8))) int i32x8v;
+
+// ---- MX quantization -----------------------------------------------------
+// x f32 [R][K] row-major, K % 64 == 0. Outputs:
+//   q  u8 [R][K]      shuffled e4m3 payload ([q0|q2|q1|q3] per 64-group)
+//   sc u8 [R][K/64][2] E8M0 block scales (kb = 0: quarters {0,2}; 1: {1,3})
+__global__ __launch_bounds__(THREADS) void quantize_mx_kernel(
+    const float* __restrict__ x, unsigned char* __restrict__ q,
+    unsigned char* __restrict__ sc, int R, int K) {
+  const int groups = K >> 6;
+  const long total = (long)R * groups * 2;  // one thread per (row, group, kb)
+  for (long idx = (long)blockIdx.x * THREADS + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * THREADS) {
+    const int kb = (int)(idx & 1);
+    const int g = (int)((idx >> 1) % groups);
+    const int r = (int)(idx / (2 * groups));
+    const float* src = x + (long)r * K + g * 64;
+    // the block's 32 elements: quarters kb and kb+2
+    float amax = 0.f;
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      const int q0 = kb + 2 * h;  // quarter index (kb, kb+2)
+      for (int j = 0; j < 16; ++j)
+        amax = fmaxf(amax, fabsf(src[q0 * 16 + j]));
+    }
+    // E8M0 shared scale: 2^e with amax/2^e <= 448 (e4m3 max)
+    int e = 0;
+    if (amax > 0.f) {
+      int ex;
+      (void)frexpf(amax, &ex);      // amax = m * 2^ex, m in [0.5, 1)
+      e = ex - 9;                   // amax / 2^e in [256, 512) -> <= 448 after sat
+    }
+    if (e < -127) e = -127;
+    if (e > 127) e = 127;
+    sc[((long)r * groups + g) * 2 + kb] = (unsigned char)(e + 127);
+    const float inv = exp2f((float)-e);
+    unsigned char* dst = q + (long)r * K + g * 64 + kb * 32;
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      const int q0 = kb + 2 * h;
+      for (int j = 0; j < 16; ++j) {
+        float v = src[q0 * 16 + j] * inv;
+        // e4m3 with saturation (reuse bit algorithm from fp8_gemm.hip)
+        union { float f; uint32_t u; } c;
+        c.f = v;
+        const uint32_t sign = (c.u >> 24) & 0x80;
+        float ax = fabsf(v);
+        unsigned char out;
+        if (!(ax == ax)) {
+          out = 0x7f;
+        } else {
+          if (ax > 448.f) ax = 448.f;
+          c.f = ax;
+          int32_t ee = (int32_t)((c.u >> 23) & 0xff) - 127;
+          uint32_t m = c.u & 0x7fffff;
+          uint32_t o;
+          if (ax == 0.f) {
+            o = 0;
+          } else if (ee < -6) {
+            float scaled = ax * 512.f;
+            uint32_t qq = (uint32_t)scaled;
+            float frac = scaled - (float)qq;
+            if (frac > 0.5f || (frac == 0.5f && (qq & 1))) ++qq;
+            o = (qq >= 8) ? (1u << 3) : qq;
+          } else {
+            uint32_t mant = m >> 20;
+            uint32_t rest = m & 0xfffff;
+            if (rest > 0x80000 || (rest == 0x80000 && (mant & 1))) {
+              ++mant;
+              if (mant == 8) { mant = 0; ++ee; }
+            }
+            if (ee > 8) { ee = 8; mant = 7; }
+            o = ((uint32_t)(ee + 7) << 3) | mant;
+          }
+          out = (unsigned char)o;
+        }
+        dst[h * 16 + j] = (unsigned char)(sign | out);
+      }
+    }
+  }
+}
+
+// ---- MX GEMM (v2 register-staged structure, 256x256 tile, BK=64 elems) ----
+
+constexpr int BM = 256, BN = 256, BK = 64;  // 64 fp8 bytes per row per tile
+constexpr int THREADS2 = 512;
+constexpr int ROWB = BK + 16;               // 80-B padded LDS rows
+constexpr int CHUNKS = BM * BK / 16 / THREADS2;  // 16-B chunks/thread = 2
+
+__device__ __forceinline__ void stage_mx(const unsigned char* __restrict__ src,
+                                         int rows_total, int K, int row0,
+                                         int k0, i32x4 (&regs)[CHUNKS]) {
+  const int t = threadIdx.x;
+#pragma unroll
+  for (int i = 0; i < CHUNKS; ++i) {
+    const int c = t + i * THREADS2;
+    const int r = c >> 2;
+    const int cell = (c & 3) * 16;
+    const int grow = row0 + r;
+    const int gk = k0 + cell;
+    i32x4 v = {0, 0, 0, 0};
+    if (grow < rows_total && gk + 16 <= K)
+      v = *reinterpret_cast<const i32x4*>(src + (long)grow * K + gk);
+    regs[i] = v;
+  }
+}
+
+__device__ __forceinline__ void write_mx_lds(const i32x4 (&regs)[CHUNKS],
+                                             char* __restrict__ lds) {
+  const int t = threadIdx.x;
+#pragma unroll
+  for (int i = 0; i < CHUNKS; ++i) {
+    const int c = t + i * THREADS2;
+    const int r = c >> 2;
+    const int cell = (c & 3) * 16;
+    *reinterpret_cast<i32x4*>(lds + r * ROWB + cell) = regs[i];
+  }
+}
+
+__global__ __launch_bounds__(THREADS2, 2) void gemm_mx_kernel(
+    const unsigned char* __restrict__ A, const unsigned char* __restrict__ sA,
+    const unsigned char* __restrict__ Bt, const unsigned char* __restrict__ sB,
+    float* __restrict__ C, int M, int N, int K, unsigned tiles_m,
+    unsigned tiles_n) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* const lA = smem;                      // [256][80]
+  char* const lB = smem + BM * ROWB;
+
+  const unsigned nwg = tiles_m * tiles_n;
+  const unsigned bid = lpx_xcd_swizzle(blockIdx.x, nwg);
+  const int bm = (int)(bid / tiles_n) * BM;
+  const int bn = (int)(bid % tiles_n) * BN;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wr = wid >> 2, wc = wid & 3;  // 2x4 wave grid; 128x64 per wave
+
+  const int fr = lane & 31;   // fragment row/col (32-wide)
+  const int fkb = lane >> 5;  // 0/1: which interleaved 32-byte half
+
+  // acc: 4 m-frags x 2 n-frags of 32x32 -> 8 x f32x16 = 128 VGPR
+  f32x16 acc[4][2];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int e = 0; e < 16; ++e) acc[i][j][e] = 0.f;
+
+  const int groups = K >> 6;
+
+  i32x4 aRegs[CHUNKS], bRegs[CHUNKS];
+  stage_mx(A, M, K, bm, 0, aRegs);
+  stage_mx(Bt, N, K, bn, 0, bRegs);
+  write_mx_lds(aRegs, lA);
+  write_mx_lds(bRegs, lB);
+  __syncthreads();
+
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);  // T5 static form
+
+  const int ksteps = groups;
+  for (int ks = 0; ks < ksteps; ++ks) {
+    if (ks + 1 < ksteps) {
+      stage_mx(A, M, K, bm, (ks + 1) * BK, aRegs);
+      stage_mx(Bt, N, K, bn, (ks + 1) * BK, bRegs);
+    }
+    // fragments: 32 B per lane = 2 x b128 at [row][fkb*32 .. +32)
+    i32x8v aF[4], bF[2];
+    int sa[4], sb[2];
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf) {
+      const int r = wr * 128 + mf * 32 + fr;
+      const char* p = lA + r * ROWB + fkb * 32;
+      i32x4 lo = *reinterpret_cast<const i32x4*>(p);
+      i32x4 hi = *reinterpret_cast<const i32x4*>(p + 16);
+      aF[mf] = i32x8v{lo.x, lo.y, lo.z, lo.w, hi.x, hi.y, hi.z, hi.w};
+      sa[mf] = sA[((long)(bm + r) * groups + ks) * 2 + fkb];
+    }
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf) {
+      const int r = wc * 64 + nf * 32 + fr;
+      const char* p = lB + r * ROWB + fkb * 32;
+      i32x4 lo = *reinterpret_cast<const i32x4*>(p);
+      i32x4 hi = *reinterpret_cast<const i32x4*>(p + 16);
+      bF[nf] = i32x8v{lo.x, lo.y, lo.z, lo.w, hi.x, hi.y, hi.z, hi.w};
+      sb[nf] = sB[((long)(bn + r) * groups + ks) * 2 + fkb];
+    }
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+      for (int nf = 0; nf < 2; ++nf)
+        acc[mf][nf] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+            aF[mf], bF[nf], acc[mf][nf], 0, 0, 0, sa[mf], 0, sb[nf]);
+    __syncthreads();
+    if (ks + 1 < ksteps) {
+      write_mx_lds(aRegs, lA);
+      write_mx_lds(bRegs, lB);
+    }
+    __syncthreads();
+  }
+
+  // epilogue: 32x32 D map per fragment
+  const int ocol_l = lane & 31;
+  const int orow_base = 4 * (lane >> 5);
+#pragma unroll
+  for (int mf = 0; mf < 4; ++mf) {
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int row = bm + wr * 128 + mf * 32 + (r & 3) + 8 * (r >> 2) +
+                        orow_base;
+        const int col = bn + wc * 64 + nf * 32 + ocol_l;
+        if (row < M && col < N) C[(long)row * N + col] = acc[mf][nf][r];
+      }
+    }
+  }
+}
+
+}  // namespace
+
+hipError_t quantize_mx_fp8(const float* x, unsigned char* q, unsigned char* sc,
+                           int R, int K, hipStream_t stream) {
+  if (K % 64 != 0) return hipErrorInvalidValue;
+  const long total = (long)R * (K >> 6) * 2;
+  unsigned blocks = (unsigned)((total + THREADS - 1) / THREADS);
+  if (blocks > 4096u) blocks = 4096u;
+ hipLaunchKernelGGL(( quantize_mx_kernel), dim3(dim3(blocks ? blocks : 1)), dim3(dim3(THREADS)), 0, stream, 
+      x, q, sc, R, K);
+  return hipGetLastError();
+}
+
+hipError_t gemm_mx_fp8_nt(const unsigned char* A, const unsigned char* sA,
+                          const unsigned char* Bt, const unsigned char* sB,
+                          float* C, int M, int N, int K, hipStream_t stream) {
+  if (K % 64 != 0) return hipErrorInvalidValue;
+  const unsigned tiles_m = lpx_ceil_div(M, BM), tiles_n = lpx_ceil_div(N, BN);
+  const size_t lds = (size_t)2 * BM * ROWB;  // 40 KiB
+ hipLaunchKernelGGL(( gemm_mx_kernel), dim3(dim3(tiles_m * tiles_n)), dim3(dim3(THREADS2)), lds, stream, 
+      A, sA, Bt, sB, C, M, N, K, tiles_m, tiles_n);
+  return hipGetLastError();
+}
+
+}  // namespace lpx

@@ -117,4 +117,14 @@ hipError_t gemm_fp8_nt8ph(const unsigned char* A, const unsigned char* Bt,
                           float* C, const float* bias, int M, int N, int K,
                           float descale, bool relu, hipStream_t stream);
 
+// ---- MX block-scaled fp8 (v_mfma_scale_f32_32x32x64_f8f6f4) --------------
+// x f32 [R,K] (K%64==0) -> q e4m3 bytes [R,K] in MFMA-shuffled order
+// ([q0|q2|q1|q3] per 64-group) + sc E8M0 [R, K/64, 2] block scales.
+hipError_t quantize_mx_fp8(const float* x, unsigned char* q, unsigned char* sc,
+                           int R, int K, hipStream_t stream);
+// C f32 [M,N] = dequant(A) @ dequant(Bt)^T — scales applied in hardware.
+hipError_t gemm_mx_fp8_nt(const unsigned char* A, const unsigned char* sA,
+                          const unsigned char* Bt, const unsigned char* sB,
+                          float* C, int M, int N, int K, hipStream_t stream);
+
 }  // namespace lpx

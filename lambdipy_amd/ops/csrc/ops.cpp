@@ -276,6 +276,45 @@ at::Tensor gemm_fp8(const at::Tensor& A, const at::Tensor& Bt, double descale,
   return C;
 }
 
+std::vector<at::Tensor> quantize_mx(const at::Tensor& x) {
+  CHECK_DEV(x);
+  TORCH_CHECK(x.scalar_type() == at::kFloat && x.dim() == 2 &&
+              x.size(1) % 64 == 0, "quantize_mx expects f32 [R, K%64==0]");
+  const int64_t R = x.size(0), K = x.size(1);
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
+  auto q = at::empty({R, K}, x.options().dtype(at::kByte));
+  auto sc = at::empty({R, K / 64, 2}, x.options().dtype(at::kByte));
+  check_hip(lpx::quantize_mx_fp8(x.data_ptr<float>(), q.data_ptr<uint8_t>(),
+                                 sc.data_ptr<uint8_t>(), (int)R, (int)K,
+                                 current_stream(x)),
+            "quantize_mx_fp8");
+  return {q, sc};
+}
+
+at::Tensor gemm_mx(const at::Tensor& A, const at::Tensor& sA,
+                   const at::Tensor& Bt, const at::Tensor& sB) {
+  CHECK_DEV(A);
+  CHECK_DEV(Bt);
+  CHECK_DEV(sA);
+  CHECK_DEV(sB);
+  TORCH_CHECK(A.scalar_type() == at::kByte && Bt.scalar_type() == at::kByte &&
+              sA.scalar_type() == at::kByte && sB.scalar_type() == at::kByte,
+              "gemm_mx expects byte tensors");
+  TORCH_CHECK(A.dim() == 2 && Bt.dim() == 2 && A.size(1) == Bt.size(1) &&
+              A.size(1) % 64 == 0, "gemm_mx expects A [M,K], Bt [N,K], K%64==0");
+  const int64_t M = A.size(0), K = A.size(1), N = Bt.size(0);
+  TORCH_CHECK(sA.numel() == M * (K / 64) * 2 && sB.numel() == N * (K / 64) * 2,
+              "scale shapes must be [rows, K/64, 2]");
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(A.device());
+  auto C = at::empty({M, N}, A.options().dtype(at::kFloat));
+  check_hip(lpx::gemm_mx_fp8_nt(A.data_ptr<uint8_t>(), sA.data_ptr<uint8_t>(),
+                                Bt.data_ptr<uint8_t>(), sB.data_ptr<uint8_t>(),
+                                C.data_ptr<float>(), (int)M, (int)N, (int)K,
+                                current_stream(A)),
+            "gemm_mx_fp8_nt");
+  return C;
+}
+
 at::Tensor rf_predict8(const at::Tensor& nodes8, const at::Tensor& leaf_values,
                        const at::Tensor& roots, const at::Tensor& X,
                        int64_t ilp) {
@@ -449,6 +488,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_fp8", &gemm_fp8, "fp8 e4m3 MFMA GEMM with fused descale/bias/relu",
         py::arg("A"), py::arg("Bt"), py::arg("descale") = 1.0,
         py::arg("bias") = py::none(), py::arg("relu") = false);
+  m.def("quantize_mx", &quantize_mx,
+        "f32 [R,K] -> (e4m3 bytes shuffled, E8M0 [R,K/64,2] block scales)",
+        py::arg("x"));
+  m.def("gemm_mx", &gemm_mx,
+        "MX block-scaled fp8 GEMM (scales applied in hardware)", py::arg("A"),
+        py::arg("sA"), py::arg("Bt"), py::arg("sB"));
   m.def("gemm_bf16_tune8", &gemm_bf16_tune8, "8-phase GEMM tuning variants",
         py::arg("A"), py::arg("Bt"), py::arg("tune") = 0);
   m.def("build_info", &build_info);

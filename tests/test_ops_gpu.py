@@ -244,6 +244,71 @@ def test_gemm_fp8_vs_dequant_ref(ops):
     assert err < 1e-3, err
 
 
+def _dequant_mx(q, sc):
+    """Reconstruct f32 from MX-quantized (shuffled) bytes + E8M0 scales.
+
+    Shuffled 64-group order is [q0|q2|q1|q3]; block kb covers shuffled
+    bytes [kb*32, +32) (= original quarters {kb, kb+2})."""
+    R, K = q.shape
+    g = K // 64
+    d = q.view(R, g, 4, 16).view(torch.float8_e4m3fn).float()
+    scales = torch.pow(2.0, sc.float() - 127.0)          # [R, g, 2]
+    d = d * scales.repeat_interleave(2, dim=2).unsqueeze(-1)
+    return d[:, :, [0, 2, 1, 3], :].reshape(R, K)        # unshuffle
+
+
+def test_quantize_mx_roundtrip(ops):
+    e = ops.ext()
+    g = torch.Generator(device="cpu").manual_seed(7)
+    # wide dynamic range across blocks: per-row magnitude ramp
+    x = (torch.randn(64, 256, generator=g) *
+         torch.logspace(-6, 6, 64, base=2.0, generator=None).unsqueeze(1)).to(DEV)
+    q, sc = e.quantize_mx(x)
+    xd = _dequant_mx(q, sc)
+    # per-element rel error bounded by e4m3 step within each 32-elem block
+    err = (xd - x).abs()
+    amax = x.abs().amax().clamp(min=1e-30)
+    # block-relative: compare against per-block amax
+    blk = x.view(64, 4, 4, 16)[:, :, [0, 2, 1, 3], :].reshape(64, 4, 2, 32)
+    blk_amax = blk.abs().amax(dim=-1, keepdim=True)
+    err_blk = (err.view(64, 4, 4, 16)[:, :, [0, 2, 1, 3], :]
+               .reshape(64, 4, 2, 32) / blk_amax.clamp(min=1e-30))
+    assert err_blk.max().item() < 0.07, err_blk.max().item()
+    del amax
+
+
+def test_gemm_mx_vs_dequant_ref(ops):
+    e = ops.ext()
+    g = torch.Generator(device="cpu").manual_seed(8)
+    M, N, K = 512, 512, 256
+    A = (torch.randn(M, K, generator=g) * 3).to(DEV)
+    Bt = torch.randn(N, K, generator=g).to(DEV)
+    qa, sa = e.quantize_mx(A)
+    qb, sb = e.quantize_mx(Bt)
+    C = e.gemm_mx(qa, sa, qb, sb)
+    ref = _dequant_mx(qa, sa) @ _dequant_mx(qb, sb).t()
+    err = (C - ref).abs().max().item() / (ref.abs().max().item() + 1e-6)
+    assert err < 1e-3, err
+    # end-to-end quantization error vs exact f32 matmul stays small
+    exact = A @ Bt.t()
+    rel = (C - exact).norm().item() / exact.norm().item()
+    assert rel < 0.02, rel
+
+
+def test_gemm_mx_tail_shapes(ops):
+    e = ops.ext()
+    g = torch.Generator(device="cpu").manual_seed(9)
+    M, N, K = 300, 200, 192  # non-multiples of the 256x256 tile
+    A = torch.randn(M, K, generator=g).to(DEV)
+    Bt = torch.randn(N, K, generator=g).to(DEV)
+    qa, sa = e.quantize_mx(A)
+    qb, sb = e.quantize_mx(Bt)
+    C = e.gemm_mx(qa, sa, qb, sb)
+    ref = _dequant_mx(qa, sa) @ _dequant_mx(qb, sb).t()
+    err = (C - ref).abs().max().item() / (ref.abs().max().item() + 1e-6)
+    assert err < 1e-3, err
+
+
 def test_gemm_fp8_bias_relu(ops):
     e = ops.ext()
     g = torch.Generator(device="cpu").manual_seed(2)
