@@ -203,6 +203,36 @@ def fetch(name, output, release_dir):
     click.echo(f"fetched {dest}")
 
 
+@main.command("strip-archs")
+@click.argument("paths", nargs=-1, type=click.Path(exists=True))
+@click.option("--arch", "archs", multiple=True, default=("gfx950",),
+              show_default=True, help="gfx arch(s) to keep")
+@click.option("-o", "--output", default=None,
+              help="output path (single input only; default: in place)")
+@click.option("--json", "as_json", is_flag=True)
+def strip_archs(paths, archs, output, as_json):
+    """Strip non-target GPU code from HIP shared libraries.
+
+    ROCm DSOs carry clang-offload fat binaries for ~14 gfx architectures;
+    for MI355X-only layers everything but gfx950 is dead weight (e.g.
+    libtorch_hip.so: 251 MB of fatbin -> ~26 MB). Rewrites the ELF in
+    place: rebundles each fat-binary entry, repoints the registration
+    wrappers/relocs, and compacts the file."""
+    from .packager.fatbin import strip_fatbin
+
+    if output and len(paths) != 1:
+        raise click.ClickException("-o requires exactly one input")
+    rows = []
+    for p in paths:
+        st = strip_fatbin(p, output if output else None, archs=tuple(archs))
+        rows.append(st)
+        if not as_json:
+            click.echo(f"{p}: {st['old_size']/1e6:.1f} -> {st['new_size']/1e6:.1f} MB "
+                       f"({st['stripped']}/{st['bundles']} bundles stripped)")
+    if as_json:
+        click.echo(json.dumps(rows, indent=1))
+
+
 @main.command()
 def recipes():
     """List the recipe registry."""

@@ -3,15 +3,17 @@
 // The ONLY path to the ~4.6 PF fp8 ceiling (guide §3: non-scaled fp8 runs at
 // the bf16 rate; MX K=64/128 doubles it). Empirically derived semantics
 // (scripts/mx_sweep.py, one-lane scale perturbation with quarter-marked
-// data): for the 32x32x64 shape, lane l sources A row (l&31); its 32 operand
-// bytes cover k-quarters {kb, kb+2} (kb = l>>5, quarters = 16 elements), and
-// its SINGLE scale byte (always byte0 of the scale VGPR; byte1 unused)
-// applies to exactly those 32 elements. So:
-//   - operands are packed SHUFFLED per 64-k group: [q0 | q2 | q1 | q3]
-//     (lane kb=0 reads bytes [0,32) = original k [0,16)+[32,48); kb=1 reads
-//     [32,64) = k [16,32)+[48,64)) — then each lane's bytes are contiguous;
-//   - MX blocks are (row, 64-k-group, kb): 32 elements with one shared E8M0
-//     scale, quantized by quantize_mx_fp8 below (amax -> 2^e, e4m3 payload);
+// data): for the 32x32x64 shape, lane l sources row/col (l&31); lane pair
+// {l, l+32} concatenates to that row's 64 k-bytes. Lane l's SINGLE scale
+// byte (always byte0 of the scale VGPR; byte1 unused) applies to BYTE-HALF
+// (l>>5) of BOTH lanes of its row — with plain contiguous packing the scale
+// blocks would be the interleaved quarters {kb, kb+2}. We pack each 64-k
+// group SHUFFLED as [q0 | q2 | q1 | q3] (quarter = 16 elements) instead,
+// which makes the hardware scale blocks equal the standard contiguous OCP
+// MX-32 blocks {2kb, 2kb+1}:
+//   - quantize_mx_fp8 quantizes per contiguous 32-element block (amax ->
+//     E8M0 2^e, e4m3 payload), writing quarters to their shuffled slots;
+//   - scale sc[r][g][kb] covers original k [g*64 + kb*32, +32);
 //   - the MFMA applies scales in hardware; NO epilogue descale is needed.
 // C/D layout (32x32): col = l&31, row = (r&3) + 8*(r>>2) + 4*(l>>5) — probed.
 
@@ -39,32 +41,32 @@ __global__ __launch_bounds__(THREADS) void quantize_mx_kernel(
     const int kb = (int)(idx & 1);
     const int g = (int)((idx >> 1) % groups);
     const int r = (int)(idx / (2 * groups));
-    const float* src = x + (long)r * K + g * 64;
-    // the block's 32 elements: quarters kb and kb+2
+    const float* src = x + (long)r * K + g * 64 + kb * 32;
+    // standard contiguous MX-32 block: original k [g*64 + kb*32, +32).
+    // The MFMA applies lane-kb's scale to byte-half kb of BOTH k-lanes;
+    // with the [q0|q2|q1|q3] data shuffle that is exactly this block.
     float amax = 0.f;
-#pragma unroll
-    for (int h = 0; h < 2; ++h) {
-      const int q0 = kb + 2 * h;  // quarter index (kb, kb+2)
-      for (int j = 0; j < 16; ++j)
-        amax = fmaxf(amax, fabsf(src[q0 * 16 + j]));
-    }
-    // E8M0 shared scale: 2^e with amax/2^e <= 448 (e4m3 max)
+    for (int j = 0; j < 32; ++j) amax = fmaxf(amax, fabsf(src[j]));
+    // E8M0 shared scale: 2^e, non-clipping (amax/2^e <= 448, the e4m3 max)
     int e = 0;
     if (amax > 0.f) {
       int ex;
       (void)frexpf(amax, &ex);      // amax = m * 2^ex, m in [0.5, 1)
-      e = ex - 9;                   // amax / 2^e in [256, 512) -> <= 448 after sat
+      e = ex - 9;                   // amax / 2^e in [256, 512)
+      if (amax * exp2f((float)-e) > 448.f) ++e;  // avoid saturating block max
     }
     if (e < -127) e = -127;
     if (e > 127) e = 127;
     sc[((long)r * groups + g) * 2 + kb] = (unsigned char)(e + 127);
     const float inv = exp2f((float)-e);
-    unsigned char* dst = q + (long)r * K + g * 64 + kb * 32;
+    unsigned char* base = q + (long)r * K + g * 64;
 #pragma unroll
     for (int h = 0; h < 2; ++h) {
-      const int q0 = kb + 2 * h;
+      const int qq = 2 * kb + h;                  // original quarter index
+      const int pos = ((qq & 1) << 1) | (qq >> 1);  // shuffled slot
+      unsigned char* dst = base + pos * 16;
       for (int j = 0; j < 16; ++j) {
-        float v = src[q0 * 16 + j] * inv;
+        float v = src[h * 16 + j] * inv;
         // e4m3 with saturation (reuse bit algorithm from fp8_gemm.hip)
         union { float f; uint32_t u; } c;
         c.f = v;
@@ -99,7 +101,7 @@ __global__ __launch_bounds__(THREADS) void quantize_mx_kernel(
           }
           out = (unsigned char)o;
         }
-        dst[h * 16 + j] = (unsigned char)(sign | out);
+        dst[j] = (unsigned char)(sign | out);
       }
     }
   }
@@ -142,6 +144,27 @@ __device__ __forceinline__ void write_mx_lds(const i32x4 (&regs)[CHUNKS],
   }
 }
 
+// stage the tile's 2x256 scale bytes (per side) into LDS registers
+__device__ __forceinline__ unsigned stage_sc(const unsigned char* __restrict__ sc,
+                                             int rows_total, int groups,
+                                             int row0, int ks) {
+  // threads 0..127 cover one side's 256 rows x 2 scales as u16 pairs
+  const int t = threadIdx.x;
+  if (t >= 128) return 0;
+  const int r = row0 + t * 2;
+  unsigned v = 0x7f7f;  // scale 1.0 pair for out-of-range rows
+  if (r + 1 < rows_total) {
+    const unsigned short a =
+        *reinterpret_cast<const unsigned short*>(sc + ((long)r * groups + ks) * 2);
+    const unsigned short b = *reinterpret_cast<const unsigned short*>(
+        sc + ((long)(r + 1) * groups + ks) * 2);
+    v = (unsigned)a | ((unsigned)b << 16);
+  } else if (r < rows_total) {
+    v = *reinterpret_cast<const unsigned short*>(sc + ((long)r * groups + ks) * 2);
+  }
+  return v;
+}
+
 __global__ __launch_bounds__(THREADS2, 2) void gemm_mx_kernel(
     const unsigned char* __restrict__ A, const unsigned char* __restrict__ sA,
     const unsigned char* __restrict__ Bt, const unsigned char* __restrict__ sB,
@@ -150,6 +173,9 @@ __global__ __launch_bounds__(THREADS2, 2) void gemm_mx_kernel(
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* const lA = smem;                      // [256][80]
   char* const lB = smem + BM * ROWB;
+  unsigned char* const lsA =
+      reinterpret_cast<unsigned char*>(smem + 2 * BM * ROWB);  // [256][2]
+  unsigned char* const lsB = lsA + BM * 2;
 
   const unsigned nwg = tiles_m * tiles_n;
   const unsigned bid = lpx_xcd_swizzle(blockIdx.x, nwg);
@@ -175,10 +201,17 @@ __global__ __launch_bounds__(THREADS2, 2) void gemm_mx_kernel(
   const int groups = K >> 6;
 
   i32x4 aRegs[CHUNKS], bRegs[CHUNKS];
+  unsigned scA, scB;
   stage_mx(A, M, K, bm, 0, aRegs);
   stage_mx(Bt, N, K, bn, 0, bRegs);
+  scA = stage_sc(sA, M, groups, bm, 0);
+  scB = stage_sc(sB, N, groups, bn, 0);
   write_mx_lds(aRegs, lA);
   write_mx_lds(bRegs, lB);
+  if (threadIdx.x < 128) {
+    reinterpret_cast<unsigned*>(lsA)[threadIdx.x] = scA;
+    reinterpret_cast<unsigned*>(lsB)[threadIdx.x] = scB;
+  }
   __syncthreads();
 
   if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
@@ -189,6 +222,8 @@ __global__ __launch_bounds__(THREADS2, 2) void gemm_mx_kernel(
     if (ks + 1 < ksteps) {
       stage_mx(A, M, K, bm, (ks + 1) * BK, aRegs);
       stage_mx(Bt, N, K, bn, (ks + 1) * BK, bRegs);
+      scA = stage_sc(sA, M, groups, bm, ks + 1);
+      scB = stage_sc(sB, N, groups, bn, ks + 1);
     }
     // fragments: 32 B per lane = 2 x b128 at [row][fkb*32 .. +32)
     i32x8v aF[4], bF[2];
@@ -200,7 +235,7 @@ __global__ __launch_bounds__(THREADS2, 2) void gemm_mx_kernel(
       i32x4 lo = *reinterpret_cast<const i32x4*>(p);
       i32x4 hi = *reinterpret_cast<const i32x4*>(p + 16);
       aF[mf] = i32x8v{lo.x, lo.y, lo.z, lo.w, hi.x, hi.y, hi.z, hi.w};
-      sa[mf] = sA[((long)(bm + r) * groups + ks) * 2 + fkb];
+      sa[mf] = lsA[r * 2 + fkb];
     }
 #pragma unroll
     for (int nf = 0; nf < 2; ++nf) {
@@ -209,7 +244,7 @@ __global__ __launch_bounds__(THREADS2, 2) void gemm_mx_kernel(
       i32x4 lo = *reinterpret_cast<const i32x4*>(p);
       i32x4 hi = *reinterpret_cast<const i32x4*>(p + 16);
       bF[nf] = i32x8v{lo.x, lo.y, lo.z, lo.w, hi.x, hi.y, hi.z, hi.w};
-      sb[nf] = sB[((long)(bn + r) * groups + ks) * 2 + fkb];
+      sb[nf] = lsB[r * 2 + fkb];
     }
 #pragma unroll
     for (int mf = 0; mf < 4; ++mf)
@@ -221,6 +256,10 @@ __global__ __launch_bounds__(THREADS2, 2) void gemm_mx_kernel(
     if (ks + 1 < ksteps) {
       write_mx_lds(aRegs, lA);
       write_mx_lds(bRegs, lB);
+      if (threadIdx.x < 128) {
+        reinterpret_cast<unsigned*>(lsA)[threadIdx.x] = scA;
+        reinterpret_cast<unsigned*>(lsB)[threadIdx.x] = scB;
+      }
     }
     __syncthreads();
   }
@@ -261,7 +300,7 @@ hipError_t gemm_mx_fp8_nt(const unsigned char* A, const unsigned char* sA,
                           float* C, int M, int N, int K, hipStream_t stream) {
   if (K % 64 != 0) return hipErrorInvalidValue;
   const unsigned tiles_m = lpx_ceil_div(M, BM), tiles_n = lpx_ceil_div(N, BN);
-  const size_t lds = (size_t)2 * BM * ROWB;  // 40 KiB
+  const size_t lds = (size_t)2 * BM * ROWB + 4 * BM;  // 40 KiB + scales
   gemm_mx_kernel<<<dim3(tiles_m * tiles_n), dim3(THREADS2), lds, stream>>>(
       A, sA, Bt, sB, C, M, N, K, tiles_m, tiles_n);
   return hipGetLastError();

@@ -1,9 +1,11 @@
 """A/B: MX block-scaled fp8 GEMM vs plain fp8 v2/v3 (TFLOP/s)."""
+import sys
 import time
 
 import torch
 
-from lambdipy_amd import ops
+sys.path.insert(0, "/root/repo")
+from lambdipy_amd import ops  # noqa: E402
 
 e = ops.ext()
 

@@ -1,0 +1,68 @@
+// Raw MFMA issue-rate probe: scaled 32x32x64 f8f6f4 vs plain 16x16x32 fp8.
+// Register-resident operands, 8 independent accumulator chains per thread.
+// Build (on a GPU box): hipcc --offload-arch=gfx950 -O3 -shared -fPIC
+//   mfma_rate.hip -o mfma_rate.so
+#include <hip/hip_runtime.h>
+
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(16))) float f32x16;
+typedef __attribute__((ext_vector_type(8))) int i32x8v;
+
+__global__ __launch_bounds__(512) void rate_scaled(float* out, int iters) {
+  i32x8v a = {0, 0, 0, 0, 0, 0, 0, 0}, b = a;
+  f32x16 acc[8];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int e = 0; e < 16; ++e) acc[i][e] = 0.f;
+  for (int it = 0; it < iters; ++it) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+      acc[i] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+          a, b, acc[i], 0, 0, 0, 127, 0, 127);
+  }
+  float s = 0.f;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) s += acc[i][0];
+  if (s != 0.f) out[0] = s;  // never taken; keeps the loop alive
+}
+
+__global__ __launch_bounds__(512) void rate_plain(float* out, int iters) {
+  long a = 0, b = 0;
+  f32x4 acc[8];
+#pragma unroll
+  for (int i = 0; i < 8; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+  for (int it = 0; it < iters; ++it) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+      acc[i] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(a, b, acc[i], 0, 0, 0);
+  }
+  float s = 0.f;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) s += acc[i][0];
+  if (s != 0.f) out[0] = s;
+}
+
+extern "C" int run_rate(int which, int blocks, int iters, float* ms_out) {
+  float* d;
+  if (hipMalloc(&d, 4) != hipSuccess) return 1;
+  hipEvent_t e0, e1;
+  hipEventCreate(&e0);
+  hipEventCreate(&e1);
+  // warmup
+  if (which == 0)
+    rate_scaled<<<blocks, 512>>>(d, iters / 10 + 1);
+  else
+    rate_plain<<<blocks, 512>>>(d, iters / 10 + 1);
+  hipDeviceSynchronize();
+  hipEventRecord(e0);
+  if (which == 0)
+    rate_scaled<<<blocks, 512>>>(d, iters);
+  else
+    rate_plain<<<blocks, 512>>>(d, iters);
+  hipEventRecord(e1);
+  if (hipDeviceSynchronize() != hipSuccess) return 2;
+  hipEventElapsedTime(ms_out, e0, e1);
+  hipFree(d);
+  return 0;
+}

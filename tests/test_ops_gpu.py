@@ -247,13 +247,13 @@ def test_gemm_fp8_vs_dequant_ref(ops):
 def _dequant_mx(q, sc):
     """Reconstruct f32 from MX-quantized (shuffled) bytes + E8M0 scales.
 
-    Shuffled 64-group order is [q0|q2|q1|q3]; block kb covers shuffled
-    bytes [kb*32, +32) (= original quarters {kb, kb+2})."""
+    Shuffled 64-group order is [q0|q2|q1|q3]; scale kb covers the contiguous
+    original block {q_2kb, q_2kb+1} = shuffled slots {kb, kb+2}."""
     R, K = q.shape
     g = K // 64
     d = q.view(R, g, 4, 16).view(torch.float8_e4m3fn).float()
     scales = torch.pow(2.0, sc.float() - 127.0)          # [R, g, 2]
-    d = d * scales.repeat_interleave(2, dim=2).unsqueeze(-1)
+    d = d * scales[:, :, [0, 1, 0, 1]].unsqueeze(-1)     # slots q0,q2,q1,q3
     return d[:, :, [0, 2, 1, 3], :].reshape(R, K)        # unshuffle
 
 
@@ -262,19 +262,15 @@ def test_quantize_mx_roundtrip(ops):
     g = torch.Generator(device="cpu").manual_seed(7)
     # wide dynamic range across blocks: per-row magnitude ramp
     x = (torch.randn(64, 256, generator=g) *
-         torch.logspace(-6, 6, 64, base=2.0, generator=None).unsqueeze(1)).to(DEV)
+         torch.logspace(-6, 6, 64, base=2.0).unsqueeze(1)).to(DEV)
     q, sc = e.quantize_mx(x)
     xd = _dequant_mx(q, sc)
-    # per-element rel error bounded by e4m3 step within each 32-elem block
-    err = (xd - x).abs()
-    amax = x.abs().amax().clamp(min=1e-30)
-    # block-relative: compare against per-block amax
-    blk = x.view(64, 4, 4, 16)[:, :, [0, 2, 1, 3], :].reshape(64, 4, 2, 32)
-    blk_amax = blk.abs().amax(dim=-1, keepdim=True)
-    err_blk = (err.view(64, 4, 4, 16)[:, :, [0, 2, 1, 3], :]
-               .reshape(64, 4, 2, 32) / blk_amax.clamp(min=1e-30))
-    assert err_blk.max().item() < 0.07, err_blk.max().item()
-    del amax
+    # per-element error bounded by the e4m3 step within each contiguous
+    # 32-element MX block (relative to the block amax)
+    err = (xd - x).abs().view(64, 4, 2, 32)
+    blk_amax = x.view(64, 4, 2, 32).abs().amax(dim=-1, keepdim=True)
+    err_blk = err / blk_amax.clamp(min=1e-30)
+    assert err_blk.max().item() < 0.08, err_blk.max().item()
 
 
 def test_gemm_mx_vs_dequant_ref(ops):
@@ -289,10 +285,11 @@ def test_gemm_mx_vs_dequant_ref(ops):
     ref = _dequant_mx(qa, sa) @ _dequant_mx(qb, sb).t()
     err = (C - ref).abs().max().item() / (ref.abs().max().item() + 1e-6)
     assert err < 1e-3, err
-    # end-to-end quantization error vs exact f32 matmul stays small
+    # end-to-end quantization error vs exact f32 matmul: ~4% Frobenius is
+    # the expected e4m3 noise floor for random data (element RMS ~3%)
     exact = A @ Bt.t()
     rel = (C - exact).norm().item() / exact.norm().item()
-    assert rel < 0.02, rel
+    assert rel < 0.06, rel
 
 
 def test_gemm_mx_tail_shapes(ops):
