@@ -1,0 +1,231 @@
+#include "hip/hip_runtime.h"
+// MX block-scaled fp8 GEMM v2 for gfx950 — glds 8-phase pipeline.
+//
+// Same quadrant-phase schedule as gemm_bf16_8ph.hip / fp8_gemm_8ph.hip
+// (one half-tile glds prefetch per phase, counted vmcnt across raw
+// barriers, T5 static young-half priority) but built around
+// v_mfma_scale_f32_32x32x64_f8f6f4, which issues at 2x the plain-fp8 rate
+// (4.90 vs 2.43 PF measured register-resident, scripts/mfma_rate.py).
+//
+// Differences from the plain fp8 8-phase kernel:
+//   - fragments are 32 B/lane (2 consecutive 16-B cells), so the LDS cell
+//     permutation is rotation by (row>>1)&7 — (row&7) would 2-way conflict
+//     the 16-lane b128 quarters (rows step 2 within a bank half).
+//   - wave tile is 64x32 of each 128x128 quadrant: 2x1 fragments of 32x32.
+//   - per stage call each thread issues 3 glds (2x16-B data + 1x4-B scale
+//     dword), keeping vmcnt counts uniform; scale dwords land in a 512-B
+//     per-half LDS strip ([128 rows][4 bytes], tile-kt's 2 groups x 2
+//     blocks), duplicated by wave pairs (same-source benign).
+//   - operands/scales are the shuffled MX packing of quantize_mx_fp8
+//     (fp8_mx.hip); scales are applied by the MFMA, no epilogue descale.
+// Requires M,N % 256 == 0, K % 128 == 0 (launcher falls back to the
+// register-staged MX kernel for other shapes).
+
+#include "lpx_common.h"
+
+namespace lpx {
+
+namespace {
+
+constexpr int BMX = 256, BKX = 128;  // 128 fp8 elements = 128 B per row
+constexpr int THREADSX = 512;
+constexpr int HALF_DATA = 128 * 128;               // bytes per half-tile
+constexpr int HALF_SC = 512;                       // scale strip per half
+constexpr int HALF_BYTES = HALF_DATA + HALF_SC;
+constexpr int PARITY_BYTES = 4 * HALF_BYTES;
+
+typedef __attribute__((ext_vector_type(8))) int i32x8v;
+
+// logical 16-B cell -> physical: rotation by (row>>1) within the row
+__device__ __forceinline__ int phys_cellx(int row, int cell) {
+  return row * 8 + ((cell + (row >> 1)) & 7);
+}
+// inverse: which logical cell belongs at linear slot q = row*8 + pc
+__device__ __forceinline__ int src_cellx(int q) {
+  const int row = q >> 3;
+  return ((q & 7) - (row >> 1)) & 7;
+}
+
+// Stage one half-tile (128 rows x 128 B data + 512-B scale strip) for
+// `tile` with 3 glds per thread. hh: 0/1 = A halves, 2/3 = B halves.
+__device__ __forceinline__ void stage_halfx(
+    const unsigned char* __restrict__ A, const unsigned char* __restrict__ sA,
+    const unsigned char* __restrict__ Bt, const unsigned char* __restrict__ sB,
+    long K, int bm, int bn, int tile, int hh, int ksteps,
+    char* __restrict__ smem, int w, int lane) {
+  const int tc = tile < ksteps ? tile : ksteps - 1;
+  const long k0 = (long)tc * BKX;
+  const unsigned char* src = (hh < 2) ? A : Bt;
+  const unsigned char* ssrc = (hh < 2) ? sA : sB;
+  const int row0 = ((hh < 2) ? bm : bn) + (hh & 1) * 128;
+  char* const base = smem + (size_t)(tile & 1) * PARITY_BYTES +
+                     (size_t)hh * HALF_BYTES;
+  // data: 1024 cells of 16 B; each thread fetches the global cell that
+  // belongs at its linear LDS slot (rotation absorbed on the source side)
+#pragma unroll
+  for (int j = 0; j < 2; ++j) {
+    const int qbase = (j * 8 + w) * 64;
+    const int q = qbase + lane;
+    const int r = q >> 3, cell = src_cellx(q);
+    const unsigned char* g = src + (long)(row0 + r) * K + (k0 + cell * 16);
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)g,
+        (__attribute__((address_space(3))) void*)(base + qbase * 16), 16, 0, 0);
+  }
+  // scales: one dword per row ([2 groups][2 blocks] for this k-tile);
+  // wave pairs duplicate the 128 dwords (same source, benign).
+  const int di = (w & 1) * 64 + lane;                       // dword 0..127
+  const long scstride = (K >> 6) * 2;                       // bytes per row
+  const unsigned char* g = ssrc + (long)(row0 + di) * scstride + (long)tc * 4;
+  __builtin_amdgcn_global_load_lds(
+      (const __attribute__((address_space(1))) void*)g,
+      (__attribute__((address_space(3))) void*)(base + HALF_DATA +
+                                                (w & 1) * 256),
+      4, 0, 0);
+}
+
+__device__ __forceinline__ void stage_slotx(int t, int p, int* tile, int* hh) {
+  switch (p) {
+    case 0: *tile = t + 1; *hh = 3; break;
+    case 1: *tile = t + 1; *hh = 1; break;
+    case 2: *tile = t + 2; *hh = 0; break;
+    default: *tile = t + 2; *hh = 2; break;
+  }
+}
+
+// 32-B fragment (2 x b128) for row_in_half, k-group kg, block half fkb
+__device__ __forceinline__ i32x8v fragx(const char* __restrict__ smem,
+                                        int parity, int hh, int row_in_half,
+                                        int kg, int fkb) {
+  const char* half = smem + (size_t)parity * PARITY_BYTES +
+                     (size_t)hh * HALF_BYTES;
+  const int c0 = kg * 4 + fkb * 2;  // logical cells c0, c0+1
+  const i32x4 lo = *reinterpret_cast<const i32x4*>(
+      half + phys_cellx(row_in_half, c0) * 16);
+  const i32x4 hi = *reinterpret_cast<const i32x4*>(
+      half + phys_cellx(row_in_half, c0 + 1) * 16);
+  return i32x8v{lo.x, lo.y, lo.z, lo.w, hi.x, hi.y, hi.z, hi.w};
+}
+
+__device__ __forceinline__ int scalex(const char* __restrict__ smem,
+                                      int parity, int hh, int row_in_half,
+                                      int kg, int fkb) {
+  const char* half = smem + (size_t)parity * PARITY_BYTES +
+                     (size_t)hh * HALF_BYTES;
+  return (int)*reinterpret_cast<const unsigned char*>(
+      half + HALF_DATA + row_in_half * 4 + kg * 2 + fkb);
+}
+
+__global__ __launch_bounds__(THREADSX, 1) void gemm_mx_8ph_kernel(
+    const unsigned char* __restrict__ A, const unsigned char* __restrict__ sA,
+    const unsigned char* __restrict__ Bt, const unsigned char* __restrict__ sB,
+    float* __restrict__ C, int M, int N, int K, unsigned tiles_m,
+    unsigned tiles_n) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+
+  const unsigned nwg = tiles_m * tiles_n;
+  const unsigned bid = lpx_xcd_swizzle(blockIdx.x, nwg);
+  const int bm = (int)(bid / tiles_n) * BMX;
+  const int bn = (int)(bid % tiles_n) * BMX;
+
+  const int lane = threadIdx.x & 63;
+  const int w = threadIdx.x >> 6;
+  const int wsr = w >> 2, wsc = w & 3;  // 2x4 waves; 64x32 of each quadrant
+  const int fr = lane & 31;
+  const int fkb = lane >> 5;
+
+  f32x16 acc[4][2];  // [quadrant][m-frag], 1 n-frag per wave
+#pragma unroll
+  for (int p = 0; p < 4; ++p)
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+      for (int e = 0; e < 16; ++e) acc[p][i][e] = 0.f;
+
+  const int ksteps = (K + BKX - 1) / BKX;
+
+  stage_halfx(A, sA, Bt, sB, K, bm, bn, 0, 0, ksteps, smem, w, lane);
+  stage_halfx(A, sA, Bt, sB, K, bm, bn, 0, 2, ksteps, smem, w, lane);
+  stage_halfx(A, sA, Bt, sB, K, bm, bn, 0, 3, ksteps, smem, w, lane);
+  stage_halfx(A, sA, Bt, sB, K, bm, bn, 0, 1, ksteps, smem, w, lane);
+  stage_halfx(A, sA, Bt, sB, K, bm, bn, 1, 0, ksteps, smem, w, lane);
+  stage_halfx(A, sA, Bt, sB, K, bm, bn, 1, 2, ksteps, smem, w, lane);
+  asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);  // T5 static young-half priority
+
+  for (int t = 0; t < ksteps; ++t) {
+    const int parity = t & 1;
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      const int rh = (p >> 1) & 1;
+      const int ch = p & 1;
+      i32x8v aF[2][2], bF[2];
+      int sa[2][2], sb[2];
+#pragma unroll
+      for (int kg = 0; kg < 2; ++kg) {
+#pragma unroll
+        for (int mf = 0; mf < 2; ++mf) {
+          const int r = wsr * 64 + mf * 32 + fr;
+          aF[mf][kg] = fragx(smem, parity, rh, r, kg, fkb);
+          sa[mf][kg] = scalex(smem, parity, rh, r, kg, fkb);
+        }
+        const int rb = wsc * 32 + fr;
+        bF[kg] = fragx(smem, parity, 2 + ch, rb, kg, fkb);
+        sb[kg] = scalex(smem, parity, 2 + ch, rb, kg, fkb);
+      }
+      int st, sh;
+      stage_slotx(t, p, &st, &sh);
+      stage_halfx(A, sA, Bt, sB, K, bm, bn, st, sh, ksteps, smem, w, lane);
+
+      __builtin_amdgcn_s_barrier();
+#pragma unroll
+      for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+        for (int kg = 0; kg < 2; ++kg)
+          acc[p][mf] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+              aF[mf][kg], bF[kg], acc[p][mf], 0, 0, 0, sa[mf][kg], 0, sb[kg]);
+      asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+
+  const int ocol_l = lane & 31;
+  const int orow_base = 4 * (lane >> 5);
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    const int rh = (p >> 1) & 1, ch = p & 1;
+#pragma unroll
+    for (int mf = 0; mf < 2; ++mf) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int row = bm + rh * 128 + wsr * 64 + mf * 32 + (r & 3) +
+                        8 * (r >> 2) + orow_base;
+        const int col = bn + ch * 128 + wsc * 32 + ocol_l;
+        C[(long)row * N + col] = acc[p][mf][r];
+      }
+    }
+  }
+}
+
+}  // namespace
+
+bool gemm_mx_8ph_supported(int M, int N, int K) {
+  return M % BMX == 0 && N % BMX == 0 && K % BKX == 0 && K >= BKX;
+}
+
+hipError_t gemm_mx_fp8_nt8ph(const unsigned char* A, const unsigned char* sA,
+                             const unsigned char* Bt, const unsigned char* sB,
+                             float* C, int M, int N, int K,
+                             hipStream_t stream) {
+  const unsigned tiles_m = (unsigned)(M / BMX), tiles_n = (unsigned)(N / BMX);
+  const size_t lds = 2 * PARITY_BYTES;  // 132 KiB
+ hipLaunchKernelGGL(( gemm_mx_8ph_kernel), dim3(dim3(tiles_m * tiles_n)), dim3(dim3(THREADSX)), lds, stream, 
+      A, sA, Bt, sB, C, M, N, K, tiles_m, tiles_n);
+  return hipGetLastError();
+}
+
+}  // namespace lpx

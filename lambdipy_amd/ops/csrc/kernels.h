@@ -126,5 +126,11 @@ hipError_t quantize_mx_fp8(const float* x, unsigned char* q, unsigned char* sc,
 hipError_t gemm_mx_fp8_nt(const unsigned char* A, const unsigned char* sA,
                           const unsigned char* Bt, const unsigned char* sB,
                           float* C, int M, int N, int K, hipStream_t stream);
+// v2: glds 8-phase pipeline (M,N%256==0, K%128==0), 2x MFMA rate.
+bool gemm_mx_8ph_supported(int M, int N, int K);
+hipError_t gemm_mx_fp8_nt8ph(const unsigned char* A, const unsigned char* sA,
+                             const unsigned char* Bt, const unsigned char* sB,
+                             float* C, int M, int N, int K,
+                             hipStream_t stream);
 
 }  // namespace lpx

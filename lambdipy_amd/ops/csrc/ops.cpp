@@ -307,11 +307,19 @@ at::Tensor gemm_mx(const at::Tensor& A, const at::Tensor& sA,
               "scale shapes must be [rows, K/64, 2]");
   c10::hip::HIPGuardMasqueradingAsCUDA guard(A.device());
   auto C = at::empty({M, N}, A.options().dtype(at::kFloat));
-  check_hip(lpx::gemm_mx_fp8_nt(A.data_ptr<uint8_t>(), sA.data_ptr<uint8_t>(),
-                                Bt.data_ptr<uint8_t>(), sB.data_ptr<uint8_t>(),
-                                C.data_ptr<float>(), (int)M, (int)N, (int)K,
-                                current_stream(A)),
-            "gemm_mx_fp8_nt");
+  if (lpx::gemm_mx_8ph_supported((int)M, (int)N, (int)K))
+    check_hip(lpx::gemm_mx_fp8_nt8ph(
+                  A.data_ptr<uint8_t>(), sA.data_ptr<uint8_t>(),
+                  Bt.data_ptr<uint8_t>(), sB.data_ptr<uint8_t>(),
+                  C.data_ptr<float>(), (int)M, (int)N, (int)K,
+                  current_stream(A)),
+              "gemm_mx_fp8_nt8ph");
+  else
+    check_hip(lpx::gemm_mx_fp8_nt(A.data_ptr<uint8_t>(), sA.data_ptr<uint8_t>(),
+                                  Bt.data_ptr<uint8_t>(), sB.data_ptr<uint8_t>(),
+                                  C.data_ptr<float>(), (int)M, (int)N, (int)K,
+                                  current_stream(A)),
+              "gemm_mx_fp8_nt");
   return C;
 }
 
