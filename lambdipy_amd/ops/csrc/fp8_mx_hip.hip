@@ -27,6 +27,7 @@ namespace {
 constexpr int THREADS = 256;
 
 typedef __attribute__((ext_vector_type(8))) int i32x8v;
+typedef __attribute__((ext_vector_type(2))) short s16x2;
 
 // ---- MX quantization -----------------------------------------------------
 // x f32 [R][K] row-major, K % 64 == 0. Outputs:
@@ -46,8 +47,15 @@ __global__ __launch_bounds__(THREADS) void quantize_mx_kernel(
     // standard contiguous MX-32 block: original k [g*64 + kb*32, +32).
     // The MFMA applies lane-kb's scale to byte-half kb of BOTH k-lanes;
     // with the [q0|q2|q1|q3] data shuffle that is exactly this block.
+    const f32x4* src4 = reinterpret_cast<const f32x4*>(src);
+    f32x4 buf[8];
     float amax = 0.f;
-    for (int j = 0; j < 32; ++j) amax = fmaxf(amax, fabsf(src[j]));
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      buf[i] = src4[i];
+#pragma unroll
+      for (int j = 0; j < 4; ++j) amax = fmaxf(amax, fabsf(buf[i][j]));
+    }
     // E8M0 shared scale: 2^e, non-clipping (amax/2^e <= 448, the e4m3 max)
     int e = 0;
     if (amax > 0.f) {
@@ -59,51 +67,28 @@ __global__ __launch_bounds__(THREADS) void quantize_mx_kernel(
     if (e < -127) e = -127;
     if (e > 127) e = 127;
     sc[((long)r * groups + g) * 2 + kb] = (unsigned char)(e + 127);
-    const float inv = exp2f((float)-e);
+    const float scl = exp2f((float)e);  // hw cvt divides by the scale operand
     unsigned char* base = q + (long)r * K + g * 64;
 #pragma unroll
     for (int h = 0; h < 2; ++h) {
-      const int qq = 2 * kb + h;                  // original quarter index
+      const int qq = 2 * kb + h;                    // original quarter index
       const int pos = ((qq & 1) << 1) | (qq >> 1);  // shuffled slot
-      unsigned char* dst = base + pos * 16;
-      for (int j = 0; j < 16; ++j) {
-        float v = src[h * 16 + j] * inv;
-        // e4m3 with saturation (reuse bit algorithm from fp8_gemm.hip)
-        union { float f; uint32_t u; } c;
-        c.f = v;
-        const uint32_t sign = (c.u >> 24) & 0x80;
-        float ax = fabsf(v);
-        unsigned char out;
-        if (!(ax == ax)) {
-          out = 0x7f;
-        } else {
-          if (ax > 448.f) ax = 448.f;
-          c.f = ax;
-          int32_t ee = (int32_t)((c.u >> 23) & 0xff) - 127;
-          uint32_t m = c.u & 0x7fffff;
-          uint32_t o;
-          if (ax == 0.f) {
-            o = 0;
-          } else if (ee < -6) {
-            float scaled = ax * 512.f;
-            uint32_t qq = (uint32_t)scaled;
-            float frac = scaled - (float)qq;
-            if (frac > 0.5f || (frac == 0.5f && (qq & 1))) ++qq;
-            o = (qq >= 8) ? (1u << 3) : qq;
-          } else {
-            uint32_t mant = m >> 20;
-            uint32_t rest = m & 0xfffff;
-            if (rest > 0x80000 || (rest == 0x80000 && (mant & 1))) {
-              ++mant;
-              if (mant == 8) { mant = 0; ++ee; }
-            }
-            if (ee > 8) { ee = 8; mant = 7; }
-            o = ((uint32_t)(ee + 7) << 3) | mant;
-          }
-          out = (unsigned char)o;
-        }
-        dst[j] = (unsigned char)(sign | out);
+      // v_cvt_scalef32_pk_fp8_f32: RNE fp8(src/scale), 2 elems per call
+      s16x2 pk[4];
+#pragma unroll
+      for (int jj = 0; jj < 4; ++jj) {
+        const f32x4 v = buf[h * 4 + jj];
+        s16x2 t = {0, 0};
+        t = __builtin_amdgcn_cvt_scalef32_pk_fp8_f32(t, v[0], v[1], scl, false);
+        t = __builtin_amdgcn_cvt_scalef32_pk_fp8_f32(t, v[2], v[3], scl, true);
+        pk[jj] = t;
       }
+      i32x4 out;
+      out.x = (int)(unsigned short)pk[0][0] | ((int)(unsigned short)pk[0][1] << 16);
+      out.y = (int)(unsigned short)pk[1][0] | ((int)(unsigned short)pk[1][1] << 16);
+      out.z = (int)(unsigned short)pk[2][0] | ((int)(unsigned short)pk[2][1] << 16);
+      out.w = (int)(unsigned short)pk[3][0] | ((int)(unsigned short)pk[3][1] << 16);
+      *reinterpret_cast<i32x4*>(base + pos * 16) = out;
     }
   }
 }

@@ -198,6 +198,24 @@ def assemble_layer(
         report.stripped_files += sn
         report.stripped_bytes += sb
 
+    # strip non-gfx950 fat-binary entries (HIP multi-arch DSOs)
+    if any(registry.get(d.recipe).strip_archs for d in resolution.recipe_hits if registry.get(d.recipe)):
+        from .fatbin import strip_fatbin
+
+        for root, _dirs, files in os.walk(build_dir):
+            for fn in files:
+                if ".so" not in fn:
+                    continue
+                p = os.path.join(root, fn)
+                try:
+                    st = strip_fatbin(p)
+                except Exception as exc:  # noqa: BLE001 - leave lib intact
+                    log.warning("strip-archs skipped %s: %s", p, exc)
+                    continue
+                if st["new_size"] < st["old_size"]:
+                    report.stripped_files += 1
+                    report.stripped_bytes += st["old_size"] - st["new_size"]
+
     if handler_files:
         for rel, local in handler_files.items():
             out = os.path.join(build_dir, rel)

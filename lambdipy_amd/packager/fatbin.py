@@ -159,9 +159,9 @@ def _list_targets(path: str, bundler: str) -> List[str]:
 
 
 def _rebundle(blob: bytes, keep_archs: Tuple[str, ...], tmp: str,
-              bundler: str) -> Optional[bytes]:
+              bundler: str, tag: str, level: int) -> Optional[bytes]:
     """Rebundle keeping host + keep_archs. None => keep original bytes."""
-    src = os.path.join(tmp, "in.bundle")
+    src = os.path.join(tmp, f"in{tag}.bundle")
     with open(src, "wb") as f:
         f.write(blob)
     targets = _list_targets(src, bundler)
@@ -171,25 +171,26 @@ def _rebundle(blob: bytes, keep_archs: Tuple[str, ...], tmp: str,
     if not keep or len(keep) + len(host) == len(targets):
         return None  # nothing to strip (or target arch absent: leave intact)
     sel = host[:1] + keep
-    outs = [os.path.join(tmp, f"t{i}.bin") for i in range(len(sel))]
+    outs = [os.path.join(tmp, f"t{tag}_{i}.bin") for i in range(len(sel))]
     cmd = [bundler, "--type=o", "--unbundle", f"--input={src}",
            "--targets=" + ",".join(sel)] + [f"--output={o}" for o in outs]
     subprocess.run(cmd, check=True, capture_output=True)
-    dst = os.path.join(tmp, "out.bundle")
+    dst = os.path.join(tmp, f"out{tag}.bundle")
     cmd = [bundler, "--type=o", "--targets=" + ",".join(sel),
-           f"--output={dst}", "-compress", "-compression-level=19"]
+           f"--output={dst}", "-compress", f"-compression-level={level}"]
     cmd += [f"--input={o}" for o in outs]
     subprocess.run(cmd, check=True, capture_output=True)
     with open(dst, "rb") as f:
         new = f.read()
-    for o in outs:
+    for o in outs + [src, dst]:
         os.unlink(o)
     return new if len(new) < len(blob) else blob
 
 
 def strip_fatbin(src: str, dst: Optional[str] = None,
                  archs: Tuple[str, ...] = ("gfx950",),
-                 bundler: str = BUNDLER, verbose: bool = False) -> Dict:
+                 bundler: str = BUNDLER, verbose: bool = False,
+                 level: int = 9, jobs: int = 8) -> Dict:
     """Strip non-`archs` GPU code from a HIP DSO. Returns a stats dict.
 
     Writes to `dst` (default: in-place via temp + rename). No-op (still
@@ -227,20 +228,30 @@ def strip_fatbin(src: str, dst: Optional[str] = None,
         wrappers.append((p + 8, ptr))
     stats["bundles"] = len(wrappers)
 
-    # 2. rebundle each
+    # 2. rebundle each (subprocess-bound -> thread pool)
     fat_off = _vaddr_to_off(e, fat.addr)
     new_blobs: Dict[int, bytes] = {}  # old vaddr -> new bytes
+    uniq = []
+    for _, va in wrappers:
+        if va not in new_blobs:
+            new_blobs[va] = b""
+            uniq.append(va)
+    raw = bytes(data)
     with tempfile.TemporaryDirectory(prefix="lpx_fatbin_") as tmp:
-        for _, va in wrappers:
+        def work(va):
             off = fat_off + (va - fat.addr)
-            ext = _bundle_extent(bytes(data), off)
-            blob = bytes(data[off:off + ext])
-            new = _rebundle(blob, archs, tmp, bundler)
-            if new is None:
-                new = blob
-            else:
-                stats["stripped"] += 1
-            new_blobs[va] = new
+            blob = raw[off:off + _bundle_extent(raw, off)]
+            return va, blob, _rebundle(blob, archs, tmp, bundler,
+                                       f"{va:x}", level)
+
+        from concurrent.futures import ThreadPoolExecutor
+        with ThreadPoolExecutor(max_workers=jobs) as pool:
+            for va, blob, new in pool.map(work, uniq):
+                if new is None:
+                    new = blob
+                else:
+                    stats["stripped"] += 1
+                new_blobs[va] = new
 
     # 3. lay out the new section (64-B aligned bundles)
     layout: Dict[int, int] = {}

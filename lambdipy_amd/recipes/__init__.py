@@ -60,6 +60,7 @@ class BuildRecipe:
     excludes: List[str] = field(default_factory=list)
     prune: List[str] = field(default_factory=list)
     strip_native: bool = True
+    strip_archs: bool = False  # strip non-gfx950 HIP fat-binary entries
     vendor_upstream: bool = True  # False: upstream pkg is runtime-provided
     runtime_requires: List[str] = field(default_factory=list)
     description: str = ""
@@ -88,6 +89,7 @@ class BuildRecipe:
             excludes=[canonicalize_name(e) for e in d.get("excludes", [])],
             prune=list(d.get("prune", [])),
             strip_native=bool(d.get("strip_native", True)),
+            strip_archs=bool(d.get("strip_archs", False)),
             vendor_upstream=bool(d.get("vendor_upstream", True)),
             runtime_requires=list(d.get("runtime_requires", [])),
             description=d.get("description", ""),
