@@ -530,6 +530,10 @@ hipError_t im2col_fp8(const float* x, unsigned char* out, int N, int H, int W,
 hipError_t gemm_fp8_nt(const unsigned char* A, const unsigned char* Bt,
                        float* C, const float* bias, int M, int N, int K,
                        float descale, bool relu, hipStream_t stream) {
+  // scaled-MFMA path first: 2x the 16x16x32 instruction rate (unit scales)
+  if (M >= 512 && N >= 512 && M % 256 == 0 && N % 256 == 0 && K % 128 == 0 &&
+      K >= 128)
+    return gemm_fp8_nt8phx(A, Bt, C, bias, M, N, K, descale, relu, stream);
   if (M >= 1024 && N >= 1024 && gemm_fp8_8ph_supported(M, N, K))
     return gemm_fp8_nt8ph(A, Bt, C, bias, M, N, K, descale, relu, stream);
   const bool big = (M >= 512 && N >= 512 && K >= 64);

@@ -46,8 +46,20 @@ __device__ __forceinline__ int src_cellx(int q) {
   return ((q & 7) - (row >> 1)) & 7;
 }
 
+template <int N>
+__device__ __forceinline__ void waitvm() {
+  if constexpr (N == 0) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  else if constexpr (N == 4) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  else if constexpr (N == 6) asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+  else if constexpr (N == 8) asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+  else if constexpr (N == 12) asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
+  else static_assert(N == 0, "unsupported vmcnt");
+}
+
 // Stage one half-tile (128 rows x 128 B data + 512-B scale strip) for
-// `tile` with 3 glds per thread. hh: 0/1 = A halves, 2/3 = B halves.
+// `tile` with 3 glds per thread (2 when SC=false: no scale dword).
+// hh: 0/1 = A halves, 2/3 = B halves.
+template <bool SC>
 __device__ __forceinline__ void stage_halfx(
     const unsigned char* __restrict__ A, const unsigned char* __restrict__ sA,
     const unsigned char* __restrict__ Bt, const unsigned char* __restrict__ sB,
@@ -72,16 +84,18 @@ __device__ __forceinline__ void stage_halfx(
         (const __attribute__((address_space(1))) void*)g,
         (__attribute__((address_space(3))) void*)(base + qbase * 16), 16, 0, 0);
   }
-  // scales: one dword per row ([2 groups][2 blocks] for this k-tile);
-  // wave pairs duplicate the 128 dwords (same source, benign).
-  const int di = (w & 1) * 64 + lane;                       // dword 0..127
-  const long scstride = (K >> 6) * 2;                       // bytes per row
-  const unsigned char* g = ssrc + (long)(row0 + di) * scstride + (long)tc * 4;
-  __builtin_amdgcn_global_load_lds(
-      (const __attribute__((address_space(1))) void*)g,
-      (__attribute__((address_space(3))) void*)(base + HALF_DATA +
-                                                (w & 1) * 256),
-      4, 0, 0);
+  if constexpr (SC) {
+    // scales: one dword per row ([2 groups][2 blocks] for this k-tile);
+    // wave pairs duplicate the 128 dwords (same source, benign).
+    const int di = (w & 1) * 64 + lane;                     // dword 0..127
+    const long scstride = (K >> 6) * 2;                     // bytes per row
+    const unsigned char* g = ssrc + (long)(row0 + di) * scstride + (long)tc * 4;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)g,
+        (__attribute__((address_space(3))) void*)(base + HALF_DATA +
+                                                  (w & 1) * 256),
+        4, 0, 0);
+  }
 }
 
 __device__ __forceinline__ void stage_slotx(int t, int p, int* tile, int* hh) {
@@ -116,11 +130,16 @@ __device__ __forceinline__ int scalex(const char* __restrict__ smem,
       half + HALF_DATA + row_in_half * 4 + kg * 2 + fkb);
 }
 
+// SC=true: per-block MX scales from LDS (sA/sB used; descale/bias ignored).
+// SC=false: unit scales (literal 127) — plain per-tensor fp8 GEMM on the
+// 2x-rate scaled MFMA (the k-permutation is self-consistent between A and
+// B, so unshuffled operands contract correctly) with fused descale/bias/relu.
+template <bool SC, bool BIAS, bool RELU>
 __global__ __launch_bounds__(THREADSX, 1) void gemm_mx_8ph_kernel(
     const unsigned char* __restrict__ A, const unsigned char* __restrict__ sA,
     const unsigned char* __restrict__ Bt, const unsigned char* __restrict__ sB,
-    float* __restrict__ C, int M, int N, int K, unsigned tiles_m,
-    unsigned tiles_n) {
+    float* __restrict__ C, const float* __restrict__ bias, float descale,
+    int M, int N, int K, unsigned tiles_m, unsigned tiles_n) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
 
   const unsigned nwg = tiles_m * tiles_n;
@@ -144,13 +163,13 @@ __global__ __launch_bounds__(THREADSX, 1) void gemm_mx_8ph_kernel(
 
   const int ksteps = (K + BKX - 1) / BKX;
 
-  stage_halfx(A, sA, Bt, sB, K, bm, bn, 0, 0, ksteps, smem, w, lane);
-  stage_halfx(A, sA, Bt, sB, K, bm, bn, 0, 2, ksteps, smem, w, lane);
-  stage_halfx(A, sA, Bt, sB, K, bm, bn, 0, 3, ksteps, smem, w, lane);
-  stage_halfx(A, sA, Bt, sB, K, bm, bn, 0, 1, ksteps, smem, w, lane);
-  stage_halfx(A, sA, Bt, sB, K, bm, bn, 1, 0, ksteps, smem, w, lane);
-  stage_halfx(A, sA, Bt, sB, K, bm, bn, 1, 2, ksteps, smem, w, lane);
-  asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+  stage_halfx<SC>(A, sA, Bt, sB, K, bm, bn, 0, 0, ksteps, smem, w, lane);
+  stage_halfx<SC>(A, sA, Bt, sB, K, bm, bn, 0, 2, ksteps, smem, w, lane);
+  stage_halfx<SC>(A, sA, Bt, sB, K, bm, bn, 0, 3, ksteps, smem, w, lane);
+  stage_halfx<SC>(A, sA, Bt, sB, K, bm, bn, 0, 1, ksteps, smem, w, lane);
+  stage_halfx<SC>(A, sA, Bt, sB, K, bm, bn, 1, 0, ksteps, smem, w, lane);
+  stage_halfx<SC>(A, sA, Bt, sB, K, bm, bn, 1, 2, ksteps, smem, w, lane);
+  waitvm<SC ? 6 : 4>();
   __builtin_amdgcn_s_barrier();
 
   if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
@@ -170,15 +189,15 @@ __global__ __launch_bounds__(THREADSX, 1) void gemm_mx_8ph_kernel(
         for (int mf = 0; mf < 2; ++mf) {
           const int r = wsr * 64 + mf * 32 + fr;
           aF[mf][kg] = fragx(smem, parity, rh, r, kg, fkb);
-          sa[mf][kg] = scalex(smem, parity, rh, r, kg, fkb);
+          sa[mf][kg] = SC ? scalex(smem, parity, rh, r, kg, fkb) : 127;
         }
         const int rb = wsc * 32 + fr;
         bF[kg] = fragx(smem, parity, 2 + ch, rb, kg, fkb);
-        sb[kg] = scalex(smem, parity, 2 + ch, rb, kg, fkb);
+        sb[kg] = SC ? scalex(smem, parity, 2 + ch, rb, kg, fkb) : 127;
       }
       int st, sh;
       stage_slotx(t, p, &st, &sh);
-      stage_halfx(A, sA, Bt, sB, K, bm, bn, st, sh, ksteps, smem, w, lane);
+      stage_halfx<SC>(A, sA, Bt, sB, K, bm, bn, st, sh, ksteps, smem, w, lane);
 
       __builtin_amdgcn_s_barrier();
 #pragma unroll
@@ -187,11 +206,11 @@ __global__ __launch_bounds__(THREADSX, 1) void gemm_mx_8ph_kernel(
         for (int kg = 0; kg < 2; ++kg)
           acc[p][mf] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
               aF[mf][kg], bF[kg], acc[p][mf], 0, 0, 0, sa[mf][kg], 0, sb[kg]);
-      asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
+      waitvm<SC ? 12 : 8>();
       __builtin_amdgcn_s_barrier();
     }
   }
-  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  waitvm<0>();
 
   const int ocol_l = lane & 31;
   const int orow_base = 4 * (lane >> 5);
@@ -205,7 +224,11 @@ __global__ __launch_bounds__(THREADSX, 1) void gemm_mx_8ph_kernel(
         const int row = bm + rh * 128 + wsr * 64 + mf * 32 + (r & 3) +
                         8 * (r >> 2) + orow_base;
         const int col = bn + ch * 128 + wsc * 32 + ocol_l;
-        C[(long)row * N + col] = acc[p][mf][r];
+        float v = acc[p][mf][r];
+        if constexpr (!SC) v *= descale;
+        if constexpr (BIAS) v += bias[col];
+        if constexpr (RELU) v = v > 0.f ? v : 0.f;
+        C[(long)row * N + col] = v;
       }
     }
   }
@@ -223,8 +246,31 @@ hipError_t gemm_mx_fp8_nt8ph(const unsigned char* A, const unsigned char* sA,
                              hipStream_t stream) {
   const unsigned tiles_m = (unsigned)(M / BMX), tiles_n = (unsigned)(N / BMX);
   const size_t lds = 2 * PARITY_BYTES;  // 132 KiB
- hipLaunchKernelGGL(( gemm_mx_8ph_kernel), dim3(dim3(tiles_m * tiles_n)), dim3(dim3(THREADSX)), lds, stream, 
-      A, sA, Bt, sB, C, M, N, K, tiles_m, tiles_n);
+ hipLaunchKernelGGL(( gemm_mx_8ph_kernel<true, false, false>)
+      , dim3(dim3(tiles_m * tiles_n)), dim3(dim3(THREADSX)), lds, stream, 
+          A, sA, Bt, sB, C, nullptr, 1.f, M, N, K, tiles_m, tiles_n);
+  return hipGetLastError();
+}
+
+// Per-tensor-scale fp8 GEMM on the 2x-rate scaled MFMA (unit block scales).
+hipError_t gemm_fp8_nt8phx(const unsigned char* A, const unsigned char* Bt,
+                           float* C, const float* bias, int M, int N, int K,
+                           float descale, bool relu, hipStream_t stream) {
+  const unsigned tiles_m = (unsigned)(M / BMX), tiles_n = (unsigned)(N / BMX);
+  const size_t lds = 2 * PARITY_BYTES;
+  dim3 grid(tiles_m * tiles_n), block(THREADSX);
+  if (bias && relu)
+   hipLaunchKernelGGL(( gemm_mx_8ph_kernel<false, true, true>), dim3(grid), dim3(block), lds, stream, 
+        A, nullptr, Bt, nullptr, C, bias, descale, M, N, K, tiles_m, tiles_n);
+  else if (bias)
+   hipLaunchKernelGGL(( gemm_mx_8ph_kernel<false, true, false>), dim3(grid), dim3(block), lds, stream, 
+        A, nullptr, Bt, nullptr, C, bias, descale, M, N, K, tiles_m, tiles_n);
+  else if (relu)
+   hipLaunchKernelGGL(( gemm_mx_8ph_kernel<false, false, true>), dim3(grid), dim3(block), lds, stream, 
+        A, nullptr, Bt, nullptr, C, nullptr, descale, M, N, K, tiles_m, tiles_n);
+  else
+   hipLaunchKernelGGL(( gemm_mx_8ph_kernel<false, false, false>), dim3(grid), dim3(block), lds, stream, 
+        A, nullptr, Bt, nullptr, C, nullptr, descale, M, N, K, tiles_m, tiles_n);
   return hipGetLastError();
 }
 

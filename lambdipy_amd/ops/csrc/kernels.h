@@ -132,5 +132,10 @@ hipError_t gemm_mx_fp8_nt8ph(const unsigned char* A, const unsigned char* sA,
                              const unsigned char* Bt, const unsigned char* sB,
                              float* C, int M, int N, int K,
                              hipStream_t stream);
+// per-tensor-scale fp8 GEMM on the scaled MFMA with unit block scales
+// (same shape constraints as gemm_mx_8ph_supported).
+hipError_t gemm_fp8_nt8phx(const unsigned char* A, const unsigned char* Bt,
+                           float* C, const float* bias, int M, int N, int K,
+                           float descale, bool relu, hipStream_t stream);
 
 }  // namespace lpx

@@ -306,6 +306,27 @@ def test_gemm_mx_tail_shapes(ops):
     assert err < 1e-3, err
 
 
+def test_gemm_fp8_scaled_mfma_path(ops):
+    # 256-multiple shapes dispatch to the unit-scale scaled-MFMA kernel
+    e = ops.ext()
+    g = torch.Generator(device="cpu").manual_seed(11)
+    M, N, K = 512, 512, 384
+    A = torch.randn(M, K, generator=g).to(DEV)
+    Bt = torch.randn(N, K, generator=g).to(DEV)
+    bias = torch.randn(N, generator=g).to(DEV)
+    sa, sb = 0.25, 0.5
+    Aq = e.quantize_fp8(A, sa)
+    Btq = e.quantize_fp8(Bt, sb)
+    Ad = Aq.view(torch.float8_e4m3fn).float() * sa
+    Btd = Btq.view(torch.float8_e4m3fn).float() * sb
+    C = e.gemm_fp8(Aq, Btq, sa * sb)
+    ref = Ad @ Btd.t()
+    assert (C - ref).abs().max().item() / ref.abs().max().item() < 1e-3
+    C2 = e.gemm_fp8(Aq, Btq, sa * sb, bias, True)
+    ref2 = torch.relu(ref + bias)
+    assert (C2 - ref2).abs().max().item() < 1e-2
+
+
 def test_gemm_fp8_bias_relu(ops):
     e = ops.ext()
     g = torch.Generator(device="cpu").manual_seed(2)
