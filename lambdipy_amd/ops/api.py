@@ -151,7 +151,8 @@ def rf_predict_proba_sum(nodes: torch.Tensor, leaf_values: torch.Tensor,
         assert depth is not None
 
         def fn(n, lv, r, x):
-            return e.rf_predict_heap(n, lv, depth, x)
+            # top=-1: two-level kernel, auto-picked LDS-resident top levels
+            return e.rf_predict_heap(n, lv, depth, x, 1, -1)
     elif packed and chunks is not None:
         ctb, max_nodes = chunks
 

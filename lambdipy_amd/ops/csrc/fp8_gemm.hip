@@ -478,6 +478,107 @@ __global__ __launch_bounds__(THREADS2, 2) void conv_fp8_imp_kernel(
   }
 }
 
+// ---- implicit conv on the scaled MFMA (unit block scales) ----------------
+// Same staging/padded-LDS geometry, but v_mfma_scale_f32_32x32x64_f8f6f4 at
+// 2x the 16x16x32 rate: wave tile 128x64 = 4x2 fragments of 32x32, one
+// MFMA per fragment pair per 64-B k-tile (the operand k-permutation is
+// self-consistent between A and B). 80-B padded rows are conflict-free for
+// the 16-lane b128 quarters (bank starts 20r mod 64 tile perfectly).
+
+typedef __attribute__((ext_vector_type(8))) int fp8frag32;
+
+__device__ __forceinline__ fp8frag32 frag32(const char* __restrict__ lds,
+                                            int row, int kb) {
+  const char* p = lds + row * ROWB2 + kb * 32;
+  const i32x4 lo = *reinterpret_cast<const i32x4*>(p);
+  const i32x4 hi = *reinterpret_cast<const i32x4*>(p + 16);
+  return fp8frag32{lo.x, lo.y, lo.z, lo.w, hi.x, hi.y, hi.z, hi.w};
+}
+
+template <bool RELU, bool BIAS>
+__global__ __launch_bounds__(THREADS2, 2) void conv_fp8_impx_kernel(
+    const unsigned char* __restrict__ xq, const unsigned char* __restrict__ Wq,
+    float* __restrict__ C, const float* __restrict__ bias, ConvGeom g, int M,
+    int N, int K, float descale, unsigned tiles_m, unsigned tiles_n) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* const lA = smem;
+  char* const lB = smem + BM2 * ROWB2;
+
+  const unsigned nwg = tiles_m * tiles_n;
+  const unsigned bid = lpx_xcd_swizzle(blockIdx.x, nwg);
+  const int bm = (int)(bid / tiles_n) * BM2;
+  const int bn = (int)(bid % tiles_n) * BN2;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wr = wid >> 2, wc = wid & 3;  // wave tile 128x64
+  const int fr = lane & 31;
+  const int fkb = lane >> 5;
+
+  f32x16 acc[4][2];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int e = 0; e < 16; ++e) acc[i][j][e] = 0.f;
+
+  i32x4 aRegs[CHUNKS2], bRegs[CHUNKS2];
+  stage_imp(xq, g, M, K, bm, 0, aRegs);
+  load_fp8_regs(Wq, N, K, bn, 0, bRegs);
+  write_fp8_lds(aRegs, lA);
+  write_fp8_lds(bRegs, lB);
+  __syncthreads();
+
+  const int ksteps = (K + BK2 - 1) / BK2;
+  for (int ks = 0; ks < ksteps; ++ks) {
+    if (ks + 1 < ksteps) {
+      stage_imp(xq, g, M, K, bm, (ks + 1) * BK2, aRegs);
+      load_fp8_regs(Wq, N, K, bn, (ks + 1) * BK2, bRegs);
+    }
+    fp8frag32 aF[4], bF[2];
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf)
+      aF[mf] = frag32(lA, wr * 128 + mf * 32 + fr, fkb);
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf)
+      bF[nf] = frag32(lB, wc * 64 + nf * 32 + fr, fkb);
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+      for (int nf = 0; nf < 2; ++nf)
+        acc[mf][nf] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+            aF[mf], bF[nf], acc[mf][nf], 0, 0, 0, 127, 0, 127);
+    __syncthreads();
+    if (ks + 1 < ksteps) {
+      write_fp8_lds(aRegs, lA);
+      write_fp8_lds(bRegs, lB);
+    }
+    __syncthreads();
+  }
+
+  const int ocol_l = lane & 31;
+  const int orow_base = 4 * (lane >> 5);
+#pragma unroll
+  for (int mf = 0; mf < 4; ++mf) {
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int row = bm + wr * 128 + mf * 32 + (r & 3) + 8 * (r >> 2) +
+                        orow_base;
+        const int col = bn + wc * 64 + nf * 32 + ocol_l;
+        if (row < M && col < N) {
+          float v = acc[mf][nf][r] * descale;
+          if constexpr (BIAS) v += bias[col];
+          if constexpr (RELU) v = v > 0.f ? v : 0.f;
+          C[(long)row * N + col] = v;
+        }
+      }
+    }
+  }
+}
+
 }  // namespace
 
 hipError_t conv_fp8_implicit(const unsigned char* xq, const unsigned char* Wq,
@@ -493,16 +594,16 @@ hipError_t conv_fp8_implicit(const unsigned char* xq, const unsigned char* Wq,
   const size_t lds = (size_t)2 * BM2 * ROWB2;
   dim3 grid(tiles_m * tiles_n), block(THREADS2);
   if (bias && relu)
-    conv_fp8_imp_kernel<true, true><<<grid, block, lds, stream>>>(
+    conv_fp8_impx_kernel<true, true><<<grid, block, lds, stream>>>(
         xq, Wq, C, bias, g, M, KO, K, descale, tiles_m, tiles_n);
   else if (bias)
-    conv_fp8_imp_kernel<false, true><<<grid, block, lds, stream>>>(
+    conv_fp8_impx_kernel<false, true><<<grid, block, lds, stream>>>(
         xq, Wq, C, bias, g, M, KO, K, descale, tiles_m, tiles_n);
   else if (relu)
-    conv_fp8_imp_kernel<true, false><<<grid, block, lds, stream>>>(
+    conv_fp8_impx_kernel<true, false><<<grid, block, lds, stream>>>(
         xq, Wq, C, bias, g, M, KO, K, descale, tiles_m, tiles_n);
   else
-    conv_fp8_imp_kernel<false, false><<<grid, block, lds, stream>>>(
+    conv_fp8_impx_kernel<false, false><<<grid, block, lds, stream>>>(
         xq, Wq, C, bias, g, M, KO, K, descale, tiles_m, tiles_n);
   return hipGetLastError();
 }

@@ -78,6 +78,12 @@ hipError_t rf_predict_heap_f32(const int* nodes_heap, const float* leaf_values,
                                int n_trees, int depth, const float* X,
                                float* out, int S, int F, int C, int silp,
                                hipStream_t stream);
+// v5: two-level heap — top `top` levels served from LDS (top<0: auto-pick
+// largest fitting 2-blocks/CU; falls back to v4 when nothing fits).
+hipError_t rf_predict_heap2_f32(const int* nodes_heap, const float* leaf_values,
+                                int n_trees, int depth, const float* X,
+                                float* out, int S, int F, int C, int silp,
+                                int top, hipStream_t stream);
 // Forest-in-LDS chunked traversal (v3): whole-tree chunks staged in LDS
 // (host-computed chunk_tree_begin[n_chunks+1]); out accumulates across
 // chunks. One block per CU.

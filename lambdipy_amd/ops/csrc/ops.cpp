@@ -394,7 +394,7 @@ at::Tensor gemm_bf16_tune8(const at::Tensor& A, const at::Tensor& Bt, int64_t tu
 
 at::Tensor rf_predict_heap(const at::Tensor& nodes_heap,
                            const at::Tensor& leaf_values, int64_t depth,
-                           const at::Tensor& X, int64_t silp) {
+                           const at::Tensor& X, int64_t silp, int64_t top) {
   CHECK_DEV(nodes_heap);
   CHECK_DEV(leaf_values);
   CHECK_DEV(X);
@@ -410,12 +410,20 @@ at::Tensor rf_predict_heap(const at::Tensor& nodes_heap,
               "leaf_values rows != n_trees * 2^depth");
   c10::hip::HIPGuardMasqueradingAsCUDA guard(X.device());
   auto out = at::empty({S, C}, X.options());
-  check_hip(lpx::rf_predict_heap_f32(nodes_heap.data_ptr<int>(),
-                                     leaf_values.data_ptr<float>(), (int)n_trees,
-                                     (int)depth, X.data_ptr<float>(),
-                                     out.data_ptr<float>(), (int)S, (int)F,
-                                     (int)C, (int)silp, current_stream(X)),
-            "rf_predict_heap_f32");
+  if (top == -2)
+    check_hip(lpx::rf_predict_heap_f32(
+                  nodes_heap.data_ptr<int>(), leaf_values.data_ptr<float>(),
+                  (int)n_trees, (int)depth, X.data_ptr<float>(),
+                  out.data_ptr<float>(), (int)S, (int)F, (int)C, (int)silp,
+                  current_stream(X)),
+              "rf_predict_heap_f32");
+  else
+    check_hip(lpx::rf_predict_heap2_f32(
+                  nodes_heap.data_ptr<int>(), leaf_values.data_ptr<float>(),
+                  (int)n_trees, (int)depth, X.data_ptr<float>(),
+                  out.data_ptr<float>(), (int)S, (int)F, (int)C, (int)silp,
+                  (int)top, current_stream(X)),
+              "rf_predict_heap2_f32");
   return out;
 }
 
@@ -474,7 +482,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rf_predict_heap", &rf_predict_heap,
         "RandomForest per-class sums (heap layout, complete forests)",
         py::arg("nodes_heap"), py::arg("leaf_values"), py::arg("depth"),
-        py::arg("X"), py::arg("silp") = 1);
+        py::arg("X"), py::arg("silp") = 1, py::arg("top") = -2);
   m.def("rf_predict8_lds", &rf_predict8_lds,
         "RandomForest per-class sums (forest chunks staged in LDS)",
         py::arg("nodes8"), py::arg("leaf_values"), py::arg("roots"),

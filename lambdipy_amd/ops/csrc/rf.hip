@@ -19,6 +19,7 @@
 //     similar-depth CART trees.
 
 #include "lpx_common.h"
+#include "kernels.h"
 
 namespace lpx {
 
@@ -353,7 +354,150 @@ __global__ __launch_bounds__(THREADS) void rf_heap_kernel(
   }
 }
 
+// v5: two-level heap traversal — the top `top` levels of every tree are
+// cached in LDS (loaded once per block), the remaining levels read from
+// L2/HBM. The kernel is L2-request-bound (243 Ms/s = ~5 TB/s of 8-B node
+// fetches at the bench shape), so serving the hottest, perfectly-shared
+// top nodes from LDS removes top/depth of that traffic.
+template <bool STAGE_X, int SILP>
+__global__ __launch_bounds__(THREADS) void rf_heap2_kernel(
+    const i32x2* __restrict__ nodes, const float* __restrict__ leaf_values,
+    int n_trees, int depth, int top, const float* __restrict__ X,
+    float* __restrict__ out, int S, int F, int C) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* xl = reinterpret_cast<float*>(smem);
+  const int rowf = F + 1;
+  const size_t xbytes = STAGE_X ? (size_t)THREADS * SILP * rowf * 4 : 0;
+  i32x2* ltop = reinterpret_cast<i32x2*>(smem + xbytes);
+  const int t = threadIdx.x;
+  const int tree_stride = (1 << depth) - 1;
+  const int top_stride = (1 << top) - 1;
+  const int n_leaves = 1 << depth;
+
+  {  // cooperative top-level load (once per block; heap prefix per tree)
+    const long total = (long)n_trees * top_stride;
+    for (long i = t; i < total; i += THREADS) {
+      const int tr = (int)(i / top_stride), nd = (int)(i % top_stride);
+      ltop[i] = nodes[(long)tr * tree_stride + nd];
+    }
+    __syncthreads();
+  }
+
+  for (int tile = blockIdx.x * (THREADS * SILP); tile < S;
+       tile += gridDim.x * (THREADS * SILP)) {
+    const int tile_n = min(THREADS * SILP, S - tile);
+    if (STAGE_X) {
+      const long total = (long)tile_n * F;
+      for (long i = t; i < total; i += THREADS) {
+        const int r = (int)(i / F), c = (int)(i - (long)r * F);
+        xl[r * rowf + c] = X[((long)tile + r) * F + c];
+      }
+      __syncthreads();
+    }
+
+    int s[SILP];
+    const float* xrow[SILP];
+#pragma unroll
+    for (int j = 0; j < SILP; ++j) {
+      s[j] = tile + t + j * THREADS;
+      xrow[j] = STAGE_X ? (xl + (t + j * THREADS) * rowf)
+                        : (X + (long)s[j] * F);
+    }
+    float acc[SILP][CMAX];
+#pragma unroll
+    for (int j = 0; j < SILP; ++j)
+#pragma unroll
+      for (int c = 0; c < CMAX; ++c) acc[j][c] = 0.f;
+
+    for (int tr = 0; tr < n_trees; ++tr) {
+      const i32x2* ttop = ltop + (long)tr * top_stride;
+      const i32x2* tb = nodes + (long)tr * tree_stride;
+      int nd[SILP];
+#pragma unroll
+      for (int j = 0; j < SILP; ++j) nd[j] = 0;
+      for (int d = 0; d < top; ++d) {
+#pragma unroll
+        for (int j = 0; j < SILP; ++j) {
+          if (s[j] < S) {
+            const i32x2 n = ttop[nd[j]];
+            const float xv = xrow[j][(unsigned)n.x];
+            nd[j] = 2 * nd[j] + 1 + (xv <= __int_as_float(n.y) ? 0 : 1);
+          }
+        }
+      }
+      for (int d = top; d < depth; ++d) {
+#pragma unroll
+        for (int j = 0; j < SILP; ++j) {
+          if (s[j] < S) {
+            const i32x2 n = tb[nd[j]];
+            const float xv = xrow[j][(unsigned)n.x];
+            nd[j] = 2 * nd[j] + 1 + (xv <= __int_as_float(n.y) ? 0 : 1);
+          }
+        }
+      }
+#pragma unroll
+      for (int j = 0; j < SILP; ++j) {
+        if (s[j] < S) {
+          const int leaf = nd[j] - tree_stride;
+          const float* lv = leaf_values + ((long)tr * n_leaves + leaf) * C;
+#pragma unroll
+          for (int c = 0; c < CMAX; ++c)
+            if (c < C) acc[j][c] += lv[c];
+        }
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < SILP; ++j)
+      if (s[j] < S)
+#pragma unroll
+        for (int c = 0; c < CMAX; ++c)
+          if (c < C) out[(long)s[j] * C + c] = acc[j][c];
+    if (STAGE_X) __syncthreads();
+  }
+}
+
 }  // namespace
+
+hipError_t rf_predict_heap2_f32(const int* nodes_heap, const float* leaf_values,
+                                int n_trees, int depth, const float* X,
+                                float* out, int S, int F, int C, int silp,
+                                int top, hipStream_t stream) {
+  if (C > CMAX || depth < 1 || depth > 26) return hipErrorInvalidValue;
+  const int SILP = (silp == 2) ? 2 : (silp == 4 ? 4 : 1);
+  const bool stage = F <= F_LDS_MAX;
+  const size_t xbytes = stage ? (size_t)THREADS * SILP * (F + 1) * 4 : 0;
+  if (top < 0) {
+    // largest top with total LDS <= 78 KiB (2 blocks/CU)
+    top = 0;
+    while (top < depth &&
+           xbytes + (size_t)n_trees * ((2u << top) - 1) * 8 <= 78 * 1024)
+      ++top;
+  }
+  if (top > depth) top = depth;
+  const size_t lds = xbytes + (size_t)n_trees * ((1u << top) - 1) * 8;
+  if (lds > 160 * 1024 || top < 1)
+    return rf_predict_heap_f32(nodes_heap, leaf_values, n_trees, depth, X, out,
+                               S, F, C, silp, stream);
+  unsigned blocks = lpx_ceil_div(S, THREADS * SILP);
+  if (blocks > 2048u) blocks = 2048u;
+  if (blocks == 0) blocks = 1;
+  const auto* nh = reinterpret_cast<const i32x2*>(nodes_heap);
+#define LPX_RFH2_CASE(I)                                                       \
+  if (SILP == I) {                                                             \
+    if (stage)                                                                 \
+      rf_heap2_kernel<true, I><<<dim3(blocks), dim3(THREADS), lds, stream>>>(  \
+          nh, leaf_values, n_trees, depth, top, X, out, S, F, C);              \
+    else                                                                       \
+      rf_heap2_kernel<false, I><<<dim3(blocks), dim3(THREADS), lds, stream>>>( \
+          nh, leaf_values, n_trees, depth, top, X, out, S, F, C);              \
+    return hipGetLastError();                                                  \
+  }
+  LPX_RFH2_CASE(1)
+  LPX_RFH2_CASE(2)
+  LPX_RFH2_CASE(4)
+#undef LPX_RFH2_CASE
+  return hipErrorInvalidValue;
+}
 
 hipError_t rf_predict_heap_f32(const int* nodes_heap, const float* leaf_values,
                                int n_trees, int depth, const float* X,

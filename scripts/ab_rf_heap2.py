@@ -1,0 +1,44 @@
+"""A/B: two-level heap RF kernel (top levels in LDS) vs v4, silp x top sweep."""
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, "/root/repo")
+from lambdipy_amd import ops  # noqa: E402
+from lambdipy_amd.shims.lpx_sklearn import ForestSoA  # noqa: E402
+
+e = ops.ext()
+
+
+def bench(fn, iters=10, warmup=3):
+    for _ in range(warmup):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / iters
+
+
+def main():
+    f = ForestSoA.random(256, 10, 32, 8, seed=17, device="cuda")
+    assert f.layout == "heap", f.layout
+    X = torch.randn(1 << 20, 32, device="cuda")
+    S = X.shape[0]
+    n, lv = f.nodes, f.leaf_values
+    ref = e.rf_predict_heap(n, lv, 10, X, 1, -2)
+    for silp in (1, 2, 4):
+        for top in (-2, -1, 2, 3, 4, 5):
+            try:
+                out = e.rf_predict_heap(n, lv, 10, X, silp, top)
+                ok = torch.allclose(out, ref, atol=1e-4)
+                dt = bench(lambda: e.rf_predict_heap(n, lv, 10, X, silp, top))
+                print(f"silp={silp} top={top:2d}: {S/dt/1e6:7.1f} Ms/s  ok={ok}")
+            except RuntimeError as ex:
+                print(f"silp={silp} top={top:2d}: ERR {ex}")
+
+
+if __name__ == "__main__":
+    main()
