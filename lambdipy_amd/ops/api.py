@@ -151,8 +151,11 @@ def rf_predict_proba_sum(nodes: torch.Tensor, leaf_values: torch.Tensor,
         assert depth is not None
 
         def fn(n, lv, r, x):
-            # top=-1: two-level kernel, auto-picked LDS-resident top levels
-            return e.rf_predict_heap(n, lv, depth, x, 1, -1)
+            # top=-2: single-level v4 kernel. The two-level variant (top>=0)
+            # measured SLOWER (146 vs 243 Ms/s at the bench shape): the top
+            # nodes are already L1-hot and the extra LDS halves occupancy,
+            # which the latency-bound bottom levels need. Kept for A/B.
+            return e.rf_predict_heap(n, lv, depth, x, 1, -2)
     elif packed and chunks is not None:
         ctb, max_nodes = chunks
 

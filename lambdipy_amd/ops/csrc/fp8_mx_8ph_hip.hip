@@ -29,10 +29,10 @@ namespace {
 
 constexpr int BMX = 256, BKX = 128;  // 128 fp8 elements = 128 B per row
 constexpr int THREADSX = 512;
-constexpr int HALF_DATA = 128 * 128;               // bytes per half-tile
-constexpr int HALF_SC = 512;                       // scale strip per half
-constexpr int HALF_BYTES = HALF_DATA + HALF_SC;
+constexpr int HALF_BYTES = 128 * 128;              // data bytes per half-tile
 constexpr int PARITY_BYTES = 4 * HALF_BYTES;
+constexpr int SC_TILE = 2048;  // per-tile scale block [A 256 rows | B 256]x4B
+constexpr int SC_BASE = 2 * PARITY_BYTES;
 
 typedef __attribute__((ext_vector_type(8))) int i32x8v;
 
@@ -52,23 +52,20 @@ __device__ __forceinline__ void waitvm() {
   else if constexpr (N == 4) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
   else if constexpr (N == 6) asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
   else if constexpr (N == 8) asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+  else if constexpr (N == 9) asm volatile("s_waitcnt vmcnt(9)" ::: "memory");
   else if constexpr (N == 12) asm volatile("s_waitcnt vmcnt(12)" ::: "memory");
   else static_assert(N == 0, "unsupported vmcnt");
 }
 
-// Stage one half-tile (128 rows x 128 B data + 512-B scale strip) for
-// `tile` with 3 glds per thread (2 when SC=false: no scale dword).
+// Stage one half-tile (128 rows x 128 B data, 2 glds per thread).
 // hh: 0/1 = A halves, 2/3 = B halves.
-template <bool SC>
 __device__ __forceinline__ void stage_halfx(
-    const unsigned char* __restrict__ A, const unsigned char* __restrict__ sA,
-    const unsigned char* __restrict__ Bt, const unsigned char* __restrict__ sB,
+    const unsigned char* __restrict__ A, const unsigned char* __restrict__ Bt,
     long K, int bm, int bn, int tile, int hh, int ksteps,
     char* __restrict__ smem, int w, int lane) {
   const int tc = tile < ksteps ? tile : ksteps - 1;
   const long k0 = (long)tc * BKX;
   const unsigned char* src = (hh < 2) ? A : Bt;
-  const unsigned char* ssrc = (hh < 2) ? sA : sB;
   const int row0 = ((hh < 2) ? bm : bn) + (hh & 1) * 128;
   char* const base = smem + (size_t)(tile & 1) * PARITY_BYTES +
                      (size_t)hh * HALF_BYTES;
@@ -84,18 +81,26 @@ __device__ __forceinline__ void stage_halfx(
         (const __attribute__((address_space(1))) void*)g,
         (__attribute__((address_space(3))) void*)(base + qbase * 16), 16, 0, 0);
   }
-  if constexpr (SC) {
-    // scales: one dword per row ([2 groups][2 blocks] for this k-tile);
-    // wave pairs duplicate the 128 dwords (same source, benign).
-    const int di = (w & 1) * 64 + lane;                     // dword 0..127
-    const long scstride = (K >> 6) * 2;                     // bytes per row
-    const unsigned char* g = ssrc + (long)(row0 + di) * scstride + (long)tc * 4;
-    __builtin_amdgcn_global_load_lds(
-        (const __attribute__((address_space(1))) void*)g,
-        (__attribute__((address_space(3))) void*)(base + HALF_DATA +
-                                                  (w & 1) * 256),
-        4, 0, 0);
-  }
+}
+
+// Stage the WHOLE tile's scales (2 KB: [A rows 0..255 | B rows 0..255] x
+// one dword per row = this k-tile's [2 groups][2 blocks]) with exactly ONE
+// 4-B glds per thread — no duplication. Issued only in phase 0, so the
+// 4-call vmcnt window always contains one scale load (constant counts).
+__device__ __forceinline__ void stage_scalesx(
+    const unsigned char* __restrict__ sA, const unsigned char* __restrict__ sB,
+    long K, int bm, int bn, int tile, int ksteps, char* __restrict__ smem,
+    int w, int lane) {
+  const int tc = tile < ksteps ? tile : ksteps - 1;
+  const int t = w * 64 + lane;
+  const long scstride = (K >> 6) * 2;  // scale bytes per row
+  const unsigned char* g =
+      (t < 256) ? sA + (long)(bm + t) * scstride + (long)tc * 4
+                : sB + (long)(bn + (t - 256)) * scstride + (long)tc * 4;
+  char* const base = smem + SC_BASE + (size_t)(tile & 1) * SC_TILE + w * 256;
+  __builtin_amdgcn_global_load_lds(
+      (const __attribute__((address_space(1))) void*)g,
+      (__attribute__((address_space(3))) void*)base, 4, 0, 0);
 }
 
 __device__ __forceinline__ void stage_slotx(int t, int p, int* tile, int* hh) {
@@ -124,10 +129,9 @@ __device__ __forceinline__ i32x8v fragx(const char* __restrict__ smem,
 __device__ __forceinline__ int scalex(const char* __restrict__ smem,
                                       int parity, int hh, int row_in_half,
                                       int kg, int fkb) {
-  const char* half = smem + (size_t)parity * PARITY_BYTES +
-                     (size_t)hh * HALF_BYTES;
+  const int row = (hh < 2 ? 0 : 1024) + ((hh & 1) * 128 + row_in_half) * 4;
   return (int)*reinterpret_cast<const unsigned char*>(
-      half + HALF_DATA + row_in_half * 4 + kg * 2 + fkb);
+      smem + SC_BASE + (size_t)parity * SC_TILE + row + kg * 2 + fkb);
 }
 
 // SC=true: per-block MX scales from LDS (sA/sB used; descale/bias ignored).
@@ -163,13 +167,15 @@ __global__ __launch_bounds__(THREADSX, 1) void gemm_mx_8ph_kernel(
 
   const int ksteps = (K + BKX - 1) / BKX;
 
-  stage_halfx<SC>(A, sA, Bt, sB, K, bm, bn, 0, 0, ksteps, smem, w, lane);
-  stage_halfx<SC>(A, sA, Bt, sB, K, bm, bn, 0, 2, ksteps, smem, w, lane);
-  stage_halfx<SC>(A, sA, Bt, sB, K, bm, bn, 0, 3, ksteps, smem, w, lane);
-  stage_halfx<SC>(A, sA, Bt, sB, K, bm, bn, 0, 1, ksteps, smem, w, lane);
-  stage_halfx<SC>(A, sA, Bt, sB, K, bm, bn, 1, 0, ksteps, smem, w, lane);
-  stage_halfx<SC>(A, sA, Bt, sB, K, bm, bn, 1, 2, ksteps, smem, w, lane);
-  waitvm<SC ? 6 : 4>();
+  stage_halfx(A, Bt, K, bm, bn, 0, 0, ksteps, smem, w, lane);
+  stage_halfx(A, Bt, K, bm, bn, 0, 2, ksteps, smem, w, lane);
+  stage_halfx(A, Bt, K, bm, bn, 0, 3, ksteps, smem, w, lane);
+  stage_halfx(A, Bt, K, bm, bn, 0, 1, ksteps, smem, w, lane);
+  if constexpr (SC)
+    stage_scalesx(sA, sB, K, bm, bn, 0, ksteps, smem, w, lane);
+  stage_halfx(A, Bt, K, bm, bn, 1, 0, ksteps, smem, w, lane);
+  stage_halfx(A, Bt, K, bm, bn, 1, 2, ksteps, smem, w, lane);
+  waitvm<4>();
   __builtin_amdgcn_s_barrier();
 
   if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
@@ -197,7 +203,10 @@ __global__ __launch_bounds__(THREADSX, 1) void gemm_mx_8ph_kernel(
       }
       int st, sh;
       stage_slotx(t, p, &st, &sh);
-      stage_halfx<SC>(A, sA, Bt, sB, K, bm, bn, st, sh, ksteps, smem, w, lane);
+      stage_halfx(A, Bt, K, bm, bn, st, sh, ksteps, smem, w, lane);
+      if constexpr (SC)
+        if (p == 0)  // tile t+1's full scale block, 1 glds/thread, no dup
+          stage_scalesx(sA, sB, K, bm, bn, t + 1, ksteps, smem, w, lane);
 
       __builtin_amdgcn_s_barrier();
 #pragma unroll
@@ -206,7 +215,16 @@ __global__ __launch_bounds__(THREADSX, 1) void gemm_mx_8ph_kernel(
         for (int kg = 0; kg < 2; ++kg)
           acc[p][mf] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
               aF[mf][kg], bF[kg], acc[p][mf], 0, 0, 0, sa[mf][kg], 0, sb[kg]);
-      waitvm<SC ? 12 : 8>();
+      // SC: per-tile-cycle call events are [3,2,2,2]; any 4-call window
+      // holds one phase-0, so 9 is constant — except after p3, where the
+      // next phase reads tile t+1's scales (issued at p0, 4 calls back):
+      // tighten to 6 so that load is complete.
+      if constexpr (SC) {
+        if (p == 3) waitvm<6>();
+        else waitvm<9>();
+      } else {
+        waitvm<8>();
+      }
       __builtin_amdgcn_s_barrier();
     }
   }
@@ -245,7 +263,7 @@ hipError_t gemm_mx_fp8_nt8ph(const unsigned char* A, const unsigned char* sA,
                              float* C, int M, int N, int K,
                              hipStream_t stream) {
   const unsigned tiles_m = (unsigned)(M / BMX), tiles_n = (unsigned)(N / BMX);
-  const size_t lds = 2 * PARITY_BYTES;  // 132 KiB
+  const size_t lds = 2 * PARITY_BYTES + 2 * SC_TILE;  // 132 KiB
  hipLaunchKernelGGL(( gemm_mx_8ph_kernel<true, false, false>)
       , dim3(dim3(tiles_m * tiles_n)), dim3(dim3(THREADSX)), lds, stream, 
           A, sA, Bt, sB, C, nullptr, 1.f, M, N, K, tiles_m, tiles_n);
@@ -257,7 +275,7 @@ hipError_t gemm_fp8_nt8phx(const unsigned char* A, const unsigned char* Bt,
                            float* C, const float* bias, int M, int N, int K,
                            float descale, bool relu, hipStream_t stream) {
   const unsigned tiles_m = (unsigned)(M / BMX), tiles_n = (unsigned)(N / BMX);
-  const size_t lds = 2 * PARITY_BYTES;
+  const size_t lds = 2 * PARITY_BYTES;  // 128 KiB, no scale region
   dim3 grid(tiles_m * tiles_n), block(THREADSX);
   if (bias && relu)
    hipLaunchKernelGGL(( gemm_mx_8ph_kernel<false, true, true>), dim3(grid), dim3(block), lds, stream, 

@@ -26,6 +26,13 @@ Mechanics (empirically verified against ROCm 7.2 libtorch_hip.so):
     obtained by dropping PT_NOTE (not used at runtime).
 
 The result dlopens and runs kernels exactly as before on gfx950 machines.
+
+When the DSO has no PT_NOTE to reuse (e.g. libmagma.so), the section is
+rewritten in place and its tail zero-filled: the file size is unchanged but
+the zeros cost nothing after zip/zstd in the layer artifact. Measured on
+the ROCm 7.2 torch dist (torch/lib): 5.43 -> 3.73 GB on disk
+(libtorch_hip 423->200 MB, rocsolver 760->90, rocsparse 392->80,
+rocrand 354->42, MIOpen 954->797, rocblas 55->30; magma zero-filled).
 """
 
 from __future__ import annotations
