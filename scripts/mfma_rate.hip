@@ -27,6 +27,44 @@ __global__ __launch_bounds__(512) void rate_scaled(float* out, int iters) {
   if (s != 0.f) out[0] = s;  // never taken; keeps the loop alive
 }
 
+__global__ __launch_bounds__(512) void rate_scaled_fp4(float* out, int iters) {
+  i32x8v a = {0, 0, 0, 0, 0, 0, 0, 0}, b = a;
+  f32x16 acc[8];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int e = 0; e < 16; ++e) acc[i][e] = 0.f;
+  for (int it = 0; it < iters; ++it) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+      acc[i] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+          a, b, acc[i], 4, 4, 0, 127, 0, 127);  // cbsz/blgp 4 = e2m1 fp4
+  }
+  float s = 0.f;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) s += acc[i][0];
+  if (s != 0.f) out[0] = s;
+}
+
+typedef __attribute__((ext_vector_type(4))) float f32x4v;
+
+__global__ __launch_bounds__(512) void rate_scaled_16x128(float* out, int iters) {
+  i32x8v a = {0, 0, 0, 0, 0, 0, 0, 0}, b = a;
+  f32x4v acc[8];
+#pragma unroll
+  for (int i = 0; i < 8; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+  for (int it = 0; it < iters; ++it) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+      acc[i] = __builtin_amdgcn_mfma_scale_f32_16x16x128_f8f6f4(
+          a, b, acc[i], 0, 0, 0, 127, 0, 127);
+  }
+  float s = 0.f;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) s += acc[i][0];
+  if (s != 0.f) out[0] = s;
+}
+
 __global__ __launch_bounds__(512) void rate_plain(float* out, int iters) {
   long a = 0, b = 0;
   f32x4 acc[8];
@@ -49,17 +87,16 @@ extern "C" int run_rate(int which, int blocks, int iters, float* ms_out) {
   hipEvent_t e0, e1;
   hipEventCreate(&e0);
   hipEventCreate(&e1);
-  // warmup
-  if (which == 0)
-    rate_scaled<<<blocks, 512>>>(d, iters / 10 + 1);
-  else
-    rate_plain<<<blocks, 512>>>(d, iters / 10 + 1);
+  auto launch = [&](int n) {
+    if (which == 0) rate_scaled<<<blocks, 512>>>(d, n);
+    else if (which == 1) rate_plain<<<blocks, 512>>>(d, n);
+    else if (which == 2) rate_scaled_fp4<<<blocks, 512>>>(d, n);
+    else rate_scaled_16x128<<<blocks, 512>>>(d, n);
+  };
+  launch(iters / 10 + 1);
   hipDeviceSynchronize();
   hipEventRecord(e0);
-  if (which == 0)
-    rate_scaled<<<blocks, 512>>>(d, iters);
-  else
-    rate_plain<<<blocks, 512>>>(d, iters);
+  launch(iters);
   hipEventRecord(e1);
   if (hipDeviceSynchronize() != hipSuccess) return 2;
   hipEventElapsedTime(ms_out, e0, e1);
