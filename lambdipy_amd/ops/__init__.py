@@ -17,11 +17,14 @@ from .api import (
     fft,
     gemm_bf16,
     gemm_mx,
+    gemm_mx_fp4,
     gemv,
     ifft,
     matmul_bf16,
     matmul_mx,
+    matmul_mx_fp4,
     quantize_mx,
+    quantize_mx_fp4,
     pairwise_sqeuclidean,
     require_extension,
     rf_predict_proba_sum,
@@ -38,6 +41,9 @@ __all__ = [
     "quantize_mx",
     "gemm_mx",
     "matmul_mx",
+    "quantize_mx_fp4",
+    "gemm_mx_fp4",
+    "matmul_mx_fp4",
     "gemv",
     "spmv_csr",
     "fft",

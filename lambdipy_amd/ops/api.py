@@ -90,6 +90,26 @@ def matmul_mx(A: torch.Tensor, B: torch.Tensor) -> torch.Tensor:
     return gemm_mx(qa, sa, qb, sb)
 
 
+def quantize_mx_fp4(x: torch.Tensor):
+    """f32 [R, K%64==0] -> (e2m1 nibbles [R, K/2] in MFMA-shuffled order,
+    E8M0 scales [R, K/64, 2]).  The fp4 MFMA path runs at 2x the fp8 rate."""
+    q, sc = require_extension().quantize_mx4(x.contiguous())
+    return q, sc
+
+
+def gemm_mx_fp4(A: torch.Tensor, sA: torch.Tensor, Bt: torch.Tensor,
+                sB: torch.Tensor) -> torch.Tensor:
+    """MX block-scaled fp4 GEMM (M,N%256, K%128); f32 out, hw scales."""
+    return require_extension().gemm_mx4(A, sA, Bt, sB)
+
+
+def matmul_mx_fp4(A: torch.Tensor, B: torch.Tensor) -> torch.Tensor:
+    """Convenience: quantize both f32 operands to MX fp4 and multiply."""
+    qa, sa = quantize_mx_fp4(A)
+    qb, sb = quantize_mx_fp4(B.t().contiguous())
+    return gemm_mx_fp4(qa, sa, qb, sb)
+
+
 # ---- sparse --------------------------------------------------------------
 
 def spmv_csr(indptr: torch.Tensor, indices: torch.Tensor, data: torch.Tensor,

@@ -1,0 +1,320 @@
+// MX block-scaled fp4 (OCP e2m1) GEMM for gfx950 — the full-rate path.
+//
+// v_mfma_scale_f32_32x32x64_f8f6f4 with cbsz/blgp=4 (e2m1) issues at 2x the
+// fp8 rate (9.67 vs 4.89 PF register-resident, scripts/mfma_rate.py; dense
+// fp4 silicon peak is ~10 PF). Operands shrink to 16 B/lane (4 VGPRs — the
+// disassembly shows v[16:19] operands), halving LDS and HBM traffic too.
+//
+// Layout mirrors fp8_mx*.hip: lane pair {l, l+32} concatenates to row
+// (l&31)'s 32 k-bytes (64 e2m1 elements); lane l's scale byte applies to
+// byte-half (l>>5) of both lanes. Packing each 64-element group as
+// [q0|q2|q1|q3] (quarters = 16 elements = 8 bytes) makes the hardware scale
+// blocks the standard contiguous MX-32 blocks. Nibble order within a block
+// is irrelevant to the product (A and B share the permutation and the
+// block scale is constant), but quantizer and tests use low-nibble-first
+// (the v_cvt_scalef32_pk_fp4_f32 convention).
+//
+// 8-phase glds pipeline: half-tile = 128 rows x 64 B = one 16-B glds per
+// thread per stage call; per-tile-cycle call events [2,1,1,1] keep the
+// counted vmcnt constant at 5 (3 after p3, covering the next tile's
+// scales). LDS cell rotation is ((row>>1)+(row>>2))&3: rows 4, 8 and 12
+// apart land on the same 16-bank base, and this rotation separates all of
+// them (plain (row>>1)&3 collides at +8). 68 KiB LDS -> 2 blocks/CU.
+
+#include "lpx_common.h"
+
+namespace lpx {
+
+namespace {
+
+constexpr int THREADS = 256;
+
+typedef __attribute__((ext_vector_type(8))) int i32x8v;
+typedef __attribute__((ext_vector_type(2))) short s16x2;
+
+// ---- MX fp4 quantization -------------------------------------------------
+// x f32 [R][K], K % 64 == 0 -> q [R][K/2] shuffled nibbles + sc [R][K/64][2]
+__global__ __launch_bounds__(THREADS) void quantize_mx4_kernel(
+    const float* __restrict__ x, unsigned char* __restrict__ q,
+    unsigned char* __restrict__ sc, int R, int K) {
+  const int groups = K >> 6;
+  const long total = (long)R * groups * 2;  // thread per (row, group, block)
+  for (long idx = (long)blockIdx.x * THREADS + threadIdx.x; idx < total;
+       idx += (long)gridDim.x * THREADS) {
+    const int kb = (int)(idx & 1);
+    const int g = (int)((idx >> 1) % groups);
+    const int r = (int)(idx / (2 * groups));
+    const float* src = x + (long)r * K + g * 64 + kb * 32;
+    const f32x4* src4 = reinterpret_cast<const f32x4*>(src);
+    f32x4 buf[8];
+    float amax = 0.f;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      buf[i] = src4[i];
+#pragma unroll
+      for (int j = 0; j < 4; ++j) amax = fmaxf(amax, fabsf(buf[i][j]));
+    }
+    // E8M0 scale, non-clipping: amax/2^e <= 6 (e2m1 max)
+    int e = 0;
+    if (amax > 0.f) {
+      int ex;
+      (void)frexpf(amax, &ex);       // amax = m * 2^ex, m in [0.5, 1)
+      e = ex - 3;                    // amax / 2^e in [4, 8)
+      if (amax * exp2f((float)-e) > 6.f) ++e;  // -> (3, 6], no saturation
+    }
+    if (e < -127) e = -127;
+    if (e > 127) e = 127;
+    sc[((long)r * groups + g) * 2 + kb] = (unsigned char)(e + 127);
+    const float scl = exp2f((float)e);  // hw cvt divides by scale
+    unsigned char* base = q + (long)r * (K >> 1) + g * 32;
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      const int qq = 2 * kb + h;                    // original quarter
+      const int pos = ((qq & 1) << 1) | (qq >> 1);  // shuffled slot
+      // quarter = 16 elements -> 8 bytes = 2 u32 (4 cvt calls each; the
+      // byte-select operand must be a literal)
+      unsigned int w[2] = {0, 0};
+#pragma unroll
+      for (int u = 0; u < 2; ++u) {
+        const f32x4 v0 = buf[h * 4 + u * 2];
+        const f32x4 v1 = buf[h * 4 + u * 2 + 1];
+        unsigned int acc = 0;
+        acc = __builtin_amdgcn_cvt_scalef32_pk_fp4_f32(acc, v0[0], v0[1], scl, 0);
+        acc = __builtin_amdgcn_cvt_scalef32_pk_fp4_f32(acc, v0[2], v0[3], scl, 1);
+        acc = __builtin_amdgcn_cvt_scalef32_pk_fp4_f32(acc, v1[0], v1[1], scl, 2);
+        acc = __builtin_amdgcn_cvt_scalef32_pk_fp4_f32(acc, v1[2], v1[3], scl, 3);
+        w[u] = acc;
+      }
+      unsigned int* dst =
+          reinterpret_cast<unsigned int*>(base + pos * 8);
+      dst[0] = w[0];
+      dst[1] = w[1];
+    }
+  }
+}
+
+// ---- 8-phase MX fp4 GEMM -------------------------------------------------
+
+constexpr int BM4 = 256, BK4 = 128;  // 128 elements = 64 B per row
+constexpr int THREADS4 = 512;
+constexpr int HALF4 = 128 * 64;      // bytes per half-tile
+constexpr int PARITY4 = 4 * HALF4;   // 32 KiB
+constexpr int SC_TILE4 = 2048;
+constexpr int SC_BASE4 = 2 * PARITY4;
+
+__device__ __forceinline__ int rot4(int row) {
+  return ((row >> 1) + (row >> 2)) & 3;
+}
+__device__ __forceinline__ int phys_cell4(int row, int cell) {
+  return row * 4 + ((cell + rot4(row)) & 3);
+}
+__device__ __forceinline__ int src_cell4(int q) {
+  const int row = q >> 2;
+  return ((q & 3) - rot4(row)) & 3;
+}
+
+template <int N>
+__device__ __forceinline__ void waitvm4() {
+  if constexpr (N == 0) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  else if constexpr (N == 2) asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+  else if constexpr (N == 3) asm volatile("s_waitcnt vmcnt(3)" ::: "memory");
+  else if constexpr (N == 5) asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
+  else static_assert(N == 0, "unsupported vmcnt");
+}
+
+// one 16-B glds per thread: 512 cells = 128 rows x 4
+__device__ __forceinline__ void stage_half4(
+    const unsigned char* __restrict__ A, const unsigned char* __restrict__ Bt,
+    long Kb, int bm, int bn, int tile, int hh, int ksteps,
+    char* __restrict__ smem, int t) {
+  const int tc = tile < ksteps ? tile : ksteps - 1;
+  const long k0 = (long)tc * 64;  // bytes per BK4 tile
+  const unsigned char* src = (hh < 2) ? A : Bt;
+  const int row0 = ((hh < 2) ? bm : bn) + (hh & 1) * 128;
+  char* const base = smem + (size_t)(tile & 1) * PARITY4 +
+                     (size_t)hh * HALF4;
+  const int r = t >> 2, cell = src_cell4(t);
+  const unsigned char* g = src + (long)(row0 + r) * Kb + (k0 + cell * 16);
+  // LDS dest must be WAVE-UNIFORM (hw adds lane*16): wave w owns cells
+  // [w*64, +64) -> base + w*1024.
+  __builtin_amdgcn_global_load_lds(
+      (const __attribute__((address_space(1))) void*)g,
+      (__attribute__((address_space(3))) void*)(base + (t >> 6) * 1024),
+      16, 0, 0);
+}
+
+__device__ __forceinline__ void stage_scales4(
+    const unsigned char* __restrict__ sA, const unsigned char* __restrict__ sB,
+    int groups, int bm, int bn, int tile, int ksteps, char* __restrict__ smem,
+    int w, int lane) {
+  const int tc = tile < ksteps ? tile : ksteps - 1;
+  const int t = w * 64 + lane;
+  const long scstride = (long)groups * 2;
+  const unsigned char* g =
+      (t < 256) ? sA + (long)(bm + t) * scstride + (long)tc * 4
+                : sB + (long)(bn + (t - 256)) * scstride + (long)tc * 4;
+  char* const base = smem + SC_BASE4 + (size_t)(tile & 1) * SC_TILE4 + w * 256;
+  __builtin_amdgcn_global_load_lds(
+      (const __attribute__((address_space(1))) void*)g,
+      (__attribute__((address_space(3))) void*)base, 4, 0, 0);
+}
+
+__device__ __forceinline__ void stage_slot4(int t, int p, int* tile, int* hh) {
+  switch (p) {
+    case 0: *tile = t + 1; *hh = 3; break;
+    case 1: *tile = t + 1; *hh = 1; break;
+    case 2: *tile = t + 2; *hh = 0; break;
+    default: *tile = t + 2; *hh = 2; break;
+  }
+}
+
+// 16-B fragment: one b128; regs [0..3] carry the data (fp4 operands use 4
+// VGPRs), [4..7] are ignored by the instruction.
+__device__ __forceinline__ i32x8v frag4(const char* __restrict__ smem,
+                                        int parity, int hh, int row_in_half,
+                                        int kg, int fkb) {
+  const char* half = smem + (size_t)parity * PARITY4 + (size_t)hh * HALF4;
+  const i32x4 v = *reinterpret_cast<const i32x4*>(
+      half + phys_cell4(row_in_half, kg * 2 + fkb) * 16);
+  return i32x8v{v.x, v.y, v.z, v.w, 0, 0, 0, 0};
+}
+
+__device__ __forceinline__ int scale4(const char* __restrict__ smem,
+                                      int parity, int hh, int row_in_half,
+                                      int kg, int fkb) {
+  const int row = (hh < 2 ? 0 : 1024) + ((hh & 1) * 128 + row_in_half) * 4;
+  return (int)*reinterpret_cast<const unsigned char*>(
+      smem + SC_BASE4 + (size_t)parity * SC_TILE4 + row + kg * 2 + fkb);
+}
+
+__global__ __launch_bounds__(THREADS4, 2) void gemm_mx4_8ph_kernel(
+    const unsigned char* __restrict__ A, const unsigned char* __restrict__ sA,
+    const unsigned char* __restrict__ Bt, const unsigned char* __restrict__ sB,
+    float* __restrict__ C, int M, int N, int K, unsigned tiles_m,
+    unsigned tiles_n) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+
+  const unsigned nwg = tiles_m * tiles_n;
+  const unsigned bid = lpx_xcd_swizzle(blockIdx.x, nwg);
+  const int bm = (int)(bid / tiles_n) * BM4;
+  const int bn = (int)(bid % tiles_n) * BM4;
+
+  const int lane = threadIdx.x & 63;
+  const int w = threadIdx.x >> 6;
+  const int wsr = w >> 2, wsc = w & 3;
+  const int fr = lane & 31;
+  const int fkb = lane >> 5;
+  const long Kb = K >> 1;  // bytes per row
+  const int groups = K >> 6;
+
+  f32x16 acc[4][2];
+#pragma unroll
+  for (int p = 0; p < 4; ++p)
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+      for (int e = 0; e < 16; ++e) acc[p][i][e] = 0.f;
+
+  const int ksteps = (K + BK4 - 1) / BK4;
+  const int t512 = threadIdx.x;
+
+  stage_half4(A, Bt, Kb, bm, bn, 0, 0, ksteps, smem, t512);
+  stage_half4(A, Bt, Kb, bm, bn, 0, 2, ksteps, smem, t512);
+  stage_half4(A, Bt, Kb, bm, bn, 0, 3, ksteps, smem, t512);
+  stage_half4(A, Bt, Kb, bm, bn, 0, 1, ksteps, smem, t512);
+  stage_scales4(sA, sB, groups, bm, bn, 0, ksteps, smem, w, lane);
+  stage_half4(A, Bt, Kb, bm, bn, 1, 0, ksteps, smem, t512);
+  stage_half4(A, Bt, Kb, bm, bn, 1, 2, ksteps, smem, t512);
+  waitvm4<2>();
+  __builtin_amdgcn_s_barrier();
+
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);  // T5 static young-half priority
+
+  for (int t = 0; t < ksteps; ++t) {
+    const int parity = t & 1;
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      const int rh = (p >> 1) & 1;
+      const int ch = p & 1;
+      i32x8v aF[2][2], bF[2];
+      int sa[2][2], sb[2];
+#pragma unroll
+      for (int kg = 0; kg < 2; ++kg) {
+#pragma unroll
+        for (int mf = 0; mf < 2; ++mf) {
+          const int r = wsr * 64 + mf * 32 + fr;
+          aF[mf][kg] = frag4(smem, parity, rh, r, kg, fkb);
+          sa[mf][kg] = scale4(smem, parity, rh, r, kg, fkb);
+        }
+        const int rb = wsc * 32 + fr;
+        bF[kg] = frag4(smem, parity, 2 + ch, rb, kg, fkb);
+        sb[kg] = scale4(smem, parity, 2 + ch, rb, kg, fkb);
+      }
+      int st, sh;
+      stage_slot4(t, p, &st, &sh);
+      stage_half4(A, Bt, Kb, bm, bn, st, sh, ksteps, smem, t512);
+      if (p == 0)
+        stage_scales4(sA, sB, groups, bm, bn, t + 1, ksteps, smem, w, lane);
+
+      __builtin_amdgcn_s_barrier();
+#pragma unroll
+      for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+        for (int kg = 0; kg < 2; ++kg)
+          acc[p][mf] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+              aF[mf][kg], bF[kg], acc[p][mf], 4, 4, 0, sa[mf][kg], 0, sb[kg]);
+      if (p == 3) waitvm4<3>();
+      else waitvm4<5>();
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+  waitvm4<0>();
+
+  const int ocol_l = lane & 31;
+  const int orow_base = 4 * (lane >> 5);
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    const int rh = (p >> 1) & 1, ch = p & 1;
+#pragma unroll
+    for (int mf = 0; mf < 2; ++mf) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int row = bm + rh * 128 + wsr * 64 + mf * 32 + (r & 3) +
+                        8 * (r >> 2) + orow_base;
+        const int col = bn + ch * 128 + wsc * 32 + ocol_l;
+        C[(long)row * N + col] = acc[p][mf][r];
+      }
+    }
+  }
+}
+
+}  // namespace
+
+hipError_t quantize_mx_fp4(const float* x, unsigned char* q, unsigned char* sc,
+                           int R, int K, hipStream_t stream) {
+  if (K % 64 != 0) return hipErrorInvalidValue;
+  const long total = (long)R * (K >> 6) * 2;
+  unsigned blocks = (unsigned)((total + THREADS - 1) / THREADS);
+  if (blocks > 4096u) blocks = 4096u;
+  quantize_mx4_kernel<<<dim3(blocks ? blocks : 1), dim3(THREADS), 0, stream>>>(
+      x, q, sc, R, K);
+  return hipGetLastError();
+}
+
+bool gemm_mx_fp4_supported(int M, int N, int K) {
+  return M % BM4 == 0 && N % BM4 == 0 && K % BK4 == 0 && K >= BK4;
+}
+
+hipError_t gemm_mx_fp4_nt8ph(const unsigned char* A, const unsigned char* sA,
+                             const unsigned char* Bt, const unsigned char* sB,
+                             float* C, int M, int N, int K,
+                             hipStream_t stream) {
+  const unsigned tiles_m = (unsigned)(M / BM4), tiles_n = (unsigned)(N / BM4);
+  const size_t lds = 2 * PARITY4 + 2 * SC_TILE4;  // 68 KiB -> 2 blocks/CU
+  gemm_mx4_8ph_kernel<<<dim3(tiles_m * tiles_n), dim3(THREADS4), lds, stream>>>(
+      A, sA, Bt, sB, C, M, N, K, tiles_m, tiles_n);
+  return hipGetLastError();
+}
+
+}  // namespace lpx

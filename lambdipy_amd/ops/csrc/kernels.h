@@ -144,4 +144,14 @@ hipError_t gemm_fp8_nt8phx(const unsigned char* A, const unsigned char* Bt,
                            float* C, const float* bias, int M, int N, int K,
                            float descale, bool relu, hipStream_t stream);
 
+// ---- MX block-scaled fp4 (e2m1, 2x the fp8 MFMA rate) --------------------
+// x f32 [R,K] (K%64==0) -> q nibbles [R,K/2] (shuffled) + sc [R,K/64,2].
+hipError_t quantize_mx_fp4(const float* x, unsigned char* q, unsigned char* sc,
+                           int R, int K, hipStream_t stream);
+bool gemm_mx_fp4_supported(int M, int N, int K);
+hipError_t gemm_mx_fp4_nt8ph(const unsigned char* A, const unsigned char* sA,
+                             const unsigned char* Bt, const unsigned char* sB,
+                             float* C, int M, int N, int K,
+                             hipStream_t stream);
+
 }  // namespace lpx

@@ -323,6 +323,47 @@ at::Tensor gemm_mx(const at::Tensor& A, const at::Tensor& sA,
   return C;
 }
 
+std::vector<at::Tensor> quantize_mx4(const at::Tensor& x) {
+  CHECK_DEV(x);
+  TORCH_CHECK(x.scalar_type() == at::kFloat && x.dim() == 2 &&
+              x.size(1) % 64 == 0, "quantize_mx4 expects f32 [R, K%64==0]");
+  const int64_t R = x.size(0), K = x.size(1);
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
+  auto q = at::empty({R, K / 2}, x.options().dtype(at::kByte));
+  auto sc = at::empty({R, K / 64, 2}, x.options().dtype(at::kByte));
+  check_hip(lpx::quantize_mx_fp4(x.data_ptr<float>(), q.data_ptr<uint8_t>(),
+                                 sc.data_ptr<uint8_t>(), (int)R, (int)K,
+                                 current_stream(x)),
+            "quantize_mx_fp4");
+  return {q, sc};
+}
+
+at::Tensor gemm_mx4(const at::Tensor& A, const at::Tensor& sA,
+                    const at::Tensor& Bt, const at::Tensor& sB) {
+  CHECK_DEV(A);
+  CHECK_DEV(Bt);
+  CHECK_DEV(sA);
+  CHECK_DEV(sB);
+  TORCH_CHECK(A.scalar_type() == at::kByte && Bt.scalar_type() == at::kByte &&
+              sA.scalar_type() == at::kByte && sB.scalar_type() == at::kByte,
+              "gemm_mx4 expects byte tensors");
+  TORCH_CHECK(A.dim() == 2 && Bt.dim() == 2 && A.size(1) == Bt.size(1),
+              "gemm_mx4 expects A [M,K/2], Bt [N,K/2]");
+  const int64_t M = A.size(0), K = A.size(1) * 2, N = Bt.size(0);
+  TORCH_CHECK(lpx::gemm_mx_fp4_supported((int)M, (int)N, (int)K),
+              "gemm_mx4 requires M,N%256==0 and K%128==0");
+  TORCH_CHECK(sA.numel() == M * (K / 64) * 2 && sB.numel() == N * (K / 64) * 2,
+              "scale shapes must be [rows, K/64, 2]");
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(A.device());
+  auto C = at::empty({M, N}, A.options().dtype(at::kFloat));
+  check_hip(lpx::gemm_mx_fp4_nt8ph(A.data_ptr<uint8_t>(), sA.data_ptr<uint8_t>(),
+                                   Bt.data_ptr<uint8_t>(), sB.data_ptr<uint8_t>(),
+                                   C.data_ptr<float>(), (int)M, (int)N, (int)K,
+                                   current_stream(A)),
+            "gemm_mx_fp4_nt8ph");
+  return C;
+}
+
 at::Tensor rf_predict8(const at::Tensor& nodes8, const at::Tensor& leaf_values,
                        const at::Tensor& roots, const at::Tensor& X,
                        int64_t ilp) {
@@ -509,6 +550,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("x"));
   m.def("gemm_mx", &gemm_mx,
         "MX block-scaled fp8 GEMM (scales applied in hardware)", py::arg("A"),
+        py::arg("sA"), py::arg("Bt"), py::arg("sB"));
+  m.def("quantize_mx4", &quantize_mx4,
+        "f32 [R,K] -> (e2m1 nibbles shuffled [R,K/2], E8M0 scales)",
+        py::arg("x"));
+  m.def("gemm_mx4", &gemm_mx4,
+        "MX block-scaled fp4 GEMM (2x fp8 MFMA rate)", py::arg("A"),
         py::arg("sA"), py::arg("Bt"), py::arg("sB"));
   m.def("gemm_bf16_tune8", &gemm_bf16_tune8, "8-phase GEMM tuning variants",
         py::arg("A"), py::arg("Bt"), py::arg("tune") = 0);

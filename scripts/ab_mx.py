@@ -37,6 +37,10 @@ def main():
         print(f"n={n} gemm_fp8(v3) {fl/dt:8.0f} TF  {dt*1e3:.3f} ms")
         dt = bench(lambda: e.quantize_mx(a))
         print(f"n={n} quantize_mx  {2*n*n*4/dt/1e12:8.2f} TB/s rd+wr  {dt*1e3:.3f} ms")
+        qa4, sa4 = e.quantize_mx4(a)
+        qb4, sb4 = e.quantize_mx4(b)
+        dt = bench(lambda: e.gemm_mx4(qa4, sa4, qb4, sb4))
+        print(f"n={n} gemm_mx4     {fl/dt:8.0f} TF  {dt*1e3:.3f} ms")
 
 
 if __name__ == "__main__":

@@ -306,6 +306,55 @@ def test_gemm_mx_tail_shapes(ops):
     assert err < 1e-3, err
 
 
+_E2M1 = [0.0, 0.5, 1.0, 1.5, 2.0, 3.0, 4.0, 6.0,
+         -0.0, -0.5, -1.0, -1.5, -2.0, -3.0, -4.0, -6.0]
+
+
+def _dequant_mx4(q, sc, K):
+    """Reconstruct f32 from MX-fp4 nibbles + E8M0 scales (low nibble first;
+    shuffled [q0|q2|q1|q3] quarters of 16 elements = 8 bytes)."""
+    R = q.shape[0]
+    g = K // 64
+    lut = torch.tensor(_E2M1, device=q.device)
+    lo = lut[(q & 0xF).long()]
+    hi = lut[(q >> 4).long()]
+    d = torch.stack([lo, hi], dim=-1).reshape(R, g, 4, 16)
+    scales = torch.pow(2.0, sc.float() - 127.0)
+    d = d * scales[:, :, [0, 1, 0, 1]].unsqueeze(-1)
+    return d[:, :, [0, 2, 1, 3], :].reshape(R, K)
+
+
+def test_quantize_mx_fp4_roundtrip(ops):
+    e = ops.ext()
+    g = torch.Generator(device="cpu").manual_seed(12)
+    x = (torch.randn(64, 256, generator=g) *
+         torch.logspace(-6, 6, 64, base=2.0).unsqueeze(1)).to(DEV)
+    q, sc = e.quantize_mx4(x)
+    xd = _dequant_mx4(q, sc, 256)
+    err = (xd - x).abs().view(64, 4, 2, 32)
+    blk_amax = x.view(64, 4, 2, 32).abs().amax(dim=-1, keepdim=True)
+    # e2m1 worst step/amax: gap 2 around 4..6, amax in (3,6] -> <= 1/3
+    assert (err / blk_amax.clamp(min=1e-30)).max().item() < 0.35
+
+
+def test_gemm_mx_fp4_vs_dequant_ref(ops):
+    e = ops.ext()
+    g = torch.Generator(device="cpu").manual_seed(13)
+    M, N, K = 512, 512, 384
+    A = (torch.randn(M, K, generator=g) * 2).to(DEV)
+    Bt = torch.randn(N, K, generator=g).to(DEV)
+    qa, sa = e.quantize_mx4(A)
+    qb, sb = e.quantize_mx4(Bt)
+    C = e.gemm_mx4(qa, sa, qb, sb)
+    ref = _dequant_mx4(qa, sa, K) @ _dequant_mx4(qb, sb, K).t()
+    err = (C - ref).abs().max().item() / (ref.abs().max().item() + 1e-6)
+    assert err < 1e-3, err
+    # fp4 quantization noise vs exact matmul (e2m1 element RMS ~10%)
+    exact = A @ Bt.t()
+    rel = (C - exact).norm().item() / exact.norm().item()
+    assert rel < 0.2, rel
+
+
 def test_gemm_fp8_scaled_mfma_path(ops):
     # 256-multiple shapes dispatch to the unit-scale scaled-MFMA kernel
     e = ops.ext()
