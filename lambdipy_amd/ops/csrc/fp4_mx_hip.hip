@@ -6,14 +6,14 @@
 // fp4 silicon peak is ~10 PF). Operands shrink to 16 B/lane (4 VGPRs — the
 // disassembly shows v[16:19] operands), halving LDS and HBM traffic too.
 //
-// Layout mirrors fp8_mx*.hip: lane pair {l, l+32} concatenates to row
-// (l&31)'s 32 k-bytes (64 e2m1 elements); lane l's scale byte applies to
-// byte-half (l>>5) of both lanes. Packing each 64-element group as
-// [q0|q2|q1|q3] (quarters = 16 elements = 8 bytes) makes the hardware scale
-// blocks the standard contiguous MX-32 blocks. Nibble order within a block
-// is irrelevant to the product (A and B share the permutation and the
-// block scale is constant), but quantizer and tests use low-nibble-first
-// (the v_cvt_scalef32_pk_fp4_f32 convention).
+// fp4 scale semantics DIFFER from fp8 (scripts/fp4_diag.py, hypothesis H1
+// exact to 0.00000): lane l's scale byte applies to lane l's OWN 16 operand
+// bytes = 32 e2m1 elements = exactly one MX-32 block. So operands pack
+// LINEARLY (no quarter shuffle): lane kb = l>>5 holds contiguous original
+// k [kb*32, +32) of its row, and sc[r][g][kb] covers that block. Nibble
+// order within a block is irrelevant to the product (A and B share the
+// permutation and the block scale is constant), but quantizer and tests
+// use low-nibble-first (the v_cvt_scalef32_pk_fp4_f32 convention).
 //
 // 8-phase glds pipeline: half-tile = 128 rows x 64 B = one 16-B glds per
 // thread per stage call; per-tile-cycle call events [2,1,1,1] keep the
@@ -34,7 +34,7 @@ typedef __attribute__((ext_vector_type(8))) int i32x8v;
 typedef __attribute__((ext_vector_type(2))) short s16x2;
 
 // ---- MX fp4 quantization -------------------------------------------------
-// x f32 [R][K], K % 64 == 0 -> q [R][K/2] shuffled nibbles + sc [R][K/64][2]
+// x f32 [R][K], K % 64 == 0 -> q [R][K/2] linear nibbles + sc [R][K/64][2]
 __global__ __launch_bounds__(THREADS) void quantize_mx4_kernel(
     const float* __restrict__ x, unsigned char* __restrict__ q,
     unsigned char* __restrict__ sc, int R, int K) {
@@ -70,8 +70,7 @@ __global__ __launch_bounds__(THREADS) void quantize_mx4_kernel(
     unsigned char* base = q + (long)r * (K >> 1) + g * 32;
 #pragma unroll
     for (int h = 0; h < 2; ++h) {
-      const int qq = 2 * kb + h;                    // original quarter
-      const int pos = ((qq & 1) << 1) | (qq >> 1);  // shuffled slot
+      const int pos = 2 * kb + h;  // LINEAR slot (fp4 lanes own their block)
       // quarter = 16 elements -> 8 bytes = 2 u32 (4 cvt calls each; the
       // byte-select operand must be a literal)
       unsigned int w[2] = {0, 0};

@@ -496,11 +496,13 @@ __device__ __forceinline__ fp8frag32 frag32(const char* __restrict__ lds,
   return fp8frag32{lo.x, lo.y, lo.z, lo.w, hi.x, hi.y, hi.z, hi.w};
 }
 
-template <bool RELU, bool BIAS>
+template <bool RELU, bool BIAS, bool OUT8 = false>
 __global__ __launch_bounds__(THREADS2, 2) void conv_fp8_impx_kernel(
     const unsigned char* __restrict__ xq, const unsigned char* __restrict__ Wq,
-    float* __restrict__ C, const float* __restrict__ bias, ConvGeom g, int M,
-    int N, int K, float descale, unsigned tiles_m, unsigned tiles_n) {
+    float* __restrict__ C, unsigned char* __restrict__ C8,
+    const float* __restrict__ bias, ConvGeom g, int M,
+    int N, int K, float descale, float out_inv, unsigned tiles_m,
+    unsigned tiles_n) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* const lA = smem;
   char* const lB = smem + BM2 * ROWB2;
@@ -573,7 +575,10 @@ __global__ __launch_bounds__(THREADS2, 2) void conv_fp8_impx_kernel(
           float v = acc[mf][nf][r] * descale;
           if constexpr (BIAS) v += bias[col];
           if constexpr (RELU) v = v > 0.f ? v : 0.f;
-          C[(long)row * N + col] = v;
+          if constexpr (OUT8)
+            C8[(long)row * N + col] = lpx_f32_to_e4m3(v * out_inv);
+          else
+            C[(long)row * N + col] = v;
         }
       }
     }
@@ -596,16 +601,51 @@ hipError_t conv_fp8_implicit(const unsigned char* xq, const unsigned char* Wq,
   dim3 grid(tiles_m * tiles_n), block(THREADS2);
   if (bias && relu)
    hipLaunchKernelGGL(( conv_fp8_impx_kernel<true, true>), dim3(grid), dim3(block), lds, stream, 
-        xq, Wq, C, bias, g, M, KO, K, descale, tiles_m, tiles_n);
+        xq, Wq, C, nullptr, bias, g, M, KO, K, descale, 1.f, tiles_m, tiles_n);
   else if (bias)
    hipLaunchKernelGGL(( conv_fp8_impx_kernel<false, true>), dim3(grid), dim3(block), lds, stream, 
-        xq, Wq, C, bias, g, M, KO, K, descale, tiles_m, tiles_n);
+        xq, Wq, C, nullptr, bias, g, M, KO, K, descale, 1.f, tiles_m, tiles_n);
   else if (relu)
    hipLaunchKernelGGL(( conv_fp8_impx_kernel<true, false>), dim3(grid), dim3(block), lds, stream, 
-        xq, Wq, C, bias, g, M, KO, K, descale, tiles_m, tiles_n);
+        xq, Wq, C, nullptr, bias, g, M, KO, K, descale, 1.f, tiles_m, tiles_n);
   else
    hipLaunchKernelGGL(( conv_fp8_impx_kernel<false, false>), dim3(grid), dim3(block), lds, stream, 
-        xq, Wq, C, bias, g, M, KO, K, descale, tiles_m, tiles_n);
+        xq, Wq, C, nullptr, bias, g, M, KO, K, descale, 1.f, tiles_m, tiles_n);
+  return hipGetLastError();
+}
+
+// Fused epilogue quantization: out bytes = e4m3(conv_out * out_inv) — the
+// next layer's pre-quantized input (skips a full f32 write + read + u8
+// write per layer pair on the ResNet path).
+hipError_t conv_fp8_implicit_q8(const unsigned char* xq,
+                                const unsigned char* Wq, unsigned char* C8,
+                                const float* bias, int Nb, int H, int W,
+                                int Cin, int R, int S, int HO, int WO,
+                                int stride, int pad, int KO, float descale,
+                                bool relu, float out_inv, hipStream_t stream) {
+  if (Cin % 16 != 0) return hipErrorInvalidValue;
+  ConvGeom g{Nb, H, W, Cin, R, S, HO, WO, stride, pad};
+  const int M = Nb * HO * WO;
+  const int K = R * S * Cin;
+  const unsigned tiles_m = lpx_ceil_div(M, BM2), tiles_n = lpx_ceil_div(KO, BN2);
+  const size_t lds = (size_t)2 * BM2 * ROWB2;
+  dim3 grid(tiles_m * tiles_n), block(THREADS2);
+  if (bias && relu)
+   hipLaunchKernelGGL(( conv_fp8_impx_kernel<true, true, true>), dim3(grid), dim3(block), lds, stream, 
+        xq, Wq, nullptr, C8, bias, g, M, KO, K, descale, out_inv, tiles_m,
+        tiles_n);
+  else if (bias)
+   hipLaunchKernelGGL(( conv_fp8_impx_kernel<false, true, true>), dim3(grid), dim3(block), lds, stream, 
+        xq, Wq, nullptr, C8, bias, g, M, KO, K, descale, out_inv, tiles_m,
+        tiles_n);
+  else if (relu)
+   hipLaunchKernelGGL(( conv_fp8_impx_kernel<true, false, true>), dim3(grid), dim3(block), lds, stream, 
+        xq, Wq, nullptr, C8, bias, g, M, KO, K, descale, out_inv, tiles_m,
+        tiles_n);
+  else
+   hipLaunchKernelGGL(( conv_fp8_impx_kernel<false, false, true>), dim3(grid), dim3(block), lds, stream, 
+        xq, Wq, nullptr, C8, bias, g, M, KO, K, descale, out_inv, tiles_m,
+        tiles_n);
   return hipGetLastError();
 }
 

@@ -138,11 +138,12 @@ __device__ __forceinline__ int scalex(const char* __restrict__ smem,
 // SC=false: unit scales (literal 127) — plain per-tensor fp8 GEMM on the
 // 2x-rate scaled MFMA (the k-permutation is self-consistent between A and
 // B, so unshuffled operands contract correctly) with fused descale/bias/relu.
-template <bool SC, bool BIAS, bool RELU>
+template <bool SC, bool BIAS, bool RELU, bool OUT8 = false>
 __global__ __launch_bounds__(THREADSX, 1) void gemm_mx_8ph_kernel(
     const unsigned char* __restrict__ A, const unsigned char* __restrict__ sA,
     const unsigned char* __restrict__ Bt, const unsigned char* __restrict__ sB,
-    float* __restrict__ C, const float* __restrict__ bias, float descale,
+    float* __restrict__ C, unsigned char* __restrict__ C8,
+    const float* __restrict__ bias, float descale, float out_inv,
     int M, int N, int K, unsigned tiles_m, unsigned tiles_n) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
 
@@ -246,7 +247,10 @@ __global__ __launch_bounds__(THREADSX, 1) void gemm_mx_8ph_kernel(
         if constexpr (!SC) v *= descale;
         if constexpr (BIAS) v += bias[col];
         if constexpr (RELU) v = v > 0.f ? v : 0.f;
-        C[(long)row * N + col] = v;
+        if constexpr (OUT8)
+          C8[(long)row * N + col] = lpx_f32_to_e4m3(v * out_inv);
+        else
+          C[(long)row * N + col] = v;
       }
     }
   }
@@ -266,7 +270,8 @@ hipError_t gemm_mx_fp8_nt8ph(const unsigned char* A, const unsigned char* sA,
   const size_t lds = 2 * PARITY_BYTES + 2 * SC_TILE;  // 132 KiB
  hipLaunchKernelGGL(( gemm_mx_8ph_kernel<true, false, false>)
       , dim3(dim3(tiles_m * tiles_n)), dim3(dim3(THREADSX)), lds, stream, 
-          A, sA, Bt, sB, C, nullptr, 1.f, M, N, K, tiles_m, tiles_n);
+          A, sA, Bt, sB, C, nullptr, nullptr, 1.f, 1.f, M, N, K, tiles_m,
+          tiles_n);
   return hipGetLastError();
 }
 
@@ -279,16 +284,47 @@ hipError_t gemm_fp8_nt8phx(const unsigned char* A, const unsigned char* Bt,
   dim3 grid(tiles_m * tiles_n), block(THREADSX);
   if (bias && relu)
    hipLaunchKernelGGL(( gemm_mx_8ph_kernel<false, true, true>), dim3(grid), dim3(block), lds, stream, 
-        A, nullptr, Bt, nullptr, C, bias, descale, M, N, K, tiles_m, tiles_n);
+        A, nullptr, Bt, nullptr, C, nullptr, bias, descale, 1.f, M, N, K,
+        tiles_m, tiles_n);
   else if (bias)
    hipLaunchKernelGGL(( gemm_mx_8ph_kernel<false, true, false>), dim3(grid), dim3(block), lds, stream, 
-        A, nullptr, Bt, nullptr, C, bias, descale, M, N, K, tiles_m, tiles_n);
+        A, nullptr, Bt, nullptr, C, nullptr, bias, descale, 1.f, M, N, K,
+        tiles_m, tiles_n);
   else if (relu)
    hipLaunchKernelGGL(( gemm_mx_8ph_kernel<false, false, true>), dim3(grid), dim3(block), lds, stream, 
-        A, nullptr, Bt, nullptr, C, nullptr, descale, M, N, K, tiles_m, tiles_n);
+        A, nullptr, Bt, nullptr, C, nullptr, nullptr, descale, 1.f, M, N, K,
+        tiles_m, tiles_n);
   else
    hipLaunchKernelGGL(( gemm_mx_8ph_kernel<false, false, false>), dim3(grid), dim3(block), lds, stream, 
-        A, nullptr, Bt, nullptr, C, nullptr, descale, M, N, K, tiles_m, tiles_n);
+        A, nullptr, Bt, nullptr, C, nullptr, nullptr, descale, 1.f, M, N, K,
+        tiles_m, tiles_n);
+  return hipGetLastError();
+}
+
+// Same with fused epilogue quantization (out bytes = e4m3(v * out_inv)).
+hipError_t gemm_fp8_nt8phx_q8(const unsigned char* A, const unsigned char* Bt,
+                              unsigned char* C8, const float* bias, int M,
+                              int N, int K, float descale, bool relu,
+                              float out_inv, hipStream_t stream) {
+  const unsigned tiles_m = (unsigned)(M / BMX), tiles_n = (unsigned)(N / BMX);
+  const size_t lds = 2 * PARITY_BYTES;
+  dim3 grid(tiles_m * tiles_n), block(THREADSX);
+  if (bias && relu)
+   hipLaunchKernelGGL(( gemm_mx_8ph_kernel<false, true, true, true>), dim3(grid), dim3(block), lds, stream, 
+        A, nullptr, Bt, nullptr, nullptr, C8, bias, descale, out_inv, M, N, K,
+        tiles_m, tiles_n);
+  else if (bias)
+   hipLaunchKernelGGL(( gemm_mx_8ph_kernel<false, true, false, true>), dim3(grid), dim3(block), lds, stream, 
+        A, nullptr, Bt, nullptr, nullptr, C8, bias, descale, out_inv, M, N, K,
+        tiles_m, tiles_n);
+  else if (relu)
+   hipLaunchKernelGGL(( gemm_mx_8ph_kernel<false, false, true, true>), dim3(grid), dim3(block), lds, stream, 
+        A, nullptr, Bt, nullptr, nullptr, C8, nullptr, descale, out_inv, M, N,
+        K, tiles_m, tiles_n);
+  else
+   hipLaunchKernelGGL(( gemm_mx_8ph_kernel<false, false, false, true>), dim3(grid), dim3(block), lds, stream, 
+        A, nullptr, Bt, nullptr, nullptr, C8, nullptr, descale, out_inv, M, N,
+        K, tiles_m, tiles_n);
   return hipGetLastError();
 }
 

@@ -143,6 +143,17 @@ hipError_t gemm_mx_fp8_nt8ph(const unsigned char* A, const unsigned char* sA,
 hipError_t gemm_fp8_nt8phx(const unsigned char* A, const unsigned char* Bt,
                            float* C, const float* bias, int M, int N, int K,
                            float descale, bool relu, hipStream_t stream);
+// fused epilogue quantize: out = e4m3((descale*acc + bias, relu) * out_inv)
+hipError_t gemm_fp8_nt8phx_q8(const unsigned char* A, const unsigned char* Bt,
+                              unsigned char* C8, const float* bias, int M,
+                              int N, int K, float descale, bool relu,
+                              float out_inv, hipStream_t stream);
+hipError_t conv_fp8_implicit_q8(const unsigned char* xq,
+                                const unsigned char* Wq, unsigned char* C8,
+                                const float* bias, int Nb, int H, int W,
+                                int Cin, int R, int S, int HO, int WO,
+                                int stride, int pad, int KO, float descale,
+                                bool relu, float out_inv, hipStream_t stream);
 
 // ---- MX block-scaled fp4 (e2m1, 2x the fp8 MFMA rate) --------------------
 // x f32 [R,K] (K%64==0) -> q nibbles [R,K/2] (shuffled) + sc [R,K/64,2].

@@ -73,6 +73,16 @@ __device__ __forceinline__ unsigned lpx_xcd_swizzle(unsigned bid, unsigned nwg) 
   return base + idx;
 }
 
+// ---- f32 -> OCP e4m3 via the hw scaled conversion (RNE). The hw op
+// returns NaN above the e4m3 max, so clamp to +-448 first (satfinite).
+__device__ __forceinline__ unsigned char lpx_f32_to_e4m3(float x) {
+  typedef __attribute__((ext_vector_type(2))) short lpx_s16x2;
+  x = fminf(fmaxf(x, -448.f), 448.f);
+  lpx_s16x2 t = {0, 0};
+  t = __builtin_amdgcn_cvt_scalef32_pk_fp8_f32(t, x, x, 1.0f, false);
+  return (unsigned char)(t[0] & 0xff);
+}
+
 // ---- grid helper ---------------------------------------------------------
 static inline unsigned lpx_ceil_div(unsigned a, unsigned b) { return (a + b - 1) / b; }
 

@@ -276,6 +276,74 @@ at::Tensor gemm_fp8(const at::Tensor& A, const at::Tensor& Bt, double descale,
   return C;
 }
 
+at::Tensor gemm_fp8_q8(const at::Tensor& A, const at::Tensor& Bt,
+                       double descale, const c10::optional<at::Tensor>& bias,
+                       bool relu, double out_scale) {
+  CHECK_DEV(A);
+  CHECK_DEV(Bt);
+  TORCH_CHECK(A.scalar_type() == at::kByte && Bt.scalar_type() == at::kByte,
+              "gemm_fp8_q8 expects e4m3 bytes");
+  TORCH_CHECK(A.dim() == 2 && Bt.dim() == 2 && A.size(1) == Bt.size(1),
+              "gemm_fp8_q8 expects A [M,K], Bt [N,K]");
+  const int64_t M = A.size(0), K = A.size(1), N = Bt.size(0);
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(A.device());
+  const float* bptr = nullptr;
+  if (bias.has_value()) {
+    CHECK_DEV(bias.value());
+    bptr = bias->data_ptr<float>();
+  }
+  auto C8 = at::empty({M, N}, A.options().dtype(at::kByte));
+  const float oinv = (float)(1.0 / out_scale);
+  if (M % 256 == 0 && N % 256 == 0 && K % 128 == 0 && K >= 128) {
+    check_hip(lpx::gemm_fp8_nt8phx_q8(A.data_ptr<uint8_t>(),
+                                      Bt.data_ptr<uint8_t>(),
+                                      C8.data_ptr<uint8_t>(), bptr, (int)M,
+                                      (int)N, (int)K, (float)descale, relu,
+                                      oinv, current_stream(A)),
+              "gemm_fp8_nt8phx_q8");
+  } else {
+    // unfused fallback for unaligned shapes: f32 GEMM + quantize pass
+    auto C = at::empty({M, N}, A.options().dtype(at::kFloat));
+    check_hip(lpx::gemm_fp8_nt(A.data_ptr<uint8_t>(), Bt.data_ptr<uint8_t>(),
+                               C.data_ptr<float>(), bptr, (int)M, (int)N,
+                               (int)K, (float)descale, relu,
+                               current_stream(A)),
+              "gemm_fp8_nt");
+    check_hip(lpx::quantize_fp8(C.data_ptr<float>(), C8.data_ptr<uint8_t>(),
+                                C.numel(), oinv, current_stream(A)),
+              "quantize_fp8");
+  }
+  return C8;
+}
+
+at::Tensor conv_fp8_q8(const at::Tensor& xq, const at::Tensor& Wq, int64_t H,
+                       int64_t W, int64_t Cin, int64_t R, int64_t S,
+                       int64_t stride, int64_t pad, double descale,
+                       const c10::optional<at::Tensor>& bias, bool relu,
+                       double out_scale) {
+  CHECK_DEV(xq);
+  CHECK_DEV(Wq);
+  TORCH_CHECK(xq.scalar_type() == at::kByte && Wq.scalar_type() == at::kByte,
+              "conv_fp8_q8 expects e4m3 bytes");
+  const int64_t numel = xq.numel();
+  const int64_t Nb = numel / (H * W * Cin);
+  const int64_t KO = Wq.size(0);
+  const int64_t HO = (H + 2 * pad - R) / stride + 1;
+  const int64_t WO = (W + 2 * pad - S) / stride + 1;
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(xq.device());
+  const float* bptr = nullptr;
+  if (bias.has_value()) bptr = bias->data_ptr<float>();
+  auto C8 = at::empty({Nb * HO * WO, KO}, xq.options());
+  check_hip(lpx::conv_fp8_implicit_q8(
+                xq.data_ptr<uint8_t>(), Wq.data_ptr<uint8_t>(),
+                C8.data_ptr<uint8_t>(), bptr, (int)Nb, (int)H, (int)W,
+                (int)Cin, (int)R, (int)S, (int)HO, (int)WO, (int)stride,
+                (int)pad, (int)KO, (float)descale, relu,
+                (float)(1.0 / out_scale), current_stream(xq)),
+            "conv_fp8_implicit_q8");
+  return C8;
+}
+
 std::vector<at::Tensor> quantize_mx(const at::Tensor& x) {
   CHECK_DEV(x);
   TORCH_CHECK(x.scalar_type() == at::kFloat && x.dim() == 2 &&
@@ -545,6 +613,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_fp8", &gemm_fp8, "fp8 e4m3 MFMA GEMM with fused descale/bias/relu",
         py::arg("A"), py::arg("Bt"), py::arg("descale") = 1.0,
         py::arg("bias") = py::none(), py::arg("relu") = false);
+  m.def("gemm_fp8_q8", &gemm_fp8_q8,
+        "fp8 GEMM with fused epilogue quantization (e4m3 out)", py::arg("A"),
+        py::arg("Bt"), py::arg("descale"), py::arg("bias") = c10::nullopt,
+        py::arg("relu") = false, py::arg("out_scale") = 1.0);
+  m.def("conv_fp8_q8", &conv_fp8_q8,
+        "implicit fp8 conv with fused epilogue quantization", py::arg("xq"),
+        py::arg("Wq"), py::arg("H"), py::arg("W"), py::arg("Cin"),
+        py::arg("R"), py::arg("S"), py::arg("stride"), py::arg("pad"),
+        py::arg("descale"), py::arg("bias") = c10::nullopt,
+        py::arg("relu") = false, py::arg("out_scale") = 1.0);
   m.def("quantize_mx", &quantize_mx,
         "f32 [R,K] -> (e4m3 bytes shuffled, E8M0 [R,K/64,2] block scales)",
         py::arg("x"));

@@ -60,6 +60,11 @@ class Conv2dFP8:
                      * 0.01).to(device)
         self.sa: Optional[float] = None  # frozen activation scale
 
+    def out_shape(self, H: int, W: int) -> Tuple[int, int]:
+        ho = (H + 2 * self.pad - self.k) // self.stride + 1
+        wo = (W + 2 * self.pad - self.k) // self.stride + 1
+        return ho, wo
+
     def __call__(self, x: torch.Tensor) -> torch.Tensor:
         # x: [N, H, W, Cin] f32
         N, H, W, C = x.shape
@@ -67,8 +72,7 @@ class Conv2dFP8:
         if self.sa is None:
             self.sa = _act_scale(x) * _CAL_HEADROOM
         sa = self.sa
-        ho = (H + 2 * self.pad - self.k) // self.stride + 1
-        wo = (W + 2 * self.pad - self.k) // self.stride + 1
+        ho, wo = self.out_shape(H, W)
         if self.k == 1 and self.stride == 1 and self.pad == 0:
             aq = e.quantize_fp8(x.reshape(-1, C).contiguous(), sa)
             out = e.gemm_fp8(aq, self.wq, sa * self.wscale, self.bias, self.relu)
@@ -84,6 +88,32 @@ class Conv2dFP8:
             out = e.gemm_fp8(aq, self.wq, sa * self.wscale, self.bias, self.relu)
         return out.view(N, ho, wo, self.cout)
 
+    # -- fused-chain entry points (inputs/outputs as e4m3 bytes) -----------
+    # xq is NHWC e4m3 quantized at self.sa (the producer wrote it with
+    # out_scale = self.sa). out_scale=None -> f32 result; else e4m3 bytes.
+    def run_q(self, xq: torch.Tensor, N: int, H: int, W: int,
+              out_scale: Optional[float]):
+        e = _ops.ext()
+        desc = self.sa * self.wscale
+        ho, wo = self.out_shape(H, W)
+        if self.k == 1 and self.stride == 1 and self.pad == 0:
+            a2 = xq.reshape(-1, self.cin)
+            if out_scale is None:
+                out = e.gemm_fp8(a2, self.wq, desc, self.bias, self.relu)
+            else:
+                out = e.gemm_fp8_q8(a2, self.wq, desc, self.bias, self.relu,
+                                    out_scale)
+        else:
+            if out_scale is None:
+                out = e.conv_fp8(xq, self.wq, H, W, self.cin, self.k, self.k,
+                                 self.stride, self.pad, desc, self.bias,
+                                 self.relu)
+            else:
+                out = e.conv_fp8_q8(xq, self.wq, H, W, self.cin, self.k,
+                                    self.k, self.stride, self.pad, desc,
+                                    self.bias, self.relu, out_scale)
+        return out.view(N, ho, wo, self.cout), ho, wo
+
 
 class Bottleneck:
     def __init__(self, cin: int, mid: int, cout: int, stride: int, device, gen):
@@ -96,7 +126,17 @@ class Bottleneck:
 
     def __call__(self, x: torch.Tensor) -> torch.Tensor:
         idn = x if self.down is None else self.down(x)
-        y = self.conv3(self.conv2(self.conv1(x)))
+        if self.conv2.sa is not None and self.conv3.sa is not None:
+            # fused chain: conv1 and conv2 emit e4m3 at the NEXT conv's
+            # frozen input scale — skips two f32 write/read/quantize passes
+            e = _ops.ext()
+            N, H, W, C = x.shape
+            xq = e.quantize_fp8(x.contiguous(), self.conv1.sa)
+            q1, h, w = self.conv1.run_q(xq, N, H, W, self.conv2.sa)
+            q2, h, w = self.conv2.run_q(q1.contiguous(), N, h, w, self.conv3.sa)
+            y, h, w = self.conv3.run_q(q2.contiguous(), N, h, w, None)
+        else:
+            y = self.conv3(self.conv2(self.conv1(x)))  # calibration pass
         return F.relu(y + idn)
 
 

@@ -312,16 +312,16 @@ _E2M1 = [0.0, 0.5, 1.0, 1.5, 2.0, 3.0, 4.0, 6.0,
 
 def _dequant_mx4(q, sc, K):
     """Reconstruct f32 from MX-fp4 nibbles + E8M0 scales (low nibble first;
-    shuffled [q0|q2|q1|q3] quarters of 16 elements = 8 bytes)."""
+    LINEAR layout — fp4 lanes own their whole 32-element block, so no
+    quarter shuffle; see scripts/fp4_diag.py)."""
     R = q.shape[0]
     g = K // 64
     lut = torch.tensor(_E2M1, device=q.device)
     lo = lut[(q & 0xF).long()]
     hi = lut[(q >> 4).long()]
-    d = torch.stack([lo, hi], dim=-1).reshape(R, g, 4, 16)
+    d = torch.stack([lo, hi], dim=-1).reshape(R, g, 2, 32)
     scales = torch.pow(2.0, sc.float() - 127.0)
-    d = d * scales[:, :, [0, 1, 0, 1]].unsqueeze(-1)
-    return d[:, :, [0, 2, 1, 3], :].reshape(R, K)
+    return (d * scales.unsqueeze(-1)).reshape(R, K)
 
 
 def test_quantize_mx_fp4_roundtrip(ops):
@@ -374,6 +374,48 @@ def test_gemm_fp8_scaled_mfma_path(ops):
     C2 = e.gemm_fp8(Aq, Btq, sa * sb, bias, True)
     ref2 = torch.relu(ref + bias)
     assert (C2 - ref2).abs().max().item() < 1e-2
+
+
+def test_gemm_fp8_q8_matches_two_step(ops):
+    e = ops.ext()
+    g = torch.Generator(device="cpu").manual_seed(14)
+    for M, N, K in ((512, 512, 256), (300, 200, 192)):  # fused + fallback
+        A = torch.randn(M, K, generator=g).to(DEV)
+        Bt = torch.randn(N, K, generator=g).to(DEV)
+        Aq = e.quantize_fp8(A, 1.0)
+        Btq = e.quantize_fp8(Bt, 1.0)
+        out_scale = 0.37
+        q8 = e.gemm_fp8_q8(Aq, Btq, 1.0, None, True, out_scale)
+        ref = e.quantize_fp8(e.gemm_fp8(Aq, Btq, 1.0, None, True), out_scale)
+        mism = (q8 != ref).float().mean().item()
+        assert mism < 2e-3, f"{M}x{N}x{K}: {mism*100:.3f}% byte mismatch"
+
+
+def test_conv_fp8_q8_matches_two_step(ops):
+    e = ops.ext()
+    g = torch.Generator(device="cpu").manual_seed(15)
+    N, H, W, C, KO = 2, 14, 14, 32, 64
+    x = torch.randn(N, H, W, C, generator=g).to(DEV)
+    w = torch.randn(KO, 3 * 3 * C, generator=g).to(DEV)
+    xq = e.quantize_fp8(x, 1.0)
+    wq = e.quantize_fp8(w, 1.0)
+    out_scale = 0.5
+    q8 = e.conv_fp8_q8(xq, wq, H, W, C, 3, 3, 1, 1, 1.0, None, True, out_scale)
+    f32 = e.conv_fp8(xq, wq, H, W, C, 3, 3, 1, 1, 1.0, None, True)
+    ref = e.quantize_fp8(f32, out_scale)
+    mism = (q8 != ref.view_as(q8)).float().mean().item()
+    assert mism < 2e-3, f"{mism*100:.3f}% byte mismatch"
+
+
+def test_resnet_bottleneck_fused_consistency(ops):
+    from lambdipy_amd.shims.lpx_torch import Bottleneck
+    g = torch.Generator().manual_seed(3)
+    blk = Bottleneck(64, 16, 64, 1, DEV, g)
+    x = torch.randn(2, 8, 8, 64, generator=torch.Generator().manual_seed(4)).to(DEV)
+    y_cal = blk(x)        # calibration pass (unfused)
+    y_fused = blk(x)      # fused chain, same input
+    rel = (y_fused - y_cal).norm().item() / (y_cal.norm().item() + 1e-9)
+    assert rel < 0.05, rel
 
 
 def test_gemm_fp8_bias_relu(ops):
