@@ -155,6 +155,11 @@ hipError_t conv_fp8_implicit_q8(const unsigned char* xq,
                                 int stride, int pad, int KO, float descale,
                                 bool relu, float out_inv, hipStream_t stream);
 
+// NHWC f32 max-pool (k<=8, C%4==0) — avoids the NCHW permute round trip.
+hipError_t maxpool2d_nhwc_f32(const float* x, float* out, int N, int H, int W,
+                              int C, int k, int stride, int pad,
+                              hipStream_t stream);
+
 // ---- MX block-scaled fp4 (e2m1, 2x the fp8 MFMA rate) --------------------
 // x f32 [R,K] (K%64==0) -> q nibbles [R,K/2] (shuffled) + sc [R,K/64,2].
 hipError_t quantize_mx_fp4(const float* x, unsigned char* q, unsigned char* sc,

@@ -276,6 +276,23 @@ at::Tensor gemm_fp8(const at::Tensor& A, const at::Tensor& Bt, double descale,
   return C;
 }
 
+at::Tensor maxpool2d_nhwc(const at::Tensor& x, int64_t k, int64_t stride,
+                          int64_t pad) {
+  CHECK_DEV(x);
+  TORCH_CHECK(x.scalar_type() == at::kFloat && x.dim() == 4 &&
+              x.size(3) % 4 == 0, "maxpool2d_nhwc expects f32 NHWC, C%4==0");
+  const int64_t N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+  const int64_t HO = (H + 2 * pad - k) / stride + 1;
+  const int64_t WO = (W + 2 * pad - k) / stride + 1;
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
+  auto out = at::empty({N, HO, WO, C}, x.options());
+  check_hip(lpx::maxpool2d_nhwc_f32(x.data_ptr<float>(), out.data_ptr<float>(),
+                                    (int)N, (int)H, (int)W, (int)C, (int)k,
+                                    (int)stride, (int)pad, current_stream(x)),
+            "maxpool2d_nhwc_f32");
+  return out;
+}
+
 at::Tensor gemm_fp8_q8(const at::Tensor& A, const at::Tensor& Bt,
                        double descale, const c10::optional<at::Tensor>& bias,
                        bool relu, double out_scale) {
@@ -613,6 +630,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_fp8", &gemm_fp8, "fp8 e4m3 MFMA GEMM with fused descale/bias/relu",
         py::arg("A"), py::arg("Bt"), py::arg("descale") = 1.0,
         py::arg("bias") = py::none(), py::arg("relu") = false);
+  m.def("maxpool2d_nhwc", &maxpool2d_nhwc, "NHWC f32 max pool",
+        py::arg("x"), py::arg("k"), py::arg("stride"), py::arg("pad"));
   m.def("gemm_fp8_q8", &gemm_fp8_q8,
         "fp8 GEMM with fused epilogue quantization (e4m3 out)", py::arg("A"),
         py::arg("Bt"), py::arg("descale"), py::arg("bias") = c10::nullopt,

@@ -179,8 +179,8 @@ class ResNet50FP8:
         """x: [N, 3, 224, 224] f32 (NCHW input, converted to NHWC inside)."""
         x = x.permute(0, 2, 3, 1).contiguous()  # NHWC
         x = self.conv1(x)
-        # maxpool 3x3 s2 p1 (torch elementwise op, NCHW round trip)
-        x = F.max_pool2d(x.permute(0, 3, 1, 2), 3, 2, 1).permute(0, 2, 3, 1)
+        # maxpool 3x3 s2 p1: native NHWC kernel (no NCHW permute round trip)
+        x = _ops.ext().maxpool2d_nhwc(x.contiguous(), 3, 2, 1)
         x = x.contiguous()
         for b in self.blocks:
             x = b(x)

@@ -376,6 +376,16 @@ def test_gemm_fp8_scaled_mfma_path(ops):
     assert (C2 - ref2).abs().max().item() < 1e-2
 
 
+def test_maxpool_nhwc_matches_torch(ops):
+    e = ops.ext()
+    g = torch.Generator(device="cpu").manual_seed(16)
+    x = torch.randn(2, 13, 13, 32, generator=g).to(DEV)
+    got = e.maxpool2d_nhwc(x, 3, 2, 1)
+    import torch.nn.functional as F
+    ref = F.max_pool2d(x.permute(0, 3, 1, 2), 3, 2, 1).permute(0, 2, 3, 1)
+    assert torch.equal(got, ref)
+
+
 def test_gemm_fp8_q8_matches_two_step(ops):
     e = ops.ext()
     g = torch.Generator(device="cpu").manual_seed(14)
