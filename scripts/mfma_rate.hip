@@ -81,6 +81,42 @@ __global__ __launch_bounds__(512) void rate_plain(float* out, int iters) {
   if (s != 0.f) out[0] = s;
 }
 
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+
+__global__ __launch_bounds__(512) void rate_bf16_16(float* out, int iters) {
+  bf16x8 a = {0, 0, 0, 0, 0, 0, 0, 0}, b = a;
+  f32x4v acc[8];
+#pragma unroll
+  for (int i = 0; i < 8; ++i) acc[i] = {0.f, 0.f, 0.f, 0.f};
+  for (int it = 0; it < iters; ++it) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+      acc[i] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[i], 0, 0, 0);
+  }
+  float s = 0.f;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) s += acc[i][0];
+  if (s != 0.f) out[0] = s;
+}
+
+__global__ __launch_bounds__(512) void rate_bf16_32(float* out, int iters) {
+  bf16x8 a = {0, 0, 0, 0, 0, 0, 0, 0}, b = a;
+  f32x16 acc[8];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int e = 0; e < 16; ++e) acc[i][e] = 0.f;
+  for (int it = 0; it < iters; ++it) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+      acc[i] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a, b, acc[i], 0, 0, 0);
+  }
+  float s = 0.f;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) s += acc[i][0];
+  if (s != 0.f) out[0] = s;
+}
+
 extern "C" int run_rate(int which, int blocks, int iters, float* ms_out) {
   float* d;
   if (hipMalloc(&d, 4) != hipSuccess) return 1;
@@ -91,7 +127,9 @@ extern "C" int run_rate(int which, int blocks, int iters, float* ms_out) {
     if (which == 0) rate_scaled<<<blocks, 512>>>(d, n);
     else if (which == 1) rate_plain<<<blocks, 512>>>(d, n);
     else if (which == 2) rate_scaled_fp4<<<blocks, 512>>>(d, n);
-    else rate_scaled_16x128<<<blocks, 512>>>(d, n);
+    else if (which == 3) rate_scaled_16x128<<<blocks, 512>>>(d, n);
+    else if (which == 4) rate_bf16_16<<<blocks, 512>>>(d, n);
+    else rate_bf16_32<<<blocks, 512>>>(d, n);
   };
   launch(iters / 10 + 1);
   hipDeviceSynchronize();
