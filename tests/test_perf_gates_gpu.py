@@ -49,6 +49,53 @@ def test_rf_traversal_floor():
     assert ms > 140, f"RF traversal at {ms:.0f} Msamples/s (floor 140; measured 243)"
 
 
+def test_gemm_fp8_scaled_floor():
+    from lambdipy_amd import ops
+
+    e = ops.ext()
+    n = 4096
+    a = torch.randn(n, n, device=DEV)
+    aq = e.quantize_fp8(a, 1.0)
+    dt = _time_op(lambda: e.gemm_fp8(aq, aq, 1.0))
+    tf = 2 * n**3 / dt / 1e12
+    assert tf > 900, f"fp8 GEMM at {tf:.0f} TF (floor 900; measured ~1600)"
+
+
+def test_gemm_mx_floor():
+    from lambdipy_amd import ops
+
+    e = ops.ext()
+    n = 4096
+    a = torch.randn(n, n, device=DEV)
+    qa, sa = e.quantize_mx(a)
+    dt = _time_op(lambda: e.gemm_mx(qa, sa, qa, sa))
+    tf = 2 * n**3 / dt / 1e12
+    assert tf > 1000, f"MX fp8 GEMM at {tf:.0f} TF (floor 1000; measured ~1780)"
+
+
+def test_gemm_mx_fp4_floor():
+    from lambdipy_amd import ops
+
+    e = ops.ext()
+    n = 4096
+    a = torch.randn(n, n, device=DEV)
+    qa, sa = e.quantize_mx4(a)
+    dt = _time_op(lambda: e.gemm_mx4(qa, sa, qa, sa))
+    tf = 2 * n**3 / dt / 1e12
+    assert tf > 1300, f"MX fp4 GEMM at {tf:.0f} TF (floor 1300; measured ~2300)"
+
+
+def test_resnet50_fp8_floor():
+    from lambdipy_amd.shims.lpx_torch import ResNet50FP8
+
+    m = ResNet50FP8(device=DEV, seed=0)
+    x = torch.randn(64, 3, 224, 224, device=DEV)
+    m(x)  # calibration
+    dt = _time_op(lambda: m(x), iters=5, warmup=2)
+    ips = 64 / dt
+    assert ips > 7000, f"ResNet-50 fp8 at {ips:.0f} img/s (floor 7000; measured ~11600)"
+
+
 def test_gemv_bw_floor():
     from lambdipy_amd import ops
 
