@@ -75,6 +75,24 @@ def main():
     for _ in range(5):
         ops.pairwise_sqeuclidean(P, Q)
     torch.cuda.synchronize()
+
+    # MX fp8 + fp4 GEMMs (scaled MFMA) and the hw quantizers
+    Af = torch.randn(4096, 4096, device=dev)
+    qa, sa = e.quantize_mx(Af)
+    for _ in range(10):
+        e.gemm_mx(qa, sa, qa, sa)
+    qa4, sa4 = e.quantize_mx4(Af)
+    for _ in range(10):
+        e.gemm_mx4(qa4, sa4, qa4, sa4)
+    torch.cuda.synchronize()
+
+    # ResNet-50 fp8 end-to-end (fused quantize chains + NHWC maxpool)
+    from lambdipy_amd.shims.lpx_torch import ResNet50FP8
+    m = ResNet50FP8(device=dev, seed=0)
+    xr = torch.randn(64, 3, 224, 224, device=dev)
+    for _ in range(3):
+        m(xr)
+    torch.cuda.synchronize()
     print("profile workload done")
 
 
