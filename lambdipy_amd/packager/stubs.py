@@ -27,8 +27,11 @@ _SYM_RE = re.compile(
     r"(GLOBAL|WEAK)\s+(DEFAULT|PROTECTED)\s+(\d+|[A-Z]+)\s+(\S+)")
 
 
-def exported_symbols(so_path: str) -> List[Tuple[str, str, int]]:
-    """(name, type, size) for dynamic symbols DEFINED by the DSO."""
+def exported_symbols(so_path: str) -> List[Tuple[str, str, int, str]]:
+    """(name, type, size, version) for dynamic symbols DEFINED by the DSO.
+
+    version is '' for unversioned symbols; importers that reference
+    name@VERSION need the stub to re-define the same version node."""
     out = subprocess.run(["readelf", "-W", "--dyn-syms", so_path],
                          capture_output=True, text=True, check=True).stdout
     syms = []
@@ -39,15 +42,17 @@ def exported_symbols(so_path: str) -> List[Tuple[str, str, int]]:
         size, typ, _bind, _vis, shndx, name = m.groups()
         if shndx == "UND":
             continue  # imported, not exported
+        ver = ""
         if "@" in name:
-            name = name.split("@")[0]
-        syms.append((name, typ, int(size)))
-    # dedupe (versioned aliases)
-    seen = {}
-    for name, typ, size in syms:
-        if name not in seen or seen[name][1] < size:
-            seen[name] = (typ, size)
-    return [(n, t, s) for n, (t, s) in sorted(seen.items())]
+            name, _, ver = name.partition("@")
+            ver = ver.lstrip("@")
+        syms.append((name, typ, int(size), ver))
+    # dedupe (keep the default-versioned / larger entry)
+    seen: Dict[str, Tuple[str, int, str]] = {}
+    for name, typ, size, ver in syms:
+        if name not in seen or seen[name][1] < size or (ver and not seen[name][2]):
+            seen[name] = (typ, size, ver)
+    return [(n, t, s, v) for n, (t, s, v) in sorted(seen.items())]
 
 
 def soname_of(so_path: str) -> Optional[str]:
@@ -72,15 +77,16 @@ def make_stub(src_so: str, out_so: str, cc: str = "gcc") -> Dict:
     """
     syms = exported_symbols(src_so)
     soname = soname_of(src_so) or os.path.basename(src_so)
+    kept = [s for s in syms
+            if re.fullmatch(r"[A-Za-z_$.][A-Za-z0-9_$.]*", s[0])]
+    versions: Dict[str, List[str]] = {}
     with tempfile.TemporaryDirectory(prefix="lpx_stub_") as tmp:
         csrc = os.path.join(tmp, "stub.c")
         with open(csrc, "w") as f:
             f.write("__attribute__((noreturn)) static void lpx_stub_trap(void)"
                     "{__builtin_trap();}\n")
-            for name, typ, size in syms:
-                if not re.fullmatch(r"[A-Za-z_$.][A-Za-z0-9_$.]*", name):
-                    continue  # skip unrepresentable names
-                cname = name.replace(".", "_dot_").replace("$", "_dlr_")
+            for i, (name, typ, size, ver) in enumerate(kept):
+                cname = f"lpx_s{i}"
                 if typ == "OBJECT" or typ == "TLS":
                     sz = max(int(size), 1)
                     f.write(f'char {cname}[{sz}] __asm__("{name}") '
@@ -88,9 +94,24 @@ def make_stub(src_so: str, out_so: str, cc: str = "gcc") -> Dict:
                 else:
                     f.write(f'void {cname}(void) __asm__("{name}");\n'
                             f'void {cname}(void) {{ lpx_stub_trap(); }}\n')
-        subprocess.run([cc, "-shared", "-fPIC", "-O0", "-fno-asynchronous-unwind-tables",
-                        f"-Wl,-soname,{soname}", "-o", out_so, csrc],
-                       check=True, capture_output=True)
+                if ver:
+                    versions.setdefault(ver, []).append(name)
+        cmd = [cc, "-shared", "-fPIC", "-O0",
+               "-fno-asynchronous-unwind-tables",
+               f"-Wl,-soname,{soname}", "-o", out_so, csrc]
+        if versions:
+            # replicate the original version nodes so name@VERSION refs
+            # resolve (ld.so asserts otherwise)
+            vs = os.path.join(tmp, "stub.ver")
+            with open(vs, "w") as f:
+                for ver, names in versions.items():
+                    f.write(ver + " {\nglobal:\n")
+                    for n in names:
+                        f.write(f'  "{n}";\n')
+                    f.write("};\n")
+            cmd.insert(-2, f"-Wl,--version-script={vs}")
+        subprocess.run(cmd, check=True, capture_output=True)
     return {"src": src_so, "out": out_so, "soname": soname,
-            "symbols": len(syms), "old_size": os.path.getsize(src_so),
+            "symbols": len(kept), "versions": sorted(versions),
+            "old_size": os.path.getsize(src_so),
             "new_size": os.path.getsize(out_so)}
