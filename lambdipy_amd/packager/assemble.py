@@ -198,6 +198,28 @@ def assemble_layer(
         report.stripped_files += sn
         report.stripped_bytes += sb
 
+    # stub out never-called native dependencies (e.g. libmagma in a torch
+    # layer: 1.3 GB -> 4 MB; the loader only needs the symbol set)
+    stub_names = set()
+    for d in resolution.recipe_hits:
+        r = registry.get(d.recipe)
+        if r:
+            stub_names.update(r.stub_native)
+    if stub_names:
+        from .stubs import make_stub
+
+        for root, _dirs, files in os.walk(build_dir):
+            for fn in files:
+                if fn in stub_names:
+                    p = os.path.join(root, fn)
+                    try:
+                        st = make_stub(p, p)
+                    except Exception as exc:  # noqa: BLE001
+                        log.warning("stub skipped %s: %s", p, exc)
+                        continue
+                    report.stripped_files += 1
+                    report.stripped_bytes += st["old_size"] - st["new_size"]
+
     # strip non-gfx950 fat-binary entries (HIP multi-arch DSOs)
     if any(registry.get(d.recipe).strip_archs for d in resolution.recipe_hits if registry.get(d.recipe)):
         from .fatbin import strip_fatbin

@@ -40,8 +40,8 @@ def exported_symbols(so_path: str) -> List[Tuple[str, str, int, str]]:
         if not m:
             continue
         size, typ, _bind, _vis, shndx, name = m.groups()
-        if shndx == "UND":
-            continue  # imported, not exported
+        if shndx in ("UND", "ABS"):
+            continue  # imported / version-definition pseudo-symbols
         ver = ""
         if "@" in name:
             name, _, ver = name.partition("@")
@@ -97,8 +97,7 @@ def make_stub(src_so: str, out_so: str, cc: str = "gcc") -> Dict:
                 if ver:
                     versions.setdefault(ver, []).append(name)
         cmd = [cc, "-shared", "-fPIC", "-O0",
-               "-fno-asynchronous-unwind-tables",
-               f"-Wl,-soname,{soname}", "-o", out_so, csrc]
+               "-fno-asynchronous-unwind-tables", f"-Wl,-soname,{soname}"]
         if versions:
             # replicate the original version nodes so name@VERSION refs
             # resolve (ld.so asserts otherwise)
@@ -109,7 +108,8 @@ def make_stub(src_so: str, out_so: str, cc: str = "gcc") -> Dict:
                     for n in names:
                         f.write(f'  "{n}";\n')
                     f.write("};\n")
-            cmd.insert(-2, f"-Wl,--version-script={vs}")
+            cmd.append(f"-Wl,--version-script={vs}")
+        cmd += ["-o", out_so, csrc]
         subprocess.run(cmd, check=True, capture_output=True)
     return {"src": src_so, "out": out_so, "soname": soname,
             "symbols": len(kept), "versions": sorted(versions),

@@ -61,6 +61,7 @@ class BuildRecipe:
     prune: List[str] = field(default_factory=list)
     strip_native: bool = True
     strip_archs: bool = False  # strip non-gfx950 HIP fat-binary entries
+    stub_native: List[str] = field(default_factory=list)  # deps to stub out
     vendor_upstream: bool = True  # False: upstream pkg is runtime-provided
     runtime_requires: List[str] = field(default_factory=list)
     description: str = ""
@@ -90,6 +91,7 @@ class BuildRecipe:
             prune=list(d.get("prune", [])),
             strip_native=bool(d.get("strip_native", True)),
             strip_archs=bool(d.get("strip_archs", False)),
+            stub_native=list(d.get("stub_native", [])),
             vendor_upstream=bool(d.get("vendor_upstream", True)),
             runtime_requires=list(d.get("runtime_requires", [])),
             description=d.get("description", ""),
