@@ -206,14 +206,14 @@ def assemble_layer(
         if r:
             stub_names.update(r.stub_native)
     if stub_names:
-        from .stubs import make_stub
+        from .stubs import DEFAULT_BENIGN, make_stub
 
         for root, _dirs, files in os.walk(build_dir):
             for fn in files:
                 if fn in stub_names:
                     p = os.path.join(root, fn)
                     try:
-                        st = make_stub(p, p)
+                        st = make_stub(p, p, benign=DEFAULT_BENIGN.get(fn, ()))
                     except Exception as exc:  # noqa: BLE001
                         log.warning("stub skipped %s: %s", p, exc)
                         continue

@@ -51,6 +51,7 @@ RAW_MAGIC = b"__CLANG_OFFLOAD_BUNDLE__"
 CCOB_MAGIC = b"CCOB"
 PAGE = 4096
 PT_LOAD, PT_NOTE = 1, 4
+R_X86_64_64 = 1
 R_X86_64_RELATIVE = 8
 
 
@@ -216,22 +217,44 @@ def strip_fatbin(src: str, dst: Optional[str] = None,
         return stats
 
     # 1. wrappers -> bundle vaddrs. The in-file ptr field may be zero (set
-    # only at load time), so prefer the RELATIVE reloc addend targeting it.
+    # only at load time), so resolve via the reloc targeting it: either a
+    # RELATIVE addend (libtorch_hip style) or an R_X86_64_64 against a
+    # dynsym whose st_value is the bundle vaddr (librccl style).
     rela = e.section(".rela.dyn")
+    dynsym = e.section(".dynsym")
     rel_by_off: Dict[int, int] = {}
+    sym_by_off: Dict[int, Tuple[int, int]] = {}  # ptr-field vaddr -> (sym, add)
     if rela is not None:
         for off in range(rela.offset, rela.offset + rela.size, 24):
             r_offset, r_info, r_addend = struct.unpack("<QQQ", data[off:off + 24])
-            if (r_info & 0xffffffff) == R_X86_64_RELATIVE:
+            rtype = r_info & 0xffffffff
+            if rtype == R_X86_64_RELATIVE:
                 rel_by_off[r_offset] = r_addend
+            elif rtype == R_X86_64_64:
+                sym_by_off[r_offset] = (r_info >> 32, r_addend)
+
+    def sym_value(idx: int) -> int:
+        off = dynsym.offset + idx * 24
+        return struct.unpack("<Q", data[off + 8:off + 16])[0]
+
+    def set_sym_value(idx: int, v: int) -> None:
+        struct.pack_into("<Q", data, dynsym.offset + idx * 24 + 8, v)
+
     seg_off = _vaddr_to_off(e, seg.addr)
     wrappers = []  # (file_off_of_ptr_field, bundle_vaddr)
+    wrapper_syms: Dict[int, int] = {}  # bundle vaddr -> dynsym index
     for i in range(seg.size // 24):
         p = seg_off + i * 24
         magic, _ver, ptr = struct.unpack("<IIQ", data[p:p + 16])
         if magic != WRAPPER_MAGIC:
             raise ValueError(f"bad wrapper magic at {p:#x}")
-        ptr = rel_by_off.get(seg.addr + i * 24 + 8, ptr)
+        field_va = seg.addr + i * 24 + 8
+        if field_va in rel_by_off:
+            ptr = rel_by_off[field_va]
+        elif field_va in sym_by_off and dynsym is not None:
+            sidx, add = sym_by_off[field_va]
+            ptr = sym_value(sidx) + add
+            wrapper_syms[ptr - add] = sidx
         wrappers.append((p + 8, ptr))
     stats["bundles"] = len(wrappers)
 
@@ -281,12 +304,16 @@ def strip_fatbin(src: str, dst: Optional[str] = None,
     for ptr_field_off, va in wrappers:
         struct.pack_into("<Q", data, ptr_field_off, layout[va])
 
-    # 5. patch RELATIVE reloc addends that point at old bundle starts
+    # 5. patch RELATIVE reloc addends that point at old bundle starts, and
+    # dynsym st_values for symbol-addressed bundles (R_X86_64_64 style)
     if rela is not None:
         for off in range(rela.offset, rela.offset + rela.size, 24):
             r_offset, r_info, r_addend = struct.unpack("<QQQ", data[off:off + 24])
             if (r_info & 0xffffffff) == R_X86_64_RELATIVE and r_addend in layout:
                 struct.pack_into("<Q", data, off + 16, layout[r_addend])
+    for old_va, sidx in wrapper_syms.items():
+        if old_va in layout:
+            set_sym_value(sidx, layout[old_va])
 
     # 6. splice: [fat_off, fat_off+keep) := out; drop the next `delta` bytes.
     # The segment split needs a spare phdr slot (we drop one PT_NOTE). If

@@ -22,6 +22,12 @@ import subprocess
 import tempfile
 from typing import Dict, List, Optional, Tuple
 
+# init/teardown hooks the host library calls unconditionally at startup:
+# these must succeed (return 0) rather than trap.
+DEFAULT_BENIGN: Dict[str, Tuple[str, ...]] = {
+    "libmagma.so": ("magma_init", "magma_finalize"),
+}
+
 _SYM_RE = re.compile(
     r"^\s*\d+:\s+[0-9a-f]+\s+(\d+)\s+(FUNC|OBJECT|NOTYPE|GNU_IFUNC|TLS)\s+"
     r"(GLOBAL|WEAK)\s+(DEFAULT|PROTECTED)\s+(\d+|[A-Z]+)\s+(\S+)")
@@ -68,12 +74,15 @@ def needed_of(so_path: str) -> List[str]:
     return re.findall(r"NEEDED\)?\s+Shared library: \[([^\]]+)\]", out)
 
 
-def make_stub(src_so: str, out_so: str, cc: str = "gcc") -> Dict:
+def make_stub(src_so: str, out_so: str, cc: str = "gcc",
+              benign: Tuple[str, ...] = ()) -> Dict:
     """Generate a stub .so exporting src_so's symbol set under its SONAME.
 
     Functions trap when called (the layer is built on the premise that the
     dependency is never exercised); objects become zeroed storage of the
-    original size so data references stay in-bounds.
+    original size so data references stay in-bounds. Symbols in `benign`
+    return 0 instead of trapping — for init/finalize hooks the host calls
+    unconditionally (e.g. torch's CUDA init calls magma_init()).
     """
     syms = exported_symbols(src_so)
     soname = soname_of(src_so) or os.path.basename(src_so)
@@ -91,6 +100,9 @@ def make_stub(src_so: str, out_so: str, cc: str = "gcc") -> Dict:
                     sz = max(int(size), 1)
                     f.write(f'char {cname}[{sz}] __asm__("{name}") '
                             f'__attribute__((aligned(16)));\n')
+                elif name in benign:
+                    f.write(f'long {cname}(void) __asm__("{name}");\n'
+                            f'long {cname}(void) {{ return 0; }}\n')
                 else:
                     f.write(f'void {cname}(void) __asm__("{name}");\n'
                             f'void {cname}(void) {{ lpx_stub_trap(); }}\n')
