@@ -288,6 +288,201 @@ __global__ __launch_bounds__(THREADS4, 2) void gemm_mx4_8ph_kernel(
   }
 }
 
+// ---- BK=256 (wide) variant -----------------------------------------------
+// 128-B rows = 256 elements per k-tile: double the MFMA work per barrier
+// pair (8 scaled fp4 MFMAs/phase/wave — the same compute-per-phase balance
+// that takes the unit-scale fp8 kernel to 2.2 PF). LDS 136 KiB -> 1
+// block/CU. Cell rotation (row>>1)&7 is the fp8-MX one (b128-quarter
+// conflict-free for 128-B rows).
+
+constexpr int BK4W = 256;              // elements; 128 B per row
+constexpr int HALF4W = 128 * 128;      // bytes per half-tile
+constexpr int PARITY4W = 4 * HALF4W;   // 64 KiB
+constexpr int SC_TILE4W = 4096;        // [A 256x8 B | B 256x8 B]
+constexpr int SC_BASE4W = 2 * PARITY4W;
+
+__device__ __forceinline__ int phys_cellw(int row, int cell) {
+  return row * 8 + ((cell + (row >> 1)) & 7);
+}
+__device__ __forceinline__ int src_cellw(int q) {
+  const int row = q >> 3;
+  return ((q & 7) - (row >> 1)) & 7;
+}
+
+template <int N>
+__device__ __forceinline__ void waitvmw() {
+  if constexpr (N == 0) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  else if constexpr (N == 6) asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+  else if constexpr (N == 10) asm volatile("s_waitcnt vmcnt(10)" ::: "memory");
+  else static_assert(N == 0, "unsupported vmcnt");
+}
+
+__device__ __forceinline__ void stage_halfw(
+    const unsigned char* __restrict__ A, const unsigned char* __restrict__ Bt,
+    long Kb, int bm, int bn, int tile, int hh, int ksteps,
+    char* __restrict__ smem, int w, int lane) {
+  const int tc = tile < ksteps ? tile : ksteps - 1;
+  const long k0 = (long)tc * 128;  // bytes per BK4W tile
+  const unsigned char* src = (hh < 2) ? A : Bt;
+  const int row0 = ((hh < 2) ? bm : bn) + (hh & 1) * 128;
+  char* const base = smem + (size_t)(tile & 1) * PARITY4W +
+                     (size_t)hh * HALF4W;
+#pragma unroll
+  for (int j = 0; j < 2; ++j) {
+    const int qbase = (j * 8 + w) * 64;
+    const int q = qbase + lane;
+    const int r = q >> 3, cell = src_cellw(q);
+    const unsigned char* g = src + (long)(row0 + r) * Kb + (k0 + cell * 16);
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)g,
+        (__attribute__((address_space(3))) void*)(base + qbase * 16), 16, 0, 0);
+  }
+}
+
+// the tile's 4-KB scale block: 2 x 4-B glds per thread
+__device__ __forceinline__ void stage_scalesw(
+    const unsigned char* __restrict__ sA, const unsigned char* __restrict__ sB,
+    int groups, int bm, int bn, int tile, int ksteps, char* __restrict__ smem,
+    int w, int lane) {
+  const int tc = tile < ksteps ? tile : ksteps - 1;
+  const long scstride = (long)groups * 2;
+  char* const base = smem + SC_BASE4W + (size_t)(tile & 1) * SC_TILE4W;
+#pragma unroll
+  for (int j = 0; j < 2; ++j) {
+    const int d = j * 512 + w * 64 + lane;  // dword 0..1023
+    const int side = d >> 9;                // 0 = A, 1 = B
+    const int r = (d & 511) >> 1, dj = d & 1;
+    const unsigned char* g =
+        (side == 0 ? sA + (long)(bm + r) * scstride
+                   : sB + (long)(bn + r) * scstride) +
+        (long)tc * 8 + dj * 4;
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)g,
+        (__attribute__((address_space(3))) void*)(base + (j * 8 + w) * 256),
+        4, 0, 0);
+  }
+}
+
+__device__ __forceinline__ i32x8v fragw(const char* __restrict__ smem,
+                                        int parity, int hh, int row_in_half,
+                                        int kg, int fkb) {
+  const char* half = smem + (size_t)parity * PARITY4W + (size_t)hh * HALF4W;
+  const i32x4 v = *reinterpret_cast<const i32x4*>(
+      half + phys_cellw(row_in_half, kg * 2 + fkb) * 16);
+  return i32x8v{v.x, v.y, v.z, v.w, 0, 0, 0, 0};
+}
+
+__device__ __forceinline__ int scalew(const char* __restrict__ smem,
+                                      int parity, int hh, int row_in_half,
+                                      int kg, int fkb) {
+  const int row = (hh < 2 ? 0 : 2048) + ((hh & 1) * 128 + row_in_half) * 8;
+  return (int)*reinterpret_cast<const unsigned char*>(
+      smem + SC_BASE4W + (size_t)parity * SC_TILE4W + row + kg * 2 + fkb);
+}
+
+__global__ __launch_bounds__(THREADS4, 1) void gemm_mx4w_8ph_kernel(
+    const unsigned char* __restrict__ A, const unsigned char* __restrict__ sA,
+    const unsigned char* __restrict__ Bt, const unsigned char* __restrict__ sB,
+    float* __restrict__ C, int M, int N, int K, unsigned tiles_m,
+    unsigned tiles_n) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+
+  const unsigned nwg = tiles_m * tiles_n;
+  const unsigned bid = lpx_xcd_swizzle(blockIdx.x, nwg);
+  const int bm = (int)(bid / tiles_n) * BM4;
+  const int bn = (int)(bid % tiles_n) * BM4;
+
+  const int lane = threadIdx.x & 63;
+  const int w = threadIdx.x >> 6;
+  const int wsr = w >> 2, wsc = w & 3;
+  const int fr = lane & 31;
+  const int fkb = lane >> 5;
+  const long Kb = K >> 1;
+  const int groups = K >> 6;
+
+  f32x16 acc[4][2];
+#pragma unroll
+  for (int p = 0; p < 4; ++p)
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+      for (int e = 0; e < 16; ++e) acc[p][i][e] = 0.f;
+
+  const int ksteps = (K + BK4W - 1) / BK4W;
+
+  stage_halfw(A, Bt, Kb, bm, bn, 0, 0, ksteps, smem, w, lane);
+  stage_halfw(A, Bt, Kb, bm, bn, 0, 2, ksteps, smem, w, lane);
+  stage_halfw(A, Bt, Kb, bm, bn, 0, 3, ksteps, smem, w, lane);
+  stage_halfw(A, Bt, Kb, bm, bn, 0, 1, ksteps, smem, w, lane);
+  stage_scalesw(sA, sB, groups, bm, bn, 0, ksteps, smem, w, lane);
+  stage_halfw(A, Bt, Kb, bm, bn, 1, 0, ksteps, smem, w, lane);
+  stage_halfw(A, Bt, Kb, bm, bn, 1, 2, ksteps, smem, w, lane);
+  waitvmw<6>();  // first 10 of 16 prologue events (tile0 data + scales)
+  __builtin_amdgcn_s_barrier();
+
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);
+
+  for (int t = 0; t < ksteps; ++t) {
+    const int parity = t & 1;
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      const int rh = (p >> 1) & 1;
+      const int ch = p & 1;
+      i32x8v aF[2][4], bF[4];
+      int sa[2][4], sb[4];
+#pragma unroll
+      for (int kg = 0; kg < 4; ++kg) {
+#pragma unroll
+        for (int mf = 0; mf < 2; ++mf) {
+          const int r = wsr * 64 + mf * 32 + fr;
+          aF[mf][kg] = fragw(smem, parity, rh, r, kg, fkb);
+          sa[mf][kg] = scalew(smem, parity, rh, r, kg, fkb);
+        }
+        const int rb = wsc * 32 + fr;
+        bF[kg] = fragw(smem, parity, 2 + ch, rb, kg, fkb);
+        sb[kg] = scalew(smem, parity, 2 + ch, rb, kg, fkb);
+      }
+      int st, sh;
+      stage_slot4(t, p, &st, &sh);
+      stage_halfw(A, Bt, Kb, bm, bn, st, sh, ksteps, smem, w, lane);
+      if (p == 0)
+        stage_scalesw(sA, sB, groups, bm, bn, t + 1, ksteps, smem, w, lane);
+
+      __builtin_amdgcn_s_barrier();
+#pragma unroll
+      for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+        for (int kg = 0; kg < 4; ++kg)
+          acc[p][mf] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+              aF[mf][kg], bF[kg], acc[p][mf], 4, 4, 0, sa[mf][kg], 0, sb[kg]);
+      // events per tile cycle: [4,2,2,2] -> any 4-call window = 10; after
+      // p3 exclude p0(t) (next phase reads tile t+1's scales) -> 6
+      if (p == 3) waitvmw<6>();
+      else waitvmw<10>();
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+  waitvmw<0>();
+
+  const int ocol_l = lane & 31;
+  const int orow_base = 4 * (lane >> 5);
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    const int rh = (p >> 1) & 1, ch = p & 1;
+#pragma unroll
+    for (int mf = 0; mf < 2; ++mf) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int row = bm + rh * 128 + wsr * 64 + mf * 32 + (r & 3) +
+                        8 * (r >> 2) + orow_base;
+        const int col = bn + ch * 128 + wsc * 32 + ocol_l;
+        C[(long)row * N + col] = acc[p][mf][r];
+      }
+    }
+  }
+}
+
 }  // namespace
 
 hipError_t quantize_mx_fp4(const float* x, unsigned char* q, unsigned char* sc,
@@ -310,6 +505,13 @@ hipError_t gemm_mx_fp4_nt8ph(const unsigned char* A, const unsigned char* sA,
                              float* C, int M, int N, int K,
                              hipStream_t stream) {
   const unsigned tiles_m = (unsigned)(M / BM4), tiles_n = (unsigned)(N / BM4);
+  if (K % BK4W == 0) {  // wide (BK=256) kernel: 2x MFMA work per barrier
+    const size_t lds = 2 * PARITY4W + 2 * SC_TILE4W;  // 136 KiB
+    gemm_mx4w_8ph_kernel<<<dim3(tiles_m * tiles_n), dim3(THREADS4), lds,
+                           stream>>>(A, sA, Bt, sB, C, M, N, K, tiles_m,
+                                     tiles_n);
+    return hipGetLastError();
+  }
   const size_t lds = 2 * PARITY4 + 2 * SC_TILE4;  // 68 KiB -> 2 blocks/CU
   gemm_mx4_8ph_kernel<<<dim3(tiles_m * tiles_n), dim3(THREADS4), lds, stream>>>(
       A, sA, Bt, sB, C, M, N, K, tiles_m, tiles_n);
