@@ -506,7 +506,10 @@ hipError_t gemm_mx_fp4_nt8ph(const unsigned char* A, const unsigned char* sA,
                              float* C, int M, int N, int K,
                              hipStream_t stream) {
   const unsigned tiles_m = (unsigned)(M / BM4), tiles_n = (unsigned)(N / BM4);
-  if (K % BK4W == 0) {  // wide (BK=256) kernel: 2x MFMA work per barrier
+  // wide kernel runs 1 block/CU (136 KiB LDS): wins only when K is large
+  // enough to amortize the reduced latency hiding (measured: 2644 vs 2514
+  // TF @8192, but 2237 vs 2314 @4096)
+  if (K % BK4W == 0 && K >= 6144) {
     const size_t lds = 2 * PARITY4W + 2 * SC_TILE4W;  // 136 KiB
    hipLaunchKernelGGL(( gemm_mx4w_8ph_kernel), dim3(dim3(tiles_m * tiles_n)), dim3(dim3(THREADS4)), lds,
                            stream, A, sA, Bt, sB, C, M, N, K, tiles_m,
