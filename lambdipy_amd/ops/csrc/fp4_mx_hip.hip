@@ -179,12 +179,13 @@ __device__ __forceinline__ i32x8v frag4(const char* __restrict__ smem,
   return i32x8v{v.x, v.y, v.z, v.w, 0, 0, 0, 0};
 }
 
-__device__ __forceinline__ int scale4(const char* __restrict__ smem,
-                                      int parity, int hh, int row_in_half,
-                                      int kg, int fkb) {
+// the row's 4 scale bytes as one word (extract (w>>8*(kg*2+fkb))&0xff)
+__device__ __forceinline__ unsigned scaleword4(const char* __restrict__ smem,
+                                               int parity, int hh,
+                                               int row_in_half) {
   const int row = (hh < 2 ? 0 : 1024) + ((hh & 1) * 128 + row_in_half) * 4;
-  return (int)*reinterpret_cast<const unsigned char*>(
-      smem + SC_BASE4 + (size_t)parity * SC_TILE4 + row + kg * 2 + fkb);
+  return *reinterpret_cast<const unsigned*>(
+      smem + SC_BASE4 + (size_t)parity * SC_TILE4 + row);
 }
 
 __global__ __launch_bounds__(THREADS4, 2) void gemm_mx4_8ph_kernel(
@@ -239,17 +240,22 @@ __global__ __launch_bounds__(THREADS4, 2) void gemm_mx4_8ph_kernel(
       const int ch = p & 1;
       i32x8v aF[2][2], bF[2];
       int sa[2][2], sb[2];
+      unsigned aw[2], bw;
+#pragma unroll
+      for (int mf = 0; mf < 2; ++mf)
+        aw[mf] = scaleword4(smem, parity, rh, wsr * 64 + mf * 32 + fr);
+      bw = scaleword4(smem, parity, 2 + ch, wsc * 32 + fr);
 #pragma unroll
       for (int kg = 0; kg < 2; ++kg) {
 #pragma unroll
         for (int mf = 0; mf < 2; ++mf) {
           const int r = wsr * 64 + mf * 32 + fr;
           aF[mf][kg] = frag4(smem, parity, rh, r, kg, fkb);
-          sa[mf][kg] = scale4(smem, parity, rh, r, kg, fkb);
+          sa[mf][kg] = (int)((aw[mf] >> (8 * (kg * 2 + fkb))) & 0xff);
         }
         const int rb = wsc * 32 + fr;
         bF[kg] = frag4(smem, parity, 2 + ch, rb, kg, fkb);
-        sb[kg] = scale4(smem, parity, 2 + ch, rb, kg, fkb);
+        sb[kg] = (int)((bw >> (8 * (kg * 2 + fkb))) & 0xff);
       }
       int st, sh;
       stage_slot4(t, p, &st, &sh);
@@ -373,12 +379,15 @@ __device__ __forceinline__ i32x8v fragw(const char* __restrict__ smem,
   return i32x8v{v.x, v.y, v.z, v.w, 0, 0, 0, 0};
 }
 
-__device__ __forceinline__ int scalew(const char* __restrict__ smem,
-                                      int parity, int hh, int row_in_half,
-                                      int kg, int fkb) {
+// the row's 8 scale bytes ([4 groups][2 blocks]) as a dword pair
+__device__ __forceinline__ void scalewordw(const char* __restrict__ smem,
+                                           int parity, int hh,
+                                           int row_in_half, unsigned out[2]) {
   const int row = (hh < 2 ? 0 : 2048) + ((hh & 1) * 128 + row_in_half) * 8;
-  return (int)*reinterpret_cast<const unsigned char*>(
-      smem + SC_BASE4W + (size_t)parity * SC_TILE4W + row + kg * 2 + fkb);
+  const unsigned* p = reinterpret_cast<const unsigned*>(
+      smem + SC_BASE4W + (size_t)parity * SC_TILE4W + row);
+  out[0] = p[0];
+  out[1] = p[1];
 }
 
 __global__ __launch_bounds__(THREADS4, 1) void gemm_mx4w_8ph_kernel(
@@ -432,17 +441,23 @@ __global__ __launch_bounds__(THREADS4, 1) void gemm_mx4w_8ph_kernel(
       const int ch = p & 1;
       i32x8v aF[2][4], bF[4];
       int sa[2][4], sb[4];
+      unsigned aw[2][2], bw[2];
+#pragma unroll
+      for (int mf = 0; mf < 2; ++mf)
+        scalewordw(smem, parity, rh, wsr * 64 + mf * 32 + fr, aw[mf]);
+      scalewordw(smem, parity, 2 + ch, wsc * 32 + fr, bw);
 #pragma unroll
       for (int kg = 0; kg < 4; ++kg) {
+        const int sh8 = 8 * (((kg & 1) * 2 + fkb));
 #pragma unroll
         for (int mf = 0; mf < 2; ++mf) {
           const int r = wsr * 64 + mf * 32 + fr;
           aF[mf][kg] = fragw(smem, parity, rh, r, kg, fkb);
-          sa[mf][kg] = scalew(smem, parity, rh, r, kg, fkb);
+          sa[mf][kg] = (int)((aw[mf][kg >> 1] >> sh8) & 0xff);
         }
         const int rb = wsc * 32 + fr;
         bF[kg] = fragw(smem, parity, 2 + ch, rb, kg, fkb);
-        sb[kg] = scalew(smem, parity, 2 + ch, rb, kg, fkb);
+        sb[kg] = (int)((bw[kg >> 1] >> sh8) & 0xff);
       }
       int st, sh;
       stage_slot4(t, p, &st, &sh);

@@ -126,12 +126,14 @@ __device__ __forceinline__ i32x8v fragx(const char* __restrict__ smem,
   return i32x8v{lo.x, lo.y, lo.z, lo.w, hi.x, hi.y, hi.z, hi.w};
 }
 
-__device__ __forceinline__ int scalex(const char* __restrict__ smem,
-                                      int parity, int hh, int row_in_half,
-                                      int kg, int fkb) {
+// the row's 4 scale bytes ([2 groups][2 blocks]) as one word; extract with
+// (w >> 8*(kg*2+fkb)) & 0xff — one LDS read instead of 2+ byte reads.
+__device__ __forceinline__ unsigned scaleword(const char* __restrict__ smem,
+                                              int parity, int hh,
+                                              int row_in_half) {
   const int row = (hh < 2 ? 0 : 1024) + ((hh & 1) * 128 + row_in_half) * 4;
-  return (int)*reinterpret_cast<const unsigned char*>(
-      smem + SC_BASE + (size_t)parity * SC_TILE + row + kg * 2 + fkb);
+  return *reinterpret_cast<const unsigned*>(
+      smem + SC_BASE + (size_t)parity * SC_TILE + row);
 }
 
 // SC=true: per-block MX scales from LDS (sA/sB used; descale/bias ignored).
@@ -190,17 +192,24 @@ __global__ __launch_bounds__(THREADSX, 1) void gemm_mx_8ph_kernel(
       const int ch = p & 1;
       i32x8v aF[2][2], bF[2];
       int sa[2][2], sb[2];
+      unsigned aw[2] = {0, 0}, bw = 0;
+      if constexpr (SC) {
+#pragma unroll
+        for (int mf = 0; mf < 2; ++mf)
+          aw[mf] = scaleword(smem, parity, rh, wsr * 64 + mf * 32 + fr);
+        bw = scaleword(smem, parity, 2 + ch, wsc * 32 + fr);
+      }
 #pragma unroll
       for (int kg = 0; kg < 2; ++kg) {
 #pragma unroll
         for (int mf = 0; mf < 2; ++mf) {
           const int r = wsr * 64 + mf * 32 + fr;
           aF[mf][kg] = fragx(smem, parity, rh, r, kg, fkb);
-          sa[mf][kg] = SC ? scalex(smem, parity, rh, r, kg, fkb) : 127;
+          sa[mf][kg] = SC ? (int)((aw[mf] >> (8 * (kg * 2 + fkb))) & 0xff) : 127;
         }
         const int rb = wsc * 32 + fr;
         bF[kg] = fragx(smem, parity, 2 + ch, rb, kg, fkb);
-        sb[kg] = SC ? scalex(smem, parity, 2 + ch, rb, kg, fkb) : 127;
+        sb[kg] = SC ? (int)((bw >> (8 * (kg * 2 + fkb))) & 0xff) : 127;
       }
       int st, sh;
       stage_slotx(t, p, &st, &sh);
