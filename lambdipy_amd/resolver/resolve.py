@@ -207,7 +207,15 @@ def resolve(
             requested_by=tuple(sorted(parents.get(name, set()))),
         )
 
-        if follow_transitive:
+        # A runtime-provided package (recipe vendor_upstream=false, e.g.
+        # torch) brings its own runtime deps with it — do NOT vendor its
+        # transitive closure (torch would otherwise drag 650 MB of
+        # triton-rocm + sympy into the layer).
+        runtime_provided = False
+        if pinned[name].recipe and registry:
+            rec = registry.get(pinned[name].recipe)
+            runtime_provided = bool(rec) and not rec.vendor_upstream
+        if follow_transitive and not runtime_provided:
             raw_reqs = (
                 _wheel_requires(chosen.wheel_path)
                 if chosen.wheel_path
