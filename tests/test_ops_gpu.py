@@ -376,6 +376,25 @@ def test_gemm_fp8_scaled_mfma_path(ops):
     assert (C2 - ref2).abs().max().item() < 1e-2
 
 
+def test_determinism_double_run(ops):
+    """SURVEY §4.2 race check: every kernel is bitwise deterministic across
+    back-to-back runs on identical inputs (no atomics/race in the hot path
+    except the reduction kernels, which are unordered by design and tested
+    for tolerance elsewhere)."""
+    from lambdipy_amd.shims.lpx_sklearn import ForestSoA
+    e = ops.ext()
+    g = torch.Generator(device="cpu").manual_seed(21)
+    a = torch.randn(512, 512, generator=g).to(DEV).to(torch.bfloat16)
+    assert torch.equal(ops.gemm_bf16(a, a), ops.gemm_bf16(a, a))
+    aq = e.quantize_fp8(a.float(), 1.0)
+    assert torch.equal(e.gemm_fp8(aq, aq, 1.0), e.gemm_fp8(aq, aq, 1.0))
+    qa, sa = e.quantize_mx(a.float().contiguous())
+    assert torch.equal(e.gemm_mx(qa, sa, qa, sa), e.gemm_mx(qa, sa, qa, sa))
+    f = ForestSoA.random(32, 8, 16, 4, seed=3, device=DEV)
+    X = torch.randn(10000, 16, generator=g).to(DEV)
+    assert torch.equal(f.predict_proba_t(X), f.predict_proba_t(X))
+
+
 def test_maxpool_nhwc_matches_torch(ops):
     e = ops.ext()
     g = torch.Generator(device="cpu").manual_seed(16)
