@@ -233,6 +233,28 @@ def strip_archs(paths, archs, output, as_json):
         click.echo(json.dumps(rows, indent=1))
 
 
+@main.command("stub-deps")
+@click.argument("paths", nargs=-1, type=click.Path(exists=True))
+@click.option("-o", "--output", default=None,
+              help="output path (single input only; default: in place)")
+def stub_deps(paths, output):
+    """Replace shared libraries with symbol stubs (same SONAME).
+
+    For never-called DT_NEEDED dependencies in a deployment layer (e.g.
+    libmagma.so 1.3 GB -> 4 MB): functions trap if called, data symbols
+    become zeroed storage, version nodes are replicated."""
+    from .packager.stubs import DEFAULT_BENIGN, make_stub
+
+    if output and len(paths) != 1:
+        raise click.ClickException("-o requires exactly one input")
+    for p in paths:
+        base = os.path.basename(p)
+        st = make_stub(p, output if output else p,
+                       benign=DEFAULT_BENIGN.get(base, ()))
+        click.echo(f"{p}: {st['old_size']/1e6:.1f} MB -> {st['new_size']/1e3:.0f} KB "
+                   f"({st['symbols']} symbols, versions={st['versions'] or '-'})")
+
+
 @main.command()
 def recipes():
     """List the recipe registry."""
