@@ -153,7 +153,7 @@ __device__ __forceinline__ void stage_scales4(
   const unsigned char* g =
       (t < 256) ? sA + (long)(bm + t) * scstride + (long)tc * 4
                 : sB + (long)(bn + (t - 256)) * scstride + (long)tc * 4;
-  char* const base = smem + SC_BASE4 + (size_t)(tile & 1) * SC_TILE4 + w * 256;
+  char* const base = smem + SC_BASE4 + (size_t)(tile % 3) * SC_TILE4 + w * 256;
   __builtin_amdgcn_global_load_lds(
       (const __attribute__((address_space(1))) void*)g,
       (__attribute__((address_space(3))) void*)base, 4, 0, 0);
@@ -181,11 +181,11 @@ __device__ __forceinline__ i32x8v frag4(const char* __restrict__ smem,
 
 // the row's 4 scale bytes as one word (extract (w>>8*(kg*2+fkb))&0xff)
 __device__ __forceinline__ unsigned scaleword4(const char* __restrict__ smem,
-                                               int parity, int hh,
+                                               int slot3, int hh,
                                                int row_in_half) {
   const int row = (hh < 2 ? 0 : 1024) + ((hh & 1) * 128 + row_in_half) * 4;
   return *reinterpret_cast<const unsigned*>(
-      smem + SC_BASE4 + (size_t)parity * SC_TILE4 + row);
+      smem + SC_BASE4 + (size_t)slot3 * SC_TILE4 + row);
 }
 
 __global__ __launch_bounds__(THREADS4, 2) void gemm_mx4_8ph_kernel(
@@ -224,9 +224,10 @@ __global__ __launch_bounds__(THREADS4, 2) void gemm_mx4_8ph_kernel(
   stage_half4(A, Bt, Kb, bm, bn, 0, 3, ksteps, smem, t512);
   stage_half4(A, Bt, Kb, bm, bn, 0, 1, ksteps, smem, t512);
   stage_scales4(sA, sB, groups, bm, bn, 0, ksteps, smem, w, lane);
+  stage_scales4(sA, sB, groups, bm, bn, 1, ksteps, smem, w, lane);
   stage_half4(A, Bt, Kb, bm, bn, 1, 0, ksteps, smem, t512);
   stage_half4(A, Bt, Kb, bm, bn, 1, 2, ksteps, smem, t512);
-  waitvm4<2>();
+  waitvm4<3>();  // 9 prologue events; tile0 data (4) + sc0 (5th) required
   __builtin_amdgcn_s_barrier();
 
   if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
@@ -241,10 +242,11 @@ __global__ __launch_bounds__(THREADS4, 2) void gemm_mx4_8ph_kernel(
       i32x8v aF[2][2], bF[2];
       int sa[2][2], sb[2];
       unsigned aw[2], bw;
+      const int slot3 = t % 3;
 #pragma unroll
       for (int mf = 0; mf < 2; ++mf)
-        aw[mf] = scaleword4(smem, parity, rh, wsr * 64 + mf * 32 + fr);
-      bw = scaleword4(smem, parity, 2 + ch, wsc * 32 + fr);
+        aw[mf] = scaleword4(smem, slot3, rh, wsr * 64 + mf * 32 + fr);
+      bw = scaleword4(smem, slot3, 2 + ch, wsc * 32 + fr);
 #pragma unroll
       for (int kg = 0; kg < 2; ++kg) {
 #pragma unroll
@@ -260,8 +262,8 @@ __global__ __launch_bounds__(THREADS4, 2) void gemm_mx4_8ph_kernel(
       int st, sh;
       stage_slot4(t, p, &st, &sh);
       stage_half4(A, Bt, Kb, bm, bn, st, sh, ksteps, smem, t512);
-      if (p == 0)
-        stage_scales4(sA, sB, groups, bm, bn, t + 1, ksteps, smem, w, lane);
+      if (p == 0)  // tri-buffered slot (t+2)%3, never the one being read
+        stage_scales4(sA, sB, groups, bm, bn, t + 2, ksteps, smem, w, lane);
 
       __builtin_amdgcn_s_barrier();
 #pragma unroll
@@ -270,8 +272,7 @@ __global__ __launch_bounds__(THREADS4, 2) void gemm_mx4_8ph_kernel(
         for (int kg = 0; kg < 2; ++kg)
           acc[p][mf] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
               aF[mf][kg], bF[kg], acc[p][mf], 4, 4, 0, sa[mf][kg], 0, sb[kg]);
-      if (p == 3) waitvm4<3>();
-      else waitvm4<5>();
+      waitvm4<5>();  // constant: window [2,1,1,1]; scales 8 calls back
       __builtin_amdgcn_s_barrier();
     }
   }
@@ -353,7 +354,7 @@ __device__ __forceinline__ void stage_scalesw(
     int w, int lane) {
   const int tc = tile < ksteps ? tile : ksteps - 1;
   const long scstride = (long)groups * 2;
-  char* const base = smem + SC_BASE4W + (size_t)(tile & 1) * SC_TILE4W;
+  char* const base = smem + SC_BASE4W + (size_t)(tile % 3) * SC_TILE4W;
 #pragma unroll
   for (int j = 0; j < 2; ++j) {
     const int d = j * 512 + w * 64 + lane;  // dword 0..1023
@@ -381,11 +382,11 @@ __device__ __forceinline__ i32x8v fragw(const char* __restrict__ smem,
 
 // the row's 8 scale bytes ([4 groups][2 blocks]) as a dword pair
 __device__ __forceinline__ void scalewordw(const char* __restrict__ smem,
-                                           int parity, int hh,
+                                           int slot3, int hh,
                                            int row_in_half, unsigned out[2]) {
   const int row = (hh < 2 ? 0 : 2048) + ((hh & 1) * 128 + row_in_half) * 8;
   const unsigned* p = reinterpret_cast<const unsigned*>(
-      smem + SC_BASE4W + (size_t)parity * SC_TILE4W + row);
+      smem + SC_BASE4W + (size_t)slot3 * SC_TILE4W + row);
   out[0] = p[0];
   out[1] = p[1];
 }
@@ -425,9 +426,10 @@ __global__ __launch_bounds__(THREADS4, 1) void gemm_mx4w_8ph_kernel(
   stage_halfw(A, Bt, Kb, bm, bn, 0, 3, ksteps, smem, w, lane);
   stage_halfw(A, Bt, Kb, bm, bn, 0, 1, ksteps, smem, w, lane);
   stage_scalesw(sA, sB, groups, bm, bn, 0, ksteps, smem, w, lane);
+  stage_scalesw(sA, sB, groups, bm, bn, 1, ksteps, smem, w, lane);
   stage_halfw(A, Bt, Kb, bm, bn, 1, 0, ksteps, smem, w, lane);
   stage_halfw(A, Bt, Kb, bm, bn, 1, 2, ksteps, smem, w, lane);
-  waitvmw<6>();  // first 10 of 16 prologue events (tile0 data + scales)
+  waitvmw<6>();  // 16 events; tile0 data (8) + sc0 (10th) required
   __builtin_amdgcn_s_barrier();
 
   if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
@@ -442,10 +444,11 @@ __global__ __launch_bounds__(THREADS4, 1) void gemm_mx4w_8ph_kernel(
       i32x8v aF[2][4], bF[4];
       int sa[2][4], sb[4];
       unsigned aw[2][2], bw[2];
+      const int slot3 = t % 3;
 #pragma unroll
       for (int mf = 0; mf < 2; ++mf)
-        scalewordw(smem, parity, rh, wsr * 64 + mf * 32 + fr, aw[mf]);
-      scalewordw(smem, parity, 2 + ch, wsc * 32 + fr, bw);
+        scalewordw(smem, slot3, rh, wsr * 64 + mf * 32 + fr, aw[mf]);
+      scalewordw(smem, slot3, 2 + ch, wsc * 32 + fr, bw);
 #pragma unroll
       for (int kg = 0; kg < 4; ++kg) {
         const int sh8 = 8 * (((kg & 1) * 2 + fkb));
@@ -462,8 +465,8 @@ __global__ __launch_bounds__(THREADS4, 1) void gemm_mx4w_8ph_kernel(
       int st, sh;
       stage_slot4(t, p, &st, &sh);
       stage_halfw(A, Bt, Kb, bm, bn, st, sh, ksteps, smem, w, lane);
-      if (p == 0)
-        stage_scalesw(sA, sB, groups, bm, bn, t + 1, ksteps, smem, w, lane);
+      if (p == 0)  // tri-buffered slot (t+2)%3
+        stage_scalesw(sA, sB, groups, bm, bn, t + 2, ksteps, smem, w, lane);
 
       __builtin_amdgcn_s_barrier();
 #pragma unroll
@@ -472,10 +475,9 @@ __global__ __launch_bounds__(THREADS4, 1) void gemm_mx4w_8ph_kernel(
         for (int kg = 0; kg < 4; ++kg)
           acc[p][mf] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
               aF[mf][kg], bF[kg], acc[p][mf], 4, 4, 0, sa[mf][kg], 0, sb[kg]);
-      // events per tile cycle: [4,2,2,2] -> any 4-call window = 10; after
-      // p3 exclude p0(t) (next phase reads tile t+1's scales) -> 6
-      if (p == 3) waitvmw<6>();
-      else waitvmw<10>();
+      // events per tile cycle: [4,2,2,2] -> any 4-call window = 10;
+      // tri-buffered scales are 8+ calls back, no tight p3 wait
+      waitvmw<10>();
       __builtin_amdgcn_s_barrier();
     }
   }
@@ -525,13 +527,13 @@ hipError_t gemm_mx_fp4_nt8ph(const unsigned char* A, const unsigned char* sA,
   // enough to amortize the reduced latency hiding (measured: 2644 vs 2514
   // TF @8192, but 2237 vs 2314 @4096)
   if (K % BK4W == 0 && K >= 6144) {
-    const size_t lds = 2 * PARITY4W + 2 * SC_TILE4W;  // 136 KiB
+    const size_t lds = 2 * PARITY4W + 3 * SC_TILE4W;  // 140 KiB
    hipLaunchKernelGGL(( gemm_mx4w_8ph_kernel), dim3(dim3(tiles_m * tiles_n)), dim3(dim3(THREADS4)), lds,
                            stream, A, sA, Bt, sB, C, M, N, K, tiles_m,
                                      tiles_n);
     return hipGetLastError();
   }
-  const size_t lds = 2 * PARITY4 + 2 * SC_TILE4;  // 68 KiB -> 2 blocks/CU
+  const size_t lds = 2 * PARITY4 + 3 * SC_TILE4;  // 70 KiB -> 2 blocks/CU
  hipLaunchKernelGGL(( gemm_mx4_8ph_kernel), dim3(dim3(tiles_m * tiles_n)), dim3(dim3(THREADS4)), lds, stream, 
       A, sA, Bt, sB, C, M, N, K, tiles_m, tiles_n);
   return hipGetLastError();

@@ -33,6 +33,9 @@ constexpr int HALF_BYTES = 128 * 128;              // data bytes per half-tile
 constexpr int PARITY_BYTES = 4 * HALF_BYTES;
 constexpr int SC_TILE = 2048;  // per-tile scale block [A 256 rows | B 256]x4B
 constexpr int SC_BASE = 2 * PARITY_BYTES;
+// scales are TRI-buffered (slot = tile % 3): phase 0 of tile t stages tile
+// t+2's block, so the write slot never aliases the slot being read and
+// every phase keeps the same relaxed vmcnt (no tight p3 wait).
 
 typedef __attribute__((ext_vector_type(8))) int i32x8v;
 
@@ -50,6 +53,7 @@ template <int N>
 __device__ __forceinline__ void waitvm() {
   if constexpr (N == 0) asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   else if constexpr (N == 4) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  else if constexpr (N == 5) asm volatile("s_waitcnt vmcnt(5)" ::: "memory");
   else if constexpr (N == 6) asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
   else if constexpr (N == 8) asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
   else if constexpr (N == 9) asm volatile("s_waitcnt vmcnt(9)" ::: "memory");
@@ -97,7 +101,7 @@ __device__ __forceinline__ void stage_scalesx(
   const unsigned char* g =
       (t < 256) ? sA + (long)(bm + t) * scstride + (long)tc * 4
                 : sB + (long)(bn + (t - 256)) * scstride + (long)tc * 4;
-  char* const base = smem + SC_BASE + (size_t)(tile & 1) * SC_TILE + w * 256;
+  char* const base = smem + SC_BASE + (size_t)(tile % 3) * SC_TILE + w * 256;
   __builtin_amdgcn_global_load_lds(
       (const __attribute__((address_space(1))) void*)g,
       (__attribute__((address_space(3))) void*)base, 4, 0, 0);
@@ -129,11 +133,11 @@ __device__ __forceinline__ i32x8v fragx(const char* __restrict__ smem,
 // the row's 4 scale bytes ([2 groups][2 blocks]) as one word; extract with
 // (w >> 8*(kg*2+fkb)) & 0xff — one LDS read instead of 2+ byte reads.
 __device__ __forceinline__ unsigned scaleword(const char* __restrict__ smem,
-                                              int parity, int hh,
+                                              int slot3, int hh,
                                               int row_in_half) {
   const int row = (hh < 2 ? 0 : 1024) + ((hh & 1) * 128 + row_in_half) * 4;
   return *reinterpret_cast<const unsigned*>(
-      smem + SC_BASE + (size_t)parity * SC_TILE + row);
+      smem + SC_BASE + (size_t)slot3 * SC_TILE + row);
 }
 
 // SC=true: per-block MX scales from LDS (sA/sB used; descale/bias ignored).
@@ -174,11 +178,13 @@ __global__ __launch_bounds__(THREADSX, 1) void gemm_mx_8ph_kernel(
   stage_halfx(A, Bt, K, bm, bn, 0, 2, ksteps, smem, w, lane);
   stage_halfx(A, Bt, K, bm, bn, 0, 3, ksteps, smem, w, lane);
   stage_halfx(A, Bt, K, bm, bn, 0, 1, ksteps, smem, w, lane);
-  if constexpr (SC)
+  if constexpr (SC) {
     stage_scalesx(sA, sB, K, bm, bn, 0, ksteps, smem, w, lane);
+    stage_scalesx(sA, sB, K, bm, bn, 1, ksteps, smem, w, lane);
+  }
   stage_halfx(A, Bt, K, bm, bn, 1, 0, ksteps, smem, w, lane);
   stage_halfx(A, Bt, K, bm, bn, 1, 2, ksteps, smem, w, lane);
-  waitvm<4>();
+  waitvm<SC ? 5 : 4>();
   __builtin_amdgcn_s_barrier();
 
   if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
@@ -194,10 +200,11 @@ __global__ __launch_bounds__(THREADSX, 1) void gemm_mx_8ph_kernel(
       int sa[2][2], sb[2];
       unsigned aw[2] = {0, 0}, bw = 0;
       if constexpr (SC) {
+        const int slot3 = t % 3;
 #pragma unroll
         for (int mf = 0; mf < 2; ++mf)
-          aw[mf] = scaleword(smem, parity, rh, wsr * 64 + mf * 32 + fr);
-        bw = scaleword(smem, parity, 2 + ch, wsc * 32 + fr);
+          aw[mf] = scaleword(smem, slot3, rh, wsr * 64 + mf * 32 + fr);
+        bw = scaleword(smem, slot3, 2 + ch, wsc * 32 + fr);
       }
 #pragma unroll
       for (int kg = 0; kg < 2; ++kg) {
@@ -215,8 +222,9 @@ __global__ __launch_bounds__(THREADSX, 1) void gemm_mx_8ph_kernel(
       stage_slotx(t, p, &st, &sh);
       stage_halfx(A, Bt, K, bm, bn, st, sh, ksteps, smem, w, lane);
       if constexpr (SC)
-        if (p == 0)  // tile t+1's full scale block, 1 glds/thread, no dup
-          stage_scalesx(sA, sB, K, bm, bn, t + 1, ksteps, smem, w, lane);
+        if (p == 0)  // tile t+2's scale block into slot (t+2)%3 (never the
+                     // slot being read), 1 glds/thread, no duplication
+          stage_scalesx(sA, sB, K, bm, bn, t + 2, ksteps, smem, w, lane);
 
       __builtin_amdgcn_s_barrier();
 #pragma unroll
@@ -226,15 +234,11 @@ __global__ __launch_bounds__(THREADSX, 1) void gemm_mx_8ph_kernel(
           acc[p][mf] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
               aF[mf][kg], bF[kg], acc[p][mf], 0, 0, 0, sa[mf][kg], 0, sb[kg]);
       // SC: per-tile-cycle call events are [3,2,2,2]; any 4-call window
-      // holds one phase-0, so 9 is constant — except after p3, where the
-      // next phase reads tile t+1's scales (issued at p0, 4 calls back):
-      // tighten to 6 so that load is complete.
-      if constexpr (SC) {
-        if (p == 3) waitvm<6>();
-        else waitvm<9>();
-      } else {
-        waitvm<8>();
-      }
+      // holds one phase-0, so 9 is constant. Tile t+1's scales were staged
+      // at p0 of t-1 (8 calls back, outside the window) — tri-buffered
+      // slots make that prefetch distance safe.
+      if constexpr (SC) waitvm<9>();
+      else waitvm<8>();
       __builtin_amdgcn_s_barrier();
     }
   }
@@ -276,7 +280,7 @@ hipError_t gemm_mx_fp8_nt8ph(const unsigned char* A, const unsigned char* sA,
                              float* C, int M, int N, int K,
                              hipStream_t stream) {
   const unsigned tiles_m = (unsigned)(M / BMX), tiles_n = (unsigned)(N / BMX);
-  const size_t lds = 2 * PARITY_BYTES + 2 * SC_TILE;  // 132 KiB
+  const size_t lds = 2 * PARITY_BYTES + 3 * SC_TILE;  // 134 KiB
  hipLaunchKernelGGL(( gemm_mx_8ph_kernel<true, false, false>)
       , dim3(dim3(tiles_m * tiles_n)), dim3(dim3(THREADSX)), lds, stream, 
           A, sA, Bt, sB, C, nullptr, nullptr, 1.f, 1.f, M, N, K, tiles_m,
