@@ -69,9 +69,10 @@ def gemv(A: torch.Tensor, x: torch.Tensor) -> torch.Tensor:
 
 def quantize_mx(x: torch.Tensor):
     """f32 [R, K%64==0] -> (e4m3 bytes in MFMA-shuffled order, E8M0 scales
-    [R, K/64, 2]).  One shared 2^e scale per 32-element MX block; block b of
-    group g covers k quarters {b, b+2} (the hardware lane coverage of
-    v_mfma_scale_f32_32x32x64_f8f6f4, decoded in scripts/mx_sweep.py)."""
+    TILE-MAJOR [ceil(K/128), R, 4] so GEMM scale staging is coalesced).
+    One shared 2^e scale per contiguous 32-element MX block (hardware scale
+    coverage of v_mfma_scale_f32_32x32x64_f8f6f4 decoded in
+    scripts/mx_sweep.py)."""
     q, sc = require_extension().quantize_mx(x.contiguous())
     return q, sc
 
@@ -91,8 +92,8 @@ def matmul_mx(A: torch.Tensor, B: torch.Tensor) -> torch.Tensor:
 
 
 def quantize_mx_fp4(x: torch.Tensor):
-    """f32 [R, K%64==0] -> (e2m1 nibbles [R, K/2] in MFMA-shuffled order,
-    E8M0 scales [R, K/64, 2]).  The fp4 MFMA path runs at 2x the fp8 rate."""
+    """f32 [R, K%64==0] -> (e2m1 nibbles [R, K/2], linear order, E8M0
+    scales TILE-MAJOR [ceil(K/128), R, 4]). 2x the fp8 MFMA rate."""
     q, sc = require_extension().quantize_mx4(x.contiguous())
     return q, sc
 

@@ -34,7 +34,8 @@ typedef __attribute__((ext_vector_type(8))) int i32x8v;
 typedef __attribute__((ext_vector_type(2))) short s16x2;
 
 // ---- MX fp4 quantization -------------------------------------------------
-// x f32 [R][K], K % 64 == 0 -> q [R][K/2] linear nibbles + sc [R][K/64][2]
+// x f32 [R][K], K%64==0 -> q [R][K/2] linear nibbles + tile-major scales
+// [ceil(groups/2)][R][4] (contiguous per GEMM k-tile -> coalesced staging)
 __global__ __launch_bounds__(THREADS) void quantize_mx4_kernel(
     const float* __restrict__ x, unsigned char* __restrict__ q,
     unsigned char* __restrict__ sc, int R, int K) {
@@ -65,7 +66,8 @@ __global__ __launch_bounds__(THREADS) void quantize_mx4_kernel(
     }
     if (e < -127) e = -127;
     if (e > 127) e = 127;
-    sc[((long)r * groups + g) * 2 + kb] = (unsigned char)(e + 127);
+    sc[((long)(g >> 1) * R + r) * 4 + (g & 1) * 2 + kb] =
+        (unsigned char)(e + 127);  // tile-major [ceil(groups/2)][R][4]
     const float scl = exp2f((float)e);  // hw cvt divides by scale
     unsigned char* base = q + (long)r * (K >> 1) + g * 32;
 #pragma unroll
@@ -145,14 +147,14 @@ __device__ __forceinline__ void stage_half4(
 
 __device__ __forceinline__ void stage_scales4(
     const unsigned char* __restrict__ sA, const unsigned char* __restrict__ sB,
-    int groups, int bm, int bn, int tile, int ksteps, char* __restrict__ smem,
-    int w, int lane) {
+    int Mtot, int Ntot, int bm, int bn, int tile, int ksteps,
+    char* __restrict__ smem, int w, int lane) {
   const int tc = tile < ksteps ? tile : ksteps - 1;
   const int t = w * 64 + lane;
-  const long scstride = (long)groups * 2;
+  // tile-major scales -> coalesced
   const unsigned char* g =
-      (t < 256) ? sA + (long)(bm + t) * scstride + (long)tc * 4
-                : sB + (long)(bn + (t - 256)) * scstride + (long)tc * 4;
+      (t < 256) ? sA + ((long)tc * Mtot + bm + t) * 4
+                : sB + ((long)tc * Ntot + bn + (t - 256)) * 4;
   char* const base = smem + SC_BASE4 + (size_t)(tile % 3) * SC_TILE4 + w * 256;
   __builtin_amdgcn_global_load_lds(
       (const __attribute__((address_space(1))) void*)g,
@@ -223,8 +225,8 @@ __global__ __launch_bounds__(THREADS4, 2) void gemm_mx4_8ph_kernel(
   stage_half4(A, Bt, Kb, bm, bn, 0, 2, ksteps, smem, t512);
   stage_half4(A, Bt, Kb, bm, bn, 0, 3, ksteps, smem, t512);
   stage_half4(A, Bt, Kb, bm, bn, 0, 1, ksteps, smem, t512);
-  stage_scales4(sA, sB, groups, bm, bn, 0, ksteps, smem, w, lane);
-  stage_scales4(sA, sB, groups, bm, bn, 1, ksteps, smem, w, lane);
+  stage_scales4(sA, sB, M, N, bm, bn, 0, ksteps, smem, w, lane);
+  stage_scales4(sA, sB, M, N, bm, bn, 1, ksteps, smem, w, lane);
   stage_half4(A, Bt, Kb, bm, bn, 1, 0, ksteps, smem, t512);
   stage_half4(A, Bt, Kb, bm, bn, 1, 2, ksteps, smem, t512);
   waitvm4<3>();  // 9 prologue events; tile0 data (4) + sc0 (5th) required
@@ -263,7 +265,7 @@ __global__ __launch_bounds__(THREADS4, 2) void gemm_mx4_8ph_kernel(
       stage_slot4(t, p, &st, &sh);
       stage_half4(A, Bt, Kb, bm, bn, st, sh, ksteps, smem, t512);
       if (p == 0)  // tri-buffered slot (t+2)%3, never the one being read
-        stage_scales4(sA, sB, groups, bm, bn, t + 2, ksteps, smem, w, lane);
+        stage_scales4(sA, sB, M, N, bm, bn, t + 2, ksteps, smem, w, lane);
 
       __builtin_amdgcn_s_barrier();
 #pragma unroll
@@ -350,20 +352,21 @@ __device__ __forceinline__ void stage_halfw(
 // the tile's 4-KB scale block: 2 x 4-B glds per thread
 __device__ __forceinline__ void stage_scalesw(
     const unsigned char* __restrict__ sA, const unsigned char* __restrict__ sB,
-    int groups, int bm, int bn, int tile, int ksteps, char* __restrict__ smem,
-    int w, int lane) {
+    int Mtot, int Ntot, int bm, int bn, int tile, int ksteps,
+    char* __restrict__ smem, int w, int lane) {
   const int tc = tile < ksteps ? tile : ksteps - 1;
-  const long scstride = (long)groups * 2;
   char* const base = smem + SC_BASE4W + (size_t)(tile % 3) * SC_TILE4W;
+  // BK=256 tile = scale-kt pair {2*tc, 2*tc+1}; each dword coalesced in
+  // the tile-major layout. LDS image must stay [row][8B]: dword d of the
+  // 4-KB block = (side, r, dj) with source block kt = 2*tc + dj.
 #pragma unroll
   for (int j = 0; j < 2; ++j) {
     const int d = j * 512 + w * 64 + lane;  // dword 0..1023
     const int side = d >> 9;                // 0 = A, 1 = B
     const int r = (d & 511) >> 1, dj = d & 1;
     const unsigned char* g =
-        (side == 0 ? sA + (long)(bm + r) * scstride
-                   : sB + (long)(bn + r) * scstride) +
-        (long)tc * 8 + dj * 4;
+        (side == 0 ? sA + ((long)(2 * tc + dj) * Mtot + bm + r) * 4
+                   : sB + ((long)(2 * tc + dj) * Ntot + bn + r) * 4);
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) void*)g,
         (__attribute__((address_space(3))) void*)(base + (j * 8 + w) * 256),
@@ -425,8 +428,8 @@ __global__ __launch_bounds__(THREADS4, 1) void gemm_mx4w_8ph_kernel(
   stage_halfw(A, Bt, Kb, bm, bn, 0, 2, ksteps, smem, w, lane);
   stage_halfw(A, Bt, Kb, bm, bn, 0, 3, ksteps, smem, w, lane);
   stage_halfw(A, Bt, Kb, bm, bn, 0, 1, ksteps, smem, w, lane);
-  stage_scalesw(sA, sB, groups, bm, bn, 0, ksteps, smem, w, lane);
-  stage_scalesw(sA, sB, groups, bm, bn, 1, ksteps, smem, w, lane);
+  stage_scalesw(sA, sB, M, N, bm, bn, 0, ksteps, smem, w, lane);
+  stage_scalesw(sA, sB, M, N, bm, bn, 1, ksteps, smem, w, lane);
   stage_halfw(A, Bt, Kb, bm, bn, 1, 0, ksteps, smem, w, lane);
   stage_halfw(A, Bt, Kb, bm, bn, 1, 2, ksteps, smem, w, lane);
   waitvmw<6>();  // 16 events; tile0 data (8) + sc0 (10th) required
@@ -466,7 +469,7 @@ __global__ __launch_bounds__(THREADS4, 1) void gemm_mx4w_8ph_kernel(
       stage_slot4(t, p, &st, &sh);
       stage_halfw(A, Bt, Kb, bm, bn, st, sh, ksteps, smem, w, lane);
       if (p == 0)  // tri-buffered slot (t+2)%3
-        stage_scalesw(sA, sB, groups, bm, bn, t + 2, ksteps, smem, w, lane);
+        stage_scalesw(sA, sB, M, N, bm, bn, t + 2, ksteps, smem, w, lane);
 
       __builtin_amdgcn_s_barrier();
 #pragma unroll

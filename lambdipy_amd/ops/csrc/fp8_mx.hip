@@ -31,7 +31,11 @@ typedef __attribute__((ext_vector_type(2))) short s16x2;
 // ---- MX quantization -----------------------------------------------------
 // x f32 [R][K] row-major, K % 64 == 0. Outputs:
 //   q  u8 [R][K]      shuffled e4m3 payload ([q0|q2|q1|q3] per 64-group)
-//   sc u8 [R][K/64][2] E8M0 block scales (kb = 0: quarters {0,2}; 1: {1,3})
+//   sc E8M0 block scales, TILE-MAJOR [ceil(groups/2)][R][4]: one BK=128
+//   GEMM tile's scale block is contiguous (rows x 4 B), so the GEMM's
+//   scale staging is fully coalesced (row-major scales would cost ~16x
+//   their bytes in L2 lines). Byte (g&1)*2 + kb of row slot r covers
+//   original k [g*64 + kb*32, +32).
 __global__ __launch_bounds__(THREADS) void quantize_mx_kernel(
     const float* __restrict__ x, unsigned char* __restrict__ q,
     unsigned char* __restrict__ sc, int R, int K) {
@@ -65,7 +69,8 @@ __global__ __launch_bounds__(THREADS) void quantize_mx_kernel(
     }
     if (e < -127) e = -127;
     if (e > 127) e = 127;
-    sc[((long)r * groups + g) * 2 + kb] = (unsigned char)(e + 127);
+    sc[((long)(g >> 1) * R + r) * 4 + (g & 1) * 2 + kb] =
+        (unsigned char)(e + 127);
     const float scl = exp2f((float)e);  // hw cvt divides by the scale operand
     unsigned char* base = q + (long)r * K + g * 64;
 #pragma unroll
@@ -138,14 +143,16 @@ __device__ __forceinline__ unsigned stage_sc(const unsigned char* __restrict__ s
   if (t >= 128) return 0;
   const int r = row0 + t * 2;
   unsigned v = 0x7f7f;  // scale 1.0 pair for out-of-range rows
+  const unsigned char* base =
+      sc + ((long)(ks >> 1) * rows_total) * 4 + (ks & 1) * 2;
   if (r + 1 < rows_total) {
     const unsigned short a =
-        *reinterpret_cast<const unsigned short*>(sc + ((long)r * groups + ks) * 2);
-    const unsigned short b = *reinterpret_cast<const unsigned short*>(
-        sc + ((long)(r + 1) * groups + ks) * 2);
+        *reinterpret_cast<const unsigned short*>(base + (long)r * 4);
+    const unsigned short b =
+        *reinterpret_cast<const unsigned short*>(base + (long)(r + 1) * 4);
     v = (unsigned)a | ((unsigned)b << 16);
   } else if (r < rows_total) {
-    v = *reinterpret_cast<const unsigned short*>(sc + ((long)r * groups + ks) * 2);
+    v = *reinterpret_cast<const unsigned short*>(base + (long)r * 4);
   }
   return v;
 }

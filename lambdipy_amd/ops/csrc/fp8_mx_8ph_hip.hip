@@ -93,14 +93,14 @@ __device__ __forceinline__ void stage_halfx(
 // 4-call vmcnt window always contains one scale load (constant counts).
 __device__ __forceinline__ void stage_scalesx(
     const unsigned char* __restrict__ sA, const unsigned char* __restrict__ sB,
-    long K, int bm, int bn, int tile, int ksteps, char* __restrict__ smem,
-    int w, int lane) {
+    int Mtot, int Ntot, int bm, int bn, int tile, int ksteps,
+    char* __restrict__ smem, int w, int lane) {
   const int tc = tile < ksteps ? tile : ksteps - 1;
   const int t = w * 64 + lane;
-  const long scstride = (K >> 6) * 2;  // scale bytes per row
+  // tile-major scales: the tile's block is contiguous -> coalesced glds
   const unsigned char* g =
-      (t < 256) ? sA + (long)(bm + t) * scstride + (long)tc * 4
-                : sB + (long)(bn + (t - 256)) * scstride + (long)tc * 4;
+      (t < 256) ? sA + ((long)tc * Mtot + bm + t) * 4
+                : sB + ((long)tc * Ntot + bn + (t - 256)) * 4;
   char* const base = smem + SC_BASE + (size_t)(tile % 3) * SC_TILE + w * 256;
   __builtin_amdgcn_global_load_lds(
       (const __attribute__((address_space(1))) void*)g,
@@ -179,8 +179,8 @@ __global__ __launch_bounds__(THREADSX, 1) void gemm_mx_8ph_kernel(
   stage_halfx(A, Bt, K, bm, bn, 0, 3, ksteps, smem, w, lane);
   stage_halfx(A, Bt, K, bm, bn, 0, 1, ksteps, smem, w, lane);
   if constexpr (SC) {
-    stage_scalesx(sA, sB, K, bm, bn, 0, ksteps, smem, w, lane);
-    stage_scalesx(sA, sB, K, bm, bn, 1, ksteps, smem, w, lane);
+    stage_scalesx(sA, sB, M, N, bm, bn, 0, ksteps, smem, w, lane);
+    stage_scalesx(sA, sB, M, N, bm, bn, 1, ksteps, smem, w, lane);
   }
   stage_halfx(A, Bt, K, bm, bn, 1, 0, ksteps, smem, w, lane);
   stage_halfx(A, Bt, K, bm, bn, 1, 2, ksteps, smem, w, lane);
@@ -224,7 +224,7 @@ __global__ __launch_bounds__(THREADSX, 1) void gemm_mx_8ph_kernel(
       if constexpr (SC)
         if (p == 0)  // tile t+2's scale block into slot (t+2)%3 (never the
                      // slot being read), 1 glds/thread, no duplication
-          stage_scalesx(sA, sB, K, bm, bn, t + 2, ksteps, smem, w, lane);
+          stage_scalesx(sA, sB, M, N, bm, bn, t + 2, ksteps, smem, w, lane);
 
       __builtin_amdgcn_s_barrier();
 #pragma unroll

@@ -368,7 +368,8 @@ std::vector<at::Tensor> quantize_mx(const at::Tensor& x) {
   const int64_t R = x.size(0), K = x.size(1);
   c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
   auto q = at::empty({R, K}, x.options().dtype(at::kByte));
-  auto sc = at::empty({R, K / 64, 2}, x.options().dtype(at::kByte));
+  // tile-major scales: [ceil(groups/2)][R][4] (coalesced GEMM staging)
+  auto sc = at::empty({(K / 64 + 1) / 2, R, 4}, x.options().dtype(at::kByte));
   check_hip(lpx::quantize_mx_fp8(x.data_ptr<float>(), q.data_ptr<uint8_t>(),
                                  sc.data_ptr<uint8_t>(), (int)R, (int)K,
                                  current_stream(x)),
@@ -388,8 +389,9 @@ at::Tensor gemm_mx(const at::Tensor& A, const at::Tensor& sA,
   TORCH_CHECK(A.dim() == 2 && Bt.dim() == 2 && A.size(1) == Bt.size(1) &&
               A.size(1) % 64 == 0, "gemm_mx expects A [M,K], Bt [N,K], K%64==0");
   const int64_t M = A.size(0), K = A.size(1), N = Bt.size(0);
-  TORCH_CHECK(sA.numel() == M * (K / 64) * 2 && sB.numel() == N * (K / 64) * 2,
-              "scale shapes must be [rows, K/64, 2]");
+  const int64_t kt = (K / 64 + 1) / 2;
+  TORCH_CHECK(sA.numel() == kt * M * 4 && sB.numel() == kt * N * 4,
+              "scale shapes must be [ceil(K/128), rows, 4] (tile-major)");
   c10::hip::HIPGuardMasqueradingAsCUDA guard(A.device());
   auto C = at::empty({M, N}, A.options().dtype(at::kFloat));
   if (lpx::gemm_mx_8ph_supported((int)M, (int)N, (int)K))
@@ -415,7 +417,8 @@ std::vector<at::Tensor> quantize_mx4(const at::Tensor& x) {
   const int64_t R = x.size(0), K = x.size(1);
   c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
   auto q = at::empty({R, K / 2}, x.options().dtype(at::kByte));
-  auto sc = at::empty({R, K / 64, 2}, x.options().dtype(at::kByte));
+  // tile-major scales: [ceil(groups/2)][R][4]
+  auto sc = at::empty({(K / 64 + 1) / 2, R, 4}, x.options().dtype(at::kByte));
   check_hip(lpx::quantize_mx_fp4(x.data_ptr<float>(), q.data_ptr<uint8_t>(),
                                  sc.data_ptr<uint8_t>(), (int)R, (int)K,
                                  current_stream(x)),
@@ -437,8 +440,9 @@ at::Tensor gemm_mx4(const at::Tensor& A, const at::Tensor& sA,
   const int64_t M = A.size(0), K = A.size(1) * 2, N = Bt.size(0);
   TORCH_CHECK(lpx::gemm_mx_fp4_supported((int)M, (int)N, (int)K),
               "gemm_mx4 requires M,N%256==0 and K%128==0");
-  TORCH_CHECK(sA.numel() == M * (K / 64) * 2 && sB.numel() == N * (K / 64) * 2,
-              "scale shapes must be [rows, K/64, 2]");
+  const int64_t kt = (K / 64 + 1) / 2;
+  TORCH_CHECK(sA.numel() == kt * M * 4 && sB.numel() == kt * N * 4,
+              "scale shapes must be [ceil(K/128), rows, 4] (tile-major)");
   c10::hip::HIPGuardMasqueradingAsCUDA guard(A.device());
   auto C = at::empty({M, N}, A.options().dtype(at::kFloat));
   check_hip(lpx::gemm_mx_fp4_nt8ph(A.data_ptr<uint8_t>(), sA.data_ptr<uint8_t>(),

@@ -26,7 +26,9 @@ def dequant(q, sc, K, smap):
     lo = E2M1[(q & 0xF).long()]
     hi = E2M1[(q >> 4).long()]
     d = torch.stack([lo, hi], dim=-1).reshape(R, g, 4, 16)
-    scales = torch.pow(2.0, sc.float() - 127.0)
+    # tile-major sc [KT, R, 4] -> [R, g, 2]
+    s = sc.permute(1, 0, 2).reshape(R, -1, 2, 2).reshape(R, -1, 2)[:, :g]
+    scales = torch.pow(2.0, s.float() - 127.0)
     d = d * scales[:, :, smap].unsqueeze(-1)
     return d[:, :, [0, 2, 1, 3], :].reshape(R, K)
 

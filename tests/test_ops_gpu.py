@@ -244,6 +244,13 @@ def test_gemm_fp8_vs_dequant_ref(ops):
     assert err < 1e-3, err
 
 
+def _mx_scales(sc, R, groups):
+    """Tile-major [KT, R, 4] E8M0 bytes -> [R, groups, 2] scale factors."""
+    s = sc.permute(1, 0, 2).reshape(R, -1, 2, 2)   # [R, KT, ghalf, kb]
+    s = s.reshape(R, s.shape[1] * 2, 2)[:, :groups]
+    return torch.pow(2.0, s.float() - 127.0)
+
+
 def _dequant_mx(q, sc):
     """Reconstruct f32 from MX-quantized (shuffled) bytes + E8M0 scales.
 
@@ -252,7 +259,7 @@ def _dequant_mx(q, sc):
     R, K = q.shape
     g = K // 64
     d = q.view(R, g, 4, 16).view(torch.float8_e4m3fn).float()
-    scales = torch.pow(2.0, sc.float() - 127.0)          # [R, g, 2]
+    scales = _mx_scales(sc, R, g)                        # [R, g, 2]
     d = d * scales[:, :, [0, 1, 0, 1]].unsqueeze(-1)     # slots q0,q2,q1,q3
     return d[:, :, [0, 2, 1, 3], :].reshape(R, K)        # unshuffle
 
@@ -320,7 +327,7 @@ def _dequant_mx4(q, sc, K):
     lo = lut[(q & 0xF).long()]
     hi = lut[(q >> 4).long()]
     d = torch.stack([lo, hi], dim=-1).reshape(R, g, 2, 32)
-    scales = torch.pow(2.0, sc.float() - 127.0)
+    scales = _mx_scales(sc, R, g)
     return (d * scales.unsqueeze(-1)).reshape(R, K)
 
 
