@@ -356,20 +356,19 @@ __device__ __forceinline__ void stage_scalesw(
     char* __restrict__ smem, int w, int lane) {
   const int tc = tile < ksteps ? tile : ksteps - 1;
   char* const base = smem + SC_BASE4W + (size_t)(tile % 3) * SC_TILE4W;
-  // BK=256 tile = scale-kt pair {2*tc, 2*tc+1}; each dword coalesced in
-  // the tile-major layout. LDS image must stay [row][8B]: dword d of the
-  // 4-KB block = (side, r, dj) with source block kt = 2*tc + dj.
+  // BK=256 tile = scale-kt pair {2*tc, 2*tc+1}. glds j stages block
+  // kt = 2*tc+j for ALL threads (fully coalesced: 256 consecutive dwords
+  // per side); LDS image = [dj 2 KB][A rows | B rows] x 4 B.
 #pragma unroll
   for (int j = 0; j < 2; ++j) {
-    const int d = j * 512 + w * 64 + lane;  // dword 0..1023
-    const int side = d >> 9;                // 0 = A, 1 = B
-    const int r = (d & 511) >> 1, dj = d & 1;
+    const int t = w * 64 + lane;            // dword 0..511 within the block
+    const int side = t >> 8, r = t & 255;
     const unsigned char* g =
-        (side == 0 ? sA + ((long)(2 * tc + dj) * Mtot + bm + r) * 4
-                   : sB + ((long)(2 * tc + dj) * Ntot + bn + r) * 4);
+        (side == 0 ? sA + ((long)(2 * tc + j) * Mtot + bm + r) * 4
+                   : sB + ((long)(2 * tc + j) * Ntot + bn + r) * 4);
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) void*)g,
-        (__attribute__((address_space(3))) void*)(base + (j * 8 + w) * 256),
+        (__attribute__((address_space(3))) void*)(base + j * 2048 + w * 256),
         4, 0, 0);
   }
 }
@@ -387,11 +386,11 @@ __device__ __forceinline__ i32x8v fragw(const char* __restrict__ smem,
 __device__ __forceinline__ void scalewordw(const char* __restrict__ smem,
                                            int slot3, int hh,
                                            int row_in_half, unsigned out[2]) {
-  const int row = (hh < 2 ? 0 : 2048) + ((hh & 1) * 128 + row_in_half) * 8;
-  const unsigned* p = reinterpret_cast<const unsigned*>(
-      smem + SC_BASE4W + (size_t)slot3 * SC_TILE4W + row);
-  out[0] = p[0];
-  out[1] = p[1];
+  // LDS image: [dj 2 KB][A 1 KB | B 1 KB], one dword per row
+  const int row = (hh < 2 ? 0 : 1024) + ((hh & 1) * 128 + row_in_half) * 4;
+  const char* base = smem + SC_BASE4W + (size_t)slot3 * SC_TILE4W + row;
+  out[0] = *reinterpret_cast<const unsigned*>(base);
+  out[1] = *reinterpret_cast<const unsigned*>(base + 2048);
 }
 
 __global__ __launch_bounds__(THREADS4, 1) void gemm_mx4w_8ph_kernel(
