@@ -525,10 +525,11 @@ hipError_t gemm_mx_fp4_nt8ph(const unsigned char* A, const unsigned char* sA,
                              float* C, int M, int N, int K,
                              hipStream_t stream) {
   const unsigned tiles_m = (unsigned)(M / BM4), tiles_n = (unsigned)(N / BM4);
-  // wide kernel runs 1 block/CU (140 KiB LDS); A/B via LPX_FP4_WIDE=0/1
+  // BK=128 wins at every measured shape after the coalesced tile-major
+  // scale staging (2956 vs 2043 TF @8192: 2 blocks/CU beats the wide
+  // kernel's per-phase work). Wide kept behind LPX_FP4_WIDE=1 for A/B.
   const char* wenv = getenv("LPX_FP4_WIDE");
-  const bool use_wide =
-      wenv ? (wenv[0] == '1') : (K % BK4W == 0 && K >= 6144);
+  const bool use_wide = wenv && wenv[0] == '1';
   if (use_wide && K % BK4W == 0) {
     const size_t lds = 2 * PARITY4W + 3 * SC_TILE4W;  // 140 KiB
    hipLaunchKernelGGL(( gemm_mx4w_8ph_kernel), dim3(dim3(tiles_m * tiles_n)), dim3(dim3(THREADS4)), lds,
