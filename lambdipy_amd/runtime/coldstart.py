@@ -143,15 +143,21 @@ def cold_start(
         layer_mb = os.path.getsize(layer) / 2**20
         tmpdir = tempfile.mkdtemp(prefix="lpx-layer-")
         ready_file = os.path.join(tmpdir, ".lpx_ready")
-        _parallel_extract(layer, tmpdir,
-                          only=lambda n: n.startswith("lpx_handlers/")
-                          or n == "lpx_fanout.py")
+        # the torch-free rf_lite handler imports ONLY lpx_rt + the AOT
+        # .hsaco: extract those eagerly too and skip the ready-marker wait
+        # entirely (the bulk unzip overlaps the whole GPU cold start).
+        if handler == "rf_lite":
+            def early(n):
+                return (n.startswith("lpx_handlers/") or n == "lpx_fanout.py"
+                        or n == "lpx_rt.py" or n.startswith("lpx_ops/"))
+        else:
+            def early(n):
+                return n.startswith("lpx_handlers/") or n == "lpx_fanout.py"
+        _parallel_extract(layer, tmpdir, only=early)
         layer_dir = tmpdir
 
         def deferred_extract():
-            _parallel_extract(layer, tmpdir,
-                              only=lambda n: not (n.startswith("lpx_handlers/")
-                                                  or n == "lpx_fanout.py"))
+            _parallel_extract(layer, tmpdir, only=lambda n: not early(n))
             with open(ready_file, "w") as f:
                 f.write("ready")
     else:
@@ -166,7 +172,7 @@ def cold_start(
         os.pathsep + base_env["PYTHONPATH"] if base_env.get("PYTHONPATH") else ""
     )
     base_env.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
-    if ready_file:
+    if ready_file and handler != "rf_lite":  # rf_lite's files are all early
         base_env["LPX_READY_FILE"] = ready_file
 
     code = f"import lpx_handlers.{mod} as h; h.main()"
