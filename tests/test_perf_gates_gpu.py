@@ -82,7 +82,7 @@ def test_gemm_mx_fp4_floor():
     qa, sa = e.quantize_mx4(a)
     dt = _time_op(lambda: e.gemm_mx4(qa, sa, qa, sa))
     tf = 2 * n**3 / dt / 1e12
-    assert tf > 1300, f"MX fp4 GEMM at {tf:.0f} TF (floor 1300; measured ~2300)"
+    assert tf > 1500, f"MX fp4 GEMM at {tf:.0f} TF (floor 1500; measured ~2400-2950)"
 
 
 def test_resnet50_fp8_floor():
