@@ -170,10 +170,15 @@ __global__ __launch_bounds__(THREADS8, 2) void gemm_bf16_8ph_kernel(
             acc[p][mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                 aF[mf][ks], bF[nf][ks], acc[p][mf][nf], 0, 0, 0);
       if constexpr (PRIO == 0) __builtin_amdgcn_s_setprio(0);
-      if constexpr (VM == 8)
+      if constexpr (VM == 60) {
+        // guide-template placement: wait only at phases 1/3 (twice per
+        // K-tile); every read is >=4 calls back, allow-last-3 covers it
+        if (p & 1) asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+      } else if constexpr (VM == 8) {
         asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
-      else
+      } else {
         asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+      }
       __builtin_amdgcn_s_barrier();
     }
   }
@@ -237,6 +242,9 @@ hipError_t gemm_bf16_nt8ph_tune(const uint16_t* A, const uint16_t* Bt,
     case 3: return launch8ph<false, 8, false>(A, Bt, C, M, N, K, stream);
     case 4: return launch8ph<false, 8, false, 1>(A, Bt, C, M, N, K, stream);
     case 5: return launch8ph<false, 8, false, 2>(A, Bt, C, M, N, K, stream);
+    case 6: return launch8ph<false, 60, true, 1>(A, Bt, C, M, N, K, stream);
+    case 7: return launch8ph<false, 60, true, 0>(A, Bt, C, M, N, K, stream);
+    case 8: return launch8ph<false, 60, false, 1>(A, Bt, C, M, N, K, stream);
     default: return launch8ph<false, 6, true>(A, Bt, C, M, N, K, stream);
   }
 }
