@@ -85,7 +85,7 @@ __device__ __forceinline__ const bf16frag* frag_ptr(const char* __restrict__ sme
       smem + (size_t)parity * PARITY_BYTES + (size_t)hh * HALF_BYTES + q * 16);
 }
 
-template <bool BF16_OUT, int VM, bool LG0, int PRIO>
+template <bool BF16_OUT, int VM, bool LG0, int PRIO, int ILV = 0>
 __global__ __launch_bounds__(THREADS8, 2) void gemm_bf16_8ph_kernel(
     const uint16_t* __restrict__ A, const uint16_t* __restrict__ Bt,
     void* __restrict__ Cout, int M, int N, int K, unsigned tiles_m,
@@ -139,10 +139,10 @@ __global__ __launch_bounds__(THREADS8, 2) void gemm_bf16_8ph_kernel(
     for (int p = 0; p < 4; ++p) {
       const int rh = (p >> 1) & 1;  // quadrant row half
       const int ch = p & 1;        // quadrant col half
-      // 12 fragment reads for this quadrant
+      // fragment reads for this quadrant (ILV=1: defer the ks=1 half past
+      // the barrier so its ds_read latency hides under the ks=0 MFMAs)
       bf16frag aF[4][2], bF[2][2];
-#pragma unroll
-      for (int ks = 0; ks < 2; ++ks) {
+      auto read_ks = [&](int ks) {
         const int cell = ks * 4 + fkb;
 #pragma unroll
         for (int mf = 0; mf < 4; ++mf)
@@ -152,7 +152,9 @@ __global__ __launch_bounds__(THREADS8, 2) void gemm_bf16_8ph_kernel(
         for (int nf = 0; nf < 2; ++nf)
           bF[nf][ks] = *frag_ptr(smem, parity, 2 + ch,
                                  wsc * 32 + nf * 16 + fr, cell);
-      }
+      };
+      read_ks(0);
+      if constexpr (ILV == 0) read_ks(1);
       // stage the slot this phase frees
       int st, sh;
       stage_slot(t, p, &st, &sh);
@@ -161,14 +163,26 @@ __global__ __launch_bounds__(THREADS8, 2) void gemm_bf16_8ph_kernel(
       __builtin_amdgcn_s_barrier();
       if constexpr (LG0) asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
       if constexpr (PRIO == 0) __builtin_amdgcn_s_setprio(1);
+      if constexpr (ILV == 1) {
+        read_ks(1);
 #pragma unroll
-      for (int mf = 0; mf < 4; ++mf)
+        for (int ks = 0; ks < 2; ++ks)
 #pragma unroll
-        for (int nf = 0; nf < 2; ++nf)
+          for (int mf = 0; mf < 4; ++mf)
 #pragma unroll
-          for (int ks = 0; ks < 2; ++ks)
-            acc[p][mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                aF[mf][ks], bF[nf][ks], acc[p][mf][nf], 0, 0, 0);
+            for (int nf = 0; nf < 2; ++nf)
+              acc[p][mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  aF[mf][ks], bF[nf][ks], acc[p][mf][nf], 0, 0, 0);
+      } else {
+#pragma unroll
+        for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+          for (int nf = 0; nf < 2; ++nf)
+#pragma unroll
+            for (int ks = 0; ks < 2; ++ks)
+              acc[p][mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  aF[mf][ks], bF[nf][ks], acc[p][mf][nf], 0, 0, 0);
+      }
       if constexpr (PRIO == 0) __builtin_amdgcn_s_setprio(0);
       if constexpr (VM == 60) {
         // guide-template placement: wait only at phases 1/3 (twice per
@@ -212,12 +226,13 @@ __global__ __launch_bounds__(THREADS8, 2) void gemm_bf16_8ph_kernel(
 
 }  // namespace
 
-template <bool BF16_OUT, int VM = 60, bool LG0 = false, int PRIO = 1>
+template <bool BF16_OUT, int VM = 60, bool LG0 = false, int PRIO = 1,
+          int ILV = 0>
 static hipError_t launch8ph(const uint16_t* A, const uint16_t* Bt, void* C,
                             int M, int N, int K, hipStream_t stream) {
   const unsigned tiles_m = (unsigned)(M / BM8), tiles_n = (unsigned)(N / BM8);
   const size_t lds = 2 * PARITY_BYTES;  // 128 KiB
-  gemm_bf16_8ph_kernel<BF16_OUT, VM, LG0, PRIO><<<dim3(tiles_m * tiles_n),
+  gemm_bf16_8ph_kernel<BF16_OUT, VM, LG0, PRIO, ILV><<<dim3(tiles_m * tiles_n),
                                                   dim3(THREADS8), lds, stream>>>(
       A, Bt, C, M, N, K, tiles_m, tiles_n);
   return hipGetLastError();
@@ -245,6 +260,8 @@ hipError_t gemm_bf16_nt8ph_tune(const uint16_t* A, const uint16_t* Bt,
     case 6: return launch8ph<false, 60, true, 1>(A, Bt, C, M, N, K, stream);
     case 7: return launch8ph<false, 60, true, 0>(A, Bt, C, M, N, K, stream);
     case 8: return launch8ph<false, 60, false, 1>(A, Bt, C, M, N, K, stream);
+    case 9: return launch8ph<false, 60, false, 1, 1>(A, Bt, C, M, N, K, stream);
+    case 10: return launch8ph<false, 8, false, 1, 1>(A, Bt, C, M, N, K, stream);
     default: return launch8ph<false, 6, true>(A, Bt, C, M, N, K, stream);
   }
 }
