@@ -249,8 +249,7 @@ __global__ __launch_bounds__(THREADS4, 2) void gemm_mx4_8ph_kernel(
       for (int mf = 0; mf < 2; ++mf)
         aw[mf] = scaleword4(smem, slot3, rh, wsr * 64 + mf * 32 + fr);
       bw = scaleword4(smem, slot3, 2 + ch, wsc * 32 + fr);
-#pragma unroll
-      for (int kg = 0; kg < 2; ++kg) {
+      auto read_kg = [&](int kg) {
 #pragma unroll
         for (int mf = 0; mf < 2; ++mf) {
           const int r = wsr * 64 + mf * 32 + fr;
@@ -260,7 +259,8 @@ __global__ __launch_bounds__(THREADS4, 2) void gemm_mx4_8ph_kernel(
         const int rb = wsc * 32 + fr;
         bF[kg] = frag4(smem, parity, 2 + ch, rb, kg, fkb);
         sb[kg] = (int)((bw >> (8 * (kg * 2 + fkb))) & 0xff);
-      }
+      };
+      read_kg(0);  // kg=1 deferred past the barrier (bf16-twin +5%)
       int st, sh;
       stage_slot4(t, p, &st, &sh);
       stage_half4(A, Bt, Kb, bm, bn, st, sh, ksteps, smem, t512);
@@ -268,10 +268,11 @@ __global__ __launch_bounds__(THREADS4, 2) void gemm_mx4_8ph_kernel(
         stage_scales4(sA, sB, M, N, bm, bn, t + 2, ksteps, smem, w, lane);
 
       __builtin_amdgcn_s_barrier();
+      read_kg(1);
 #pragma unroll
-      for (int mf = 0; mf < 2; ++mf)
+      for (int kg = 0; kg < 2; ++kg)
 #pragma unroll
-        for (int kg = 0; kg < 2; ++kg)
+        for (int mf = 0; mf < 2; ++mf)
           acc[p][mf] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
               aF[mf][kg], bF[kg], acc[p][mf], 4, 4, 0, sa[mf][kg], 0, sb[kg]);
       waitvm4<5>();  // constant: window [2,1,1,1]; scales 8 calls back

@@ -214,8 +214,7 @@ __global__ __launch_bounds__(THREADSX, 1) void gemm_mx_8ph_kernel(
       int sa[2][2], sb[2];
       const unsigned* aw = awT[rh];
       const unsigned bw = bwT[ch];
-#pragma unroll
-      for (int kg = 0; kg < 2; ++kg) {
+      auto read_kg = [&](int kg) {
 #pragma unroll
         for (int mf = 0; mf < 2; ++mf) {
           const int r = wsr * 64 + mf * 32 + fr;
@@ -225,7 +224,9 @@ __global__ __launch_bounds__(THREADSX, 1) void gemm_mx_8ph_kernel(
         const int rb = wsc * 32 + fr;
         bF[kg] = fragx(smem, parity, 2 + ch, rb, kg, fkb);
         sb[kg] = SC ? (int)((bw >> (8 * (kg * 2 + fkb))) & 0xff) : 127;
-      }
+      };
+      read_kg(0);  // kg=1 reads deferred past the barrier (latency hides
+                   // under the kg=0 MFMAs — measured +5% on the bf16 twin)
       int st, sh;
       stage_slotx(t, p, &st, &sh);
       stage_halfx(A, Bt, K, bm, bn, st, sh, ksteps, smem, w, lane);
@@ -235,10 +236,11 @@ __global__ __launch_bounds__(THREADSX, 1) void gemm_mx_8ph_kernel(
           stage_scalesx(sA, sB, M, N, bm, bn, t + 2, ksteps, smem, w, lane);
 
       __builtin_amdgcn_s_barrier();
+      read_kg(1);
 #pragma unroll
-      for (int mf = 0; mf < 2; ++mf)
+      for (int kg = 0; kg < 2; ++kg)
 #pragma unroll
-        for (int kg = 0; kg < 2; ++kg)
+        for (int mf = 0; mf < 2; ++mf)
           acc[p][mf] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
               aF[mf][kg], bF[kg], acc[p][mf], 0, 0, 0, sa[mf][kg], 0, sb[kg]);
       // SC: per-tile-cycle call events are [3,2,2,2]; any 4-call window

@@ -227,7 +227,7 @@ __global__ __launch_bounds__(THREADS8, 2) void gemm_bf16_8ph_kernel(
 }  // namespace
 
 template <bool BF16_OUT, int VM = 60, bool LG0 = false, int PRIO = 1,
-          int ILV = 0>
+          int ILV = 1>
 static hipError_t launch8ph(const uint16_t* A, const uint16_t* Bt, void* C,
                             int M, int N, int K, hipStream_t stream) {
   const unsigned tiles_m = (unsigned)(M / BM8), tiles_n = (unsigned)(N / BM8);
