@@ -175,6 +175,9 @@ class ResNet50FP8:
                                    device=self.device)
         self.fc_sa: Optional[float] = None
         self.n_classes = n_classes
+        self._graph = None
+        self._static_in = None
+        self._static_out = None
 
     @property
     def weight_tensors(self) -> List[torch.Tensor]:
@@ -187,7 +190,18 @@ class ResNet50FP8:
         return ts
 
     def __call__(self, x: torch.Tensor) -> torch.Tensor:
-        """x: [N, 3, 224, 224] f32 (NCHW input, converted to NHWC inside)."""
+        """x: [N, 3, 224, 224] f32 (NCHW input, converted to NHWC inside).
+
+        After capture_graph(), calls with the captured batch shape replay
+        the recorded hipGraph (one launch for the whole ~120-kernel
+        forward — wins where small batches are launch-bound)."""
+        if self._graph is not None and x.shape == self._static_in.shape:
+            self._static_in.copy_(x)
+            self._graph.replay()
+            return self._static_out
+        return self._forward(x)
+
+    def _forward(self, x: torch.Tensor) -> torch.Tensor:
         x = x.permute(0, 2, 3, 1).contiguous()  # NHWC
         x = self.conv1(x)
         # maxpool 3x3 s2 p1: native NHWC kernel (no NCHW permute round trip)
@@ -203,3 +217,24 @@ class ResNet50FP8:
         logits = e.gemm_fp8(e.quantize_fp8(x.contiguous(), sa), self.fcq,
                             sa * self.fcscale, self.fc_bias, False)
         return logits
+
+    def capture_graph(self, batch: int) -> None:
+        """Record the forward as a hipGraph (torch.cuda.CUDAGraph on ROCm).
+
+        Requires the calibration pass to have run (frozen activation
+        scales: the captured graph must be sync-free). The whole fused
+        forward replays as a single graph launch afterwards."""
+        assert self.fc_sa is not None, "run one calibration batch first"
+        self._static_in = torch.zeros(batch, 3, 224, 224,
+                                      device=self.device)
+        # warmup on a side stream (allocator state for capture)
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                self._forward(self._static_in)
+        torch.cuda.current_stream().wait_stream(s)
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            self._static_out = self._forward(self._static_in)
+        self._graph = g

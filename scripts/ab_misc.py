@@ -43,3 +43,25 @@ for n in (1024, 4096):
     dt = timeit(lambda: ops.fft(xc), iters=10)
     tb = 2 * xc.numel() * 8 / dt / 1e12
     print(f"PERF fft batch=4096 n={n}: {dt*1e3:.3f} ms  {tb:.2f} TB/s eff")
+
+# hipGraph-captured ResNet (single graph launch per forward)
+try:
+    from lambdipy_amd.shims.lpx_torch import ResNet50FP8 as _R
+    for bs in (32, 64, 128):
+        m2 = _R(device="cuda", seed=0)
+        x2 = torch.randn(bs, 3, 224, 224, device="cuda")
+        m2(x2)  # calibration
+        m2.capture_graph(bs)
+        for _ in range(3):
+            m2(x2)
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        for _ in range(10):
+            m2(x2)
+        torch.cuda.synchronize()
+        dt = (time.perf_counter() - t0) / 10
+        print(f"PERF resnet50-fp8-graph bs={bs}: {dt*1e3:.2f} ms  {bs/dt:.0f} img/s")
+        del m2
+        torch.cuda.empty_cache()
+except Exception as ex:
+    print("graph capture failed:", ex)
