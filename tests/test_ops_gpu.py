@@ -402,6 +402,18 @@ def test_determinism_double_run(ops):
     assert torch.equal(f.predict_proba_t(X), f.predict_proba_t(X))
 
 
+def test_resnet_graph_capture_matches_eager(ops):
+    from lambdipy_amd.shims.lpx_torch import ResNet50FP8
+    m = ResNet50FP8(device=DEV, seed=0)
+    x = torch.randn(8, 3, 224, 224, device=DEV)
+    m(x)  # calibration
+    eager = m(x).clone()
+    m.capture_graph(8)
+    replay = m(x).clone()
+    rel = (replay - eager).norm().item() / (eager.norm().item() + 1e-9)
+    assert rel < 1e-4, rel
+
+
 def test_maxpool_nhwc_matches_torch(ops):
     e = ops.ext()
     g = torch.Generator(device="cpu").manual_seed(16)
