@@ -130,7 +130,8 @@ __device__ __forceinline__ void stage_fp8_nt(const unsigned char* __restrict__ s
 template <bool RELU, bool BIAS>
 __global__ __launch_bounds__(THREADS) void gemm_fp8_nt_kernel(
     const unsigned char* __restrict__ A, const unsigned char* __restrict__ Bt,
-    float* __restrict__ C, const float* __restrict__ bias, int M, int N,
+    float* __restrict__ C, const float* __restrict__ bias,
+    const float* __restrict__ resid, int M, int N,
     int K, float descale, unsigned tiles_m, unsigned tiles_n) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* const sbase = smem;
@@ -202,6 +203,7 @@ __global__ __launch_bounds__(THREADS) void gemm_fp8_nt_kernel(
         if (row < M && col < N) {
           float v = acc[mf][nf][r] * descale;
           if constexpr (BIAS) v += bias[col];
+          if (resid) v += resid[(long)row * N + col];  // fused residual add
           if constexpr (RELU) v = v > 0.f ? v : 0.f;
           C[(long)row * N + col] = v;
         }
@@ -257,7 +259,8 @@ __device__ __forceinline__ void write_fp8_lds(const i32x4 (&regs)[CHUNKS2],
 template <bool RELU, bool BIAS>
 __global__ __launch_bounds__(THREADS2, 2) void gemm_fp8_nt256_kernel(
     const unsigned char* __restrict__ A, const unsigned char* __restrict__ Bt,
-    float* __restrict__ C, const float* __restrict__ bias, int M, int N,
+    float* __restrict__ C, const float* __restrict__ bias,
+    const float* __restrict__ resid, int M, int N,
     int K, float descale, unsigned tiles_m, unsigned tiles_n) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* const lA = smem;
@@ -337,6 +340,7 @@ __global__ __launch_bounds__(THREADS2, 2) void gemm_fp8_nt256_kernel(
         if (row < M && col < N) {
           float v = acc[mf][nf][r] * descale;
           if constexpr (BIAS) v += bias[col];
+          if (resid) v += resid[(long)row * N + col];  // fused residual add
           if constexpr (RELU) v = v > 0.f ? v : 0.f;
           C[(long)row * N + col] = v;
         }
@@ -671,11 +675,23 @@ hipError_t im2col_fp8(const float* x, unsigned char* out, int N, int H, int W,
 hipError_t gemm_fp8_nt(const unsigned char* A, const unsigned char* Bt,
                        float* C, const float* bias, int M, int N, int K,
                        float descale, bool relu, hipStream_t stream) {
+  return gemm_fp8_nt_res(A, Bt, C, bias, nullptr, M, N, K, descale, relu,
+                         stream);
+}
+
+// resid != nullptr fuses `C += resid` (pre-relu) into the epilogue — the
+// ResNet bottleneck's identity add (saves a full elementwise read+write
+// pass over the block output).
+hipError_t gemm_fp8_nt_res(const unsigned char* A, const unsigned char* Bt,
+                           float* C, const float* bias, const float* resid,
+                           int M, int N, int K, float descale, bool relu,
+                           hipStream_t stream) {
   // scaled-MFMA path first: 2x the 16x16x32 instruction rate (unit scales)
   if (M >= 512 && N >= 512 && M % 256 == 0 && N % 256 == 0 && K % 128 == 0 &&
       K >= 128)
-    return gemm_fp8_nt8phx(A, Bt, C, bias, M, N, K, descale, relu, stream);
-  if (M >= 1024 && N >= 1024 && gemm_fp8_8ph_supported(M, N, K))
+    return gemm_fp8_nt8phx_res(A, Bt, C, bias, resid, M, N, K, descale, relu,
+                               stream);
+  if (!resid && M >= 1024 && N >= 1024 && gemm_fp8_8ph_supported(M, N, K))
     return gemm_fp8_nt8ph(A, Bt, C, bias, M, N, K, descale, relu, stream);
   const bool big = (M >= 512 && N >= 512 && K >= 64);
   if (big) {
@@ -684,16 +700,16 @@ hipError_t gemm_fp8_nt(const unsigned char* A, const unsigned char* Bt,
     dim3 grid(tiles_m * tiles_n), block(THREADS2);
     if (bias && relu)
       gemm_fp8_nt256_kernel<true, true><<<grid, block, lds, stream>>>(
-          A, Bt, C, bias, M, N, K, descale, tiles_m, tiles_n);
+          A, Bt, C, bias, resid, M, N, K, descale, tiles_m, tiles_n);
     else if (bias)
       gemm_fp8_nt256_kernel<false, true><<<grid, block, lds, stream>>>(
-          A, Bt, C, bias, M, N, K, descale, tiles_m, tiles_n);
+          A, Bt, C, bias, resid, M, N, K, descale, tiles_m, tiles_n);
     else if (relu)
       gemm_fp8_nt256_kernel<true, false><<<grid, block, lds, stream>>>(
-          A, Bt, C, bias, M, N, K, descale, tiles_m, tiles_n);
+          A, Bt, C, bias, resid, M, N, K, descale, tiles_m, tiles_n);
     else
       gemm_fp8_nt256_kernel<false, false><<<grid, block, lds, stream>>>(
-          A, Bt, C, bias, M, N, K, descale, tiles_m, tiles_n);
+          A, Bt, C, bias, resid, M, N, K, descale, tiles_m, tiles_n);
     return hipGetLastError();
   }
   const unsigned tiles_m = lpx_ceil_div(M, BM), tiles_n = lpx_ceil_div(N, BN);
@@ -701,16 +717,16 @@ hipError_t gemm_fp8_nt(const unsigned char* A, const unsigned char* Bt,
   dim3 grid(tiles_m * tiles_n), block(THREADS);
   if (bias && relu)
     gemm_fp8_nt_kernel<true, true><<<grid, block, lds, stream>>>(
-        A, Bt, C, bias, M, N, K, descale, tiles_m, tiles_n);
+        A, Bt, C, bias, resid, M, N, K, descale, tiles_m, tiles_n);
   else if (bias)
     gemm_fp8_nt_kernel<false, true><<<grid, block, lds, stream>>>(
-        A, Bt, C, bias, M, N, K, descale, tiles_m, tiles_n);
+        A, Bt, C, bias, resid, M, N, K, descale, tiles_m, tiles_n);
   else if (relu)
     gemm_fp8_nt_kernel<true, false><<<grid, block, lds, stream>>>(
-        A, Bt, C, bias, M, N, K, descale, tiles_m, tiles_n);
+        A, Bt, C, bias, resid, M, N, K, descale, tiles_m, tiles_n);
   else
     gemm_fp8_nt_kernel<false, false><<<grid, block, lds, stream>>>(
-        A, Bt, C, bias, M, N, K, descale, tiles_m, tiles_n);
+        A, Bt, C, bias, resid, M, N, K, descale, tiles_m, tiles_n);
   return hipGetLastError();
 }
 

@@ -21,6 +21,7 @@
 // register-staged MX kernel for other shapes).
 
 #include "lpx_common.h"
+#include "kernels.h"
 
 namespace lpx {
 
@@ -148,7 +149,8 @@ __global__ __launch_bounds__(THREADSX, 1) void gemm_mx_8ph_kernel(
     const unsigned char* __restrict__ A, const unsigned char* __restrict__ sA,
     const unsigned char* __restrict__ Bt, const unsigned char* __restrict__ sB,
     float* __restrict__ C, unsigned char* __restrict__ C8,
-    const float* __restrict__ bias, float descale, float out_inv,
+    const float* __restrict__ bias, const float* __restrict__ resid,
+    float descale, float out_inv,
     int M, int N, int K, unsigned tiles_m, unsigned tiles_n) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
 
@@ -268,6 +270,7 @@ __global__ __launch_bounds__(THREADSX, 1) void gemm_mx_8ph_kernel(
         float v = acc[p][mf][r];
         if constexpr (!SC) v *= descale;
         if constexpr (BIAS) v += bias[col];
+        if (resid) v += resid[(long)row * N + col];  // fused residual add
         if constexpr (RELU) v = v > 0.f ? v : 0.f;
         if constexpr (OUT8)
           C8[(long)row * N + col] = lpx_f32_to_e4m3(v * out_inv);
@@ -292,7 +295,7 @@ hipError_t gemm_mx_fp8_nt8ph(const unsigned char* A, const unsigned char* sA,
   const size_t lds = 2 * PARITY_BYTES + 3 * SC_TILE;  // 134 KiB
   gemm_mx_8ph_kernel<true, false, false>
       <<<dim3(tiles_m * tiles_n), dim3(THREADSX), lds, stream>>>(
-          A, sA, Bt, sB, C, nullptr, nullptr, 1.f, 1.f, M, N, K, tiles_m,
+          A, sA, Bt, sB, C, nullptr, nullptr, nullptr, 1.f, 1.f, M, N, K, tiles_m,
           tiles_n);
   return hipGetLastError();
 }
@@ -301,24 +304,32 @@ hipError_t gemm_mx_fp8_nt8ph(const unsigned char* A, const unsigned char* sA,
 hipError_t gemm_fp8_nt8phx(const unsigned char* A, const unsigned char* Bt,
                            float* C, const float* bias, int M, int N, int K,
                            float descale, bool relu, hipStream_t stream) {
+  return gemm_fp8_nt8phx_res(A, Bt, C, bias, nullptr, M, N, K, descale, relu,
+                             stream);
+}
+
+hipError_t gemm_fp8_nt8phx_res(const unsigned char* A, const unsigned char* Bt,
+                               float* C, const float* bias, const float* resid,
+                               int M, int N, int K, float descale, bool relu,
+                               hipStream_t stream) {
   const unsigned tiles_m = (unsigned)(M / BMX), tiles_n = (unsigned)(N / BMX);
   const size_t lds = 2 * PARITY_BYTES;  // 128 KiB, no scale region
   dim3 grid(tiles_m * tiles_n), block(THREADSX);
   if (bias && relu)
     gemm_mx_8ph_kernel<false, true, true><<<grid, block, lds, stream>>>(
-        A, nullptr, Bt, nullptr, C, nullptr, bias, descale, 1.f, M, N, K,
+        A, nullptr, Bt, nullptr, C, nullptr, bias, resid, descale, 1.f, M, N, K,
         tiles_m, tiles_n);
   else if (bias)
     gemm_mx_8ph_kernel<false, true, false><<<grid, block, lds, stream>>>(
-        A, nullptr, Bt, nullptr, C, nullptr, bias, descale, 1.f, M, N, K,
+        A, nullptr, Bt, nullptr, C, nullptr, bias, resid, descale, 1.f, M, N, K,
         tiles_m, tiles_n);
   else if (relu)
     gemm_mx_8ph_kernel<false, false, true><<<grid, block, lds, stream>>>(
-        A, nullptr, Bt, nullptr, C, nullptr, nullptr, descale, 1.f, M, N, K,
+        A, nullptr, Bt, nullptr, C, nullptr, nullptr, resid, descale, 1.f, M, N, K,
         tiles_m, tiles_n);
   else
     gemm_mx_8ph_kernel<false, false, false><<<grid, block, lds, stream>>>(
-        A, nullptr, Bt, nullptr, C, nullptr, nullptr, descale, 1.f, M, N, K,
+        A, nullptr, Bt, nullptr, C, nullptr, nullptr, resid, descale, 1.f, M, N, K,
         tiles_m, tiles_n);
   return hipGetLastError();
 }
@@ -333,19 +344,19 @@ hipError_t gemm_fp8_nt8phx_q8(const unsigned char* A, const unsigned char* Bt,
   dim3 grid(tiles_m * tiles_n), block(THREADSX);
   if (bias && relu)
     gemm_mx_8ph_kernel<false, true, true, true><<<grid, block, lds, stream>>>(
-        A, nullptr, Bt, nullptr, nullptr, C8, bias, descale, out_inv, M, N, K,
+        A, nullptr, Bt, nullptr, nullptr, C8, bias, nullptr, descale, out_inv, M, N, K,
         tiles_m, tiles_n);
   else if (bias)
     gemm_mx_8ph_kernel<false, true, false, true><<<grid, block, lds, stream>>>(
-        A, nullptr, Bt, nullptr, nullptr, C8, bias, descale, out_inv, M, N, K,
+        A, nullptr, Bt, nullptr, nullptr, C8, bias, nullptr, descale, out_inv, M, N, K,
         tiles_m, tiles_n);
   else if (relu)
     gemm_mx_8ph_kernel<false, false, true, true><<<grid, block, lds, stream>>>(
-        A, nullptr, Bt, nullptr, nullptr, C8, nullptr, descale, out_inv, M, N,
+        A, nullptr, Bt, nullptr, nullptr, C8, nullptr, nullptr, descale, out_inv, M, N,
         K, tiles_m, tiles_n);
   else
     gemm_mx_8ph_kernel<false, false, false, true><<<grid, block, lds, stream>>>(
-        A, nullptr, Bt, nullptr, nullptr, C8, nullptr, descale, out_inv, M, N,
+        A, nullptr, Bt, nullptr, nullptr, C8, nullptr, nullptr, descale, out_inv, M, N,
         K, tiles_m, tiles_n);
   return hipGetLastError();
 }

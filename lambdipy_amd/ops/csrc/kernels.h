@@ -110,6 +110,11 @@ hipError_t im2col_fp8(const float* x, unsigned char* out, int N, int H, int W,
 hipError_t gemm_fp8_nt(const unsigned char* A, const unsigned char* Bt,
                        float* C, const float* bias, int M, int N, int K,
                        float descale, bool relu, hipStream_t stream);
+// resid != nullptr: fused `C += resid` before relu (bottleneck identity add)
+hipError_t gemm_fp8_nt_res(const unsigned char* A, const unsigned char* Bt,
+                           float* C, const float* bias, const float* resid,
+                           int M, int N, int K, float descale, bool relu,
+                           hipStream_t stream);
 // Implicit-GEMM fp8 conv: xq NHWC e4m3 (pre-quantized), Wq [KO, R*S*C];
 // out f32 [N*HO*WO, KO]; requires C % 16 == 0.
 hipError_t conv_fp8_implicit(const unsigned char* xq, const unsigned char* Wq,
@@ -143,6 +148,10 @@ hipError_t gemm_mx_fp8_nt8ph(const unsigned char* A, const unsigned char* sA,
 hipError_t gemm_fp8_nt8phx(const unsigned char* A, const unsigned char* Bt,
                            float* C, const float* bias, int M, int N, int K,
                            float descale, bool relu, hipStream_t stream);
+hipError_t gemm_fp8_nt8phx_res(const unsigned char* A, const unsigned char* Bt,
+                               float* C, const float* bias, const float* resid,
+                               int M, int N, int K, float descale, bool relu,
+                               hipStream_t stream);
 // fused epilogue quantize: out = e4m3((descale*acc + bias, relu) * out_inv)
 hipError_t gemm_fp8_nt8phx_q8(const unsigned char* A, const unsigned char* Bt,
                               unsigned char* C8, const float* bias, int M,

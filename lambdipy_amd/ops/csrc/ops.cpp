@@ -253,7 +253,8 @@ at::Tensor im2col_fp8(const at::Tensor& x, int64_t R, int64_t S, int64_t stride,
 }
 
 at::Tensor gemm_fp8(const at::Tensor& A, const at::Tensor& Bt, double descale,
-                    const c10::optional<at::Tensor>& bias, bool relu) {
+                    const c10::optional<at::Tensor>& bias, bool relu,
+                    const c10::optional<at::Tensor>& resid) {
   CHECK_DEV(A);
   CHECK_DEV(Bt);
   TORCH_CHECK(A.scalar_type() == at::kByte && Bt.scalar_type() == at::kByte,
@@ -269,10 +270,18 @@ at::Tensor gemm_fp8(const at::Tensor& A, const at::Tensor& Bt, double descale,
     TORCH_CHECK(bias->numel() == N, "bias must be [N]");
     bptr = bias->data_ptr<float>();
   }
-  check_hip(lpx::gemm_fp8_nt(A.data_ptr<uint8_t>(), Bt.data_ptr<uint8_t>(),
-                             C.data_ptr<float>(), bptr, (int)M, (int)N, (int)K,
-                             (float)descale, relu, current_stream(A)),
-            "gemm_fp8_nt");
+  const float* rptr = nullptr;
+  if (resid.has_value()) {
+    CHECK_DEV(resid.value());
+    TORCH_CHECK(resid->numel() == M * N && resid->is_contiguous(),
+                "resid must be contiguous [M, N]");
+    rptr = resid->data_ptr<float>();
+  }
+  check_hip(lpx::gemm_fp8_nt_res(A.data_ptr<uint8_t>(), Bt.data_ptr<uint8_t>(),
+                                 C.data_ptr<float>(), bptr, rptr, (int)M,
+                                 (int)N, (int)K, (float)descale, relu,
+                                 current_stream(A)),
+            "gemm_fp8_nt_res");
   return C;
 }
 
@@ -631,9 +640,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("R"), py::arg("S"), py::arg("stride"), py::arg("pad"),
         py::arg("descale") = 1.0, py::arg("bias") = py::none(),
         py::arg("relu") = false);
-  m.def("gemm_fp8", &gemm_fp8, "fp8 e4m3 MFMA GEMM with fused descale/bias/relu",
+  m.def("gemm_fp8", &gemm_fp8,
+        "fp8 e4m3 MFMA GEMM with fused descale/bias/resid/relu",
         py::arg("A"), py::arg("Bt"), py::arg("descale") = 1.0,
-        py::arg("bias") = py::none(), py::arg("relu") = false);
+        py::arg("bias") = py::none(), py::arg("relu") = false,
+        py::arg("resid") = py::none());
   m.def("maxpool2d_nhwc", &maxpool2d_nhwc, "NHWC f32 max pool",
         py::arg("x"), py::arg("k"), py::arg("stride"), py::arg("pad"));
   m.def("gemm_fp8_q8", &gemm_fp8_q8,

@@ -97,26 +97,32 @@ class Conv2dFP8:
     # xq is NHWC e4m3 quantized at self.sa (the producer wrote it with
     # out_scale = self.sa). out_scale=None -> f32 result; else e4m3 bytes.
     def run_q(self, xq: torch.Tensor, N: int, H: int, W: int,
-              out_scale: Optional[float]):
+              out_scale: Optional[float],
+              resid: Optional[torch.Tensor] = None,
+              relu: Optional[bool] = None):
         e = _ops.ext()
         desc = self.sa * self.wscale
         ho, wo = self.out_shape(H, W)
+        if relu is None:
+            relu = self.relu
         if self.k == 1 and self.stride == 1 and self.pad == 0:
             a2 = xq.reshape(-1, self.cin)
             if out_scale is None:
-                out = e.gemm_fp8(a2, self.wq, desc, self.bias, self.relu)
+                out = e.gemm_fp8(a2, self.wq, desc, self.bias, relu,
+                                 resid.reshape(-1, self.cout).contiguous()
+                                 if resid is not None else None)
             else:
-                out = e.gemm_fp8_q8(a2, self.wq, desc, self.bias, self.relu,
+                out = e.gemm_fp8_q8(a2, self.wq, desc, self.bias, relu,
                                     out_scale)
         else:
             if out_scale is None:
                 out = e.conv_fp8(xq, self.wq, H, W, self.cin, self.k, self.k,
                                  self.stride, self.pad, desc, self.bias,
-                                 self.relu)
+                                 relu)
             else:
                 out = e.conv_fp8_q8(xq, self.wq, H, W, self.cin, self.k,
                                     self.k, self.stride, self.pad, desc,
-                                    self.bias, self.relu, out_scale)
+                                    self.bias, relu, out_scale)
         return out.view(N, ho, wo, self.cout), ho, wo
 
 
@@ -139,9 +145,12 @@ class Bottleneck:
             xq = e.quantize_fp8(x.contiguous(), self.conv1.sa)
             q1, h, w = self.conv1.run_q(xq, N, H, W, self.conv2.sa)
             q2, h, w = self.conv2.run_q(q1.contiguous(), N, h, w, self.conv3.sa)
-            y, h, w = self.conv3.run_q(q2.contiguous(), N, h, w, None)
-        else:
-            y = self.conv3(self.conv2(self.conv1(x)))  # calibration pass
+            # conv3 fuses the identity add + relu into its epilogue (conv3
+            # is always 1x1, so the fused-resid GEMM path applies)
+            y, h, w = self.conv3.run_q(q2.contiguous(), N, h, w, None,
+                                       resid=idn, relu=True)
+            return y
+        y = self.conv3(self.conv2(self.conv1(x)))  # calibration pass
         return F.relu(y + idn)
 
 
