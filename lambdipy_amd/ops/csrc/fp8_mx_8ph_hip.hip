@@ -145,7 +145,8 @@ __device__ __forceinline__ unsigned scaleword(const char* __restrict__ smem,
 // SC=false: unit scales (literal 127) — plain per-tensor fp8 GEMM on the
 // 2x-rate scaled MFMA (the k-permutation is self-consistent between A and
 // B, so unshuffled operands contract correctly) with fused descale/bias/relu.
-template <bool SC, bool BIAS, bool RELU, bool OUT8 = false>
+template <bool SC, bool BIAS, bool RELU, bool OUT8 = false,
+          bool RES = false>
 __global__ __launch_bounds__(THREADSX, 1) void gemm_mx_8ph_kernel(
     const unsigned char* __restrict__ A, const unsigned char* __restrict__ sA,
     const unsigned char* __restrict__ Bt, const unsigned char* __restrict__ sB,
@@ -271,7 +272,7 @@ __global__ __launch_bounds__(THREADSX, 1) void gemm_mx_8ph_kernel(
         float v = acc[p][mf][r];
         if constexpr (!SC) v *= descale;
         if constexpr (BIAS) v += bias[col];
-        if (resid) v += resid[(long)row * N + col];  // fused residual add
+        if constexpr (RES) v += resid[(long)row * N + col];  // identity add
         if constexpr (RELU) v = v > 0.f ? v : 0.f;
         if constexpr (OUT8)
           C8[(long)row * N + col] = lpx_f32_to_e4m3(v * out_inv);
@@ -316,6 +317,31 @@ hipError_t gemm_fp8_nt8phx_res(const unsigned char* A, const unsigned char* Bt,
   const unsigned tiles_m = (unsigned)(M / BMX), tiles_n = (unsigned)(N / BMX);
   const size_t lds = 2 * PARITY_BYTES;  // 128 KiB, no scale region
   dim3 grid(tiles_m * tiles_n), block(THREADSX);
+  if (resid) {
+    // RES is a template param: a runtime per-element branch in the
+    // unrolled epilogue generated conditional loads and cost ~4% e2e
+    if (bias && relu)
+     hipLaunchKernelGGL(( gemm_mx_8ph_kernel<false, true, true, false, true>)
+          , dim3(grid), dim3(block), lds, stream, A, nullptr, Bt, nullptr, C, nullptr,
+                                         bias, resid, descale, 1.f, M, N, K,
+                                         tiles_m, tiles_n);
+    else if (bias)
+     hipLaunchKernelGGL(( gemm_mx_8ph_kernel<false, true, false, false, true>)
+          , dim3(grid), dim3(block), lds, stream, A, nullptr, Bt, nullptr, C, nullptr,
+                                         bias, resid, descale, 1.f, M, N, K,
+                                         tiles_m, tiles_n);
+    else if (relu)
+     hipLaunchKernelGGL(( gemm_mx_8ph_kernel<false, false, true, false, true>)
+          , dim3(grid), dim3(block), lds, stream, A, nullptr, Bt, nullptr, C, nullptr,
+                                         nullptr, resid, descale, 1.f, M, N,
+                                         K, tiles_m, tiles_n);
+    else
+     hipLaunchKernelGGL(( gemm_mx_8ph_kernel<false, false, false, false, true>)
+          , dim3(grid), dim3(block), lds, stream, A, nullptr, Bt, nullptr, C, nullptr,
+                                         nullptr, resid, descale, 1.f, M, N,
+                                         K, tiles_m, tiles_n);
+    return hipGetLastError();
+  }
   if (bias && relu)
    hipLaunchKernelGGL(( gemm_mx_8ph_kernel<false, true, true>), dim3(grid), dim3(block), lds, stream, 
         A, nullptr, Bt, nullptr, C, nullptr, bias, resid, descale, 1.f, M, N, K,
