@@ -600,6 +600,10 @@ hipError_t conv_fp8_implicit(const unsigned char* xq, const unsigned char* Wq,
   ConvGeom g{Nb, H, W, Cin, R, S, HO, WO, stride, pad};
   const int M = Nb * HO * WO;
   const int K = R * S * Cin;
+  if (conv_fp8_8ph_supported(M, KO, K, Cin))  // 8-phase glds pipeline
+    return conv_fp8_implicit_8ph(xq, Wq, C, nullptr, bias, Nb, H, W, Cin, R,
+                                 S, HO, WO, stride, pad, KO, descale, relu,
+                                 1.f, stream);
   const unsigned tiles_m = lpx_ceil_div(M, BM2), tiles_n = lpx_ceil_div(KO, BN2);
   const size_t lds = (size_t)2 * BM2 * ROWB2;
   dim3 grid(tiles_m * tiles_n), block(THREADS2);
@@ -631,6 +635,10 @@ hipError_t conv_fp8_implicit_q8(const unsigned char* xq,
   ConvGeom g{Nb, H, W, Cin, R, S, HO, WO, stride, pad};
   const int M = Nb * HO * WO;
   const int K = R * S * Cin;
+  if (conv_fp8_8ph_supported(M, KO, K, Cin))
+    return conv_fp8_implicit_8ph(xq, Wq, nullptr, C8, bias, Nb, H, W, Cin, R,
+                                 S, HO, WO, stride, pad, KO, descale, relu,
+                                 out_inv, stream);
   const unsigned tiles_m = lpx_ceil_div(M, BM2), tiles_n = lpx_ceil_div(KO, BN2);
   const size_t lds = (size_t)2 * BM2 * ROWB2;
   dim3 grid(tiles_m * tiles_n), block(THREADS2);

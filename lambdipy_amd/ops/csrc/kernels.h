@@ -157,6 +157,16 @@ hipError_t gemm_fp8_nt8phx_q8(const unsigned char* A, const unsigned char* Bt,
                               unsigned char* C8, const float* bias, int M,
                               int N, int K, float descale, bool relu,
                               float out_inv, hipStream_t stream);
+// 8-phase glds implicit conv (scaled MFMA, zero-page padding redirect);
+// M%256, KO%256, K=R*S*C%128, C%16. C8 != nullptr => fused e4m3 out.
+bool conv_fp8_8ph_supported(int M, int KO, int K, int Cin);
+hipError_t conv_fp8_implicit_8ph(const unsigned char* xq,
+                                 const unsigned char* Wq, float* C,
+                                 unsigned char* C8, const float* bias, int Nb,
+                                 int H, int W, int Cin, int R, int S, int HO,
+                                 int WO, int stride, int pad, int KO,
+                                 float descale, bool relu, float out_inv,
+                                 hipStream_t stream);
 hipError_t conv_fp8_implicit_q8(const unsigned char* xq,
                                 const unsigned char* Wq, unsigned char* C8,
                                 const float* bias, int Nb, int H, int W,

@@ -402,6 +402,28 @@ def test_determinism_double_run(ops):
     assert torch.equal(f.predict_proba_t(X), f.predict_proba_t(X))
 
 
+def test_conv_fp8_8ph_vs_unfold_ref(ops):
+    # shape chosen to hit the 8-phase glds conv (M%256, KO%256, K%128)
+    e = ops.ext()
+    g = torch.Generator(device="cpu").manual_seed(17)
+    N, H, W, C, KO = 4, 16, 16, 128, 256
+    x = torch.randn(N, H, W, C, generator=g).to(DEV)
+    w = torch.randn(KO, 3 * 3 * C, generator=g).to(DEV)
+    xq = e.quantize_fp8(x, 1.0)
+    wq = e.quantize_fp8(w, 1.0)
+    got = e.conv_fp8(xq, wq, H, W, C, 3, 3, 1, 1, 1.0)
+    xd = xq.view(torch.float8_e4m3fn).float()
+    wd = wq.view(torch.float8_e4m3fn).float()
+    # reference: unfold on NCHW then matmul
+    unf = torch.nn.functional.unfold(
+        xd.permute(0, 3, 1, 2), 3, padding=1)  # [N, C*3*3, L] (C-major taps)
+    # our patch layout is (r, s, c) row-major with c fastest: rearrange
+    unf = unf.view(N, C, 3, 3, -1).permute(0, 4, 2, 3, 1).reshape(-1, 3 * 3 * C)
+    ref = unf @ wd.t()
+    err = (got - ref).abs().max().item() / (ref.abs().max().item() + 1e-6)
+    assert err < 1e-3, err
+
+
 def test_resnet_graph_capture_matches_eager(ops):
     from lambdipy_amd.shims.lpx_torch import ResNet50FP8
     m = ResNet50FP8(device=DEV, seed=0)
