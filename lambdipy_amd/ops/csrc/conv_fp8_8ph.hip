@@ -47,36 +47,61 @@ __device__ __forceinline__ void waitvmc() {
   else static_assert(N == 0, "unsupported vmcnt");
 }
 
-// A half: 128 output-pixel rows x 128 patch bytes, gather-addressed
+// Per-thread output-pixel context: each thread stages the SAME 4 rows
+// every call (2 j-slots x 2 A-halves), so the m -> (n, ho, wo) divisions
+// hoist out of the phase loop entirely.
+struct RowCtx {
+  long n_off;   // n * H (pre-scaled row base)
+  int pho, pwo; // ho*stride - pad, wo*stride - pad
+  int okm;      // m < M
+};
+
+__device__ __forceinline__ void make_row_ctx(const ConvGeomC g, int M, int bm,
+                                             int w, int lane, RowCtx (&c)[4]) {
+#pragma unroll
+  for (int half = 0; half < 2; ++half)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      const int q = (j * 8 + w) * 64 + lane;
+      const int m = bm + half * 128 + (q >> 3);
+      const int wo = m % g.WO;
+      const int ho = (m / g.WO) % g.HO;
+      const int n = m / (g.WO * g.HO);
+      c[half * 2 + j].n_off = (long)n * g.H;
+      c[half * 2 + j].pho = ho * g.stride - g.pad;
+      c[half * 2 + j].pwo = wo * g.stride - g.pad;
+      c[half * 2 + j].okm = m < M;
+    }
+}
+
+// A half: 128 output-pixel rows x 128 patch bytes, gather-addressed.
+// S_T (tap count) is a template so rs/S_T folds; C is a power of two
+// (shift clog2) — no runtime div/mod in the phase path.
+template <int S_T>
 __device__ __forceinline__ void stage_act(
     const unsigned char* __restrict__ xq, const unsigned char* __restrict__ zp,
-    const ConvGeomC g, int M, int K, int bm, int tile, int half, int ksteps,
-    char* __restrict__ smem, int w, int lane) {
+    const ConvGeomC g, int clog2, const RowCtx (&ctx)[4], int tile, int half,
+    int ksteps, char* __restrict__ smem, int w, int lane) {
   const int tc = tile < ksteps ? tile : ksteps - 1;
   const int k0 = tc * BKC;
-  const int row0 = bm + half * 128;
   char* const base = smem + (size_t)(tile & 1) * PARITY_BYTES +
                      (size_t)half * HALF_BYTES;
 #pragma unroll
   for (int j = 0; j < 2; ++j) {
     const int qbase = (j * 8 + w) * 64;
     const int q = qbase + lane;
-    const int r = q >> 3, cell = src_cellc(q);
-    const int m = row0 + r;
+    const int cell = src_cellc(q);
     const int k = k0 + cell * 16;
-    // decompose patch index -> (tap r_f, s_f, channel c0)
-    const int c0 = k % g.C;
-    const int rs = k / g.C;
-    const int s_f = rs % g.S;
-    const int r_f = rs / g.S;
-    const int wo = m % g.WO;
-    const int ho = (m / g.WO) % g.HO;
-    const int n = m / (g.WO * g.HO);
-    const int hi = ho * g.stride + r_f - g.pad;
-    const int wi = wo * g.stride + s_f - g.pad;
-    const bool ok = (m < M) & (hi >= 0) & (hi < g.H) & (wi >= 0) & (wi < g.W);
+    const int c0 = k & ((1 << clog2) - 1);
+    const int rs = k >> clog2;
+    const int r_f = rs / S_T;
+    const int s_f = rs - r_f * S_T;
+    const RowCtx& rc = ctx[half * 2 + j];
+    const int hi = rc.pho + r_f;
+    const int wi = rc.pwo + s_f;
+    const bool ok = rc.okm & (hi >= 0) & (hi < g.H) & (wi >= 0) & (wi < g.W);
     const unsigned char* src =
-        ok ? xq + (((long)n * g.H + hi) * g.W + wi) * g.C + c0
+        ok ? xq + (((rc.n_off + hi) * g.W + wi) << clog2) + c0
            : zp + (lane & 3) * 16;  // zero page (padding -> zeros in LDS)
     __builtin_amdgcn_global_load_lds(
         (const __attribute__((address_space(1))) void*)src,
@@ -114,13 +139,15 @@ __device__ __forceinline__ void stage_slotc(int t, int p, int* tile, int* hh) {
   }
 }
 
+template <int S_T>
 __device__ __forceinline__ void stage_any(
     const unsigned char* __restrict__ xq, const unsigned char* __restrict__ zp,
-    const unsigned char* __restrict__ Wq, const ConvGeomC g, int M, int K,
-    int bm, int bn, int tile, int hh, int ksteps, char* __restrict__ smem,
+    const unsigned char* __restrict__ Wq, const ConvGeomC g, int clog2,
+    const RowCtx (&ctx)[4], int K,
+    int bn, int tile, int hh, int ksteps, char* __restrict__ smem,
     int w, int lane) {
   if (hh < 2)
-    stage_act(xq, zp, g, M, K, bm, tile, hh, ksteps, smem, w, lane);
+    stage_act<S_T>(xq, zp, g, clog2, ctx, tile, hh, ksteps, smem, w, lane);
   else
     stage_wgt(Wq, K, bn, tile, hh - 2, ksteps, smem, w, lane);
 }
@@ -138,12 +165,12 @@ __device__ __forceinline__ i32x8v fragc(const char* __restrict__ smem,
   return i32x8v{lo.x, lo.y, lo.z, lo.w, hi.x, hi.y, hi.z, hi.w};
 }
 
-template <bool BIAS, bool RELU, bool OUT8>
+template <bool BIAS, bool RELU, bool OUT8, int S_T>
 __global__ __launch_bounds__(THREADSC, 1) void conv_fp8_8ph_kernel(
     const unsigned char* __restrict__ xq, const unsigned char* __restrict__ Wq,
     const unsigned char* __restrict__ zp, float* __restrict__ C,
     unsigned char* __restrict__ C8, const float* __restrict__ bias,
-    ConvGeomC g, int M, int N, int K, float descale, float out_inv,
+    ConvGeomC g, int clog2, int M, int N, int K, float descale, float out_inv,
     unsigned tiles_m, unsigned tiles_n) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
 
@@ -168,12 +195,15 @@ __global__ __launch_bounds__(THREADSC, 1) void conv_fp8_8ph_kernel(
 
   const int ksteps = (K + BKC - 1) / BKC;
 
-  stage_any(xq, zp, Wq, g, M, K, bm, bn, 0, 0, ksteps, smem, w, lane);
-  stage_any(xq, zp, Wq, g, M, K, bm, bn, 0, 2, ksteps, smem, w, lane);
-  stage_any(xq, zp, Wq, g, M, K, bm, bn, 0, 3, ksteps, smem, w, lane);
-  stage_any(xq, zp, Wq, g, M, K, bm, bn, 0, 1, ksteps, smem, w, lane);
-  stage_any(xq, zp, Wq, g, M, K, bm, bn, 1, 0, ksteps, smem, w, lane);
-  stage_any(xq, zp, Wq, g, M, K, bm, bn, 1, 2, ksteps, smem, w, lane);
+  RowCtx ctx[4];
+  make_row_ctx(g, M, bm, w, lane, ctx);
+
+  stage_any<S_T>(xq, zp, Wq, g, clog2, ctx, K, bn, 0, 0, ksteps, smem, w, lane);
+  stage_any<S_T>(xq, zp, Wq, g, clog2, ctx, K, bn, 0, 2, ksteps, smem, w, lane);
+  stage_any<S_T>(xq, zp, Wq, g, clog2, ctx, K, bn, 0, 3, ksteps, smem, w, lane);
+  stage_any<S_T>(xq, zp, Wq, g, clog2, ctx, K, bn, 0, 1, ksteps, smem, w, lane);
+  stage_any<S_T>(xq, zp, Wq, g, clog2, ctx, K, bn, 1, 0, ksteps, smem, w, lane);
+  stage_any<S_T>(xq, zp, Wq, g, clog2, ctx, K, bn, 1, 2, ksteps, smem, w, lane);
   waitvmc<4>();
   __builtin_amdgcn_s_barrier();
 
@@ -197,7 +227,8 @@ __global__ __launch_bounds__(THREADSC, 1) void conv_fp8_8ph_kernel(
       read_kg(0);  // kg=1 deferred past the barrier (bf16-twin +5%)
       int st, sh;
       stage_slotc(t, p, &st, &sh);
-      stage_any(xq, zp, Wq, g, M, K, bm, bn, st, sh, ksteps, smem, w, lane);
+      stage_any<S_T>(xq, zp, Wq, g, clog2, ctx, K, bn, st, sh, ksteps, smem,
+                     w, lane);
 
       __builtin_amdgcn_s_barrier();
       read_kg(1);
@@ -254,7 +285,7 @@ unsigned char* conv_zero_page() {
 
 bool conv_fp8_8ph_supported(int M, int KO, int K, int Cin) {
   return M % BMC == 0 && KO % BMC == 0 && K % BKC == 0 && K >= BKC &&
-         Cin % 16 == 0;
+         Cin >= 16 && (Cin & (Cin - 1)) == 0;  // pow2 C: shift addressing
 }
 
 hipError_t conv_fp8_implicit_8ph(const unsigned char* xq,
@@ -269,26 +300,40 @@ hipError_t conv_fp8_implicit_8ph(const unsigned char* xq,
   unsigned char* zp = conv_zero_page();
   if (!zp) return hipErrorOutOfMemory;
   ConvGeomC g{Nb, H, W, Cin, R, S, HO, WO, stride, pad};
+  int clog2 = 0;
+  while ((1 << clog2) < Cin) ++clog2;
+  if (S != 1 && S != 3 && S != 7) return hipErrorInvalidValue;
   const unsigned tiles_m = (unsigned)(M / BMC), tiles_n = (unsigned)(KO / BMC);
   const size_t lds = 2 * PARITY_BYTES;  // 128 KiB
   dim3 grid(tiles_m * tiles_n), block(THREADSC);
-#define LPX_C8PH(B_, R_, O_)                                                   \
-  conv_fp8_8ph_kernel<B_, R_, O_><<<grid, block, lds, stream>>>(               \
-      xq, Wq, zp, C, C8, bias, g, M, KO, K, descale, out_inv, tiles_m,         \
-      tiles_n)
+#define LPX_C8PH_S(B_, R_, O_)                                                 \
+  do {                                                                         \
+    if (S == 3)                                                                \
+      conv_fp8_8ph_kernel<B_, R_, O_, 3><<<grid, block, lds, stream>>>(        \
+          xq, Wq, zp, C, C8, bias, g, clog2, M, KO, K, descale, out_inv,       \
+          tiles_m, tiles_n);                                                   \
+    else if (S == 1)                                                           \
+      conv_fp8_8ph_kernel<B_, R_, O_, 1><<<grid, block, lds, stream>>>(        \
+          xq, Wq, zp, C, C8, bias, g, clog2, M, KO, K, descale, out_inv,       \
+          tiles_m, tiles_n);                                                   \
+    else                                                                       \
+      conv_fp8_8ph_kernel<B_, R_, O_, 7><<<grid, block, lds, stream>>>(        \
+          xq, Wq, zp, C, C8, bias, g, clog2, M, KO, K, descale, out_inv,       \
+          tiles_m, tiles_n);                                                   \
+  } while (0)
   const bool b = bias != nullptr, o8 = C8 != nullptr;
   if (o8) {
-    if (b && relu) LPX_C8PH(true, true, true);
-    else if (b) LPX_C8PH(true, false, true);
-    else if (relu) LPX_C8PH(false, true, true);
-    else LPX_C8PH(false, false, true);
+    if (b && relu) LPX_C8PH_S(true, true, true);
+    else if (b) LPX_C8PH_S(true, false, true);
+    else if (relu) LPX_C8PH_S(false, true, true);
+    else LPX_C8PH_S(false, false, true);
   } else {
-    if (b && relu) LPX_C8PH(true, true, false);
-    else if (b) LPX_C8PH(true, false, false);
-    else if (relu) LPX_C8PH(false, true, false);
-    else LPX_C8PH(false, false, false);
+    if (b && relu) LPX_C8PH_S(true, true, false);
+    else if (b) LPX_C8PH_S(true, false, false);
+    else if (relu) LPX_C8PH_S(false, true, false);
+    else LPX_C8PH_S(false, false, false);
   }
-#undef LPX_C8PH
+#undef LPX_C8PH_S
   return hipGetLastError();
 }
 
