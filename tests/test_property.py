@@ -29,13 +29,13 @@ def test_heap_forest_cpu_reference_consistent(trees, depth, feats, classes,
     assert out1.shape == (samples, classes)
     assert np.isfinite(out1).all()
     if f.layout == "heap":
-        # heap-generated leaf rows are probability distributions ->
-        # per-sample class sums equal the tree count
-        assert np.allclose(out1.sum(axis=1), trees, rtol=1e-4)
+        # heap-generated leaf rows are probability distributions and the
+        # reference returns the across-tree MEAN -> rows sum to 1
+        assert np.allclose(out1.sum(axis=1), 1.0, rtol=1e-4)
 
 
 @settings(max_examples=50, deadline=None)
-@given(name=st.from_regex(r"[A-Za-z][A-Za-z0-9_.\-]{0,20}", fullmatch=True),
+@given(name=st.from_regex(r"[A-Za-z]([A-Za-z0-9_.\-]{0,18}[A-Za-z0-9])?", fullmatch=True),
        lo=st.integers(0, 20), hi=st.integers(21, 99))
 def test_requirement_line_roundtrip(name, lo, hi):
     line = f"{name}>={lo},<{hi}"
