@@ -281,8 +281,8 @@ __global__ __launch_bounds__(THREADS) void rf_lds_kernel(
 // [t * 2^depth + leaf_idx]. Used for ForestSoA.random forests (and heapified
 // sklearn trees when padding stays cheap); ragged forests use rf8_kernel.
 
-template <bool STAGE_X, int SILP>
-__global__ __launch_bounds__(THREADS) void rf_heap_kernel(
+template <bool STAGE_X, int SILP, int BLK = THREADS>
+__global__ __launch_bounds__(BLK) void rf_heap_kernel(
     const i32x2* __restrict__ nodes, const float* __restrict__ leaf_values,
     int n_trees, int depth, const float* __restrict__ X,
     float* __restrict__ out, int S, int F, int C) {
@@ -293,12 +293,12 @@ __global__ __launch_bounds__(THREADS) void rf_heap_kernel(
   const int tree_stride = (1 << depth) - 1;  // internal nodes per tree
   const int n_leaves = 1 << depth;
 
-  for (int tile = blockIdx.x * (THREADS * SILP); tile < S;
-       tile += gridDim.x * (THREADS * SILP)) {
-    const int tile_n = min(THREADS * SILP, S - tile);
+  for (int tile = blockIdx.x * (BLK * SILP); tile < S;
+       tile += gridDim.x * (BLK * SILP)) {
+    const int tile_n = min(BLK * SILP, S - tile);
     if (STAGE_X) {
       const long total = (long)tile_n * F;
-      for (long i = t; i < total; i += THREADS) {
+      for (long i = t; i < total; i += BLK) {
         const int r = (int)(i / F), c = (int)(i - (long)r * F);
         xl[r * rowf + c] = X[((long)tile + r) * F + c];
       }
@@ -309,8 +309,8 @@ __global__ __launch_bounds__(THREADS) void rf_heap_kernel(
     const float* xrow[SILP];
 #pragma unroll
     for (int j = 0; j < SILP; ++j) {
-      s[j] = tile + t + j * THREADS;
-      xrow[j] = STAGE_X ? (xl + (t + j * THREADS) * rowf)
+      s[j] = tile + t + j * BLK;
+      xrow[j] = STAGE_X ? (xl + (t + j * BLK) * rowf)
                         : (X + (long)s[j] * F);
     }
     float acc[SILP][CMAX];
@@ -528,6 +528,32 @@ hipError_t rf_predict_heap_f32(const int* nodes_heap, const float* leaf_values,
   LPX_RFH_CASE(2)
   LPX_RFH_CASE(4)
 #undef LPX_RFH_CASE
+  if (silp == 101 || silp == 102) {
+    // A/B: 128-thread blocks (half the X-stage LDS -> 9-10 blocks/CU of
+    // resident waves vs 4 at 256 threads)
+    const size_t l2 = stage ? (size_t)128 * (F + 1) * sizeof(float) : 0;
+    unsigned b2 = lpx_ceil_div(S, 128);
+    if (b2 > 8192u) b2 = 8192u;
+    if (silp == 101) {
+      if (stage)
+       hipLaunchKernelGGL(( rf_heap_kernel<true, 1, 128>), dim3(dim3(b2)), dim3(dim3(128)), l2, stream, 
+            nh, leaf_values, n_trees, depth, X, out, S, F, C);
+      else
+       hipLaunchKernelGGL(( rf_heap_kernel<false, 1, 128>), dim3(dim3(b2)), dim3(dim3(128)), 0, stream, 
+            nh, leaf_values, n_trees, depth, X, out, S, F, C);
+    } else {
+      b2 = lpx_ceil_div(S, 64);
+      if (b2 > 16384u) b2 = 16384u;
+      const size_t l3 = stage ? (size_t)64 * (F + 1) * sizeof(float) : 0;
+      if (stage)
+       hipLaunchKernelGGL(( rf_heap_kernel<true, 1, 64>), dim3(dim3(b2)), dim3(dim3(64)), l3, stream, 
+            nh, leaf_values, n_trees, depth, X, out, S, F, C);
+      else
+       hipLaunchKernelGGL(( rf_heap_kernel<false, 1, 64>), dim3(dim3(b2)), dim3(dim3(64)), 0, stream, 
+            nh, leaf_values, n_trees, depth, X, out, S, F, C);
+    }
+    return hipGetLastError();
+  }
   return hipErrorInvalidValue;
 }
 

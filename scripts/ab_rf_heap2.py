@@ -29,8 +29,8 @@ def main():
     S = X.shape[0]
     n, lv = f.nodes, f.leaf_values
     ref = e.rf_predict_heap(n, lv, 10, X, 1, -2)
-    for silp in (1, 2, 4):
-        for top in (-2, -1, 2, 3, 4, 5):
+    for silp in (1, 101, 102):
+        for top in (-2,):
             try:
                 out = e.rf_predict_heap(n, lv, 10, X, silp, top)
                 ok = torch.allclose(out, ref, atol=1e-4)
