@@ -65,6 +65,25 @@ __global__ __launch_bounds__(512) void rate_scaled_16x128(float* out, int iters)
   if (s != 0.f) out[0] = s;
 }
 
+__global__ __launch_bounds__(512) void rate_mixed84(float* out, int iters) {
+  i32x8v a = {0, 0, 0, 0, 0, 0, 0, 0}, b = a;
+  f32x16 acc[8];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int e = 0; e < 16; ++e) acc[i][e] = 0.f;
+  for (int it = 0; it < iters; ++it) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+      acc[i] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+          a, b, acc[i], 0, 4, 0, 127, 0, 127);  // A=e4m3, B=e2m1
+  }
+  float s = 0.f;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) s += acc[i][0];
+  if (s != 0.f) out[0] = s;
+}
+
 __global__ __launch_bounds__(512) void rate_plain(float* out, int iters) {
   long a = 0, b = 0;
   f32x4 acc[8];
@@ -129,7 +148,8 @@ extern "C" int run_rate(int which, int blocks, int iters, float* ms_out) {
     else if (which == 2) rate_scaled_fp4<<<blocks, 512>>>(d, n);
     else if (which == 3) rate_scaled_16x128<<<blocks, 512>>>(d, n);
     else if (which == 4) rate_bf16_16<<<blocks, 512>>>(d, n);
-    else rate_bf16_32<<<blocks, 512>>>(d, n);
+    else if (which == 5) rate_bf16_32<<<blocks, 512>>>(d, n);
+    else rate_mixed84<<<blocks, 512>>>(d, n);
   };
   launch(iters / 10 + 1);
   hipDeviceSynchronize();

@@ -18,7 +18,8 @@ for which, name, flops_per in ((0, "scaled 32x32x64 fp8", 8 * 131072),
                                (2, "scaled 32x32x64 fp4", 8 * 131072),
                                (3, "scaled 16x16x128 fp8", 8 * 65536),
                                (4, "plain 16x16x32 bf16", 8 * 16384),
-                               (5, "plain 32x32x16 bf16", 8 * 32768)):
+                               (5, "plain 32x32x16 bf16", 8 * 32768),
+                               (6, "scaled 32x32x64 fp8xfp4", 8 * 131072)):
     ms = ctypes.c_float()
     rc = lib.run_rate(which, BLOCKS, ITERS, ctypes.byref(ms))
     assert rc == 0, rc
