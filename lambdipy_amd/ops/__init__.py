@@ -18,11 +18,13 @@ from .api import (
     gemm_bf16,
     gemm_mx,
     gemm_mx_fp4,
+    gemm_mx_w4,
     gemv,
     ifft,
     matmul_bf16,
     matmul_mx,
     matmul_mx_fp4,
+    matmul_mx_w4,
     quantize_mx,
     quantize_mx_fp4,
     pairwise_sqeuclidean,
@@ -43,7 +45,9 @@ __all__ = [
     "matmul_mx",
     "quantize_mx_fp4",
     "gemm_mx_fp4",
+    "gemm_mx_w4",
     "matmul_mx_fp4",
+    "matmul_mx_w4",
     "gemv",
     "spmv_csr",
     "fft",

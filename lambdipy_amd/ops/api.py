@@ -111,6 +111,20 @@ def matmul_mx_fp4(A: torch.Tensor, B: torch.Tensor) -> torch.Tensor:
     return gemm_mx_fp4(qa, sa, qb, sb)
 
 
+def gemm_mx_w4(A: torch.Tensor, sA: torch.Tensor, Bt4: torch.Tensor,
+               sB: torch.Tensor) -> torch.Tensor:
+    """Mixed MX GEMM: fp8-MX activations x fp4-MX weights — the full fp8
+    MFMA rate with half the weight bytes (weight-heavy inference GEMMs)."""
+    return require_extension().gemm_mx_w4(A, sA, Bt4, sB)
+
+
+def matmul_mx_w4(A: torch.Tensor, B: torch.Tensor) -> torch.Tensor:
+    """Convenience: activations to fp8-MX, weights to fp4-MX, multiply."""
+    qa, sa = quantize_mx(A)
+    qb, sb = quantize_mx_fp4(B.t().contiguous())
+    return gemm_mx_w4(qa, sa, qb, sb)
+
+
 # ---- sparse --------------------------------------------------------------
 
 def spmv_csr(indptr: torch.Tensor, indices: torch.Tensor, data: torch.Tensor,

@@ -184,6 +184,13 @@ hipError_t maxpool2d_nhwc_f32(const float* x, float* out, int N, int H, int W,
 hipError_t quantize_mx_fp4(const float* x, unsigned char* q, unsigned char* sc,
                            int R, int K, hipStream_t stream);
 bool gemm_mx_fp4_supported(int M, int N, int K);
+// mixed: fp8-MX activations x fp4-MX weights at the full fp8 rate with
+// half the weight bytes (probe: 4.96 PF).
+bool gemm_mx_w4_supported(int M, int N, int K);
+hipError_t gemm_mx_w4_nt8ph(const unsigned char* A, const unsigned char* sA,
+                            const unsigned char* Bt, const unsigned char* sB,
+                            float* C, int M, int N, int K,
+                            hipStream_t stream);
 hipError_t gemm_mx_fp4_nt8ph(const unsigned char* A, const unsigned char* sA,
                              const unsigned char* Bt, const unsigned char* sB,
                              float* C, int M, int N, int K,

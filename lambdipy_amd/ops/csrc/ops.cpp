@@ -462,6 +462,32 @@ at::Tensor gemm_mx4(const at::Tensor& A, const at::Tensor& sA,
   return C;
 }
 
+at::Tensor gemm_mx_w4(const at::Tensor& A, const at::Tensor& sA,
+                      const at::Tensor& Bt4, const at::Tensor& sB) {
+  CHECK_DEV(A);
+  CHECK_DEV(Bt4);
+  CHECK_DEV(sA);
+  CHECK_DEV(sB);
+  TORCH_CHECK(A.scalar_type() == at::kByte && Bt4.scalar_type() == at::kByte,
+              "gemm_mx_w4 expects byte tensors");
+  const int64_t M = A.size(0), K = A.size(1), N = Bt4.size(0);
+  TORCH_CHECK(Bt4.size(1) * 2 == K,
+              "gemm_mx_w4: Bt must be fp4 nibbles [N, K/2]");
+  TORCH_CHECK(lpx::gemm_mx_w4_supported((int)M, (int)N, (int)K),
+              "gemm_mx_w4 requires M,N%256==0 and K%128==0");
+  const int64_t kt = (K / 64 + 1) / 2;
+  TORCH_CHECK(sA.numel() == kt * M * 4 && sB.numel() == kt * N * 4,
+              "scale shapes must be [ceil(K/128), rows, 4] (tile-major)");
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(A.device());
+  auto C = at::empty({M, N}, A.options().dtype(at::kFloat));
+  check_hip(lpx::gemm_mx_w4_nt8ph(A.data_ptr<uint8_t>(), sA.data_ptr<uint8_t>(),
+                                  Bt4.data_ptr<uint8_t>(),
+                                  sB.data_ptr<uint8_t>(), C.data_ptr<float>(),
+                                  (int)M, (int)N, (int)K, current_stream(A)),
+            "gemm_mx_w4_nt8ph");
+  return C;
+}
+
 at::Tensor rf_predict8(const at::Tensor& nodes8, const at::Tensor& leaf_values,
                        const at::Tensor& roots, const at::Tensor& X,
                        int64_t ilp) {
@@ -666,6 +692,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("quantize_mx4", &quantize_mx4,
         "f32 [R,K] -> (e2m1 nibbles shuffled [R,K/2], E8M0 scales)",
         py::arg("x"));
+  m.def("gemm_mx_w4", &gemm_mx_w4,
+        "mixed MX GEMM: fp8 activations x fp4 weights (full fp8 rate, half "
+        "weight bytes)", py::arg("A"), py::arg("sA"), py::arg("Bt4"),
+        py::arg("sB"));
   m.def("gemm_mx4", &gemm_mx4,
         "MX block-scaled fp4 GEMM (2x fp8 MFMA rate)", py::arg("A"),
         py::arg("sA"), py::arg("Bt"), py::arg("sB"));

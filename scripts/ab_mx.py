@@ -41,6 +41,8 @@ def main():
         qb4, sb4 = e.quantize_mx4(b)
         dt = bench(lambda: e.gemm_mx4(qa4, sa4, qb4, sb4))
         print(f"n={n} gemm_mx4     {fl/dt:8.0f} TF  {dt*1e3:.3f} ms")
+        dt = bench(lambda: e.gemm_mx_w4(qa, sa, qb4, sb4))
+        print(f"n={n} gemm_mx_w4   {fl/dt:8.0f} TF  {dt*1e3:.3f} ms")
 
 
 if __name__ == "__main__":

@@ -362,6 +362,25 @@ def test_gemm_mx_fp4_vs_dequant_ref(ops):
     assert rel < 0.2, rel
 
 
+def test_gemm_mx_w4_vs_dequant_ref(ops):
+    # mixed: fp8-MX activations x fp4-MX weights
+    e = ops.ext()
+    g = torch.Generator(device="cpu").manual_seed(19)
+    M, N, K = 512, 512, 384
+    A = (torch.randn(M, K, generator=g) * 2).to(DEV)
+    Bt = torch.randn(N, K, generator=g).to(DEV)
+    qa, sa = e.quantize_mx(A)
+    qb, sb = e.quantize_mx4(Bt)
+    C = e.gemm_mx_w4(qa, sa, qb, sb)
+    ref = _dequant_mx(qa, sa) @ _dequant_mx4(qb, sb, K).t()
+    err = (C - ref).abs().max().item() / (ref.abs().max().item() + 1e-6)
+    assert err < 1e-3, err
+    # vs exact f32: activations e4m3 + weights e2m1 -> ~e2m1-dominated noise
+    exact = A @ Bt.t()
+    rel = (C - exact).norm().item() / exact.norm().item()
+    assert rel < 0.15, rel
+
+
 def test_gemm_fp8_scaled_mfma_path(ops):
     # 256-multiple shapes dispatch to the unit-scale scaled-MFMA kernel
     e = ops.ext()
