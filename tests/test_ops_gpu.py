@@ -381,6 +381,20 @@ def test_gemm_mx_w4_vs_dequant_ref(ops):
     assert rel < 0.15, rel
 
 
+def test_linear_w4_matches_f32(ops):
+    from lambdipy_amd.shims.lpx_torch import LinearW4
+    g = torch.Generator(device="cpu").manual_seed(20)
+    M, IN, OUT = 256, 256, 256
+    w = (torch.randn(OUT, IN, generator=g) * 0.1).to(DEV)
+    b = torch.randn(OUT, generator=g).to(DEV)
+    x = torch.randn(M, IN, generator=g).to(DEV)
+    lin = LinearW4(w, b)
+    y = lin(x)
+    ref = x @ w.t() + b
+    rel = (y - ref).norm().item() / ref.norm().item()
+    assert rel < 0.15, rel  # e2m1 weights dominate the quantization noise
+
+
 def test_gemm_fp8_scaled_mfma_path(ops):
     # 256-multiple shapes dispatch to the unit-scale scaled-MFMA kernel
     e = ops.ext()
