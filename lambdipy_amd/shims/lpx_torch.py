@@ -136,13 +136,18 @@ class Bottleneck:
             self.down = Conv2dFP8(cin, cout, 1, stride, 0, False, device, gen)
 
     def __call__(self, x: torch.Tensor) -> torch.Tensor:
-        idn = x if self.down is None else self.down(x)
         if self.conv2.sa is not None and self.conv3.sa is not None:
             # fused chain: conv1 and conv2 emit e4m3 at the NEXT conv's
             # frozen input scale — skips two f32 write/read/quantize passes
             e = _ops.ext()
             N, H, W, C = x.shape
             xq = e.quantize_fp8(x.contiguous(), self.conv1.sa)
+            if self.down is None:
+                idn = x
+            else:
+                # conv1 and down calibrate on the SAME tensor, so their
+                # frozen scales are identical — reuse the quantized bytes
+                idn, _, _ = self.down.run_q(xq, N, H, W, None)
             q1, h, w = self.conv1.run_q(xq, N, H, W, self.conv2.sa)
             q2, h, w = self.conv2.run_q(q1.contiguous(), N, h, w, self.conv3.sa)
             # conv3 fuses the identity add + relu into its epilogue when the
@@ -158,6 +163,7 @@ class Bottleneck:
                 return y
             y, h, w = c3.run_q(q2.contiguous(), N, h, w, None)
             return F.relu(y + idn)
+        idn = x if self.down is None else self.down(x)
         y = self.conv3(self.conv2(self.conv1(x)))  # calibration pass
         return F.relu(y + idn)
 
