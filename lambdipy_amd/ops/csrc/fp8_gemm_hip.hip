@@ -350,6 +350,171 @@ __global__ __launch_bounds__(THREADS2, 2) void gemm_fp8_nt256_kernel(
   }
 }
 
+// ---- tall-skinny fp8 GEMM (N <= 64) on the scaled MFMA -------------------
+// The 1x1 convs with tiny KO (ResNet layer1: N=64, M~400k, K=64..147) fall
+// off every 256-wide tile; the 128^2 v1 kernel wastes half its tile and
+// runs the slow 16x16x32 op. This kernel: 256x64 tile, 8 waves of 32x64
+// (1x2 fragments of 32x32 on the scaled MFMA, unit scales), register-staged
+// double buffer, 80-B padded LDS rows (b128-quarter conflict-free), 50 KiB
+// LDS -> 3 blocks/CU. Handles arbitrary K and N <= 64 tails by zero-fill.
+
+constexpr int BMT = 256, BNT = 64, BKT = 64;   // elements; 64 B rows
+constexpr int ROWT = BKT + 16;                 // 80-B padded
+constexpr int THREADST = 512;
+constexpr int CHT = BMT * BKT / 16 / THREADST; // 2 chunks/thread (A)
+
+typedef __attribute__((ext_vector_type(8))) int fp8ts_frag;
+
+__device__ __forceinline__ void ts_stage_a(const unsigned char* __restrict__ A,
+                                           int M, int K, int bm, int k0,
+                                           i32x4 (&regs)[CHT]) {
+  const int t = threadIdx.x;
+#pragma unroll
+  for (int i = 0; i < CHT; ++i) {
+    const int c = t + i * THREADST;
+    const int r = c >> 2, cell = (c & 3) * 16;
+    const int grow = bm + r, gk = k0 + cell;
+    i32x4 v = {0, 0, 0, 0};
+    if (grow < M) {
+      if (gk + 16 <= K) {
+        v = *reinterpret_cast<const i32x4*>(A + (long)grow * K + gk);
+      } else if (gk < K) {
+        unsigned char z[16] = {0};
+        for (int j = 0; j < 16 && gk + j < K; ++j)
+          z[j] = A[(long)grow * K + gk + j];
+        v = *reinterpret_cast<const i32x4*>(z);
+      }
+    }
+    regs[i] = v;
+  }
+}
+
+__device__ __forceinline__ i32x4 ts_stage_b(const unsigned char* __restrict__ Bt,
+                                            int N, int K, int bn, int k0) {
+  const int t = threadIdx.x;
+  i32x4 v = {0, 0, 0, 0};
+  if (t < 256) {
+    const int r = t >> 2, cell = (t & 3) * 16;
+    const int grow = bn + r, gk = k0 + cell;
+    if (grow < N) {
+      if (gk + 16 <= K) {
+        v = *reinterpret_cast<const i32x4*>(Bt + (long)grow * K + gk);
+      } else if (gk < K) {
+        unsigned char z[16] = {0};
+        for (int j = 0; j < 16 && gk + j < K; ++j)
+          z[j] = Bt[(long)grow * K + gk + j];
+        v = *reinterpret_cast<const i32x4*>(z);
+      }
+    }
+  }
+  return v;
+}
+
+template <bool RELU, bool BIAS>
+__global__ __launch_bounds__(THREADST, 3) void gemm_fp8_ts_kernel(
+    const unsigned char* __restrict__ A, const unsigned char* __restrict__ Bt,
+    float* __restrict__ C, const float* __restrict__ bias, int M, int N,
+    int K, float descale, unsigned tiles_m, unsigned tiles_n) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int ABUF = BMT * ROWT;       // 20 KiB
+  const int BBUF = BNT * ROWT;       // 5 KiB
+  char* const lA = smem;             // [2][256][80]
+  char* const lB = smem + 2 * ABUF;  // [2][64][80]
+
+  const unsigned nwg = tiles_m * tiles_n;
+  const unsigned bid = lpx_xcd_swizzle(blockIdx.x, nwg);
+  const int bm = (int)(bid / tiles_n) * BMT;
+  const int bn = (int)(bid % tiles_n) * BNT;
+
+  const int lane = threadIdx.x & 63;
+  const int w = threadIdx.x >> 6;    // 8 waves, 32 rows each
+  const int fr = lane & 31;
+  const int fkb = lane >> 5;
+
+  f32x16 acc[2];
+#pragma unroll
+  for (int j = 0; j < 2; ++j)
+#pragma unroll
+    for (int e = 0; e < 16; ++e) acc[j][e] = 0.f;
+
+  i32x4 aR[CHT];
+  i32x4 bR;
+  ts_stage_a(A, M, K, bm, 0, aR);
+  bR = ts_stage_b(Bt, N, K, bn, 0);
+  {
+    const int t = threadIdx.x;
+#pragma unroll
+    for (int i = 0; i < CHT; ++i) {
+      const int c = t + i * THREADST;
+      *reinterpret_cast<i32x4*>(lA + (c >> 2) * ROWT + (c & 3) * 16) = aR[i];
+    }
+    if (t < 256)
+      *reinterpret_cast<i32x4*>(lB + (t >> 2) * ROWT + (t & 3) * 16) = bR;
+  }
+  __syncthreads();
+
+  const int ksteps = (K + BKT - 1) / BKT;
+  int cur = 0;
+  for (int ks = 0; ks < ksteps; ++ks) {
+    if (ks + 1 < ksteps) {
+      ts_stage_a(A, M, K, bm, (ks + 1) * BKT, aR);
+      bR = ts_stage_b(Bt, N, K, bn, (ks + 1) * BKT);
+    }
+    const char* aCur = lA + cur * ABUF;
+    const char* bCur = lB + cur * BBUF;
+    fp8ts_frag aF, bF[2];
+    {
+      const char* p = aCur + (w * 32 + fr) * ROWT + fkb * 32;
+      const i32x4 lo = *reinterpret_cast<const i32x4*>(p);
+      const i32x4 hi = *reinterpret_cast<const i32x4*>(p + 16);
+      aF = fp8ts_frag{lo.x, lo.y, lo.z, lo.w, hi.x, hi.y, hi.z, hi.w};
+    }
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf) {
+      const char* p = bCur + (nf * 32 + fr) * ROWT + fkb * 32;
+      const i32x4 lo = *reinterpret_cast<const i32x4*>(p);
+      const i32x4 hi = *reinterpret_cast<const i32x4*>(p + 16);
+      bF[nf] = fp8ts_frag{lo.x, lo.y, lo.z, lo.w, hi.x, hi.y, hi.z, hi.w};
+    }
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf)
+      acc[nf] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+          aF, bF[nf], acc[nf], 0, 0, 0, 127, 0, 127);
+    __syncthreads();
+    if (ks + 1 < ksteps) {
+      const int t = threadIdx.x;
+      char* aN = lA + (cur ^ 1) * ABUF;
+#pragma unroll
+      for (int i = 0; i < CHT; ++i) {
+        const int c = t + i * THREADST;
+        *reinterpret_cast<i32x4*>(aN + (c >> 2) * ROWT + (c & 3) * 16) = aR[i];
+      }
+      if (t < 256)
+        *reinterpret_cast<i32x4*>(lB + (cur ^ 1) * BBUF + (t >> 2) * ROWT +
+                                  (t & 3) * 16) = bR;
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  const int ocol_l = lane & 31;
+  const int orow_base = 4 * (lane >> 5);
+#pragma unroll
+  for (int nf = 0; nf < 2; ++nf) {
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int row = bm + w * 32 + (r & 3) + 8 * (r >> 2) + orow_base;
+      const int col = bn + nf * 32 + ocol_l;
+      if (row < M && col < N) {
+        float v = acc[nf][r] * descale;
+        if constexpr (BIAS) v += bias[col];
+        if constexpr (RELU) v = v > 0.f ? v : 0.f;
+        C[(long)row * N + col] = v;
+      }
+    }
+  }
+}
+
 // ---- implicit-GEMM fp8 conv (v2 structure) -------------------------------
 // out[m = (n,ho,wo)][ko] = sum_{r,s,c} xq[n][ho*st+r-p][wo*st+s-p][c] * w[ko][r,s,c]
 // The A staging gathers 16-B chunks of pre-quantized NHWC activations
@@ -707,6 +872,25 @@ hipError_t gemm_fp8_nt_res(const unsigned char* A, const unsigned char* Bt,
                                stream);
   if (!resid && M >= 1024 && N >= 1024 && gemm_fp8_8ph_supported(M, N, K))
     return gemm_fp8_nt8ph(A, Bt, C, bias, M, N, K, descale, relu, stream);
+  if (!resid && N <= 64 && M >= 2048) {
+    // tall-skinny (ResNet small-KO 1x1 convs): 256x64 scaled-MFMA tile
+    const unsigned tm = lpx_ceil_div(M, BMT), tn = lpx_ceil_div(N, BNT);
+    const size_t tlds = (size_t)2 * (BMT + BNT) * ROWT;  // 50 KiB
+    dim3 tg(tm * tn), tb(THREADST);
+    if (bias && relu)
+     hipLaunchKernelGGL(( gemm_fp8_ts_kernel<true, true>), dim3(tg), dim3(tb), tlds, stream, 
+          A, Bt, C, bias, M, N, K, descale, tm, tn);
+    else if (bias)
+     hipLaunchKernelGGL(( gemm_fp8_ts_kernel<false, true>), dim3(tg), dim3(tb), tlds, stream, 
+          A, Bt, C, bias, M, N, K, descale, tm, tn);
+    else if (relu)
+     hipLaunchKernelGGL(( gemm_fp8_ts_kernel<true, false>), dim3(tg), dim3(tb), tlds, stream, 
+          A, Bt, C, bias, M, N, K, descale, tm, tn);
+    else
+     hipLaunchKernelGGL(( gemm_fp8_ts_kernel<false, false>), dim3(tg), dim3(tb), tlds, stream, 
+          A, Bt, C, bias, M, N, K, descale, tm, tn);
+    return hipGetLastError();
+  }
   const bool big = (M >= 512 && N >= 512 && K >= 64);
   if (big) {
     const unsigned tiles_m = lpx_ceil_div(M, BM2), tiles_n = lpx_ceil_div(N, BN2);

@@ -381,6 +381,23 @@ def test_gemm_mx_w4_vs_dequant_ref(ops):
     assert rel < 0.15, rel
 
 
+def test_gemm_fp8_tall_skinny(ops):
+    # N<=64, large M (ResNet layer1 1x1 conv shape class, K-tail 147)
+    e = ops.ext()
+    g = torch.Generator(device="cpu").manual_seed(22)
+    for M, N, K in ((4096, 64, 147), (2048, 48, 64), (2304, 64, 128)):
+        A = torch.randn(M, K, generator=g).to(DEV)
+        Bt = torch.randn(N, K, generator=g).to(DEV)
+        bias = torch.randn(N, generator=g).to(DEV)
+        Aq = e.quantize_fp8(A, 1.0)
+        Btq = e.quantize_fp8(Bt, 1.0)
+        C = e.gemm_fp8(Aq, Btq, 1.0, bias, True)
+        ref = torch.relu(Aq.view(torch.float8_e4m3fn).float()
+                         @ Btq.view(torch.float8_e4m3fn).float().t() + bias)
+        err = (C - ref).abs().max().item()
+        assert err < 1e-2, (M, N, K, err)
+
+
 def test_linear_w4_matches_f32(ops):
     from lambdipy_amd.shims.lpx_torch import LinearW4
     g = torch.Generator(device="cpu").manual_seed(20)
