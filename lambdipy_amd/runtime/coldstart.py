@@ -146,10 +146,14 @@ def cold_start(
         # the torch-free rf_lite handler imports ONLY lpx_rt + the AOT
         # .hsaco: extract those eagerly too and skip the ready-marker wait
         # entirely (the bulk unzip overlaps the whole GPU cold start).
+        # The torch extension .so in lpx_ops/ (~14 MB) is NOT part of the
+        # rf_lite path — extracting it eagerly would serialize ~0.1 s of
+        # inflate before the spawn, so it goes to the deferred bulk pass.
         if handler == "rf_lite":
             def early(n):
                 return (n.startswith("lpx_handlers/") or n == "lpx_fanout.py"
-                        or n == "lpx_rt.py" or n.startswith("lpx_ops/"))
+                        or n == "lpx_rt.py"
+                        or (n.startswith("lpx_ops/") and not n.endswith(".so")))
         else:
             def early(n):
                 return n.startswith("lpx_handlers/") or n == "lpx_fanout.py"
