@@ -220,12 +220,29 @@ def cold_start(
     report = {}
     total_ms = None
     try:
-        out0, err0 = procs[0].communicate(timeout=timeout_s)
-        total_ms = (time.perf_counter() - t0) * 1e3
-        for line in out0.splitlines():
+        # t1 is taken when the LPX_RESULT line ARRIVES (the docstring
+        # contract), not when the process exits — exit adds ~50 ms of
+        # Python + HIP runtime teardown that is not cold-start latency.
+        import selectors
+
+        sel = selectors.DefaultSelector()
+        sel.register(procs[0].stdout, selectors.EVENT_READ)
+        deadline = time.perf_counter() + timeout_s
+        lines0 = []
+        while True:
+            if not sel.select(timeout=max(0.0, deadline - time.perf_counter())):
+                raise subprocess.TimeoutExpired(argv, timeout_s)
+            line = procs[0].stdout.readline()
+            if line == "":  # EOF before LPX_RESULT — collect stderr below
+                break
+            lines0.append(line)
             if line.startswith("LPX_RESULT "):
+                total_ms = (time.perf_counter() - t0) * 1e3
                 report = json.loads(line[len("LPX_RESULT "):])
                 break
+        sel.close()
+        rest0, err0 = procs[0].communicate(timeout=timeout_s)
+        out0 = "".join(lines0) + (rest0 or "")
         if not report:
             raise RuntimeError(
                 f"handler produced no LPX_RESULT (rc={procs[0].returncode})\n"
