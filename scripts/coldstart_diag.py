@@ -59,7 +59,17 @@ def main():
     if "--one" in sys.argv:
         print("DIAG " + json.dumps(probe()), flush=True)
         return
-    for name, env in VARIANTS.items():
+    # interleaved repeats: the FIRST process on a box absorbs one-time
+    # driver/KFD warm-up, so run each variant several times alternating
+    order = [(n, VARIANTS[n]) for n in VARIANTS]
+    reps = int(os.environ.get("LPX_DIAG_REPS", "1"))
+    for r in range(reps):
+        for name, env in order:
+            _run(name + f"#{r}", env)
+
+
+def _run(name, env):
+    if True:
         e = dict(os.environ, **env)
         out = subprocess.run(
             [sys.executable, os.path.abspath(__file__), "--one"],
