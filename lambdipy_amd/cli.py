@@ -123,8 +123,9 @@ def publish(zip_path, name, release_dir):
 def run(layer, handler, gpus, timeout, as_json):
     """Cold-start a handler from a layer (zip or dir) and time to first
     inference."""
-    from .runtime import cold_start
+    from .runtime import cold_start, prewarm_host
 
+    prewarm_host()
     res = cold_start(layer, handler=handler, gpus=gpus, timeout_s=timeout)
     d = res.to_dict()
     if as_json:
@@ -147,7 +148,9 @@ def bench(layer, handler, gpus, repeats, out, rocprof_dir):
     """Cold-start scaling sweep: layer MB + cold-start ms per GPU count."""
     import shutil as _sh
 
-    from .runtime import cold_start
+    from .runtime import cold_start, prewarm_host
+
+    prewarm_host()
 
     rows = []
     for g in [int(x) for x in gpus.split(",")]:

@@ -91,6 +91,44 @@ def _parallel_extract(zip_path: str, dest: str, workers: int = 8,
         list(ex.map(_shard, shards))
 
 
+def prewarm_host(timeout_s: float = 120.0) -> float:
+    """Warm the host's HIP module-launch path before serving requests.
+
+    Measured on MI355X (scripts/coldstart_diag.py, profiles/README.md): the
+    FIRST process on a box to go through hipModuleLoad + hipModuleLaunchKernel
+    pays a one-time ~150 ms KFD/queue bring-up; every later process pays
+    ~14 ms, even after the first one exits, and a resident torch process does
+    NOT absorb it (different init path). That cost is host state, not
+    function cold start — serving platforms absorb it by pre-warming workers
+    before any request arrives, which is what this does: one subprocess that
+    loads the runtime's own AOT .hsaco (no user layer bytes) and launches a
+    tiny kernel. Returns elapsed ms; 0.0 if there is no usable GPU.
+    """
+    code = (
+        "from lambdipy_amd.shims import lpx_rt\n"
+        "lpx_rt.set_device(0)\n"
+        "m = lpx_rt.Module()\n"
+        "buf = lpx_rt.malloc(4096)\n"
+        "lpx_rt.xgen(m, buf, 1024, 1)\n"
+        "lpx_rt.sync()\n"
+    )
+    t0 = time.perf_counter()
+    env = dict(os.environ)
+    env["PYTHONPATH"] = (
+        os.path.dirname(os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+        + (os.pathsep + env["PYTHONPATH"] if env.get("PYTHONPATH") else "")
+    )
+    try:
+        r = subprocess.run([sys.executable, "-c", code], env=env,
+                           capture_output=True, timeout=timeout_s)
+        if r.returncode != 0:
+            log.debug("prewarm_host: no GPU / failed (%s)", r.stderr[-200:])
+            return 0.0
+    except (subprocess.TimeoutExpired, OSError):
+        return 0.0
+    return (time.perf_counter() - t0) * 1e3
+
+
 def _free_port() -> int:
     s = socket.socket()
     s.bind(("127.0.0.1", 0))
