@@ -250,10 +250,24 @@ def cold_start(
                                  text=True)
             )
 
+    # bulk extraction runs CONCURRENTLY with the handler (zlib releases the
+    # GIL): first-inference latency must not wait on layer bytes the handler
+    # does not read — the rf_lite path needs only its eager set, and the
+    # torch handlers gate themselves on the ready marker this thread writes.
+    extract_thread = None
+    extract_err: List[BaseException] = []
+    tex0 = time.perf_counter()
     if deferred_extract is not None:
-        tex0 = time.perf_counter()
-        deferred_extract()
-        unzip_ms += (time.perf_counter() - tex0) * 1e3
+        import threading
+
+        def _bg_extract():
+            try:
+                deferred_extract()
+            except BaseException as e:  # surfaced after the result read
+                extract_err.append(e)
+
+        extract_thread = threading.Thread(target=_bg_extract, daemon=True)
+        extract_thread.start()
 
     report = {}
     total_ms = None
@@ -279,6 +293,11 @@ def cold_start(
                 report = json.loads(line[len("LPX_RESULT "):])
                 break
         sel.close()
+        if extract_thread is not None:
+            extract_thread.join(timeout=timeout_s)
+            unzip_ms += (time.perf_counter() - tex0) * 1e3
+            if extract_err:
+                raise extract_err[0]
         rest0, err0 = procs[0].communicate(timeout=timeout_s)
         out0 = "".join(lines0) + (rest0 or "")
         if not report:
@@ -297,6 +316,8 @@ def cold_start(
         for p in procs:
             if p.poll() is None:
                 p.kill()
+        if extract_thread is not None and extract_thread.is_alive():
+            extract_thread.join(timeout=60)  # never rmtree under the writer
         if tmpdir and not keep_dir:
             shutil.rmtree(tmpdir, ignore_errors=True)
 
