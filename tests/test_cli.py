@@ -76,3 +76,30 @@ def test_fetch_missing(runner, tmp_path):
     r = runner.invoke(cli_main, ["fetch", "nope", "--release-dir",
                                  str(tmp_path / "empty")])
     assert r.exit_code != 0
+
+
+def test_cold_start_t1_at_result_not_exit(tmp_path):
+    """t1 must be taken when LPX_RESULT arrives, not at process exit
+    (coldstart.py streams stdout; a handler that lingers after printing
+    must not inflate total_ms), and bulk extraction must overlap."""
+    import json as _json
+    import zipfile
+
+    from lambdipy_amd.runtime import cold_start
+
+    zp = tmp_path / "layer.zip"
+    handler = (
+        "import json, time\n"
+        "def main():\n"
+        "    print('LPX_RESULT ' + json.dumps({'handler': 'lingerer'}),"
+        " flush=True)\n"
+        "    time.sleep(1.5)\n"  # teardown time that must NOT be timed
+    )
+    with zipfile.ZipFile(zp, "w") as zf:
+        zf.writestr("lpx_handlers/__init__.py", "")
+        zf.writestr("lpx_handlers/linger_handler.py", handler)
+        zf.writestr("bulk/pad.bin", b"\0" * (1 << 20))  # deferred-set file
+    res = cold_start(str(zp), handler="linger_handler", gpus=1, timeout_s=60)
+    assert res.handler_report == {"handler": "lingerer"}
+    # the 1.5 s linger must not be inside total_ms
+    assert res.total_ms < 1200, res.total_ms
