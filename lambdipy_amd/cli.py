@@ -73,6 +73,12 @@ def build(requirements, output, zip_path, handlers, no_transitive, as_json):
         res, output, zip_path=zip_path, wheel_builder=build_recipe_artifacts
     )
     if handlers:
+        handlers = list(handlers)
+        if "rf" in handlers and "rf_lite" not in handlers:
+            # the torch-free fast path of the same config (~50 KB: lpx_rt +
+            # AOT .hsaco) — always ship it next to the torch rf handler so
+            # `lpx run/bench --handler rf_lite` works on any rf layer
+            handlers.append("rf_lite")
         _install_handlers(output, handlers)
         if zip_path:  # re-zip with handlers included
             from .packager import make_zip, zip_report
