@@ -78,6 +78,15 @@ hipError_t rf_predict_heap_f32(const int* nodes_heap, const float* leaf_values,
                                int n_trees, int depth, const float* X,
                                float* out, int S, int F, int C, int silp,
                                hipStream_t stream);
+// v6: child-pair heap — pairs = int32 [n_trees*(2^(depth-1)-1), 4] (node n's
+// two children as one 16-B record), roots = int32 [n_trees, 2]; one random
+// 16-B request per level instead of one 8-B (the v4 kernel is L2
+// request-bound).
+hipError_t rf_predict_heap_pair_f32(const int* pairs, const int* roots,
+                                    const float* leaf_values, int n_trees,
+                                    int depth, const float* X, float* out,
+                                    int S, int F, int C, int silp,
+                                    hipStream_t stream);
 // v5: two-level heap — top `top` levels served from LDS (top<0: auto-pick
 // largest fitting 2-blocks/CU; falls back to v4 when nothing fits).
 hipError_t rf_predict_heap2_f32(const int* nodes_heap, const float* leaf_values,

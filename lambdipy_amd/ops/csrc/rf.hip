@@ -354,6 +354,101 @@ __global__ __launch_bounds__(BLK) void rf_heap_kernel(
   }
 }
 
+// v6: child-PAIR heap traversal — nodes repacked so node n's two children
+// (heap 2n+1, 2n+2) sit in ONE 16-B record pairs[tr][n] (n < 2^(depth-1)-1;
+// roots in a separate [n_trees] i32x2 array). A level step loads the child
+// pair of the CURRENT node while comparing it, then selects the next node
+// from registers: depth-1 random 16-B L2 requests per tree instead of
+// depth 8-B requests. The v4 kernel is L2 REQUEST-bound (0.895 wait frac,
+// invariant to block size / occupancy — profiles/README), and the L2 moves
+// a full line per random request either way, so ~10% fewer requests plus a
+// root served from a 2-KB cached array is a direct request-count cut.
+template <bool STAGE_X, int SILP>
+__global__ __launch_bounds__(THREADS) void rf_heap_pair_kernel(
+    const i32x4* __restrict__ pairs, const i32x2* __restrict__ roots,
+    const float* __restrict__ leaf_values, int n_trees, int depth,
+    const float* __restrict__ X, float* __restrict__ out, int S, int F,
+    int C) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* xl = reinterpret_cast<float*>(smem);
+  const int t = threadIdx.x;
+  const int rowf = F + 1;
+  const int tree_stride = (1 << depth) - 1;
+  const int ppt = (1 << (depth - 1)) - 1;  // pairs per tree
+  const int n_leaves = 1 << depth;
+
+  for (int tile = blockIdx.x * (THREADS * SILP); tile < S;
+       tile += gridDim.x * (THREADS * SILP)) {
+    const int tile_n = min(THREADS * SILP, S - tile);
+    if (STAGE_X) {
+      const long total = (long)tile_n * F;
+      for (long i = t; i < total; i += THREADS) {
+        const int r = (int)(i / F), c = (int)(i - (long)r * F);
+        xl[r * rowf + c] = X[((long)tile + r) * F + c];
+      }
+      __syncthreads();
+    }
+
+    int s[SILP];
+    const float* xrow[SILP];
+#pragma unroll
+    for (int j = 0; j < SILP; ++j) {
+      s[j] = tile + t + j * THREADS;
+      xrow[j] = STAGE_X ? (xl + (t + j * THREADS) * rowf)
+                        : (X + (long)s[j] * F);
+    }
+    float acc[SILP][CMAX];
+#pragma unroll
+    for (int j = 0; j < SILP; ++j)
+#pragma unroll
+      for (int c = 0; c < CMAX; ++c) acc[j][c] = 0.f;
+
+    for (int tr = 0; tr < n_trees; ++tr) {
+      const i32x4* tp = pairs + (long)tr * ppt;
+      const i32x2 root = roots[tr];
+      int nd[SILP];        // current heap index
+      int cf[SILP];        // current node feature
+      float ct[SILP];      // current node threshold
+#pragma unroll
+      for (int j = 0; j < SILP; ++j) {
+        nd[j] = 0;
+        cf[j] = root.x;
+        ct[j] = __int_as_float(root.y);
+      }
+      for (int d = 0; d < depth - 1; ++d) {
+#pragma unroll
+        for (int j = 0; j < SILP; ++j) {
+          if (s[j] < S) {
+            const i32x4 pr = tp[nd[j]];  // children of current node
+            const int go = (xrow[j][(unsigned)cf[j]] <= ct[j]) ? 0 : 1;
+            nd[j] = 2 * nd[j] + 1 + go;
+            cf[j] = go ? pr.z : pr.x;
+            ct[j] = __int_as_float(go ? pr.w : pr.y);
+          }
+        }
+      }
+#pragma unroll
+      for (int j = 0; j < SILP; ++j) {
+        if (s[j] < S) {
+          const int go = (xrow[j][(unsigned)cf[j]] <= ct[j]) ? 0 : 1;
+          const int leaf = 2 * nd[j] + 1 + go - tree_stride;
+          const float* lv = leaf_values + ((long)tr * n_leaves + leaf) * C;
+#pragma unroll
+          for (int c = 0; c < CMAX; ++c)
+            if (c < C) acc[j][c] += lv[c];
+        }
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < SILP; ++j)
+      if (s[j] < S)
+#pragma unroll
+        for (int c = 0; c < CMAX; ++c)
+          if (c < C) out[(long)s[j] * C + c] = acc[j][c];
+    if (STAGE_X) __syncthreads();
+  }
+}
+
 // v5: two-level heap traversal — the top `top` levels of every tree are
 // cached in LDS (loaded once per block), the remaining levels read from
 // L2/HBM. The kernel is L2-request-bound (243 Ms/s = ~5 TB/s of 8-B node
@@ -457,6 +552,40 @@ __global__ __launch_bounds__(THREADS) void rf_heap2_kernel(
 }
 
 }  // namespace
+
+hipError_t rf_predict_heap_pair_f32(const int* pairs, const int* roots,
+                                    const float* leaf_values, int n_trees,
+                                    int depth, const float* X, float* out,
+                                    int S, int F, int C, int silp,
+                                    hipStream_t stream) {
+  if (C > CMAX || depth < 2 || depth > 26) return hipErrorInvalidValue;
+  const int SILP = (silp == 2) ? 2 : (silp == 4 ? 4 : 1);
+  const bool stage = F <= F_LDS_MAX;
+  const size_t lds = stage ? (size_t)THREADS * SILP * (F + 1) * 4 : 0;
+  if (lds > 160 * 1024) return hipErrorInvalidValue;
+  unsigned blocks = lpx_ceil_div(S, THREADS * SILP);
+  if (blocks > 4096u) blocks = 4096u;
+  if (blocks == 0) blocks = 1;
+  const auto* pp = reinterpret_cast<const i32x4*>(pairs);
+  const auto* rr = reinterpret_cast<const i32x2*>(roots);
+#define LPX_RFHP_CASE(I)                                                      \
+  if (SILP == I) {                                                            \
+    if (stage)                                                                \
+      rf_heap_pair_kernel<true, I>                                            \
+          <<<dim3(blocks), dim3(THREADS), lds, stream>>>(                     \
+              pp, rr, leaf_values, n_trees, depth, X, out, S, F, C);          \
+    else                                                                      \
+      rf_heap_pair_kernel<false, I>                                           \
+          <<<dim3(blocks), dim3(THREADS), 0, stream>>>(                       \
+              pp, rr, leaf_values, n_trees, depth, X, out, S, F, C);          \
+    return hipGetLastError();                                                 \
+  }
+  LPX_RFHP_CASE(1)
+  LPX_RFHP_CASE(2)
+  LPX_RFHP_CASE(4)
+#undef LPX_RFHP_CASE
+  return hipErrorInvalidValue;
+}
 
 hipError_t rf_predict_heap2_f32(const int* nodes_heap, const float* leaf_values,
                                 int n_trees, int depth, const float* X,

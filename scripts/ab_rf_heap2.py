@@ -39,6 +39,26 @@ def main():
             except RuntimeError as ex:
                 print(f"silp={silp} top={top:2d}: ERR {ex}")
 
+    # v6 child-pair layout: repack heap nodes -> (pairs, roots)
+    depth = 10
+    stride = (1 << depth) - 1
+    T = n.shape[0] // stride
+    nh = n.view(T, stride, 2)
+    ppt = (1 << (depth - 1)) - 1
+    idx = torch.arange(ppt, device=n.device)
+    pairs = torch.cat([nh[:, 2 * idx + 1, :], nh[:, 2 * idx + 2, :]],
+                      dim=2).reshape(T * ppt, 4).contiguous()
+    roots = nh[:, 0, :].contiguous()
+    for silp in (1, 2, 4):
+        try:
+            out = e.rf_predict_heap_pair(pairs, roots, lv, depth, X, silp)
+            ok = torch.allclose(out, ref, atol=1e-4)
+            dt = bench(lambda: e.rf_predict_heap_pair(pairs, roots, lv, depth,
+                                                      X, silp))
+            print(f"pair silp={silp}: {S/dt/1e6:7.1f} Ms/s  ok={ok}")
+        except RuntimeError as ex:
+            print(f"pair silp={silp}: ERR {ex}")
+
 
 if __name__ == "__main__":
     main()
