@@ -78,6 +78,14 @@ hipError_t rf_predict_heap_f32(const int* nodes_heap, const float* leaf_values,
                                int n_trees, int depth, const float* X,
                                float* out, int S, int F, int C, int silp,
                                hipStream_t stream);
+// v7: quad-record heap — quads = int32 [n_trees*(2^depth-1)/3, 16]
+// (64-B records: children + grandchildren of each even-level node); two
+// levels per random fetch. Even depth only.
+hipError_t rf_predict_heap_quad_f32(const int* quads, const int* roots,
+                                    const float* leaf_values, int n_trees,
+                                    int depth, const float* X, float* out,
+                                    int S, int F, int C, int silp,
+                                    hipStream_t stream);
 // v6: child-pair heap — pairs = int32 [n_trees*(2^(depth-1)-1), 4] (node n's
 // two children as one 16-B record), roots = int32 [n_trees, 2]; one random
 // 16-B request per level instead of one 8-B (the v4 kernel is L2

@@ -592,6 +592,38 @@ at::Tensor rf_predict_heap(const at::Tensor& nodes_heap,
   return out;
 }
 
+at::Tensor rf_predict_heap_quad(const at::Tensor& quads,
+                                const at::Tensor& roots,
+                                const at::Tensor& leaf_values, int64_t depth,
+                                const at::Tensor& X, int64_t silp) {
+  CHECK_DEV(quads);
+  CHECK_DEV(roots);
+  CHECK_DEV(leaf_values);
+  CHECK_DEV(X);
+  TORCH_CHECK(quads.scalar_type() == at::kInt && quads.dim() == 2 &&
+              quads.size(1) == 16, "quads must be int32 [n, 16]");
+  TORCH_CHECK(roots.scalar_type() == at::kInt && roots.dim() == 2 &&
+              roots.size(1) == 2, "roots must be int32 [n_trees, 2]");
+  TORCH_CHECK(X.scalar_type() == at::kFloat && X.dim() == 2,
+              "X must be f32 [S,F]");
+  const int64_t S = X.size(0), F = X.size(1), C = leaf_values.size(1);
+  TORCH_CHECK(C <= 8, "rf_predict_heap_quad: C<=8 per call");
+  const int64_t n_trees = roots.size(0);
+  const int64_t qpt = ((1LL << depth) - 1) / 3;
+  TORCH_CHECK(quads.size(0) == n_trees * qpt, "quads/depth mismatch");
+  TORCH_CHECK(leaf_values.size(0) == n_trees * (1LL << depth),
+              "leaf_values rows != n_trees * 2^depth");
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(X.device());
+  auto out = at::empty({S, C}, X.options());
+  check_hip(lpx::rf_predict_heap_quad_f32(
+                quads.data_ptr<int>(), roots.data_ptr<int>(),
+                leaf_values.data_ptr<float>(), (int)n_trees, (int)depth,
+                X.data_ptr<float>(), out.data_ptr<float>(), (int)S, (int)F,
+                (int)C, (int)silp, current_stream(X)),
+            "rf_predict_heap_quad_f32");
+  return out;
+}
+
 at::Tensor rf_predict_heap_pair(const at::Tensor& pairs,
                                 const at::Tensor& roots,
                                 const at::Tensor& leaf_values, int64_t depth,
@@ -680,6 +712,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "RandomForest per-class sums (heap layout, complete forests)",
         py::arg("nodes_heap"), py::arg("leaf_values"), py::arg("depth"),
         py::arg("X"), py::arg("silp") = 1, py::arg("top") = -2);
+  m.def("rf_predict_heap_quad", &rf_predict_heap_quad,
+        "RandomForest per-class sums (quad-record heap layout)",
+        py::arg("quads"), py::arg("roots"), py::arg("leaf_values"),
+        py::arg("depth"), py::arg("X"), py::arg("silp") = 1);
   m.def("rf_predict_heap_pair", &rf_predict_heap_pair,
         "RandomForest per-class sums (child-pair heap layout)",
         py::arg("pairs"), py::arg("roots"), py::arg("leaf_values"),

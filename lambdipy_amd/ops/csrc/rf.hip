@@ -449,6 +449,115 @@ __global__ __launch_bounds__(THREADS) void rf_heap_pair_kernel(
   }
 }
 
+// v7: QUAD records — two tree levels per random fetch. Record r for an
+// even-level node n (level d=2k, within-level offset o, record slot
+// (4^k-1)/3 + o) is 64 B: {children(n) [2 nodes], grandchildren(n)
+// [4 nodes], 16 B pad}. One 64-B-aligned record read = 3 dwordx4 loads of
+// the SAME line (first misses L1 -> one L2 request, the rest hit L1), and
+// serves TWO compares: depth/2 L2 requests per tree instead of depth-1.
+// Slot recurrence mirrors the heap with branching factor 4:
+// slot' = 4*slot + 1 + (2*b1 + b2). Requires even depth (launcher
+// falls back to the pair kernel otherwise). Memory: 2x the flat heap
+// (each node appears in its parent's and grandparent's records).
+template <bool STAGE_X, int SILP>
+__global__ __launch_bounds__(THREADS) void rf_heap_quad_kernel(
+    const i32x4* __restrict__ quads, const i32x2* __restrict__ roots,
+    const float* __restrict__ leaf_values, int n_trees, int depth,
+    const float* __restrict__ X, float* __restrict__ out, int S, int F,
+    int C) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* xl = reinterpret_cast<float*>(smem);
+  const int t = threadIdx.x;
+  const int rowf = F + 1;
+  const int tree_stride = (1 << depth) - 1;
+  const int qpt = ((1 << depth) - 1) / 3;  // (4^(depth/2) - 1) / 3 records
+  const int n_leaves = 1 << depth;
+
+  for (int tile = blockIdx.x * (THREADS * SILP); tile < S;
+       tile += gridDim.x * (THREADS * SILP)) {
+    const int tile_n = min(THREADS * SILP, S - tile);
+    if (STAGE_X) {
+      const long total = (long)tile_n * F;
+      for (long i = t; i < total; i += THREADS) {
+        const int r = (int)(i / F), c = (int)(i - (long)r * F);
+        xl[r * rowf + c] = X[((long)tile + r) * F + c];
+      }
+      __syncthreads();
+    }
+
+    int s[SILP];
+    const float* xrow[SILP];
+#pragma unroll
+    for (int j = 0; j < SILP; ++j) {
+      s[j] = tile + t + j * THREADS;
+      xrow[j] = STAGE_X ? (xl + (t + j * THREADS) * rowf)
+                        : (X + (long)s[j] * F);
+    }
+    float acc[SILP][CMAX];
+#pragma unroll
+    for (int j = 0; j < SILP; ++j)
+#pragma unroll
+      for (int c = 0; c < CMAX; ++c) acc[j][c] = 0.f;
+
+    for (int tr = 0; tr < n_trees; ++tr) {
+      const i32x4* tq = quads + (long)tr * qpt * 4;
+      const i32x2 root = roots[tr];
+      int slot[SILP], nn[SILP], cf[SILP];
+      float ct[SILP];
+#pragma unroll
+      for (int j = 0; j < SILP; ++j) {
+        slot[j] = 0;
+        nn[j] = 0;
+        cf[j] = root.x;
+        ct[j] = __int_as_float(root.y);
+      }
+      for (int k = 0; k < (depth >> 1) - 1; ++k) {  // full two-level steps
+#pragma unroll
+        for (int j = 0; j < SILP; ++j) {
+          if (s[j] < S) {
+            const i32x4* rec = tq + slot[j] * 4;
+            const i32x4 ch = rec[0];
+            const i32x4 g01 = rec[1];
+            const i32x4 g23 = rec[2];
+            const int b1 = (xrow[j][(unsigned)cf[j]] <= ct[j]) ? 0 : 1;
+            const int chf = b1 ? ch.z : ch.x;
+            const float cht = __int_as_float(b1 ? ch.w : ch.y);
+            const int b2 = (xrow[j][(unsigned)chf] <= cht) ? 0 : 1;
+            const int q = 2 * b1 + b2;
+            const i32x4 gg = (q < 2) ? g01 : g23;
+            cf[j] = (q & 1) ? gg.z : gg.x;
+            ct[j] = __int_as_float((q & 1) ? gg.w : gg.y);
+            slot[j] = 4 * slot[j] + 1 + q;
+            nn[j] = 4 * nn[j] + 3 + q;
+          }
+        }
+      }
+#pragma unroll
+      for (int j = 0; j < SILP; ++j) {  // partial step: children then leaves
+        if (s[j] < S) {
+          const i32x4 ch = tq[slot[j] * 4];
+          const int b1 = (xrow[j][(unsigned)cf[j]] <= ct[j]) ? 0 : 1;
+          const int chf = b1 ? ch.z : ch.x;
+          const float cht = __int_as_float(b1 ? ch.w : ch.y);
+          const int b2 = (xrow[j][(unsigned)chf] <= cht) ? 0 : 1;
+          const int leaf = 4 * nn[j] + 3 + 2 * b1 + b2 - tree_stride;
+          const float* lv = leaf_values + ((long)tr * n_leaves + leaf) * C;
+#pragma unroll
+          for (int c = 0; c < CMAX; ++c)
+            if (c < C) acc[j][c] += lv[c];
+        }
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < SILP; ++j)
+      if (s[j] < S)
+#pragma unroll
+        for (int c = 0; c < CMAX; ++c)
+          if (c < C) out[(long)s[j] * C + c] = acc[j][c];
+    if (STAGE_X) __syncthreads();
+  }
+}
+
 // v5: two-level heap traversal — the top `top` levels of every tree are
 // cached in LDS (loaded once per block), the remaining levels read from
 // L2/HBM. The kernel is L2-request-bound (243 Ms/s = ~5 TB/s of 8-B node
@@ -552,6 +661,41 @@ __global__ __launch_bounds__(THREADS) void rf_heap2_kernel(
 }
 
 }  // namespace
+
+hipError_t rf_predict_heap_quad_f32(const int* quads, const int* roots,
+                                    const float* leaf_values, int n_trees,
+                                    int depth, const float* X, float* out,
+                                    int S, int F, int C, int silp,
+                                    hipStream_t stream) {
+  if (C > CMAX || depth < 4 || depth > 26 || (depth & 1))
+    return hipErrorInvalidValue;
+  const int SILP = (silp == 2) ? 2 : (silp == 4 ? 4 : 1);
+  const bool stage = F <= F_LDS_MAX;
+  const size_t lds = stage ? (size_t)THREADS * SILP * (F + 1) * 4 : 0;
+  if (lds > 160 * 1024) return hipErrorInvalidValue;
+  unsigned blocks = lpx_ceil_div(S, THREADS * SILP);
+  if (blocks > 4096u) blocks = 4096u;
+  if (blocks == 0) blocks = 1;
+  const auto* qq = reinterpret_cast<const i32x4*>(quads);
+  const auto* rr = reinterpret_cast<const i32x2*>(roots);
+#define LPX_RFHQ_CASE(I)                                                      \
+  if (SILP == I) {                                                            \
+    if (stage)                                                                \
+      rf_heap_quad_kernel<true, I>                                            \
+          <<<dim3(blocks), dim3(THREADS), lds, stream>>>(                     \
+              qq, rr, leaf_values, n_trees, depth, X, out, S, F, C);          \
+    else                                                                      \
+      rf_heap_quad_kernel<false, I>                                           \
+          <<<dim3(blocks), dim3(THREADS), 0, stream>>>(                       \
+              qq, rr, leaf_values, n_trees, depth, X, out, S, F, C);          \
+    return hipGetLastError();                                                 \
+  }
+  LPX_RFHQ_CASE(1)
+  LPX_RFHQ_CASE(2)
+  LPX_RFHQ_CASE(4)
+#undef LPX_RFHQ_CASE
+  return hipErrorInvalidValue;
+}
 
 hipError_t rf_predict_heap_pair_f32(const int* pairs, const int* roots,
                                     const float* leaf_values, int n_trees,

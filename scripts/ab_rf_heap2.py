@@ -59,6 +59,30 @@ def main():
         except RuntimeError as ex:
             print(f"pair silp={silp}: ERR {ex}")
 
+    # v7 quad records: 64-B records {children, grandchildren} per even-level
+    # node; record slot recurrence slot' = 4*slot + 1 + (2*b1+b2)
+    qpt = ((1 << depth) - 1) // 3
+    quads = torch.zeros(T, qpt, 16, dtype=torch.int32, device=n.device)
+    for k in range(depth // 2):
+        base = (4**k - 1) // 3
+        o = torch.arange(4**k, device=n.device)
+        nn = (1 << (2 * k)) - 1 + o  # heap index of the even-level node
+        quads[:, base + o, 0:2] = nh[:, 2 * nn + 1, :]
+        quads[:, base + o, 2:4] = nh[:, 2 * nn + 2, :]
+        if 2 * k + 2 < depth:
+            for j in range(4):
+                quads[:, base + o, 4 + 2 * j:6 + 2 * j] = nh[:, 4 * nn + 3 + j, :]
+    quads = quads.reshape(T * qpt, 16).contiguous()
+    for silp in (1, 2, 4):
+        try:
+            out = e.rf_predict_heap_quad(quads, roots, lv, depth, X, silp)
+            ok = torch.allclose(out, ref, atol=1e-4)
+            dt = bench(lambda: e.rf_predict_heap_quad(quads, roots, lv, depth,
+                                                      X, silp))
+            print(f"quad silp={silp}: {S/dt/1e6:7.1f} Ms/s  ok={ok}")
+        except RuntimeError as ex:
+            print(f"quad silp={silp}: ERR {ex}")
+
 
 if __name__ == "__main__":
     main()
