@@ -613,6 +613,63 @@ def test_rf_heap_deterministic_and_silp(ops):
     assert torch.equal(p1, p2) and torch.equal(p1, p4)
 
 
+def _heap_pairs_roots(nodes, depth):
+    """Repack flat heap nodes [T*stride, 2] into the v6 child-pair layout."""
+    stride = (1 << depth) - 1
+    T = nodes.shape[0] // stride
+    nh = nodes.view(T, stride, 2)
+    ppt = (1 << (depth - 1)) - 1
+    idx = torch.arange(ppt, device=nodes.device)
+    pairs = (torch.cat([nh[:, 2 * idx + 1, :], nh[:, 2 * idx + 2, :]], dim=2)
+             .reshape(T * ppt, 4).contiguous())
+    return pairs, nh[:, 0, :].contiguous()
+
+
+def test_rf_heap_pair_matches_v4(ops):
+    """v6 child-pair layout (the routed default) against the flat v4 kernel."""
+    from lambdipy_amd.shims.lpx_sklearn import ForestSoA
+
+    e = ops.ext()
+    for depth in (5, 10):  # odd and even depths
+        f = ForestSoA.random(n_trees=32, depth=depth, n_features=16,
+                             n_classes=6, seed=4, device=DEV)
+        X = _rand((32768, 16))
+        ref = e.rf_predict_heap(f.nodes, f.leaf_values, depth, X, 1, -2)
+        pairs, roots = _heap_pairs_roots(f.nodes, depth)
+        got = e.rf_predict_heap_pair(pairs, roots, f.leaf_values, depth, X, 1)
+        assert torch.allclose(got, ref, atol=1e-5), (got - ref).abs().max()
+
+
+def test_rf_heap_quad_matches_v4(ops):
+    """v7 quad-record layout (A/B only — slower, kept for evidence)."""
+    from lambdipy_amd.shims.lpx_sklearn import ForestSoA
+
+    e = ops.ext()
+    depth = 8
+    f = ForestSoA.random(n_trees=16, depth=depth, n_features=12, n_classes=4,
+                         seed=5, device=DEV)
+    X = _rand((16384, 12))
+    ref = e.rf_predict_heap(f.nodes, f.leaf_values, depth, X, 1, -2)
+    stride = (1 << depth) - 1
+    T = f.nodes.shape[0] // stride
+    nh = f.nodes.view(T, stride, 2)
+    qpt = ((1 << depth) - 1) // 3
+    quads = torch.zeros(T, qpt, 16, dtype=torch.int32, device=DEV)
+    for k in range(depth // 2):
+        base = (4 ** k - 1) // 3
+        o = torch.arange(4 ** k, device=DEV)
+        nn = (1 << (2 * k)) - 1 + o
+        quads[:, base + o, 0:2] = nh[:, 2 * nn + 1, :]
+        quads[:, base + o, 2:4] = nh[:, 2 * nn + 2, :]
+        if 2 * k + 2 < depth:
+            for j in range(4):
+                quads[:, base + o, 4 + 2 * j:6 + 2 * j] = nh[:, 4 * nn + 3 + j, :]
+    quads = quads.reshape(T * qpt, 16).contiguous()
+    roots = nh[:, 0, :].contiguous()
+    got = e.rf_predict_heap_quad(quads, roots, f.leaf_values, depth, X, 1)
+    assert torch.allclose(got, ref, atol=1e-5), (got - ref).abs().max()
+
+
 def test_conv_fp8_implicit_vs_im2col(ops):
     e = ops.ext()
     g = torch.Generator().manual_seed(7)
