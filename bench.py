@@ -33,7 +33,7 @@ if REPO not in sys.path:
 import torch  # noqa: E402
 
 
-def measure_coldstart_and_layer(handler: str, per_gpu_batch: int) -> dict:
+def measure_coldstart_and_layer(per_gpu_batch: int) -> dict:
     """Build (or reuse) the sklearn layer + measure one cold start on 1 GPU."""
     from lambdipy_amd.builder import build_recipe_artifacts
     from lambdipy_amd.packager import assemble_layer, make_zip, zip_report
@@ -75,7 +75,7 @@ def measure_coldstart_and_layer(handler: str, per_gpu_batch: int) -> dict:
 def main() -> int:
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--steps", type=int, default=200)
     p.add_argument("--warmup", type=int, default=5)
     p.add_argument("--batch", type=int, default=1_048_576, help="samples per GPU")
     p.add_argument("--trees", type=int, default=256)
@@ -151,11 +151,20 @@ def main() -> int:
 
     if ctx.is_root:
         extras = {}
+        # self-certification: observed backend, world size and RCCL version
+        # (proves N ranks actually initialized a communicator, not a fallback)
+        comm = {"backend": ctx.backend, "world_size": n}
+        if n > 1 and ctx.backend == "nccl":
+            try:
+                comm["rccl_version"] = list(torch.cuda.nccl.version())
+            except Exception:  # noqa: BLE001 — informational only
+                pass
+        extras["comm"] = comm
         if not args.no_coldstart and n == 1 and "RANK" not in os.environ:
             try:
-                extras = measure_coldstart_and_layer("rf", args.batch)
+                extras.update(measure_coldstart_and_layer(args.batch))
             except Exception as e:  # noqa: BLE001
-                extras = {"coldstart_error": str(e)[-500:]}
+                extras = {"comm": comm, "coldstart_error": str(e)[-500:]}
         line = {
             "metric": "sklearn_rf_fanout_samples_per_s",
             "value": round(samples_per_s, 1),

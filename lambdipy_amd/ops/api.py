@@ -188,9 +188,11 @@ def rf_predict_proba_sum(nodes: torch.Tensor, leaf_values: torch.Tensor,
         # v6 child-pair layout by default (250 vs 242 Ms/s at the bench
         # shape: one random 16-B L2 request per level instead of one 8-B —
         # the kernel is L2 REQUEST-bound, profiles/README). The repack is a
-        # one-time gather cached ON the nodes tensor; callers must not
-        # mutate nodes after the first predict (our broadcast happens
-        # before). LPX_RF_PAIR=0 falls back to the v4 single-node kernel.
+        # one-time gather cached ON the nodes tensor, keyed by torch's
+        # mutation counter (_version): any in-place write (e.g. a new
+        # broadcast of model bytes into the same tensor) invalidates the
+        # cache and the pair layout is rebuilt on the next predict.
+        # LPX_RF_PAIR=0 falls back to the v4 single-node kernel.
         # The two-level LDS variant (top>=0) measured SLOWER (146 Ms/s):
         # top nodes are already L1-hot and the extra LDS halves occupancy.
         use_pair = depth >= 2 and os.environ.get("LPX_RF_PAIR", "1") == "1"
@@ -198,7 +200,7 @@ def rf_predict_proba_sum(nodes: torch.Tensor, leaf_values: torch.Tensor,
         def fn(n, lv, r, x):
             if use_pair:
                 cached = getattr(n, "_lpx_pair_cache", None)
-                if cached is None:
+                if cached is None or cached[0] != n._version:
                     stride = (1 << depth) - 1
                     T = n.shape[0] // stride
                     nh = n.view(T, stride, 2)
@@ -207,9 +209,9 @@ def rf_predict_proba_sum(nodes: torch.Tensor, leaf_values: torch.Tensor,
                     pairs = (torch.cat([nh[:, 2 * idx + 1, :],
                                         nh[:, 2 * idx + 2, :]], dim=2)
                              .reshape(T * ppt, 4).contiguous())
-                    cached = (pairs, nh[:, 0, :].contiguous())
+                    cached = (n._version, pairs, nh[:, 0, :].contiguous())
                     n._lpx_pair_cache = cached
-                pairs, rts = cached
+                _, pairs, rts = cached
                 return e.rf_predict_heap_pair(pairs, rts, lv, depth, x, 1)
             return e.rf_predict_heap(n, lv, depth, x, 1, -2)
     elif packed and chunks is not None:

@@ -105,6 +105,16 @@ def scatter_rows(ctx: FanoutContext, X_root: Optional[torch.Tensor],
     if ctx.world_size == 1:
         assert X_root is not None
         return X_root
+    # The path choice must be DETERMINISTIC and identical on every rank: a
+    # per-call try/except here could diverge across ranks (root fails, peers
+    # wait in scatter) and deadlock the job — so the decision is made from
+    # the backend name alone, and a real scatter failure propagates loudly.
+    if dist.get_backend() not in ("nccl", "gloo"):
+        full = X_root if ctx.is_root else torch.empty(
+            (sum(counts),) + row_shape, dtype=dtype, device=ctx.device)
+        dist.broadcast(full, src=0)
+        off = sum(counts[: ctx.rank])
+        return full[off : off + counts[ctx.rank]].contiguous()
     recv = torch.empty((maxc,) + row_shape, dtype=dtype, device=ctx.device)
     scatter_list = None
     if ctx.is_root:
@@ -115,15 +125,7 @@ def scatter_rows(ctx: FanoutContext, X_root: Optional[torch.Tensor],
             t[:c] = X_root[off : off + c]
             scatter_list.append(t)
             off += c
-    try:
-        dist.scatter(recv, scatter_list, src=0)
-    except Exception:
-        # broadcast + slice fallback
-        full = X_root if ctx.is_root else torch.empty(
-            (sum(counts),) + row_shape, dtype=dtype, device=ctx.device)
-        dist.broadcast(full, src=0)
-        off = sum(counts[: ctx.rank])
-        return full[off : off + counts[ctx.rank]].contiguous()
+    dist.scatter(recv, scatter_list, src=0)
     return recv[: counts[ctx.rank]]
 
 

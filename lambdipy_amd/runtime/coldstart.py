@@ -269,6 +269,28 @@ def cold_start(
         extract_thread = threading.Thread(target=_bg_extract, daemon=True)
         extract_thread.start()
 
+    # Drain every pipe we are not actively reading (rank-0 stderr, all other
+    # ranks' stdout/stderr) on background threads: a handler that emits >64 KB
+    # of warnings/RCCL debug before LPX_RESULT must not deadlock the harness
+    # on a full pipe.
+    import threading
+
+    def _drain(stream, buf):
+        try:
+            for ln in stream:
+                buf.append(ln)
+        except ValueError:  # closed under us at teardown
+            pass
+
+    drain_threads: List[threading.Thread] = []
+    err0_buf: List[str] = []
+    for i, p in enumerate(procs):
+        for stream, buf in (((p.stderr, err0_buf),) if i == 0
+                            else ((p.stdout, []), (p.stderr, []))):
+            th = threading.Thread(target=_drain, args=(stream, buf), daemon=True)
+            th.start()
+            drain_threads.append(th)
+
     report = {}
     total_ms = None
     try:
@@ -293,14 +315,22 @@ def cold_start(
                 report = json.loads(line[len("LPX_RESULT "):])
                 break
         sel.close()
+        # hand rank-0's remaining stdout to a drain thread too
+        th = threading.Thread(target=_drain, args=(procs[0].stdout, lines0),
+                              daemon=True)
+        th.start()
+        drain_threads.append(th)
         if extract_thread is not None:
             extract_thread.join(timeout=timeout_s)
             unzip_ms += (time.perf_counter() - tex0) * 1e3
             if extract_err:
                 raise extract_err[0]
-        rest0, err0 = procs[0].communicate(timeout=timeout_s)
-        out0 = "".join(lines0) + (rest0 or "")
+        procs[0].wait(timeout=timeout_s)
         if not report:
+            for th in drain_threads:
+                th.join(timeout=5)
+            out0 = "".join(lines0)
+            err0 = "".join(err0_buf)
             raise RuntimeError(
                 f"handler produced no LPX_RESULT (rc={procs[0].returncode})\n"
                 f"stdout:\n{out0[-4000:]}\nstderr:\n{err0[-4000:]}"
@@ -308,10 +338,12 @@ def cold_start(
         rcs = [procs[0].returncode]
         for p in procs[1:]:
             try:
-                p.communicate(timeout=60)
+                p.wait(timeout=60)
             except subprocess.TimeoutExpired:
                 p.kill()
             rcs.append(p.returncode)
+        for th in drain_threads:
+            th.join(timeout=5)
     finally:
         for p in procs:
             if p.poll() is None:

@@ -21,7 +21,26 @@ def hip():
     global _hip
     if _hip is None:
         _hip = ctypes.CDLL("libamdhip64.so", mode=ctypes.RTLD_GLOBAL)
+        vp, i32, u32, sz = (ctypes.c_void_p, ctypes.c_int, ctypes.c_uint,
+                            ctypes.c_size_t)
         _hip.hipGetErrorString.restype = ctypes.c_char_p
+        _hip.hipGetErrorString.argtypes = [i32]
+        _hip.hipGetDeviceCount.argtypes = [ctypes.POINTER(i32)]
+        _hip.hipSetDevice.argtypes = [i32]
+        _hip.hipMalloc.argtypes = [ctypes.POINTER(vp), sz]
+        _hip.hipMemcpy.argtypes = [vp, vp, sz, i32]
+        _hip.hipDeviceSynchronize.argtypes = []
+        _hip.hipModuleLoad.argtypes = [ctypes.POINTER(vp), ctypes.c_char_p]
+        _hip.hipModuleGetFunction.argtypes = [ctypes.POINTER(vp), vp,
+                                              ctypes.c_char_p]
+        _hip.hipModuleLaunchKernel.argtypes = [
+            vp, u32, u32, u32, u32, u32, u32, u32, vp,
+            ctypes.POINTER(vp), ctypes.POINTER(vp)]
+        for f in (_hip.hipGetDeviceCount, _hip.hipSetDevice, _hip.hipMalloc,
+                  _hip.hipMemcpy, _hip.hipDeviceSynchronize,
+                  _hip.hipModuleLoad, _hip.hipModuleGetFunction,
+                  _hip.hipModuleLaunchKernel):
+            f.restype = i32
     return _hip
 
 
@@ -227,41 +246,119 @@ def random_heap_forest_np(n_trees: int, depth: int, n_features: int,
 NCCL_FLOAT32 = 7
 NCCL_INT32 = 2
 
+_rccl_typed = False
+
+
+def _rccl_typed_lib():
+    """librccl with explicit argtypes/restypes on every entry point we call.
+
+    ctypes' default int-promotion happens to work on x86-64 for most of
+    these, but one wrong-width argument (e.g. count as c_int instead of
+    c_size_t) would corrupt silently — declare everything."""
+    global _rccl_typed
+    rc = rccl()
+    if not _rccl_typed:
+        vp, i32, sz = ctypes.c_void_p, ctypes.c_int, ctypes.c_size_t
+        rc.ncclGetErrorString.restype = ctypes.c_char_p
+        rc.ncclGetErrorString.argtypes = [i32]
+        rc.ncclGetVersion.restype = i32
+        rc.ncclGetVersion.argtypes = [ctypes.POINTER(i32)]
+        rc.ncclCommInitAll.restype = i32
+        rc.ncclCommInitAll.argtypes = [ctypes.POINTER(vp), i32,
+                                       ctypes.POINTER(i32)]
+        rc.ncclCommDestroy.restype = i32
+        rc.ncclCommDestroy.argtypes = [vp]
+        rc.ncclGroupStart.restype = i32
+        rc.ncclGroupStart.argtypes = []
+        rc.ncclGroupEnd.restype = i32
+        rc.ncclGroupEnd.argtypes = []
+        # (send, recv, count, dtype, [root,] comm, stream)
+        rc.ncclBroadcast.restype = i32
+        rc.ncclBroadcast.argtypes = [vp, vp, sz, i32, i32, vp, vp]
+        rc.ncclAllGather.restype = i32
+        rc.ncclAllGather.argtypes = [vp, vp, sz, i32, vp, vp]
+        _rccl_typed = True
+    return rc
+
+
+def _nck(r: int, what: str) -> None:
+    if r != 0:
+        try:
+            msg = _rccl_typed_lib().ncclGetErrorString(r)
+            msg = msg.decode() if msg else str(r)
+        except Exception:  # noqa: BLE001 — error-path best effort
+            msg = str(r)
+        raise HipError(f"{what} failed: nccl error {r} ({msg})")
+
+
+def nccl_version() -> int:
+    """RCCL version code (e.g. 22807); 0 if the library is unusable."""
+    try:
+        v = ctypes.c_int(0)
+        _nck(_rccl_typed_lib().ncclGetVersion(ctypes.byref(v)), "ncclGetVersion")
+        return v.value
+    except Exception:  # noqa: BLE001 — probe, not a hard dependency
+        return 0
+
 
 class Comms:
     """ncclCommInitAll over g devices in ONE process (xGMI fan-out without a
-    process-per-GPU rendezvous — the cold-start-cheap topology)."""
+    process-per-GPU rendezvous — the cold-start-cheap topology).
 
-    def __init__(self, ndev: int):
+    init_timeout_s: watchdog on comm init — ncclCommInitAll can hang
+    indefinitely when a device is wedged; a serving process must die loudly
+    (the cold-start harness sees EOF and reports) rather than hang the box.
+    """
+
+    def __init__(self, ndev: int, init_timeout_s: float = 120.0):
+        import threading
+
         self.ndev = ndev
         self.comms = (ctypes.c_void_p * ndev)()
         devs = (ctypes.c_int * ndev)(*range(ndev))
-        r = rccl().ncclCommInitAll(self.comms, ndev, devs)
-        if r != 0:
-            raise HipError(f"ncclCommInitAll failed: {r}")
+        rc = _rccl_typed_lib()
+        done = threading.Event()
+
+        def _watchdog():
+            if not done.wait(init_timeout_s):
+                import sys as _sys
+                print(f"lpx_rt.Comms: ncclCommInitAll({ndev}) exceeded "
+                      f"{init_timeout_s}s watchdog — aborting process",
+                      file=_sys.stderr, flush=True)
+                os._exit(124)
+
+        wd = threading.Thread(target=_watchdog, daemon=True)
+        wd.start()
+        try:
+            _nck(rc.ncclCommInitAll(self.comms, ndev, devs),
+                 f"ncclCommInitAll({ndev})")
+        finally:
+            done.set()
 
     def broadcast(self, bufs: Sequence[int], count: int, dtype: int,
                   root: int = 0) -> None:
-        rc = rccl()
-        rc.ncclGroupStart()
+        rc = _rccl_typed_lib()
+        _nck(rc.ncclGroupStart(), "ncclGroupStart")
         for d in range(self.ndev):
             set_device(d)
             # each rank passes device-local pointers; only root's send matters
-            rc.ncclBroadcast(ctypes.c_void_p(bufs[d]), ctypes.c_void_p(bufs[d]),
-                             ctypes.c_size_t(count), dtype, root,
-                             self.comms[d], None)
-        rc.ncclGroupEnd()
+            _nck(rc.ncclBroadcast(bufs[d], bufs[d], count, dtype, root,
+                                  self.comms[d], None),
+                 f"ncclBroadcast(dev {d})")
+        _nck(rc.ncclGroupEnd(), "ncclGroupEnd")
 
     def all_gather(self, send: Sequence[int], recv: Sequence[int],
                    count: int, dtype: int) -> None:
-        rc = rccl()
-        rc.ncclGroupStart()
+        rc = _rccl_typed_lib()
+        _nck(rc.ncclGroupStart(), "ncclGroupStart")
         for d in range(self.ndev):
             set_device(d)
-            rc.ncclAllGather(ctypes.c_void_p(send[d]), ctypes.c_void_p(recv[d]),
-                             ctypes.c_size_t(count), dtype, self.comms[d], None)
-        rc.ncclGroupEnd()
+            _nck(rc.ncclAllGather(send[d], recv[d], count, dtype,
+                                  self.comms[d], None),
+                 f"ncclAllGather(dev {d})")
+        _nck(rc.ncclGroupEnd(), "ncclGroupEnd")
 
     def destroy(self) -> None:
         for d in range(self.ndev):
-            rccl().ncclCommDestroy(self.comms[d])
+            _nck(_rccl_typed_lib().ncclCommDestroy(self.comms[d]),
+                 f"ncclCommDestroy(dev {d})")

@@ -137,7 +137,9 @@ class ForestSoA:
         new_roots = np.empty(self.n_trees, dtype=np.int32)
         pos = 0
         roots = self.roots.cpu().numpy()
-        thr_bits = nodes[:, 1].astype(np.int64).view(np.uint64).astype(np.uint32)             if False else nodes[:, 1].view(np.uint32) if nodes[:, 1].flags['C_CONTIGUOUS']             else np.ascontiguousarray(nodes[:, 1]).view(np.uint32)
+        thr_col = nodes[:, 1]
+        thr_bits = (thr_col.view(np.uint32) if thr_col.flags['C_CONTIGUOUS']
+                    else np.ascontiguousarray(thr_col).view(np.uint32))
         for t, root in enumerate(roots):
             new_roots[t] = pos
             stack = [int(root)]
