@@ -65,6 +65,65 @@ def ensure_standalone_hsaco(allow_build: bool = True) -> str:
     return HSACO_PATH
 
 
+NN_LIB_PATH = os.path.join(OPS_DIR, "liblpx_nn_gfx950.so")
+
+
+def ensure_nn_lib(allow_build: bool = True) -> str:
+    """The torch-free serving runtime library (liblpx_nn_gfx950.so).
+
+    Compiled with hipcc straight from the SAME kernel sources the torch
+    extension uses (ops/csrc/*.hip) plus the C API (nn_capi.cpp) — one
+    kernel code base, two front ends. AOT gfx950, ships in-tree (git-ignored
+    .so travels with snapshots / into recipe wheels). Parallel per-file
+    compiles with mtime-based reuse; objects cached next to the sources in
+    ops/csrc/.nnobj/."""
+    csrc = os.path.join(OPS_DIR, "csrc")
+    sources = sorted(
+        os.path.join(csrc, f)
+        for f in os.listdir(csrc)
+        if (f.endswith(".hip") or f == "nn_capi.cpp")
+        and not f.endswith("_hip.hip") and f != "standalone.hip"
+    )
+    headers = [os.path.join(csrc, h) for h in ("kernels.h", "lpx_common.h")]
+    hdr_mtime = max(os.path.getmtime(h) for h in headers)
+    if os.path.exists(NN_LIB_PATH):
+        lib_m = os.path.getmtime(NN_LIB_PATH)
+        if lib_m >= hdr_mtime and all(
+            lib_m >= os.path.getmtime(s) for s in sources
+        ):
+            return NN_LIB_PATH
+    if not allow_build:
+        raise RuntimeError("liblpx_nn missing and allow_build=False")
+    log.info("building liblpx_nn_gfx950.so (hipcc, %d sources)...", len(sources))
+    objdir = os.path.join(csrc, ".nnobj")
+    os.makedirs(objdir, exist_ok=True)
+    flags = ["--offload-arch=gfx950", "-O3", "-std=c++17", "-fPIC",
+             "-fvisibility=hidden", "-I", csrc]
+    procs = []
+    objs = []
+    for s in sources:
+        o = os.path.join(objdir, os.path.basename(s) + ".o")
+        objs.append(o)
+        if os.path.exists(o) and os.path.getmtime(o) >= max(
+            os.path.getmtime(s), hdr_mtime
+        ):
+            continue
+        procs.append((s, subprocess.Popen(
+            ["hipcc", "-c", s, "-o", o] + flags,
+            stdout=subprocess.PIPE, stderr=subprocess.PIPE)))
+    for s, p in procs:
+        out, err = p.communicate()
+        if p.returncode != 0:
+            raise RuntimeError(
+                f"hipcc failed on {s}:\n{err.decode()[-4000:]}")
+    subprocess.run(
+        ["hipcc", "-shared", "-fPIC", "--offload-arch=gfx950",
+         "-o", NN_LIB_PATH] + objs,
+        check=True,
+    )
+    return NN_LIB_PATH
+
+
 def ensure_ops_extension(allow_build: bool = True) -> str:
     """Return the built extension path, building it in-tree if needed."""
     p = ops_extension_path()
@@ -98,11 +157,13 @@ def emit_ops_wheel(dist_dir: str) -> str:
 
     so = ensure_ops_extension()
     hsaco = ensure_standalone_hsaco()
+    nnlib = ensure_nn_lib()
     files = {
         "lpx_ops/__init__.py": _read(os.path.join(OPS_DIR, "__init__.py")),
         "lpx_ops/api.py": _read(os.path.join(OPS_DIR, "api.py")),
         f"lpx_ops/{os.path.basename(so)}": _read(so),
         f"lpx_ops/{os.path.basename(hsaco)}": _read(hsaco),
+        f"lpx_ops/{os.path.basename(nnlib)}": _read(nnlib),
     }
     return write_wheel(
         dist_dir, "lpx_ops", files, platform_specific=True,

@@ -43,6 +43,19 @@ def _install_handlers(build_dir: str, handlers):
                  os.path.join(build_dir, "lpx_fanout.py"))
     shutil.copy2(os.path.join(pkg, "shims", "lpx_rt.py"),
                  os.path.join(build_dir, "lpx_rt.py"))
+    shutil.copy2(os.path.join(pkg, "shims", "lpx_nn.py"),
+                 os.path.join(build_dir, "lpx_nn.py"))
+    # NN lite handlers dlopen liblpx_nn (AOT gfx950): ship it unless the
+    # layer's lpx_ops wheel already carries it
+    needs_nn = any(h in ("matmul_lite", "fftspmv_lite", "resnet50_lite")
+                   for h in handlers)
+    in_ops = os.path.exists(os.path.join(build_dir, "lpx_ops",
+                                         "liblpx_nn_gfx950.so"))
+    if needs_nn and not in_ops:
+        from .builder.wheel_builder import ensure_nn_lib
+
+        lib = ensure_nn_lib()
+        shutil.copy2(lib, os.path.join(build_dir, os.path.basename(lib)))
 
 
 @click.group()
@@ -74,11 +87,12 @@ def build(requirements, output, zip_path, handlers, no_transitive, as_json):
     )
     if handlers:
         handlers = list(handlers)
-        if "rf" in handlers and "rf_lite" not in handlers:
-            # the torch-free fast path of the same config (~50 KB: lpx_rt +
-            # AOT .hsaco) — always ship it next to the torch rf handler so
-            # `lpx run/bench --handler rf_lite` works on any rf layer
-            handlers.append("rf_lite")
+        # every torch handler gets its torch-free fast-path twin shipped
+        # alongside (lpx_rt/lpx_nn + AOT artifacts, ~2 MB total) so
+        # `lpx run/bench --handler <x>_lite` works on any such layer
+        for h in ("rf", "matmul", "fftspmv", "resnet50"):
+            if h in handlers and h + "_lite" not in handlers:
+                handlers.append(h + "_lite")
         _install_handlers(output, handlers)
         if zip_path:  # re-zip with handlers included
             from .packager import make_zip, zip_report

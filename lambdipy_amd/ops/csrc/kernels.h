@@ -213,4 +213,31 @@ hipError_t gemm_mx_fp4_nt8ph(const unsigned char* A, const unsigned char* sA,
                              float* C, int M, int N, int K,
                              hipStream_t stream);
 
+// ---- torch-free serving utilities (nn_extra.hip; liblpx_nn C API) -------
+// Device randn (xorshift128 + Box-Muller), scaled: out[i] ~ N(0, scale^2).
+hipError_t randn_f32(float* out, long n, unsigned seed, float scale,
+                     hipStream_t stream);
+// Interleaved complex64 with re ~ N(0,1), im = 0.
+hipError_t randn_c64(float* out, long n_complex, unsigned seed,
+                     hipStream_t stream);
+// Abs-max via atomicMax on float bits; caller zeroes out_bits first.
+hipError_t amax_f32(const float* x, unsigned* out_bits, long n,
+                    hipStream_t stream);
+hipError_t checksum_f32(const float* x, double* out, long n,
+                        hipStream_t stream);
+// NHWC global average pool: [N, HW, C] -> [N, C], C % 4 == 0.
+hipError_t gap_nhwc_f32(const float* x, float* out, int N, int HW, int C,
+                        hipStream_t stream);
+hipError_t add_relu_f32(const float* a, const float* b, float* out, long n,
+                        hipStream_t stream);
+hipError_t argmax_rows_f32(const float* x, int* out, long rows, int cols,
+                           hipStream_t stream);
+hipError_t f32_to_bf16(const float* x, uint16_t* out, long n,
+                       hipStream_t stream);
+// out[k] = (cos, sin)(-2 pi k / n) for k < n/2 (fft_c64's twiddle table).
+hipError_t fft_twiddles(float* out, int n, hipStream_t stream);
+// Synthetic fixed-nnz CSR (sorted random columns, N(0,1) data), nnz <= 64.
+hipError_t csrgen_fixed(int* indptr, int* indices, float* data, int rows,
+                        int nnz, int ncols, unsigned seed, hipStream_t stream);
+
 }  // namespace lpx

@@ -28,11 +28,21 @@ log = get_logger("runtime")
 HANDLERS = {
     "requests": "requests_handler",
     "matmul": "matmul_handler",
+    "matmul_lite": "matmul_lite_handler",
     "fftspmv": "fftspmv_handler",
+    "fftspmv_lite": "fftspmv_lite_handler",
     "rf": "rf_handler",
     "rf_lite": "rf_lite_handler",
     "resnet50": "resnet50_handler",
+    "resnet50_lite": "resnet50_lite_handler",
 }
+
+# torch-free handlers (ctypes over AOT gfx950 artifacts): all their layer
+# bytes are extracted eagerly and they skip the ready-marker wait — the bulk
+# unzip overlaps the whole GPU cold start.
+LITE_HANDLERS = {"rf_lite", "matmul_lite", "fftspmv_lite", "resnet50_lite"}
+# single-process multi-GPU (ncclCommInitAll) env knob per lite handler
+LITE_GPUS_ENV = {"rf_lite": "LPX_RF_GPUS", "resnet50_lite": "LPX_RESNET_GPUS"}
 
 
 @dataclass
@@ -181,17 +191,21 @@ def cold_start(
         layer_mb = os.path.getsize(layer) / 2**20
         tmpdir = tempfile.mkdtemp(prefix="lpx-layer-")
         ready_file = os.path.join(tmpdir, ".lpx_ready")
-        # the torch-free rf_lite handler imports ONLY lpx_rt + the AOT
-        # .hsaco: extract those eagerly too and skip the ready-marker wait
-        # entirely (the bulk unzip overlaps the whole GPU cold start).
-        # The torch extension .so in lpx_ops/ (~14 MB) is NOT part of the
-        # rf_lite path — extracting it eagerly would serialize ~0.1 s of
-        # inflate before the spawn, so it goes to the deferred bulk pass.
-        if handler == "rf_lite":
+        # the torch-free lite handlers import ONLY lpx_rt/lpx_nn + their AOT
+        # artifacts (.hsaco / liblpx_nn): extract those eagerly and skip the
+        # ready-marker wait entirely (the bulk unzip overlaps the whole GPU
+        # cold start). The torch extension .so in lpx_ops/ (~14 MB) is NOT
+        # part of any lite path — extracting it eagerly would serialize
+        # ~0.1 s of inflate before the spawn, so it stays in the deferred
+        # bulk pass; liblpx_nn (~2 MB) IS needed before the first launch.
+        if handler in LITE_HANDLERS:
             def early(n):
+                base = os.path.basename(n)
                 return (n.startswith("lpx_handlers/") or n == "lpx_fanout.py"
-                        or n == "lpx_rt.py"
-                        or (n.startswith("lpx_ops/") and not n.endswith(".so")))
+                        or n == "lpx_rt.py" or n == "lpx_nn.py"
+                        or base.startswith("liblpx_nn")
+                        or (n.startswith("lpx_ops/")
+                            and not base.startswith("_lpx_ops")))
         else:
             def early(n):
                 return n.startswith("lpx_handlers/") or n == "lpx_fanout.py"
@@ -214,7 +228,7 @@ def cold_start(
         os.pathsep + base_env["PYTHONPATH"] if base_env.get("PYTHONPATH") else ""
     )
     base_env.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
-    if ready_file and handler != "rf_lite":  # rf_lite's files are all early
+    if ready_file and handler not in LITE_HANDLERS:  # lite files are all early
         base_env["LPX_READY_FILE"] = ready_file
 
     code = f"import lpx_handlers.{mod} as h; h.main()"
@@ -222,9 +236,9 @@ def cold_start(
     if wrap_cmd:  # e.g. rocprofv3 --kernel-trace --stats -d DIR --
         argv = list(wrap_cmd) + argv
     procs = []
-    if handler == "rf_lite" and gpus > 1:
+    if handler in LITE_GPUS_ENV and gpus > 1:
         # single-process multi-GPU (ncclCommInitAll) — no per-rank spawn
-        e = dict(base_env, LPX_RF_GPUS=str(gpus))
+        e = dict(base_env, **{LITE_GPUS_ENV[handler]: str(gpus)})
         procs.append(
             subprocess.Popen(argv, env=e, stdout=subprocess.PIPE,
                              stderr=subprocess.PIPE, text=True)

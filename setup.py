@@ -23,6 +23,7 @@ sources = [
     for f in sorted(os.listdir(CSRC))
     if f.endswith((".cpp", ".hip")) and not f.endswith(("_hip.cpp", "_hip.hip"))
     and f != "standalone.hip"  # --genco code object, built by the builder
+    and f != "nn_capi.cpp"  # liblpx_nn C API, built by builder.ensure_nn_lib
 ]
 
 setup(
