@@ -64,8 +64,9 @@ def main():
 
 
 @main.command()
-@click.option("-r", "--requirements", default="requirements.txt",
-              type=click.Path(exists=True), help="requirements file to build")
+@click.option("-r", "--requirements", default=None,
+              help="project file (requirements.txt, Pipfile or Pipfile.lock; "
+                   "default: auto-detect in cwd)")
 @click.option("-o", "--output", default="build", help="layer output directory")
 @click.option("-z", "--zip", "zip_path", default=None,
               help="also produce a zip artifact at this path")
@@ -77,9 +78,19 @@ def build(requirements, output, zip_path, handlers, no_transitive, as_json):
     """Resolve requirements, build recipe wheels, assemble the layer."""
     from .builder import build_recipe_artifacts
     from .packager import assemble_layer
-    from .resolver import parse_requirements_file, resolve
+    from .resolver import parse_project_file, resolve
 
-    reqs = parse_requirements_file(requirements)
+    if requirements is None:
+        for cand in ("requirements.txt", "Pipfile.lock", "Pipfile"):
+            if os.path.exists(cand):
+                requirements = cand
+                break
+        else:
+            raise click.ClickException(
+                "no requirements.txt / Pipfile(.lock) found (use -r)")
+    elif not os.path.exists(requirements):
+        raise click.ClickException(f"{requirements!r} does not exist")
+    reqs = parse_project_file(requirements)
     res = resolve(reqs, follow_transitive=not no_transitive)
     log.info("resolved %d deps (%d recipe hits)", len(res.deps), len(res.recipe_hits))
     report = assemble_layer(
@@ -109,29 +120,43 @@ def build(requirements, output, zip_path, handlers, no_transitive, as_json):
 
 
 @main.command()
-@click.argument("zip_path", type=click.Path(exists=True))
-@click.option("--name", default=None, help="release name (default: zip filename)")
+@click.argument("zip_path", required=False, type=click.Path(exists=True))
+@click.option("--recipe", "recipe_name", default=None,
+              help="publish a recipe's per-version ROCm wheels instead of a layer")
+@click.option("--name", default=None, help="release name (default: filename)")
 @click.option("--release-dir", default=DEFAULT_RELEASE_DIR, show_default=True)
-def publish(zip_path, name, release_dir):
-    """Store a built layer in the local artifact store.
+def publish(zip_path, recipe_name, name, release_dir):
+    """Store artifacts in the local keyed store.
 
-    Offline analog of the reference's GitHub-Releases upload:
-    artifacts land in a content-addressed local store that later builds reuse."""
-    import hashlib
+    Offline analog of the reference's GitHub-Releases upload (SURVEY.md
+    §3.2): assets are keyed by (name, version, python tag, gfx arch) with a
+    sha256 sidecar that fetch verifies. Two asset kinds, mirroring the
+    reference's publish path: finished layer zips (positional arg) and
+    per-recipe build artifacts (--recipe emits + publishes the recipe's
+    ROCm wheels, the analog of recipe x version release assets)."""
+    from .store import publish_file
 
-    os.makedirs(release_dir, exist_ok=True)
-    h = hashlib.sha256()
-    with open(zip_path, "rb") as f:
-        for chunk in iter(lambda: f.read(1 << 20), b""):
-            h.update(chunk)
-    digest = h.hexdigest()[:16]
-    base = name or os.path.basename(zip_path)
-    dest = os.path.join(release_dir, f"{base}.{digest}.zip")
-    shutil.copy2(zip_path, dest)
-    meta = {"name": base, "sha256_16": digest, "bytes": os.path.getsize(zip_path)}
-    with open(dest + ".json", "w") as f:
-        json.dump(meta, f)
-    click.echo(f"published {dest}")
+    if recipe_name:
+        from .builder.wheel_builder import emit_ops_wheel, emit_shim_wheel
+        from .builder.wheelgen import WHEEL_CACHE
+        from .recipes import get_registry
+
+        recipe = get_registry().get(recipe_name)
+        if recipe is None:
+            raise click.ClickException(f"unknown recipe {recipe_name!r}")
+        wheels = [emit_ops_wheel(WHEEL_CACHE)]
+        if recipe.shim:
+            wheels.append(emit_shim_wheel(recipe, WHEEL_CACHE))
+        for w in wheels:
+            a = publish_file(w, release_dir)
+            click.echo(f"published {a.name} {a.version} "
+                       f"(py={a.python or '-'}, arch={a.arch or '-'}) "
+                       f"-> {a.path}")
+        return
+    if not zip_path:
+        raise click.ClickException("pass a layer zip or --recipe NAME")
+    a = publish_file(zip_path, release_dir, name=name)
+    click.echo(f"published {a.path}")
 
 
 @main.command()
@@ -193,36 +218,24 @@ def bench(layer, handler, gpus, repeats, out, rocprof_dir):
 
 @main.command()
 @click.argument("name")
-@click.option("-o", "--output", default=None, help="destination zip path")
+@click.option("-o", "--output", default=None, help="destination path")
+@click.option("--version", default=None, help="exact version to match")
+@click.option("--python", "py_tag", default=None, help="python tag, e.g. cp310")
+@click.option("--arch", default=None, help="gfx arch, e.g. gfx950")
 @click.option("--release-dir", default=DEFAULT_RELEASE_DIR, show_default=True)
-def fetch(name, output, release_dir):
-    """Fetch a published layer from the local artifact store.
+def fetch(name, output, version, py_tag, arch, release_dir):
+    """Fetch a published artifact from the local keyed store.
 
     Offline analog of the reference's prebuilt GitHub-Releases download
-    (SURVEY.md §2.1 R4): picks the newest artifact whose name matches,
-    verifies its content hash against the stored metadata, and copies it."""
-    import hashlib
+    (SURVEY.md §2.1 R4): matches on (name, version, python tag, gfx arch),
+    verifies the stored sha256 against the file, and copies it out."""
+    from .store import StoreError, fetch_asset
 
-    candidates = sorted(
-        (f for f in os.listdir(release_dir) if f.endswith(".zip")
-         and f.startswith(name)),
-        key=lambda f: os.path.getmtime(os.path.join(release_dir, f)),
-    ) if os.path.isdir(release_dir) else []
-    if not candidates:
-        raise click.ClickException(f"no published artifact matches {name!r}")
-    src = os.path.join(release_dir, candidates[-1])
-    meta_path = src + ".json"
-    if os.path.exists(meta_path):
-        with open(meta_path) as f:
-            meta = json.load(f)
-        h = hashlib.sha256()
-        with open(src, "rb") as f:
-            for chunk in iter(lambda: f.read(1 << 20), b""):
-                h.update(chunk)
-        if h.hexdigest()[:16] != meta.get("sha256_16"):
-            raise click.ClickException(f"hash mismatch for {src} (corrupt store)")
-    dest = output or os.path.basename(src)
-    shutil.copy2(src, dest)
+    try:
+        dest = fetch_asset(name, output, release_dir, version=version,
+                           python=py_tag, arch=arch)
+    except StoreError as e:
+        raise click.ClickException(str(e))
     click.echo(f"fetched {dest}")
 
 

@@ -97,8 +97,12 @@ class BuildRecipe:
             description=d.get("description", ""),
             path=path,
         )
-        if r.kind == "native" and not r.kernels:
-            raise RecipeError(f"recipe {r.name}: native recipes must list kernels")
+        # native + shim => kernels must be declared (the shim routes hot ops
+        # to them); native WITHOUT a shim is legal — a strip/prune-only
+        # recipe (e.g. pandas, whose numerics ride the numpy shim).
+        if r.kind == "native" and r.shim and not r.kernels:
+            raise RecipeError(
+                f"recipe {r.name}: native recipes with a shim must list kernels")
         return r
 
     def supports(self, version: str) -> bool:

@@ -104,6 +104,76 @@ def parse_requirements(text: str, *, origin: str = "<string>") -> List[Requireme
     return reqs
 
 
+def parse_pipfile_lock(path: str, *, dev: bool = False) -> List[Requirement]:
+    """Parse a Pipfile.lock (the reference reads
+    Pipfile(.lock) as an alternative front end). The lock file is already
+    fully pinned, so entries map 1:1 onto Requirement objects; the
+    ``develop`` section is excluded unless dev=True (build layers ship
+    runtime deps only, like the reference)."""
+    import json
+
+    with open(path) as f:
+        data = json.load(f)
+    reqs: List[Requirement] = []
+    sections = ["default"] + (["develop"] if dev else [])
+    for section in sections:
+        for name, spec in (data.get(section) or {}).items():
+            if not isinstance(spec, dict):
+                spec = {"version": spec}
+            ver = spec.get("version") or ""
+            if ver == "*":
+                ver = ""
+            reqs.append(Requirement(
+                name=canonicalize_name(name),
+                raw=f"{name}{ver}",
+                specifier=ver,
+                extras=tuple(sorted(spec.get("extras") or ())),
+                marker=spec.get("markers"),
+            ))
+    return reqs
+
+
+def parse_pipfile(path: str, *, dev: bool = False) -> List[Requirement]:
+    """Parse a Pipfile (TOML). Entries are either version strings
+    (``numpy = "*"``) or tables (``{version = ">=1.0", markers = "..."}``)."""
+    import tomli
+
+    with open(path, "rb") as f:
+        data = tomli.load(f)
+    reqs: List[Requirement] = []
+    sections = ["packages"] + (["dev-packages"] if dev else [])
+    for section in sections:
+        for name, spec in (data.get(section) or {}).items():
+            if isinstance(spec, str):
+                spec = {"version": spec}
+            ver = spec.get("version") or ""
+            if ver == "*":
+                ver = ""
+            reqs.append(Requirement(
+                name=canonicalize_name(name),
+                raw=f"{name}{ver}",
+                specifier=ver,
+                extras=tuple(sorted(spec.get("extras") or ())),
+                marker=spec.get("markers"),
+            ))
+    return reqs
+
+
+def parse_project_file(path: str) -> List[Requirement]:
+    """Front-door dispatch on the project file type: requirements.txt,
+    Pipfile.lock (preferred over Pipfile when both exist, like the
+    reference's pipenv flow) or Pipfile."""
+    base = os.path.basename(path)
+    if base == "Pipfile.lock":
+        return parse_pipfile_lock(path)
+    if base == "Pipfile":
+        lock = os.path.join(os.path.dirname(path) or ".", "Pipfile.lock")
+        if os.path.exists(lock):
+            return parse_pipfile_lock(lock)
+        return parse_pipfile(path)
+    return parse_requirements_file(path)
+
+
 def parse_requirements_file(path: str, _seen: Optional[set] = None) -> List[Requirement]:
     """Parse a requirements file, following ``-r`` includes relative to it."""
     _seen = _seen if _seen is not None else set()

@@ -52,7 +52,9 @@ def test_publish(runner, tmp_path):
     rel = tmp_path / "releases"
     r = runner.invoke(cli_main, ["publish", str(zp), "--release-dir", str(rel)])
     assert r.exit_code == 0, r.output
-    assert any(f.endswith(".zip") for f in os.listdir(rel))
+    # store keys assets as <name>.zip.<sha16> + sidecar json
+    assert any(".zip." in f and not f.endswith(".json")
+               for f in os.listdir(rel))
 
 
 def test_publish_fetch_roundtrip(runner, tmp_path):

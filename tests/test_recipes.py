@@ -35,11 +35,16 @@ def test_recipe_schema_validation():
         BuildRecipe.from_dict(
             {"recipe_format": 1, "name": "x", "versions": ">=1", "kind": "weird"}
         )
-    # native without kernels
+    # native WITH a shim but without kernels (shim needs ops to route to);
+    # shimless native recipes (strip/prune-only, e.g. pandas) are legal
     with pytest.raises(RecipeError):
         BuildRecipe.from_dict(
-            {"recipe_format": 1, "name": "x", "versions": ">=1", "kind": "native"}
+            {"recipe_format": 1, "name": "x", "versions": ">=1",
+             "kind": "native", "shim": "lpx_x"}
         )
+    BuildRecipe.from_dict(
+        {"recipe_format": 1, "name": "x", "versions": ">=1", "kind": "native"}
+    )
 
 
 def test_provides_defaults_to_name():
