@@ -299,7 +299,33 @@ class ForestSoA:
     # default; set use_lds_chunks=True to A/B the LDS variant.
     use_lds_chunks = False
 
+    def _predict_proba_t_cpu(self, X: torch.Tensor) -> torch.Tensor:
+        """Plain-torch heap-forest reference (CPU tensors only): the same
+        f32 traversal semantics as the gfx950 kernel (right iff x > thr).
+        Lets the multi-process bench/fanout path run under gloo on CPU —
+        it is NOT a GPU fallback (ops.api stays fail-loud on GPU boxes)."""
+        assert self.layout == "heap", "CPU reference supports heap forests"
+        d = self.depth
+        stride = (1 << d) - 1
+        T = self.nodes.shape[0] // stride
+        nh = self.nodes.view(T, stride, 2)
+        feat = nh[..., 0].long()
+        thr = nh[..., 1].contiguous().view(torch.float32)
+        idx = torch.zeros(T, X.shape[0], dtype=torch.long)
+        for _ in range(d):
+            f = feat.gather(1, idx)                      # [T,S]
+            t = thr.gather(1, idx)
+            xv = X.gather(1, f.t()).t()                  # X[s, f[tr,s]]
+            idx = 2 * idx + 1 + (xv > t).long()
+        leaf = idx - stride
+        C = self.leaf_values.shape[1]
+        lv = self.leaf_values.view(T, 1 << d, C)
+        gathered = lv.gather(1, leaf.unsqueeze(-1).expand(T, -1, C))
+        return gathered.sum(dim=0) / float(self.n_trees)
+
     def predict_proba_t(self, X: torch.Tensor) -> torch.Tensor:
+        if self.nodes.device.type == "cpu" and self.layout == "heap":
+            return self._predict_proba_t_cpu(X)
         if self.layout == "heap":
             sums = _ops.rf_predict_proba_sum(self.nodes, self.leaf_values,
                                              self.roots, X, layout="heap",

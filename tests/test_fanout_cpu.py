@@ -128,3 +128,42 @@ def test_scatter_rows_gloo():
     for p, (o, e) in zip(procs, outs):
         assert p.returncode == 0, f"{o}\n{e}"
     assert any("SCATTER_OK" in o for o, _ in outs)
+
+
+def test_bench_script_world8_gloo(tmp_path):
+    """Run bench.py ITSELF at world_size=8 over gloo on CPU — the exact
+    step function (scatter -> predict -> all_gather -> MAX-reduce of
+    elapsed) the driver's 8-GPU SCALE run executes over RCCL
+    (VERDICT round-1 next-step 2: every collective call site covered by a
+    test that runs this round)."""
+    world = 8
+    port = 29820
+    procs = []
+    for r in range(world):
+        env = dict(
+            os.environ,
+            RANK=str(r),
+            LOCAL_RANK=str(r),
+            WORLD_SIZE=str(world),
+            MASTER_ADDR="127.0.0.1",
+            MASTER_PORT=str(port),
+        )
+        procs.append(
+            subprocess.Popen(
+                [sys.executable, os.path.join(REPO, "bench.py"),
+                 "--gpus", str(world), "--steps", "3", "--warmup", "1",
+                 "--batch", "512", "--trees", "8", "--depth", "4",
+                 "--features", "8", "--no-coldstart"],
+                env=env, stdout=subprocess.PIPE, stderr=subprocess.PIPE,
+                text=True, cwd=REPO)
+        )
+    outs = [p.communicate(timeout=300) for p in procs]
+    for p, (o, e) in zip(procs, outs):
+        assert p.returncode == 0, f"bench rank failed:\n{o[-2000:]}\n{e[-2000:]}"
+    lines = [l for l in outs[0][0].splitlines() if l.startswith("{")]
+    assert lines, outs[0][0]
+    d = json.loads(lines[-1])
+    assert d["n_gpus"] == 8
+    assert d["config"]["global_batch"] == 512 * 8
+    assert d["comm"]["world_size"] == 8 and d["comm"]["backend"] == "gloo"
+    assert d["value"] > 0 and d["scaling"] == "weak"

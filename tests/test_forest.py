@@ -131,3 +131,48 @@ def test_pack_roundtrip_equivalence():
     got = soa_predict_proba_cpu(f, X[:80])
     want = clf.predict_proba(X[:80]).astype(np.float32)
     np.testing.assert_allclose(got, want, rtol=1e-5, atol=1e-6)
+
+
+def test_cpu_heap_reference_matches_naive():
+    """The torch CPU heap reference (used by the gloo bench path) vs a
+    per-sample python traversal."""
+    import torch
+
+    from lambdipy_amd.shims.lpx_sklearn import ForestSoA
+
+    f = ForestSoA.random(n_trees=5, depth=4, n_features=6, n_classes=3,
+                         seed=11, device="cpu")
+    g = torch.Generator().manual_seed(12)
+    X = torch.randn(17, 6, generator=g)
+    proba = f.predict_proba_t(X)
+    stride = (1 << 4) - 1
+    nh = f.nodes.view(5, stride, 2)
+    lv = f.leaf_values.view(5, 1 << 4, 3)
+    for s in range(17):
+        acc = torch.zeros(3)
+        for t in range(5):
+            nd = 0
+            for _ in range(4):
+                feat = int(nh[t, nd, 0])
+                thr = nh[t, nd, 1].view(torch.float32).item()
+                nd = 2 * nd + 1 + (0 if X[s, feat].item() <= thr else 1)
+            acc += lv[t, nd - stride]
+        assert torch.allclose(proba[s], acc / 5, atol=1e-6), f"sample {s}"
+
+
+@pytest.mark.gpu
+def test_gpu_heap_kernel_matches_cpu_reference():
+    """HIP heap kernel vs the plain-torch CPU reference on the same forest."""
+    import torch
+
+    from lambdipy_amd.shims.lpx_sklearn import ForestSoA
+
+    f = ForestSoA.random(n_trees=32, depth=8, n_features=16, n_classes=8,
+                         seed=3, device="cuda")
+    g = torch.Generator().manual_seed(4)
+    X = torch.randn(4096, 16, generator=g)
+    proba_gpu = f.predict_proba_t(X.cuda()).cpu()
+    fc = ForestSoA.random(n_trees=32, depth=8, n_features=16, n_classes=8,
+                          seed=3, device="cpu")
+    proba_cpu = fc.predict_proba_t(X)
+    assert torch.allclose(proba_gpu, proba_cpu, atol=1e-5)
