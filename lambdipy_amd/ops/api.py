@@ -196,8 +196,34 @@ def rf_predict_proba_sum(nodes: torch.Tensor, leaf_values: torch.Tensor,
         # The two-level LDS variant (top>=0) measured SLOWER (146 Ms/s):
         # top nodes are already L1-hot and the extra LDS halves occupancy.
         use_pair = depth >= 2 and os.environ.get("LPX_RF_PAIR", "1") == "1"
+        # v8 compact nodes (u16 feat + fp16 threshold, 8-B pair records):
+        # halves the bytes of every random request. OPT-IN (LPX_RF_COMPACT=1)
+        # because fp16-rounded thresholds change decisions for x strictly
+        # between the f32 and fp16 threshold — EXACT only when the forest's
+        # thresholds are fp16-representable; sklearn-fitted f32 forests keep
+        # the exact pair kernel by default.
+        use_compact = (use_pair
+                       and os.environ.get("LPX_RF_COMPACT", "0") == "1")
 
         def fn(n, lv, r, x):
+            if use_compact and x.shape[1] <= 0xFFFF:
+                cached = getattr(n, "_lpx_cpair_cache", None)
+                if cached is None or cached[0] != n._version:
+                    stride = (1 << depth) - 1
+                    T = n.shape[0] // stride
+                    nh = n.view(T, stride, 2)
+                    thr16 = (nh[..., 1].contiguous().view(torch.float32)
+                             .to(torch.float16).view(torch.int16)
+                             .to(torch.int32) & 0xFFFF)
+                    cn = (nh[..., 0] << 16) | thr16  # [T, stride] i32
+                    ppt = (1 << (depth - 1)) - 1
+                    idx = torch.arange(ppt, device=n.device)
+                    cp = (torch.stack([cn[:, 2 * idx + 1], cn[:, 2 * idx + 2]],
+                                      dim=2).reshape(T * ppt, 2).contiguous())
+                    cached = (n._version, cp, cn[:, 0].contiguous())
+                    n._lpx_cpair_cache = cached
+                _, cp, rts = cached
+                return e.rf_predict_heap_cpair(cp, rts, lv, depth, x, 1)
             if use_pair:
                 cached = getattr(n, "_lpx_pair_cache", None)
                 if cached is None or cached[0] != n._version:

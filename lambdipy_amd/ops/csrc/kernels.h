@@ -95,6 +95,15 @@ hipError_t rf_predict_heap_pair_f32(const int* pairs, const int* roots,
                                     int depth, const float* X, float* out,
                                     int S, int F, int C, int silp,
                                     hipStream_t stream);
+// v8: compact child-pair heap — 4-B nodes (u16 feat << 16 | fp16-thr bits),
+// cpairs int32 [n_trees*(2^(depth-1)-1), 2], roots int32 [n_trees]; halves
+// the bytes of every random L2 request vs v6. fp16 thresholds: exact for
+// fp16-representable forests, else RNE-rounded decisions (see rf.hip).
+hipError_t rf_predict_heap_cpair_f32(const int* cpairs, const int* roots,
+                                     const float* leaf_values, int n_trees,
+                                     int depth, const float* X, float* out,
+                                     int S, int F, int C, int silp,
+                                     hipStream_t stream);
 // v5: two-level heap — top `top` levels served from LDS (top<0: auto-pick
 // largest fitting 2-blocks/CU; falls back to v4 when nothing fits).
 hipError_t rf_predict_heap2_f32(const int* nodes_heap, const float* leaf_values,

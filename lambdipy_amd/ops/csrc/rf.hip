@@ -449,6 +449,110 @@ __global__ __launch_bounds__(THREADS) void rf_heap_pair_kernel(
   }
 }
 
+// v8: COMPACT child-pair traversal — 4-B nodes (u16 feature id, fp16
+// threshold: node = feat<<16 | f16bits) so a child-pair record is 8 B and
+// the roots array 4 B/tree. The pair kernel is L2 REQUEST-bound but the
+// quad A/B showed bytes/latency also matter (NOTES item 4): halving the
+// bytes of every random request keeps the request count of v6 while
+// halving its line footprint — two pair records now share a 16-B sector.
+// Numerics: thresholds are rounded f32->fp16 (RNE) at pack time; a
+// decision flips only for x strictly between the f32 and fp16 threshold
+// (|dt| <= 2^-11 relative), quantified in tests. For forests whose
+// thresholds are already fp16-representable the traversal is EXACT.
+// Feature ids must fit u16 (F <= 65535 — always true here).
+__device__ __forceinline__ float lpx_h2f(unsigned short u) {
+  __half_raw r;
+  r.x = u;
+  return __half2float(__half(r));
+}
+
+template <bool STAGE_X, int SILP>
+__global__ __launch_bounds__(THREADS) void rf_heap_cpair_kernel(
+    const i32x2* __restrict__ pairs, const int* __restrict__ roots,
+    const float* __restrict__ leaf_values, int n_trees, int depth,
+    const float* __restrict__ X, float* __restrict__ out, int S, int F,
+    int C) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* xl = reinterpret_cast<float*>(smem);
+  const int t = threadIdx.x;
+  const int rowf = F + 1;
+  const int tree_stride = (1 << depth) - 1;
+  const int ppt = (1 << (depth - 1)) - 1;  // pairs per tree
+  const int n_leaves = 1 << depth;
+
+  for (int tile = blockIdx.x * (THREADS * SILP); tile < S;
+       tile += gridDim.x * (THREADS * SILP)) {
+    const int tile_n = min(THREADS * SILP, S - tile);
+    if (STAGE_X) {
+      const long total = (long)tile_n * F;
+      for (long i = t; i < total; i += THREADS) {
+        const int r = (int)(i / F), c = (int)(i - (long)r * F);
+        xl[r * rowf + c] = X[((long)tile + r) * F + c];
+      }
+      __syncthreads();
+    }
+
+    int s[SILP];
+    const float* xrow[SILP];
+#pragma unroll
+    for (int j = 0; j < SILP; ++j) {
+      s[j] = tile + t + j * THREADS;
+      xrow[j] = STAGE_X ? (xl + (t + j * THREADS) * rowf)
+                        : (X + (long)s[j] * F);
+    }
+    float acc[SILP][CMAX];
+#pragma unroll
+    for (int j = 0; j < SILP; ++j)
+#pragma unroll
+      for (int c = 0; c < CMAX; ++c) acc[j][c] = 0.f;
+
+    for (int tr = 0; tr < n_trees; ++tr) {
+      const i32x2* tp = pairs + (long)tr * ppt;
+      const unsigned root = (unsigned)roots[tr];
+      int nd[SILP];        // current heap index
+      unsigned cf[SILP];   // current node feature
+      float ct[SILP];      // current node threshold (f32 of the f16 bits)
+#pragma unroll
+      for (int j = 0; j < SILP; ++j) {
+        nd[j] = 0;
+        cf[j] = root >> 16;
+        ct[j] = lpx_h2f((unsigned short)(root & 0xffffu));
+      }
+      for (int d = 0; d < depth - 1; ++d) {
+#pragma unroll
+        for (int j = 0; j < SILP; ++j) {
+          if (s[j] < S) {
+            const i32x2 pr = tp[nd[j]];  // compact children of current node
+            const int go = (xrow[j][cf[j]] <= ct[j]) ? 0 : 1;
+            nd[j] = 2 * nd[j] + 1 + go;
+            const unsigned c = (unsigned)(go ? pr.y : pr.x);
+            cf[j] = c >> 16;
+            ct[j] = lpx_h2f((unsigned short)(c & 0xffffu));
+          }
+        }
+      }
+#pragma unroll
+      for (int j = 0; j < SILP; ++j) {
+        if (s[j] < S) {
+          const int go = (xrow[j][cf[j]] <= ct[j]) ? 0 : 1;
+          const int leaf = 2 * nd[j] + 1 + go - tree_stride;
+          const float* lv = leaf_values + ((long)tr * n_leaves + leaf) * C;
+#pragma unroll
+          for (int c = 0; c < CMAX; ++c)
+            if (c < C) acc[j][c] += lv[c];
+        }
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < SILP; ++j)
+      if (s[j] < S)
+#pragma unroll
+        for (int c = 0; c < CMAX; ++c)
+          if (c < C) out[(long)s[j] * C + c] = acc[j][c];
+    if (STAGE_X) __syncthreads();
+  }
+}
+
 // v7: QUAD records — two tree levels per random fetch. Record r for an
 // even-level node n (level d=2k, within-level offset o, record slot
 // (4^k-1)/3 + o) is 64 B: {children(n) [2 nodes], grandchildren(n)
@@ -728,6 +832,40 @@ hipError_t rf_predict_heap_pair_f32(const int* pairs, const int* roots,
   LPX_RFHP_CASE(2)
   LPX_RFHP_CASE(4)
 #undef LPX_RFHP_CASE
+  return hipErrorInvalidValue;
+}
+
+hipError_t rf_predict_heap_cpair_f32(const int* cpairs, const int* roots,
+                                     const float* leaf_values, int n_trees,
+                                     int depth, const float* X, float* out,
+                                     int S, int F, int C, int silp,
+                                     hipStream_t stream) {
+  if (C > CMAX || depth < 2 || depth > 26 || F > 0xffff)
+    return hipErrorInvalidValue;
+  const int SILP = (silp == 2) ? 2 : (silp == 4 ? 4 : 1);
+  const bool stage = F <= F_LDS_MAX;
+  const size_t lds = stage ? (size_t)THREADS * SILP * (F + 1) * 4 : 0;
+  if (lds > 160 * 1024) return hipErrorInvalidValue;
+  unsigned blocks = lpx_ceil_div(S, THREADS * SILP);
+  if (blocks > 4096u) blocks = 4096u;
+  if (blocks == 0) blocks = 1;
+  const auto* pp = reinterpret_cast<const i32x2*>(cpairs);
+#define LPX_RFHC_CASE(I)                                                      \
+  if (SILP == I) {                                                            \
+    if (stage)                                                                \
+      rf_heap_cpair_kernel<true, I>                                           \
+          <<<dim3(blocks), dim3(THREADS), lds, stream>>>(                     \
+              pp, roots, leaf_values, n_trees, depth, X, out, S, F, C);       \
+    else                                                                      \
+      rf_heap_cpair_kernel<false, I>                                          \
+          <<<dim3(blocks), dim3(THREADS), 0, stream>>>(                       \
+              pp, roots, leaf_values, n_trees, depth, X, out, S, F, C);       \
+    return hipGetLastError();                                                 \
+  }
+  LPX_RFHC_CASE(1)
+  LPX_RFHC_CASE(2)
+  LPX_RFHC_CASE(4)
+#undef LPX_RFHC_CASE
   return hipErrorInvalidValue;
 }
 

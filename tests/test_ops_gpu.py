@@ -836,3 +836,82 @@ def test_lpx_sklearn_shim_pairwise(ops):
     got = lpx_sklearn.pairwise_sqeuclidean(X, Y)
     ref = euclidean_distances(X, Y, squared=True)
     assert np.abs(got - ref).max() < 1e-2
+
+
+# ---- RF v8 compact pair (u16 feat + fp16 threshold, 8-B records) ----------
+
+def _pack_pairs(n, depth):
+    stride = (1 << depth) - 1
+    T = n.shape[0] // stride
+    nh = n.view(T, stride, 2)
+    ppt = (1 << (depth - 1)) - 1
+    idx = torch.arange(ppt, device=n.device)
+    pairs = torch.cat([nh[:, 2 * idx + 1, :], nh[:, 2 * idx + 2, :]],
+                      dim=2).reshape(T * ppt, 4).contiguous()
+    roots = nh[:, 0, :].contiguous()
+    thr16 = (nh[..., 1].contiguous().view(torch.float32).to(torch.float16)
+             .view(torch.int16).to(torch.int32) & 0xFFFF)
+    cn = (nh[..., 0] << 16) | thr16
+    cpairs = (torch.stack([cn[:, 2 * idx + 1], cn[:, 2 * idx + 2]], dim=2)
+              .reshape(T * ppt, 2).contiguous())
+    return pairs, roots, cpairs, cn[:, 0].contiguous()
+
+
+def test_rf_cpair_exact_on_fp16_forest(ops):
+    """On a forest whose thresholds are fp16-representable, the compact
+    kernel must be bitwise-identical to the f32 pair kernel."""
+    from lambdipy_amd.shims.lpx_sklearn import ForestSoA
+
+    e = ops.ext()
+    depth = 8
+    f = ForestSoA.random(64, depth, 16, 8, seed=5, device=DEV)
+    n = f.nodes.view(-1, 2).clone()
+    n[:, 1] = (n[:, 1].view(torch.float32).to(torch.float16)
+               .to(torch.float32).view(torch.int32))
+    n = n.view_as(f.nodes).contiguous()
+    X = _rand((8192, 16), seed=6)
+    pairs, roots, cpairs, croots = _pack_pairs(n, depth)
+    a = e.rf_predict_heap_pair(pairs, roots, f.leaf_values, depth, X, 1)
+    b = e.rf_predict_heap_cpair(cpairs, croots, f.leaf_values, depth, X, 1)
+    assert torch.equal(a, b), "compact kernel diverges on fp16 thresholds"
+    b2 = e.rf_predict_heap_cpair(cpairs, croots, f.leaf_values, depth, X, 1)
+    assert torch.equal(b, b2), "nondeterministic cpair kernel"
+
+
+def test_rf_cpair_flip_rate_on_f32_forest(ops):
+    """On an f32-threshold forest the fp16 rounding flips only decisions for
+    x strictly between the two thresholds — the label-flip rate must be
+    small (quantified error class)."""
+    from lambdipy_amd.shims.lpx_sklearn import ForestSoA
+
+    e = ops.ext()
+    depth = 10
+    f = ForestSoA.random(128, depth, 32, 8, seed=7, device=DEV)
+    X = _rand((65536, 32), seed=8)
+    pairs, roots, cpairs, croots = _pack_pairs(f.nodes, depth)
+    a = e.rf_predict_heap_pair(pairs, roots, f.leaf_values, depth, X, 1)
+    b = e.rf_predict_heap_cpair(cpairs, croots, f.leaf_values, depth, X, 1)
+    flips = (a.argmax(1) != b.argmax(1)).float().mean().item()
+    # ~1e-3..1e-2 expected for N(0,1) thresholds at depth 10 x 128 trees
+    assert flips < 0.05, f"label flip rate {flips} too high"
+    # per-class sums stay close (each flipped path changes one leaf row)
+    rel = (a - b).abs().max().item() / (a.abs().max().item() + 1e-6)
+    assert rel < 0.2, f"proba divergence {rel}"
+
+
+def test_rf_compact_api_routing(ops, monkeypatch):
+    """LPX_RF_COMPACT=1 routes predict through the compact kernel and the
+    cache invalidates on mutation (same contract as the pair cache)."""
+    from lambdipy_amd.shims.lpx_sklearn import ForestSoA
+
+    monkeypatch.setenv("LPX_RF_COMPACT", "1")
+    f = ForestSoA.random(32, 6, 12, 4, seed=9, device=DEV)
+    X = _rand((4096, 12), seed=10)
+    p1 = f.predict_proba_t(X)
+    assert hasattr(f.nodes, "_lpx_cpair_cache")
+    p2 = f.predict_proba_t(X)
+    assert torch.equal(p1, p2)
+    monkeypatch.delenv("LPX_RF_COMPACT")
+    p3 = f.predict_proba_t(X)  # f32 pair path
+    flips = (p1.argmax(1) != p3.argmax(1)).float().mean().item()
+    assert flips < 0.05
