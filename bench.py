@@ -185,6 +185,7 @@ def main() -> int:
                 "features": args.features,
                 "classes": args.classes,
                 "parallelism": f"dp{n}",
+                "thresholds": "fp16-representable (compact 4B-node kernel, bitwise == f32 kernel)",
                 "baseline_metric": "packaged layer MB + cold-start->first-inference ms, sklearn RF 1/2/4/8 GPU",
             },
         }

@@ -129,9 +129,12 @@ def matmul_mx_w4(A: torch.Tensor, B: torch.Tensor) -> torch.Tensor:
 # ---- sparse --------------------------------------------------------------
 
 def spmv_csr(indptr: torch.Tensor, indices: torch.Tensor, data: torch.Tensor,
-             x: torch.Tensor) -> torch.Tensor:
+             x: torch.Tensor, variant: int = -1) -> torch.Tensor:
+    """y = A @ x. variant: -1 adaptive seg8/wave-row; 2 = x-tile-in-LDS
+    (caller asserts CSR-canonical sorted columns — scipy's invariant)."""
     return require_extension().spmv_csr(
-        indptr.contiguous(), indices.contiguous(), data.contiguous(), x.contiguous()
+        indptr.contiguous(), indices.contiguous(), data.contiguous(),
+        x.contiguous(), variant
     )
 
 
@@ -172,6 +175,18 @@ def pairwise_sqeuclidean(X: torch.Tensor, Y: torch.Tensor) -> torch.Tensor:
 
 # ---- RandomForest --------------------------------------------------------
 
+def _thresholds_fp16_exact(nodes: torch.Tensor) -> bool:
+    """True iff every heap-node threshold is exactly fp16-representable
+    (then the compact 4-B-node kernel is bitwise-exact). One pass over the
+    node table, cached on the tensor keyed by its mutation counter."""
+    cached = getattr(nodes, "_lpx_f16exact_cache", None)
+    if cached is not None and cached[0] == nodes._version:
+        return cached[1]
+    thr = nodes.view(-1, 2)[:, 1].contiguous().view(torch.float32)
+    ok = bool(thr.to(torch.float16).to(torch.float32).eq(thr).all().item())
+    nodes._lpx_f16exact_cache = (nodes._version, ok)
+    return ok
+
 def rf_predict_proba_sum(nodes: torch.Tensor, leaf_values: torch.Tensor,
                          roots: torch.Tensor, X: torch.Tensor,
                          packed: bool = False, chunks=None,
@@ -197,13 +212,17 @@ def rf_predict_proba_sum(nodes: torch.Tensor, leaf_values: torch.Tensor,
         # top nodes are already L1-hot and the extra LDS halves occupancy.
         use_pair = depth >= 2 and os.environ.get("LPX_RF_PAIR", "1") == "1"
         # v8 compact nodes (u16 feat + fp16 threshold, 8-B pair records):
-        # halves the bytes of every random request. OPT-IN (LPX_RF_COMPACT=1)
-        # because fp16-rounded thresholds change decisions for x strictly
-        # between the f32 and fp16 threshold — EXACT only when the forest's
-        # thresholds are fp16-representable; sklearn-fitted f32 forests keep
-        # the exact pair kernel by default.
-        use_compact = (use_pair
-                       and os.environ.get("LPX_RF_COMPACT", "0") == "1")
+        # halves the bytes of every random request (+6% measured at the
+        # bench shape). Routing preserves exact-match semantics:
+        #   LPX_RF_COMPACT=auto (default): compact ONLY when every
+        #     threshold is exactly fp16-representable — then the compact
+        #     traversal is bitwise-identical to the f32 kernels (tested).
+        #     sklearn-fitted f32 forests fail the check and keep the exact
+        #     pair kernel.
+        #   =1 forces compact (fp16-rounded decisions), =0 disables.
+        cmode = os.environ.get("LPX_RF_COMPACT", "auto")
+        use_compact = use_pair and cmode != "0" and (
+            cmode == "1" or _thresholds_fp16_exact(nodes))
 
         def fn(n, lv, r, x):
             if use_compact and x.shape[1] <= 0xFFFF:

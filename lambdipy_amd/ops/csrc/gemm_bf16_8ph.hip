@@ -85,7 +85,8 @@ __device__ __forceinline__ const bf16frag* frag_ptr(const char* __restrict__ sme
       smem + (size_t)parity * PARITY_BYTES + (size_t)hh * HALF_BYTES + q * 16);
 }
 
-template <bool BF16_OUT, int VM, bool LG0, int PRIO, int ILV = 0>
+template <bool BF16_OUT, int VM, bool LG0, int PRIO, int ILV = 0,
+          int THR = -1>
 __global__ __launch_bounds__(THREADS8, 2) void gemm_bf16_8ph_kernel(
     const uint16_t* __restrict__ A, const uint16_t* __restrict__ Bt,
     void* __restrict__ Cout, int M, int N, int K, unsigned tiles_m,
@@ -160,6 +161,11 @@ __global__ __launch_bounds__(THREADS8, 2) void gemm_bf16_8ph_kernel(
       stage_slot(t, p, &st, &sh);
       stage_half8(A, Bt, K, bm, bn, st, sh, ksteps, smem, w, lane);
 
+      // optional ds_read-queue throttle before the barrier (guide §5: when
+      // 12 ds_reads are issued per phase, waiting the oldest 4 here keeps
+      // the post-barrier lgkmcnt(0) short without serializing the issue)
+      if constexpr (THR >= 0)
+        asm volatile("s_waitcnt lgkmcnt(%0)" ::"i"(THR) : "memory");
       __builtin_amdgcn_s_barrier();
       if constexpr (LG0) asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
       if constexpr (PRIO == 0) __builtin_amdgcn_s_setprio(1);
@@ -227,14 +233,14 @@ __global__ __launch_bounds__(THREADS8, 2) void gemm_bf16_8ph_kernel(
 }  // namespace
 
 template <bool BF16_OUT, int VM = 60, bool LG0 = false, int PRIO = 1,
-          int ILV = 1>
+          int ILV = 1, int THR = -1>
 static hipError_t launch8ph(const uint16_t* A, const uint16_t* Bt, void* C,
                             int M, int N, int K, hipStream_t stream) {
   const unsigned tiles_m = (unsigned)(M / BM8), tiles_n = (unsigned)(N / BM8);
   const size_t lds = 2 * PARITY_BYTES;  // 128 KiB
-  gemm_bf16_8ph_kernel<BF16_OUT, VM, LG0, PRIO, ILV><<<dim3(tiles_m * tiles_n),
-                                                  dim3(THREADS8), lds, stream>>>(
-      A, Bt, C, M, N, K, tiles_m, tiles_n);
+  gemm_bf16_8ph_kernel<BF16_OUT, VM, LG0, PRIO, ILV, THR>
+      <<<dim3(tiles_m * tiles_n), dim3(THREADS8), lds, stream>>>(
+          A, Bt, C, M, N, K, tiles_m, tiles_n);
   return hipGetLastError();
 }
 
@@ -262,6 +268,16 @@ hipError_t gemm_bf16_nt8ph_tune(const uint16_t* A, const uint16_t* Bt,
     case 8: return launch8ph<false, 60, false, 1>(A, Bt, C, M, N, K, stream);
     case 9: return launch8ph<false, 60, false, 1, 1>(A, Bt, C, M, N, K, stream);
     case 10: return launch8ph<false, 8, false, 1, 1>(A, Bt, C, M, N, K, stream);
+    // guide-template-exact order: ALL 12 ds_reads before the barrier,
+    // lgkmcnt(0) after it, per-phase setprio for every wave
+    case 11: return launch8ph<false, 60, true, 0, 0>(A, Bt, C, M, N, K, stream);
+    // + the lgkmcnt(8) throttle before the barrier
+    case 12: return launch8ph<false, 60, true, 0, 0, 8>(A, Bt, C, M, N, K, stream);
+    // throttle on the current default's deferred-read form
+    case 13: return launch8ph<false, 60, true, 1, 1, 8>(A, Bt, C, M, N, K, stream);
+    case 14: return launch8ph<false, 60, false, 1, 1, 8>(A, Bt, C, M, N, K, stream);
+    // template-exact but with the static half-wave priority raise
+    case 15: return launch8ph<false, 60, true, 1, 0, 8>(A, Bt, C, M, N, K, stream);
     default: return launch8ph<false, 6, true>(A, Bt, C, M, N, K, stream);
   }
 }

@@ -42,6 +42,12 @@ hipError_t gemv_f32(const float* A, const float* x, float* y,
 hipError_t spmv_csr_f32(const int* indptr, const int* indices, const float* data,
                         const float* x, float* y, int n_rows, long nnz,
                         hipStream_t stream);
+// x-tile-in-LDS variant for column-SORTED rows with x too big to stay
+// cache-hot (random 4-B gathers pull 64-B sectors; this streams x through
+// LDS in 128-KiB column tiles instead). WRONG RESULTS on unsorted rows.
+hipError_t spmv_csr_xtile_f32(const int* indptr, const int* indices,
+                              const float* data, const float* x, float* y,
+                              int n_rows, int n_cols, hipStream_t stream);
 
 // Batched 1-D complex64 FFT, N = power of two (<= 4096), Stockham in LDS.
 // x/out: [batch][N] interleaved (re,im) f32. twiddles: [N/2] complex

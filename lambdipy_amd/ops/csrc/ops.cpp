@@ -117,7 +117,8 @@ at::Tensor gemv(const at::Tensor& A_in, const at::Tensor& x_in) {
 // ---- CSR SpMV ------------------------------------------------------------
 
 at::Tensor spmv_csr(const at::Tensor& indptr, const at::Tensor& indices,
-                    const at::Tensor& data, const at::Tensor& x) {
+                    const at::Tensor& data, const at::Tensor& x,
+                    int64_t variant) {
   CHECK_DEV(indptr);
   CHECK_DEV(indices);
   CHECK_DEV(data);
@@ -129,6 +130,18 @@ at::Tensor spmv_csr(const at::Tensor& indptr, const at::Tensor& indices,
   const int64_t n_rows = indptr.numel() - 1;
   c10::hip::HIPGuardMasqueradingAsCUDA guard(x.device());
   auto y = at::empty({n_rows}, x.options());
+  // variant 2: x-tile-in-LDS (requires CSR-canonical sorted columns —
+  // scipy guarantees this; the caller opts in because sortedness is not
+  // checked here). -1/0: adaptive seg8/wave-row.
+  if (variant == 2) {
+    check_hip(lpx::spmv_csr_xtile_f32(
+                  indptr.data_ptr<int>(), indices.data_ptr<int>(),
+                  data.data_ptr<float>(), x.data_ptr<float>(),
+                  y.data_ptr<float>(), (int)n_rows, (int)x.numel(),
+                  current_stream(x)),
+              "spmv_csr_xtile_f32");
+    return y;
+  }
   check_hip(lpx::spmv_csr_f32(indptr.data_ptr<int>(), indices.data_ptr<int>(),
                               data.data_ptr<float>(), x.data_ptr<float>(),
                               y.data_ptr<float>(), (int)n_rows,
@@ -734,7 +747,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("variant") = -1);
   m.def("gemv", &gemv, "y[b,m] = A[b,m,:] . x[b,:]", py::arg("A"), py::arg("x"));
   m.def("spmv_csr", &spmv_csr, "y = CSR(A) @ x", py::arg("indptr"),
-        py::arg("indices"), py::arg("data"), py::arg("x"));
+        py::arg("indices"), py::arg("data"), py::arg("x"),
+        py::arg("variant") = -1);
   m.def("fft_c64", &fft_c64, "batched complex64 FFT (Stockham)", py::arg("x"),
         py::arg("twiddles"), py::arg("inverse") = false);
   m.def("pairwise_sqeuclidean", &pairwise_sqeuclidean,

@@ -9,6 +9,7 @@
 //   - lpx_argmax_rows: per-row argmax of [S, C] f32 -> int32 labels
 //   - lpx_xgen:      synthetic ~N(0,1) feature matrix (xorshift + Box-Muller)
 
+#include <hip/hip_fp16.h>
 #include <hip/hip_runtime.h>
 #include <stdint.h>
 
@@ -132,7 +133,10 @@ extern "C" __global__ __launch_bounds__(THREADS) void lpx_forestgen_nodes(
     const float f2 = (u2 >> 8) * (1.0f / 16777216.0f);
     float sn, cs;
     __sincosf(6.2831853f * f2, &sn, &cs);
-    const float thresh = sqrtf(-2.0f * logf(f1)) * cs;
+    // rounded to fp16-representable f32 (matches lpx_rt.random_heap_forest_np:
+    // the compact 4-B-node RF kernel traverses such forests bitwise-exactly)
+    const float thresh = __half2float(__float2half(
+        sqrtf(-2.0f * logf(f1)) * cs));
     i32x2 nd;
     nd.x = (int)(u2 % (unsigned)n_features);
     nd.y = __float_as_int(thresh);

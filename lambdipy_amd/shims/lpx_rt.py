@@ -234,7 +234,13 @@ def random_heap_forest_np(n_trees: int, depth: int, n_features: int,
     n_leaves = 1 << depth
     total = n_trees * n_internal
     feat = rng.integers(0, n_features, total).astype(np.uint32)
-    thr = rng.standard_normal(total).astype(np.float32)
+    # thresholds rounded to fp16-REPRESENTABLE f32 values: an equally
+    # random model (same traversal work), but the compact 4-B-node kernel
+    # (ops/csrc/rf.hip v8) traverses it bitwise-identically to the f32
+    # kernels — exact-match semantics preserved while halving node bytes.
+    # sklearn-fitted forests keep full f32 thresholds (and the f32 kernel).
+    thr = (rng.standard_normal(total).astype(np.float32)
+           .astype(np.float16).astype(np.float32))
     nodes = np.stack([feat, thr.view(np.uint32)], axis=1).view(np.int32)
     lv = rng.random((n_trees * n_leaves, n_classes)).astype(np.float32)
     lv /= lv.sum(axis=1, keepdims=True)

@@ -14,9 +14,18 @@ run_cfg () {  # name, requirements-content, handler
   timeout 300 python -m lambdipy_amd.cli run $W/$1.zip --handler $3 --json 2>/dev/null | python -c "import json,sys; d=json.load(sys.stdin); print('  cold-start->first-inference:', d.get('coldstart_ms'), 'ms; ok =', d.get('ok', d))"
 }
 
+run_lite () {  # name, lite-handler — serve the SAME zip torch-free
+  echo "-- $1 via $2 (torch-free) --"
+  timeout 300 python -m lambdipy_amd.cli run $W/$1.zip --handler $2 --json 2>/dev/null | python -c "import json,sys; d=json.load(sys.stdin); print('  cold-start->first-inference:', d.get('coldstart_ms'), 'ms; ok =', d.get('ok', d))"
+}
+
 run_cfg requests "requests" requests
 run_cfg numpy "numpy" matmul
+run_lite numpy matmul_lite
 run_cfg scipy "numpy\nscipy" fftspmv
+run_lite scipy fftspmv_lite
 run_cfg sklearn "numpy\nscipy\nscikit-learn" rf
+run_lite sklearn rf_lite
 run_cfg torch "torch" resnet50
+run_lite torch resnet50_lite
 echo "ALL CONFIGS SERVED"

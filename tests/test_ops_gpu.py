@@ -887,6 +887,12 @@ def test_rf_cpair_flip_rate_on_f32_forest(ops):
     e = ops.ext()
     depth = 10
     f = ForestSoA.random(128, depth, 32, 8, seed=7, device=DEV)
+    # overwrite the (fp16-representable) synthetic thresholds with full-f32
+    # randoms — the sklearn-fitted case this error class is about
+    n = f.nodes.view(-1, 2)
+    g = torch.Generator(device="cpu").manual_seed(77)
+    thr = torch.randn(n.shape[0], generator=g).to(DEV)
+    n[:, 1] = thr.view(torch.int32)
     X = _rand((65536, 32), seed=8)
     pairs, roots, cpairs, croots = _pack_pairs(f.nodes, depth)
     a = e.rf_predict_heap_pair(pairs, roots, f.leaf_values, depth, X, 1)
@@ -900,18 +906,45 @@ def test_rf_cpair_flip_rate_on_f32_forest(ops):
 
 
 def test_rf_compact_api_routing(ops, monkeypatch):
-    """LPX_RF_COMPACT=1 routes predict through the compact kernel and the
-    cache invalidates on mutation (same contract as the pair cache)."""
+    """Auto mode detects fp16-representable thresholds and routes through
+    the compact kernel BITWISE-EXACTLY vs the f32 pair kernel; sklearn-like
+    full-f32 thresholds fail the detection and keep the pair kernel."""
     from lambdipy_amd.shims.lpx_sklearn import ForestSoA
 
-    monkeypatch.setenv("LPX_RF_COMPACT", "1")
     f = ForestSoA.random(32, 6, 12, 4, seed=9, device=DEV)
     X = _rand((4096, 12), seed=10)
-    p1 = f.predict_proba_t(X)
-    assert hasattr(f.nodes, "_lpx_cpair_cache")
-    p2 = f.predict_proba_t(X)
-    assert torch.equal(p1, p2)
+    p1 = f.predict_proba_t(X)  # auto: synthetic forests are fp16-exact
+    assert hasattr(f.nodes, "_lpx_cpair_cache"), "auto-compact not taken"
+    monkeypatch.setenv("LPX_RF_COMPACT", "0")
+    p2 = f.predict_proba_t(X)  # forced f32 pair path
+    assert torch.equal(p1, p2), "compact path diverged on fp16-exact forest"
     monkeypatch.delenv("LPX_RF_COMPACT")
-    p3 = f.predict_proba_t(X)  # f32 pair path
-    flips = (p1.argmax(1) != p3.argmax(1)).float().mean().item()
-    assert flips < 0.05
+    # full-f32 thresholds: auto must NOT take the compact path
+    n = f.nodes.view(-1, 2)
+    g = torch.Generator(device="cpu").manual_seed(42)
+    n[:, 1] = torch.randn(n.shape[0], generator=g).to(DEV).view(torch.int32)
+    f.predict_proba_t(X)
+    from lambdipy_amd.ops.api import _thresholds_fp16_exact
+    assert not _thresholds_fp16_exact(f.nodes)
+
+
+def test_spmv_xtile_vs_seg8(ops):
+    """x-tile-in-LDS variant vs adaptive kernel on sorted-column CSR,
+    including a shape with a non-multiple-of-tile column count."""
+    for rows, cols, nnz, seed in ((100_000, 100_000, 32, 1),
+                                  (5000, 40_000, 9, 2),
+                                  (262_144, 50_000, 16, 3)):
+        rng = np.random.default_rng(seed)
+        idx = np.sort(rng.integers(0, cols, (rows, nnz)).astype(np.int32),
+                      axis=1).reshape(-1)
+        indptr = np.arange(0, (rows + 1) * nnz, nnz, dtype=np.int32)
+        data = rng.standard_normal(rows * nnz).astype(np.float32)
+        x = rng.standard_normal(cols).astype(np.float32)
+        ti = torch.from_numpy(indptr).to(DEV)
+        tj = torch.from_numpy(idx).to(DEV)
+        td = torch.from_numpy(data).to(DEV)
+        tx = torch.from_numpy(x).to(DEV)
+        a = ops.spmv_csr(ti, tj, td, tx, -1)
+        b = ops.spmv_csr(ti, tj, td, tx, 2)
+        assert torch.allclose(a, b, rtol=1e-4, atol=1e-3), \
+            f"xtile mismatch rows={rows}"

@@ -85,11 +85,13 @@ def test_fuzz_gemm_fp8(ops, m, n, k, seed):
     aq = e.quantize_fp8(A, sa)
     bq = e.quantize_fp8(B, sb)
     C = e.gemm_fp8(aq, bq, sa * sb, None, False)
-    ref = A @ B.t()
-    # e4m3 has ~2 mantissa-bit precision: tolerance scales with K
-    tol = 0.07 * max(1.0, np.sqrt(k / 64.0))
+    # reference on the DEQUANTIZED operands: isolates the GEMM numerics
+    # from e4m3 rounding (f32 accumulation in both -> tight tolerance)
+    aqf = aq.view(torch.float8_e4m3fn).float() * sa
+    bqf = bq.view(torch.float8_e4m3fn).float() * sb
+    ref = aqf @ bqf.t()
     err = (C - ref).abs().max().item()
-    assert err / (ref.abs().max().item() + 1e-6) < tol, \
+    assert err / (ref.abs().max().item() + 1e-6) < 1e-3, \
         f"({m},{n},{k}) rel err {err/(ref.abs().max().item()+1e-6)}"
 
 
