@@ -345,23 +345,32 @@ class Comms:
                   root: int = 0) -> None:
         rc = _rccl_typed_lib()
         _nck(rc.ncclGroupStart(), "ncclGroupStart")
-        for d in range(self.ndev):
-            set_device(d)
-            # each rank passes device-local pointers; only root's send matters
-            _nck(rc.ncclBroadcast(bufs[d], bufs[d], count, dtype, root,
-                                  self.comms[d], None),
-                 f"ncclBroadcast(dev {d})")
+        try:
+            for d in range(self.ndev):
+                set_device(d)
+                # each rank passes device-local pointers; only root's send
+                # matters
+                _nck(rc.ncclBroadcast(bufs[d], bufs[d], count, dtype, root,
+                                      self.comms[d], None),
+                     f"ncclBroadcast(dev {d})")
+        except Exception:
+            rc.ncclGroupEnd()  # never leave the group open on error
+            raise
         _nck(rc.ncclGroupEnd(), "ncclGroupEnd")
 
     def all_gather(self, send: Sequence[int], recv: Sequence[int],
                    count: int, dtype: int) -> None:
         rc = _rccl_typed_lib()
         _nck(rc.ncclGroupStart(), "ncclGroupStart")
-        for d in range(self.ndev):
-            set_device(d)
-            _nck(rc.ncclAllGather(send[d], recv[d], count, dtype,
-                                  self.comms[d], None),
-                 f"ncclAllGather(dev {d})")
+        try:
+            for d in range(self.ndev):
+                set_device(d)
+                _nck(rc.ncclAllGather(send[d], recv[d], count, dtype,
+                                      self.comms[d], None),
+                     f"ncclAllGather(dev {d})")
+        except Exception:
+            rc.ncclGroupEnd()
+            raise
         _nck(rc.ncclGroupEnd(), "ncclGroupEnd")
 
     def destroy(self) -> None:
