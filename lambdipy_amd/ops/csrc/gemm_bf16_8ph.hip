@@ -86,7 +86,7 @@ __device__ __forceinline__ const bf16frag* frag_ptr(const char* __restrict__ sme
 }
 
 template <bool BF16_OUT, int VM, bool LG0, int PRIO, int ILV = 0,
-          int THR = -1>
+          int THR = -1, bool ASM = false>
 __global__ __launch_bounds__(THREADS8, 2) void gemm_bf16_8ph_kernel(
     const uint16_t* __restrict__ A, const uint16_t* __restrict__ Bt,
     void* __restrict__ Cout, int M, int N, int K, unsigned tiles_m,
@@ -140,6 +140,58 @@ __global__ __launch_bounds__(THREADS8, 2) void gemm_bf16_8ph_kernel(
     for (int p = 0; p < 4; ++p) {
       const int rh = (p >> 1) & 1;  // quadrant row half
       const int ch = p & 1;        // quadrant col half
+      // ASM variant (tune 16): one inline-asm block per k-half — 6
+      // ds_read_b128 issued back-to-back, then the 8 MFMAs gated by
+      // DESCENDING counted lgkmcnt (4/2/1/0: wait only for the operands
+      // the next MFMA pair needs; LDS ops retire in order). The compiler's
+      // own schedule drains lgkmcnt(0) before each MFMA group (see
+      // disassembly), serializing read latency the counted form hides.
+      auto asm_ks = [&](int ks) {
+        const int cell = ks * 4 + fkb;
+        const size_t pbase = (size_t)parity * PARITY_BYTES;
+        unsigned aA[4], aB[2];
+#pragma unroll
+        for (int mf = 0; mf < 4; ++mf)
+          aA[mf] = (unsigned)(size_t)(
+              __attribute__((address_space(3))) char*)(
+              smem + pbase + (size_t)rh * HALF_BYTES +
+              phys_cell(wsr * 64 + mf * 16 + fr, cell) * 16);
+#pragma unroll
+        for (int nf = 0; nf < 2; ++nf)
+          aB[nf] = (unsigned)(size_t)(
+              __attribute__((address_space(3))) char*)(
+              smem + pbase + (size_t)(2 + ch) * HALF_BYTES +
+              phys_cell(wsc * 32 + nf * 16 + fr, cell) * 16);
+        bf16frag A0, A1, A2, A3, B0, B1;
+        asm volatile(
+            "ds_read_b128 %[A0], %[aA0]\n\t"
+            "ds_read_b128 %[B0], %[aB0]\n\t"
+            "ds_read_b128 %[A1], %[aA1]\n\t"
+            "ds_read_b128 %[B1], %[aB1]\n\t"
+            "ds_read_b128 %[A2], %[aA2]\n\t"
+            "ds_read_b128 %[A3], %[aA3]\n\t"
+            "s_waitcnt lgkmcnt(4)\n\t"
+            "v_mfma_f32_16x16x32_bf16 %[C00], %[A0], %[B0], %[C00]\n\t"
+            "s_waitcnt lgkmcnt(2)\n\t"
+            "v_mfma_f32_16x16x32_bf16 %[C01], %[A0], %[B1], %[C01]\n\t"
+            "v_mfma_f32_16x16x32_bf16 %[C10], %[A1], %[B0], %[C10]\n\t"
+            "v_mfma_f32_16x16x32_bf16 %[C11], %[A1], %[B1], %[C11]\n\t"
+            "s_waitcnt lgkmcnt(1)\n\t"
+            "v_mfma_f32_16x16x32_bf16 %[C20], %[A2], %[B0], %[C20]\n\t"
+            "v_mfma_f32_16x16x32_bf16 %[C21], %[A2], %[B1], %[C21]\n\t"
+            "s_waitcnt lgkmcnt(0)\n\t"
+            "v_mfma_f32_16x16x32_bf16 %[C30], %[A3], %[B0], %[C30]\n\t"
+            "v_mfma_f32_16x16x32_bf16 %[C31], %[A3], %[B1], %[C31]"
+            : [C00] "+v"(acc[p][0][0]), [C01] "+v"(acc[p][0][1]),
+              [C10] "+v"(acc[p][1][0]), [C11] "+v"(acc[p][1][1]),
+              [C20] "+v"(acc[p][2][0]), [C21] "+v"(acc[p][2][1]),
+              [C30] "+v"(acc[p][3][0]), [C31] "+v"(acc[p][3][1]),
+              [A0] "=&v"(A0), [A1] "=&v"(A1), [A2] "=&v"(A2),
+              [A3] "=&v"(A3), [B0] "=&v"(B0), [B1] "=&v"(B1)
+            : [aA0] "v"(aA[0]), [aA1] "v"(aA[1]), [aA2] "v"(aA[2]),
+              [aA3] "v"(aA[3]), [aB0] "v"(aB[0]), [aB1] "v"(aB[1])
+            : "memory");
+      };
       // fragment reads for this quadrant (ILV=1: defer the ks=1 half past
       // the barrier so its ds_read latency hides under the ks=0 MFMAs)
       bf16frag aF[4][2], bF[2][2];
@@ -154,6 +206,14 @@ __global__ __launch_bounds__(THREADS8, 2) void gemm_bf16_8ph_kernel(
           bF[nf][ks] = *frag_ptr(smem, parity, 2 + ch,
                                  wsc * 32 + nf * 16 + fr, cell);
       };
+      if constexpr (ASM) {
+        asm_ks(0);  // reads + MFMAs of the first k-half (ends lgkm-clean)
+        int st0, sh0;
+        stage_slot(t, p, &st0, &sh0);
+        stage_half8(A, Bt, K, bm, bn, st0, sh0, ksteps, smem, w, lane);
+        __builtin_amdgcn_s_barrier();
+        asm_ks(1);
+      } else {
       read_ks(0);
       if constexpr (ILV == 0) read_ks(1);
       // stage the slot this phase frees
@@ -190,6 +250,7 @@ __global__ __launch_bounds__(THREADS8, 2) void gemm_bf16_8ph_kernel(
                   aF[mf][ks], bF[nf][ks], acc[p][mf][nf], 0, 0, 0);
       }
       if constexpr (PRIO == 0) __builtin_amdgcn_s_setprio(0);
+      }
       if constexpr (VM == 60) {
         // guide-template placement: wait only at phases 1/3 (twice per
         // K-tile); every read is >=4 calls back, allow-last-3 covers it
@@ -233,12 +294,12 @@ __global__ __launch_bounds__(THREADS8, 2) void gemm_bf16_8ph_kernel(
 }  // namespace
 
 template <bool BF16_OUT, int VM = 60, bool LG0 = false, int PRIO = 1,
-          int ILV = 1, int THR = -1>
+          int ILV = 1, int THR = -1, bool ASM = false>
 static hipError_t launch8ph(const uint16_t* A, const uint16_t* Bt, void* C,
                             int M, int N, int K, hipStream_t stream) {
   const unsigned tiles_m = (unsigned)(M / BM8), tiles_n = (unsigned)(N / BM8);
   const size_t lds = 2 * PARITY_BYTES;  // 128 KiB
-  gemm_bf16_8ph_kernel<BF16_OUT, VM, LG0, PRIO, ILV, THR>
+  gemm_bf16_8ph_kernel<BF16_OUT, VM, LG0, PRIO, ILV, THR, ASM>
       <<<dim3(tiles_m * tiles_n), dim3(THREADS8), lds, stream>>>(
           A, Bt, C, M, N, K, tiles_m, tiles_n);
   return hipGetLastError();
@@ -250,7 +311,10 @@ bool gemm_bf16_8ph_supported(int M, int N, int K) {
 
 hipError_t gemm_bf16_nt8ph(const uint16_t* A, const uint16_t* Bt, float* C,
                            int M, int N, int K, hipStream_t stream) {
-  return launch8ph<false>(A, Bt, C, M, N, K, stream);
+  // tune-14 config (r2 A/B): vmcnt(6) at odd phases, deferred ks=1 reads,
+  // static half-wave priority, lgkmcnt(8) throttle before the barrier —
+  // 1139/1144 TF median @4096/8192 vs 1102/1133 for the old default.
+  return launch8ph<false, 60, false, 1, 1, 8>(A, Bt, C, M, N, K, stream);
 }
 
 // tuning variants for within-probe A/B (scripts/ab_gemm8.py)
@@ -278,6 +342,9 @@ hipError_t gemm_bf16_nt8ph_tune(const uint16_t* A, const uint16_t* Bt,
     case 14: return launch8ph<false, 60, false, 1, 1, 8>(A, Bt, C, M, N, K, stream);
     // template-exact but with the static half-wave priority raise
     case 15: return launch8ph<false, 60, true, 1, 0, 8>(A, Bt, C, M, N, K, stream);
+    // hand-asm phase body: counted descending lgkmcnt between ds_reads
+    // and MFMAs (the compiler's schedule drains lgkmcnt(0) per group)
+    case 16: return launch8ph<false, 60, false, 1, 1, -1, true>(A, Bt, C, M, N, K, stream);
     default: return launch8ph<false, 6, true>(A, Bt, C, M, N, K, stream);
   }
 }

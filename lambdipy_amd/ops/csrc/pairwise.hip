@@ -114,6 +114,119 @@ __global__ __launch_bounds__(THREADS) void pairwise_kernel(
   }
 }
 
+// ---- v2: 128x128 tile on v_mfma_f32_32x32x2_f32 --------------------------
+// The v1 kernel is LDS-read bound: 16x16x4 consumes one scalar b32 A-read
+// per 2048 FLOP. The 32x32x2 op doubles FLOPs per operand register (4096)
+// and the 128^2 tile halves staging traffic per FLOP; 4 waves, each owning
+// a 64x64 output (2x2 fragments of 32x32), BK=16 double-buffered (~17 KiB
+// LDS). Same exact-f32 numerics (fmaf chains) as v1.
+constexpr int BM2 = 128, BK2 = 16;
+constexpr int ROWF2 = BK2 + 1;
+
+__device__ __forceinline__ void stage_f32_128(const float* __restrict__ src,
+                                              int rows_total, int K, int row0,
+                                              int k0, float* __restrict__ lds) {
+  // 128 rows x 16 f32 = 8 KiB; 256 threads x 2 float4 chunks
+  const int t = threadIdx.x;
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    const int c = t + i * THREADS;
+    const int r = c >> 2, kb = (c & 3) * 4;
+    const int grow = row0 + r, gk = k0 + kb;
+    f32x4 v = {0.f, 0.f, 0.f, 0.f};
+    if (grow < rows_total) {
+      if (gk + 4 <= K) {
+        v = *reinterpret_cast<const f32x4*>(src + (long)grow * K + gk);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          if (gk + j < K) v[j] = src[(long)grow * K + gk + j];
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < 4; ++j) lds[r * ROWF2 + kb + j] = v[j];
+  }
+}
+
+__global__ __launch_bounds__(THREADS) void pairwise2_kernel(
+    const float* __restrict__ X, const float* __restrict__ Y,
+    const float* __restrict__ xsq, const float* __restrict__ ysq,
+    float* __restrict__ D, int M, int N, int K, unsigned tiles_m,
+    unsigned tiles_n) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* const fbase = reinterpret_cast<float*>(smem);
+  const int BUF = BM2 * ROWF2;  // floats
+
+  const unsigned nwg = tiles_m * tiles_n;
+  const unsigned bid = lpx_xcd_swizzle(blockIdx.x, nwg);
+  const int bm = (int)(bid / tiles_n) * BM2;
+  const int bn = (int)(bid % tiles_n) * BM2;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wr = wid >> 1, wc = wid & 1;  // wave covers 64x64
+
+  f32x16 acc[2][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int e = 0; e < 16; ++e) acc[i][j][e] = 0.f;
+
+  const int fr = lane & 31;      // fragment row
+  const int fk = lane >> 5;      // k-lane (0..1)
+
+  int cur = 0;
+  stage_f32_128(X, M, K, bm, 0, fbase);
+  stage_f32_128(Y, N, K, bn, 0, fbase + 2 * BUF);
+  __syncthreads();
+
+  const int ksteps = (K + BK2 - 1) / BK2;
+  for (int ks = 0; ks < ksteps; ++ks) {
+    float* const xCur = fbase + cur * BUF;
+    float* const yCur = fbase + 2 * BUF + cur * BUF;
+    if (ks + 1 < ksteps) {
+      stage_f32_128(X, M, K, bm, (ks + 1) * BK2, fbase + (cur ^ 1) * BUF);
+      stage_f32_128(Y, N, K, bn, (ks + 1) * BK2,
+                    fbase + 2 * BUF + (cur ^ 1) * BUF);
+    }
+#pragma unroll
+    for (int kk = 0; kk < BK2 / 2; ++kk) {
+      const int kof = kk * 2 + fk;
+      float a0 = xCur[(wr * 64 + fr) * ROWF2 + kof];
+      float a1 = xCur[(wr * 64 + 32 + fr) * ROWF2 + kof];
+      float b0 = yCur[(wc * 64 + fr) * ROWF2 + kof];
+      float b1 = yCur[(wc * 64 + 32 + fr) * ROWF2 + kof];
+      acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc[0][0], 0, 0, 0);
+      acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc[0][1], 0, 0, 0);
+      acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc[1][0], 0, 0, 0);
+      acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc[1][1], 0, 0, 0);
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  // 32x32 D fragment: col = lane&31, rows = 4*(lane>>5) + (r&3) + 8*(r>>2)
+  const int ocol_l = lane & 31;
+  const int orow_base = 4 * (lane >> 5);
+#pragma unroll
+  for (int mf = 0; mf < 2; ++mf) {
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int row = bm + wr * 64 + mf * 32 + orow_base + (r & 3) + 8 * (r >> 2);
+        const int col = bn + wc * 64 + nf * 32 + ocol_l;
+        if (row < M && col < N) {
+          float d = xsq[row] - 2.0f * acc[mf][nf][r] + ysq[col];
+          D[(long)row * N + col] = d > 0.f ? d : 0.f;
+        }
+      }
+    }
+  }
+}
+
 __global__ __launch_bounds__(THREADS) void row_sqnorm_kernel(
     const float* __restrict__ X, float* __restrict__ out, int M, long K) {
   const int lane = threadIdx.x & 63;
@@ -141,6 +254,15 @@ hipError_t pairwise_sqeuclidean_f32(const float* X, const float* Y,
                                     const float* xsq, const float* ysq,
                                     float* D, int M, int N, int K,
                                     hipStream_t stream) {
+  const char* pv = getenv("LPX_PW_V1");
+  if ((M >= 256 && N >= 256) && !(pv && pv[0] == '1')) {
+    // v2: 32x32x2 MFMA, 128^2 tile (LDS-read pressure halved per FLOP)
+    const unsigned tm = lpx_ceil_div(M, BM2), tn = lpx_ceil_div(N, BM2);
+    const size_t lds = sizeof(float) * 4u * BM2 * ROWF2;
+    pairwise2_kernel<<<dim3(tm * tn), dim3(THREADS), lds, stream>>>(
+        X, Y, xsq, ysq, D, M, N, K, tm, tn);
+    return hipGetLastError();
+  }
   const unsigned tiles_m = lpx_ceil_div(M, BM), tiles_n = lpx_ceil_div(N, BN);
   const size_t lds = sizeof(float) * 4u * BM * ROWF;
   pairwise_kernel<<<dim3(tiles_m * tiles_n), dim3(THREADS), lds, stream>>>(
