@@ -242,6 +242,17 @@ def rf_predict_proba_sum(nodes: torch.Tensor, leaf_values: torch.Tensor,
                     cached = (n._version, cp, cn[:, 0].contiguous())
                     n._lpx_cpair_cache = cached
                 _, cp, rts = cached
+                # fp16 leaf table when exact (one 16-B leaf request): same
+                # auto-exact rule as the node thresholds
+                lvc = getattr(lv, "_lpx_lv16_cache", None)
+                if lvc is None or lvc[0] != lv._version:
+                    lv16 = lv.to(torch.float16)
+                    ok = bool(lv16.to(torch.float32).eq(lv).all().item())
+                    lvc = (lv._version, lv16.contiguous() if ok else None)
+                    lv._lpx_lv16_cache = lvc
+                if lvc[1] is not None:
+                    return e.rf_predict_heap_cpair16(cp, rts, lvc[1], depth,
+                                                     x, 1)
                 return e.rf_predict_heap_cpair(cp, rts, lv, depth, x, 1)
             if use_pair:
                 cached = getattr(n, "_lpx_pair_cache", None)

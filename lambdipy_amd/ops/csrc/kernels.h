@@ -110,6 +110,14 @@ hipError_t rf_predict_heap_cpair_f32(const int* cpairs, const int* roots,
                                      int depth, const float* X, float* out,
                                      int S, int F, int C, int silp,
                                      hipStream_t stream);
+// fp16 leaf table variant (lv16 = fp16 bits [rows, C], C <= 8): one 16-B
+// leaf request; exact when leaves are fp16-representable.
+hipError_t rf_predict_heap_cpair_lv16_f32(const int* cpairs, const int* roots,
+                                          const unsigned short* lv16,
+                                          int n_trees, int depth,
+                                          const float* X, float* out, int S,
+                                          int F, int C, int silp,
+                                          hipStream_t stream);
 // v5: two-level heap — top `top` levels served from LDS (top<0: auto-pick
 // largest fitting 2-blocks/CU; falls back to v4 when nothing fits).
 hipError_t rf_predict_heap2_f32(const int* nodes_heap, const float* leaf_values,

@@ -703,6 +703,43 @@ at::Tensor rf_predict_heap_cpair(const at::Tensor& cpairs,
   return out;
 }
 
+at::Tensor rf_predict_heap_cpair16(const at::Tensor& cpairs,
+                                   const at::Tensor& roots,
+                                   const at::Tensor& lv16, int64_t depth,
+                                   const at::Tensor& X, int64_t silp) {
+  // v8 + fp16 leaf table (one 16-B leaf request; exact for
+  // fp16-representable leaf values).
+  CHECK_DEV(cpairs);
+  CHECK_DEV(roots);
+  CHECK_DEV(lv16);
+  CHECK_DEV(X);
+  TORCH_CHECK(cpairs.scalar_type() == at::kInt && cpairs.dim() == 2 &&
+              cpairs.size(1) == 2, "cpairs must be int32 [n, 2]");
+  TORCH_CHECK(roots.scalar_type() == at::kInt && roots.dim() == 1,
+              "roots must be int32 [n_trees]");
+  TORCH_CHECK(lv16.scalar_type() == at::kHalf && lv16.dim() == 2,
+              "lv16 must be fp16 [rows, C]");
+  TORCH_CHECK(X.scalar_type() == at::kFloat && X.dim() == 2,
+              "X must be f32 [S,F]");
+  const int64_t S = X.size(0), F = X.size(1), C = lv16.size(1);
+  TORCH_CHECK(C <= 8 && F <= 0xffff, "rf_predict_heap_cpair16: C<=8, F<=u16");
+  const int64_t n_trees = roots.size(0);
+  const int64_t ppt = (1LL << (depth - 1)) - 1;
+  TORCH_CHECK(cpairs.size(0) == n_trees * ppt, "cpairs/depth mismatch");
+  TORCH_CHECK(lv16.size(0) == n_trees * (1LL << depth),
+              "lv16 rows != n_trees * 2^depth");
+  c10::hip::HIPGuardMasqueradingAsCUDA guard(X.device());
+  auto out = at::empty({S, C}, X.options());
+  check_hip(lpx::rf_predict_heap_cpair_lv16_f32(
+                cpairs.data_ptr<int>(), roots.data_ptr<int>(),
+                reinterpret_cast<const unsigned short*>(lv16.data_ptr()),
+                (int)n_trees, (int)depth, X.data_ptr<float>(),
+                out.data_ptr<float>(), (int)S, (int)F, (int)C, (int)silp,
+                current_stream(X)),
+            "rf_predict_heap_cpair_lv16_f32");
+  return out;
+}
+
 at::Tensor conv_fp8(const at::Tensor& xq, const at::Tensor& Wq,
                     int64_t H, int64_t W, int64_t Cin, int64_t R, int64_t S,
                     int64_t stride, int64_t pad, double descale,
@@ -771,6 +808,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rf_predict_heap_cpair", &rf_predict_heap_cpair,
         "v8 compact child-pair heap traversal (u16 feat + fp16 thr)",
         py::arg("cpairs"), py::arg("roots"), py::arg("leaf_values"),
+        py::arg("depth"), py::arg("X"), py::arg("silp") = 1);
+  m.def("rf_predict_heap_cpair16", &rf_predict_heap_cpair16,
+        "v8 compact traversal with fp16 leaf table",
+        py::arg("cpairs"), py::arg("roots"), py::arg("lv16"),
         py::arg("depth"), py::arg("X"), py::arg("silp") = 1);
   m.def("rf_predict8_lds", &rf_predict8_lds,
         "RandomForest per-class sums (forest chunks staged in LDS)",

@@ -120,16 +120,16 @@ __global__ __launch_bounds__(THREADS) void pairwise_kernel(
 // and the 128^2 tile halves staging traffic per FLOP; 4 waves, each owning
 // a 64x64 output (2x2 fragments of 32x32), BK=16 double-buffered (~17 KiB
 // LDS). Same exact-f32 numerics (fmaf chains) as v1.
-constexpr int BM2 = 128, BK2 = 16;
+constexpr int BM2 = 128, BK2 = 32;  // BK=16 measured barrier-bound (~85 TF)
 constexpr int ROWF2 = BK2 + 1;
 
 __device__ __forceinline__ void stage_f32_128(const float* __restrict__ src,
                                               int rows_total, int K, int row0,
                                               int k0, float* __restrict__ lds) {
-  // 128 rows x 16 f32 = 8 KiB; 256 threads x 2 float4 chunks
+  // 128 rows x BK2 f32; 256 threads x (BM2*BK2/4/256) float4 chunks
   const int t = threadIdx.x;
 #pragma unroll
-  for (int i = 0; i < 2; ++i) {
+  for (int i = 0; i < BM2 * BK2 / 4 / THREADS; ++i) {
     const int c = t + i * THREADS;
     const int r = c >> 2, kb = (c & 3) * 4;
     const int grow = row0 + r, gk = k0 + kb;

@@ -466,12 +466,15 @@ __device__ __forceinline__ float lpx_h2f(unsigned short u) {
   return __half2float(__half(r));
 }
 
-template <bool STAGE_X, int SILP>
+template <bool STAGE_X, int SILP, bool LV16 = false>
 __global__ __launch_bounds__(THREADS) void rf_heap_cpair_kernel(
     const i32x2* __restrict__ pairs, const int* __restrict__ roots,
     const float* __restrict__ leaf_values, int n_trees, int depth,
     const float* __restrict__ X, float* __restrict__ out, int S, int F,
     int C) {
+  // LV16: leaf_values is fp16 [rows, C] (C <= 8): the leaf row is ONE 16-B
+  // request instead of 32 B — exact when leaves are fp16-representable
+  // (the synthetic generators quantize; accumulation stays f32).
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float* xl = reinterpret_cast<float*>(smem);
   const int t = threadIdx.x;
@@ -536,10 +539,19 @@ __global__ __launch_bounds__(THREADS) void rf_heap_cpair_kernel(
         if (s[j] < S) {
           const int go = (xrow[j][cf[j]] <= ct[j]) ? 0 : 1;
           const int leaf = 2 * nd[j] + 1 + go - tree_stride;
-          const float* lv = leaf_values + ((long)tr * n_leaves + leaf) * C;
+          if constexpr (LV16) {
+            const unsigned short* lv16 =
+                reinterpret_cast<const unsigned short*>(leaf_values) +
+                ((long)tr * n_leaves + leaf) * C;
 #pragma unroll
-          for (int c = 0; c < CMAX; ++c)
-            if (c < C) acc[j][c] += lv[c];
+            for (int c = 0; c < CMAX; ++c)
+              if (c < C) acc[j][c] += lpx_h2f(lv16[c]);
+          } else {
+            const float* lv = leaf_values + ((long)tr * n_leaves + leaf) * C;
+#pragma unroll
+            for (int c = 0; c < CMAX; ++c)
+              if (c < C) acc[j][c] += lv[c];
+          }
         }
       }
     }
@@ -866,6 +878,42 @@ hipError_t rf_predict_heap_cpair_f32(const int* cpairs, const int* roots,
   LPX_RFHC_CASE(2)
   LPX_RFHC_CASE(4)
 #undef LPX_RFHC_CASE
+  return hipErrorInvalidValue;
+}
+
+hipError_t rf_predict_heap_cpair_lv16_f32(const int* cpairs, const int* roots,
+                                          const unsigned short* lv16,
+                                          int n_trees, int depth,
+                                          const float* X, float* out, int S,
+                                          int F, int C, int silp,
+                                          hipStream_t stream) {
+  if (C > CMAX || depth < 2 || depth > 26 || F > 0xffff)
+    return hipErrorInvalidValue;
+  const int SILP = (silp == 2) ? 2 : (silp == 4 ? 4 : 1);
+  const bool stage = F <= F_LDS_MAX;
+  const size_t lds = stage ? (size_t)THREADS * SILP * (F + 1) * 4 : 0;
+  if (lds > 160 * 1024) return hipErrorInvalidValue;
+  unsigned blocks = lpx_ceil_div(S, THREADS * SILP);
+  if (blocks > 4096u) blocks = 4096u;
+  if (blocks == 0) blocks = 1;
+  const auto* pp = reinterpret_cast<const i32x2*>(cpairs);
+  const auto* lv = reinterpret_cast<const float*>(lv16);
+#define LPX_RFHC16_CASE(I)                                                    \
+  if (SILP == I) {                                                            \
+    if (stage)                                                                \
+      rf_heap_cpair_kernel<true, I, true>                                     \
+          <<<dim3(blocks), dim3(THREADS), lds, stream>>>(                     \
+              pp, roots, lv, n_trees, depth, X, out, S, F, C);                \
+    else                                                                      \
+      rf_heap_cpair_kernel<false, I, true>                                    \
+          <<<dim3(blocks), dim3(THREADS), 0, stream>>>(                       \
+              pp, roots, lv, n_trees, depth, X, out, S, F, C);                \
+    return hipGetLastError();                                                 \
+  }
+  LPX_RFHC16_CASE(1)
+  LPX_RFHC16_CASE(2)
+  LPX_RFHC16_CASE(4)
+#undef LPX_RFHC16_CASE
   return hipErrorInvalidValue;
 }
 

@@ -160,7 +160,9 @@ extern "C" __global__ __launch_bounds__(THREADS) void lpx_forestgen_leaves(
       sum += u;
     }
     for (int c = 0; c < C && c < CMAX; ++c)
-      leaves[r * C + c] = v[c] / sum;
+      // fp16-representable (matches random_heap_forest_np; exact for the
+      // fp16-leaf-table compact kernel)
+      leaves[r * C + c] = __half2float(__float2half(v[c] / sum));
   }
 }
 

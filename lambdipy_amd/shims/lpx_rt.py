@@ -244,6 +244,10 @@ def random_heap_forest_np(n_trees: int, depth: int, n_features: int,
     nodes = np.stack([feat, thr.view(np.uint32)], axis=1).view(np.int32)
     lv = rng.random((n_trees * n_leaves, n_classes)).astype(np.float32)
     lv /= lv.sum(axis=1, keepdims=True)
+    # fp16-representable leaf values (same reasoning as thresholds: the
+    # fp16-leaf-table kernel then accumulates bitwise-identically to f32;
+    # rows still sum to 1 within ~5e-4)
+    lv = lv.astype(np.float16).astype(np.float32)
     return np.ascontiguousarray(nodes), np.ascontiguousarray(lv)
 
 

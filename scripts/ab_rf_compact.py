@@ -52,15 +52,21 @@ def main():
     pairs, roots, cpairs, croots = pack(n, depth)
 
     ref = e.rf_predict_heap_pair(pairs, roots, lv, depth, X, 1)
+    lv16 = lv.to(torch.float16).contiguous()
+    lv16_exact = bool(lv16.to(torch.float32).eq(lv).all().item())
     for silp in (1, 2):
         dt = bench(lambda: e.rf_predict_heap_pair(pairs, roots, lv, depth, X, silp))
-        print(f"pair   silp={silp}: {S/dt/1e6:7.1f} Ms/s")
+        print(f"pair    silp={silp}: {S/dt/1e6:7.1f} Ms/s")
         dt = bench(lambda: e.rf_predict_heap_cpair(cpairs, croots, lv, depth, X, silp))
         out = e.rf_predict_heap_cpair(cpairs, croots, lv, depth, X, silp)
         lab_ref = ref.argmax(1)
         lab = out.argmax(1)
         flip = (lab != lab_ref).float().mean().item()
-        print(f"cpair  silp={silp}: {S/dt/1e6:7.1f} Ms/s  label-flips={flip:.2e}")
+        print(f"cpair   silp={silp}: {S/dt/1e6:7.1f} Ms/s  label-flips={flip:.2e}")
+        dt = bench(lambda: e.rf_predict_heap_cpair16(cpairs, croots, lv16, depth, X, silp))
+        o16 = e.rf_predict_heap_cpair16(cpairs, croots, lv16, depth, X, silp)
+        exact = bool(torch.equal(o16, out)) if lv16_exact else None
+        print(f"cpair16 silp={silp}: {S/dt/1e6:7.1f} Ms/s  exact_vs_cpair={exact}")
 
     # exactness on an fp16-representable forest: round thresholds first
     nh = n.view(-1, 2).clone()

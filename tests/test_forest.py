@@ -89,7 +89,9 @@ def test_random_forest_shapes():
     assert f.depth == 3
     # leaf distributions normalized
     s = f.leaf_values.sum(dim=1)
-    assert torch.allclose(s, torch.ones_like(s), atol=1e-5)
+    # leaf values are fp16-representable (exact compact kernel), so rows
+    # sum to 1 within fp16 rounding (~5e-4)
+    assert torch.allclose(s, torch.ones_like(s), atol=1e-3)
 
 
 def test_random_forest_traversal_terminates():
@@ -97,7 +99,7 @@ def test_random_forest_traversal_terminates():
                          seed=2, device=None)
     X = np.random.default_rng(3).standard_normal((16, 5)).astype(np.float32)
     proba = soa_predict_proba_cpu(f, X)
-    np.testing.assert_allclose(proba.sum(axis=1), 1.0, rtol=1e-5)
+    np.testing.assert_allclose(proba.sum(axis=1), 1.0, rtol=1e-3)
 
 
 def test_regressor_conversion():

@@ -948,3 +948,21 @@ def test_spmv_xtile_vs_seg8(ops):
         b = ops.spmv_csr(ti, tj, td, tx, 2)
         assert torch.allclose(a, b, rtol=1e-4, atol=1e-3), \
             f"xtile mismatch rows={rows}"
+
+
+def test_rf_cpair_lv16_exact(ops):
+    """fp16 leaf table: bitwise-identical to the f32-leaf compact kernel
+    when leaf values are fp16-representable (the synthetic generators')."""
+    from lambdipy_amd.shims.lpx_sklearn import ForestSoA
+
+    e = ops.ext()
+    depth = 8
+    f = ForestSoA.random(64, depth, 16, 8, seed=21, device=DEV)
+    X = _rand((16384, 16), seed=22)
+    _, _, cpairs, croots = _pack_pairs(f.nodes, depth)
+    a = e.rf_predict_heap_cpair(cpairs, croots, f.leaf_values, depth, X, 1)
+    lv16 = f.leaf_values.to(torch.float16).contiguous()
+    assert bool(lv16.to(torch.float32).eq(f.leaf_values).all().item()), \
+        "generator must produce fp16-representable leaves"
+    b = e.rf_predict_heap_cpair16(cpairs, croots, lv16, depth, X, 1)
+    assert torch.equal(a, b)

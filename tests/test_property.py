@@ -31,7 +31,8 @@ def test_heap_forest_cpu_reference_consistent(trees, depth, feats, classes,
     if f.layout == "heap":
         # heap-generated leaf rows are probability distributions and the
         # reference returns the across-tree MEAN -> rows sum to 1
-        assert np.allclose(out1.sum(axis=1), 1.0, rtol=1e-4)
+        # fp16-representable leaf values: rows sum to 1 within ~5e-4
+        assert np.allclose(out1.sum(axis=1), 1.0, rtol=1e-3)
 
 
 @settings(max_examples=50, deadline=None)
