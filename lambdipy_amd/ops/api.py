@@ -155,16 +155,72 @@ def _twiddles(n: int, device: torch.device) -> torch.Tensor:
 
 
 def fft(x: torch.Tensor) -> torch.Tensor:
-    """Batched 1-D FFT over the last axis (complex64, N = power of two)."""
+    """Batched 1-D FFT over the last axis. Power-of-two N runs the LDS
+    Stockham kernel directly; any other N runs Bluestein's chirp-z over the
+    same kernel (scipy.fft parity for arbitrary sizes — SURVEY.md §2.2 N6).
+    """
     x = x.contiguous()
     n = x.shape[-1]
-    return require_extension().fft_c64(x, _twiddles(n, x.device), False)
+    if n & (n - 1) == 0:
+        return require_extension().fft_c64(x, _twiddles(n, x.device), False)
+    return _bluestein(x, inverse=False)
 
 
 def ifft(x: torch.Tensor) -> torch.Tensor:
     x = x.contiguous()
     n = x.shape[-1]
-    return require_extension().fft_c64(x, _twiddles(n, x.device), True)
+    if n & (n - 1) == 0:
+        return require_extension().fft_c64(x, _twiddles(n, x.device), True)
+    return _bluestein(x, inverse=True)
+
+
+_bluestein_cache = {}
+
+
+def _bluestein_plan(n: int, device: torch.device):
+    """(chirp a_k = e^{-i pi k^2 / n}, FFT(b) for the padded conv kernel,
+    m): Bluestein expresses DFT_n as a circular convolution that runs on
+    the power-of-two kernel. Chirp computed in f64 (k^2/n up to ~n, f32
+    phase would lose the FFT's accuracy)."""
+    key = (n, str(device))
+    plan = _bluestein_cache.get(key)
+    if plan is None:
+        m = 1
+        while m < 2 * n - 1:
+            m <<= 1
+        k = torch.arange(n, dtype=torch.float64)
+        ang = math.pi * (k * k % (2 * n)) / n  # mod keeps f64 phase exact
+        a = torch.complex(torch.cos(ang), -torch.sin(ang))  # e^{-i pi k^2/n}
+        b = torch.zeros(m, dtype=torch.complex128)
+        conj = torch.complex(torch.cos(ang), torch.sin(ang))
+        b[:n] = conj
+        b[m - n + 1:] = conj[1:].flip(0)
+        a32 = a.to(torch.complex64).to(device)
+        b32 = b.to(torch.complex64).to(device)
+        e = require_extension()
+        fb = e.fft_c64(b32.unsqueeze(0).contiguous(), _twiddles(m, device),
+                       False)[0]
+        plan = (a32, fb, m)
+        _bluestein_cache[key] = plan
+    return plan
+
+
+def _bluestein(x: torch.Tensor, inverse: bool) -> torch.Tensor:
+    n = x.shape[-1]
+    shape = x.shape
+    x2 = x.reshape(-1, n)
+    if inverse:  # IDFT_n(x) = conj(DFT_n(conj(x))) / n
+        return (_bluestein(x2.conj(), False).conj() / n).reshape(shape)
+    a, fb, m = _bluestein_plan(n, x.device)
+    e = require_extension()
+    tw = _twiddles(m, x.device)
+    ap = torch.zeros(x2.shape[0], m, dtype=torch.complex64, device=x.device)
+    ap[:, :n] = x2 * a
+    # circular convolution via the pow-2 kernel: conv = IFFT(FFT(ap) * fb)
+    fa = e.fft_c64(ap.contiguous(), tw, False)
+    prod = (fa * fb).contiguous()
+    conv = e.fft_c64(prod, tw, True)
+    return (conv[:, :n] * a).reshape(shape)
 
 
 # ---- pairwise ------------------------------------------------------------

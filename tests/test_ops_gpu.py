@@ -966,3 +966,21 @@ def test_rf_cpair_lv16_exact(ops):
         "generator must produce fp16-representable leaves"
     b = e.rf_predict_heap_cpair16(cpairs, croots, lv16, depth, X, 1)
     assert torch.equal(a, b)
+
+
+# ---- non-power-of-two FFT (Bluestein over the pow-2 kernel) ---------------
+
+@pytest.mark.parametrize("batch,n", [(4, 100), (8, 720), (2, 1000), (16, 97),
+                                     (3, 1536), (1, 3), (4, 6000)])
+def test_fft_nonpow2_vs_torch(ops, batch, n):
+    g = torch.Generator(device="cpu").manual_seed(n)
+    x = (torch.randn(batch, n, generator=g) +
+         1j * torch.randn(batch, n, generator=g)).to(torch.complex64).to(DEV)
+    got = ops.fft(x)
+    ref = torch.fft.fft(x.to(torch.complex128))
+    err = (got.to(torch.complex128) - ref).abs().max().item()
+    scale = ref.abs().max().item()
+    assert err / scale < 5e-4, f"n={n}: rel {err/scale}"
+    # inverse round trip
+    rt = ops.ifft(got)
+    assert (rt - x).abs().max().item() / (x.abs().max().item() + 1e-6) < 1e-3
