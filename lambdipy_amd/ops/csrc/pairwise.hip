@@ -127,11 +127,12 @@ __device__ __forceinline__ void stage_f32_128(const float* __restrict__ src,
                                               int rows_total, int K, int row0,
                                               int k0, float* __restrict__ lds) {
   // 128 rows x BK2 f32; 256 threads x (BM2*BK2/4/256) float4 chunks
+  constexpr int CPR = BK2 / 4;  // 16-B chunks per row
   const int t = threadIdx.x;
 #pragma unroll
   for (int i = 0; i < BM2 * BK2 / 4 / THREADS; ++i) {
     const int c = t + i * THREADS;
-    const int r = c >> 2, kb = (c & 3) * 4;
+    const int r = c / CPR, kb = (c % CPR) * 4;
     const int grow = row0 + r, gk = k0 + kb;
     f32x4 v = {0.f, 0.f, 0.f, 0.f};
     if (grow < rows_total) {

@@ -36,7 +36,7 @@ def test_gemm_bf16_floor():
     b = torch.randn(n, n, device=DEV, dtype=torch.bfloat16)
     dt = _time_op(lambda: ops.gemm_bf16(a, b))
     tf = 2 * n**3 / dt / 1e12
-    assert tf > 600, f"bf16 GEMM at {tf:.0f} TF (floor 600; measured ~1100)"
+    assert tf > 800, f"bf16 GEMM at {tf:.0f} TF (floor 800; measured ~1100-1150)"
 
 
 def test_rf_traversal_floor():
@@ -46,7 +46,7 @@ def test_rf_traversal_floor():
     X = torch.randn(1 << 20, 32, device=DEV)
     dt = _time_op(lambda: f.predict_proba_t(X))
     ms = (1 << 20) / dt / 1e6
-    assert ms > 140, f"RF traversal at {ms:.0f} Msamples/s (floor 140; measured 243)"
+    assert ms > 200, f"RF traversal at {ms:.0f} Msamples/s (floor 200; measured 276-279 r2)"
 
 
 def test_gemm_fp8_scaled_floor():
