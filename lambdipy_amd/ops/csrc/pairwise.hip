@@ -192,17 +192,31 @@ __global__ __launch_bounds__(THREADS) void pairwise2_kernel(
       stage_f32_128(Y, N, K, bn, (ks + 1) * BK2,
                     fbase + 2 * BUF + (cur ^ 1) * BUF);
     }
+    // software-pipelined operand regs: read kk+1 while the kk MFMAs run
+    // (the un-pipelined form exposes the ds_read latency every kk)
+    const int xoff0 = (wr * 64 + fr) * ROWF2 + fk;
+    const int xoff1 = (wr * 64 + 32 + fr) * ROWF2 + fk;
+    const int yoff0 = (wc * 64 + fr) * ROWF2 + fk;
+    const int yoff1 = (wc * 64 + 32 + fr) * ROWF2 + fk;
+    float a0 = xCur[xoff0], a1 = xCur[xoff1];
+    float b0 = yCur[yoff0], b1 = yCur[yoff1];
 #pragma unroll
     for (int kk = 0; kk < BK2 / 2; ++kk) {
-      const int kof = kk * 2 + fk;
-      float a0 = xCur[(wr * 64 + fr) * ROWF2 + kof];
-      float a1 = xCur[(wr * 64 + 32 + fr) * ROWF2 + kof];
-      float b0 = yCur[(wc * 64 + fr) * ROWF2 + kof];
-      float b1 = yCur[(wc * 64 + 32 + fr) * ROWF2 + kof];
+      float na0, na1, nb0, nb1;
+      if (kk + 1 < BK2 / 2) {
+        const int kof = (kk + 1) * 2;
+        na0 = xCur[xoff0 + kof];
+        na1 = xCur[xoff1 + kof];
+        nb0 = yCur[yoff0 + kof];
+        nb1 = yCur[yoff1 + kof];
+      }
       acc[0][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b0, acc[0][0], 0, 0, 0);
       acc[0][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a0, b1, acc[0][1], 0, 0, 0);
       acc[1][0] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b0, acc[1][0], 0, 0, 0);
       acc[1][1] = __builtin_amdgcn_mfma_f32_32x32x2f32(a1, b1, acc[1][1], 0, 0, 0);
+      if (kk + 1 < BK2 / 2) {
+        a0 = na0; a1 = na1; b0 = nb0; b1 = nb1;
+      }
     }
     __syncthreads();
     cur ^= 1;
@@ -255,8 +269,11 @@ hipError_t pairwise_sqeuclidean_f32(const float* X, const float* Y,
                                     const float* xsq, const float* ysq,
                                     float* D, int M, int N, int K,
                                     hipStream_t stream) {
-  const char* pv = getenv("LPX_PW_V1");
-  if ((M >= 256 && N >= 256) && !(pv && pv[0] == '1')) {
+  // v2 measured SLOWER than v1 at every tested shape (52-74 vs 64-87 TF,
+  // ab_pairwise.py round 2) despite halving LDS-read pressure per FLOP —
+  // kept opt-in for iteration; v1 (64^2, 16x16x4) stays the default.
+  const char* pv = getenv("LPX_PW_V2");
+  if ((M >= 256 && N >= 256) && (pv && pv[0] == '1')) {
     // v2: 32x32x2 MFMA, 128^2 tile (LDS-read pressure halved per FLOP)
     const unsigned tm = lpx_ceil_div(M, BM2), tn = lpx_ceil_div(N, BM2);
     const size_t lds = sizeof(float) * 4u * BM2 * ROWF2;
