@@ -3,7 +3,7 @@
 # (separate runs — --pmc must not be combined with trace domains on this pool).
 set -x
 cd /tmp && export TMPDIR=/tmp
-OUT=/root/repo/gpurun_out/prof2
+OUT=/root/repo/gpurun_out/prof_r02
 mkdir -p $OUT
 
 timeout 400 rocprofv3 --kernel-trace --stats --output-format csv -d $OUT -o trace -- \
@@ -19,5 +19,8 @@ echo "pmc rc=$?" >> $OUT/pmc.log
 
 ls -la $OUT
 # summarize the stats csv (top kernels)
-for f in $OUT/*stats*.csv; do echo "== $f =="; head -25 "$f"; done
+for f in $OUT/*stats*.csv; do echo "== $f =="; head -30 "$f"; done
+for f in $OUT/*counter_collection*.csv; do
+  python /root/repo/scripts/pmc_summarize.py "$f" $OUT/pmc_summary_r02.json &&     cat $OUT/pmc_summary_r02.json
+done
 tail -5 $OUT/trace.log $OUT/pmc.log
