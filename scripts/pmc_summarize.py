@@ -17,9 +17,14 @@ def main(path, out):
             ctr = row.get("Counter_Name") or row.get("Counter Name") or ""
             val = float(row.get("Counter_Value") or row.get("Counter Value")
                         or 0)
-            m = re.search(r"(\w+)\(", name)
-            short = m.group(1) if m else name.split("(")[0].split("<")[0]
-            short = short.split("::")[-1]
+            n = name.replace("void ", "").replace("(anonymous namespace)::",
+                                                  "")
+            cut = len(n)
+            for ch in "<(":
+                i = n.find(ch)
+                if i != -1:
+                    cut = min(cut, i)
+            short = n[:cut].split("::")[-1].strip()
             per[short][ctr] += val
     out_d = {}
     for k, c in sorted(per.items()):
