@@ -56,6 +56,15 @@ def _install_handlers(build_dir: str, handlers):
 
         lib = ensure_nn_lib()
         shutil.copy2(lib, os.path.join(build_dir, os.path.basename(lib)))
+    # rf_lite hipModuleLoads the standalone .hsaco — same rule (a handlers-
+    # only layer must be self-contained; real recipe layers get it via the
+    # lpx_ops wheel)
+    if any(h in ("rf_lite", "rf") for h in handlers) and not os.path.exists(
+            os.path.join(build_dir, "lpx_ops", "standalone_gfx950.hsaco")):
+        from .builder.wheel_builder import ensure_standalone_hsaco
+
+        hs = ensure_standalone_hsaco()
+        shutil.copy2(hs, os.path.join(build_dir, os.path.basename(hs)))
 
 
 @click.group()

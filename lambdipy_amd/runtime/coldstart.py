@@ -204,6 +204,7 @@ def cold_start(
                 return (n.startswith("lpx_handlers/") or n == "lpx_fanout.py"
                         or n == "lpx_rt.py" or n == "lpx_nn.py"
                         or base.startswith("liblpx_nn")
+                        or base.endswith(".hsaco")
                         or (n.startswith("lpx_ops/")
                             and not base.startswith("_lpx_ops")))
         else:

@@ -54,6 +54,11 @@ def test_install_handlers_ships_nn_runtime(tmp_path):
     assert os.path.exists(os.path.join(d, "lpx_nn.py"))
     assert os.path.exists(os.path.join(d, "lpx_rt.py"))
     assert os.path.exists(os.path.join(d, "liblpx_nn_gfx950.so"))
+    # a handlers-only rf_lite layer must be self-contained too
+    d2 = str(tmp_path / "layer2")
+    os.makedirs(d2)
+    _install_handlers(d2, ["rf_lite"])
+    assert os.path.exists(os.path.join(d2, "standalone_gfx950.hsaco"))
     assert os.path.exists(
         os.path.join(d, "lpx_handlers", "resnet50_lite_handler.py"))
 
