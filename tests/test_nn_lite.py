@@ -85,9 +85,11 @@ def test_early_extract_covers_nn_files():
         return (n.startswith("lpx_handlers/") or n == "lpx_fanout.py"
                 or n == "lpx_rt.py" or n == "lpx_nn.py"
                 or base.startswith("liblpx_nn")
+                or base.endswith(".hsaco")
                 or (n.startswith("lpx_ops/")
                     and not base.startswith("_lpx_ops")))
 
+    assert early("standalone_gfx950.hsaco")
     assert early("lpx_nn.py")
     assert early("liblpx_nn_gfx950.so")
     assert early("lpx_ops/liblpx_nn_gfx950.so")
