@@ -85,6 +85,149 @@ __device__ __forceinline__ const bf16frag* frag_ptr(const char* __restrict__ sme
       smem + (size_t)parity * PARITY_BYTES + (size_t)hh * HALF_BYTES + q * 16);
 }
 
+// ---- BK=32 half-LDS variant (tune 17) ------------------------------------
+// Same 4-phase quadrant schedule with 32-deep K-steps: half-tiles are 8 KiB
+// (1 glds/thread), LDS drops 128->64 KiB so the kernel runs 2 blocks/CU
+// (4 waves/SIMD) — the extra block interleaves across the other's barriers,
+// trading per-phase MFMA count (8 vs 16) for occupancy-hidden latency.
+constexpr int BK32 = 32;
+constexpr int HALF32 = 128 * 64;            // 128 rows x 64 B
+constexpr int PAR32 = 4 * HALF32;           // A-lo A-hi B-lo B-hi
+
+__device__ __forceinline__ int phys_cell32(int row, int cell) {
+  return row * 4 + ((cell + row) & 3);      // 4 16-B cells per 64-B row
+}
+__device__ __forceinline__ int src_cell32(int q) {
+  const int row = q >> 2;
+  return ((q & 3) - row) & 3;
+}
+
+__device__ __forceinline__ void stage_half32(const uint16_t* __restrict__ A,
+                                             const uint16_t* __restrict__ Bt,
+                                             long K, int bm, int bn, int tile,
+                                             int hh, int ksteps,
+                                             char* __restrict__ smem, int w,
+                                             int lane) {
+  const int tc = tile < ksteps ? tile : ksteps - 1;  // clamp (never read)
+  const long k0 = (long)tc * BK32;
+  const uint16_t* src = (hh < 2) ? A : Bt;
+  const int row0 = ((hh < 2) ? bm : bn) + (hh & 1) * 128;
+  char* const base = smem + (size_t)(tile & 1) * PAR32 +
+                     (size_t)hh * HALF32;
+  const int q = w * 64 + lane;               // one 16-B cell per thread
+  const int r = q >> 2, cell = src_cell32(q);
+  const uint16_t* g = src + (long)(row0 + r) * K + (k0 + cell * 8);
+  __builtin_amdgcn_global_load_lds(
+      (const __attribute__((address_space(1))) void*)g,
+      (__attribute__((address_space(3))) void*)(base + q * 16), 16, 0, 0);
+}
+
+__device__ __forceinline__ const bf16frag* frag_ptr32(
+    const char* __restrict__ smem, int parity, int hh, int row_in_half,
+    int cell) {
+  const int q = phys_cell32(row_in_half, cell);
+  return reinterpret_cast<const bf16frag*>(
+      smem + (size_t)parity * PAR32 + (size_t)hh * HALF32 + q * 16);
+}
+
+template <bool BF16_OUT>
+__global__ __launch_bounds__(THREADS8, 2) void gemm_bf16_8ph32_kernel(
+    const uint16_t* __restrict__ A, const uint16_t* __restrict__ Bt,
+    void* __restrict__ Cout, int M, int N, int K, unsigned tiles_m,
+    unsigned tiles_n) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+
+  const unsigned nwg = tiles_m * tiles_n;
+  const unsigned bid = lpx_xcd_swizzle(blockIdx.x, nwg);
+  const int bm = (int)(bid / tiles_n) * BM8;
+  const int bn = (int)(bid % tiles_n) * BM8;
+
+  const int lane = threadIdx.x & 63;
+  const int w = threadIdx.x >> 6;
+  const int wsr = w >> 2, wsc = w & 3;
+  const int fr = lane & 15;
+  const int fkb = lane >> 4;
+
+  f32x4 acc[4][4][2];
+#pragma unroll
+  for (int q = 0; q < 4; ++q)
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 2; ++j) acc[q][i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int ksteps = (K + BK32 - 1) / BK32;
+
+  // prologue mirrors the BK64 schedule (tile0 all 4 halves + tile1 lo pair)
+  stage_half32(A, Bt, K, bm, bn, 0, 0, ksteps, smem, w, lane);
+  stage_half32(A, Bt, K, bm, bn, 0, 2, ksteps, smem, w, lane);
+  stage_half32(A, Bt, K, bm, bn, 0, 3, ksteps, smem, w, lane);
+  stage_half32(A, Bt, K, bm, bn, 0, 1, ksteps, smem, w, lane);
+  stage_half32(A, Bt, K, bm, bn, 1, 0, ksteps, smem, w, lane);
+  stage_half32(A, Bt, K, bm, bn, 1, 2, ksteps, smem, w, lane);
+  asm volatile("s_waitcnt vmcnt(2)" ::: "memory");  // tile0 A-lo/B-lo landed
+  __builtin_amdgcn_s_barrier();
+
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);
+
+  for (int t = 0; t < ksteps; ++t) {
+    const int parity = t & 1;
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      const int rh = (p >> 1) & 1;
+      const int ch = p & 1;
+      bf16frag aF[4], bF[2];
+      const int cell = fkb;                  // 4 cells = the full 32-k step
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf)
+        aF[mf] = *frag_ptr32(smem, parity, rh, wsr * 64 + mf * 16 + fr, cell);
+#pragma unroll
+      for (int nf = 0; nf < 2; ++nf)
+        bF[nf] = *frag_ptr32(smem, parity, 2 + ch, wsc * 32 + nf * 16 + fr,
+                             cell);
+      int st, sh;
+      stage_slot(t, p, &st, &sh);
+      stage_half32(A, Bt, K, bm, bn, st, sh, ksteps, smem, w, lane);
+      __builtin_amdgcn_s_barrier();
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 2; ++nf)
+          acc[p][mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              aF[mf], bF[nf], acc[p][mf][nf], 0, 0, 0);
+      if (p & 1) asm volatile("s_waitcnt vmcnt(3)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+
+  const int orow_l = (lane >> 4) * 4;
+  const int ocol_l = lane & 15;
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    const int rh = (p >> 1) & 1, ch = p & 1;
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf) {
+#pragma unroll
+      for (int nf = 0; nf < 2; ++nf) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int row = bm + rh * 128 + wsr * 64 + mf * 16 + orow_l + r;
+          const int col = bn + ch * 128 + wsc * 32 + nf * 16 + ocol_l;
+          if constexpr (BF16_OUT)
+            reinterpret_cast<uint16_t*>(Cout)[(long)row * N + col] =
+                (uint16_t)lpx_f32_to_bf16(acc[p][mf][nf][r]);
+          else
+            reinterpret_cast<float*>(Cout)[(long)row * N + col] =
+                acc[p][mf][nf][r];
+        }
+      }
+    }
+  }
+}
+
 template <bool BF16_OUT, int VM, bool LG0, int PRIO, int ILV = 0,
           int THR = -1, bool ASM = false>
 __global__ __launch_bounds__(THREADS8, 2) void gemm_bf16_8ph_kernel(
@@ -345,6 +488,15 @@ hipError_t gemm_bf16_nt8ph_tune(const uint16_t* A, const uint16_t* Bt,
     // hand-asm phase body: counted descending lgkmcnt between ds_reads
     // and MFMAs (the compiler's schedule drains lgkmcnt(0) per group)
     case 16: return launch8ph<false, 60, false, 1, 1, -1, true>(A, Bt, C, M, N, K, stream);
+    // BK=32 half-LDS (64 KiB -> 2 blocks/CU) variant
+    case 17: {
+      if (K % BK32 != 0) return hipErrorInvalidValue;
+      const unsigned tm = (unsigned)(M / BM8), tn = (unsigned)(N / BM8);
+      gemm_bf16_8ph32_kernel<false>
+          <<<dim3(tm * tn), dim3(THREADS8), 2 * PAR32, stream>>>(
+              A, Bt, C, M, N, K, tm, tn);
+      return hipGetLastError();
+    }
     default: return launch8ph<false, 6, true>(A, Bt, C, M, N, K, stream);
   }
 }
