@@ -182,3 +182,17 @@ def test_pandas_layer_assembles(tmp_path):
     assert os.path.isdir(os.path.join(d, "pandas"))
     assert not os.path.isdir(os.path.join(d, "pandas", "tests"))
     assert any(x["name"] == "pandas" for x in report.deps)
+
+
+def test_grpcio_recipe(tmp_path):
+    from lambdipy_amd.packager import assemble_layer
+    from lambdipy_amd.recipes import get_registry
+    from lambdipy_amd.resolver import parse_requirements, resolve
+
+    r = get_registry().get("grpcio")
+    assert r is not None and r.kind == "native" and r.strip_native
+    res = resolve(parse_requirements("grpcio\n"), follow_transitive=False)
+    assert any(d.recipe == "grpcio" for d in res.recipe_hits)
+    d = str(tmp_path / "build")
+    assemble_layer(res, d)
+    assert os.path.isdir(os.path.join(d, "grpc"))
