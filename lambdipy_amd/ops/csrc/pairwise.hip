@@ -242,6 +242,125 @@ __global__ __launch_bounds__(THREADS) void pairwise2_kernel(
   }
 }
 
+// ---- v3: 256x256 tile, 512 threads, 32x32x2 MFMA -------------------------
+// v1 is HBM-staging-bound (64^2 tile re-reads X/Y 128x at 8192^2: 2 GB);
+// the 256^2 tile cuts staging 4x (0.5 GB -> ~80 us floor at 8192^2 x 256)
+// and pushes the kernel into the MFMA-bound regime. 8 waves (2x4), each
+// owning 128x64 (4x2 fragments of 32x32); BK=32 double-buffered = ~135 KiB
+// LDS -> 1 block/CU (2 waves/SIMD, same occupancy class as the GEMM
+// pipeline). Same exact-f32 numerics.
+constexpr int BM3 = 256, BK3 = 32;
+constexpr int ROWF3 = BK3 + 1;
+constexpr int THREADS3 = 512;
+
+__device__ __forceinline__ void stage_f32_256(const float* __restrict__ src,
+                                              int rows_total, int K, int row0,
+                                              int k0, float* __restrict__ lds) {
+  constexpr int CPR = BK3 / 4;  // 16-B chunks per row
+  const int t = threadIdx.x;
+#pragma unroll
+  for (int i = 0; i < BM3 * BK3 / 4 / THREADS3; ++i) {
+    const int c = t + i * THREADS3;
+    const int r = c / CPR, kb = (c % CPR) * 4;
+    const int grow = row0 + r, gk = k0 + kb;
+    f32x4 v = {0.f, 0.f, 0.f, 0.f};
+    if (grow < rows_total) {
+      if (gk + 4 <= K) {
+        v = *reinterpret_cast<const f32x4*>(src + (long)grow * K + gk);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          if (gk + j < K) v[j] = src[(long)grow * K + gk + j];
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < 4; ++j) lds[r * ROWF3 + kb + j] = v[j];
+  }
+}
+
+__global__ __launch_bounds__(THREADS3) void pairwise3_kernel(
+    const float* __restrict__ X, const float* __restrict__ Y,
+    const float* __restrict__ xsq, const float* __restrict__ ysq,
+    float* __restrict__ D, int M, int N, int K, unsigned tiles_m,
+    unsigned tiles_n) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* const fbase = reinterpret_cast<float*>(smem);
+  const int BUF = BM3 * ROWF3;  // floats
+
+  const unsigned nwg = tiles_m * tiles_n;
+  const unsigned bid = lpx_xcd_swizzle(blockIdx.x, nwg);
+  const int bm = (int)(bid / tiles_n) * BM3;
+  const int bn = (int)(bid % tiles_n) * BM3;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wr = wid >> 2, wc = wid & 3;  // 2x4 waves; 128x64 per wave
+
+  f32x16 acc[4][2];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int e = 0; e < 16; ++e) acc[i][j][e] = 0.f;
+
+  const int fr = lane & 31;
+  const int fk = lane >> 5;
+
+  int cur = 0;
+  stage_f32_256(X, M, K, bm, 0, fbase);
+  stage_f32_256(Y, N, K, bn, 0, fbase + 2 * BUF);
+  __syncthreads();
+
+  const int ksteps = (K + BK3 - 1) / BK3;
+  for (int ks = 0; ks < ksteps; ++ks) {
+    float* const xCur = fbase + cur * BUF;
+    float* const yCur = fbase + 2 * BUF + cur * BUF;
+    if (ks + 1 < ksteps) {
+      stage_f32_256(X, M, K, bm, (ks + 1) * BK3, fbase + (cur ^ 1) * BUF);
+      stage_f32_256(Y, N, K, bn, (ks + 1) * BK3,
+                    fbase + 2 * BUF + (cur ^ 1) * BUF);
+    }
+    const int xo0 = (wr * 128 + fr) * ROWF3 + fk;
+    const int yo0 = (wc * 64 + fr) * ROWF3 + fk;
+#pragma unroll
+    for (int kk = 0; kk < BK3 / 2; ++kk) {
+      const int kof = kk * 2;
+      float a[4], b[2];
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf) a[mf] = xCur[xo0 + mf * 32 * ROWF3 + kof];
+#pragma unroll
+      for (int nf = 0; nf < 2; ++nf) b[nf] = yCur[yo0 + nf * 32 * ROWF3 + kof];
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 2; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_32x32x2f32(
+              a[mf], b[nf], acc[mf][nf], 0, 0, 0);
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  const int ocol_l = lane & 31;
+  const int orow_base = 4 * (lane >> 5);
+#pragma unroll
+  for (int mf = 0; mf < 4; ++mf) {
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int row = bm + wr * 128 + mf * 32 + orow_base + (r & 3) + 8 * (r >> 2);
+        const int col = bn + wc * 64 + nf * 32 + ocol_l;
+        if (row < M && col < N) {
+          float d = xsq[row] - 2.0f * acc[mf][nf][r] + ysq[col];
+          D[(long)row * N + col] = d > 0.f ? d : 0.f;
+        }
+      }
+    }
+  }
+}
+
 __global__ __launch_bounds__(THREADS) void row_sqnorm_kernel(
     const float* __restrict__ X, float* __restrict__ out, int M, long K) {
   const int lane = threadIdx.x & 63;
@@ -278,6 +397,14 @@ hipError_t pairwise_sqeuclidean_f32(const float* X, const float* Y,
     const unsigned tm = lpx_ceil_div(M, BM2), tn = lpx_ceil_div(N, BM2);
     const size_t lds = sizeof(float) * 4u * BM2 * ROWF2;
     pairwise2_kernel<<<dim3(tm * tn), dim3(THREADS), lds, stream>>>(
+        X, Y, xsq, ysq, D, M, N, K, tm, tn);
+    return hipGetLastError();
+  }
+  if ((M >= 256 && N >= 256) && (pv && pv[0] == '3')) {
+    // v3: 256^2 tile, 8 waves (staging cut 4x vs v1 -> MFMA-bound regime)
+    const unsigned tm = lpx_ceil_div(M, BM3), tn = lpx_ceil_div(N, BM3);
+    const size_t lds = sizeof(float) * 4u * BM3 * ROWF3;
+    pairwise3_kernel<<<dim3(tm * tn), dim3(THREADS3), lds, stream>>>(
         X, Y, xsq, ysq, D, M, N, K, tm, tn);
     return hipGetLastError();
   }

@@ -27,11 +27,11 @@ def main():
         X = torch.randn(m, k, device="cuda")
         Y = torch.randn(n, k, device="cuda")
         ref = None
-        for env, name in ((None, "v2"), ("1", "v1")):
+        for env, name in ((None, "v1"), ("1", "v2"), ("3", "v3")):
             if env is None:
-                os.environ.pop("LPX_PW_V1", None)
+                os.environ.pop("LPX_PW_V2", None)
             else:
-                os.environ["LPX_PW_V1"] = env
+                os.environ["LPX_PW_V2"] = env
             D = ops.pairwise_sqeuclidean(X, Y)
             if ref is None:
                 ref = torch.cdist(X.double(), Y.double()) ** 2
@@ -40,7 +40,7 @@ def main():
             dt = timeit(lambda: ops.pairwise_sqeuclidean(X, Y))
             tf = 2.0 * m * n * k / dt / 1e12
             print(f"{name} m={m} n={n} k={k}: {dt*1e3:7.3f} ms {tf:6.1f} TF ok={ok}")
-        os.environ.pop("LPX_PW_V1", None)
+        os.environ.pop("LPX_PW_V2", None)
 
 
 if __name__ == "__main__":
