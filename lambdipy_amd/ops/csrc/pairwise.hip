@@ -361,6 +361,122 @@ __global__ __launch_bounds__(THREADS3) void pairwise3_kernel(
   }
 }
 
+// ---- v4: 256x256 tile, 512 threads, 16x16x4 MFMA -------------------------
+// v3's geometry with v1's MFMA: the 128x64 wave tile amortizes operand
+// reads over 8x4 fragments (32 MFMAs per 12 scalar reads = 2.7/read vs
+// v1's 1.0) and BK=16 keeps LDS at ~70 KiB -> 2 blocks/CU (4 waves/SIMD,
+// double v3's latency-hiding). Same exact-f32 numerics.
+constexpr int BM4 = 256, BK4 = 16;
+constexpr int ROWF4 = BK4 + 1;
+
+__device__ __forceinline__ void stage_f32_256b(const float* __restrict__ src,
+                                               int rows_total, int K,
+                                               int row0, int k0,
+                                               float* __restrict__ lds) {
+  constexpr int CPR = BK4 / 4;
+  const int t = threadIdx.x;
+#pragma unroll
+  for (int i = 0; i < BM4 * BK4 / 4 / THREADS3; ++i) {
+    const int c = t + i * THREADS3;
+    const int r = c / CPR, kb = (c % CPR) * 4;
+    const int grow = row0 + r, gk = k0 + kb;
+    f32x4 v = {0.f, 0.f, 0.f, 0.f};
+    if (grow < rows_total) {
+      if (gk + 4 <= K) {
+        v = *reinterpret_cast<const f32x4*>(src + (long)grow * K + gk);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          if (gk + j < K) v[j] = src[(long)grow * K + gk + j];
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < 4; ++j) lds[r * ROWF4 + kb + j] = v[j];
+  }
+}
+
+__global__ __launch_bounds__(THREADS3) void pairwise4_kernel(
+    const float* __restrict__ X, const float* __restrict__ Y,
+    const float* __restrict__ xsq, const float* __restrict__ ysq,
+    float* __restrict__ D, int M, int N, int K, unsigned tiles_m,
+    unsigned tiles_n) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* const fbase = reinterpret_cast<float*>(smem);
+  const int BUF = BM4 * ROWF4;
+
+  const unsigned nwg = tiles_m * tiles_n;
+  const unsigned bid = lpx_xcd_swizzle(blockIdx.x, nwg);
+  const int bm = (int)(bid / tiles_n) * BM4;
+  const int bn = (int)(bid % tiles_n) * BM4;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wr = wid >> 2, wc = wid & 3;  // 2x4 waves; 128x64 per wave
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int fr = lane & 15, fk = lane >> 4;  // fragment row, k-lane (0..3)
+
+  int cur = 0;
+  stage_f32_256b(X, M, K, bm, 0, fbase);
+  stage_f32_256b(Y, N, K, bn, 0, fbase + 2 * BUF);
+  __syncthreads();
+
+  const int ksteps = (K + BK4 - 1) / BK4;
+  for (int ks = 0; ks < ksteps; ++ks) {
+    float* const xCur = fbase + cur * BUF;
+    float* const yCur = fbase + 2 * BUF + cur * BUF;
+    if (ks + 1 < ksteps) {
+      stage_f32_256b(X, M, K, bm, (ks + 1) * BK4, fbase + (cur ^ 1) * BUF);
+      stage_f32_256b(Y, N, K, bn, (ks + 1) * BK4,
+                     fbase + 2 * BUF + (cur ^ 1) * BUF);
+    }
+    const int xo0 = (wr * 128 + fr) * ROWF4 + fk;
+    const int yo0 = (wc * 64 + fr) * ROWF4 + fk;
+#pragma unroll
+    for (int kk = 0; kk < BK4 / 4; ++kk) {
+      const int kof = kk * 4;
+      float a[8], b[4];
+#pragma unroll
+      for (int mf = 0; mf < 8; ++mf)
+        a[mf] = xCur[xo0 + mf * 16 * ROWF4 + kof];
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf)
+        b[nf] = yCur[yo0 + nf * 16 * ROWF4 + kof];
+#pragma unroll
+      for (int mf = 0; mf < 8; ++mf)
+#pragma unroll
+        for (int nf = 0; nf < 4; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x4f32(
+              a[mf], b[nf], acc[mf][nf], 0, 0, 0);
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  const int orow_l = (lane >> 4) * 4;
+  const int ocol_l = lane & 15;
+#pragma unroll
+  for (int mf = 0; mf < 8; ++mf) {
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = bm + wr * 128 + mf * 16 + orow_l + r;
+        const int col = bn + wc * 64 + nf * 16 + ocol_l;
+        if (row < M && col < N) {
+          float d = xsq[row] - 2.0f * acc[mf][nf][r] + ysq[col];
+          D[(long)row * N + col] = d > 0.f ? d : 0.f;
+        }
+      }
+    }
+  }
+}
+
 __global__ __launch_bounds__(THREADS) void row_sqnorm_kernel(
     const float* __restrict__ X, float* __restrict__ out, int M, long K) {
   const int lane = threadIdx.x & 63;
@@ -405,6 +521,14 @@ hipError_t pairwise_sqeuclidean_f32(const float* X, const float* Y,
     const unsigned tm = lpx_ceil_div(M, BM3), tn = lpx_ceil_div(N, BM3);
     const size_t lds = sizeof(float) * 4u * BM3 * ROWF3;
     pairwise3_kernel<<<dim3(tm * tn), dim3(THREADS3), lds, stream>>>(
+        X, Y, xsq, ysq, D, M, N, K, tm, tn);
+    return hipGetLastError();
+  }
+  if ((M >= 256 && N >= 256) && (pv && pv[0] == '4')) {
+    // v4: 256^2 tile, 16x16x4, 2 blocks/CU
+    const unsigned tm = lpx_ceil_div(M, BM4), tn = lpx_ceil_div(N, BM4);
+    const size_t lds = sizeof(float) * 4u * BM4 * ROWF4;
+    pairwise4_kernel<<<dim3(tm * tn), dim3(THREADS3), lds, stream>>>(
         X, Y, xsq, ysq, D, M, N, K, tm, tn);
     return hipGetLastError();
   }

@@ -27,7 +27,7 @@ def main():
         X = torch.randn(m, k, device="cuda")
         Y = torch.randn(n, k, device="cuda")
         ref = None
-        for env, name in ((None, "v1"), ("1", "v2"), ("3", "v3")):
+        for env, name in ((None, "v1"), ("3", "v3"), ("4", "v4")):
             if env is None:
                 os.environ.pop("LPX_PW_V2", None)
             else:
