@@ -504,10 +504,14 @@ hipError_t pairwise_sqeuclidean_f32(const float* X, const float* Y,
                                     const float* xsq, const float* ysq,
                                     float* D, int M, int N, int K,
                                     hipStream_t stream) {
-  // v2 measured SLOWER than v1 at every tested shape (52-74 vs 64-87 TF,
-  // ab_pairwise.py round 2) despite halving LDS-read pressure per FLOP —
-  // kept opt-in for iteration; v1 (64^2, 16x16x4) stays the default.
+  // Variant ladder (ab_pairwise.py, r2 within-probe): the kernel family is
+  // ds_read/MFMA-latency-bound, not HBM-bound — big tiles cut staging 4x
+  // yet only win where the K loop is long enough to amortize their lower
+  // occupancy. Measured: v1 63-82 TF, v2 52-74, v3 51-85, v4 52-87.
+  // Dispatch: v4 (256^2, 16x16x4, 2 blocks/CU) for long-K large shapes,
+  // v1 (64^2, 4 blocks/CU) otherwise; LPX_PW_V2=1/3/4 force a variant.
   const char* pv = getenv("LPX_PW_V2");
+  if (!pv && K >= 192 && M >= 1024 && N >= 1024) pv = "4";
   if ((M >= 256 && N >= 256) && (pv && pv[0] == '1')) {
     // v2: 32x32x2 MFMA, 128^2 tile (LDS-read pressure halved per FLOP)
     const unsigned tm = lpx_ceil_div(M, BM2), tn = lpx_ceil_div(N, BM2);

@@ -466,7 +466,7 @@ __device__ __forceinline__ float lpx_h2f(unsigned short u) {
   return __half2float(__half(r));
 }
 
-template <bool STAGE_X, int SILP, bool LV16 = false>
+template <bool STAGE_X, int SILP, bool LV16 = false, int TILP = 1>
 __global__ __launch_bounds__(THREADS) void rf_heap_cpair_kernel(
     const i32x2* __restrict__ pairs, const int* __restrict__ roots,
     const float* __restrict__ leaf_values, int n_trees, int depth,
@@ -509,48 +509,72 @@ __global__ __launch_bounds__(THREADS) void rf_heap_cpair_kernel(
 #pragma unroll
       for (int c = 0; c < CMAX; ++c) acc[j][c] = 0.f;
 
-    for (int tr = 0; tr < n_trees; ++tr) {
-      const i32x2* tp = pairs + (long)tr * ppt;
-      const unsigned root = (unsigned)roots[tr];
-      int nd[SILP];        // current heap index
-      unsigned cf[SILP];   // current node feature
-      float ct[SILP];      // current node threshold (f32 of the f16 bits)
+    // TILP independent tree chains per lane fill the fetch latency the
+    // serial per-tree walk exposes (the kernel is latency-bound: PMC
+    // wait_frac 0.85 with ONE outstanding request per lane). No extra LDS
+    // (same sample row serves every chain), unlike sample-ILP.
+    const int tr_main = n_trees - (n_trees % TILP);
+    for (int tr = 0; tr < n_trees; tr += (tr < tr_main ? TILP : 1)) {
+      const int nt = (tr < tr_main) ? TILP : 1;
+      int nd[SILP][TILP];
+      unsigned cf[SILP][TILP];
+      float ct[SILP][TILP];
 #pragma unroll
-      for (int j = 0; j < SILP; ++j) {
-        nd[j] = 0;
-        cf[j] = root >> 16;
-        ct[j] = lpx_h2f((unsigned short)(root & 0xffffu));
+      for (int ti = 0; ti < TILP; ++ti) {
+        const unsigned root =
+            (unsigned)roots[tr + (ti < nt ? ti : 0)];
+#pragma unroll
+        for (int j = 0; j < SILP; ++j) {
+          nd[j][ti] = 0;
+          cf[j][ti] = root >> 16;
+          ct[j][ti] = lpx_h2f((unsigned short)(root & 0xffffu));
+        }
       }
       for (int d = 0; d < depth - 1; ++d) {
 #pragma unroll
         for (int j = 0; j < SILP; ++j) {
           if (s[j] < S) {
-            const i32x2 pr = tp[nd[j]];  // compact children of current node
-            const int go = (xrow[j][cf[j]] <= ct[j]) ? 0 : 1;
-            nd[j] = 2 * nd[j] + 1 + go;
-            const unsigned c = (unsigned)(go ? pr.y : pr.x);
-            cf[j] = c >> 16;
-            ct[j] = lpx_h2f((unsigned short)(c & 0xffffu));
+            i32x2 pr[TILP];
+#pragma unroll
+            for (int ti = 0; ti < TILP; ++ti)
+              if (ti < nt)
+                pr[ti] = pairs[(long)(tr + ti) * ppt + nd[j][ti]];
+#pragma unroll
+            for (int ti = 0; ti < TILP; ++ti) {
+              if (ti < nt) {
+                const int go = (xrow[j][cf[j][ti]] <= ct[j][ti]) ? 0 : 1;
+                nd[j][ti] = 2 * nd[j][ti] + 1 + go;
+                const unsigned c = (unsigned)(go ? pr[ti].y : pr[ti].x);
+                cf[j][ti] = c >> 16;
+                ct[j][ti] = lpx_h2f((unsigned short)(c & 0xffffu));
+              }
+            }
           }
         }
       }
 #pragma unroll
       for (int j = 0; j < SILP; ++j) {
         if (s[j] < S) {
-          const int go = (xrow[j][cf[j]] <= ct[j]) ? 0 : 1;
-          const int leaf = 2 * nd[j] + 1 + go - tree_stride;
-          if constexpr (LV16) {
-            const unsigned short* lv16 =
-                reinterpret_cast<const unsigned short*>(leaf_values) +
-                ((long)tr * n_leaves + leaf) * C;
 #pragma unroll
-            for (int c = 0; c < CMAX; ++c)
-              if (c < C) acc[j][c] += lpx_h2f(lv16[c]);
-          } else {
-            const float* lv = leaf_values + ((long)tr * n_leaves + leaf) * C;
+          for (int ti = 0; ti < TILP; ++ti) {
+            if (ti < nt) {
+              const int go = (xrow[j][cf[j][ti]] <= ct[j][ti]) ? 0 : 1;
+              const int leaf = 2 * nd[j][ti] + 1 + go - tree_stride;
+              if constexpr (LV16) {
+                const unsigned short* lv16 =
+                    reinterpret_cast<const unsigned short*>(leaf_values) +
+                    ((long)(tr + ti) * n_leaves + leaf) * C;
 #pragma unroll
-            for (int c = 0; c < CMAX; ++c)
-              if (c < C) acc[j][c] += lv[c];
+                for (int c = 0; c < CMAX; ++c)
+                  if (c < C) acc[j][c] += lpx_h2f(lv16[c]);
+              } else {
+                const float* lv =
+                    leaf_values + ((long)(tr + ti) * n_leaves + leaf) * C;
+#pragma unroll
+                for (int c = 0; c < CMAX; ++c)
+                  if (c < C) acc[j][c] += lv[c];
+              }
+            }
           }
         }
       }
@@ -854,6 +878,8 @@ hipError_t rf_predict_heap_cpair_f32(const int* cpairs, const int* roots,
                                      hipStream_t stream) {
   if (C > CMAX || depth < 2 || depth > 26 || F > 0xffff)
     return hipErrorInvalidValue;
+  // silp 21/41 encode tree-ILP 2/4 (SILP stays 1)
+  const int TILP = (silp == 21) ? 2 : (silp == 41 ? 4 : 1);
   const int SILP = (silp == 2) ? 2 : (silp == 4 ? 4 : 1);
   const bool stage = F <= F_LDS_MAX;
   const size_t lds = stage ? (size_t)THREADS * SILP * (F + 1) * 4 : 0;
@@ -862,21 +888,23 @@ hipError_t rf_predict_heap_cpair_f32(const int* cpairs, const int* roots,
   if (blocks > 4096u) blocks = 4096u;
   if (blocks == 0) blocks = 1;
   const auto* pp = reinterpret_cast<const i32x2*>(cpairs);
-#define LPX_RFHC_CASE(I)                                                      \
-  if (SILP == I) {                                                            \
+#define LPX_RFHC_CASE(I, T)                                                   \
+  if (SILP == I && TILP == T) {                                               \
     if (stage)                                                                \
-      rf_heap_cpair_kernel<true, I>                                           \
+      rf_heap_cpair_kernel<true, I, false, T>                                 \
           <<<dim3(blocks), dim3(THREADS), lds, stream>>>(                     \
               pp, roots, leaf_values, n_trees, depth, X, out, S, F, C);       \
     else                                                                      \
-      rf_heap_cpair_kernel<false, I>                                          \
+      rf_heap_cpair_kernel<false, I, false, T>                                \
           <<<dim3(blocks), dim3(THREADS), 0, stream>>>(                       \
               pp, roots, leaf_values, n_trees, depth, X, out, S, F, C);       \
     return hipGetLastError();                                                 \
   }
-  LPX_RFHC_CASE(1)
-  LPX_RFHC_CASE(2)
-  LPX_RFHC_CASE(4)
+  LPX_RFHC_CASE(1, 1)
+  LPX_RFHC_CASE(2, 1)
+  LPX_RFHC_CASE(4, 1)
+  LPX_RFHC_CASE(1, 2)
+  LPX_RFHC_CASE(1, 4)
 #undef LPX_RFHC_CASE
   return hipErrorInvalidValue;
 }
@@ -889,6 +917,8 @@ hipError_t rf_predict_heap_cpair_lv16_f32(const int* cpairs, const int* roots,
                                           hipStream_t stream) {
   if (C > CMAX || depth < 2 || depth > 26 || F > 0xffff)
     return hipErrorInvalidValue;
+  // silp 21/41 encode tree-ILP 2/4 (SILP stays 1)
+  const int TILP = (silp == 21) ? 2 : (silp == 41 ? 4 : 1);
   const int SILP = (silp == 2) ? 2 : (silp == 4 ? 4 : 1);
   const bool stage = F <= F_LDS_MAX;
   const size_t lds = stage ? (size_t)THREADS * SILP * (F + 1) * 4 : 0;
@@ -898,21 +928,23 @@ hipError_t rf_predict_heap_cpair_lv16_f32(const int* cpairs, const int* roots,
   if (blocks == 0) blocks = 1;
   const auto* pp = reinterpret_cast<const i32x2*>(cpairs);
   const auto* lv = reinterpret_cast<const float*>(lv16);
-#define LPX_RFHC16_CASE(I)                                                    \
-  if (SILP == I) {                                                            \
+#define LPX_RFHC16_CASE(I, T)                                                 \
+  if (SILP == I && TILP == T) {                                               \
     if (stage)                                                                \
-      rf_heap_cpair_kernel<true, I, true>                                     \
+      rf_heap_cpair_kernel<true, I, true, T>                                  \
           <<<dim3(blocks), dim3(THREADS), lds, stream>>>(                     \
               pp, roots, lv, n_trees, depth, X, out, S, F, C);                \
     else                                                                      \
-      rf_heap_cpair_kernel<false, I, true>                                    \
+      rf_heap_cpair_kernel<false, I, true, T>                                 \
           <<<dim3(blocks), dim3(THREADS), 0, stream>>>(                       \
               pp, roots, lv, n_trees, depth, X, out, S, F, C);                \
     return hipGetLastError();                                                 \
   }
-  LPX_RFHC16_CASE(1)
-  LPX_RFHC16_CASE(2)
-  LPX_RFHC16_CASE(4)
+  LPX_RFHC16_CASE(1, 1)
+  LPX_RFHC16_CASE(2, 1)
+  LPX_RFHC16_CASE(4, 1)
+  LPX_RFHC16_CASE(1, 2)
+  LPX_RFHC16_CASE(1, 4)
 #undef LPX_RFHC16_CASE
   return hipErrorInvalidValue;
 }
