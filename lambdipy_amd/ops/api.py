@@ -306,10 +306,12 @@ def rf_predict_proba_sum(nodes: torch.Tensor, leaf_values: torch.Tensor,
                     ok = bool(lv16.to(torch.float32).eq(lv).all().item())
                     lvc = (lv._version, lv16.contiguous() if ok else None)
                     lv._lpx_lv16_cache = lvc
+                # tree-ILP 4: four independent chains per lane fill the
+                # fetch latency (A/B: 280 -> 314 Ms/s at the bench shape)
                 if lvc[1] is not None:
                     return e.rf_predict_heap_cpair16(cp, rts, lvc[1], depth,
-                                                     x, 1)
-                return e.rf_predict_heap_cpair(cp, rts, lv, depth, x, 1)
+                                                     x, 41)
+                return e.rf_predict_heap_cpair(cp, rts, lv, depth, x, 41)
             if use_pair:
                 cached = getattr(n, "_lpx_pair_cache", None)
                 if cached is None or cached[0] != n._version:

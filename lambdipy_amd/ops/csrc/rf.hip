@@ -363,7 +363,7 @@ __global__ __launch_bounds__(BLK) void rf_heap_kernel(
 // invariant to block size / occupancy — profiles/README), and the L2 moves
 // a full line per random request either way, so ~10% fewer requests plus a
 // root served from a 2-KB cached array is a direct request-count cut.
-template <bool STAGE_X, int SILP>
+template <bool STAGE_X, int SILP, int TILP = 1>
 __global__ __launch_bounds__(THREADS) void rf_heap_pair_kernel(
     const i32x4* __restrict__ pairs, const i32x2* __restrict__ roots,
     const float* __restrict__ leaf_values, int n_trees, int depth,
@@ -403,39 +403,62 @@ __global__ __launch_bounds__(THREADS) void rf_heap_pair_kernel(
 #pragma unroll
       for (int c = 0; c < CMAX; ++c) acc[j][c] = 0.f;
 
-    for (int tr = 0; tr < n_trees; ++tr) {
-      const i32x4* tp = pairs + (long)tr * ppt;
-      const i32x2 root = roots[tr];
-      int nd[SILP];        // current heap index
-      int cf[SILP];        // current node feature
-      float ct[SILP];      // current node threshold
+    // TILP independent tree chains per lane (latency-bound kernel; see the
+    // compact variant): no extra LDS, registers only.
+    const int tr_main = n_trees - (n_trees % TILP);
+    for (int tr = 0; tr < n_trees; tr += (tr < tr_main ? TILP : 1)) {
+      const int nt = (tr < tr_main) ? TILP : 1;
+      int nd[SILP][TILP];
+      int cf[SILP][TILP];
+      float ct[SILP][TILP];
 #pragma unroll
-      for (int j = 0; j < SILP; ++j) {
-        nd[j] = 0;
-        cf[j] = root.x;
-        ct[j] = __int_as_float(root.y);
+      for (int ti = 0; ti < TILP; ++ti) {
+        const i32x2 root = roots[tr + (ti < nt ? ti : 0)];
+#pragma unroll
+        for (int j = 0; j < SILP; ++j) {
+          nd[j][ti] = 0;
+          cf[j][ti] = root.x;
+          ct[j][ti] = __int_as_float(root.y);
+        }
       }
       for (int d = 0; d < depth - 1; ++d) {
 #pragma unroll
         for (int j = 0; j < SILP; ++j) {
           if (s[j] < S) {
-            const i32x4 pr = tp[nd[j]];  // children of current node
-            const int go = (xrow[j][(unsigned)cf[j]] <= ct[j]) ? 0 : 1;
-            nd[j] = 2 * nd[j] + 1 + go;
-            cf[j] = go ? pr.z : pr.x;
-            ct[j] = __int_as_float(go ? pr.w : pr.y);
+            i32x4 pr[TILP];
+#pragma unroll
+            for (int ti = 0; ti < TILP; ++ti)
+              if (ti < nt)
+                pr[ti] = pairs[(long)(tr + ti) * ppt + nd[j][ti]];
+#pragma unroll
+            for (int ti = 0; ti < TILP; ++ti) {
+              if (ti < nt) {
+                const int go =
+                    (xrow[j][(unsigned)cf[j][ti]] <= ct[j][ti]) ? 0 : 1;
+                nd[j][ti] = 2 * nd[j][ti] + 1 + go;
+                cf[j][ti] = go ? pr[ti].z : pr[ti].x;
+                ct[j][ti] = __int_as_float(go ? pr[ti].w : pr[ti].y);
+              }
+            }
           }
         }
       }
 #pragma unroll
       for (int j = 0; j < SILP; ++j) {
         if (s[j] < S) {
-          const int go = (xrow[j][(unsigned)cf[j]] <= ct[j]) ? 0 : 1;
-          const int leaf = 2 * nd[j] + 1 + go - tree_stride;
-          const float* lv = leaf_values + ((long)tr * n_leaves + leaf) * C;
 #pragma unroll
-          for (int c = 0; c < CMAX; ++c)
-            if (c < C) acc[j][c] += lv[c];
+          for (int ti = 0; ti < TILP; ++ti) {
+            if (ti < nt) {
+              const int go =
+                  (xrow[j][(unsigned)cf[j][ti]] <= ct[j][ti]) ? 0 : 1;
+              const int leaf = 2 * nd[j][ti] + 1 + go - tree_stride;
+              const float* lv =
+                  leaf_values + ((long)(tr + ti) * n_leaves + leaf) * C;
+#pragma unroll
+              for (int c = 0; c < CMAX; ++c)
+                if (c < C) acc[j][c] += lv[c];
+            }
+          }
         }
       }
     }
@@ -843,6 +866,7 @@ hipError_t rf_predict_heap_pair_f32(const int* pairs, const int* roots,
                                     int S, int F, int C, int silp,
                                     hipStream_t stream) {
   if (C > CMAX || depth < 2 || depth > 26) return hipErrorInvalidValue;
+  const int TILP = (silp == 21) ? 2 : (silp == 41 ? 4 : 1);
   const int SILP = (silp == 2) ? 2 : (silp == 4 ? 4 : 1);
   const bool stage = F <= F_LDS_MAX;
   const size_t lds = stage ? (size_t)THREADS * SILP * (F + 1) * 4 : 0;
@@ -852,21 +876,23 @@ hipError_t rf_predict_heap_pair_f32(const int* pairs, const int* roots,
   if (blocks == 0) blocks = 1;
   const auto* pp = reinterpret_cast<const i32x4*>(pairs);
   const auto* rr = reinterpret_cast<const i32x2*>(roots);
-#define LPX_RFHP_CASE(I)                                                      \
-  if (SILP == I) {                                                            \
+#define LPX_RFHP_CASE(I, T)                                                   \
+  if (SILP == I && TILP == T) {                                               \
     if (stage)                                                                \
-      rf_heap_pair_kernel<true, I>                                            \
+      rf_heap_pair_kernel<true, I, T>                                         \
           <<<dim3(blocks), dim3(THREADS), lds, stream>>>(                     \
               pp, rr, leaf_values, n_trees, depth, X, out, S, F, C);          \
     else                                                                      \
-      rf_heap_pair_kernel<false, I>                                           \
+      rf_heap_pair_kernel<false, I, T>                                        \
           <<<dim3(blocks), dim3(THREADS), 0, stream>>>(                       \
               pp, rr, leaf_values, n_trees, depth, X, out, S, F, C);          \
     return hipGetLastError();                                                 \
   }
-  LPX_RFHP_CASE(1)
-  LPX_RFHP_CASE(2)
-  LPX_RFHP_CASE(4)
+  LPX_RFHP_CASE(1, 1)
+  LPX_RFHP_CASE(2, 1)
+  LPX_RFHP_CASE(4, 1)
+  LPX_RFHP_CASE(1, 2)
+  LPX_RFHP_CASE(1, 4)
 #undef LPX_RFHP_CASE
   return hipErrorInvalidValue;
 }
