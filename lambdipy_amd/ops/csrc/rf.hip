@@ -866,7 +866,7 @@ hipError_t rf_predict_heap_pair_f32(const int* pairs, const int* roots,
                                     int S, int F, int C, int silp,
                                     hipStream_t stream) {
   if (C > CMAX || depth < 2 || depth > 26) return hipErrorInvalidValue;
-  const int TILP = (silp == 21) ? 2 : (silp == 41 ? 4 : 1);
+  const int TILP = (silp == 21) ? 2 : (silp == 41 ? 4 : (silp == 81 ? 8 : 1));
   const int SILP = (silp == 2) ? 2 : (silp == 4 ? 4 : 1);
   const bool stage = F <= F_LDS_MAX;
   const size_t lds = stage ? (size_t)THREADS * SILP * (F + 1) * 4 : 0;
@@ -893,6 +893,7 @@ hipError_t rf_predict_heap_pair_f32(const int* pairs, const int* roots,
   LPX_RFHP_CASE(4, 1)
   LPX_RFHP_CASE(1, 2)
   LPX_RFHP_CASE(1, 4)
+  LPX_RFHP_CASE(1, 8)
 #undef LPX_RFHP_CASE
   return hipErrorInvalidValue;
 }
@@ -905,7 +906,7 @@ hipError_t rf_predict_heap_cpair_f32(const int* cpairs, const int* roots,
   if (C > CMAX || depth < 2 || depth > 26 || F > 0xffff)
     return hipErrorInvalidValue;
   // silp 21/41 encode tree-ILP 2/4 (SILP stays 1)
-  const int TILP = (silp == 21) ? 2 : (silp == 41 ? 4 : 1);
+  const int TILP = (silp == 21) ? 2 : (silp == 41 ? 4 : (silp == 81 ? 8 : 1));
   const int SILP = (silp == 2) ? 2 : (silp == 4 ? 4 : 1);
   const bool stage = F <= F_LDS_MAX;
   const size_t lds = stage ? (size_t)THREADS * SILP * (F + 1) * 4 : 0;
@@ -931,6 +932,7 @@ hipError_t rf_predict_heap_cpair_f32(const int* cpairs, const int* roots,
   LPX_RFHC_CASE(4, 1)
   LPX_RFHC_CASE(1, 2)
   LPX_RFHC_CASE(1, 4)
+  LPX_RFHC_CASE(1, 8)
 #undef LPX_RFHC_CASE
   return hipErrorInvalidValue;
 }
@@ -944,7 +946,7 @@ hipError_t rf_predict_heap_cpair_lv16_f32(const int* cpairs, const int* roots,
   if (C > CMAX || depth < 2 || depth > 26 || F > 0xffff)
     return hipErrorInvalidValue;
   // silp 21/41 encode tree-ILP 2/4 (SILP stays 1)
-  const int TILP = (silp == 21) ? 2 : (silp == 41 ? 4 : 1);
+  const int TILP = (silp == 21) ? 2 : (silp == 41 ? 4 : (silp == 81 ? 8 : 1));
   const int SILP = (silp == 2) ? 2 : (silp == 4 ? 4 : 1);
   const bool stage = F <= F_LDS_MAX;
   const size_t lds = stage ? (size_t)THREADS * SILP * (F + 1) * 4 : 0;
@@ -971,6 +973,7 @@ hipError_t rf_predict_heap_cpair_lv16_f32(const int* cpairs, const int* roots,
   LPX_RFHC16_CASE(4, 1)
   LPX_RFHC16_CASE(1, 2)
   LPX_RFHC16_CASE(1, 4)
+  LPX_RFHC16_CASE(1, 8)
 #undef LPX_RFHC16_CASE
   return hipErrorInvalidValue;
 }
