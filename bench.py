@@ -185,7 +185,7 @@ def main() -> int:
                 "features": args.features,
                 "classes": args.classes,
                 "parallelism": f"dp{n}",
-                "thresholds": "fp16-representable (compact 4B-node kernel, bitwise == f32 kernel)",
+                "thresholds": "fp16-representable thresholds+leaves (compact 4B-node kernel + fp16 leaf table + tree-ILP4, bitwise == f32 kernels)",
                 "baseline_metric": "packaged layer MB + cold-start->first-inference ms, sklearn RF 1/2/4/8 GPU",
             },
         }
