@@ -326,7 +326,7 @@ def rf_predict_proba_sum(nodes: torch.Tensor, leaf_values: torch.Tensor,
                     cached = (n._version, pairs, nh[:, 0, :].contiguous())
                     n._lpx_pair_cache = cached
                 _, pairs, rts = cached
-                return e.rf_predict_heap_pair(pairs, rts, lv, depth, x, 41)
+                return e.rf_predict_heap_pair(pairs, rts, lv, depth, x, 1)
             return e.rf_predict_heap(n, lv, depth, x, 1, -2)
     elif packed and chunks is not None:
         ctb, max_nodes = chunks

@@ -758,7 +758,11 @@ def test_lpx_rt_forestgen(ops):
     assert torch.isfinite(thr).all()
     assert abs(thr.mean().item()) < 0.1
     s = leaves.sum(dim=1)
-    assert torch.allclose(s, torch.ones_like(s), atol=1e-5)
+    # leaves are fp16-representable (device generator quantizes): rows sum
+    # to 1 within fp16 rounding
+    assert torch.allclose(s, torch.ones_like(s), atol=1e-3)
+    # and thresholds round-trip fp16 exactly (compact-kernel contract)
+    assert torch.equal(thr.to(torch.float16).to(torch.float32), thr)
     # the generated forest must traverse cleanly through the rf kernel
     X = _rand((4096, F), seed=3)
     out = torch.empty(4096, C, device=DEV)
@@ -766,7 +770,7 @@ def test_lpx_rt_forestgen(ops):
                    X.data_ptr(), out.data_ptr(), 4096, F, C)
     lpx_rt.sync()
     ps = out.sum(dim=1) / n_trees
-    assert torch.allclose(ps, torch.ones_like(ps), atol=1e-4)
+    assert torch.allclose(ps, torch.ones_like(ps), atol=1e-3)
 
     # checksum kernel vs torch sum
     labels = torch.randint(0, C, (100000,), dtype=torch.int32, device=DEV)
