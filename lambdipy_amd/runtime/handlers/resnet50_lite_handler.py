@@ -57,12 +57,17 @@ def main() -> dict:
     mark("hip_init_lib_load", t)
 
     t = time.perf_counter()
+    # LPX_RESNET_WEIGHTS: serve a real checkpoint (saved by
+    # ResNet50Lite.save_weights) instead of device random-init
+    wpath = os.environ.get("LPX_RESNET_WEIGHTS")
     models = []
     for d in range(g):
         lpx_rt.set_device(d)
-        models.append(lpx_nn.ResNet50Lite(n_classes=1000, seed=3,
-                                          alloc_only=(d > 0)))
+        models.append(lpx_nn.ResNet50Lite(
+            n_classes=1000, seed=3, alloc_only=(d > 0 or wpath is not None)))
     lpx_rt.set_device(0)
+    if wpath:
+        models[0].load_weights(wpath)
     models[0].stream.sync()
     mark("model_build", t)
 

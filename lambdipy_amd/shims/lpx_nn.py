@@ -314,6 +314,39 @@ class ResNet50Lite:
         for c, w, sa in zip(self.convs, s["w"], s["sa"]):
             c.wscale, c.sa = w, sa
 
+    # -- checkpointing -------------------------------------------------------
+
+    def save_weights(self, path: str) -> None:
+        """One file: JSON header (scales incl. frozen activation scales if
+        calibrated) + the raw weight slab (e4m3 weights + f32 biases)."""
+        import json
+
+        buf = (ctypes.c_ubyte * self.slab_bytes)()
+        self.stream.sync()
+        lpx_rt.memcpy_d2h(ctypes.addressof(buf), self.slab, self.slab_bytes)
+        hdr = json.dumps({"format": 1, "slab_bytes": self.slab_bytes,
+                          "n_classes": self.n_classes,
+                          "scales": self.export_scales()}).encode()
+        with open(path, "wb") as f:
+            f.write(len(hdr).to_bytes(8, "little"))
+            f.write(hdr)
+            f.write(bytes(buf))
+
+    def load_weights(self, path: str) -> None:
+        """Restore slab + scales (use with alloc_only=True for a cold start
+        that serves REAL weights: one sequential read + one H2D copy)."""
+        import json
+
+        with open(path, "rb") as f:
+            hlen = int.from_bytes(f.read(8), "little")
+            hdr = json.loads(f.read(hlen).decode())
+            assert hdr["format"] == 1 and hdr["n_classes"] == self.n_classes
+            assert hdr["slab_bytes"] == self.slab_bytes, "architecture mismatch"
+            blob = f.read(self.slab_bytes)
+        arr = (ctypes.c_ubyte * self.slab_bytes).from_buffer_copy(blob)
+        lpx_rt.memcpy_h2d(self.slab, ctypes.addressof(arr), self.slab_bytes)
+        self.import_scales(hdr["scales"])
+
     # -- layers -------------------------------------------------------------
 
     def _conv_f32(self, c: _Conv, x: int, N: int, H: int, W: int,

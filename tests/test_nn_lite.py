@@ -250,3 +250,33 @@ def test_lite_cold_start_e2e(tmp_path, handler, env):
     assert res.total_ms < 60_000
     phases = {p["phase"] for p in res.handler_report.get("phases", [])}
     assert "first_inference" in phases
+
+
+@pytest.mark.gpu
+def test_resnet50_lite_checkpoint_roundtrip(tmp_path):
+    """save_weights/load_weights: a fresh alloc-only model restored from the
+    checkpoint must produce BITWISE-identical logits (incl. frozen scales)."""
+    from lambdipy_amd.shims import lpx_nn, lpx_rt
+
+    lpx_rt.set_device(0)
+    m = lpx_nn.ResNet50Lite(n_classes=1000, seed=6)
+    N = 4
+    x = m.pool.alloc(N * 224 * 224 * 3 * 4)
+    lpx_nn._ck(lpx_nn.lib().lpxn_randn(x, N * 224 * 224 * 3, 9, 1.0,
+                                       m.stream.h), "randn")
+    m.forward(x, N)  # calibrate (freezes activation scales)
+    ref = m.forward(x, N)
+    m.stream.sync()
+    want = _d2h_f32(ref, N * 1000).copy()
+    ck = str(tmp_path / "rn50.lpxw")
+    m.save_weights(ck)
+    m2 = lpx_nn.ResNet50Lite(n_classes=1000, alloc_only=True)
+    m2.load_weights(ck)
+    assert m2.calibrated
+    x2 = m2.pool.alloc(N * 224 * 224 * 3 * 4)
+    lpx_nn._ck(lpx_nn.lib().lpxn_randn(x2, N * 224 * 224 * 3, 9, 1.0,
+                                       m2.stream.h), "randn")
+    out = m2.forward(x2, N)  # fused path directly (calibrated from ckpt)
+    m2.stream.sync()
+    got = _d2h_f32(out, N * 1000)
+    assert np.array_equal(got, want), "checkpoint roundtrip logits mismatch"
