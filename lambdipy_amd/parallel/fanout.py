@@ -129,6 +129,35 @@ def scatter_rows(ctx: FanoutContext, X_root: Optional[torch.Tensor],
     return recv[: counts[ctx.rank]]
 
 
+def scatter_rows_async(ctx: FanoutContext, X_root: Optional[torch.Tensor],
+                       counts: List[int], row_shape: Tuple[int, ...],
+                       dtype: torch.dtype):
+    """Async scatter: returns (work, padded_recv). The caller overlaps the
+    NEXT batch's scatter with the CURRENT batch's compute (serving
+    pipeline): xGMI scatter is per-peer point-to-point, so the transfer
+    runs on RCCL's comm stream concurrently with compute and the exposed
+    per-step cost is max(compute, scatter) instead of the sum. Call
+    work.wait() then slice recv[:counts[rank]].
+
+    world 1: returns (None, X_root)."""
+    if ctx.world_size == 1:
+        assert X_root is not None
+        return None, X_root
+    maxc = max(counts)
+    recv = torch.empty((maxc,) + row_shape, dtype=dtype, device=ctx.device)
+    scatter_list = None
+    if ctx.is_root:
+        scatter_list = []
+        off = 0
+        for c in counts:
+            t = torch.zeros((maxc,) + row_shape, dtype=dtype, device=ctx.device)
+            t[:c] = X_root[off : off + c]
+            scatter_list.append(t)
+            off += c
+    work = dist.scatter(recv, scatter_list, src=0, async_op=True)
+    return work, recv
+
+
 def all_gather_rows(ctx: FanoutContext, local: torch.Tensor,
                     counts: List[int]) -> torch.Tensor:
     """Gather per-rank row shards back into the full [N, ...] tensor."""
