@@ -108,7 +108,7 @@ def main() -> dict:
     # calibration runs once on GPU 0; frozen scales are shared (the weights
     # are identical on every device after the broadcast)
     lpx_rt.set_device(0)
-    logits0 = models[0].forward(X[0] + 0 * shard_el * 4, batch, res, res)
+    logits0 = models[0].forward(X[0], batch, res, res)
     scales = models[0].export_scales()
     labels, labels_all = [], []
     for d in range(g):

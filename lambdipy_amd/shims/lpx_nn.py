@@ -259,6 +259,8 @@ class ResNet50Lite:
         self.convs, self.blocks, self.conv1, self.fc = resnet50_spec(n_classes)
         self._graph = None
         self._graph_key = None
+        self._logits = 0
+        self._owned: List[int] = []
         self._lay_out_slab()
         if not alloc_only:
             self._init_weights(seed)
@@ -448,7 +450,7 @@ class ResNet50Lite:
         return out
 
     def _free_prev(self):
-        for p in getattr(self, "_owned", []):
+        for p in self._owned:
             self.pool.free(p)
         self._owned = []
 
