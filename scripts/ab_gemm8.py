@@ -45,6 +45,19 @@ for t in (0, 14, 17):
     med = sorted(times[t])[len(times[t]) // 2]
     print(f"tune {t} @4096: median {med*1e3:.3f} ms {2*4096**3/med/1e12:.0f} TF "
           f"(min {2*4096**3/max(times[t])/1e12:.0f})")
+# zero-filled vs random operands (guide §5.4 rule 25: quote random; the
+# template's 1563-zero vs 1320-random spread is data-dependent clocking)
+Z = torch.zeros(4096, 4096, device="cuda", dtype=torch.bfloat16)
+for _ in range(3):
+    e.gemm_bf16_tune8(Z, Z, 14)
+torch.cuda.synchronize()
+t0 = time.perf_counter()
+for _ in range(10):
+    e.gemm_bf16_tune8(Z, Z, 14)
+torch.cuda.synchronize()
+dz = (time.perf_counter() - t0) / 10
+print(f"tune 14 @4096 ZERO-filled: {2*4096**3/dz/1e12:.0f} TF")
+
 times = collections.defaultdict(list)
 for rnd in range(6):
     for t in (0, 14, 17):
