@@ -503,6 +503,197 @@ __global__ __launch_bounds__(THREADS4, 1) void gemm_mx4w_8ph_kernel(
   }
 }
 
+// ---- tri-parity variant: 3 data buffers, ONE barrier per phase -----------
+// The 2-parity kernel is glds/barrier latency-bound (PMC wait 0.33,
+// mfma_busy 0.79): with a third data parity the staged slot is never the
+// one being read, so the second (post-MFMA) barrier of every phase
+// disappears (4 barriers/tile instead of 8) and the counted vmcnt windows
+// deepen to 8-11 in-flight events. Costs occupancy: 102 KiB LDS -> 1
+// block/CU (2 waves/SIMD) — the A/B decides.
+constexpr int SC_BASE3 = 3 * PARITY4;
+
+__device__ __forceinline__ void stage_half4_3p(
+    const unsigned char* __restrict__ A, const unsigned char* __restrict__ Bt,
+    long Kb, int bm, int bn, int tile, int hh, int ksteps,
+    char* __restrict__ smem, int t) {
+  const int tc = tile < ksteps ? tile : ksteps - 1;
+  const long k0 = (long)tc * 64;
+  const unsigned char* src = (hh < 2) ? A : Bt;
+  const int row0 = ((hh < 2) ? bm : bn) + (hh & 1) * 128;
+  char* const base = smem + (size_t)(tile % 3) * PARITY4 +
+                     (size_t)hh * HALF4;
+  const int r = t >> 2, cell = src_cell4(t);
+  const unsigned char* g = src + (long)(row0 + r) * Kb + (k0 + cell * 16);
+  __builtin_amdgcn_global_load_lds(
+      (const __attribute__((address_space(1))) void*)g,
+      (__attribute__((address_space(3))) void*)(base + (t >> 6) * 1024),
+      16, 0, 0);
+}
+
+__device__ __forceinline__ void stage_scales4_3p(
+    const unsigned char* __restrict__ sA, const unsigned char* __restrict__ sB,
+    int Mtot, int Ntot, int bm, int bn, int tile, int ksteps,
+    char* __restrict__ smem, int w, int lane) {
+  const int tc = tile < ksteps ? tile : ksteps - 1;
+  const int t = w * 64 + lane;
+  const unsigned char* g =
+      (t < 256) ? sA + ((long)tc * Mtot + bm + t) * 4
+                : sB + ((long)tc * Ntot + bn + (t - 256)) * 4;
+  char* const base = smem + SC_BASE3 + (size_t)(tile % 3) * SC_TILE4 + w * 256;
+  __builtin_amdgcn_global_load_lds(
+      (const __attribute__((address_space(1))) void*)g,
+      (__attribute__((address_space(3))) void*)base, 4, 0, 0);
+}
+
+__device__ __forceinline__ i32x8v frag4_3p(const char* __restrict__ smem,
+                                           int par3, int hh, int row_in_half,
+                                           int kg, int fkb) {
+  const char* half = smem + (size_t)par3 * PARITY4 + (size_t)hh * HALF4;
+  const i32x4 v = *reinterpret_cast<const i32x4*>(
+      half + phys_cell4(row_in_half, kg * 2 + fkb) * 16);
+  return i32x8v{v.x, v.y, v.z, v.w, 0, 0, 0, 0};
+}
+
+__device__ __forceinline__ unsigned scaleword4_3p(const char* __restrict__ smem,
+                                                  int slot3, int hh,
+                                                  int row_in_half) {
+  const int row = (hh < 2 ? 0 : 1024) + ((hh & 1) * 128 + row_in_half) * 4;
+  return *reinterpret_cast<const unsigned*>(
+      smem + SC_BASE3 + (size_t)slot3 * SC_TILE4 + row);
+}
+
+template <int N>
+__device__ __forceinline__ void waitvm3p() {
+  asm volatile("s_waitcnt vmcnt(%0)" ::"i"(N) : "memory");
+}
+
+__global__ __launch_bounds__(THREADS4, 1) void gemm_mx4_3p_kernel(
+    const unsigned char* __restrict__ A, const unsigned char* __restrict__ sA,
+    const unsigned char* __restrict__ Bt, const unsigned char* __restrict__ sB,
+    float* __restrict__ C, int M, int N, int K, unsigned tiles_m,
+    unsigned tiles_n) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+
+  const unsigned nwg = tiles_m * tiles_n;
+  const unsigned bid = lpx_xcd_swizzle(blockIdx.x, nwg);
+  const int bm = (int)(bid / tiles_n) * BM4;
+  const int bn = (int)(bid % tiles_n) * BM4;
+
+  const int lane = threadIdx.x & 63;
+  const int w = threadIdx.x >> 6;
+  const int wsr = w >> 2, wsc = w & 3;
+  const int fr = lane & 31;
+  const int fkb = lane >> 5;
+  const long Kb = K >> 1;
+
+  f32x16 acc[4][2];
+#pragma unroll
+  for (int p = 0; p < 4; ++p)
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+      for (int e = 0; e < 16; ++e) acc[p][i][e] = 0.f;
+
+  const int ksteps = (K + BK4 - 1) / BK4;
+  const int t512 = threadIdx.x;
+
+  // prologue: t0 all, sc0, sc1, t1 all, t2 lo pair — windows derived so the
+  // steady per-phase waits stay valid from tile 0 (see counts below)
+  stage_half4_3p(A, Bt, Kb, bm, bn, 0, 0, ksteps, smem, t512);
+  stage_half4_3p(A, Bt, Kb, bm, bn, 0, 2, ksteps, smem, t512);
+  stage_half4_3p(A, Bt, Kb, bm, bn, 0, 3, ksteps, smem, t512);
+  stage_half4_3p(A, Bt, Kb, bm, bn, 0, 1, ksteps, smem, t512);
+  stage_scales4_3p(sA, sB, M, N, bm, bn, 0, ksteps, smem, w, lane);
+  stage_scales4_3p(sA, sB, M, N, bm, bn, 1, ksteps, smem, w, lane);
+  stage_half4_3p(A, Bt, Kb, bm, bn, 1, 0, ksteps, smem, t512);
+  stage_half4_3p(A, Bt, Kb, bm, bn, 1, 2, ksteps, smem, t512);
+  stage_half4_3p(A, Bt, Kb, bm, bn, 1, 3, ksteps, smem, t512);
+  stage_half4_3p(A, Bt, Kb, bm, bn, 1, 1, ksteps, smem, t512);
+  stage_half4_3p(A, Bt, Kb, bm, bn, 2, 0, ksteps, smem, t512);
+  stage_half4_3p(A, Bt, Kb, bm, bn, 2, 2, ksteps, smem, t512);
+  waitvm3p<6>();  // everything up to sc1 (t0 halves + both scale slots)
+  __builtin_amdgcn_s_barrier();
+
+  if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
+    __builtin_amdgcn_s_setprio(1);
+
+  for (int t = 0; t < ksteps; ++t) {
+    const int par3 = t % 3;
+    const int slot3 = t % 3;
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      const int rh = (p >> 1) & 1;
+      const int ch = p & 1;
+      // per-phase counted waits (events/tile = 5: p0 D+S, p1..p3 D):
+      //  p0 gates on sc(t)   @ (t-2,p0)-S: 8 younger events
+      //  p1 gates on B-hi(t) @ (t-2,p0)-D: 11
+      //  p2 gates on A-hi(t) @ (t-2,p1):   10
+      //  p3 needs nothing new (barrier only)
+      if (p == 0) waitvm3p<8>();
+      else if (p == 1) waitvm3p<11>();
+      else if (p == 2) waitvm3p<10>();
+      __builtin_amdgcn_s_barrier();
+
+      i32x8v aF[2][2], bF[2];
+      int sa[2][2], sb[2];
+      unsigned aw[2], bw;
+#pragma unroll
+      for (int mf = 0; mf < 2; ++mf)
+        aw[mf] = scaleword4_3p(smem, slot3, rh, wsr * 64 + mf * 32 + fr);
+      bw = scaleword4_3p(smem, slot3, 2 + ch, wsc * 32 + fr);
+#pragma unroll
+      for (int kg = 0; kg < 2; ++kg) {
+#pragma unroll
+        for (int mf = 0; mf < 2; ++mf) {
+          const int r = wsr * 64 + mf * 32 + fr;
+          aF[mf][kg] = frag4_3p(smem, par3, rh, r, kg, fkb);
+          sa[mf][kg] = (int)((aw[mf] >> (8 * (kg * 2 + fkb))) & 0xff);
+        }
+        const int rb = wsc * 32 + fr;
+        bF[kg] = frag4_3p(smem, par3, 2 + ch, rb, kg, fkb);
+        sb[kg] = (int)((bw >> (8 * (kg * 2 + fkb))) & 0xff);
+      }
+      // staging schedule: p0 (t+2,B-hi)+sc(t+2), p1 (t+2,A-hi),
+      // p2 (t+3,A-lo), p3 (t+3,B-lo) — (t+3) lands in parity t%3 whose
+      // lo slots were last read at p1/p2 (>=1 barrier earlier)
+      if (p == 0) {
+        stage_half4_3p(A, Bt, Kb, bm, bn, t + 2, 3, ksteps, smem, t512);
+        stage_scales4_3p(sA, sB, M, N, bm, bn, t + 2, ksteps, smem, w, lane);
+      } else if (p == 1) {
+        stage_half4_3p(A, Bt, Kb, bm, bn, t + 2, 1, ksteps, smem, t512);
+      } else if (p == 2) {
+        stage_half4_3p(A, Bt, Kb, bm, bn, t + 3, 0, ksteps, smem, t512);
+      } else {
+        stage_half4_3p(A, Bt, Kb, bm, bn, t + 3, 2, ksteps, smem, t512);
+      }
+#pragma unroll
+      for (int kg = 0; kg < 2; ++kg)
+#pragma unroll
+        for (int mf = 0; mf < 2; ++mf)
+          acc[p][mf] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+              aF[mf][kg], bF[kg], acc[p][mf], 4, 4, 0, sa[mf][kg], 0, sb[kg]);
+    }
+  }
+  waitvm3p<0>();
+
+  const int ocol_l = lane & 31;
+  const int orow_base = 4 * (lane >> 5);
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    const int rh = (p >> 1) & 1, ch = p & 1;
+#pragma unroll
+    for (int mf = 0; mf < 2; ++mf) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int row = bm + rh * 128 + wsr * 64 + mf * 32 + (r & 3) +
+                        8 * (r >> 2) + orow_base;
+        const int col = bn + ch * 128 + wsc * 32 + ocol_l;
+        C[(long)row * N + col] = acc[p][mf][r];
+      }
+    }
+  }
+}
+
 }  // namespace
 
 hipError_t quantize_mx_fp4(const float* x, unsigned char* q, unsigned char* sc,
@@ -535,6 +726,15 @@ hipError_t gemm_mx_fp4_nt8ph(const unsigned char* A, const unsigned char* sA,
     gemm_mx4w_8ph_kernel<<<dim3(tiles_m * tiles_n), dim3(THREADS4), lds,
                            stream>>>(A, sA, Bt, sB, C, M, N, K, tiles_m,
                                      tiles_n);
+    return hipGetLastError();
+  }
+  const char* p3 = getenv("LPX_FP4_3P");
+  if (p3 && p3[0] == '1') {
+    // tri-parity: 1 barrier/phase, deep vmcnt windows, 1 block/CU (A/B)
+    const size_t lds3 = 3 * PARITY4 + 3 * SC_TILE4;  // 102 KiB
+    gemm_mx4_3p_kernel<<<dim3(tiles_m * tiles_n), dim3(THREADS4), lds3,
+                         stream>>>(A, sA, Bt, sB, C, M, N, K, tiles_m,
+                                   tiles_n);
     return hipGetLastError();
   }
   const size_t lds = 2 * PARITY4 + 3 * SC_TILE4;  // 70 KiB -> 2 blocks/CU
