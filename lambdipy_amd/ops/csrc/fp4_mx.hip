@@ -728,8 +728,12 @@ hipError_t gemm_mx_fp4_nt8ph(const unsigned char* A, const unsigned char* sA,
                                      tiles_n);
     return hipGetLastError();
   }
+  // tri-parity wins on long-K shapes (2692 vs 2549 TF @8192 within-probe;
+  // loses @4096: 2364 vs 2569 — the deep windows need many K-steps to
+  // amortize the occupancy drop). LPX_FP4_3P=0/1 force.
   const char* p3 = getenv("LPX_FP4_3P");
-  if (p3 && p3[0] == '1') {
+  const bool use3p = p3 ? (p3[0] == '1') : (K >= 6144);
+  if (use3p) {
     // tri-parity: 1 barrier/phase, deep vmcnt windows, 1 block/CU (A/B)
     const size_t lds3 = 3 * PARITY4 + 3 * SC_TILE4;  // 102 KiB
     gemm_mx4_3p_kernel<<<dim3(tiles_m * tiles_n), dim3(THREADS4), lds3,

@@ -34,7 +34,7 @@ def main():
         A = torch.randn(sz, sz, device="cuda")
         qa, sa = quant(A)
         qb, sb = quant(torch.randn(sz, sz, device="cuda"))
-        os.environ.pop("LPX_FP4_3P", None)
+        os.environ["LPX_FP4_3P"] = "0"
         ref = e.gemm_mx4(qa, sa, qb, sb)
         os.environ["LPX_FP4_3P"] = "1"
         for rep in range(5):
@@ -49,7 +49,7 @@ def main():
         qa, sa = quant(A)
         qb, sb = quant(B)
         flops = 2.0 * n ** 3
-        os.environ.pop("LPX_FP4_3P", None)
+        os.environ["LPX_FP4_3P"] = "0"
         ref = e.gemm_mx4(qa, sa, qb, sb)
         dt = timeit(lambda: e.gemm_mx4(qa, sa, qb, sb))
         print(f"2p @ {n}: {flops/dt/1e12:6.0f} TF")
