@@ -285,9 +285,10 @@ __global__ __launch_bounds__(THREADS2, 2) void gemm_fp8_nt256_kernel(
     float* __restrict__ C, const float* __restrict__ bias,
     const float* __restrict__ resid, int M, int N,
     int K, float descale, unsigned tiles_m, unsigned tiles_n) {
+  // LDS double-buffered by K-step parity: one barrier per step (see the
+  // conv kernel's note); 80 KiB still fits 2 blocks/CU.
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* const lA = smem;
-  char* const lB = smem + BM2 * ROWB2;
+  const int SLOT = 2 * BM2 * ROWB2;
 
   const unsigned nwg = tiles_m * tiles_n;
   const unsigned bid = lpx_xcd_swizzle(blockIdx.x, nwg);
@@ -310,12 +311,14 @@ __global__ __launch_bounds__(THREADS2, 2) void gemm_fp8_nt256_kernel(
   i32x4 aRegs[CHUNKS2], bRegs[CHUNKS2];
   load_fp8_regs(A, M, K, bm, 0, aRegs);
   load_fp8_regs(Bt, N, K, bn, 0, bRegs);
-  write_fp8_lds(aRegs, lA);
-  write_fp8_lds(bRegs, lB);
+  write_fp8_lds(aRegs, smem);
+  write_fp8_lds(bRegs, smem + BM2 * ROWB2);
   __syncthreads();
 
   const int ksteps = (K + BK2 - 1) / BK2;
   for (int ks = 0; ks < ksteps; ++ks) {
+    char* const lA = smem + (ks & 1) * SLOT;
+    char* const lB = lA + BM2 * ROWB2;
     if (ks + 1 < ksteps) {
       load_fp8_regs(A, M, K, bm, (ks + 1) * BK2, aRegs);
       load_fp8_regs(Bt, N, K, bn, (ks + 1) * BK2, bRegs);
@@ -342,10 +345,10 @@ __global__ __launch_bounds__(THREADS2, 2) void gemm_fp8_nt256_kernel(
               __builtin_bit_cast(long, aF[mf]), __builtin_bit_cast(long, bF[nf]),
               acc[mf][nf], 0, 0, 0);
     }
-    __syncthreads();
     if (ks + 1 < ksteps) {
-      write_fp8_lds(aRegs, lA);
-      write_fp8_lds(bRegs, lB);
+      char* const nA = smem + ((ks + 1) & 1) * SLOT;
+      write_fp8_lds(aRegs, nA);
+      write_fp8_lds(bRegs, nA + BM2 * ROWB2);
     }
     __syncthreads();
   }
@@ -766,9 +769,13 @@ __global__ __launch_bounds__(THREADS2, 2) void conv_fp8_impx_kernel(
     const float* __restrict__ bias, ConvGeom g, int M,
     int N, int K, float descale, float out_inv, unsigned tiles_m,
     unsigned tiles_n) {
+  // LDS double-buffered by K-step parity: the write of tile ks+1 goes to
+  // the slot whose reads finished at the PREVIOUS step's barrier, so each
+  // step needs ONE barrier (after the writes) instead of two — the
+  // write-after-read stall between the MFMA block and the LDS fill
+  // disappears. 80 KiB still fits 2 blocks/CU.
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* const lA = smem;
-  char* const lB = smem + BM2 * ROWB2;
+  const int SLOT = 2 * BM2 * ROWB2;
 
   const unsigned nwg = tiles_m * tiles_n;
   const unsigned bid = lpx_xcd_swizzle(blockIdx.x, nwg);
@@ -794,12 +801,14 @@ __global__ __launch_bounds__(THREADS2, 2) void conv_fp8_impx_kernel(
   imp_init(cur, g, M, bm);
   imp_gather(xq, cur, g, K, aRegs);
   load_fp8_regs(Wq, N, K, bn, 0, bRegs);
-  write_fp8_lds(aRegs, lA);
-  write_fp8_lds(bRegs, lB);
+  write_fp8_lds(aRegs, smem);
+  write_fp8_lds(bRegs, smem + BM2 * ROWB2);
   __syncthreads();
 
   const int ksteps = (K + BK2 - 1) / BK2;
   for (int ks = 0; ks < ksteps; ++ks) {
+    char* const lA = smem + (ks & 1) * SLOT;
+    char* const lB = lA + BM2 * ROWB2;
     if (ks + 1 < ksteps) {
       imp_advance(cur, g);
       imp_gather(xq, cur, g, K, aRegs);
@@ -818,10 +827,10 @@ __global__ __launch_bounds__(THREADS2, 2) void conv_fp8_impx_kernel(
       for (int nf = 0; nf < 2; ++nf)
         acc[mf][nf] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
             aF[mf], bF[nf], acc[mf][nf], 0, 0, 0, 127, 0, 127);
-    __syncthreads();
     if (ks + 1 < ksteps) {
-      write_fp8_lds(aRegs, lA);
-      write_fp8_lds(bRegs, lB);
+      char* const nA = smem + ((ks + 1) & 1) * SLOT;
+      write_fp8_lds(aRegs, nA);
+      write_fp8_lds(bRegs, nA + BM2 * ROWB2);
     }
     __syncthreads();
   }
@@ -871,7 +880,7 @@ hipError_t conv_fp8_implicit(const unsigned char* xq, const unsigned char* Wq,
                                  S, HO, WO, stride, pad, KO, descale, relu,
                                  1.f, stream);
   const unsigned tiles_m = lpx_ceil_div(M, BM2), tiles_n = lpx_ceil_div(KO, BN2);
-  const size_t lds = (size_t)2 * BM2 * ROWB2;
+  const size_t lds = (size_t)4 * BM2 * ROWB2;  // 2 LDS slots (see kernel)
   dim3 grid(tiles_m * tiles_n), block(THREADS2);
   if (bias && relu)
     conv_fp8_impx_kernel<true, true><<<grid, block, lds, stream>>>(
@@ -907,7 +916,7 @@ hipError_t conv_fp8_implicit_q8(const unsigned char* xq,
                                  S, HO, WO, stride, pad, KO, descale, relu,
                                  out_inv, stream);
   const unsigned tiles_m = lpx_ceil_div(M, BM2), tiles_n = lpx_ceil_div(KO, BN2);
-  const size_t lds = (size_t)2 * BM2 * ROWB2;
+  const size_t lds = (size_t)4 * BM2 * ROWB2;  // 2 LDS slots (see kernel)
   dim3 grid(tiles_m * tiles_n), block(THREADS2);
   if (bias && relu)
     conv_fp8_impx_kernel<true, true, true><<<grid, block, lds, stream>>>(
@@ -991,7 +1000,7 @@ hipError_t gemm_fp8_nt_res(const unsigned char* A, const unsigned char* Bt,
   const bool big = (M >= 512 && N >= 512 && K >= 64);
   if (big) {
     const unsigned tiles_m = lpx_ceil_div(M, BM2), tiles_n = lpx_ceil_div(N, BN2);
-    const size_t lds = (size_t)2 * BM2 * ROWB2;  // 40 KiB
+    const size_t lds = (size_t)4 * BM2 * ROWB2;  // 80 KiB (2 LDS slots)
     dim3 grid(tiles_m * tiles_n), block(THREADS2);
     if (bias && relu)
       gemm_fp8_nt256_kernel<true, true><<<grid, block, lds, stream>>>(
