@@ -505,7 +505,9 @@ __global__ __launch_bounds__(THREADST, 3) void gemm_fp8_ts_kernel(
     for (int nf = 0; nf < 2; ++nf)
       acc[nf] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
           aF, bF[nf], acc[nf], 0, 0, 0, 127, 0, 127);
-    __syncthreads();
+    // single barrier per step: the cur^1 slot's reads finished at the
+    // PREVIOUS step's barrier, so writing it during this step's MFMAs is
+    // safe (same argument as conv_fp8_impx)
     if (ks + 1 < ksteps) {
       const int t = threadIdx.x;
       char* aN = lA + (cur ^ 1) * ABUF;
