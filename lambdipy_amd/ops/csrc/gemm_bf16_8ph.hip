@@ -488,6 +488,10 @@ hipError_t gemm_bf16_nt8ph_tune(const uint16_t* A, const uint16_t* Bt,
     // hand-asm phase body: counted descending lgkmcnt between ds_reads
     // and MFMAs (the compiler's schedule drains lgkmcnt(0) per group)
     case 16: return launch8ph<false, 60, false, 1, 1, -1, true>(A, Bt, C, M, N, K, stream);
+    // ILV=0 without the forced lgkmcnt(0): all 12 ds_reads issued before
+    // the barrier, MFMAs gated only by the compiler's own counted waits
+    case 18: return launch8ph<false, 60, false, 1, 0>(A, Bt, C, M, N, K, stream);
+    case 19: return launch8ph<false, 60, false, 1, 0, 8>(A, Bt, C, M, N, K, stream);
     // BK=32 half-LDS (64 KiB -> 2 blocks/CU) variant
     case 17: {
       if (K % BK32 != 0) return hipErrorInvalidValue;
