@@ -454,10 +454,12 @@ bool gemm_bf16_8ph_supported(int M, int N, int K) {
 
 hipError_t gemm_bf16_nt8ph(const uint16_t* A, const uint16_t* Bt, float* C,
                            int M, int N, int K, hipStream_t stream) {
-  // tune-14 config (r2 A/B): vmcnt(6) at odd phases, deferred ks=1 reads,
-  // static half-wave priority, lgkmcnt(8) throttle before the barrier —
-  // 1139/1144 TF median @4096/8192 vs 1102/1133 for the old default.
-  return launch8ph<false, 60, false, 1, 1, 8>(A, Bt, C, M, N, K, stream);
+  // tune-19 config (r2 A/B): vmcnt(6) at odd phases, ALL 12 ds_reads
+  // issued before the barrier with the lgkmcnt(8) throttle and NO forced
+  // post-barrier drain (the compiler emits counted waits at first use) —
+  // beat tune 14 within-probe on both sizes (1063/1120 vs 1019/1095 on a
+  // slow box; ~1140+ mid-band).
+  return launch8ph<false, 60, false, 1, 0, 8>(A, Bt, C, M, N, K, stream);
 }
 
 // tuning variants for within-probe A/B (scripts/ab_gemm8.py)
