@@ -494,6 +494,11 @@ hipError_t gemm_bf16_nt8ph_tune(const uint16_t* A, const uint16_t* Bt,
     // the barrier, MFMAs gated only by the compiler's own counted waits
     case 18: return launch8ph<false, 60, false, 1, 0>(A, Bt, C, M, N, K, stream);
     case 19: return launch8ph<false, 60, false, 1, 0, 8>(A, Bt, C, M, N, K, stream);
+    // throttle sweep for the ILV0 form (12 reads pre-barrier)
+    case 20: return launch8ph<false, 60, false, 1, 0, 4>(A, Bt, C, M, N, K, stream);
+    case 21: return launch8ph<false, 60, false, 1, 0, 6>(A, Bt, C, M, N, K, stream);
+    case 22: return launch8ph<false, 60, false, 1, 0, 10>(A, Bt, C, M, N, K, stream);
+    case 23: return launch8ph<false, 8, false, 1, 0, 8>(A, Bt, C, M, N, K, stream);
     // BK=32 half-LDS (64 KiB -> 2 blocks/CU) variant
     case 17: {
       if (K % BK32 != 0) return hipErrorInvalidValue;
