@@ -54,17 +54,28 @@ def main() -> dict:
         mods.append(lpx_rt.Module())
     mark("hip_init_module_load", t)
 
+    # compact serving layout by default (u16 feat | fp16 thr pair records +
+    # fp16 leaf table + tree-ILP-4 kernel — the torch extension's default,
+    # ~309 vs ~240 Ms/s); LPX_RF_COMPACT_LITE=0 falls back to the f32 heap
+    compact = os.environ.get("LPX_RF_COMPACT_LITE", "1") == "1" and depth >= 2
     n_internal = n_trees * ((1 << depth) - 1)
+    ppt = (1 << (depth - 1)) - 1
     n_leaf_rows = n_trees * (1 << depth)
-    nodes_bytes = n_internal * 8
-    leaves_bytes = n_leaf_rows * n_classes * 4
+    if compact:
+        nodes_bytes = n_trees * ppt * 8  # pair records
+        roots_bytes = n_trees * 4
+        leaves_bytes = n_leaf_rows * n_classes * 2  # fp16
+    else:
+        nodes_bytes = n_internal * 8
+        roots_bytes = 0
+        leaves_bytes = n_leaf_rows * n_classes * 4
     total = batch * g
     xel = total * n_features
 
     t = time.perf_counter()
     # one slab per device (a single hipMalloc; sub-buffers by offset)
-    sizes = [nodes_bytes, leaves_bytes, xel * 4, batch * n_classes * 4,
-             batch * 4, total * 4, 8]
+    sizes = [nodes_bytes, max(roots_bytes, 8), leaves_bytes, xel * 4,
+             batch * n_classes * 4, batch * 4, total * 4, 8]
     offs = []
     cur = 0
     for sz in sizes:
@@ -75,19 +86,25 @@ def main() -> dict:
         lpx_rt.set_device(d)
         slab = lpx_rt.malloc(cur)
         b = {k: slab + o for k, o in zip(
-            ("nodes", "leaves", "X", "out", "labels", "labels_all", "ck"),
-            offs)}
+            ("nodes", "roots", "leaves", "X", "out", "labels", "labels_all",
+             "ck"), offs)}
         dev.append(b)
     mark("alloc", t)
 
     t = time.perf_counter()
-    hostgen = os.environ.get("LPX_RF_HOSTGEN") == "1"
+    hostgen = os.environ.get("LPX_RF_HOSTGEN") == "1" and not compact
     if hostgen:
         nodes_np, leaves_np = lpx_rt.random_heap_forest_np(
             n_trees, depth, n_features, n_classes, seed=17)
         lpx_rt.set_device(0)
         lpx_rt.memcpy_h2d(dev[0]["nodes"], nodes_np.ctypes.data, nodes_bytes)
         lpx_rt.memcpy_h2d(dev[0]["leaves"], leaves_np.ctypes.data, leaves_bytes)
+    elif compact:
+        # model generated ON GPU 0, directly in the compact layout
+        lpx_rt.set_device(0)
+        lpx_rt.forestgen_compact(mods[0], dev[0]["nodes"], dev[0]["roots"],
+                                 n_trees, depth, n_features, dev[0]["leaves"],
+                                 n_leaf_rows, n_classes, 17)
     else:
         # model generated ON GPU 0 (random-init weights; no host bytes)
         lpx_rt.set_device(0)
@@ -108,8 +125,15 @@ def main() -> dict:
     if comms:
         comms.broadcast([b["nodes"] for b in dev], nodes_bytes // 4,
                         lpx_rt.NCCL_INT32, root=0)
-        comms.broadcast([b["leaves"] for b in dev], leaves_bytes // 4,
-                        lpx_rt.NCCL_FLOAT32, root=0)
+        if compact:
+            comms.broadcast([b["roots"] for b in dev], roots_bytes // 4,
+                            lpx_rt.NCCL_INT32, root=0)
+            comms.broadcast([b["leaves"] for b in dev],
+                            (leaves_bytes + 3) // 4, lpx_rt.NCCL_INT32,
+                            root=0)
+        else:
+            comms.broadcast([b["leaves"] for b in dev], leaves_bytes // 4,
+                            lpx_rt.NCCL_FLOAT32, root=0)
     mark("model_broadcast", t)
 
     t = time.perf_counter()
@@ -124,9 +148,14 @@ def main() -> dict:
     for d in range(g):
         lpx_rt.set_device(d)
         shard_ptr = dev[d]["X"] + d * batch * n_features * 4
-        lpx_rt.rf_heap(mods[d], dev[d]["nodes"], dev[d]["leaves"], n_trees,
-                       depth, shard_ptr, dev[d]["out"], batch, n_features,
-                       n_classes)
+        if compact:
+            lpx_rt.rf_cpair16(mods[d], dev[d]["nodes"], dev[d]["roots"],
+                              dev[d]["leaves"], n_trees, depth, shard_ptr,
+                              dev[d]["out"], batch, n_features, n_classes)
+        else:
+            lpx_rt.rf_heap(mods[d], dev[d]["nodes"], dev[d]["leaves"], n_trees,
+                           depth, shard_ptr, dev[d]["out"], batch, n_features,
+                           n_classes)
         lpx_rt.argmax_rows(mods[d], dev[d]["out"], dev[d]["labels"], batch,
                            n_classes)
     if comms:
@@ -149,6 +178,7 @@ def main() -> dict:
 
     out = {
         "handler": "rf_lite",
+        "kernel": "cpair16_tilp4" if compact else "heap_f32",
         "world_size": g,
         "model": {"trees": n_trees, "depth": depth, "features": n_features,
                   "classes": n_classes,

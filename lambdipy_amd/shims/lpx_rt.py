@@ -196,6 +196,25 @@ class ForestGenLeavesArgs(ctypes.Structure):
                 ("C", ctypes.c_int), ("seed", ctypes.c_uint)]
 
 
+class ForestGenCPairsArgs(ctypes.Structure):
+    _fields_ = [("cpairs", ctypes.c_void_p), ("croots", ctypes.c_void_p),
+                ("n_trees", ctypes.c_int), ("depth", ctypes.c_int),
+                ("n_features", ctypes.c_int), ("seed", ctypes.c_uint)]
+
+
+class ForestGenLeaves16Args(ctypes.Structure):
+    _fields_ = [("leaves", ctypes.c_void_p), ("n_rows", ctypes.c_long),
+                ("C", ctypes.c_int), ("seed", ctypes.c_uint)]
+
+
+class RfCpair16Args(ctypes.Structure):
+    _fields_ = [("cpairs", ctypes.c_void_p), ("croots", ctypes.c_void_p),
+                ("leaves16", ctypes.c_void_p), ("n_trees", ctypes.c_int),
+                ("depth", ctypes.c_int), ("X", ctypes.c_void_p),
+                ("out", ctypes.c_void_p), ("S", ctypes.c_int),
+                ("F", ctypes.c_int), ("C", ctypes.c_int)]
+
+
 class ChecksumArgs(ctypes.Structure):
     _fields_ = [("x", ctypes.c_void_p), ("out", ctypes.c_void_p),
                 ("n", ctypes.c_long)]
@@ -210,6 +229,36 @@ def forestgen(mod: Module, nodes: int, n_internal_total: int, n_features: int,
     b2 = min((n_leaf_rows + 255) // 256, 4096) or 1
     mod.launch("lpx_forestgen_leaves", b2, 256, 0,
                ForestGenLeavesArgs(leaves, n_leaf_rows, C, seed ^ 0x5bd1e995))
+
+
+def forestgen_compact(mod: Module, cpairs: int, croots: int, n_trees: int,
+                      depth: int, n_features: int, leaves16: int,
+                      n_leaf_rows: int, C: int, seed: int) -> None:
+    """Device-side synthetic forest DIRECTLY in the compact layout the v8
+    traversal wants (u16 feat | fp16 thr pair records, fp16 leaf table) —
+    same node/leaf values as forestgen (shared per-index hash)."""
+    ppt = (1 << (depth - 1)) - 1
+    total = n_trees * ppt
+    b1 = min((total + 255) // 256, 4096) or 1
+    mod.launch("lpx_forestgen_cpairs", b1, 256, 0,
+               ForestGenCPairsArgs(cpairs, croots, n_trees, depth,
+                                   n_features, seed))
+    b2 = min((n_leaf_rows + 255) // 256, 4096) or 1
+    mod.launch("lpx_forestgen_leaves16", b2, 256, 0,
+               ForestGenLeaves16Args(leaves16, n_leaf_rows, C,
+                                     seed ^ 0x5bd1e995))
+
+
+def rf_cpair16(mod: Module, cpairs: int, croots: int, leaves16: int,
+               n_trees: int, depth: int, X: int, out: int, S: int, F: int,
+               C: int) -> None:
+    """Compact child-pair traversal with fp16 leaves + tree-ILP 4 (the
+    serving-rate kernel; mirrors the torch extension's default)."""
+    blocks = min((S + 255) // 256, 4096) or 1
+    shared = 256 * (F + 1) * 4 if F <= 63 else 0
+    mod.launch("lpx_rf_cpair16", blocks, 256, shared,
+               RfCpair16Args(cpairs, croots, leaves16, n_trees, depth, X,
+                             out, S, F, C))
 
 
 def checksum_i32(mod: Module, x: int, out: int, n: int) -> None:

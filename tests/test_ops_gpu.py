@@ -988,3 +988,45 @@ def test_fft_nonpow2_vs_torch(ops, batch, n):
     # inverse round trip
     rt = ops.ifft(got)
     assert (rt - x).abs().max().item() / (x.abs().max().item() + 1e-6) < 1e-3
+
+
+def test_lpx_rt_cpair16_matches_heap(ops):
+    """standalone compact traversal (pair records + fp16 leaves + TILP4)
+    must be BITWISE-equal to the f32 heap kernel on the same device-
+    generated forest (shared per-index hash; leaves are fp16-exact)."""
+    from lambdipy_amd.shims import lpx_rt
+
+    mod = lpx_rt.Module()
+    n_trees, depth, F, C = 33, 9, 24, 8  # odd tree count -> TILP tail
+    stride = (1 << depth) - 1
+    ppt = (1 << (depth - 1)) - 1
+    n_int = n_trees * stride
+    n_rows = n_trees * (1 << depth)
+    nodes = torch.empty(n_int, 2, dtype=torch.int32, device=DEV)
+    leaves = torch.empty(n_rows, C, device=DEV)
+    lpx_rt.forestgen(mod, nodes.data_ptr(), n_int, F, leaves.data_ptr(),
+                     n_rows, C, 71)
+    cpairs = torch.empty(n_trees * ppt, 2, dtype=torch.int32, device=DEV)
+    croots = torch.empty(n_trees, dtype=torch.int32, device=DEV)
+    lv16 = torch.empty(n_rows, C, dtype=torch.int16, device=DEV)
+    lpx_rt.forestgen_compact(mod, cpairs.data_ptr(), croots.data_ptr(),
+                             n_trees, depth, F, lv16.data_ptr(), n_rows, C,
+                             71)
+    lpx_rt.sync()
+    # compact node values must round-trip the heap generator's values
+    heap = nodes.view(n_trees, stride, 2)
+    thr16 = (heap[..., 1].contiguous().view(torch.float32)
+             .to(torch.float16).view(torch.int16).to(torch.int32) & 0xFFFF)
+    cn = (heap[..., 0] << 16) | thr16
+    assert torch.equal(croots, cn[:, 0].to(torch.int32))
+    assert torch.equal(lv16.view(torch.float16).to(torch.float32), leaves)
+    X = _rand((20000, F), seed=72)
+    out_heap = torch.empty(20000, C, device=DEV)
+    out_c = torch.empty(20000, C, device=DEV)
+    lpx_rt.rf_heap(mod, nodes.data_ptr(), leaves.data_ptr(), n_trees, depth,
+                   X.data_ptr(), out_heap.data_ptr(), 20000, F, C)
+    lpx_rt.rf_cpair16(mod, cpairs.data_ptr(), croots.data_ptr(),
+                      lv16.data_ptr(), n_trees, depth, X.data_ptr(),
+                      out_c.data_ptr(), 20000, F, C)
+    lpx_rt.sync()
+    assert torch.equal(out_heap, out_c), "compact standalone path diverges"
