@@ -199,14 +199,22 @@ def cold_start(
         # ~0.1 s of inflate before the spawn, so it stays in the deferred
         # bulk pass; liblpx_nn (~2 MB) IS needed before the first launch.
         if handler in LITE_HANDLERS:
+            # rf_lite needs only lpx_rt + the .hsaco; the NN lite handlers
+            # also need lpx_nn + liblpx_nn (~2 MB — don't inflate it ahead
+            # of a handler that never dlopens it)
+            needs_nn = handler != "rf_lite"
+
             def early(n):
                 base = os.path.basename(n)
                 return (n.startswith("lpx_handlers/") or n == "lpx_fanout.py"
-                        or n == "lpx_rt.py" or n == "lpx_nn.py"
-                        or base.startswith("liblpx_nn")
+                        or n == "lpx_rt.py"
+                        or (needs_nn and n == "lpx_nn.py")
+                        or (needs_nn and base.startswith("liblpx_nn"))
                         or base.endswith(".hsaco")
                         or (n.startswith("lpx_ops/")
-                            and not base.startswith("_lpx_ops")))
+                            and not base.startswith("_lpx_ops")
+                            and not (not needs_nn
+                                     and base.startswith("liblpx_nn"))))
         else:
             def early(n):
                 return n.startswith("lpx_handlers/") or n == "lpx_fanout.py"
