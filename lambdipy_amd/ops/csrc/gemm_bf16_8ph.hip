@@ -277,12 +277,65 @@ __global__ __launch_bounds__(THREADS8, 2) void gemm_bf16_8ph_kernel(
       __builtin_amdgcn_s_setprio(1);
   }
 
+  // REUSE variant (tune 24): fragments live ACROSS phases — phases 0/1
+  // share the same A fragments (rh=0) and phases 0/2 the same B fragments
+  // (ch=0), so each fragment is ds_read ONCE per K-tile: 24 reads/wave/
+  // K-tile instead of 48 (the quadrant scheme re-reads half of them).
+  // Staging/barrier/vmcnt schedule unchanged.
+  bf16frag aR[4][2], bR0[2][2], bR1[2][2];
   for (int t = 0; t < ksteps; ++t) {
     const int parity = t & 1;
 #pragma unroll
     for (int p = 0; p < 4; ++p) {
       const int rh = (p >> 1) & 1;  // quadrant row half
       const int ch = p & 1;        // quadrant col half
+      if constexpr (ILV == 9) {    // REUSE schedule
+        if (p == 0 || p == 2) {
+#pragma unroll
+          for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+            for (int mf = 0; mf < 4; ++mf)
+              aR[mf][ks] = *frag_ptr(smem, parity, rh,
+                                     wsr * 64 + mf * 16 + fr, ks * 4 + fkb);
+        }
+        if (p == 0) {
+#pragma unroll
+          for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+            for (int nf = 0; nf < 2; ++nf)
+              bR0[nf][ks] = *frag_ptr(smem, parity, 2,
+                                      wsc * 32 + nf * 16 + fr, ks * 4 + fkb);
+        } else if (p == 1) {
+#pragma unroll
+          for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+            for (int nf = 0; nf < 2; ++nf)
+              bR1[nf][ks] = *frag_ptr(smem, parity, 3,
+                                      wsc * 32 + nf * 16 + fr, ks * 4 + fkb);
+        }
+        int st9, sh9;
+        stage_slot(t, p, &st9, &sh9);
+        stage_half8(A, Bt, K, bm, bn, st9, sh9, ksteps, smem, w, lane);
+        if constexpr (THR >= 0)
+          asm volatile("s_waitcnt lgkmcnt(%0)" ::"i"(THR) : "memory");
+        __builtin_amdgcn_s_barrier();
+        auto& bC = (ch == 0) ? bR0 : bR1;
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+          for (int mf = 0; mf < 4; ++mf)
+#pragma unroll
+            for (int nf = 0; nf < 2; ++nf)
+              acc[p][mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  aR[mf][ks], bC[nf][ks], acc[p][mf][nf], 0, 0, 0);
+        if constexpr (VM == 60) {
+          if (p & 1) asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+        } else {
+          asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+        }
+        __builtin_amdgcn_s_barrier();
+        continue;
+      }
       // ASM variant (tune 16): one inline-asm block per k-half — 6
       // ds_read_b128 issued back-to-back, then the 8 MFMAs gated by
       // DESCENDING counted lgkmcnt (4/2/1/0: wait only for the operands
@@ -499,6 +552,9 @@ hipError_t gemm_bf16_nt8ph_tune(const uint16_t* A, const uint16_t* Bt,
     case 21: return launch8ph<false, 60, false, 1, 0, 6>(A, Bt, C, M, N, K, stream);
     case 22: return launch8ph<false, 60, false, 1, 0, 10>(A, Bt, C, M, N, K, stream);
     case 23: return launch8ph<false, 8, false, 1, 0, 8>(A, Bt, C, M, N, K, stream);
+    // fragment-reuse schedule: 24 ds_reads/wave/K-tile instead of 48
+    case 24: return launch8ph<false, 60, false, 1, 9, 8>(A, Bt, C, M, N, K, stream);
+    case 25: return launch8ph<false, 60, false, 1, 9, -1>(A, Bt, C, M, N, K, stream);
     // BK=32 half-LDS (64 KiB -> 2 blocks/CU) variant
     case 17: {
       if (K % BK32 != 0) return hipErrorInvalidValue;
