@@ -236,30 +236,40 @@ __global__ __launch_bounds__(THREADS4, 2) void gemm_mx4_8ph_kernel(
 
   for (int t = 0; t < ksteps; ++t) {
     const int parity = t & 1;
+    const int slot3 = t % 3;
+    // scale words hoisted per tile; fragment REUSE across phases (the
+    // bf16/MX schedule): each 16-B fragment ds_read ONCE per K-tile.
+    unsigned awT[2][2], bwT[2];
+#pragma unroll
+    for (int rh = 0; rh < 2; ++rh)
+#pragma unroll
+      for (int mf = 0; mf < 2; ++mf)
+        awT[rh][mf] = scaleword4(smem, slot3, rh, wsr * 64 + mf * 32 + fr);
+#pragma unroll
+    for (int ch = 0; ch < 2; ++ch)
+      bwT[ch] = scaleword4(smem, slot3, 2 + ch, wsc * 32 + fr);
+    i32x8v aR[2][2], bR0[2], bR1[2];
 #pragma unroll
     for (int p = 0; p < 4; ++p) {
       const int rh = (p >> 1) & 1;
       const int ch = p & 1;
-      i32x8v aF[2][2], bF[2];
-      int sa[2][2], sb[2];
-      unsigned aw[2], bw;
-      const int slot3 = t % 3;
+      if (p == 0 || p == 2) {
 #pragma unroll
-      for (int mf = 0; mf < 2; ++mf)
-        aw[mf] = scaleword4(smem, slot3, rh, wsr * 64 + mf * 32 + fr);
-      bw = scaleword4(smem, slot3, 2 + ch, wsc * 32 + fr);
-      auto read_kg = [&](int kg) {
+        for (int kg = 0; kg < 2; ++kg)
 #pragma unroll
-        for (int mf = 0; mf < 2; ++mf) {
-          const int r = wsr * 64 + mf * 32 + fr;
-          aF[mf][kg] = frag4(smem, parity, rh, r, kg, fkb);
-          sa[mf][kg] = (int)((aw[mf] >> (8 * (kg * 2 + fkb))) & 0xff);
-        }
-        const int rb = wsc * 32 + fr;
-        bF[kg] = frag4(smem, parity, 2 + ch, rb, kg, fkb);
-        sb[kg] = (int)((bw >> (8 * (kg * 2 + fkb))) & 0xff);
-      };
-      read_kg(0);  // kg=1 deferred past the barrier (bf16-twin +5%)
+          for (int mf = 0; mf < 2; ++mf)
+            aR[mf][kg] = frag4(smem, parity, rh, wsr * 64 + mf * 32 + fr,
+                               kg, fkb);
+      }
+      if (p == 0) {
+#pragma unroll
+        for (int kg = 0; kg < 2; ++kg)
+          bR0[kg] = frag4(smem, parity, 2, wsc * 32 + fr, kg, fkb);
+      } else if (p == 1) {
+#pragma unroll
+        for (int kg = 0; kg < 2; ++kg)
+          bR1[kg] = frag4(smem, parity, 3, wsc * 32 + fr, kg, fkb);
+      }
       int st, sh;
       stage_slot4(t, p, &st, &sh);
       stage_half4(A, Bt, Kb, bm, bn, st, sh, ksteps, smem, t512);
@@ -267,13 +277,18 @@ __global__ __launch_bounds__(THREADS4, 2) void gemm_mx4_8ph_kernel(
         stage_scales4(sA, sB, M, N, bm, bn, t + 2, ksteps, smem, w, lane);
 
       __builtin_amdgcn_s_barrier();
-      read_kg(1);
+      auto& bC = (ch == 0) ? bR0 : bR1;
+      const unsigned* aw = awT[rh];
+      const unsigned bw = bwT[ch];
 #pragma unroll
       for (int kg = 0; kg < 2; ++kg)
 #pragma unroll
-        for (int mf = 0; mf < 2; ++mf)
+        for (int mf = 0; mf < 2; ++mf) {
+          const int sa = (int)((aw[mf] >> (8 * (kg * 2 + fkb))) & 0xff);
+          const int sb = (int)((bw >> (8 * (kg * 2 + fkb))) & 0xff);
           acc[p][mf] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
-              aF[mf][kg], bF[kg], acc[p][mf], 4, 4, 0, sa[mf][kg], 0, sb[kg]);
+              aR[mf][kg], bC[kg], acc[p][mf], 4, 4, 0, sa, 0, sb);
+        }
       waitvm4<5>();  // constant: window [2,1,1,1]; scales 8 calls back
       __builtin_amdgcn_s_barrier();
     }
@@ -617,6 +632,7 @@ __global__ __launch_bounds__(THREADS4, 1) void gemm_mx4_3p_kernel(
   if (__builtin_amdgcn_readfirstlane(threadIdx.x) >= 256)
     __builtin_amdgcn_s_setprio(1);
 
+  i32x8v aR3[2][2], bR30[2], bR31[2];
   for (int t = 0; t < ksteps; ++t) {
     const int par3 = t % 3;
     const int slot3 = t % 3;
@@ -634,25 +650,30 @@ __global__ __launch_bounds__(THREADS4, 1) void gemm_mx4_3p_kernel(
       else if (p == 2) waitvm3p<10>();
       __builtin_amdgcn_s_barrier();
 
-      i32x8v aF[2][2], bF[2];
-      int sa[2][2], sb[2];
+      // fragment reuse (see the 2-parity kernel): A pairs shared across
+      // the column-half phases, B pairs across the row-half phases
+      if (p == 0 || p == 2) {
+#pragma unroll
+        for (int kg = 0; kg < 2; ++kg)
+#pragma unroll
+          for (int mf = 0; mf < 2; ++mf)
+            aR3[mf][kg] = frag4_3p(smem, par3, rh,
+                                   wsr * 64 + mf * 32 + fr, kg, fkb);
+      }
+      if (p == 0) {
+#pragma unroll
+        for (int kg = 0; kg < 2; ++kg)
+          bR30[kg] = frag4_3p(smem, par3, 2, wsc * 32 + fr, kg, fkb);
+      } else if (p == 1) {
+#pragma unroll
+        for (int kg = 0; kg < 2; ++kg)
+          bR31[kg] = frag4_3p(smem, par3, 3, wsc * 32 + fr, kg, fkb);
+      }
       unsigned aw[2], bw;
 #pragma unroll
       for (int mf = 0; mf < 2; ++mf)
         aw[mf] = scaleword4_3p(smem, slot3, rh, wsr * 64 + mf * 32 + fr);
       bw = scaleword4_3p(smem, slot3, 2 + ch, wsc * 32 + fr);
-#pragma unroll
-      for (int kg = 0; kg < 2; ++kg) {
-#pragma unroll
-        for (int mf = 0; mf < 2; ++mf) {
-          const int r = wsr * 64 + mf * 32 + fr;
-          aF[mf][kg] = frag4_3p(smem, par3, rh, r, kg, fkb);
-          sa[mf][kg] = (int)((aw[mf] >> (8 * (kg * 2 + fkb))) & 0xff);
-        }
-        const int rb = wsc * 32 + fr;
-        bF[kg] = frag4_3p(smem, par3, 2 + ch, rb, kg, fkb);
-        sb[kg] = (int)((bw >> (8 * (kg * 2 + fkb))) & 0xff);
-      }
       // staging schedule: p0 (t+2,B-hi)+sc(t+2), p1 (t+2,A-hi),
       // p2 (t+3,A-lo), p3 (t+3,B-lo) — (t+3) lands in parity t%3 whose
       // lo slots were last read at p1/p2 (>=1 barrier earlier)
@@ -666,12 +687,16 @@ __global__ __launch_bounds__(THREADS4, 1) void gemm_mx4_3p_kernel(
       } else {
         stage_half4_3p(A, Bt, Kb, bm, bn, t + 3, 2, ksteps, smem, t512);
       }
+      auto& bC3 = (ch == 0) ? bR30 : bR31;
 #pragma unroll
       for (int kg = 0; kg < 2; ++kg)
 #pragma unroll
-        for (int mf = 0; mf < 2; ++mf)
+        for (int mf = 0; mf < 2; ++mf) {
+          const int sa = (int)((aw[mf] >> (8 * (kg * 2 + fkb))) & 0xff);
+          const int sb = (int)((bw >> (8 * (kg * 2 + fkb))) & 0xff);
           acc[p][mf] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
-              aF[mf][kg], bF[kg], acc[p][mf], 4, 4, 0, sa[mf][kg], 0, sb[kg]);
+              aR3[mf][kg], bC3[kg], acc[p][mf], 4, 4, 0, sa, 0, sb);
+        }
     }
   }
   waitvm3p<0>();

@@ -208,27 +208,36 @@ __global__ __launch_bounds__(THREADSX, 1) void gemm_mx_8ph_kernel(
       for (int ch = 0; ch < 2; ++ch)
         bwT[ch] = scaleword(smem, slot3, 2 + ch, wsc * 32 + fr);
     }
+    // fragment REUSE (the bf16 tune-25 schedule, +6-7% there): A fragment
+    // pairs are shared across the column-half phases and B pairs across
+    // the row-half phases — each 32-B fragment is ds_read ONCE per K-tile
+    // (12 fragment reads/wave/K-tile instead of 24). Staging overlap is
+    // unchanged and still safe: A-lo/B-lo's LAST LDS reads move EARLIER
+    // (phase 0), further from their p2/p3 restaging.
+    i32x8v aR[2][2], bR0[2], bR1[2];
 #pragma unroll
     for (int p = 0; p < 4; ++p) {
       const int rh = (p >> 1) & 1;
       const int ch = p & 1;
-      i32x8v aF[2][2], bF[2];
-      int sa[2][2], sb[2];
+      if (p == 0 || p == 2) {
+#pragma unroll
+        for (int kg = 0; kg < 2; ++kg)
+#pragma unroll
+          for (int mf = 0; mf < 2; ++mf)
+            aR[mf][kg] = fragx(smem, parity, rh, wsr * 64 + mf * 32 + fr,
+                               kg, fkb);
+      }
+      if (p == 0) {
+#pragma unroll
+        for (int kg = 0; kg < 2; ++kg)
+          bR0[kg] = fragx(smem, parity, 2, wsc * 32 + fr, kg, fkb);
+      } else if (p == 1) {
+#pragma unroll
+        for (int kg = 0; kg < 2; ++kg)
+          bR1[kg] = fragx(smem, parity, 3, wsc * 32 + fr, kg, fkb);
+      }
       const unsigned* aw = awT[rh];
       const unsigned bw = bwT[ch];
-      auto read_kg = [&](int kg) {
-#pragma unroll
-        for (int mf = 0; mf < 2; ++mf) {
-          const int r = wsr * 64 + mf * 32 + fr;
-          aF[mf][kg] = fragx(smem, parity, rh, r, kg, fkb);
-          sa[mf][kg] = SC ? (int)((aw[mf] >> (8 * (kg * 2 + fkb))) & 0xff) : 127;
-        }
-        const int rb = wsc * 32 + fr;
-        bF[kg] = fragx(smem, parity, 2 + ch, rb, kg, fkb);
-        sb[kg] = SC ? (int)((bw >> (8 * (kg * 2 + fkb))) & 0xff) : 127;
-      };
-      read_kg(0);  // kg=1 reads deferred past the barrier (latency hides
-                   // under the kg=0 MFMAs — measured +5% on the bf16 twin)
       int st, sh;
       stage_slotx(t, p, &st, &sh);
       stage_halfx(A, Bt, K, bm, bn, st, sh, ksteps, smem, w, lane);
@@ -238,13 +247,18 @@ __global__ __launch_bounds__(THREADSX, 1) void gemm_mx_8ph_kernel(
           stage_scalesx(sA, sB, M, N, bm, bn, t + 2, ksteps, smem, w, lane);
 
       __builtin_amdgcn_s_barrier();
-      read_kg(1);
+      auto& bC = (ch == 0) ? bR0 : bR1;
 #pragma unroll
       for (int kg = 0; kg < 2; ++kg)
 #pragma unroll
-        for (int mf = 0; mf < 2; ++mf)
+        for (int mf = 0; mf < 2; ++mf) {
+          const int sa = SC ? (int)((aw[mf] >> (8 * (kg * 2 + fkb))) & 0xff)
+                            : 127;
+          const int sb = SC ? (int)((bw >> (8 * (kg * 2 + fkb))) & 0xff)
+                            : 127;
           acc[p][mf] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
-              aF[mf][kg], bF[kg], acc[p][mf], 0, 0, 0, sa[mf][kg], 0, sb[kg]);
+              aR[mf][kg], bC[kg], acc[p][mf], 0, 0, 0, sa, 0, sb);
+        }
       // SC: per-tile-cycle call events are [3,2,2,2]; any 4-call window
       // holds one phase-0, so 9 is constant. Tile t+1's scales were staged
       // at p0 of t-1 (8 calls back, outside the window) — tri-buffered
