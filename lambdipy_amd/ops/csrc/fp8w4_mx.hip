@@ -206,24 +206,31 @@ __global__ __launch_bounds__(THREADSW, 1) void gemm_w4_kernel(
       for (int ch = 0; ch < 2; ++ch)
         bwT[ch] = scw(smem, slot3, 256 + ch * 128 + wsc * 32 + fr);
     }
+    // fragment REUSE (the bf16/MX schedule): each fragment ds_read once
+    // per K-tile — A pairs shared across column-half phases, B across
+    // row-half phases.
+    i32x8v aR[2][2], bR0[2], bR1[2];
 #pragma unroll
     for (int p = 0; p < 4; ++p) {
       const int rh = (p >> 1) & 1;
       const int ch = p & 1;
-      i32x8v aF[2][2], bF[2];
-      int sa[2][2], sb[2];
-      auto read_kg = [&](int kg) {
+      if (p == 0 || p == 2) {
 #pragma unroll
-        for (int mf = 0; mf < 2; ++mf) {
-          const int r = wsr * 64 + mf * 32 + fr;
-          aF[mf][kg] = frag_af(smem, parity, rh, r, kg, fkb);
-          sa[mf][kg] = (int)((awT[rh][mf] >> (8 * (kg * 2 + fkb))) & 0xff);
-        }
-        const int rb = wsc * 32 + fr;
-        bF[kg] = frag_bf(smem, parity, ch, rb, kg, fkb);
-        sb[kg] = (int)((bwT[ch] >> (8 * (kg * 2 + fkb))) & 0xff);
-      };
-      read_kg(0);
+        for (int kg = 0; kg < 2; ++kg)
+#pragma unroll
+          for (int mf = 0; mf < 2; ++mf)
+            aR[mf][kg] = frag_af(smem, parity, rh,
+                                 wsr * 64 + mf * 32 + fr, kg, fkb);
+      }
+      if (p == 0) {
+#pragma unroll
+        for (int kg = 0; kg < 2; ++kg)
+          bR0[kg] = frag_bf(smem, parity, 0, wsc * 32 + fr, kg, fkb);
+      } else if (p == 1) {
+#pragma unroll
+        for (int kg = 0; kg < 2; ++kg)
+          bR1[kg] = frag_bf(smem, parity, 1, wsc * 32 + fr, kg, fkb);
+      }
       int st, sh;
       slot_w(t, p, &st, &sh);
       stage_w(A, Bt, K, bm, bn, st, sh, ksteps, smem, w, lane);
@@ -231,13 +238,16 @@ __global__ __launch_bounds__(THREADSW, 1) void gemm_w4_kernel(
         stage_scw(sA, sB, M, N, bm, bn, t + 2, ksteps, smem, w, lane);
 
       __builtin_amdgcn_s_barrier();
-      read_kg(1);
+      auto& bC = (ch == 0) ? bR0 : bR1;
 #pragma unroll
       for (int kg = 0; kg < 2; ++kg)
 #pragma unroll
-        for (int mf = 0; mf < 2; ++mf)
+        for (int mf = 0; mf < 2; ++mf) {
+          const int sa = (int)((awT[rh][mf] >> (8 * (kg * 2 + fkb))) & 0xff);
+          const int sb = (int)((bwT[ch] >> (8 * (kg * 2 + fkb))) & 0xff);
           acc[p][mf] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
-              aF[mf][kg], bF[kg], acc[p][mf], 0, 4, 0, sa[mf][kg], 0, sb[kg]);
+              aR[mf][kg], bC[kg], acc[p][mf], 0, 4, 0, sa, 0, sb);
+        }
       waitvw<7>();  // constant: every 4-call window sums to 7
       __builtin_amdgcn_s_barrier();
     }
