@@ -36,7 +36,7 @@ def test_gemm_bf16_floor():
     b = torch.randn(n, n, device=DEV, dtype=torch.bfloat16)
     dt = _time_op(lambda: ops.gemm_bf16(a, b))
     tf = 2 * n**3 / dt / 1e12
-    assert tf > 800, f"bf16 GEMM at {tf:.0f} TF (floor 800; measured ~1100-1150)"
+    assert tf > 900, f"bf16 GEMM at {tf:.0f} TF (floor 900; measured 1100-1210 with the fragment-reuse schedule)"
 
 
 def test_rf_traversal_floor():
