@@ -282,21 +282,31 @@ __global__ __launch_bounds__(THREADS8, 2) void gemm_bf16_8ph_kernel(
   // (ch=0), so each fragment is ds_read ONCE per K-tile: 24 reads/wave/
   // K-tile instead of 48 (the quadrant scheme re-reads half of them).
   // Staging/barrier/vmcnt schedule unchanged.
-  bf16frag aR[4][2], bR0[2][2], bR1[2][2];
+  bf16frag aR[4][2], aR2[4][2], bR0[2][2], bR1[2][2];
   for (int t = 0; t < ksteps; ++t) {
     const int parity = t & 1;
 #pragma unroll
     for (int p = 0; p < 4; ++p) {
       const int rh = (p >> 1) & 1;  // quadrant row half
       const int ch = p & 1;        // quadrant col half
-      if constexpr (ILV == 9) {    // REUSE schedule
-        if (p == 0 || p == 2) {
+      if constexpr (ILV == 9 || ILV == 10) {  // REUSE schedule
+        // ILV 10: balance the read schedule — prefetch the rh=1 A frags
+        // during the light phase 1 (reads per phase 12/12/0/0 vs 12/4/8/0)
+        if (p == 0 || (ILV == 9 && p == 2)) {
 #pragma unroll
           for (int ks = 0; ks < 2; ++ks)
 #pragma unroll
             for (int mf = 0; mf < 4; ++mf)
               aR[mf][ks] = *frag_ptr(smem, parity, rh,
                                      wsr * 64 + mf * 16 + fr, ks * 4 + fkb);
+        }
+        if (ILV == 10 && p == 1) {
+#pragma unroll
+          for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+            for (int mf = 0; mf < 4; ++mf)
+              aR2[mf][ks] = *frag_ptr(smem, parity, 1,
+                                      wsr * 64 + mf * 16 + fr, ks * 4 + fkb);
         }
         if (p == 0) {
 #pragma unroll
@@ -320,6 +330,7 @@ __global__ __launch_bounds__(THREADS8, 2) void gemm_bf16_8ph_kernel(
           asm volatile("s_waitcnt lgkmcnt(%0)" ::"i"(THR) : "memory");
         __builtin_amdgcn_s_barrier();
         auto& bC = (ch == 0) ? bR0 : bR1;
+        auto& aC = (ILV == 10 && rh == 1) ? aR2 : aR;
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks)
 #pragma unroll
@@ -327,7 +338,7 @@ __global__ __launch_bounds__(THREADS8, 2) void gemm_bf16_8ph_kernel(
 #pragma unroll
             for (int nf = 0; nf < 2; ++nf)
               acc[p][mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                  aR[mf][ks], bC[nf][ks], acc[p][mf][nf], 0, 0, 0);
+                  aC[mf][ks], bC[nf][ks], acc[p][mf][nf], 0, 0, 0);
         if constexpr (VM == 60) {
           if (p & 1) asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
         } else {
@@ -555,6 +566,8 @@ hipError_t gemm_bf16_nt8ph_tune(const uint16_t* A, const uint16_t* Bt,
     // fragment-reuse schedule: 24 ds_reads/wave/K-tile instead of 48
     case 24: return launch8ph<false, 60, false, 1, 9, 8>(A, Bt, C, M, N, K, stream);
     case 25: return launch8ph<false, 60, false, 1, 9, -1>(A, Bt, C, M, N, K, stream);
+    // reuse + balanced read prefetch (rh1 A frags read during phase 1)
+    case 26: return launch8ph<false, 60, false, 1, 10, -1>(A, Bt, C, M, N, K, stream);
     // BK=32 half-LDS (64 KiB -> 2 blocks/CU) variant
     case 17: {
       if (K % BK32 != 0) return hipErrorInvalidValue;
