@@ -193,3 +193,44 @@ def test_fuzz_rf_heap(trees, depth, feats, classes, samples, seed):
     ref = fc.predict_proba_t(X)
     assert torch.allclose(proba, ref, atol=1e-5), \
         f"t={trees} d={depth} f={feats} c={classes} s={samples}"
+
+
+# ---- MX block-scaled GEMMs (guards the fragment-reuse kernels) -----------
+
+@settings(**GPU_SETTINGS)
+@given(mt=st.integers(1, 5), nt=st.integers(1, 5), kt=st.integers(1, 6),
+       seed=st.integers(0, 10_000))
+def test_fuzz_gemm_mx_fp8(ops, mt, nt, kt, seed):
+    from tests.test_ops_gpu import _dequant_mx
+
+    e = ops.ext()
+    m, n, k = mt * 256, nt * 256, kt * 128
+    A = _rand((m, k), seed=seed) * 2
+    Bt = _rand((n, k), seed=seed + 3)
+    qa, sa = e.quantize_mx(A)
+    qb, sb = e.quantize_mx(Bt)
+    C = e.gemm_mx(qa, sa, qb, sb)
+    C2 = e.gemm_mx(qa, sa, qb, sb)
+    assert torch.equal(C, C2), "nondeterministic MX GEMM"
+    ref = _dequant_mx(qa, sa) @ _dequant_mx(qb, sb).t()
+    err = (C - ref).abs().max().item() / (ref.abs().max().item() + 1e-6)
+    assert err < 1e-3, f"({m},{n},{k}) rel {err}"
+
+
+@settings(**GPU_SETTINGS)
+@given(mt=st.integers(1, 4), nt=st.integers(1, 4), kt=st.integers(1, 6),
+       seed=st.integers(0, 10_000))
+def test_fuzz_gemm_mx_fp4(ops, mt, nt, kt, seed):
+    e = ops.ext()
+    m, n, k = mt * 256, nt * 256, kt * 128
+    A = _rand((m, k), seed=seed)
+    Bt = _rand((n, k), seed=seed + 5)
+    qa, sa = e.quantize_mx4(A)
+    qb, sb = e.quantize_mx4(Bt)
+    C = e.gemm_mx4(qa, sa, qb, sb)
+    C2 = e.gemm_mx4(qa, sa, qb, sb)
+    assert torch.equal(C, C2), "nondeterministic fp4 GEMM"
+    # fp4 e2m1 quantization noise floor: Frobenius-relative vs exact f32
+    exact = A @ Bt.t()
+    rel = (C - exact).norm().item() / (exact.norm().item() + 1e-6)
+    assert rel < 0.25, f"({m},{n},{k}) fp4 rel {rel}"
