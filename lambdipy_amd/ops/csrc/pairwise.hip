@@ -477,6 +477,224 @@ __global__ __launch_bounds__(THREADS3) void pairwise4_kernel(
   }
 }
 
+// ---- v5/v6: bf16-split emulation (Ozaki-style) ---------------------------
+// The f32-input MFMA runs at the f32 VECTOR rate (64 FLOP/clk/SIMD, 157 TF
+// peak — MI355X_MICROARCH.md 'FP32-input MFMA'), which is why v1-v4 plateau
+// at 52-87 TF across four geometries.  The bf16 MFMA pipe is 16x faster
+// (2.5 PF), so emulate the f32 gram matrix there: split each operand value
+// exactly into bf16 parts  x = hi + mid (+ lo),  hi = trunc16(x),
+// mid = trunc16(x - hi), lo = trunc16(x - hi - mid) — truncation makes every
+// subtraction exact — and accumulate the significant cross products with
+// v_mfma_f32_16x16x32_bf16 (f32 accumulate):
+//   TERMS=4 (v5): hh + hm + mh + mm      -> |err| ~ 2^-16 |x||y| per term
+//   TERMS=6 (v6): + hl + lh              -> |err| ~ 2^-24  (f32-grade)
+// Effective ceiling = bf16 rate / TERMS = 625 / 417 TF — both far above the
+// 157 TF exact-f32 ceiling.  Splitting happens ONCE per staged tile (shared
+// by all consumer waves), not per fragment read.
+// Geometry: 256^2 tile, 512 threads (8 waves 2x4, 128x64 per wave), BK=32.
+// LDS: 16-KiB bf16 tiles [A_hi A_mid (A_lo) B_hi B_mid (B_lo)]; v5 double-
+// buffers (4 tiles x2 = 128 KiB, 1 barrier/K-step), v6 single-buffers
+// (6 tiles = 96 KiB, consume-barrier-commit-barrier).
+// LDS cells are rotated within each 128-B line (2 rows/line, period-8) so a
+// fragment read's 16 rows hit every bank quad exactly twice — the optimal
+// 2-way for a 1-KiB ds_read_b128.
+typedef __attribute__((ext_vector_type(4))) unsigned int u32x4pw;
+
+constexpr int BM5 = 256, BK5 = 32;
+constexpr int PW5_TILE = 256 * 4 * 16;  // bytes: 256 rows x 4 16-B cells
+
+// physical byte offset of (row, cell) under the 128-B-line rotation
+__device__ __forceinline__ int pw5_cell(int r, int c) {
+  return (r >> 1) * 8 + (((r & 1) * 4 + c + (r >> 1)) & 7);
+}
+
+// global->regs for this thread's two cells (zero-filled at the edges)
+__device__ __forceinline__ void pw5_load(const float* __restrict__ src,
+                                         int rows_total, int K, int row0,
+                                         int k0, float (&v)[2][8]) {
+  const int t = threadIdx.x;
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    const int q = t + i * 512;  // physical 16-B slot 0..1023
+    const int line = q >> 3;
+    const int lidx = ((q & 7) - line) & 7;  // un-rotate
+    const int r = line * 2 + (lidx >> 2), c = lidx & 3;
+    const int grow = row0 + r, gk = k0 + c * 8;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) v[i][j] = 0.f;
+    if (grow < rows_total) {
+      const float* g = src + (long)grow * K + gk;
+      if (gk + 8 <= K) {
+        *reinterpret_cast<f32x4*>(&v[i][0]) = *reinterpret_cast<const f32x4*>(g);
+        *reinterpret_cast<f32x4*>(&v[i][4]) =
+            *reinterpret_cast<const f32x4*>(g + 4);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          if (gk + j < K) v[i][j] = g[j];
+      }
+    }
+  }
+}
+
+// split regs -> bf16 tiles (hi at `tile0`, mid at +PW5_TILE, lo at +2*)
+template <int NS>
+__device__ __forceinline__ void pw5_commit(float (&v)[2][8],
+                                           char* __restrict__ tile0) {
+  const int t = threadIdx.x;
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    const int q = t + i * 512;
+    unsigned hw[4], mw[4], lw[4];
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const float x0 = v[i][2 * j], x1 = v[i][2 * j + 1];
+      const unsigned u0 = __builtin_bit_cast(unsigned, x0);
+      const unsigned u1 = __builtin_bit_cast(unsigned, x1);
+      const float h0 = __builtin_bit_cast(float, u0 & 0xFFFF0000u);
+      const float h1 = __builtin_bit_cast(float, u1 & 0xFFFF0000u);
+      const float r0 = x0 - h0, r1 = x1 - h1;  // exact
+      const unsigned m0 = __builtin_bit_cast(unsigned, r0);
+      const unsigned m1 = __builtin_bit_cast(unsigned, r1);
+      hw[j] = (u0 >> 16) | (u1 & 0xFFFF0000u);
+      mw[j] = (m0 >> 16) | (m1 & 0xFFFF0000u);
+      if (NS == 3) {
+        const float q0 = r0 - __builtin_bit_cast(float, m0 & 0xFFFF0000u);
+        const float q1 = r1 - __builtin_bit_cast(float, m1 & 0xFFFF0000u);
+        lw[j] = (__builtin_bit_cast(unsigned, q0) >> 16) |
+                (__builtin_bit_cast(unsigned, q1) & 0xFFFF0000u);
+      }
+    }
+    char* const p = tile0 + q * 16;
+    *reinterpret_cast<u32x4pw*>(p) = u32x4pw{hw[0], hw[1], hw[2], hw[3]};
+    *reinterpret_cast<u32x4pw*>(p + PW5_TILE) =
+        u32x4pw{mw[0], mw[1], mw[2], mw[3]};
+    if (NS == 3)
+      *reinterpret_cast<u32x4pw*>(p + 2 * PW5_TILE) =
+          u32x4pw{lw[0], lw[1], lw[2], lw[3]};
+  }
+}
+
+template <int TERMS>
+__global__ __launch_bounds__(THREADS3, 1) void pairwise5_kernel(
+    const float* __restrict__ X, const float* __restrict__ Y,
+    const float* __restrict__ xsq, const float* __restrict__ ysq,
+    float* __restrict__ D, int M, int N, int K, unsigned tiles_m,
+    unsigned tiles_n) {
+  constexpr int NS = (TERMS == 4) ? 2 : 3;   // splits staged per operand
+  constexpr bool DB = (TERMS == 4);          // double-buffered?
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  constexpr int PAR = 2 * NS * PW5_TILE;     // A tiles then B tiles
+
+  const unsigned nwg = tiles_m * tiles_n;
+  const unsigned bid = lpx_xcd_swizzle(blockIdx.x, nwg);
+  const int bm = (int)(bid / tiles_n) * BM5;
+  const int bn = (int)(bid % tiles_n) * BM5;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wr = wid >> 2, wc = wid & 3;  // 2x4 waves; 128x64 per wave
+  const int fr = lane & 15, fkb = lane >> 4;
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  float xv[2][8], yv[2][8];
+  pw5_load(X, M, K, bm, 0, xv);
+  pw5_load(Y, N, K, bn, 0, yv);
+  pw5_commit<NS>(xv, smem);
+  pw5_commit<NS>(yv, smem + NS * PW5_TILE);
+  __syncthreads();
+
+  const int ksteps = (K + BK5 - 1) / BK5;
+  int cur = 0;
+  for (int ks = 0; ks < ksteps; ++ks) {
+    char* const abase = smem + (DB ? cur * PAR : 0);
+    char* const bbase = abase + NS * PW5_TILE;
+    const bool more = ks + 1 < ksteps;
+    if (more) {  // prefetch next tile into regs while this one computes
+      pw5_load(X, M, K, bm, (ks + 1) * BK5, xv);
+      pw5_load(Y, N, K, bn, (ks + 1) * BK5, yv);
+    }
+
+    bf16x8 bh[4], bm_[4], bl[4];
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      const int br = wc * 64 + nf * 16 + fr;
+      const int off = pw5_cell(br, fkb) * 16;
+      bh[nf] = *reinterpret_cast<const bf16x8*>(bbase + off);
+      bm_[nf] = *reinterpret_cast<const bf16x8*>(bbase + PW5_TILE + off);
+      if constexpr (TERMS == 6)
+        bl[nf] = *reinterpret_cast<const bf16x8*>(bbase + 2 * PW5_TILE + off);
+    }
+#pragma unroll
+    for (int mf = 0; mf < 8; ++mf) {
+      const int ar = wr * 128 + mf * 16 + fr;
+      const int off = pw5_cell(ar, fkb) * 16;
+      const bf16x8 ah = *reinterpret_cast<const bf16x8*>(abase + off);
+      const bf16x8 am =
+          *reinterpret_cast<const bf16x8*>(abase + PW5_TILE + off);
+      // term-major: each acc is touched TERMS times at distance 4 MFMAs
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf)
+        acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            ah, bh[nf], acc[mf][nf], 0, 0, 0);
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf)
+        acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            ah, bm_[nf], acc[mf][nf], 0, 0, 0);
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf)
+        acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            am, bh[nf], acc[mf][nf], 0, 0, 0);
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf)
+        acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            am, bm_[nf], acc[mf][nf], 0, 0, 0);
+      if constexpr (TERMS == 6) {
+        const bf16x8 al =
+            *reinterpret_cast<const bf16x8*>(abase + 2 * PW5_TILE + off);
+#pragma unroll
+        for (int nf = 0; nf < 4; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              ah, bl[nf], acc[mf][nf], 0, 0, 0);
+#pragma unroll
+        for (int nf = 0; nf < 4; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              al, bh[nf], acc[mf][nf], 0, 0, 0);
+      }
+    }
+    if (more) {
+      if constexpr (!DB) __syncthreads();  // consumers done before overwrite
+      pw5_commit<NS>(xv, smem + (DB ? (cur ^ 1) * PAR : 0));
+      pw5_commit<NS>(yv, smem + (DB ? (cur ^ 1) * PAR : 0) + NS * PW5_TILE);
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  const int orow_l = (lane >> 4) * 4;
+  const int ocol_l = lane & 15;
+#pragma unroll
+  for (int mf = 0; mf < 8; ++mf) {
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = bm + wr * 128 + mf * 16 + orow_l + r;
+        const int col = bn + wc * 64 + nf * 16 + ocol_l;
+        if (row < M && col < N) {
+          float d = xsq[row] - 2.0f * acc[mf][nf][r] + ysq[col];
+          D[(long)row * N + col] = d > 0.f ? d : 0.f;
+        }
+      }
+    }
+  }
+}
+
 __global__ __launch_bounds__(THREADS) void row_sqnorm_kernel(
     const float* __restrict__ X, float* __restrict__ out, int M, long K) {
   const int lane = threadIdx.x & 63;
@@ -526,6 +744,18 @@ hipError_t pairwise_sqeuclidean_f32(const float* X, const float* Y,
     const size_t lds = sizeof(float) * 4u * BM3 * ROWF3;
     pairwise3_kernel<<<dim3(tm * tn), dim3(THREADS3), lds, stream>>>(
         X, Y, xsq, ysq, D, M, N, K, tm, tn);
+    return hipGetLastError();
+  }
+  if ((M >= 256 && N >= 256) && pv && (pv[0] == '5' || pv[0] == '6')) {
+    // v5/v6: bf16-split emulation (4/6 cross products on the bf16 MFMA pipe)
+    const unsigned tm = lpx_ceil_div(M, BM5), tn = lpx_ceil_div(N, BM5);
+    const size_t lds = (pv[0] == '5') ? 4u * PW5_TILE * 2 : 6u * PW5_TILE;
+    if (pv[0] == '5')
+      pairwise5_kernel<4><<<dim3(tm * tn), dim3(THREADS3), lds, stream>>>(
+          X, Y, xsq, ysq, D, M, N, K, tm, tn);
+    else
+      pairwise5_kernel<6><<<dim3(tm * tn), dim3(THREADS3), lds, stream>>>(
+          X, Y, xsq, ysq, D, M, N, K, tm, tn);
     return hipGetLastError();
   }
   if ((M >= 256 && N >= 256) && (pv && pv[0] == '4')) {

@@ -1,4 +1,5 @@
-"""A/B: pairwise v1 (16x16x4, 64^2 tile) vs v2 (32x32x2, 128^2 tile)."""
+"""A/B: pairwise variants — v1 (16x16x4 64^2), v3/v4 (256^2 f32 MFMA),
+v5/v6 (bf16-split emulation, 4/6 cross terms on the bf16 pipe)."""
 import os
 import sys
 import time
@@ -27,7 +28,7 @@ def main():
         X = torch.randn(m, k, device="cuda")
         Y = torch.randn(n, k, device="cuda")
         ref = None
-        for env, name in ((None, "v1"), ("3", "v3"), ("4", "v4")):
+        for env, name in ((None, "v1"), ("4", "v4"), ("5", "v5"), ("6", "v6")):
             if env is None:
                 os.environ.pop("LPX_PW_V2", None)
             else:
@@ -39,7 +40,8 @@ def main():
             ok = err < 1e-2 * (ref.max().item() + 1)
             dt = timeit(lambda: ops.pairwise_sqeuclidean(X, Y))
             tf = 2.0 * m * n * k / dt / 1e12
-            print(f"{name} m={m} n={n} k={k}: {dt*1e3:7.3f} ms {tf:6.1f} TF ok={ok}")
+            print(f"{name} m={m} n={n} k={k}: {dt*1e3:7.3f} ms {tf:6.1f} TF "
+                  f"ok={ok} maxerr={err:.3e}")
         os.environ.pop("LPX_PW_V2", None)
 
 
