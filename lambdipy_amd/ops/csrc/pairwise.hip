@@ -2,12 +2,15 @@
 //
 // D[i,j] = ||X[i]||^2 - 2*X[i].Y[j] + ||Y[j]||^2
 //
-// Design: GEMM-shaped. The -2ab term runs on the f32-input
-// MFMA (v_mfma_f32_16x16x4_f32 — exact f32 at the 157 TF vector-f32 rate,
-// guide §3 'FP32-input MFMA'), with row norms fused in the epilogue. 64x64
-// tile per 256-thread block, BK=16, +1-float padded LDS rows (conflict-free
-// ds_read_b32 — guide G4). sklearn.metrics.pairwise parity: exact f32
-// numerics (bitwise an fmaf chain), no bf16 downcast.
+// Design: GEMM-shaped. Two families:
+//  * v1-v4: the -2ab term on the f32-input MFMA (v_mfma_f32_16x16x4_f32 —
+//    exact f32, but capped at the 157 TF vector-f32 issue rate, guide §3
+//    'FP32-input MFMA'), row norms fused in the epilogue.
+//  * v5/v6 (default for large long-K shapes): bf16-split emulation — exact
+//    truncation splits x = hi + mid (+ lo) and 4/6 cross products on the
+//    16x bf16 MFMA pipe.  v6 keeps hi*lo and is f32-GRADE (measured maxerr
+//    equal to v1's vs float64); 129-157 TF where v1-v4 plateau at 52-87.
+// sklearn.metrics.pairwise parity: f32-accuracy numerics at every shape.
 
 #include "lpx_common.h"
 
@@ -722,14 +725,18 @@ hipError_t pairwise_sqeuclidean_f32(const float* X, const float* Y,
                                     const float* xsq, const float* ysq,
                                     float* D, int M, int N, int K,
                                     hipStream_t stream) {
-  // Variant ladder (ab_pairwise.py, r2 within-probe): the kernel family is
-  // ds_read/MFMA-latency-bound, not HBM-bound — big tiles cut staging 4x
-  // yet only win where the K loop is long enough to amortize their lower
-  // occupancy. Measured: v1 63-82 TF, v2 52-74, v3 51-85, v4 52-87.
-  // Dispatch: v4 (256^2, 16x16x4, 2 blocks/CU) for long-K large shapes,
-  // v1 (64^2, 4 blocks/CU) otherwise; LPX_PW_V2=1/3/4 force a variant.
+  // Variant ladder (ab_pairwise.py, r2 within-probe): the exact-f32 family
+  // (v1-v4) is capped by the f32 MFMA issue rate (157 TF peak; measured
+  // 52-87 across four geometries).  v6 — bf16-split emulation keeping the
+  // hi*lo terms — is f32-GRADE (maxerr matches v1's at every probed shape:
+  // both ~2^-24-class vs a float64 reference) and 1.3-1.5x faster at
+  // K>=64 on large tiles: 129.5 vs 87 TF @8192^2x256, 83 vs 64 @4096^2x128,
+  // 65 vs 60 @16384^2x64.  v5 (drops hi*lo: ~2^-16/term, maxerr ~2e-3 on
+  // those probes) is the opt-in fast mode at 156.7 TF.  v1 keeps short-K
+  // and small shapes (one 256^2 K-step can't amortize the split).
+  // LPX_PW_V2=1/3/4/5/6 force a variant.
   const char* pv = getenv("LPX_PW_V2");
-  if (!pv && K >= 192 && M >= 1024 && N >= 1024) pv = "4";
+  if (!pv && K >= 64 && M >= 2048 && N >= 2048) pv = "6";
   if ((M >= 256 && N >= 256) && (pv && pv[0] == '1')) {
     // v2: 32x32x2 MFMA, 128^2 tile (LDS-read pressure halved per FLOP)
     const unsigned tm = lpx_ceil_div(M, BM2), tn = lpx_ceil_div(N, BM2);

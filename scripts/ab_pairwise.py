@@ -23,7 +23,8 @@ def timeit(fn, iters=20, warmup=5):
 
 def main():
     for m, n, k in ((4096, 4096, 128), (8192, 8192, 256), (4096, 4096, 32),
-                    (16384, 16384, 64)):
+                    (16384, 16384, 64), (2048, 2048, 64), (2048, 2048, 256),
+                    (4096, 4096, 64)):
         torch.manual_seed(0)
         X = torch.randn(m, k, device="cuda")
         Y = torch.randn(n, k, device="cuda")

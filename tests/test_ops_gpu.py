@@ -144,6 +144,26 @@ def test_pairwise_vs_cdist(ops, m, n, k):
     assert err < 1e-2 * (ref.max().item() + 1), err
 
 
+@pytest.mark.parametrize("variant,rel_bound", [("5", 3e-4), ("6", 1e-6)])
+def test_pairwise_bf16_split_variants(ops, monkeypatch, variant, rel_bound):
+    """v5/v6 bf16-split emulation: x=hi+mid(+lo) exact truncation splits,
+    4/6 cross products on the bf16 MFMA pipe.  v6 keeps hi*lo terms ->
+    f32-grade (~2^-24); v5 drops them -> ~2^-16/term, still far inside the
+    sklearn pairwise tolerance.  Checked vs a float64 reference + both must
+    be deterministic."""
+    m, n, k = 1280, 1100, 200  # non-multiples of the 256/32 tile dims
+    X = _rand((m, k), seed=5)
+    Y = _rand((n, k), seed=6)
+    monkeypatch.setenv("LPX_PW_V2", variant)
+    D = ops.pairwise_sqeuclidean(X, Y)
+    D2 = ops.pairwise_sqeuclidean(X, Y)
+    assert torch.equal(D, D2), "nondeterministic split-pairwise"
+    ref = torch.cdist(X.double(), Y.double()) ** 2
+    err = (D.double() - ref).abs().max().item()
+    assert err < rel_bound * (ref.max().item() + 1), \
+        f"v{variant} err {err:.3e} bound {rel_bound * (ref.max().item() + 1):.3e}"
+
+
 # ---- RandomForest --------------------------------------------------------
 
 def test_rf_vs_cpu_reference(ops):

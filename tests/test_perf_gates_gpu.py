@@ -133,7 +133,9 @@ def test_pairwise_floor():
     Y = torch.randn(n, k, device=DEV)
     dt = _time_op(lambda: ops.pairwise_sqeuclidean(X, Y))
     tf = 2 * m * n * k / dt / 1e12
-    assert tf > 25, f"pairwise at {tf:.0f} TF (floor 25, f32 MFMA peak 157)"
+    # v6 bf16-split default measured 83 TF here (and 129 @8192^2x256);
+    # the old exact-f32 ceiling was 157 TF with v1-v4 plateauing at 52-87
+    assert tf > 55, f"pairwise at {tf:.0f} TF (floor 55, v6 measured 83)"
 
 
 def test_resnet50_lite_floor():
