@@ -698,6 +698,190 @@ __global__ __launch_bounds__(THREADS3, 1) void pairwise5_kernel(
   }
 }
 
+// 1024-thread staging: one 16-B cell per operand per thread
+__device__ __forceinline__ void pw7_load(const float* __restrict__ src,
+                                         int rows_total, int K, int row0,
+                                         int k0, float (&v)[1][8]) {
+  const int q = threadIdx.x;  // physical 16-B slot 0..1023
+  const int line = q >> 3;
+  const int lidx = ((q & 7) - line) & 7;
+  const int r = line * 2 + (lidx >> 2), c = lidx & 3;
+  const int grow = row0 + r, gk = k0 + c * 8;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) v[0][j] = 0.f;
+  if (grow < rows_total) {
+    const float* g = src + (long)grow * K + gk;
+    if (gk + 8 <= K) {
+      *reinterpret_cast<f32x4*>(&v[0][0]) = *reinterpret_cast<const f32x4*>(g);
+      *reinterpret_cast<f32x4*>(&v[0][4]) =
+          *reinterpret_cast<const f32x4*>(g + 4);
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        if (gk + j < K) v[0][j] = g[j];
+    }
+  }
+}
+
+template <int NS>
+__device__ __forceinline__ void pw7_commit(float (&v)[1][8],
+                                           char* __restrict__ tile0) {
+  const int q = threadIdx.x;
+  unsigned hw[4], mw[4], lw[4];
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    const float x0 = v[0][2 * j], x1 = v[0][2 * j + 1];
+    const unsigned u0 = __builtin_bit_cast(unsigned, x0);
+    const unsigned u1 = __builtin_bit_cast(unsigned, x1);
+    const float h0 = __builtin_bit_cast(float, u0 & 0xFFFF0000u);
+    const float h1 = __builtin_bit_cast(float, u1 & 0xFFFF0000u);
+    const float r0 = x0 - h0, r1 = x1 - h1;  // exact
+    const unsigned m0 = __builtin_bit_cast(unsigned, r0);
+    const unsigned m1 = __builtin_bit_cast(unsigned, r1);
+    hw[j] = (u0 >> 16) | (u1 & 0xFFFF0000u);
+    mw[j] = (m0 >> 16) | (m1 & 0xFFFF0000u);
+    if (NS == 3) {
+      const float q0 = r0 - __builtin_bit_cast(float, m0 & 0xFFFF0000u);
+      const float q1 = r1 - __builtin_bit_cast(float, m1 & 0xFFFF0000u);
+      lw[j] = (__builtin_bit_cast(unsigned, q0) >> 16) |
+              (__builtin_bit_cast(unsigned, q1) & 0xFFFF0000u);
+    }
+  }
+  char* const p = tile0 + q * 16;
+  *reinterpret_cast<u32x4pw*>(p) = u32x4pw{hw[0], hw[1], hw[2], hw[3]};
+  *reinterpret_cast<u32x4pw*>(p + PW5_TILE) =
+      u32x4pw{mw[0], mw[1], mw[2], mw[3]};
+  if (NS == 3)
+    *reinterpret_cast<u32x4pw*>(p + 2 * PW5_TILE) =
+        u32x4pw{lw[0], lw[1], lw[2], lw[3]};
+}
+
+// ---- v7/v8: bf16-split at 4 waves/SIMD -----------------------------------
+// v5/v6 run 2 waves/SIMD (234-248 VGPRs) and reach ~25% of the split-scheme
+// ceiling — latency-bound: each K-step serializes ds_reads -> MFMAs ->
+// commit -> barrier with only one other wave to hide behind.  Same 256^2
+// tile with 1024 threads (16 waves, 4x4; 64x64 per wave) quarters the
+// accumulator file (acc[4][4] = 64 VGPRs) to fit 4 waves/SIMD.
+template <int TERMS>
+__global__ __launch_bounds__(1024, 1) void pairwise7_kernel(
+    const float* __restrict__ X, const float* __restrict__ Y,
+    const float* __restrict__ xsq, const float* __restrict__ ysq,
+    float* __restrict__ D, int M, int N, int K, unsigned tiles_m,
+    unsigned tiles_n) {
+  constexpr int NS = (TERMS == 4) ? 2 : 3;
+  constexpr bool DB = (TERMS == 4);
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  constexpr int PAR = 2 * NS * PW5_TILE;
+
+  const unsigned nwg = tiles_m * tiles_n;
+  const unsigned bid = lpx_xcd_swizzle(blockIdx.x, nwg);
+  const int bm = (int)(bid / tiles_n) * BM5;
+  const int bn = (int)(bid % tiles_n) * BM5;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wr = wid >> 2, wc = wid & 3;  // 4x4 waves; 64x64 per wave
+  const int fr = lane & 15, fkb = lane >> 4;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  // staging: 1024 threads x 1 cell per operand
+  float xv[1][8], yv[1][8];
+  pw7_load(X, M, K, bm, 0, xv);
+  pw7_load(Y, N, K, bn, 0, yv);
+  pw7_commit<NS>(xv, smem);
+  pw7_commit<NS>(yv, smem + NS * PW5_TILE);
+  __syncthreads();
+
+  const int ksteps = (K + BK5 - 1) / BK5;
+  int cur = 0;
+  for (int ks = 0; ks < ksteps; ++ks) {
+    char* const abase = smem + (DB ? cur * PAR : 0);
+    char* const bbase = abase + NS * PW5_TILE;
+    const bool more = ks + 1 < ksteps;
+    if (more) {
+      pw7_load(X, M, K, bm, (ks + 1) * BK5, xv);
+      pw7_load(Y, N, K, bn, (ks + 1) * BK5, yv);
+    }
+
+    bf16x8 bh[4], bm_[4], bl[4];
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+      const int br = wc * 64 + nf * 16 + fr;
+      const int off = pw5_cell(br, fkb) * 16;
+      bh[nf] = *reinterpret_cast<const bf16x8*>(bbase + off);
+      bm_[nf] = *reinterpret_cast<const bf16x8*>(bbase + PW5_TILE + off);
+      if constexpr (TERMS == 6)
+        bl[nf] = *reinterpret_cast<const bf16x8*>(bbase + 2 * PW5_TILE + off);
+    }
+#pragma unroll
+    for (int mf = 0; mf < 4; ++mf) {
+      const int ar = wr * 64 + mf * 16 + fr;
+      const int off = pw5_cell(ar, fkb) * 16;
+      const bf16x8 ah = *reinterpret_cast<const bf16x8*>(abase + off);
+      const bf16x8 am =
+          *reinterpret_cast<const bf16x8*>(abase + PW5_TILE + off);
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf)
+        acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            ah, bh[nf], acc[mf][nf], 0, 0, 0);
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf)
+        acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            ah, bm_[nf], acc[mf][nf], 0, 0, 0);
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf)
+        acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            am, bh[nf], acc[mf][nf], 0, 0, 0);
+#pragma unroll
+      for (int nf = 0; nf < 4; ++nf)
+        acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            am, bm_[nf], acc[mf][nf], 0, 0, 0);
+      if constexpr (TERMS == 6) {
+        const bf16x8 al =
+            *reinterpret_cast<const bf16x8*>(abase + 2 * PW5_TILE + off);
+#pragma unroll
+        for (int nf = 0; nf < 4; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              ah, bl[nf], acc[mf][nf], 0, 0, 0);
+#pragma unroll
+        for (int nf = 0; nf < 4; ++nf)
+          acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              al, bh[nf], acc[mf][nf], 0, 0, 0);
+      }
+    }
+    if (more) {
+      if constexpr (!DB) __syncthreads();
+      pw7_commit<NS>(xv, smem + (DB ? (cur ^ 1) * PAR : 0));
+      pw7_commit<NS>(yv, smem + (DB ? (cur ^ 1) * PAR : 0) + NS * PW5_TILE);
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+  const int orow_l = (lane >> 4) * 4;
+  const int ocol_l = lane & 15;
+#pragma unroll
+  for (int mf = 0; mf < 4; ++mf) {
+#pragma unroll
+    for (int nf = 0; nf < 4; ++nf) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = bm + wr * 64 + mf * 16 + orow_l + r;
+        const int col = bn + wc * 64 + nf * 16 + ocol_l;
+        if (row < M && col < N) {
+          float d = xsq[row] - 2.0f * acc[mf][nf][r] + ysq[col];
+          D[(long)row * N + col] = d > 0.f ? d : 0.f;
+        }
+      }
+    }
+  }
+}
+
 __global__ __launch_bounds__(THREADS) void row_sqnorm_kernel(
     const float* __restrict__ X, float* __restrict__ out, int M, long K) {
   const int lane = threadIdx.x & 63;
@@ -751,6 +935,18 @@ hipError_t pairwise_sqeuclidean_f32(const float* X, const float* Y,
     const size_t lds = sizeof(float) * 4u * BM3 * ROWF3;
     pairwise3_kernel<<<dim3(tm * tn), dim3(THREADS3), lds, stream>>>(
         X, Y, xsq, ysq, D, M, N, K, tm, tn);
+    return hipGetLastError();
+  }
+  if ((M >= 256 && N >= 256) && pv && (pv[0] == '7' || pv[0] == '8')) {
+    // v7/v8: bf16-split, 1024 threads (4 waves/SIMD occupancy probe)
+    const unsigned tm = lpx_ceil_div(M, BM5), tn = lpx_ceil_div(N, BM5);
+    const size_t lds = (pv[0] == '7') ? 4u * PW5_TILE * 2 : 6u * PW5_TILE;
+    if (pv[0] == '7')
+      pairwise7_kernel<4><<<dim3(tm * tn), dim3(1024), lds, stream>>>(
+          X, Y, xsq, ysq, D, M, N, K, tm, tn);
+    else
+      pairwise7_kernel<6><<<dim3(tm * tn), dim3(1024), lds, stream>>>(
+          X, Y, xsq, ysq, D, M, N, K, tm, tn);
     return hipGetLastError();
   }
   if ((M >= 256 && N >= 256) && pv && (pv[0] == '5' || pv[0] == '6')) {

@@ -29,7 +29,10 @@ def main():
         X = torch.randn(m, k, device="cuda")
         Y = torch.randn(n, k, device="cuda")
         ref = None
-        for env, name in ((None, "v1"), ("4", "v4"), ("5", "v5"), ("6", "v6")):
+        import os as _os
+        _vl = _os.environ.get("PW_VARIANTS", "d,4,5,6")
+        pairs = [((None if v == "d" else v), ("default" if v == "d" else "v" + v)) for v in _vl.split(",")]
+        for env, name in pairs:
             if env is None:
                 os.environ.pop("LPX_PW_V2", None)
             else:
