@@ -808,15 +808,13 @@ __global__ __launch_bounds__(1024, 1) void pairwise7_kernel(
       pw7_load(Y, N, K, bn, (ks + 1) * BK5, yv);
     }
 
-    bf16x8 bh[4], bm_[4], bl[4];
+    bf16x8 bh[4], bm_[4];
 #pragma unroll
     for (int nf = 0; nf < 4; ++nf) {
       const int br = wc * 64 + nf * 16 + fr;
       const int off = pw5_cell(br, fkb) * 16;
       bh[nf] = *reinterpret_cast<const bf16x8*>(bbase + off);
       bm_[nf] = *reinterpret_cast<const bf16x8*>(bbase + PW5_TILE + off);
-      if constexpr (TERMS == 6)
-        bl[nf] = *reinterpret_cast<const bf16x8*>(bbase + 2 * PW5_TILE + off);
     }
 #pragma unroll
     for (int mf = 0; mf < 4; ++mf) {
@@ -841,19 +839,40 @@ __global__ __launch_bounds__(1024, 1) void pairwise7_kernel(
       for (int nf = 0; nf < 4; ++nf)
         acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
             am, bm_[nf], acc[mf][nf], 0, 0, 0);
-      if constexpr (TERMS == 6) {
+    }
+    if constexpr (TERMS == 6) {
+      // second pass for the lo cross terms with a minimal live set: every
+      // b frag re-read from LDS per use (keeping bh/bl arrays live across
+      // the pass spilled 121-127 VGPRs at the 128 budget)
+      __builtin_amdgcn_sched_barrier(0);  // pass-1 frags die here
+#pragma unroll
+      for (int mf = 0; mf < 4; ++mf) {
+        const int ar = wr * 64 + mf * 16 + fr;
+        const int off = pw5_cell(ar, fkb) * 16;
+        const bf16x8 ah = *reinterpret_cast<const bf16x8*>(abase + off);
         const bf16x8 al =
             *reinterpret_cast<const bf16x8*>(abase + 2 * PW5_TILE + off);
 #pragma unroll
-        for (int nf = 0; nf < 4; ++nf)
+        for (int nf = 0; nf < 4; ++nf) {
+          const int boff = pw5_cell(wc * 64 + nf * 16 + fr, fkb) * 16;
+          const bf16x8 blv =
+              *reinterpret_cast<const bf16x8*>(bbase + 2 * PW5_TILE + boff);
+          const bf16x8 bhv = *reinterpret_cast<const bf16x8*>(bbase + boff);
           acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              ah, bl[nf], acc[mf][nf], 0, 0, 0);
-#pragma unroll
-        for (int nf = 0; nf < 4; ++nf)
+              ah, blv, acc[mf][nf], 0, 0, 0);
           acc[mf][nf] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              al, bh[nf], acc[mf][nf], 0, 0, 0);
+              al, bhv, acc[mf][nf], 0, 0, 0);
+          // keep the scheduler from hoisting all 32 pass-2 ds_reads at
+          // once (that live-set is what spilled 105 VGPRs); 4 waves/SIMD
+          // hide the per-read latency instead
+          __builtin_amdgcn_sched_barrier(0);
+        }
       }
     }
+    // fence the scheduler: without it the commit's split arithmetic is
+    // hoisted above the MFMA passes and its 20+ temporaries stay live
+    // through them (TERMS=6 spilled 105-127 VGPRs at the 128 budget)
+    __builtin_amdgcn_sched_barrier(0);
     if (more) {
       if constexpr (!DB) __syncthreads();
       pw7_commit<NS>(xv, smem + (DB ? (cur ^ 1) * PAR : 0));

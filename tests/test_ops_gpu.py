@@ -144,7 +144,8 @@ def test_pairwise_vs_cdist(ops, m, n, k):
     assert err < 1e-2 * (ref.max().item() + 1), err
 
 
-@pytest.mark.parametrize("variant,rel_bound", [("5", 3e-4), ("6", 1e-6)])
+@pytest.mark.parametrize("variant,rel_bound", [("5", 3e-4), ("6", 1e-6),
+                                               ("7", 3e-4), ("8", 1e-6)])
 def test_pairwise_bf16_split_variants(ops, monkeypatch, variant, rel_bound):
     """v5/v6 bf16-split emulation: x=hi+mid(+lo) exact truncation splits,
     4/6 cross products on the bf16 MFMA pipe.  v6 keeps hi*lo terms ->
