@@ -127,7 +127,15 @@ class RecipeRegistry:
         return cls(recipes)
 
     def match(self, name: str, version: str) -> Optional[BuildRecipe]:
-        r = self._recipes.get(canonicalize_name(name))
+        cn = canonicalize_name(name)
+        r = self._recipes.get(cn)
+        if r is None:
+            # provides aliases: e.g. a psycopg2-binary pin resolves to the
+            # psycopg2 recipe (SURVEY.md §2.1 R3 provides/excludes info)
+            for cand in self._recipes.values():
+                if cn in (canonicalize_name(p) for p in cand.provides):
+                    r = cand
+                    break
         if r is not None and r.supports(version):
             return r
         return None

@@ -53,3 +53,16 @@ def test_provides_defaults_to_name():
     )
     assert r.name == "foo-bar"
     assert r.provides == ["foo-bar"]
+
+
+def test_provides_alias_matching():
+    """A pin on a provides alias (psycopg2-binary, PIL) resolves to the
+    canonical recipe — SURVEY.md §2.1 R3 provides/excludes info."""
+    from lambdipy_amd.recipes import RecipeRegistry
+
+    reg = RecipeRegistry.load()
+    assert reg.match("psycopg2-binary", "2.9.9").name == "psycopg2"
+    assert reg.match("PIL", "10.0.0").name == "pillow"
+    assert reg.match("Pillow", "10.4.0").name == "pillow"
+    # non-alias misses stay None
+    assert reg.match("definitely-not-a-recipe", "1.0") is None
