@@ -973,14 +973,19 @@ hipError_t gemm_fp8_nt_res(const unsigned char* A, const unsigned char* Bt,
                            float* C, const float* bias, const float* resid,
                            int M, int N, int K, float descale, bool relu,
                            hipStream_t stream) {
+  // LPX_FP8_ROUTE forces a tier for A/B: t=tall-skinny  b=nt256  n=nt
+  // (and widens that tier's shape gate to take the forced shapes).
+  const char* route = getenv("LPX_FP8_ROUTE");
+  const char rc = route ? route[0] : 0;
   // scaled-MFMA path first: 2x the 16x16x32 instruction rate (unit scales)
-  if (M >= 512 && N >= 512 && M % 256 == 0 && N % 256 == 0 && K % 128 == 0 &&
-      K >= 128)
+  if (!rc && M >= 512 && N >= 512 && M % 256 == 0 && N % 256 == 0 &&
+      K % 128 == 0 && K >= 128)
     return gemm_fp8_nt8phx_res(A, Bt, C, bias, resid, M, N, K, descale, relu,
                                stream);
-  if (!resid && M >= 1024 && N >= 1024 && gemm_fp8_8ph_supported(M, N, K))
+  if (!rc && !resid && M >= 1024 && N >= 1024 &&
+      gemm_fp8_8ph_supported(M, N, K))
     return gemm_fp8_nt8ph(A, Bt, C, bias, M, N, K, descale, relu, stream);
-  if (!resid && N <= 64 && M >= 2048) {
+  if (!resid && (rc == 't' ? (N <= 256) : (!rc && N <= 64)) && M >= 2048) {
     // tall-skinny (ResNet small-KO 1x1 convs): 256x64 scaled-MFMA tile
     const unsigned tm = lpx_ceil_div(M, BMT), tn = lpx_ceil_div(N, BNT);
     const size_t tlds = (size_t)2 * (BMT + BNT) * ROWT;  // 50 KiB
@@ -999,7 +1004,9 @@ hipError_t gemm_fp8_nt_res(const unsigned char* A, const unsigned char* Bt,
           A, Bt, C, bias, M, N, K, descale, tm, tn);
     return hipGetLastError();
   }
-  const bool big = (M >= 512 && N >= 512 && K >= 64);
+  const bool big = rc == 'n' ? false
+                             : (M >= 512 && (N >= (rc == 'b' ? 128 : 512)) &&
+                                K >= 64);
   if (big) {
     const unsigned tiles_m = lpx_ceil_div(M, BM2), tiles_n = lpx_ceil_div(N, BN2);
     const size_t lds = (size_t)4 * BM2 * ROWB2;  // 80 KiB (2 LDS slots)
