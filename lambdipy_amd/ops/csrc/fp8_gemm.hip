@@ -985,7 +985,24 @@ hipError_t gemm_fp8_nt_res(const unsigned char* A, const unsigned char* Bt,
   if (!rc && !resid && M >= 1024 && N >= 1024 &&
       gemm_fp8_8ph_supported(M, N, K))
     return gemm_fp8_nt8ph(A, Bt, C, bias, M, N, K, descale, relu, stream);
-  if (!resid && (rc == 't' ? (N <= 256) : (!rc && N <= 64)) && M >= 2048) {
+  // M-split ('s' forces, measured default below): M%256!=0 blocks the
+  // scaled-MFMA pipeline (glds staging has no row clamp) — run it on the
+  // aligned M0 rows and recurse for the <=255-row tail (lands on nt).
+  // ResNet stage-4 shapes (M=3136=12*256+64, N in {512,2048}) hit this.
+  if (rc == 's' && N >= 512 && N % 256 == 0 && K % 128 == 0 && K >= 128 &&
+      M >= 768 && M % 256 != 0) {
+    const int M0 = (M / 256) * 256;
+    hipError_t err = gemm_fp8_nt8phx_res(A, Bt, C, bias, resid, M0, N, K,
+                                         descale, relu, stream);
+    if (err != hipSuccess) return err;
+    return gemm_fp8_nt_res(A + (long)M0 * K, Bt, C + (long)M0 * N, bias,
+                           resid ? resid + (long)M0 * N : nullptr, M - M0, N,
+                           K, descale, relu, stream);
+  }
+  // tall-skinny: N<=256 measured faster than every wider-tile route for
+  // the ResNet 1x1 shapes (324 vs 219 TF @50176x128x256, 486 vs 263
+  // @12544x256x1024 — scripts/ab_fp8_route.py)
+  if (!resid && (rc == 't' || !rc) && N <= 256 && M >= 2048) {
     // tall-skinny (ResNet small-KO 1x1 convs): 256x64 scaled-MFMA tile
     const unsigned tm = lpx_ceil_div(M, BMT), tn = lpx_ceil_div(N, BNT);
     const size_t tlds = (size_t)2 * (BMT + BNT) * ROWT;  // 50 KiB

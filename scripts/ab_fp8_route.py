@@ -33,7 +33,7 @@ def main():
     # (M, N, K): stage2 conv1, stage3 conv1, stage4 conv1 (ctrl: hits nt256
     # today), giant stage1 1x1 (ctrl: ts today), FC
     shapes = [(50176, 128, 256), (12544, 256, 1024), (3136, 512, 1024),
-              (200704, 64, 256), (64, 1000, 2048)]
+              (3136, 2048, 512), (200704, 64, 256), (64, 1000, 2048)]
     for (m, n, k) in shapes:
         torch.manual_seed(0)
         A = torch.randn(m, k, device="cuda")
@@ -45,12 +45,14 @@ def main():
         ref = (aq.view(torch.float8_e4m3fn).float() * sa) @ \
               (bq.view(torch.float8_e4m3fn).float() * sb).t()
         for rc, name in ((None, "auto"), ("t", "ts"), ("b", "nt256"),
-                         ("n", "nt")):
+                         ("n", "nt"), ("s", "msplit")):
             if rc is None:
                 os.environ.pop("LPX_FP8_ROUTE", None)
             else:
                 os.environ["LPX_FP8_ROUTE"] = rc
             if rc == "t" and not (n <= 256 and m >= 2048):
+                continue
+            if rc == "s" and not (n >= 512 and n % 256 == 0 and m % 256):
                 continue
             C = e.gemm_fp8(aq, bq, sa * sb, None, False)
             err = (C - ref).abs().max().item() / (ref.abs().max().item() + 1e-6)
