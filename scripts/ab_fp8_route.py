@@ -32,8 +32,13 @@ def bench(fn, iters=20, warmup=5):
 def main():
     # (M, N, K): stage2 conv1, stage3 conv1, stage4 conv1 (ctrl: hits nt256
     # today), giant stage1 1x1 (ctrl: ts today), FC
-    shapes = [(50176, 128, 256), (12544, 256, 1024), (3136, 512, 1024),
-              (3136, 2048, 512), (200704, 64, 256), (64, 1000, 2048)]
+    import json as _json
+    env_shapes = os.environ.get("AB_SHAPES")
+    if env_shapes:
+        shapes = [tuple(t) for t in _json.loads(env_shapes)]
+    else:
+        shapes = [(50176, 128, 256), (12544, 256, 1024), (3136, 512, 1024),
+                  (3136, 2048, 512), (200704, 64, 256), (64, 1000, 2048)]
     for (m, n, k) in shapes:
         torch.manual_seed(0)
         A = torch.randn(m, k, device="cuda")
