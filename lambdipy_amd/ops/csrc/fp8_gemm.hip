@@ -985,12 +985,16 @@ hipError_t gemm_fp8_nt_res(const unsigned char* A, const unsigned char* Bt,
   if (!rc && !resid && M >= 1024 && N >= 1024 &&
       gemm_fp8_8ph_supported(M, N, K))
     return gemm_fp8_nt8ph(A, Bt, C, bias, M, N, K, descale, relu, stream);
-  // M-split ('s' forces, measured default below): M%256!=0 blocks the
-  // scaled-MFMA pipeline (glds staging has no row clamp) — run it on the
-  // aligned M0 rows and recurse for the <=255-row tail (lands on nt).
-  // ResNet stage-4 shapes (M=3136=12*256+64, N in {512,2048}) hit this.
-  if (rc == 's' && N >= 512 && N % 256 == 0 && K % 128 == 0 && K >= 128 &&
-      M >= 768 && M % 256 != 0) {
+  // M-split ('s' forces; default at chip-filling grids): M%256!=0 blocks
+  // the scaled-MFMA pipeline (glds staging has no row clamp) — run it on
+  // the aligned M0 rows and recurse for the <=255-row tail (lands on nt).
+  // Measured (ab_fp8_route.py): 815 vs 672 (nt256) vs 593 (nt) TF at
+  // 25216x1024x1024 (~396 tiles); LOSES below ~hundred tiles (195 vs 263
+  // at 4928x1024) — underfilled 8phx grids don't pay, so the default
+  // requires >=256 256^2-tiles.
+  const long t256 = (long)lpx_ceil_div(M, 256) * lpx_ceil_div(N, 256);
+  if ((rc == 's' || (!rc && t256 >= 256)) && N >= 512 && N % 256 == 0 &&
+      K % 128 == 0 && K >= 128 && M >= 768 && M % 256 != 0) {
     const int M0 = (M / 256) * 256;
     hipError_t err = gemm_fp8_nt8phx_res(A, Bt, C, bias, resid, M0, N, K,
                                          descale, relu, stream);
@@ -1021,9 +1025,14 @@ hipError_t gemm_fp8_nt_res(const unsigned char* A, const unsigned char* Bt,
           A, Bt, C, bias, M, N, K, descale, tm, tn);
     return hipGetLastError();
   }
+  // nt256 only when its 256^2 grid fills the chip: at 26-104 tiles the
+  // 128^2 nt kernel wins (136 vs 83 TF @3136x512, 316 vs 260 @3136x2048 —
+  // ResNet's stage-4 shapes); at ~400 tiles nt256 wins (370 vs 343
+  // @12608x2048).  LPX_FP8_ROUTE: b=force nt256, n=force nt.
   const bool big = rc == 'n' ? false
-                             : (M >= 512 && (N >= (rc == 'b' ? 128 : 512)) &&
-                                K >= 64);
+                 : rc == 'b' ? (M >= 512 && N >= 128 && K >= 64)
+                             : (M >= 512 && N >= 512 && K >= 64 &&
+                                t256 >= 256);
   if (big) {
     const unsigned tiles_m = lpx_ceil_div(M, BM2), tiles_n = lpx_ceil_div(N, BN2);
     const size_t lds = (size_t)4 * BM2 * ROWB2;  // 80 KiB (2 LDS slots)
