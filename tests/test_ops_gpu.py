@@ -419,6 +419,37 @@ def test_gemm_fp8_tall_skinny(ops):
         assert err < 1e-2, (M, N, K, err)
 
 
+def test_gemm_fp8_routing_tiers(ops, monkeypatch):
+    """Grid-fill-aware dispatch (scripts/ab_fp8_route.py): tall-skinny now
+    serves N<=256 by default, and LPX_FP8_ROUTE can force the M-split
+    (8phx on aligned rows + nt tail) and nt256 tiers — every route must
+    agree with the dequantized-operand reference on the same shape."""
+    e = ops.ext()
+    g = torch.Generator(device="cpu").manual_seed(31)
+
+    def check(M, N, K, route):
+        if route is None:
+            monkeypatch.delenv("LPX_FP8_ROUTE", raising=False)
+        else:
+            monkeypatch.setenv("LPX_FP8_ROUTE", route)
+        A = torch.randn(M, K, generator=g).to(DEV)
+        Bt = torch.randn(N, K, generator=g).to(DEV)
+        Aq = e.quantize_fp8(A, 1.0)
+        Btq = e.quantize_fp8(Bt, 1.0)
+        C = e.gemm_fp8(Aq, Btq, 1.0, None, False)
+        ref = (Aq.view(torch.float8_e4m3fn).float()
+               @ Btq.view(torch.float8_e4m3fn).float().t())
+        err = (C - ref).abs().max().item() / (ref.abs().max().item() + 1e-6)
+        assert err < 1e-3, (M, N, K, route, err)
+
+    check(2048, 128, 320, None)   # widened tall-skinny default (N=128)
+    check(2304, 256, 512, None)   # tall-skinny default (N=256)
+    check(832, 512, 128, "s")     # forced M-split: 8phx 768 rows + 64-row nt tail
+    check(832, 512, 96, "b")      # forced nt256
+    check(3136, 512, 96, None)    # under-filled grid -> nt
+    monkeypatch.delenv("LPX_FP8_ROUTE", raising=False)
+
+
 def test_linear_w4_matches_f32(ops):
     from lambdipy_amd.shims.lpx_torch import LinearW4
     g = torch.Generator(device="cpu").manual_seed(20)
