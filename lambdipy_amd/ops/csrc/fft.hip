@@ -27,20 +27,13 @@ __device__ __forceinline__ f32x2 cmul(f32x2 a, f32x2 b) {
 // Stockham in LDS, radix-4 stages (+ one radix-2 stage when log2(n) is odd).
 // Half the stage count / barrier count of pure radix-2 (OTFFT-style
 // recurrence; twiddles w1 = tw[p*s], w2 = tw[2*p*s], w3 = w1*w2).
-// LDS bank swizzle: pad one 8-B slot every 16 complex elements.  The
-// early stages write with 16-element (128-B, a full bank period) strides
-// between p-groups — 4-way conflicts at s=1 and s=4 (PMC measured 0.4
-// conflicts/idx before the pad).  phys(i) keeps element pairs (2i, 2i+1)
-// contiguous, so the HBM side of the i/o copies stays 16-B vectorized.
-__device__ __forceinline__ int fft_phys(int i) { return i + (i >> 4); }
-
 template <bool INV>
 __global__ __launch_bounds__(THREADS) void fft_lds_kernel(
     const float* __restrict__ xg, float* __restrict__ outg,
     const float* __restrict__ tw, int batch, int n, int log2n) {
   extern __shared__ __attribute__((aligned(16))) char smem[];
   f32x2* buf0 = reinterpret_cast<f32x2*>(smem);
-  f32x2* buf1 = buf0 + n + (n >> 4);
+  f32x2* buf1 = buf0 + n;
 
   const int t = threadIdx.x;
   const int half = n >> 1;
@@ -48,16 +41,10 @@ __global__ __launch_bounds__(THREADS) void fft_lds_kernel(
   const f32x2* twc = reinterpret_cast<const f32x2*>(tw);
 
   for (int b = blockIdx.x; b < batch; b += gridDim.x) {
-    // 16-B vectorized global load; the padded LDS slot pair stays
-    // contiguous (2i never ends a 16-block) but may be 8-B aligned,
-    // so the LDS side writes two f32x2
+    // 16-B vectorized load (two complex64 per lane — G13)
     const f32x4* src4 = reinterpret_cast<const f32x4*>(xg) + (long)b * (n >> 1);
-    for (int i = t; i < (n >> 1); i += THREADS) {
-      const f32x4 v = src4[i];
-      const int p2 = fft_phys(2 * i);
-      buf0[p2] = f32x2{v.x, v.y};
-      buf0[p2 + 1] = f32x2{v.z, v.w};
-    }
+    f32x4* b04 = reinterpret_cast<f32x4*>(buf0);
+    for (int i = t; i < (n >> 1); i += THREADS) b04[i] = src4[i];
     __syncthreads();
 
     f32x2* x = buf0;
@@ -69,10 +56,10 @@ __global__ __launch_bounds__(THREADS) void fft_lds_kernel(
       for (int u = t; u < quarter; u += THREADS) {
         const int p = u / s;
         const int q = u - p * s;
-        const f32x2 a = x[fft_phys(q + s * p)];
-        const f32x2 bv = x[fft_phys(q + s * (p + m))];
-        const f32x2 c = x[fft_phys(q + s * (p + 2 * m))];
-        const f32x2 d = x[fft_phys(q + s * (p + 3 * m))];
+        const f32x2 a = x[q + s * p];
+        const f32x2 bv = x[q + s * (p + m)];
+        const f32x2 c = x[q + s * (p + 2 * m)];
+        const f32x2 d = x[q + s * (p + 3 * m)];
         f32x2 w1 = twc[(long)p * s];
         f32x2 w2 = twc[(long)2 * p * s];
         if (INV) { w1.y = -w1.y; w2.y = -w2.y; }
@@ -83,12 +70,11 @@ __global__ __launch_bounds__(THREADS) void fft_lds_kernel(
         // jbmd = j*(b-d) for forward, -j*(b-d) for inverse
         f32x2 jbmd = {-(bv.y - d.y), bv.x - d.x};
         if (INV) jbmd = f32x2{-jbmd.x, -jbmd.y};
-        y[fft_phys(q + s * (4 * p + 0))] = f32x2{apc.x + bpd.x, apc.y + bpd.y};
-        y[fft_phys(q + s * (4 * p + 1))] =
+        y[q + s * (4 * p + 0)] = f32x2{apc.x + bpd.x, apc.y + bpd.y};
+        y[q + s * (4 * p + 1)] =
             cmul(f32x2{amc.x - jbmd.x, amc.y - jbmd.y}, w1);
-        y[fft_phys(q + s * (4 * p + 2))] =
-            cmul(f32x2{apc.x - bpd.x, apc.y - bpd.y}, w2);
-        y[fft_phys(q + s * (4 * p + 3))] =
+        y[q + s * (4 * p + 2)] = cmul(f32x2{apc.x - bpd.x, apc.y - bpd.y}, w2);
+        y[q + s * (4 * p + 3)] =
             cmul(f32x2{amc.x + jbmd.x, amc.y + jbmd.y}, w3);
       }
       __syncthreads();
@@ -100,13 +86,12 @@ __global__ __launch_bounds__(THREADS) void fft_lds_kernel(
       for (int u = t; u < half; u += THREADS) {
         const int p = u / s;
         const int q = u - p * s;
-        const f32x2 a = x[fft_phys(q + s * p)];
-        const f32x2 bv = x[fft_phys(q + s * (p + 1))];
+        const f32x2 a = x[q + s * p];
+        const f32x2 bv = x[q + s * (p + 1)];
         f32x2 w = twc[(long)p * s];
         if (INV) w.y = -w.y;
-        y[fft_phys(q + s * (2 * p))] = f32x2{a.x + bv.x, a.y + bv.y};
-        y[fft_phys(q + s * (2 * p + 1))] =
-            cmul(f32x2{a.x - bv.x, a.y - bv.y}, w);
+        y[q + s * (2 * p)] = f32x2{a.x + bv.x, a.y + bv.y};
+        y[q + s * (2 * p + 1)] = cmul(f32x2{a.x - bv.x, a.y - bv.y}, w);
       }
       __syncthreads();
       f32x2* tmp = x; x = y; y = tmp;
@@ -114,19 +99,15 @@ __global__ __launch_bounds__(THREADS) void fft_lds_kernel(
 
     f32x2* dst_lds = x;  // result lives in x after the final swap
     f32x4* dst4 = reinterpret_cast<f32x4*>(outg) + (long)b * (n >> 1);
+    const f32x4* src_lds4 = reinterpret_cast<const f32x4*>(dst_lds);
     if (INV) {
       const float inv_n = 1.0f / (float)n;
       for (int i = t; i < (n >> 1); i += THREADS) {
-        const int p2 = fft_phys(2 * i);
-        const f32x2 lo = dst_lds[p2], hi = dst_lds[p2 + 1];
-        dst4[i] = f32x4{lo.x * inv_n, lo.y * inv_n, hi.x * inv_n, hi.y * inv_n};
+        f32x4 v = src_lds4[i];
+        dst4[i] = f32x4{v.x * inv_n, v.y * inv_n, v.z * inv_n, v.w * inv_n};
       }
     } else {
-      for (int i = t; i < (n >> 1); i += THREADS) {
-        const int p2 = fft_phys(2 * i);
-        const f32x2 lo = dst_lds[p2], hi = dst_lds[p2 + 1];
-        dst4[i] = f32x4{lo.x, lo.y, hi.x, hi.y};
-      }
+      for (int i = t; i < (n >> 1); i += THREADS) dst4[i] = src_lds4[i];
     }
     __syncthreads();
   }
@@ -176,7 +157,7 @@ hipError_t fft_c64(const float* x, float* out, const float* twiddles,
                           stream);
   }
   if (n <= 4096) {
-    const size_t lds = (size_t)2 * (n + (n >> 4)) * sizeof(f32x2);  // +pad
+    const size_t lds = (size_t)2 * n * sizeof(f32x2);
     unsigned blocks = batch < 8192 ? (unsigned)batch : 8192u;
     if (inverse)
       fft_lds_kernel<true><<<dim3(blocks), dim3(THREADS), lds, stream>>>(
