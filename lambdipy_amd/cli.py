@@ -10,6 +10,7 @@ from __future__ import annotations
 import json
 import os
 import shutil
+import sys
 
 import click
 
@@ -82,12 +83,58 @@ def main():
 @click.option("--handler", "handlers", multiple=True,
               help="handler(s) to ship in the layer (rf, matmul, fftspmv, requests, resnet50)")
 @click.option("--no-transitive", is_flag=True, help="skip transitive resolution")
+@click.option("--force-build", is_flag=True,
+              help="rebuild native artifacts even if cached (the reference's "
+                   "force-docker-build switch: here it drops the in-tree "
+                   "object cache and the lpx wheel-cache entries first)")
+@click.option("--python", "py_version", default=None,
+              help="target python version, e.g. 3.10 (this offline image has "
+                   "exactly one interpreter; other versions error loudly)")
 @click.option("--json", "as_json", is_flag=True, help="print the report as JSON")
-def build(requirements, output, zip_path, handlers, no_transitive, as_json):
+def build(requirements, output, zip_path, handlers, no_transitive, force_build,
+          py_version, as_json):
     """Resolve requirements, build recipe wheels, assemble the layer."""
     from .builder import build_recipe_artifacts
     from .packager import assemble_layer
     from .resolver import parse_project_file, resolve
+
+    if py_version is not None:
+        have = f"{sys.version_info.major}.{sys.version_info.minor}"
+        if py_version not in (have, f"cp{have.replace('.', '')}"):
+            raise click.ClickException(
+                f"python {py_version} requested but only {have} is available "
+                "offline (the reference selects a docker build image; this "
+                "image ships a single interpreter)")
+    if force_build:
+        import glob as _glob
+        import subprocess as _sp
+
+        from .builder.wheel_builder import REPO_ROOT, ensure_nn_lib, \
+            ensure_standalone_hsaco
+        from .builder.wheelgen import WHEEL_CACHE
+
+        removed = 0
+        for pat in ("lpx_ops-*.whl", "lpx_shim_*-*.whl", "lpx_handlers-*.whl"):
+            for p in _glob.glob(os.path.join(WHEEL_CACHE, pat)):
+                os.unlink(p)
+                removed += 1
+        log.info("force-build: dropped %d cached wheels; recompiling "
+                 "gfx950 artifacts...", removed)
+        # full recompile now (setuptools --force ignores object mtimes);
+        # the nn lib / hsaco mtime caches are invalidated by deletion and
+        # rebuilt synchronously below, so the tree is never left bare
+        _sp.run([sys.executable, "setup.py", "build_ext", "--inplace",
+                 "--force"],
+                cwd=REPO_ROOT,
+                env=dict(os.environ, PYTORCH_ROCM_ARCH="gfx950"),
+                check=True)
+        ops_dir = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                               "ops")
+        for a in ("liblpx_nn_gfx950.so", "standalone_gfx950.hsaco"):
+            if os.path.exists(os.path.join(ops_dir, a)):
+                os.unlink(os.path.join(ops_dir, a))
+        ensure_nn_lib()
+        ensure_standalone_hsaco()
 
     if requirements is None:
         for cand in ("requirements.txt", "Pipfile.lock", "Pipfile"):
