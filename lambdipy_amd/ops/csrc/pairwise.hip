@@ -762,7 +762,7 @@ __device__ __forceinline__ void pw7_commit(float (&v)[1][8],
 // commit -> barrier with only one other wave to hide behind.  Same 256^2
 // tile with 1024 threads (16 waves, 4x4; 64x64 per wave) quarters the
 // accumulator file (acc[4][4] = 64 VGPRs) to fit 4 waves/SIMD.
-template <int TERMS>
+template <int TERMS, bool EARLY = false>
 __global__ __launch_bounds__(1024, 1) void pairwise7_kernel(
     const float* __restrict__ X, const float* __restrict__ Y,
     const float* __restrict__ xsq, const float* __restrict__ ysq,
@@ -806,6 +806,15 @@ __global__ __launch_bounds__(1024, 1) void pairwise7_kernel(
     if (more) {
       pw7_load(X, M, K, bm, (ks + 1) * BK5, xv);
       pw7_load(Y, N, K, bn, (ks + 1) * BK5, yv);
+      // EARLY: the cur^1 buffer was fully consumed at the previous
+      // barrier, so committing NOW (instead of after the MFMA block)
+      // kills the 16-reg prefetch live range that was spilling 12 VGPRs
+      // into the loop; the vmcnt wait this exposes is hidden by the
+      // other 3 waves/SIMD.
+      if constexpr (EARLY && DB) {
+        pw7_commit<NS>(xv, smem + (cur ^ 1) * PAR);
+        pw7_commit<NS>(yv, smem + (cur ^ 1) * PAR + NS * PW5_TILE);
+      }
     }
 
     bf16x8 bh[4], bm_[4];
@@ -873,7 +882,7 @@ __global__ __launch_bounds__(1024, 1) void pairwise7_kernel(
     // hoisted above the MFMA passes and its 20+ temporaries stay live
     // through them (TERMS=6 spilled 105-127 VGPRs at the 128 budget)
     __builtin_amdgcn_sched_barrier(0);
-    if (more) {
+    if (more && !(EARLY && DB)) {
       if constexpr (!DB) __syncthreads();
       pw7_commit<NS>(xv, smem + (DB ? (cur ^ 1) * PAR : 0));
       pw7_commit<NS>(yv, smem + (DB ? (cur ^ 1) * PAR : 0) + NS * PW5_TILE);
@@ -961,12 +970,17 @@ hipError_t pairwise_sqeuclidean_f32(const float* X, const float* Y,
         X, Y, xsq, ysq, D, M, N, K, tm, tn);
     return hipGetLastError();
   }
-  if ((M >= 256 && N >= 256) && pv && (pv[0] == '7' || pv[0] == '8')) {
-    // v7/v8: bf16-split, 1024 threads (4 waves/SIMD occupancy probe)
+  if ((M >= 256 && N >= 256) && pv &&
+      (pv[0] == '7' || pv[0] == '8' || pv[0] == '9')) {
+    // v7/v8: bf16-split, 1024 threads (4 waves/SIMD); '9' = v7 with the
+    // early-commit schedule (prefetch live range out of the MFMA block)
     const unsigned tm = lpx_ceil_div(M, BM5), tn = lpx_ceil_div(N, BM5);
-    const size_t lds = (pv[0] == '7') ? 4u * PW5_TILE * 2 : 6u * PW5_TILE;
+    const size_t lds = (pv[0] == '8') ? 6u * PW5_TILE : 4u * PW5_TILE * 2;
     if (pv[0] == '7')
       pairwise7_kernel<4><<<dim3(tm * tn), dim3(1024), lds, stream>>>(
+          X, Y, xsq, ysq, D, M, N, K, tm, tn);
+    else if (pv[0] == '9')
+      pairwise7_kernel<4, true><<<dim3(tm * tn), dim3(1024), lds, stream>>>(
           X, Y, xsq, ysq, D, M, N, K, tm, tn);
     else
       pairwise7_kernel<6><<<dim3(tm * tn), dim3(1024), lds, stream>>>(
