@@ -944,9 +944,10 @@ hipError_t pairwise_sqeuclidean_f32(const float* X, const float* Y,
   // both ~2^-24-class vs a float64 reference) and 1.3-1.5x faster at
   // K>=64 on large tiles: 129.5 vs 87 TF @8192^2x256, 83 vs 64 @4096^2x128,
   // 65 vs 60 @16384^2x64.  The 4-term scheme (drops hi*lo: ~2^-16/term,
-  // maxerr ~2e-3 on those probes) is the opt-in fast mode: v7 (1024 thr,
-  // 4 waves/SIMD) 180.5 TF @8192^2x256, LPX_PW_V2=7 (v5, its 512-thread
-  // 2-waves/SIMD form, 156.7).  v8 (6-term at 4 waves/SIMD) measured
+  // maxerr ~2e-3 on those probes) is the opt-in fast mode: v9 (1024 thr,
+  // 4 waves/SIMD, early-commit: 124 VGPRs zero spills) 206 TF
+  // @8192^2x256, LPX_PW_V2=9; v7 (same minus early commit, 12 in-loop
+  // spills) 184; v5 (512-thread 2-waves/SIMD) 153.  v8 (6-term at 4 waves/SIMD) measured
   // WORSE than v6 (96.5 vs 126.5 — 31 residual VGPR spills + the fenced
   // two-pass schedule cost more than the occupancy gains; negative
   // result, kept for the record).  v1 keeps short-K and small shapes

@@ -145,7 +145,8 @@ def test_pairwise_vs_cdist(ops, m, n, k):
 
 
 @pytest.mark.parametrize("variant,rel_bound", [("5", 3e-4), ("6", 1e-6),
-                                               ("7", 3e-4), ("8", 1e-6)])
+                                               ("7", 3e-4), ("8", 1e-6),
+                                               ("9", 3e-4)])
 def test_pairwise_bf16_split_variants(ops, monkeypatch, variant, rel_bound):
     """v5/v6 bf16-split emulation: x=hi+mid(+lo) exact truncation splits,
     4/6 cross products on the bf16 MFMA pipe.  v6 keeps hi*lo terms ->
