@@ -435,7 +435,7 @@ __device__ __forceinline__ i32x4 ts_stage_b(const unsigned char* __restrict__ Bt
   return v;
 }
 
-template <bool RELU, bool BIAS>
+template <bool RELU, bool BIAS, bool EARLY = false>
 __global__ __launch_bounds__(THREADST, 3) void gemm_fp8_ts_kernel(
     const unsigned char* __restrict__ A, const unsigned char* __restrict__ Bt,
     float* __restrict__ C, const float* __restrict__ bias, int M, int N,
@@ -484,6 +484,23 @@ __global__ __launch_bounds__(THREADST, 3) void gemm_fp8_ts_kernel(
     if (ks + 1 < ksteps) {
       ts_stage_a(A, M, K, bm, (ks + 1) * BKT, aR);
       bR = ts_stage_b(Bt, N, K, bn, (ks + 1) * BKT);
+      // EARLY (pairwise-v9 schedule): the cur^1 slot was consumed before
+      // the previous barrier, so commit NOW — aR/bR die before the MFMA
+      // block instead of living through it, and the pre-barrier tail is
+      // just the MFMAs; 6 waves/SIMD hide the exposed vmcnt wait.
+      if constexpr (EARLY) {
+        const int t = threadIdx.x;
+        char* aN = lA + (cur ^ 1) * ABUF;
+#pragma unroll
+        for (int i = 0; i < CHT; ++i) {
+          const int c = t + i * THREADST;
+          *reinterpret_cast<i32x4*>(aN + (c >> 2) * ROWT + (c & 3) * 16) =
+              aR[i];
+        }
+        if (t < 256)
+          *reinterpret_cast<i32x4*>(lB + (cur ^ 1) * BBUF + (t >> 2) * ROWT +
+                                    (t & 3) * 16) = bR;
+      }
     }
     const char* aCur = lA + cur * ABUF;
     const char* bCur = lB + cur * BBUF;
@@ -508,7 +525,7 @@ __global__ __launch_bounds__(THREADST, 3) void gemm_fp8_ts_kernel(
     // single barrier per step: the cur^1 slot's reads finished at the
     // PREVIOUS step's barrier, so writing it during this step's MFMAs is
     // safe (same argument as conv_fp8_impx)
-    if (ks + 1 < ksteps) {
+    if (!EARLY && ks + 1 < ksteps) {
       const int t = threadIdx.x;
       char* aN = lA + (cur ^ 1) * ABUF;
 #pragma unroll
@@ -1011,6 +1028,22 @@ hipError_t gemm_fp8_nt_res(const unsigned char* A, const unsigned char* Bt,
     const unsigned tm = lpx_ceil_div(M, BMT), tn = lpx_ceil_div(N, BNT);
     const size_t tlds = (size_t)2 * (BMT + BNT) * ROWT;  // 50 KiB
     dim3 tg(tm * tn), tb(THREADST);
+    const char* ee = getenv("LPX_FP8_EARLY");
+    if (ee && ee[0] == '1') {  // pairwise-v9 early-commit A/B
+      if (bias && relu)
+        gemm_fp8_ts_kernel<true, true, true><<<tg, tb, tlds, stream>>>(
+            A, Bt, C, bias, M, N, K, descale, tm, tn);
+      else if (bias)
+        gemm_fp8_ts_kernel<false, true, true><<<tg, tb, tlds, stream>>>(
+            A, Bt, C, bias, M, N, K, descale, tm, tn);
+      else if (relu)
+        gemm_fp8_ts_kernel<true, false, true><<<tg, tb, tlds, stream>>>(
+            A, Bt, C, bias, M, N, K, descale, tm, tn);
+      else
+        gemm_fp8_ts_kernel<false, false, true><<<tg, tb, tlds, stream>>>(
+            A, Bt, C, bias, M, N, K, descale, tm, tn);
+      return hipGetLastError();
+    }
     if (bias && relu)
       gemm_fp8_ts_kernel<true, true><<<tg, tb, tlds, stream>>>(
           A, Bt, C, bias, M, N, K, descale, tm, tn);

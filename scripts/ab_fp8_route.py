@@ -49,11 +49,18 @@ def main():
         bq = e.quantize_fp8(B, sb)
         ref = (aq.view(torch.float8_e4m3fn).float() * sa) @ \
               (bq.view(torch.float8_e4m3fn).float() * sb).t()
-        for rc, name in ((None, "auto"), ("t", "ts"), ("b", "nt256"),
-                         ("n", "nt"), ("s", "msplit")):
+        routes = [(None, "auto"), ("t", "ts"), ("b", "nt256"),
+                  ("n", "nt"), ("s", "msplit")]
+        if os.environ.get("AB_EARLY"):
+            routes = [("t", "ts"), ("t+e", "ts-early")]
+        for rc, name in routes:
+            os.environ.pop("LPX_FP8_EARLY", None)
             if rc is None:
                 os.environ.pop("LPX_FP8_ROUTE", None)
             else:
+                if rc.endswith("+e"):
+                    os.environ["LPX_FP8_EARLY"] = "1"
+                    rc = rc[:-2]
                 os.environ["LPX_FP8_ROUTE"] = rc
             if rc == "t" and not (n <= 256 and m >= 2048):
                 continue
