@@ -879,6 +879,178 @@ __global__ __launch_bounds__(THREADS2, 2) void conv_fp8_impx_kernel(
   }
 }
 
+// ---- impx2: the conv at 4 waves/SIMD ------------------------------------
+// conv_fp8_impx allocates 228 VGPRs (acc[4][2] f32x16 = 128 alone), so it
+// runs 2 waves/SIMD (512-entry file, MI355X_MICROARCH 'Register files') —
+// NOT the 2 blocks/CU its LDS would allow.  Same 256^2 tile with 1024
+// threads (16 waves, 4x4; 64x64 per wave) quarters the accumulator file
+// (acc[2][2] = 64) to fit 4 waves/SIMD — the same restructure that took
+// the bf16-split pairwise from 153 to 206 TF.
+__device__ __forceinline__ void imp_init2(ImpCursor& cu, const ConvGeom g,
+                                          int M, int row0) {
+  const int c = threadIdx.x;  // 1024 threads x 1 16-B chunk
+  const int rt = c >> 2;
+  const int cell = (c & 3) * 16;
+  const int m = row0 + rt;
+  cu.mok = m < M;
+  const int mm = cu.mok ? m : 0;
+  const int wo = mm % g.WO;
+  const int ho = (mm / g.WO) % g.HO;
+  const int n = mm / (g.WO * g.HO);
+  cu.hbase = ho * g.stride - g.pad;
+  cu.wbase = wo * g.stride - g.pad;
+  cu.addrN = (long)n * g.H * g.W * g.C;
+  cu.klin = cell;
+  cu.c0 = cell % g.C;
+  const int rs = cell / g.C;
+  cu.s = rs % g.S;
+  cu.r = rs / g.S;
+}
+
+__device__ __forceinline__ void imp_advance2(ImpCursor& cu, const ConvGeom g) {
+  cu.klin += BK2;
+  cu.c0 += BK2;
+  while (cu.c0 >= g.C) {
+    cu.c0 -= g.C;
+    if (++cu.s == g.S) {
+      cu.s = 0;
+      ++cu.r;
+    }
+  }
+}
+
+__device__ __forceinline__ i32x4 imp_gather2(const unsigned char* __restrict__ xq,
+                                             const ImpCursor& cu,
+                                             const ConvGeom g, int K) {
+  i32x4 v = {0, 0, 0, 0};
+  if (cu.mok && cu.klin + 16 <= K) {
+    const int hi = cu.hbase + cu.r;
+    const int wi = cu.wbase + cu.s;
+    if (hi >= 0 && hi < g.H && wi >= 0 && wi < g.W)
+      v = *reinterpret_cast<const i32x4*>(
+          xq + cu.addrN + ((long)hi * g.W + wi) * g.C + cu.c0);
+  }
+  return v;
+}
+
+__device__ __forceinline__ i32x4 load_fp8_regs2(
+    const unsigned char* __restrict__ src, int rows_total, int K, int row0,
+    int k0) {
+  const int c = threadIdx.x;
+  const int r = c >> 2;
+  const int cell = (c & 3) * 16;
+  const int grow = row0 + r;
+  const int gk = k0 + cell;
+  if (grow < rows_total && gk + 16 <= K)
+    return *reinterpret_cast<const i32x4*>(src + (long)grow * K + gk);
+  unsigned char z[16] = {0};
+  for (int j = 0; j < 16; ++j)
+    if (grow < rows_total && gk + j < K) z[j] = src[(long)grow * K + gk + j];
+  return *reinterpret_cast<const i32x4*>(z);
+}
+
+__device__ __forceinline__ void write_fp8_lds2(const i32x4 v,
+                                               char* __restrict__ lds) {
+  const int c = threadIdx.x;
+  *reinterpret_cast<i32x4*>(lds + (c >> 2) * ROWB2 + (c & 3) * 16) = v;
+}
+
+template <bool RELU, bool BIAS, bool OUT8 = false, bool EARLY = true>
+__global__ __launch_bounds__(1024, 1) void conv_fp8_impx2_kernel(
+    const unsigned char* __restrict__ xq, const unsigned char* __restrict__ Wq,
+    float* __restrict__ C, unsigned char* __restrict__ C8,
+    const float* __restrict__ bias, ConvGeom g, int M, int N, int K,
+    float descale, float out_inv, unsigned tiles_m, unsigned tiles_n) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const int SLOT = 2 * BM2 * ROWB2;
+
+  const unsigned nwg = tiles_m * tiles_n;
+  const unsigned bid = lpx_xcd_swizzle(blockIdx.x, nwg);
+  const int bm = (int)(bid / tiles_n) * BM2;
+  const int bn = (int)(bid % tiles_n) * BN2;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wr = wid >> 2, wc = wid & 3;  // 4x4 waves; 64x64 per wave
+  const int fr = lane & 31;
+  const int fkb = lane >> 5;
+
+  f32x16 acc[2][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int e = 0; e < 16; ++e) acc[i][j][e] = 0.f;
+
+  ImpCursor cu;
+  imp_init2(cu, g, M, bm);
+  i32x4 aReg = imp_gather2(xq, cu, g, K);
+  i32x4 bReg = load_fp8_regs2(Wq, N, K, bn, 0);
+  write_fp8_lds2(aReg, smem);
+  write_fp8_lds2(bReg, smem + BM2 * ROWB2);
+  __syncthreads();
+
+  const int ksteps = (K + BK2 - 1) / BK2;
+  for (int ks = 0; ks < ksteps; ++ks) {
+    char* const lA = smem + (ks & 1) * SLOT;
+    char* const lB = lA + BM2 * ROWB2;
+    if (ks + 1 < ksteps) {
+      imp_advance2(cu, g);
+      aReg = imp_gather2(xq, cu, g, K);
+      bReg = load_fp8_regs2(Wq, N, K, bn, (ks + 1) * BK2);
+      if constexpr (EARLY) {  // pairwise-v9: free slot, shed the live range
+        char* const nA = smem + ((ks + 1) & 1) * SLOT;
+        write_fp8_lds2(aReg, nA);
+        write_fp8_lds2(bReg, nA + BM2 * ROWB2);
+      }
+    }
+    fp8frag32 aF[2], bF[2];
+#pragma unroll
+    for (int mf = 0; mf < 2; ++mf)
+      aF[mf] = frag32(lA, wr * 64 + mf * 32 + fr, fkb);
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf)
+      bF[nf] = frag32(lB, wc * 64 + nf * 32 + fr, fkb);
+#pragma unroll
+    for (int mf = 0; mf < 2; ++mf)
+#pragma unroll
+      for (int nf = 0; nf < 2; ++nf)
+        acc[mf][nf] = __builtin_amdgcn_mfma_scale_f32_32x32x64_f8f6f4(
+            aF[mf], bF[nf], acc[mf][nf], 0, 0, 0, 127, 0, 127);
+    if (!EARLY && ks + 1 < ksteps) {
+      char* const nA = smem + ((ks + 1) & 1) * SLOT;
+      write_fp8_lds2(aReg, nA);
+      write_fp8_lds2(bReg, nA + BM2 * ROWB2);
+    }
+    __syncthreads();
+  }
+
+  const int ocol_l = lane & 31;
+  const int orow_base = 4 * (lane >> 5);
+#pragma unroll
+  for (int mf = 0; mf < 2; ++mf) {
+#pragma unroll
+    for (int nf = 0; nf < 2; ++nf) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        const int row = bm + wr * 64 + mf * 32 + (r & 3) + 8 * (r >> 2) +
+                        orow_base;
+        const int col = bn + wc * 64 + nf * 32 + ocol_l;
+        if (row < M && col < N) {
+          float v = acc[mf][nf][r] * descale;
+          if constexpr (BIAS) v += bias[col];
+          if constexpr (RELU) v = v > 0.f ? v : 0.f;
+          if constexpr (OUT8)
+            C8[(long)row * N + col] = lpx_f32_to_e4m3(v * out_inv);
+          else
+            C[(long)row * N + col] = v;
+        }
+      }
+    }
+  }
+}
+
 }  // namespace
 
 hipError_t conv_fp8_implicit(const unsigned char* xq, const unsigned char* Wq,
@@ -901,6 +1073,36 @@ hipError_t conv_fp8_implicit(const unsigned char* xq, const unsigned char* Wq,
   const unsigned tiles_m = lpx_ceil_div(M, BM2), tiles_n = lpx_ceil_div(KO, BN2);
   const size_t lds = (size_t)4 * BM2 * ROWB2;  // 2 LDS slots (see kernel)
   dim3 grid(tiles_m * tiles_n), block(THREADS2);
+  const char* cv2 = getenv("LPX_CONV_V2");
+  if (cv2 && (cv2[0] == '1' || cv2[0] == '2') && bias) {
+    // impx2 probe: 1024 threads / 4 waves per SIMD ('1' early-commit,
+    // '2' commit-after-MFMA)
+    dim3 blk2(1024);
+    if (cv2[0] == '1') {
+      if (relu)
+        conv_fp8_impx2_kernel<true, true, false, true><<<grid, blk2, lds,
+                                                         stream>>>(
+            xq, Wq, C, nullptr, bias, g, M, KO, K, descale, 1.f, tiles_m,
+            tiles_n);
+      else
+        conv_fp8_impx2_kernel<false, true, false, true><<<grid, blk2, lds,
+                                                          stream>>>(
+            xq, Wq, C, nullptr, bias, g, M, KO, K, descale, 1.f, tiles_m,
+            tiles_n);
+    } else {
+      if (relu)
+        conv_fp8_impx2_kernel<true, true, false, false><<<grid, blk2, lds,
+                                                          stream>>>(
+            xq, Wq, C, nullptr, bias, g, M, KO, K, descale, 1.f, tiles_m,
+            tiles_n);
+      else
+        conv_fp8_impx2_kernel<false, true, false, false><<<grid, blk2, lds,
+                                                           stream>>>(
+            xq, Wq, C, nullptr, bias, g, M, KO, K, descale, 1.f, tiles_m,
+            tiles_n);
+    }
+    return hipGetLastError();
+  }
   if (bias && relu)
     conv_fp8_impx_kernel<true, true><<<grid, block, lds, stream>>>(
         xq, Wq, C, nullptr, bias, g, M, KO, K, descale, 1.f, tiles_m, tiles_n);
@@ -937,6 +1139,34 @@ hipError_t conv_fp8_implicit_q8(const unsigned char* xq,
   const unsigned tiles_m = lpx_ceil_div(M, BM2), tiles_n = lpx_ceil_div(KO, BN2);
   const size_t lds = (size_t)4 * BM2 * ROWB2;  // 2 LDS slots (see kernel)
   dim3 grid(tiles_m * tiles_n), block(THREADS2);
+  const char* cv2q = getenv("LPX_CONV_V2");
+  if (cv2q && (cv2q[0] == '1' || cv2q[0] == '2') && bias) {
+    dim3 blk2(1024);
+    if (cv2q[0] == '1') {
+      if (relu)
+        conv_fp8_impx2_kernel<true, true, true, true><<<grid, blk2, lds,
+                                                        stream>>>(
+            xq, Wq, nullptr, C8, bias, g, M, KO, K, descale, out_inv,
+            tiles_m, tiles_n);
+      else
+        conv_fp8_impx2_kernel<false, true, true, true><<<grid, blk2, lds,
+                                                         stream>>>(
+            xq, Wq, nullptr, C8, bias, g, M, KO, K, descale, out_inv,
+            tiles_m, tiles_n);
+    } else {
+      if (relu)
+        conv_fp8_impx2_kernel<true, true, true, false><<<grid, blk2, lds,
+                                                         stream>>>(
+            xq, Wq, nullptr, C8, bias, g, M, KO, K, descale, out_inv,
+            tiles_m, tiles_n);
+      else
+        conv_fp8_impx2_kernel<false, true, true, false><<<grid, blk2, lds,
+                                                          stream>>>(
+            xq, Wq, nullptr, C8, bias, g, M, KO, K, descale, out_inv,
+            tiles_m, tiles_n);
+    }
+    return hipGetLastError();
+  }
   if (bias && relu)
     conv_fp8_impx_kernel<true, true, true><<<grid, block, lds, stream>>>(
         xq, Wq, nullptr, C8, bias, g, M, KO, K, descale, out_inv, tiles_m,
