@@ -1073,12 +1073,15 @@ hipError_t conv_fp8_implicit(const unsigned char* xq, const unsigned char* Wq,
   const unsigned tiles_m = lpx_ceil_div(M, BM2), tiles_n = lpx_ceil_div(KO, BN2);
   const size_t lds = (size_t)4 * BM2 * ROWB2;  // 2 LDS slots (see kernel)
   dim3 grid(tiles_m * tiles_n), block(THREADS2);
+  // impx2 (1024 threads / 4 waves per SIMD, early-commit) is the DEFAULT:
+  // +9-11% ResNet e2e within-probe over the 228-VGPR 2-waves/SIMD impx
+  // (18.8k vs 17.3k img/s bs128).  LPX_CONV_V2=0 forces legacy impx,
+  // '2' the commit-after-MFMA form (128 VGPRs + 2 spills).
   const char* cv2 = getenv("LPX_CONV_V2");
-  if (cv2 && (cv2[0] == '1' || cv2[0] == '2') && bias) {
-    // impx2 probe: 1024 threads / 4 waves per SIMD ('1' early-commit,
-    // '2' commit-after-MFMA)
+  const char cvm = cv2 ? cv2[0] : '1';
+  if (cvm != '0' && bias) {
     dim3 blk2(1024);
-    if (cv2[0] == '1') {
+    if (cvm == '1') {
       if (relu)
         conv_fp8_impx2_kernel<true, true, false, true><<<grid, blk2, lds,
                                                          stream>>>(
@@ -1140,9 +1143,10 @@ hipError_t conv_fp8_implicit_q8(const unsigned char* xq,
   const size_t lds = (size_t)4 * BM2 * ROWB2;  // 2 LDS slots (see kernel)
   dim3 grid(tiles_m * tiles_n), block(THREADS2);
   const char* cv2q = getenv("LPX_CONV_V2");
-  if (cv2q && (cv2q[0] == '1' || cv2q[0] == '2') && bias) {
+  const char cvmq = cv2q ? cv2q[0] : '1';  // impx2 default (see above)
+  if (cvmq != '0' && bias) {
     dim3 blk2(1024);
-    if (cv2q[0] == '1') {
+    if (cvmq == '1') {
       if (relu)
         conv_fp8_impx2_kernel<true, true, true, true><<<grid, blk2, lds,
                                                         stream>>>(
