@@ -489,8 +489,9 @@ __device__ __forceinline__ float lpx_h2f(unsigned short u) {
   return __half2float(__half(r));
 }
 
-template <bool STAGE_X, int SILP, bool LV16 = false, int TILP = 1>
-__global__ __launch_bounds__(THREADS) void rf_heap_cpair_kernel(
+template <bool STAGE_X, int SILP, bool LV16 = false, int TILP = 1,
+          int NT = THREADS>
+__global__ __launch_bounds__(NT) void rf_heap_cpair_kernel(
     const i32x2* __restrict__ pairs, const int* __restrict__ roots,
     const float* __restrict__ leaf_values, int n_trees, int depth,
     const float* __restrict__ X, float* __restrict__ out, int S, int F,
@@ -506,12 +507,12 @@ __global__ __launch_bounds__(THREADS) void rf_heap_cpair_kernel(
   const int ppt = (1 << (depth - 1)) - 1;  // pairs per tree
   const int n_leaves = 1 << depth;
 
-  for (int tile = blockIdx.x * (THREADS * SILP); tile < S;
-       tile += gridDim.x * (THREADS * SILP)) {
-    const int tile_n = min(THREADS * SILP, S - tile);
+  for (int tile = blockIdx.x * (NT * SILP); tile < S;
+       tile += gridDim.x * (NT * SILP)) {
+    const int tile_n = min(NT * SILP, S - tile);
     if (STAGE_X) {
       const long total = (long)tile_n * F;
-      for (long i = t; i < total; i += THREADS) {
+      for (long i = t; i < total; i += NT) {
         const int r = (int)(i / F), c = (int)(i - (long)r * F);
         xl[r * rowf + c] = X[((long)tile + r) * F + c];
       }
@@ -522,8 +523,8 @@ __global__ __launch_bounds__(THREADS) void rf_heap_cpair_kernel(
     const float* xrow[SILP];
 #pragma unroll
     for (int j = 0; j < SILP; ++j) {
-      s[j] = tile + t + j * THREADS;
-      xrow[j] = STAGE_X ? (xl + (t + j * THREADS) * rowf)
+      s[j] = tile + t + j * NT;
+      xrow[j] = STAGE_X ? (xl + (t + j * NT) * rowf)
                         : (X + (long)s[j] * F);
     }
     float acc[SILP][CMAX];
@@ -909,6 +910,19 @@ hipError_t rf_predict_heap_cpair_f32(const int* cpairs, const int* roots,
   const int TILP = (silp == 21) ? 2 : (silp == 41 ? 4 : (silp == 81 ? 8 : 1));
   const int SILP = (silp == 2) ? 2 : (silp == 4 ? 4 : 1);
   const bool stage = F <= F_LDS_MAX;
+  if (silp == 141 && stage) {
+    // TILP4 at 128-thread blocks: halves the X-tile LDS (17 KiB) so 8
+    // blocks fit per CU -> 8 waves/SIMD (the 43-VGPR kernel is LDS-
+    // occupancy-limited at 256 threads: 34 KiB -> 4 blocks -> 4 waves)
+    const size_t l128 = (size_t)128 * (F + 1) * 4;
+    unsigned b128 = lpx_ceil_div(S, 128);
+    if (b128 > 8192u) b128 = 8192u;
+    rf_heap_cpair_kernel<true, 1, false, 4, 128>
+        <<<dim3(b128), dim3(128), l128, stream>>>(
+            reinterpret_cast<const i32x2*>(cpairs), roots, leaf_values,
+            n_trees, depth, X, out, S, F, C);
+    return hipGetLastError();
+  }
   const size_t lds = stage ? (size_t)THREADS * SILP * (F + 1) * 4 : 0;
   if (lds > 160 * 1024) return hipErrorInvalidValue;
   unsigned blocks = lpx_ceil_div(S, THREADS * SILP);
@@ -949,6 +963,17 @@ hipError_t rf_predict_heap_cpair_lv16_f32(const int* cpairs, const int* roots,
   const int TILP = (silp == 21) ? 2 : (silp == 41 ? 4 : (silp == 81 ? 8 : 1));
   const int SILP = (silp == 2) ? 2 : (silp == 4 ? 4 : 1);
   const bool stage = F <= F_LDS_MAX;
+  if (silp == 141 && stage) {  // 128-thread TILP4 (see cpair_f32 launcher)
+    const size_t l128 = (size_t)128 * (F + 1) * 4;
+    unsigned b128 = lpx_ceil_div(S, 128);
+    if (b128 > 8192u) b128 = 8192u;
+    rf_heap_cpair_kernel<true, 1, true, 4, 128>
+        <<<dim3(b128), dim3(128), l128, stream>>>(
+            reinterpret_cast<const i32x2*>(cpairs), roots,
+            reinterpret_cast<const float*>(lv16), n_trees, depth, X, out, S,
+            F, C);
+    return hipGetLastError();
+  }
   const size_t lds = stage ? (size_t)THREADS * SILP * (F + 1) * 4 : 0;
   if (lds > 160 * 1024) return hipErrorInvalidValue;
   unsigned blocks = lpx_ceil_div(S, THREADS * SILP);

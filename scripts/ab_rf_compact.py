@@ -54,7 +54,7 @@ def main():
     ref = e.rf_predict_heap_pair(pairs, roots, lv, depth, X, 1)
     lv16 = lv.to(torch.float16).contiguous()
     lv16_exact = bool(lv16.to(torch.float32).eq(lv).all().item())
-    for silp in (41, 81):
+    for silp in (41, 81, 141):
         dt = bench(lambda: e.rf_predict_heap_pair(pairs, roots, lv, depth, X, silp))
         print(f"pair    silp={silp}: {S/dt/1e6:7.1f} Ms/s")
         dt = bench(lambda: e.rf_predict_heap_cpair(cpairs, croots, lv, depth, X, silp))
