@@ -62,8 +62,15 @@ def test_rccl_version_probe():
 
 
 def test_comms_single_gpu_roundtrip_and_rc_check():
-    r = subprocess.run([sys.executable, "-c", _ROUNDTRIP],
-                       capture_output=True, text=True, timeout=300)
+    # one retry: ncclCommInitAll can exceed the 120-s init watchdog on a
+    # box under memory/system pressure (observed once right after a
+    # rocprofv3 session where the WHOLE suite ran 6x slower) — a second
+    # attempt distinguishes that from a real init hang
+    for attempt in (0, 1):
+        r = subprocess.run([sys.executable, "-c", _ROUNDTRIP],
+                           capture_output=True, text=True, timeout=300)
+        if r.returncode == 0 or r.returncode != 124:
+            break
     assert r.returncode == 0, f"comms subprocess failed:\n{r.stdout[-2000:]}\n{r.stderr[-2000:]}"
     assert "COMMS_OK" in r.stdout
 
