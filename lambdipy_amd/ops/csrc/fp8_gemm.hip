@@ -792,7 +792,10 @@ __global__ __launch_bounds__(THREADS2, 2) void conv_fp8_impx_kernel(
   // the slot whose reads finished at the PREVIOUS step's barrier, so each
   // step needs ONE barrier (after the writes) instead of two — the
   // write-after-read stall between the MFMA block and the LDS fill
-  // disappears. 80 KiB still fits 2 blocks/CU.
+  // disappears.  NOTE: at 228 VGPRs this kernel runs 2 waves/SIMD
+  // (register-limited — the 80-KiB LDS would have allowed 2 blocks/CU);
+  // conv_fp8_impx2 below fixes that and is the default (LPX_CONV_V2=0
+  // selects this one).
   extern __shared__ __attribute__((aligned(16))) char smem[];
   const int SLOT = 2 * BM2 * ROWB2;
 
